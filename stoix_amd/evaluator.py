@@ -1,0 +1,110 @@
+"""Evaluator (parity with /root/reference/stoix/evaluator.py).
+
+The reference evaluates by vmapping one-episode while_loops
+(evaluator.py:87-206) and pmapping over devices; here an eval env is itself
+batched (one env slot per episode), so evaluation is: reset all episode
+slots, step the batch with an active mask until every slot has finished its
+FIRST episode, aggregating return/length at each slot's first done
+(evaluator.py:146-152 semantics). Greedy vs sampled acting mirrors
+``get_distribution_act_fn`` (evaluator.py:48-67).
+"""
+from __future__ import annotations
+
+import time
+from typing import Callable, Dict, Optional
+
+import torch
+
+from stoix_amd.envs.env import StatefulVecEnv
+from stoix_amd.types import EvalOutput
+
+
+@torch.no_grad()
+def evaluate(
+    act_fn: Callable,
+    eval_env: StatefulVecEnv,
+    greedy: bool = False,
+    max_steps: Optional[int] = None,
+    rnn: bool = False,
+) -> Dict[str, torch.Tensor]:
+    """Run one episode per env slot; returns metrics over all slots.
+
+    act_fn(obs, greedy) -> action  (or act_fn(obs, resets, state, greedy) ->
+    (action, state) when rnn=True).
+    """
+    t0 = time.perf_counter()
+    ts = eval_env.reset()
+    B = eval_env.num_envs
+    device = eval_env.device
+    ep_return = torch.zeros(B, device=device)
+    ep_length = torch.zeros(B, device=device)
+    finished = torch.zeros(B, dtype=torch.bool, device=device)
+    state = None
+    if rnn:
+        state = act_fn.initial_state(B, device)
+    steps = 0
+    limit = max_steps or (eval_env.max_episode_steps + 1)
+    resets = torch.zeros(B, dtype=torch.bool, device=device)
+    while not bool(finished.all()) and steps < limit:
+        if rnn:
+            action, state = act_fn(ts.observation, resets, state, greedy)
+        else:
+            action = act_fn(ts.observation, greedy)
+        ts = eval_env.step(action)
+        active = ~finished
+        ep_return = ep_return + ts.reward * active
+        ep_length = ep_length + active.to(ep_length.dtype)
+        finished = finished | ts.last()
+        resets = ts.last()
+        steps += 1
+    elapsed = time.perf_counter() - t0
+    metrics = {
+        "episode_return": ep_return,
+        "episode_length": ep_length,
+        "steps_per_second": torch.tensor(float(ep_length.sum()) / max(elapsed, 1e-9)),
+    }
+    thr = eval_env.solved_return_threshold
+    if thr is not None:
+        metrics["solve_rate"] = (ep_return >= thr).float()
+    return metrics
+
+
+def make_act_fn(actor: torch.nn.Module, generator: Optional[torch.Generator] = None):
+    """Greedy-mode() vs sample acting (reference evaluator.py:48-67)."""
+
+    @torch.no_grad()
+    def act(obs, greedy: bool):
+        dist = actor(obs)
+        if isinstance(dist, torch.Tensor):  # deterministic head
+            return dist
+        return dist.mode() if greedy else dist.sample(generator)
+
+    return act
+
+
+def make_recurrent_act_fn(actor, generator: Optional[torch.Generator] = None):
+    @torch.no_grad()
+    def act(obs, resets, state, greedy: bool):
+        dist, state = actor(obs.unsqueeze(0), resets.unsqueeze(0), state)
+        a = dist.mode() if greedy else dist.sample(generator)
+        return a.squeeze(0), state
+
+    act.initial_state = actor.initial_state
+    return act
+
+
+def evaluator_setup(eval_env: StatefulVecEnv, config):
+    """Build (eval_fn, absolute_eval_fn): the absolute metric runs 10x the
+    episodes on the best params (reference evaluator.py:347-416)."""
+    greedy = bool(getattr(config.arch, "evaluation_greedy", False))
+
+    def eval_fn(act_fn, rnn: bool = False):
+        return evaluate(act_fn, eval_env, greedy=greedy, rnn=rnn)
+
+    def absolute_eval_fn(act_fn, rnn: bool = False):
+        outs = []
+        for _ in range(10):
+            outs.append(evaluate(act_fn, eval_env, greedy=greedy, rnn=rnn))
+        return {k: torch.cat([o[k].reshape(-1) for o in outs]) for k in outs[0]}
+
+    return eval_fn, absolute_eval_fn

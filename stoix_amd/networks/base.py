@@ -1,0 +1,164 @@
+"""Actor/critic containers and recurrent networks.
+
+Parity with /root/reference/stoix/networks/base.py: FeedForwardActor/Critic
+:18-59, CompositeNetwork :62-86, MultiNetwork :88-121, ScannedRNN :124-159,
+RecurrentActor/Critic :162-222, chained_torsos :225-252.
+"""
+from __future__ import annotations
+
+from typing import Any, List, Optional, Sequence, Tuple
+
+import torch
+import torch.nn as nn
+
+from stoix_amd.networks.layers import StackedRNN
+
+Tensor = torch.Tensor
+
+
+class FeedForwardActor(nn.Module):
+    """input-embed -> torso -> action head (returns a Distribution)."""
+
+    def __init__(self, torso: nn.Module, action_head: nn.Module, input_layer: Optional[nn.Module] = None):
+        super().__init__()
+        self.input_layer = input_layer
+        self.torso = torso
+        self.action_head = action_head
+
+    def forward(self, obs: Any):
+        x = self.input_layer(obs) if self.input_layer is not None else obs
+        return self.action_head(self.torso(x))
+
+
+class FeedForwardCritic(nn.Module):
+    """input-embed -> torso -> critic head (scalar / distributional)."""
+
+    def __init__(self, torso: nn.Module, critic_head: nn.Module, input_layer: Optional[nn.Module] = None):
+        super().__init__()
+        self.input_layer = input_layer
+        self.torso = torso
+        self.critic_head = critic_head
+
+    def forward(self, *inputs: Any):
+        if self.input_layer is not None:
+            x = self.input_layer(*inputs)
+        else:
+            x = inputs[0] if len(inputs) == 1 else torch.cat([i.flatten(1) for i in inputs], dim=-1)
+        return self.critic_head(self.torso(x))
+
+
+class CompositeNetwork(nn.Module):
+    """Arbitrary input-layer/torso/head composition (reference base.py:62-86)."""
+
+    def __init__(self, input_layer: nn.Module, torso: nn.Module, head: nn.Module):
+        super().__init__()
+        self.input_layer = input_layer
+        self.torso = torso
+        self.head = head
+
+    def forward(self, *inputs: Any):
+        return self.head(self.torso(self.input_layer(*inputs)))
+
+
+class MultiNetwork(nn.Module):
+    """N copies of a network evaluated on the same input, outputs stacked on
+    a leading 'ensemble' dim (twin critics; reference base.py:88-121)."""
+
+    def __init__(self, networks: Sequence[nn.Module]):
+        super().__init__()
+        self.networks = nn.ModuleList(networks)
+
+    def forward(self, *inputs: Any) -> Tensor:
+        outs = [net(*inputs) for net in self.networks]
+        return torch.stack(outs, dim=0)
+
+
+class ScannedRNN(nn.Module):
+    """RNN scanned over time with per-step hidden reset on done
+    (reference base.py:124-159). Input is time-major [T, B, D]; ``resets``
+    [T, B] bool re-initialises the hidden state *before* consuming step t."""
+
+    def __init__(self, input_dim: int, hidden_dim: int, cell_type: str = "gru", num_layers: int = 1):
+        super().__init__()
+        self.rnn = StackedRNN(input_dim, hidden_dim, num_layers=num_layers, cell_type=cell_type)
+        self.hidden_dim = hidden_dim
+
+    def initial_state(self, batch: int, device) -> list:
+        return self.rnn.initial_state(batch, device)
+
+    @staticmethod
+    def _mask_state(state: Any, keep: Tensor) -> Any:
+        # keep: [B] float (1 keep, 0 reset-to-zero)
+        if isinstance(state, tuple):
+            return tuple(ScannedRNN._mask_state(s, keep) for s in state)
+        if isinstance(state, list):
+            return [ScannedRNN._mask_state(s, keep) for s in state]
+        return state * keep.unsqueeze(-1)
+
+    def forward(self, x: Tensor, resets: Tensor, state: list) -> Tuple[Tensor, list]:
+        T = x.shape[0]
+        outs = []
+        for t in range(T):
+            keep = (~resets[t].bool()).to(x.dtype)
+            state = self._mask_state(state, keep)
+            h, state = self.rnn(x[t], state)
+            outs.append(h)
+        return torch.stack(outs), state
+
+
+class RecurrentActor(nn.Module):
+    """pre-torso -> ScannedRNN -> post-torso -> head (reference base.py:162-192)."""
+
+    def __init__(self, pre_torso: nn.Module, rnn: ScannedRNN, post_torso: nn.Module, action_head: nn.Module):
+        super().__init__()
+        self.pre_torso = pre_torso
+        self.rnn = rnn
+        self.post_torso = post_torso
+        self.action_head = action_head
+
+    def initial_state(self, batch: int, device) -> list:
+        return self.rnn.initial_state(batch, device)
+
+    def forward(self, obs: Tensor, resets: Tensor, state: list):
+        # obs [T, B, D]; flatten time for the MLPs, keep for the RNN
+        T, B = obs.shape[:2]
+        z = self.pre_torso(obs.reshape(T * B, -1)).reshape(T, B, -1)
+        h, state = self.rnn(z, resets, state)
+        y = self.post_torso(h.reshape(T * B, -1))
+        dist = self.action_head(y)
+        return dist, state
+
+
+class RecurrentCritic(nn.Module):
+    def __init__(self, pre_torso: nn.Module, rnn: ScannedRNN, post_torso: nn.Module, critic_head: nn.Module):
+        super().__init__()
+        self.pre_torso = pre_torso
+        self.rnn = rnn
+        self.post_torso = post_torso
+        self.critic_head = critic_head
+
+    def initial_state(self, batch: int, device) -> list:
+        return self.rnn.initial_state(batch, device)
+
+    def forward(self, obs: Tensor, resets: Tensor, state: list):
+        T, B = obs.shape[:2]
+        z = self.pre_torso(obs.reshape(T * B, -1)).reshape(T, B, -1)
+        h, state = self.rnn(z, resets, state)
+        v = self.critic_head(self.post_torso(h.reshape(T * B, -1)))
+        if isinstance(v, torch.Tensor):
+            v = v.reshape(T, B)
+        return v, state
+
+
+class ChainedTorsos(nn.Module):
+    """Sequential composition of torsos (reference base.py:225-252)."""
+
+    def __init__(self, torsos: Sequence[nn.Module]):
+        super().__init__()
+        self.torsos = nn.ModuleList(torsos)
+        self.output_dim = getattr(torsos[-1], "output_dim", None)
+
+    def forward(self, x: Tensor) -> Tensor:
+        for t in self.torsos:
+            x = t(x)
+        return x

@@ -1,0 +1,110 @@
+"""Anakin experiment runner — the shared host loop.
+
+Structure parity with the reference's per-system ``run_experiment``
+(/root/reference/stoix/systems/ppo/anakin/ff_ppo.py:554-706): the only
+Python loop in training is over evaluation intervals; each interval runs
+``num_updates_per_eval`` fused update steps on-device, then evaluates, logs
+throughput (identical ``steps_per_second`` definition, ff_ppo.py:589-595),
+checkpoints, and tracks best params for the absolute metric
+(ff_ppo.py:673-699).
+
+Learner objects implement:
+  * ``update_step() -> Dict[str, Tensor]``  (one rollout + update, on-device)
+  * ``act_fn`` / optional ``rnn_act`` for evaluation
+  * ``state_for_checkpoint()``
+Multi-GPU: one process per GPU (torchrun), RCCL all-reduce inside the
+learner's optimiser step; rank 0 logs/evaluates.
+"""
+from __future__ import annotations
+
+import time
+from typing import Callable, Dict, Optional
+
+import torch
+
+from stoix_amd import envs as environments
+from stoix_amd.evaluator import evaluator_setup
+from stoix_amd.parallel.dist import get_dist_context
+from stoix_amd.utils.checkpointing import Checkpointer
+from stoix_amd.utils.logger import LogEvent, StoixLogger
+from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+
+def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = False) -> float:
+    ctx = get_dist_context(force_cpu=force_cpu)
+    device = ctx.device
+    config.arch.n_devices = ctx.world_size
+    check_total_timesteps(config)
+    torch.manual_seed(int(config.arch.seed) + ctx.rank)
+
+    # update_batch_size folds exactly into the env batch: the reference's
+    # vmapped independent learners stay bit-identical under per-minibatch
+    # pmean, so N replicas == one learner with N x batch.
+    num_envs = int(config.arch.num_envs) * int(getattr(config.arch, "update_batch_size", 1))
+    train_env = environments.make_single(config, num_envs, device, int(config.arch.seed) + 31 * ctx.rank)
+    eval_env = environments.make_single(
+        config, int(config.arch.num_eval_episodes), device, int(config.arch.seed) + 10_000
+    )
+
+    learner = learner_factory(config, train_env, device)
+
+    eval_fn, absolute_eval_fn = evaluator_setup(eval_env, config)
+    logger = StoixLogger(config) if ctx.is_main else None
+    checkpointer = None
+    if ctx.is_main and config.logger.checkpointing.save_model:
+        checkpointer = Checkpointer(
+            model_name=config.system.system_name,
+            metadata=config.to_plain() if hasattr(config, "to_plain") else dict(config),
+            directory=f"{logger.directory}/checkpoints",
+            max_to_keep=config.logger.checkpointing.save_args.max_to_keep,
+        )
+
+    num_updates_per_eval = int(config.arch.num_updates_per_eval)
+    steps_per_update = (
+        int(config.system.rollout_length) * num_envs * ctx.world_size
+    )
+    steps_per_eval_interval = steps_per_update * num_updates_per_eval
+
+    best_return = float("-inf")
+    best_state = None
+    final_return = 0.0
+    t_env = 0
+    rnn = bool(getattr(learner, "is_recurrent", False))
+
+    for eval_idx in range(int(config.arch.num_evaluation)):
+        t0 = time.perf_counter()
+        train_metrics: Dict[str, torch.Tensor] = {}
+        for _ in range(num_updates_per_eval):
+            train_metrics = learner.update_step()
+        if device.type == "cuda":
+            torch.cuda.synchronize(device)
+        elapsed = time.perf_counter() - t0
+        t_env += steps_per_eval_interval
+        sps = steps_per_eval_interval / elapsed
+
+        if ctx.is_main:
+            logger.log({"steps_per_second": sps, **train_metrics}, t_env, eval_idx, LogEvent.TRAIN)
+            ep_metrics = getattr(learner, "episode_metrics", None)
+            if ep_metrics:
+                logger.log(ep_metrics, t_env, eval_idx, LogEvent.ACT)
+
+            eval_metrics = eval_fn(learner.act_fn, rnn=rnn)
+            mean_return = float(eval_metrics["episode_return"].mean())
+            final_return = mean_return
+            logger.log(eval_metrics, t_env, eval_idx, LogEvent.EVAL)
+
+            if checkpointer is not None:
+                checkpointer.save(t_env, learner.state_for_checkpoint(), mean_return)
+            if mean_return >= best_return:
+                best_return = mean_return
+                best_state = learner.snapshot_params()
+
+    if ctx.is_main and bool(getattr(config.arch, "absolute_metric", True)) and best_state is not None:
+        learner.load_params(best_state)
+        abs_metrics = absolute_eval_fn(learner.act_fn, rnn=rnn)
+        logger.log(abs_metrics, t_env, int(config.arch.num_evaluation), LogEvent.ABSOLUTE)
+        final_return = float(abs_metrics["episode_return"].mean())
+
+    if ctx.is_main and logger is not None:
+        logger.close()
+    return final_return

@@ -1,0 +1,219 @@
+"""Anakin PPO (feed-forward) — discrete and continuous.
+
+Algorithm parity with /root/reference/stoix/systems/ppo/anakin/ff_ppo.py and
+ff_ppo_continuous.py (the head is chosen by the network config): rollout of
+``rollout_length`` steps storing PPOTransition with ``bootstrap_value =
+V(extras["next_obs"])`` (ff_ppo.py:113-116), truncation-aware GAE
+(ff_ppo.py:164-179), epochs x shuffled minibatches of clip-loss + entropy
+and clipped value loss (ff_ppo.py:191-235, 296-336), separate Adam chains
+with global-norm clip (ff_ppo.py:456-463), optional Welford observation
+normalisation (ff_ppo.py:90-162).
+
+MI355X design: rollout storage is preallocated [T, B, ...] device tensors
+(struct-of-arrays); the rollout and update loops are static-shaped and
+capturable into hip graphs (stoix_amd/ops/graph.py); gradients are averaged
+across GPUs with one fused flat RCCL all-reduce per minibatch
+(stoix_amd.parallel.FlatGradReducer).
+"""
+from __future__ import annotations
+
+import sys
+from typing import Dict, Optional
+
+import torch
+import torch.nn as nn
+
+from stoix_amd.config import compose
+from stoix_amd.envs.env import StatefulVecEnv, get_final_step_metrics
+from stoix_amd.networks.factory import build_actor, build_critic
+from stoix_amd.ops import multistep
+from stoix_amd.ops import running_statistics as rs
+from stoix_amd.ops.losses import clipped_value_loss, ppo_clip_loss
+from stoix_amd.parallel.dist import FlatGradReducer, broadcast_module
+from stoix_amd.systems.anakin import run_anakin_experiment
+
+Tensor = torch.Tensor
+
+
+class PPOLearner:
+    def __init__(self, config, env: StatefulVecEnv, device: torch.device):
+        self.cfg = config
+        self.sys = config.system
+        self.env = env
+        self.device = device
+        self.T = int(self.sys.rollout_length)
+        self.B = env.num_envs
+
+        obs_space, act_space = env.observation_space, env.action_space
+        self.actor = build_actor(config.network.actor_network, obs_space, act_space).to(device)
+        self.critic = build_critic(config.network.critic_network, obs_space).to(device)
+        broadcast_module(self.actor)
+        broadcast_module(self.critic)
+
+        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr), eps=1e-5)
+        self.critic_opt = torch.optim.Adam(self.critic.parameters(), lr=float(self.sys.critic_lr), eps=1e-5)
+        self.reducer = FlatGradReducer(
+            list(self.actor.parameters()) + list(self.critic.parameters()), device
+        )
+        self.gen = torch.Generator(device=device)
+        self.gen.manual_seed(int(config.arch.seed) * 7919 + 13)
+
+        self.normalize_obs = bool(getattr(self.sys, "normalize_observations", False))
+        if self.normalize_obs:
+            self.obs_stats = rs.init_state(obs_space.shape, device=device)
+
+        self.ts = env.reset()
+        self.episode_metrics: Dict[str, float] = {}
+        self._discrete = not hasattr(act_space, "shape") or len(act_space.shape) == 0
+
+        # preallocated rollout storage (struct of arrays, time-major)
+        obs_shape = obs_space.shape
+        act_shape = () if self._discrete else act_space.shape
+        act_dtype = torch.long if self._discrete else torch.float32
+        z = lambda *s, dtype=torch.float32: torch.zeros(*s, dtype=dtype, device=device)
+        self.buf_obs = z(self.T, self.B, *obs_shape)
+        self.buf_action = z(self.T, self.B, *act_shape, dtype=act_dtype)
+        self.buf_log_prob = z(self.T, self.B)
+        self.buf_value = z(self.T, self.B)
+        self.buf_bootstrap = z(self.T, self.B)
+        self.buf_reward = z(self.T, self.B)
+        self.buf_discount = z(self.T, self.B)
+        self.buf_truncated = z(self.T, self.B, dtype=torch.bool)
+
+    # ---------------------------------------------------------------- acting
+
+    @torch.no_grad()
+    def act_fn(self, obs: Tensor, greedy: bool) -> Tensor:
+        if self.normalize_obs:
+            obs = rs.normalize(obs, self.obs_stats)
+        dist = self.actor(obs)
+        return dist.mode() if greedy else dist.sample(self.gen)
+
+    # --------------------------------------------------------------- rollout
+
+    @torch.no_grad()
+    def _rollout(self) -> None:
+        ts = self.ts
+        for t in range(self.T):
+            obs = ts.observation
+            if self.normalize_obs:
+                obs = rs.normalize(obs, self.obs_stats)
+            dist = self.actor(obs)
+            value = self.critic(obs)
+            action = dist.sample(self.gen)
+            log_prob = dist.log_prob(action)
+            next_ts = self.env.step(action)
+            next_obs = next_ts.extras["next_obs"]
+            if self.normalize_obs:
+                next_obs = rs.normalize(next_obs, self.obs_stats)
+            bootstrap_value = self.critic(next_obs)
+
+            self.buf_obs[t] = obs
+            self.buf_action[t] = action
+            self.buf_log_prob[t] = log_prob
+            self.buf_value[t] = value
+            self.buf_bootstrap[t] = bootstrap_value
+            self.buf_reward[t] = next_ts.reward
+            self.buf_discount[t] = next_ts.discount
+            self.buf_truncated[t] = next_ts.truncated()
+            ts = next_ts
+        self.ts = ts
+        if self.normalize_obs:
+            self.obs_stats = rs.update(self.obs_stats, self.buf_obs, all_reduce=True)
+        em = ts.extras["episode_metrics"]
+        final, has = get_final_step_metrics(em)
+        if has:
+            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+
+    # ---------------------------------------------------------------- update
+
+    def update_step(self) -> Dict[str, Tensor]:
+        self._rollout()
+        gamma = float(self.sys.gamma)
+        adv, targets = multistep.batch_truncated_generalized_advantage_estimation(
+            self.buf_reward,
+            gamma * self.buf_discount,
+            float(self.sys.gae_lambda),
+            self.buf_value,
+            self.buf_bootstrap,
+            truncation_t=self.buf_truncated,
+            standardize_advantages=bool(self.sys.standardize_advantages),
+        )
+
+        TB = self.T * self.B
+        flat_obs = self.buf_obs.reshape(TB, *self.buf_obs.shape[2:])
+        flat_action = self.buf_action.reshape(TB, *self.buf_action.shape[2:])
+        flat_logp = self.buf_log_prob.reshape(TB)
+        flat_value = self.buf_value.reshape(TB)
+        flat_adv = adv.reshape(TB)
+        flat_targets = targets.reshape(TB)
+
+        n_mb = int(self.sys.num_minibatches)
+        mb_size = TB // n_mb
+        metrics: Dict[str, Tensor] = {}
+        for _ in range(int(self.sys.epochs)):
+            perm = torch.randperm(TB, device=self.device, generator=self.gen)
+            for mb in range(n_mb):
+                idx = perm[mb * mb_size : (mb + 1) * mb_size]
+                obs_mb = flat_obs[idx]
+                dist = self.actor(obs_mb)
+                new_logp = dist.log_prob(flat_action[idx])
+                entropy = dist.entropy().mean()
+                a_loss = ppo_clip_loss(new_logp, flat_logp[idx], flat_adv[idx], float(self.sys.clip_eps))
+                actor_loss = a_loss - float(self.sys.ent_coef) * entropy
+
+                value = self.critic(obs_mb)
+                v_loss = clipped_value_loss(value, flat_value[idx], flat_targets[idx], float(self.sys.clip_eps))
+                critic_loss = float(self.sys.vf_coef) * v_loss
+
+                self.actor_opt.zero_grad(set_to_none=True)
+                self.critic_opt.zero_grad(set_to_none=True)
+                (actor_loss + critic_loss).backward()
+                self.reducer.reduce()
+                self.reducer.wait()
+                nn.utils.clip_grad_norm_(self.actor.parameters(), float(self.sys.max_grad_norm))
+                nn.utils.clip_grad_norm_(self.critic.parameters(), float(self.sys.max_grad_norm))
+                self.actor_opt.step()
+                self.critic_opt.step()
+                metrics = {
+                    "actor_loss": a_loss.detach(),
+                    "value_loss": v_loss.detach(),
+                    "entropy": entropy.detach(),
+                }
+        return metrics
+
+    # ------------------------------------------------------------ checkpoint
+
+    def state_for_checkpoint(self):
+        return {
+            "actor": {k: v for k, v in self.actor.state_dict().items()},
+            "critic": {k: v for k, v in self.critic.state_dict().items()},
+        }
+
+    def snapshot_params(self):
+        return {
+            "actor": {k: v.clone() for k, v in self.actor.state_dict().items()},
+            "critic": {k: v.clone() for k, v in self.critic.state_dict().items()},
+        }
+
+    def load_params(self, snap) -> None:
+        self.actor.load_state_dict(snap["actor"])
+        self.critic.load_state_dict(snap["critic"])
+
+
+def learner_factory(config, env, device) -> PPOLearner:
+    return PPOLearner(config, env, device)
+
+
+def run(config) -> float:
+    return run_anakin_experiment(config, learner_factory)
+
+
+def hydra_entry_point(argv=None) -> float:
+    overrides = argv if argv is not None else sys.argv[1:]
+    cfg = compose("default/anakin/default_ff_ppo.yaml", overrides)
+    return run(cfg)
+
+
+if __name__ == "__main__":
+    hydra_entry_point()

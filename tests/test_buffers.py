@@ -1,0 +1,79 @@
+"""Replay buffer tests: item, trajectory, prioritised sum-tree."""
+import torch
+
+from stoix_amd.buffers import ItemBuffer, PrioritisedBuffer, TrajectoryBuffer
+from stoix_amd.buffers.per import SumTree
+
+
+def test_item_buffer_roundtrip():
+    buf = ItemBuffer(capacity=100, seed=0)
+    for i in range(5):
+        buf.add({"x": torch.full((10, 3), float(i)), "a": torch.full((10,), i, dtype=torch.long)})
+    assert buf.size == 50
+    s = buf.sample(32)
+    assert s["x"].shape == (32, 3)
+    # field alignment preserved
+    assert (s["x"][:, 0].long() == s["a"]).all()
+
+
+def test_item_buffer_wraps():
+    buf = ItemBuffer(capacity=16, seed=0)
+    for i in range(4):
+        buf.add({"x": torch.full((8,), float(i))})
+    assert buf.size == 16
+    s = buf.sample(64)
+    # oldest items (0) have been overwritten
+    assert (s["x"] >= 2.0).all()
+
+
+def test_trajectory_buffer_windows_contiguous():
+    buf = TrajectoryBuffer(add_batch_size=4, max_length_time_axis=64, sample_sequence_length=8, seed=0)
+    t = 0
+    for _ in range(6):
+        block = torch.arange(t, t + 8, dtype=torch.float32).unsqueeze(0).expand(4, -1)
+        buf.add({"t": block})
+        t += 8
+    s = buf.sample(16)
+    seq = s["t"]
+    assert seq.shape == (16, 8)
+    diffs = seq[:, 1:] - seq[:, :-1]
+    assert (diffs == 1).all(), "sampled windows must be time-contiguous"
+
+
+def test_sum_tree_proportional():
+    tree = SumTree(8)
+    idx = torch.arange(8)
+    prio = torch.tensor([0.0, 0.0, 1.0, 0.0, 3.0, 0.0, 0.0, 0.0])
+    tree.set(idx, prio)
+    assert abs(tree.total.item() - 4.0) < 1e-6
+    g = torch.Generator().manual_seed(0)
+    samples = tree.sample(4000, g)
+    counts = torch.bincount(samples, minlength=8).float()
+    assert counts[2] > 0 and counts[4] > 0
+    assert counts[[0, 1, 3, 5, 6, 7]].sum() == 0
+    ratio = counts[4] / counts[2]
+    assert 2.0 < ratio < 4.5  # expect ~3
+
+
+def test_sum_tree_update_changes_distribution():
+    tree = SumTree(16)
+    tree.set(torch.arange(16), torch.ones(16))
+    tree.set(torch.tensor([5]), torch.tensor([100.0]))
+    g = torch.Generator().manual_seed(1)
+    s = tree.sample(1000, g)
+    assert (s == 5).float().mean() > 0.7
+
+
+def test_prioritised_buffer_sample_and_writeback():
+    buf = PrioritisedBuffer(
+        add_batch_size=2, max_length_time_axis=32, sample_sequence_length=4, seed=0
+    )
+    for i in range(4):
+        buf.add({"x": torch.randn(2, 8, 3)})
+    assert buf.can_sample
+    s = buf.sample(8, importance_sampling_exponent=0.4)
+    assert s["x"].shape == (8, 4, 3)
+    assert (s["_weights"] <= 1.0 + 1e-6).all()
+    buf.set_priorities(s["_slots"], torch.rand(8) * 5)
+    s2 = buf.sample(8)
+    assert s2["x"].shape == (8, 4, 3)

@@ -1,0 +1,163 @@
+"""Environment-contract tests: the load-bearing autoreset/truncation
+semantics of SURVEY.md §8.7, plus per-env sanity."""
+import torch
+
+from stoix_amd.envs.ant import Ant
+from stoix_amd.envs.classic import Acrobot, CartPole, MountainCar, Pendulum
+from stoix_amd.envs.debug import DEBUG_ENVIRONMENTS
+from stoix_amd.envs.env import get_final_step_metrics
+from stoix_amd.types import StepType
+
+
+def rollout_random(env, steps=50):
+    ts = env.reset()
+    all_ts = [ts]
+    for _ in range(steps):
+        a = env.action_space.sample(env.num_envs, env.device, env.gen)
+        ts = env.step(a)
+        all_ts.append(ts)
+    return all_ts
+
+
+def test_reset_contract():
+    env = CartPole(8, seed=0)
+    ts = env.reset()
+    assert (ts.step_type == StepType.FIRST).all()
+    assert (ts.discount == 1.0).all()
+    assert ts.observation.shape == (8, 4)
+    assert "next_obs" in ts.extras and "episode_metrics" in ts.extras
+
+
+def test_termination_sets_discount_zero_and_autoresets():
+    env = CartPole(16, seed=1)
+    ts = env.reset()
+    saw_done = False
+    for _ in range(200):
+        a = env.action_space.sample(16, env.device, env.gen)
+        ts = env.step(a)
+        done = ts.done()
+        if done.any():
+            saw_done = True
+            # terminated -> discount 0 and step_type TERMINATED
+            assert (ts.step_type[done] == StepType.TERMINATED).all()
+            # autoreset: returned obs is a FRESH state (inside reset range),
+            # true final obs in extras
+            assert (ts.observation[done].abs() <= 0.05 + 1e-6).all()
+            next_obs = ts.extras["next_obs"][done]
+            # the true final obs must violate a termination bound
+            assert (
+                (next_obs[:, 0].abs() > 2.4) | (next_obs[:, 2].abs() > CartPole.THETA_LIMIT)
+            ).all()
+    assert saw_done, "random cartpole should terminate within 200 steps"
+
+
+def test_truncation_sets_truncated_steptype_discount_one():
+    env = Pendulum(4, seed=0)  # never terminates; truncates at 200
+    env.max_episode_steps = 10
+    ts = env.reset()
+    for i in range(10):
+        ts = env.step(torch.zeros(4, 1))
+    assert (ts.step_type == StepType.TRUNCATED).all()
+    assert (ts.discount == 1.0).all()
+    assert (ts.done() == False).all()  # noqa: E712  done means discount==0
+
+
+def test_episode_metrics_latch_completed_episode():
+    env = CartPole(4, seed=2)
+    env.max_episode_steps = 5
+    ts = env.reset()
+    for i in range(5):
+        ts = env.step(torch.zeros(4, dtype=torch.long))
+    em = ts.extras["episode_metrics"]
+    final, has = get_final_step_metrics(em)
+    assert has
+    # episodes truncated at length 5 with +1/step reward
+    done_mask = em["is_terminal_step"]
+    assert (em["episode_length"][done_mask] == 5).all()
+    assert (em["episode_return"][done_mask] == 5.0).all()
+
+
+def test_episode_metrics_reset_after_done():
+    env = CartPole(2, seed=3)
+    env.max_episode_steps = 3
+    env.reset()
+    for _ in range(3):
+        ts = env.step(torch.zeros(2, dtype=torch.long))
+    assert ts.extras["episode_metrics"]["is_terminal_step"].all()
+    # next episode: counters restarted
+    ts = env.step(torch.zeros(2, dtype=torch.long))
+    assert (~ts.extras["episode_metrics"]["is_terminal_step"]).all()
+    for _ in range(2):
+        ts = env.step(torch.zeros(2, dtype=torch.long))
+    em = ts.extras["episode_metrics"]
+    assert (em["episode_length"] == 3).all()
+
+
+def test_all_classic_envs_step():
+    for cls in [CartPole, Pendulum, MountainCar, Acrobot]:
+        env = cls(4, seed=0)
+        ts_list = rollout_random(env, 20)
+        obs = ts_list[-1].observation
+        assert torch.isfinite(obs).all(), cls.__name__
+        assert obs.shape[0] == 4
+
+
+def test_all_debug_envs_step():
+    for name, cls in DEBUG_ENVIRONMENTS.items():
+        env = cls(4, seed=0)
+        ts_list = rollout_random(env, 30)
+        assert torch.isfinite(ts_list[-1].observation).all(), name
+
+
+def test_identity_game_rewards():
+    env = DEBUG_ENVIRONMENTS["identity"](8, seed=0)
+    ts = env.reset()
+    # acting with the observed symbol always yields reward 1
+    for _ in range(5):
+        sym = ts.observation.argmax(-1)
+        ts = env.step(sym)
+        assert (ts.reward == 1.0).all()
+
+
+def test_exploration_chain_goal():
+    env = DEBUG_ENVIRONMENTS["exploration"](2, seed=0, chain_length=5)
+    ts = env.reset()
+    for _ in range(4):
+        ts = env.step(torch.ones(2, dtype=torch.long))
+    # reached the goal at step 4 -> terminal reward 1
+    assert (ts.reward == 1.0).all()
+    assert (ts.step_type == StepType.TERMINATED).all()
+
+
+def test_ant_steps_and_terminates_sensibly():
+    env = Ant(16, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (16, 27)
+    for _ in range(50):
+        a = env.action_space.sample(16, env.device, env.gen)
+        ts = env.step(a)
+        assert torch.isfinite(ts.observation).all()
+        assert torch.isfinite(ts.reward).all()
+    # torso stays in a plausible z band under physics (after autoresets)
+    z = ts.extras["next_obs"][:, 0]
+    assert (z > -1.0).all() and (z < 3.0).all()
+
+
+def test_ant_healthy_reward_present():
+    env = Ant(4, seed=1)
+    env.reset()
+    ts = env.step(torch.zeros(4, 8))
+    # near-static ant earns roughly the healthy bonus
+    assert (ts.reward > -5.0).all() and (ts.reward < 5.0).all()
+
+
+def test_determinism_same_seed():
+    e1 = CartPole(4, seed=7)
+    e2 = CartPole(4, seed=7)
+    ts1, ts2 = e1.reset(), e2.reset()
+    torch.testing.assert_close(ts1.observation, ts2.observation)
+    for _ in range(10):
+        a = torch.ones(4, dtype=torch.long)
+        ts1 = e1.step(a)
+        ts2 = e2.step(a)
+    torch.testing.assert_close(ts1.observation, ts2.observation)

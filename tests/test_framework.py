@@ -1,0 +1,167 @@
+"""Config composition, checkpointing, logger, networks, running stats."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from stoix_amd.config import compose
+from stoix_amd.networks.distributions import (
+    AffineTanhTransformedDistribution,
+    Categorical,
+    EpsilonGreedy,
+)
+from stoix_amd.networks.factory import build_actor, build_critic
+from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
+from stoix_amd.ops import running_statistics as rs
+from stoix_amd.utils.checkpointing import Checkpointer
+from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+
+def test_compose_defaults_and_overrides():
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/pendulum", "system.gamma=0.5", "arch.total_num_envs=64"],
+    )
+    assert cfg.env.scenario.name == "Pendulum-v1"
+    assert cfg.system.gamma == 0.5
+    assert cfg.arch.total_num_envs == 64
+    assert cfg.logger.loggers == ["console", "json"]
+
+
+def test_timestep_checker_derives_num_updates():
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["arch.total_num_envs=32", "arch.total_timesteps=40960", "system.rollout_length=16"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    assert cfg.arch.num_envs == 32
+    assert cfg.arch.num_updates == 40960 // (16 * 32)
+
+
+def test_timestep_checker_rejects_bad_divisibility():
+    cfg = compose("default/anakin/default_ff_ppo.yaml", ["arch.total_num_envs=7"])
+    cfg.arch.n_devices = 2
+    with pytest.raises(AssertionError):
+        check_total_timesteps(cfg)
+
+
+def test_checkpointer_roundtrip(tmp_path):
+    ckpt = Checkpointer("test_model", {"cfg": 1}, directory=str(tmp_path))
+    state = {"actor": {"w": torch.randn(3, 3)}, "step": torch.tensor(5)}
+    ckpt.save(100, state, metric_value=1.0)
+    template = {"actor": {"w": torch.zeros(3, 3)}, "step": torch.tensor(0)}
+    restored = ckpt.restore_params(template)
+    torch.testing.assert_close(restored["actor"]["w"], state["actor"]["w"])
+    assert restored["step"].item() == 5
+
+
+def test_checkpointer_best_retention(tmp_path):
+    ckpt = Checkpointer("m", directory=str(tmp_path), max_to_keep=1)
+    ckpt.save(1, {"w": torch.tensor([1.0])}, metric_value=5.0)
+    ckpt.save(2, {"w": torch.tensor([2.0])}, metric_value=3.0)  # worse
+    best = ckpt.restore_params({"w": torch.tensor([0.0])}, best=True)
+    assert best["w"].item() == 1.0
+    latest = ckpt.restore_params({"w": torch.tensor([0.0])})
+    assert latest["w"].item() == 2.0
+
+
+def test_checkpointer_version_gate(tmp_path):
+    import json
+
+    ckpt = Checkpointer("m", directory=str(tmp_path))
+    ckpt.save(1, {"w": torch.tensor([1.0])})
+    meta_path = os.path.join(str(tmp_path), "m", "metadata.json")
+    with open(meta_path) as f:
+        meta = json.load(f)
+    meta["checkpointer_version"] = "1.0"
+    with open(meta_path, "w") as f:
+        json.dump(meta, f)
+    with pytest.raises(ValueError):
+        ckpt.restore_params({"w": torch.tensor([0.0])})
+
+
+def test_running_statistics_welford_matches_numpy():
+    g = torch.Generator().manual_seed(0)
+    state = rs.init_state((5,))
+    chunks = [torch.randn(100, 5, generator=g) * 3 + 1 for _ in range(4)]
+    for c in chunks:
+        state = rs.update(state, c)
+    all_data = torch.cat(chunks)
+    torch.testing.assert_close(state.mean, all_data.mean(0), rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(state.std, all_data.std(0, unbiased=False), rtol=1e-3, atol=1e-4)
+    normed = rs.normalize(all_data, state)
+    assert abs(normed.mean().item()) < 1e-4
+
+
+def test_categorical_distribution():
+    logits = torch.tensor([[2.0, 0.0, -1.0]])
+    d = Categorical(logits)
+    assert d.mode().item() == 0
+    lp = d.log_prob(torch.tensor([0]))
+    probs = torch.softmax(logits, -1)
+    assert abs(lp.exp().item() - probs[0, 0].item()) < 1e-6
+    ent = d.entropy()
+    assert abs(ent.item() - (-(probs * probs.log()).sum()).item()) < 1e-5
+    g = torch.Generator().manual_seed(0)
+    samples = torch.stack([d.sample(g) for _ in range(2000)]).float()
+    assert abs((samples == 0).float().mean().item() - probs[0, 0].item()) < 0.05
+
+
+def test_tanh_normal_log_prob_consistency():
+    d = AffineTanhTransformedDistribution(
+        torch.zeros(4, 2), torch.ones(4, 2) * 0.5, -2.0, 2.0
+    )
+    a, lp = d.sample_and_log_prob(torch.Generator().manual_seed(1))
+    assert (a.abs() <= 2.0).all()
+    lp2 = d.log_prob(a)
+    torch.testing.assert_close(lp, lp2, rtol=1e-3, atol=1e-3)
+    assert d.mode().shape == (4, 2)
+
+
+def test_tanh_normal_matches_torch_reference():
+    """Cross-check log_prob against torch.distributions TransformedDistribution."""
+    loc = torch.tensor([[0.3, -0.7]])
+    scale = torch.tensor([[0.6, 1.2]])
+    mine = AffineTanhTransformedDistribution(loc, scale, -1.0, 1.0)
+    base = torch.distributions.Normal(loc, scale)
+    ref = torch.distributions.TransformedDistribution(
+        base, [torch.distributions.transforms.TanhTransform(cache_size=1)]
+    )
+    x = torch.tensor([[0.5, -0.3]])
+    torch.testing.assert_close(
+        mine.log_prob(x), ref.log_prob(x).sum(-1), rtol=1e-4, atol=1e-4
+    )
+
+
+def test_epsilon_greedy():
+    prefs = torch.tensor([[1.0, 5.0, 2.0]])
+    d = EpsilonGreedy(prefs, epsilon=0.3)
+    assert d.mode().item() == 1
+    g = torch.Generator().manual_seed(0)
+    samples = torch.stack([d.sample(g) for _ in range(3000)])
+    frac_greedy = (samples == 1).float().mean().item()
+    assert abs(frac_greedy - (0.7 + 0.1)) < 0.05
+
+
+def test_network_factory_builds_and_runs():
+    obs = BoxSpace((8,))
+    act_d = DiscreteSpace(4)
+    act_c = BoxSpace((3,), -1.0, 1.0)
+    net_cfg = {
+        "pre_torso": {"_target_": "MLPTorso", "layer_sizes": [32, 32]},
+        "action_head": {"_target_": "CategoricalHead"},
+    }
+    actor = build_actor(net_cfg, obs, act_d)
+    dist = actor(torch.randn(5, 8))
+    assert dist.sample().shape == (5,)
+    net_cfg["action_head"] = {"_target_": "NormalAffineTanhDistributionHead"}
+    actor_c = build_actor(net_cfg, obs, act_c)
+    dist = actor_c(torch.randn(5, 8))
+    assert dist.sample().shape == (5, 3)
+    critic = build_critic(
+        {"pre_torso": {"_target_": "MLPTorso", "layer_sizes": [32]}, "critic_head": {"_target_": "ScalarCriticHead"}},
+        obs,
+    )
+    assert critic(torch.randn(5, 8)).shape == (5,)

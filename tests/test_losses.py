@@ -1,0 +1,129 @@
+"""Loss-function tests: hand-computed values and invariants."""
+import math
+
+import pytest
+import torch
+
+from stoix_amd.ops import losses as L
+
+
+def test_ppo_clip_loss_values():
+    # single sample: log-ratio 0 -> ratio 1 -> loss = -A
+    lp = torch.tensor([0.0])
+    blp = torch.tensor([0.0])
+    adv = torch.tensor([2.0])
+    assert abs(L.ppo_clip_loss(lp, blp, adv, 0.2).item() + 2.0) < 1e-6
+    # big positive ratio with positive advantage clips at 1+eps
+    lp = torch.tensor([1.0])
+    out = L.ppo_clip_loss(lp, blp, adv, 0.2)
+    assert abs(out.item() + 1.2 * 2.0) < 1e-6
+    # big positive ratio with NEGATIVE advantage does NOT clip (pessimism)
+    out = L.ppo_clip_loss(lp, blp, -adv, 0.2)
+    assert abs(out.item() - math.exp(1.0) * 2.0) < 1e-4
+
+
+def test_clipped_value_loss():
+    pred = torch.tensor([2.0])
+    behav = torch.tensor([0.0])
+    tgt = torch.tensor([0.5])
+    # clipped pred = 0 + clip(2, -0.5, 0.5) = 0.5; err=(2-0.5)^2=2.25
+    # err_clipped=(0.5-0.5)^2=0 -> max = 2.25, 0.5*mean = 1.125
+    assert abs(L.clipped_value_loss(pred, behav, tgt, 0.5).item() - 1.125) < 1e-6
+
+
+def test_q_learning_loss():
+    q_tm1 = torch.tensor([[1.0, 2.0]])
+    a = torch.tensor([0])
+    r = torch.tensor([1.0])
+    d = torch.tensor([0.9])
+    q_t = torch.tensor([[3.0, 5.0]])
+    # target = 1 + .9*5 = 5.5; td = 5.5 - 1 = 4.5; mse/2 = 10.125
+    assert abs(L.q_learning(q_tm1, a, r, d, q_t).item() - 0.5 * 4.5**2) < 1e-5
+
+
+def test_double_q_learning_uses_selector_argmax():
+    q_tm1 = torch.tensor([[1.0, 2.0]])
+    a = torch.tensor([0])
+    r = torch.tensor([0.0])
+    d = torch.tensor([1.0])
+    q_t_value = torch.tensor([[10.0, 20.0]])
+    q_t_selector = torch.tensor([[5.0, 1.0]])  # argmax 0 -> value 10
+    out = L.double_q_learning(q_tm1, q_t_value, a, r, d, q_t_selector)
+    assert abs(out.item() - 0.5 * (10.0 - 1.0) ** 2) < 1e-5
+
+
+def test_categorical_l2_project_identity_and_shift():
+    z = torch.linspace(-1.0, 1.0, 5)
+    probs = torch.tensor([[0.1, 0.2, 0.4, 0.2, 0.1]])
+    # projecting onto itself is identity
+    out = L.categorical_l2_project(z.unsqueeze(0), probs, z)
+    torch.testing.assert_close(out, probs, rtol=1e-5, atol=1e-6)
+    # shift by half a bin splits mass between neighbours
+    out = L.categorical_l2_project(z.unsqueeze(0) + 0.25, probs, z)
+    assert abs(out.sum().item() - 1.0) < 1e-5
+    assert abs(out[0, 2].item() - (0.5 * 0.2 + 0.5 * 0.4)) < 1e-5
+    # clamping at the edges keeps mass
+    out = L.categorical_l2_project(z.unsqueeze(0) + 100.0, probs, z)
+    assert abs(out[0, -1].item() - 1.0) < 1e-5
+
+
+def test_categorical_double_q_learning_perfect_prediction():
+    # if predicted dist equals the projected target, CE = entropy(target)
+    atoms = torch.linspace(-1, 1, 11)
+    B, A, N = 3, 2, 11
+    g = torch.Generator().manual_seed(0)
+    logits = torch.randn(B, A, N, generator=g)
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.zeros(B)
+    d = torch.ones(B)
+    sel = torch.randn(B, A, generator=g)
+    loss = L.categorical_double_q_learning(logits, atoms, a, r, d, logits, atoms, sel)
+    assert loss.item() > 0
+
+
+def test_munchausen_reduces_to_soft_q():
+    B, A = 4, 3
+    g = torch.Generator().manual_seed(2)
+    q = torch.randn(B, A, generator=g)
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.randn(B, generator=g)
+    d = torch.full((B,), 0.9)
+    out = L.munchausen_q_learning(q, a, r, d, q, q, 0.03, 0.9, -1.0)
+    assert torch.isfinite(out)
+
+
+def test_quantile_regression_loss_zero_at_perfect():
+    taus = torch.tensor([0.25, 0.75])
+    src = torch.tensor([[1.0, 2.0]])
+    tgt = torch.tensor([[1.0, 2.0]])
+    out = L.quantile_regression_loss(src, taus, tgt, huber_param=1.0)
+    # not exactly 0 (cross terms), but small and positive
+    assert out.item() >= 0
+
+
+def test_quantile_q_learning_runs():
+    B, N, A = 4, 8, 3
+    g = torch.Generator().manual_seed(4)
+    dist = torch.randn(B, N, A, generator=g)
+    taus = (torch.arange(N, dtype=torch.float32) + 0.5) / N
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.randn(B, generator=g)
+    d = torch.full((B,), 0.99)
+    out = L.quantile_q_learning(dist, taus, a, r, d, dist, dist)
+    assert torch.isfinite(out) and out.item() > 0
+
+
+def test_dpo_loss_at_ratio_one():
+    lp = torch.tensor([0.0])
+    blp = torch.tensor([0.0])
+    adv = torch.tensor([1.5])
+    # ratio=1 -> drift terms 0 -> loss = -A
+    out = L.dpo_loss(lp, blp, adv, alpha=2.0, beta=0.6)
+    assert abs(out.item() + 1.5) < 1e-6
+
+
+def test_ppo_penalty_kl_zero_at_same_policy():
+    lp = torch.tensor([0.3, -0.2])
+    loss, kl = L.ppo_penalty_loss(lp, lp, torch.tensor([1.0, 1.0]), 3.0)
+    assert abs(kl.item()) < 1e-7
+    assert abs(loss.item() + 1.0) < 1e-6
