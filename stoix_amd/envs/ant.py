@@ -33,6 +33,7 @@ from typing import Dict, Tuple
 import torch
 
 from stoix_amd.envs.env import State, StatefulVecEnv, Tensor
+from stoix_amd.envs.hip_env import HipStepMixin
 from stoix_amd.envs.spaces import BoxSpace
 
 # state slice offsets
@@ -63,7 +64,11 @@ def quat_integrate(q: Tensor, omega: Tensor, dt: float) -> Tensor:
     return nq / nq.norm(dim=-1, keepdim=True).clamp(min=1e-8)
 
 
-class Ant(StatefulVecEnv):
+class Ant(HipStepMixin, StatefulVecEnv):
+    """GPU fast path: ops/csrc/envs.hip::ant_step_kernel (fused step)."""
+
+    HIP_KERNEL = "ant_step"
+    OBS_DIM = OBS_DIM
     max_episode_steps = 1000
 
     # body parameters
@@ -102,6 +107,10 @@ class Ant(StatefulVecEnv):
         self._hip_dir = torch.stack([torch.cos(ang), torch.sin(ang), torch.zeros(4)], dim=-1).to(
             self.device
         )  # [4, 3]
+        self._init_hip()
+
+    def _hip_action(self, action):
+        return action.to(torch.float32).contiguous()
 
     # ------------------------------------------------------------ state ops
 

@@ -1,0 +1,421 @@
+// Fused environment-step kernels for gfx950.
+//
+// Each kernel is the MI355X replacement for K1 of SURVEY.md §2.9: one launch
+// does dynamics + termination + truncation + episode-metric update +
+// autoreset for all B envs (the reference gets this fusion from XLA's
+// compilation of scan(env.step); here it is explicit). One thread per env:
+// the whole state row lives in registers; reset noise comes from Philox
+// substreams keyed by (seed, env, step_counter) so replayed hip graphs stay
+// deterministic per seed.
+#include "common.h"
+
+// ---------------------------------------------------------------- CartPole
+// Mirrors stoix_amd/envs/classic.py::CartPole (standard CartPole-v1
+// dynamics) + the StatefulVecEnv wrapper semantics (envs/env.py::step).
+extern "C" __global__ void cartpole_step_kernel(
+    float* __restrict__ state,          // [B, 4]
+    const long* __restrict__ action,    // [B]
+    int* __restrict__ step_count,       // [B]
+    float* __restrict__ ep_return,      // [B]
+    int* __restrict__ ep_length,        // [B]
+    float* __restrict__ last_ep_return, // [B]
+    int* __restrict__ last_ep_length,   // [B]
+    float* __restrict__ obs_out,        // [B, 4] post-autoreset obs
+    float* __restrict__ next_obs_out,   // [B, 4] true final obs
+    float* __restrict__ reward_out,     // [B]
+    float* __restrict__ discount_out,   // [B]
+    unsigned char* __restrict__ steptype_out, // [B]
+    unsigned char* __restrict__ done_out,     // [B] is_terminal_step
+    int B, int max_episode_steps, uint64_t seed,
+    const unsigned int* __restrict__ draw_buf) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  uint32_t draw = *draw_buf;
+
+  float x = state[b * 4 + 0];
+  float x_dot = state[b * 4 + 1];
+  float th = state[b * 4 + 2];
+  float th_dot = state[b * 4 + 3];
+
+  const float GRAV = 9.8f, MC = 1.0f, MP = 0.1f, LEN = 0.5f, FMAG = 10.0f, TAU = 0.02f;
+  const float THL = 12.0f * 2.0f * 3.14159265358979f / 360.0f, XL = 2.4f;
+
+  float force = (action[b] > 0) ? FMAG : -FMAG;
+  float costh = cosf(th), sinth = sinf(th);
+  float total_mass = MC + MP;
+  float pml = MP * LEN;
+  float temp = (force + pml * th_dot * th_dot * sinth) / total_mass;
+  float thacc = (GRAV * sinth - costh * temp) /
+                (LEN * (4.0f / 3.0f - MP * costh * costh / total_mass));
+  float xacc = temp - pml * thacc * costh / total_mass;
+  x += TAU * x_dot;
+  x_dot += TAU * xacc;
+  th += TAU * th_dot;
+  th_dot += TAU * thacc;
+
+  bool terminated = (fabsf(x) > XL) || (fabsf(th) > THL);
+  float reward = 1.0f;
+
+  int sc = step_count[b] + 1;
+  bool truncated = (sc >= max_episode_steps) && !terminated;
+  bool done = terminated || truncated;
+
+  float ret = ep_return[b] + reward;
+  int len = ep_length[b] + 1;
+  if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
+
+  // true final obs
+  next_obs_out[b * 4 + 0] = x;
+  next_obs_out[b * 4 + 1] = x_dot;
+  next_obs_out[b * 4 + 2] = th;
+  next_obs_out[b * 4 + 3] = th_dot;
+
+  if (done) {
+    Rng4 r = philox_uniform4(seed, 0u, (uint32_t)b, draw);
+    x = -0.05f + 0.1f * r.a;
+    x_dot = -0.05f + 0.1f * r.b;
+    th = -0.05f + 0.1f * r.c;
+    th_dot = -0.05f + 0.1f * r.d;
+    sc = 0; ret = 0.0f; len = 0;
+  }
+  state[b * 4 + 0] = x;
+  state[b * 4 + 1] = x_dot;
+  state[b * 4 + 2] = th;
+  state[b * 4 + 3] = th_dot;
+  obs_out[b * 4 + 0] = x;
+  obs_out[b * 4 + 1] = x_dot;
+  obs_out[b * 4 + 2] = th;
+  obs_out[b * 4 + 3] = th_dot;
+  step_count[b] = sc;
+  ep_return[b] = ret;
+  ep_length[b] = len;
+  reward_out[b] = reward;
+  discount_out[b] = terminated ? 0.0f : 1.0f;
+  steptype_out[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+  done_out[b] = done ? 1 : 0;
+}
+
+// --------------------------------------------------------------------- Ant
+// Mirrors stoix_amd/envs/ant.py (Ant-class quadruped, 4 substeps of
+// semi-implicit Euler with penalty contacts). One thread per env; the 29
+// state floats, 8 action floats and all leg intermediates live in registers.
+
+#define ANT_STATE 29
+#define ANT_OBS 27
+#define ANT_ACT 8
+
+struct V3 { float x, y, z; };
+DEV_INLINE V3 v3(float x, float y, float z) { return {x, y, z}; }
+DEV_INLINE V3 add3(V3 a, V3 b) { return {a.x + b.x, a.y + b.y, a.z + b.z}; }
+DEV_INLINE V3 scale3(V3 a, float s) { return {a.x * s, a.y * s, a.z * s}; }
+DEV_INLINE V3 cross3(V3 a, V3 b) {
+  return {a.y * b.z - a.z * b.y, a.z * b.x - a.x * b.z, a.x * b.y - a.y * b.x};
+}
+// rotate v by quaternion q = (w, x, y, z)
+DEV_INLINE V3 quat_rot(const float* q, V3 v) {
+  V3 qv = {q[1], q[2], q[3]};
+  V3 uv = cross3(qv, v);
+  V3 uuv = cross3(qv, uv);
+  return add3(v, add3(scale3(uv, 2.0f * q[0]), scale3(uuv, 2.0f)));
+}
+
+extern "C" __global__ void ant_step_kernel(
+    float* __restrict__ state,          // [B, 29]
+    const float* __restrict__ action,   // [B, 8]
+    int* __restrict__ step_count,
+    float* __restrict__ ep_return,
+    int* __restrict__ ep_length,
+    float* __restrict__ last_ep_return,
+    int* __restrict__ last_ep_length,
+    float* __restrict__ obs_out,        // [B, 27]
+    float* __restrict__ next_obs_out,   // [B, 27]
+    float* __restrict__ reward_out,
+    float* __restrict__ discount_out,
+    unsigned char* __restrict__ steptype_out,
+    unsigned char* __restrict__ done_out,
+    int B, int max_episode_steps, uint64_t seed,
+    const unsigned int* __restrict__ draw_buf) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  uint32_t draw = *draw_buf;
+
+  // constants (keep EXACTLY in sync with stoix_amd/envs/ant.py)
+  const float TORSO_MASS = 10.0f, TORSO_INERTIA = 0.4f, TORSO_Z0 = 0.55f;
+  const float HIP_RADIUS = 0.2f, L1 = 0.2f, L2 = 0.4f;
+  const float JOINT_INERTIA = 0.08f, JOINT_DAMPING = 1.2f, GEAR = 15.0f;
+  const float HIP_LIMIT = 0.6f, KNEE_LO = 0.4f, KNEE_HI = 1.4f, LIMIT_K = 40.0f;
+  const float KN = 2.0e3f, KD = 40.0f, FRICTION = 1.0f, GRAV = -9.81f;
+  const float DT = 0.05f;
+  const int SUBSTEPS = 4;
+  const float CTRL_COST = 0.5f, CONTACT_COST = 5e-4f, HEALTHY = 1.0f;
+  const float Z_MIN = 0.2f, Z_MAX = 1.0f;
+
+  float s[ANT_STATE];
+#pragma unroll
+  for (int i = 0; i < ANT_STATE; ++i) s[i] = state[b * ANT_STATE + i];
+  float a[ANT_ACT];
+#pragma unroll
+  for (int i = 0; i < ANT_ACT; ++i) a[i] = fminf(fmaxf(action[b * ANT_ACT + i], -1.0f), 1.0f);
+
+  // hip attachment angles: pi/4 + i*pi/2
+  const float SQ2 = 0.70710678118654752f;
+  const float hdx[4] = {SQ2, -SQ2, -SQ2, SQ2};
+  const float hdy[4] = {SQ2, SQ2, -SQ2, -SQ2};
+
+  float x_before = s[0];
+  float contact_mag = 0.0f;
+  const float dt = DT / SUBSTEPS;
+
+  for (int sub = 0; sub < SUBSTEPS; ++sub) {
+    float* pos = s + 0;      // 3
+    float* quat = s + 3;     // 4
+    float* linvel = s + 7;   // 3
+    float* angvel = s + 10;  // 3
+    float* qpos = s + 13;    // 8
+    float* qvel = s + 21;    // 8
+
+    // ---- foot positions & moment arms from the PRE-update joint state
+    V3 r_arm[4], foot_w[4];
+#pragma unroll
+    for (int leg = 0; leg < 4; ++leg) {
+      float base_ang = atan2f(hdy[leg], hdx[leg]);
+      float leg_ang = base_ang + qpos[leg];
+      float ca = cosf(leg_ang), sa = sinf(leg_ang);
+      float knee = qpos[4 + leg];
+      float ck = cosf(knee), sk = sinf(knee);
+      V3 body_off = {hdx[leg] * HIP_RADIUS + ca * L1 + ca * L2 * ck,
+                     hdy[leg] * HIP_RADIUS + sa * L1 + sa * L2 * ck,
+                     -L2 * sk};
+      V3 r = quat_rot(quat, body_off);
+      r_arm[leg] = r;
+      foot_w[leg] = {pos[0] + r.x, pos[1] + r.y, pos[2] + r.z};
+    }
+
+    // ---- joints: damped inertial with soft limits
+#pragma unroll
+    for (int j = 0; j < 8; ++j) {
+      float q = qpos[j];
+      float limit_tau;
+      if (j < 4) {
+        limit_tau = -LIMIT_K * (fmaxf(q - HIP_LIMIT, 0.0f) - fmaxf(-HIP_LIMIT - q, 0.0f));
+      } else {
+        limit_tau = -LIMIT_K * (fmaxf(q - KNEE_HI, 0.0f) - fmaxf(KNEE_LO - q, 0.0f));
+      }
+      float qacc = (GEAR * a[j] - JOINT_DAMPING * qvel[j] + limit_tau) / JOINT_INERTIA;
+      qvel[j] += dt * qacc;
+      qpos[j] += dt * qvel[j];
+    }
+
+    // ---- contacts
+    V3 total_f = {0, 0, 0}, total_tau = {0, 0, 0};
+#pragma unroll
+    for (int leg = 0; leg < 4; ++leg) {
+      V3 r = r_arm[leg];
+      V3 av = {angvel[0], angvel[1], angvel[2]};
+      V3 fv = add3(v3(linvel[0], linvel[1], linvel[2]), cross3(av, r));
+      float pen = fmaxf(-foot_w[leg].z, 0.0f);
+      float fn = 0.0f;
+      if (pen > 0.0f) fn = fmaxf(KN * pen - KD * fv.z, 0.0f);
+      float ftx = -FRICTION * fn * tanhf(4.0f * fv.x);
+      float fty = -FRICTION * fn * tanhf(4.0f * fv.y);
+      V3 cf = {ftx, fty, fn};
+      total_f = add3(total_f, cf);
+      total_tau = add3(total_tau, cross3(r, cf));
+      contact_mag += fabsf(cf.x) + fabsf(cf.y) + fabsf(cf.z);
+    }
+
+    // ---- torso integration
+    linvel[0] += dt * (total_f.x / TORSO_MASS);
+    linvel[1] += dt * (total_f.y / TORSO_MASS);
+    linvel[2] += dt * (total_f.z / TORSO_MASS + GRAV);
+    pos[0] += dt * linvel[0];
+    pos[1] += dt * linvel[1];
+    pos[2] += dt * linvel[2];
+    angvel[0] += dt * (total_tau.x / TORSO_INERTIA - 0.2f * angvel[0]);
+    angvel[1] += dt * (total_tau.y / TORSO_INERTIA - 0.2f * angvel[1]);
+    angvel[2] += dt * (total_tau.z / TORSO_INERTIA - 0.2f * angvel[2]);
+    // quaternion integration (world-frame omega), then renormalise
+    {
+      float w = quat[0], qx = quat[1], qy = quat[2], qz = quat[3];
+      float ox = angvel[0], oy = angvel[1], oz = angvel[2];
+      float dw = 0.5f * (-qx * ox - qy * oy - qz * oz);
+      float dx = 0.5f * (w * ox + qy * oz - qz * oy);
+      float dy = 0.5f * (w * oy + qz * ox - qx * oz);
+      float dz = 0.5f * (w * oz + qx * oy - qy * ox);
+      w += dt * dw; qx += dt * dx; qy += dt * dy; qz += dt * dz;
+      float n = sqrtf(fmaxf(w * w + qx * qx + qy * qy + qz * qz, 1e-16f));
+      quat[0] = w / n; quat[1] = qx / n; quat[2] = qy / n; quat[3] = qz / n;
+    }
+    // torso ground cushion
+    float torso_pen = fmaxf(0.12f - pos[2], 0.0f);
+    linvel[2] += dt * KN / TORSO_MASS * torso_pen;
+  }
+
+  // ---- reward & termination
+  float forward_vel = (s[0] - x_before) / DT;
+  float ctrl_cost = 0.0f;
+#pragma unroll
+  for (int i = 0; i < ANT_ACT; ++i) ctrl_cost += a[i] * a[i];
+  ctrl_cost *= CTRL_COST;
+  float cm = contact_mag / SUBSTEPS;
+  float contact_cost = CONTACT_COST * cm * cm;
+  bool finite = true;
+#pragma unroll
+  for (int i = 0; i < ANT_STATE; ++i) finite = finite && isfinite(s[i]);
+  float z = s[2];
+  bool healthy = (z > Z_MIN) && (z < Z_MAX) && finite;
+  float reward = forward_vel + HEALTHY - ctrl_cost - contact_cost;
+  if (!isfinite(reward)) reward = 0.0f;
+  bool terminated = !healthy;
+  if (!finite) {
+#pragma unroll
+    for (int i = 0; i < ANT_STATE; ++i) s[i] = isfinite(s[i]) ? s[i] : 0.0f;
+  }
+
+  // ---- wrapper semantics
+  int sc = step_count[b] + 1;
+  bool truncated = (sc >= max_episode_steps) && !terminated;
+  bool done = terminated || truncated;
+  float ret = ep_return[b] + reward;
+  int len = ep_length[b] + 1;
+  if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
+
+  // write true final obs
+  {
+    float* o = next_obs_out + b * ANT_OBS;
+    o[0] = s[2];
+    o[1] = s[3]; o[2] = s[4]; o[3] = s[5]; o[4] = s[6];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o[5 + i] = s[13 + i];
+    o[13] = s[7]; o[14] = s[8]; o[15] = s[9];
+    o[16] = s[10]; o[17] = s[11]; o[18] = s[12];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o[19 + i] = s[21 + i];
+  }
+
+  if (done) {
+    // autoreset with philox noise, mirroring Ant._reset_fn
+    Rng4 r0 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 0u);
+    Rng4 r1 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 1u);
+    Rng4 r2 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 2u);
+    Rng4 r3 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 3u);
+    Rng4 r4 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 4u);
+    float u[20] = {r0.a, r0.b, r0.c, r0.d, r1.a, r1.b, r1.c, r1.d,
+                   r2.a, r2.b, r2.c, r2.d, r3.a, r3.b, r3.c, r3.d,
+                   r4.a, r4.b, r4.c, r4.d};
+#pragma unroll
+    for (int i = 0; i < ANT_STATE; ++i) s[i] = 0.0f;
+    s[2] = TORSO_Z0;
+    s[3] = 1.0f;  // identity quaternion
+#pragma unroll
+    for (int i = 0; i < 8; ++i) s[13 + i] = -0.1f + 0.2f * u[i];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) s[17 + i] += 0.9f;  // knees bent
+#pragma unroll
+    for (int i = 0; i < 8; ++i) s[21 + i] = -0.05f + 0.1f * u[8 + i];
+#pragma unroll
+    for (int i = 0; i < 3; ++i) s[7 + i] = -0.05f + 0.1f * u[16 + i];
+    sc = 0; ret = 0.0f; len = 0;
+  }
+
+#pragma unroll
+  for (int i = 0; i < ANT_STATE; ++i) state[b * ANT_STATE + i] = s[i];
+  {
+    float* o = obs_out + b * ANT_OBS;
+    o[0] = s[2];
+    o[1] = s[3]; o[2] = s[4]; o[3] = s[5]; o[4] = s[6];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o[5 + i] = s[13 + i];
+    o[13] = s[7]; o[14] = s[8]; o[15] = s[9];
+    o[16] = s[10]; o[17] = s[11]; o[18] = s[12];
+#pragma unroll
+    for (int i = 0; i < 8; ++i) o[19 + i] = s[21 + i];
+  }
+  step_count[b] = sc;
+  ep_return[b] = ret;
+  ep_length[b] = len;
+  reward_out[b] = reward;
+  discount_out[b] = terminated ? 0.0f : 1.0f;
+  steptype_out[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+  done_out[b] = done ? 1 : 0;
+}
+
+// ----------------------------------------------------------- ant reset
+extern "C" __global__ void ant_reset_kernel(
+    float* __restrict__ state, int B, uint64_t seed, uint32_t draw) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  const float TORSO_Z0 = 0.55f;
+  float s[ANT_STATE];
+  Rng4 r0 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 0u);
+  Rng4 r1 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 1u);
+  Rng4 r2 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 2u);
+  Rng4 r3 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 3u);
+  Rng4 r4 = philox_uniform4(seed, 1u, (uint32_t)b, draw * 8u + 4u);
+  float u[20] = {r0.a, r0.b, r0.c, r0.d, r1.a, r1.b, r1.c, r1.d,
+                 r2.a, r2.b, r2.c, r2.d, r3.a, r3.b, r3.c, r3.d,
+                 r4.a, r4.b, r4.c, r4.d};
+#pragma unroll
+  for (int i = 0; i < ANT_STATE; ++i) s[i] = 0.0f;
+  s[2] = TORSO_Z0;
+  s[3] = 1.0f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s[13 + i] = -0.1f + 0.2f * u[i];
+#pragma unroll
+  for (int i = 0; i < 4; ++i) s[17 + i] += 0.9f;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) s[21 + i] = -0.05f + 0.1f * u[8 + i];
+#pragma unroll
+  for (int i = 0; i < 3; ++i) s[7 + i] = -0.05f + 0.1f * u[16 + i];
+#pragma unroll
+  for (int i = 0; i < ANT_STATE; ++i) state[b * ANT_STATE + i] = s[i];
+}
+
+extern "C" __global__ void bump_u32_kernel(unsigned int* p) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *p += 1;
+}
+
+// ----------------------------------------------------------- host launchers
+extern "C" void launch_cartpole_step(
+    float* state, const long* action, int* step_count, float* ep_return,
+    int* ep_length, float* last_ep_return, int* last_ep_length, float* obs_out,
+    float* next_obs_out, float* reward_out, float* discount_out,
+    unsigned char* steptype_out, unsigned char* done_out, int B,
+    int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
+    void* stream) {
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(cartpole_step_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, state, action, step_count, ep_return,
+                     ep_length, last_ep_return, last_ep_length, obs_out,
+                     next_obs_out, reward_out, discount_out, steptype_out,
+                     done_out, B, max_episode_steps, seed, draw_buf);
+  hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, draw_buf);
+}
+
+extern "C" void launch_ant_step(
+    float* state, const float* action, int* step_count, float* ep_return,
+    int* ep_length, float* last_ep_return, int* last_ep_length, float* obs_out,
+    float* next_obs_out, float* reward_out, float* discount_out,
+    unsigned char* steptype_out, unsigned char* done_out, int B,
+    int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
+    void* stream) {
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(ant_step_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, state, action, step_count, ep_return,
+                     ep_length, last_ep_return, last_ep_length, obs_out,
+                     next_obs_out, reward_out, discount_out, steptype_out,
+                     done_out, B, max_episode_steps, seed, draw_buf);
+  hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, draw_buf);
+}
+
+extern "C" void launch_ant_reset(float* state, int B, uint64_t seed,
+                                 uint32_t draw, void* stream) {
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(ant_reset_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, state, B, seed, draw);
+}

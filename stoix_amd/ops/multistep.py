@@ -56,6 +56,30 @@ def batch_truncated_generalized_advantage_estimation(
         (advantages [T, B], target_values = advantages + v_tm1).
     """
     T = r_t.shape[0]
+    if r_t.is_cuda:
+        from stoix_amd import ops as _ops
+
+        e = _ops.ext(required=True)
+        adv = torch.empty_like(r_t)
+        targets = torch.empty_like(r_t)
+        tr = (
+            truncation_t.to(torch.uint8).contiguous()
+            if truncation_t is not None
+            else torch.empty(0, dtype=torch.uint8, device=r_t.device)
+        )
+        e.gae(
+            r_t.contiguous(),
+            discount_t.contiguous(),
+            v_tm1.contiguous(),
+            v_t.contiguous(),
+            tr,
+            adv,
+            targets,
+            float(lambda_),
+        )
+        if standardize_advantages:
+            adv = (adv - adv.mean()) / (adv.std(unbiased=False) + 1e-8)
+        return adv, targets
     if truncation_t is None:
         cont = torch.ones_like(discount_t)
     else:
@@ -118,6 +142,13 @@ def batch_lambda_returns(
     v_t[t] = V(s_{t+1}).
     """
     T = r_t.shape[0]
+    if r_t.is_cuda:
+        from stoix_amd import ops as _ops
+
+        e = _ops.ext(required=True)
+        out = torch.empty_like(r_t)
+        e.lambda_returns(r_t.contiguous(), discount_t.contiguous(), v_t.contiguous(), out, float(lambda_))
+        return out
     out = torch.empty_like(r_t)
     acc = v_t[-1]
     for t in range(T - 1, -1, -1):
@@ -153,6 +184,16 @@ def batch_general_off_policy_returns_from_q_and_v(
     c_t[-1] is unused.
     """
     T = r_t.shape[0]
+    if r_t.is_cuda:
+        from stoix_amd import ops as _ops
+
+        e = _ops.ext(required=True)
+        out = torch.empty_like(r_t)
+        e.offpolicy_returns(
+            q_t.contiguous(), v_t.contiguous(), r_t.contiguous(),
+            discount_t.contiguous(), c_t.contiguous(), out,
+        )
+        return out
     out = torch.empty_like(r_t)
     g = r_t[T - 1] + discount_t[T - 1] * v_t[T - 1]
     out[T - 1] = g
@@ -249,6 +290,20 @@ def vtrace_td_error_and_advantage(
 
     Returns (errors = vs - v_tm1, pg_advantage, q_estimate) each [T, B].
     """
+    if r_t.is_cuda:
+        from stoix_amd import ops as _ops
+
+        e = _ops.ext(required=True)
+        errors = torch.empty_like(r_t)
+        pg_adv = torch.empty_like(r_t)
+        e.vtrace(
+            v_tm1.contiguous(), v_t.contiguous(), r_t.contiguous(),
+            discount_t.contiguous(), rho_tm1.contiguous(), errors, pg_adv,
+            float(lambda_), float(clip_rho_threshold), float(clip_pg_rho_threshold),
+        )
+        vs_t = torch.cat([errors[1:] + v_tm1[1:], v_t[-1:]], dim=0)
+        q_estimate = r_t + discount_t * vs_t
+        return errors, pg_adv, q_estimate
     rho_clip = torch.clamp(rho_tm1, max=clip_rho_threshold)
     c_t = lambda_ * torch.clamp(rho_tm1, max=1.0)
     delta = rho_clip * (r_t + discount_t * v_t - v_tm1)

@@ -50,19 +50,39 @@ class PPOLearner:
         broadcast_module(self.actor)
         broadcast_module(self.critic)
 
-        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr), eps=1e-5)
-        self.critic_opt = torch.optim.Adam(self.critic.parameters(), lr=float(self.sys.critic_lr), eps=1e-5)
+        capturable = device.type == "cuda"
+        self.actor_opt = torch.optim.Adam(
+            self.actor.parameters(), lr=float(self.sys.actor_lr), eps=1e-5, capturable=capturable
+        )
+        self.critic_opt = torch.optim.Adam(
+            self.critic.parameters(), lr=float(self.sys.critic_lr), eps=1e-5, capturable=capturable
+        )
         self.reducer = FlatGradReducer(
             list(self.actor.parameters()) + list(self.critic.parameters()), device
         )
         self.gen = torch.Generator(device=device)
         self.gen.manual_seed(int(config.arch.seed) * 7919 + 13)
 
+        # bf16 compute path (BASELINE config "Brax Ant bf16"): network fwd/bwd
+        # run under autocast-bf16 (MFMA-rate GEMMs); env physics, GAE, losses
+        # accumulate in fp32; optimiser state fp32.
+        use_bf16 = str(getattr(self.sys, "compute_dtype", "fp32")) == "bf16"
+        self.amp = (
+            torch.autocast(device_type="cuda", dtype=torch.bfloat16)
+            if use_bf16 and device.type == "cuda"
+            else None
+        )
+
         self.normalize_obs = bool(getattr(self.sys, "normalize_observations", False))
         if self.normalize_obs:
             self.obs_stats = rs.init_state(obs_space.shape, device=device)
 
         self.ts = env.reset()
+        # stable-address current-observation buffer: the graph-captured
+        # rollout reads it at t=0 and writes the final obs back at the end,
+        # so replays chain correctly (ops/graph.py).
+        self.cur_obs = self.ts.observation.clone()
+        self.collect_metrics = True
         self.episode_metrics: Dict[str, float] = {}
         self._discrete = not hasattr(act_space, "shape") or len(act_space.shape) == 0
 
@@ -93,41 +113,59 @@ class PPOLearner:
 
     @torch.no_grad()
     def _rollout(self) -> None:
-        ts = self.ts
+        obs_raw = self.cur_obs
         for t in range(self.T):
-            obs = ts.observation
+            obs = obs_raw
             if self.normalize_obs:
                 obs = rs.normalize(obs, self.obs_stats)
-            dist = self.actor(obs)
-            value = self.critic(obs)
-            action = dist.sample(self.gen)
-            log_prob = dist.log_prob(action)
+            if self.amp is not None:
+                with self.amp:
+                    dist = self.actor(obs)
+                    value = self.critic(obs)
+                    action = dist.sample(self.gen)
+                    log_prob = dist.log_prob(action)
+            else:
+                dist = self.actor(obs)
+                value = self.critic(obs)
+                action = dist.sample(self.gen)
+                log_prob = dist.log_prob(action)
             next_ts = self.env.step(action)
             next_obs = next_ts.extras["next_obs"]
             if self.normalize_obs:
                 next_obs = rs.normalize(next_obs, self.obs_stats)
-            bootstrap_value = self.critic(next_obs)
+            if self.amp is not None:
+                with self.amp:
+                    bootstrap_value = self.critic(next_obs)
+            else:
+                bootstrap_value = self.critic(next_obs)
 
             self.buf_obs[t] = obs
             self.buf_action[t] = action
-            self.buf_log_prob[t] = log_prob
-            self.buf_value[t] = value
-            self.buf_bootstrap[t] = bootstrap_value
+            self.buf_log_prob[t] = log_prob.float()
+            self.buf_value[t] = value.float()
+            self.buf_bootstrap[t] = bootstrap_value.float()
             self.buf_reward[t] = next_ts.reward
             self.buf_discount[t] = next_ts.discount
             self.buf_truncated[t] = next_ts.truncated()
-            ts = next_ts
-        self.ts = ts
+            obs_raw = next_ts.observation
+            last_ts = next_ts
+        self.cur_obs.copy_(obs_raw)
         if self.normalize_obs:
             self.obs_stats = rs.update(self.obs_stats, self.buf_obs, all_reduce=True)
-        em = ts.extras["episode_metrics"]
-        final, has = get_final_step_metrics(em)
-        if has:
-            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+        if self.collect_metrics:
+            em = last_ts.extras["episode_metrics"]
+            final, has = get_final_step_metrics(em)
+            if has:
+                self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
     # ---------------------------------------------------------------- update
 
     def update_step(self) -> Dict[str, Tensor]:
+        return self.graph_body()
+
+    def graph_body(self) -> Dict[str, Tensor]:
+        """One rollout + update. Pure device work when collect_metrics is
+        off — capturable as a single hip graph (ops/graph.py)."""
         self._rollout()
         gamma = float(self.sys.gamma)
         adv, targets = multistep.batch_truncated_generalized_advantage_estimation(
@@ -152,17 +190,31 @@ class PPOLearner:
         mb_size = TB // n_mb
         metrics: Dict[str, Tensor] = {}
         for _ in range(int(self.sys.epochs)):
-            perm = torch.randperm(TB, device=self.device, generator=self.gen)
+            if self.device.type == "cuda":
+                # capture-safe shuffle: philox rand + device argsort
+                perm = torch.argsort(torch.rand(TB, device=self.device, generator=self.gen))
+            else:
+                perm = torch.randperm(TB, device=self.device, generator=self.gen)
             for mb in range(n_mb):
                 idx = perm[mb * mb_size : (mb + 1) * mb_size]
                 obs_mb = flat_obs[idx]
-                dist = self.actor(obs_mb)
-                new_logp = dist.log_prob(flat_action[idx])
-                entropy = dist.entropy().mean()
+                if self.amp is not None:
+                    with self.amp:
+                        dist = self.actor(obs_mb)
+                        new_logp = dist.log_prob(flat_action[idx])
+                        entropy = dist.entropy().mean()
+                        value = self.critic(obs_mb)
+                    new_logp = new_logp.float()
+                    entropy = entropy.float()
+                    value = value.float()
+                else:
+                    dist = self.actor(obs_mb)
+                    new_logp = dist.log_prob(flat_action[idx])
+                    entropy = dist.entropy().mean()
+                    value = self.critic(obs_mb)
                 a_loss = ppo_clip_loss(new_logp, flat_logp[idx], flat_adv[idx], float(self.sys.clip_eps))
                 actor_loss = a_loss - float(self.sys.ent_coef) * entropy
 
-                value = self.critic(obs_mb)
                 v_loss = clipped_value_loss(value, flat_value[idx], flat_targets[idx], float(self.sys.clip_eps))
                 critic_loss = float(self.sys.vf_coef) * v_loss
 
@@ -181,6 +233,22 @@ class PPOLearner:
                     "entropy": entropy.detach(),
                 }
         return metrics
+
+    # ------------------------------------------------------- graph support
+
+    def prepare_for_graph_capture(self) -> None:
+        """Switch to capture-safe modes: default CUDA RNG (graph-aware),
+        inline all-reduce on the capture stream, no host-side metric reads."""
+        self.gen = None
+        self.collect_metrics = False
+        self.reducer._stream = None
+
+    def after_graph_replay(self) -> None:
+        # episode metrics read eagerly from the env's latched buffers
+        self.episode_metrics = {
+            "episode_return": self.env._last_ep_return.mean(),
+            "episode_length": self.env._last_ep_length.float().mean(),
+        }
 
     # ------------------------------------------------------------ checkpoint
 
