@@ -3,19 +3,24 @@
 The Anakin inner loop is launch-latency-bound: a rollout of T=128 steps with
 tiny MLPs issues thousands of kernels whose boundaries (~1.2-1.5 us each,
 MI355X_MICROARCH 'boundary') and host launch overhead (~3.3-3.8 us eager)
-dominate. Capturing the WHOLE update step (rollout + GAE + epochs of
-minibatch updates) into one hipGraph replays it with a single ~10-16 us
-host call — the MI355X equivalent of the reference's one-XLA-program design
-(SURVEY.md §3.1).
+dominate. Capture replaces thousands of host launches per update with a
+handful of graph replays (~10-16 us each) — the MI355X equivalent of the
+reference's one-XLA-program design.
 
-Capture preconditions handled here:
-  * env must have the fused HIP step path (device-side RNG counter) — the
-    torch env path has host-dependent control flow and cannot be captured;
-  * the learner's torch RNG falls back to the default CUDA generator
-    (graph-aware) instead of a user Generator;
-  * gradient all-reduce runs on the capture stream (RCCL supports graph
-    capture); Adam runs in capturable mode;
-  * episode-metric extraction (host reads) moves outside the graph.
+Two graphs per learner:
+  * rollout graph — T env steps (fused HIP kernels, device-side RNG
+    counters) + actor/critic forwards + GAE, writing stable buffers;
+  * epoch graph — num_minibatches x (forward, losses, backward, RCCL
+    all-reduce, clip, Adam) reading a static permutation buffer.
+The minibatch permutation itself is NOT capture-legal at scale (randperm /
+sort do host work), so it is refreshed eagerly into ``perm_buf`` between
+replays — one tensor write per epoch.
+
+Capture preconditions handled here: the env must have the fused HIP step
+path; the learner's RNG falls back to the default (graph-aware) CUDA
+generator; the gradient all-reduce runs inline on the capture stream (RCCL
+supports capture); Adam is in capturable mode; host-side metric reads move
+outside the graphs.
 """
 from __future__ import annotations
 
@@ -25,8 +30,8 @@ import torch
 
 
 def try_enable_graphs(learner) -> bool:
-    """Capture learner.update_step into a hip graph; monkey-patches
-    learner.update_step to replay it. Returns True on success."""
+    """Capture learner's rollout/epoch phases into hip graphs and
+    monkey-patch update_step to replay them. Returns True on success."""
     device = learner.device
     if device.type != "cuda":
         return False
@@ -35,24 +40,36 @@ def try_enable_graphs(learner) -> bool:
 
     learner.prepare_for_graph_capture()
 
-    # eager warmup on a side stream (rocBLAS/hipBLASLt workspaces, autotuning)
+    # eager warmup on a side stream (rocBLAS/hipBLASLt workspaces, autotune)
     side = torch.cuda.Stream(device)
     side.wait_stream(torch.cuda.current_stream(device))
     with torch.cuda.stream(side):
         for _ in range(2):
-            learner.graph_body()
+            learner.rollout_phase()
+            learner._new_perm()
+            learner.epoch_phase()
     torch.cuda.current_stream(device).wait_stream(side)
     torch.cuda.synchronize(device)
 
-    graph = torch.cuda.CUDAGraph()
-    with torch.cuda.graph(graph):
-        static_metrics = learner.graph_body()
+    g_rollout = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g_rollout):
+        learner.rollout_phase()
+
+    learner._new_perm()
+    g_epoch = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g_epoch):
+        static_metrics = learner.epoch_phase()
+
+    epochs = int(learner.sys.epochs)
 
     def update_step() -> Dict[str, torch.Tensor]:
-        graph.replay()
+        g_rollout.replay()
+        for _ in range(epochs):
+            learner._new_perm()
+            g_epoch.replay()
         learner.after_graph_replay()
         return static_metrics
 
-    learner._graph = graph
+    learner._graphs = (g_rollout, g_epoch)
     learner.update_step = update_step
     return True

@@ -99,6 +99,9 @@ class PPOLearner:
         self.buf_reward = z(self.T, self.B)
         self.buf_discount = z(self.T, self.B)
         self.buf_truncated = z(self.T, self.B, dtype=torch.bool)
+        self.buf_adv = z(self.T, self.B)
+        self.buf_targets = z(self.T, self.B)
+        self.perm_buf = torch.arange(self.T * self.B, device=device)
 
     # ---------------------------------------------------------------- acting
 
@@ -161,11 +164,14 @@ class PPOLearner:
     # ---------------------------------------------------------------- update
 
     def update_step(self) -> Dict[str, Tensor]:
-        return self.graph_body()
+        self.rollout_phase()
+        for _ in range(int(self.sys.epochs)):
+            self._new_perm()
+            metrics = self.epoch_phase()
+        return metrics
 
-    def graph_body(self) -> Dict[str, Tensor]:
-        """One rollout + update. Pure device work when collect_metrics is
-        off — capturable as a single hip graph (ops/graph.py)."""
+    def rollout_phase(self) -> None:
+        """Rollout + GAE into stable buffers (hip-graph capturable)."""
         self._rollout()
         gamma = float(self.sys.gamma)
         adv, targets = multistep.batch_truncated_generalized_advantage_estimation(
@@ -177,24 +183,30 @@ class PPOLearner:
             truncation_t=self.buf_truncated,
             standardize_advantages=bool(self.sys.standardize_advantages),
         )
+        self.buf_adv.copy_(adv)
+        self.buf_targets.copy_(targets)
 
+    def _new_perm(self) -> None:
+        TB = self.T * self.B
+        self.perm_buf.copy_(torch.randperm(TB, device=self.device, generator=self.gen))
+
+    def epoch_phase(self) -> Dict[str, Tensor]:
+        """One epoch of minibatch updates reading self.perm_buf
+        (hip-graph capturable; the permutation is refreshed eagerly between
+        replays — randperm is not capture-legal at scale)."""
         TB = self.T * self.B
         flat_obs = self.buf_obs.reshape(TB, *self.buf_obs.shape[2:])
         flat_action = self.buf_action.reshape(TB, *self.buf_action.shape[2:])
         flat_logp = self.buf_log_prob.reshape(TB)
         flat_value = self.buf_value.reshape(TB)
-        flat_adv = adv.reshape(TB)
-        flat_targets = targets.reshape(TB)
+        flat_adv = self.buf_adv.reshape(TB)
+        flat_targets = self.buf_targets.reshape(TB)
 
         n_mb = int(self.sys.num_minibatches)
         mb_size = TB // n_mb
         metrics: Dict[str, Tensor] = {}
-        for _ in range(int(self.sys.epochs)):
-            if self.device.type == "cuda":
-                # capture-safe shuffle: philox rand + device argsort
-                perm = torch.argsort(torch.rand(TB, device=self.device, generator=self.gen))
-            else:
-                perm = torch.randperm(TB, device=self.device, generator=self.gen)
+        if True:
+            perm = self.perm_buf
             for mb in range(n_mb):
                 idx = perm[mb * mb_size : (mb + 1) * mb_size]
                 obs_mb = flat_obs[idx]

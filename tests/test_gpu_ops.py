@@ -222,6 +222,7 @@ def test_ppo_gpu_update_step(ext):
             "arch.total_num_envs=512",
             "arch.total_timesteps=null",
             "arch.num_updates=4",
+            "arch.num_evaluation=1",
             "system.rollout_length=8",
             "system.num_minibatches=2",
             "system.epochs=1",
@@ -257,6 +258,7 @@ def test_ppo_graph_capture(ext):
             "arch.total_num_envs=512",
             "arch.total_timesteps=null",
             "arch.num_updates=4",
+            "arch.num_evaluation=1",
             "system.rollout_length=8",
             "system.num_minibatches=2",
             "system.epochs=1",

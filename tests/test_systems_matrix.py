@@ -1,0 +1,65 @@
+"""Smoke matrix: every system end-to-end at tiny scale on CPU (parity with
+the reference's bash_scripts/run-algorithms.sh)."""
+import importlib
+
+import pytest
+
+from stoix_amd.config import compose
+from stoix_amd.parallel.dist import reset_dist_context
+
+TINY = [
+    "arch.total_num_envs=8",
+    "arch.total_timesteps=null",
+    "arch.num_updates=2",
+    "arch.num_evaluation=1",
+    "arch.num_eval_episodes=4",
+    "arch.absolute_metric=false",
+    "system.rollout_length=4",
+    "logger.loggers=[]",
+    "logger.checkpointing.save_model=false",
+]
+
+SYSTEMS = [
+    ("stoix_amd.systems.ppo.ff_ppo", "default/anakin/default_ff_ppo.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.ppo.ff_ppo", "default/anakin/default_ff_ppo_continuous.yaml",
+     ["system.num_minibatches=2", "system.epochs=1", "env=classic/pendulum"]),
+    ("stoix_amd.systems.q_learning.ff_dqn", "default/anakin/default_ff_dqn.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_ddqn", "default/anakin/default_ff_ddqn.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_mdqn", "default/anakin/default_ff_mdqn.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_dqn_reg", "default/anakin/default_ff_dqn_reg.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_c51", "default/anakin/default_ff_c51.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_qr_dqn", "default/anakin/default_ff_qr_dqn.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.q_learning.ff_pqn", "default/anakin/default_ff_pqn.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.sac.ff_sac", "default/anakin/default_ff_sac.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.ddpg.ff_ddpg", "default/anakin/default_ff_ddpg.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.ddpg.ff_td3", "default/anakin/default_ff_td3.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.ddpg.ff_d4pg", "default/anakin/default_ff_d4pg.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
+    ("stoix_amd.systems.vpg.ff_reinforce", "default/anakin/default_ff_reinforce.yaml", []),
+]
+
+
+@pytest.fixture(autouse=True)
+def _fresh_dist():
+    reset_dist_context()
+    yield
+    reset_dist_context()
+
+
+@pytest.mark.parametrize("module,default,extra", SYSTEMS, ids=[f"{m.split('.')[-1]}-{d.split('_', 2)[-1][:-5]}" for m, d, e in SYSTEMS])
+def test_system_smoke(module, default, extra):
+    mod = importlib.import_module(module)
+    cfg = compose(default, TINY + extra)
+    r = mod.run(cfg)
+    assert r == r, f"{module} returned NaN"
