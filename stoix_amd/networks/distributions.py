@@ -156,10 +156,19 @@ class AffineTanhTransformedDistribution(Distribution):
         epsilon: float = 1e-3,
     ):
         self.base = Normal(loc, scale)
-        mn = torch.as_tensor(minimum, dtype=loc.dtype, device=loc.device)
-        mx = torch.as_tensor(maximum, dtype=loc.dtype, device=loc.device)
-        self.scale_affine = (mx - mn) / 2.0
-        self.shift_affine = (mx + mn) / 2.0
+        # Keep python-float bounds as scalars: creating device tensors here
+        # would issue a pageable H2D copy on every forward, which is illegal
+        # inside hip-graph capture (and needless work outside it).
+        if isinstance(minimum, torch.Tensor) or isinstance(maximum, torch.Tensor):
+            mn = torch.as_tensor(minimum, dtype=loc.dtype, device=loc.device)
+            mx = torch.as_tensor(maximum, dtype=loc.dtype, device=loc.device)
+            self.scale_affine = (mx - mn) / 2.0
+            self.shift_affine = (mx + mn) / 2.0
+            self._log_scale = torch.log(self.scale_affine)
+        else:
+            self.scale_affine = (float(maximum) - float(minimum)) / 2.0
+            self.shift_affine = (float(maximum) + float(minimum)) / 2.0
+            self._log_scale = math.log(self.scale_affine)
         self.eps = epsilon
 
     def _forward(self, u: Tensor) -> Tensor:
@@ -184,7 +193,7 @@ class AffineTanhTransformedDistribution(Distribution):
 
     def _log_prob_from_u(self, u: Tensor) -> Tensor:
         # log |d tanh/du| = log(1 - tanh(u)^2) = 2*(log2 - u - softplus(-2u))
-        log_det = 2.0 * (math.log(2.0) - u - F.softplus(-2.0 * u)) + torch.log(self.scale_affine)
+        log_det = 2.0 * (math.log(2.0) - u - F.softplus(-2.0 * u)) + self._log_scale
         return (self.base.log_prob(u) - log_det).sum(-1)
 
     def log_prob(self, value: Tensor) -> Tensor:

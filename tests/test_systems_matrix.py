@@ -47,6 +47,20 @@ SYSTEMS = [
     ("stoix_amd.systems.ddpg.ff_d4pg", "default/anakin/default_ff_d4pg.yaml",
      ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
     ("stoix_amd.systems.vpg.ff_reinforce", "default/anakin/default_ff_reinforce.yaml", []),
+    ("stoix_amd.systems.awr.ff_awr", "default/anakin/default_ff_awr.yaml",
+     ["system.num_critic_steps=2", "system.num_actor_steps=2", "system.batch_size=8",
+      "system.buffer_size=256", "system.sample_sequence_length=4"]),
+    ("stoix_amd.systems.awr.ff_awr", "default/anakin/default_ff_awr_continuous.yaml",
+     ["system.num_critic_steps=2", "system.num_actor_steps=2", "system.batch_size=8",
+      "system.buffer_size=256", "system.sample_sequence_length=4"]),
+    ("stoix_amd.systems.mpo.ff_mpo", "default/anakin/default_ff_mpo.yaml",
+     ["system.epochs=2", "system.batch_size=8", "system.buffer_size=256",
+      "system.sample_sequence_length=4", "system.num_samples=4"]),
+    ("stoix_amd.systems.mpo.ff_mpo", "default/anakin/default_ff_mpo_continuous.yaml",
+     ["system.epochs=2", "system.batch_size=8", "system.buffer_size=256",
+      "system.sample_sequence_length=4", "system.num_samples=4"]),
+    ("stoix_amd.systems.mpo.ff_vmpo", "default/anakin/default_ff_vmpo.yaml", []),
+    ("stoix_amd.systems.mpo.ff_vmpo", "default/anakin/default_ff_vmpo_continuous.yaml", []),
 ]
 
 
