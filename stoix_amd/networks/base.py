@@ -124,7 +124,7 @@ class RecurrentActor(nn.Module):
         T, B = obs.shape[:2]
         z = self.pre_torso(obs.reshape(T * B, -1)).reshape(T, B, -1)
         h, state = self.rnn(z, resets, state)
-        y = self.post_torso(h.reshape(T * B, -1))
+        y = self.post_torso(h.reshape(T * B, -1)).reshape(T, B, -1)
         dist = self.action_head(y)
         return dist, state
 

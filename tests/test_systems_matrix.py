@@ -61,6 +61,11 @@ SYSTEMS = [
       "system.sample_sequence_length=4", "system.num_samples=4"]),
     ("stoix_amd.systems.mpo.ff_vmpo", "default/anakin/default_ff_vmpo.yaml", []),
     ("stoix_amd.systems.mpo.ff_vmpo", "default/anakin/default_ff_vmpo_continuous.yaml", []),
+    ("stoix_amd.systems.q_learning.ff_rainbow", "default/anakin/default_ff_rainbow.yaml",
+     ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512",
+      "system.warmup_steps=16", "system.n_step=3"]),
+    ("stoix_amd.systems.ppo.rec_ppo", "default/anakin/default_rec_ppo.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
 ]
 
 
