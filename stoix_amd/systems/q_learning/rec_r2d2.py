@@ -1,0 +1,273 @@
+"""Anakin R2D2 (parity: /root/reference/stoix/systems/q_learning/rec_r2d2.py).
+
+Recurrent Q-network with stored per-step hidden states (dqn_types.py:18-26),
+prioritised SEQUENCE buffer, burn-in re-burning hidden states without
+gradient for online and target nets (:300-328), double-Q with n-step targets
+under the signed-hyperbolic value transform (SIGNED_HYPERBOLIC_PAIR, :18,
+344-360), priorities = eta*max + (1-eta)*mean of |TD| per sequence
+(:370-375).
+"""
+from __future__ import annotations
+
+import copy
+import sys
+from typing import Dict, Tuple
+
+import torch
+import torch.nn as nn
+
+from stoix_amd.buffers import PrioritisedBuffer
+from stoix_amd.config import compose
+from stoix_amd.envs.env import StatefulVecEnv, get_final_step_metrics
+from stoix_amd.networks.base import ScannedRNN
+from stoix_amd.networks.distributions import EpsilonGreedy
+from stoix_amd.networks.factory import build_torso
+from stoix_amd.networks.torso import orthogonal_init
+from stoix_amd.ops import multistep
+from stoix_amd.parallel.dist import FlatGradReducer, broadcast_module
+from stoix_amd.systems.anakin import run_anakin_experiment
+
+Tensor = torch.Tensor
+
+_TX_EPS = 1e-3
+
+
+def signed_hyperbolic(x: Tensor) -> Tensor:
+    """h(x) = sign(x)(sqrt(|x|+1)-1) + eps*x (Pohlen et al. 2018)."""
+    return torch.sign(x) * (torch.sqrt(x.abs() + 1.0) - 1.0) + _TX_EPS * x
+
+
+def signed_parabolic(x: Tensor) -> Tensor:
+    """h^-1 for the signed hyperbolic transform."""
+    z = torch.sqrt(1.0 + 4.0 * _TX_EPS * (x.abs() + 1.0 + _TX_EPS)) / (2.0 * _TX_EPS) - 1.0 / (
+        2.0 * _TX_EPS
+    )
+    return torch.sign(x) * (z**2 - 1.0)
+
+
+class RecurrentQNetwork(nn.Module):
+    """pre-torso -> ScannedRNN -> post-torso -> per-action Q values [T,B,A]."""
+
+    def __init__(self, net_cfg: dict, obs_dim: int, num_actions: int):
+        super().__init__()
+        self.pre = build_torso(net_cfg["pre_torso"], obs_dim)
+        rnn_cfg = dict(net_cfg.get("rnn", {}))
+        self.rnn = ScannedRNN(
+            self.pre.output_dim, rnn_cfg.get("hidden_dim", 128), rnn_cfg.get("cell_type", "gru")
+        )
+        self.post = build_torso(net_cfg["post_torso"], self.rnn.hidden_dim)
+        self.head = orthogonal_init(nn.Linear(self.post.output_dim, num_actions), scale=1.0)
+
+    def initial_state(self, batch: int, device) -> list:
+        return self.rnn.initial_state(batch, device)
+
+    def forward(self, obs: Tensor, resets: Tensor, state: list) -> Tuple[Tensor, list]:
+        T, B = obs.shape[:2]
+        z = self.pre(obs.reshape(T * B, -1)).reshape(T, B, -1)
+        h, state = self.rnn(z, resets, state)
+        q = self.head(self.post(h.reshape(T * B, -1))).reshape(T, B, -1)
+        return q, state
+
+
+class R2D2Learner:
+    is_recurrent = True
+
+    def __init__(self, config, env: StatefulVecEnv, device: torch.device):
+        self.cfg = config
+        self.sys = config.system
+        self.env = env
+        self.device = device
+        self.T = int(self.sys.rollout_length)
+        self.B = env.num_envs
+        self.gamma = float(self.sys.gamma)
+        self.n_step = int(self.sys.n_step)
+        self.burn_in = int(self.sys.burn_in_length)
+        self.learn_len = int(self.sys.sample_sequence_length) - self.burn_in
+
+        obs_dim = 1
+        for s in env.observation_space.shape:
+            obs_dim *= s
+        self.num_actions = env.action_space.num_values
+        self.q_online = RecurrentQNetwork(
+            config.network.actor_network, obs_dim, self.num_actions
+        ).to(device)
+        broadcast_module(self.q_online)
+        self.q_target = copy.deepcopy(self.q_online)
+        for p in self.q_target.parameters():
+            p.requires_grad_(False)
+
+        self.opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), eps=1e-5)
+        self.reducer = FlatGradReducer(self.q_online.parameters(), device)
+        self.gen = torch.Generator(device=device)
+        self.gen.manual_seed(int(config.arch.seed) * 7919 + 701)
+
+        self.buffer = PrioritisedBuffer(
+            add_batch_size=self.B,
+            max_length_time_axis=max(
+                int(self.sys.sample_sequence_length) + 1, int(self.sys.buffer_size) // self.B
+            ),
+            sample_sequence_length=int(self.sys.sample_sequence_length),
+            device=device,
+            seed=int(config.arch.seed) + 47,
+            priority_exponent=float(self.sys.priority_exponent),
+        )
+        self.batch_size = int(self.sys.batch_size)
+        self.train_eps = float(self.sys.training_epsilon)
+        self.eta = float(getattr(self.sys, "priority_eta", 0.9))
+
+        self.ts = env.reset()
+        self.hstate = self.q_online.initial_state(self.B, device)
+        self.prev_done = torch.zeros(self.B, dtype=torch.bool, device=device)
+        self.episode_metrics: Dict[str, Tensor] = {}
+        self._warmup()
+
+    # --------------------------------------------------------------- acting
+
+    @property
+    def act_fn(self):
+        learner = self
+
+        class _Act:
+            @staticmethod
+            def initial_state(batch, device):
+                return learner.q_online.initial_state(batch, device)
+
+            def __call__(self, obs, resets, state, greedy):
+                q, state = learner.q_online(obs.unsqueeze(0), resets.unsqueeze(0), state)
+                return q.squeeze(0).argmax(dim=-1), state
+
+        return _Act()
+
+    # -------------------------------------------------------------- rollout
+
+    @torch.no_grad()
+    def _rollout_into_buffer(self, steps: int, random_actions: bool = False) -> None:
+        ts = self.ts
+        obs_l, act_l, rew_l, disc_l, reset_l, h_l = [], [], [], [], [], []
+        state = self.hstate
+        prev_done = self.prev_done
+        for _ in range(steps):
+            obs = ts.observation
+            resets = prev_done
+            h_l.append(state[0] if isinstance(state[0], Tensor) else state[0][0])
+            q, state = self.q_online(obs.unsqueeze(0), resets.unsqueeze(0), state)
+            if random_actions:
+                action = self.env.action_space.sample(self.B, self.device, self.gen)
+            else:
+                action = EpsilonGreedy(q.squeeze(0), self.train_eps).sample(self.gen)
+            next_ts = self.env.step(action)
+            obs_l.append(obs.clone())
+            act_l.append(action)
+            rew_l.append(next_ts.reward)
+            disc_l.append(next_ts.discount)
+            reset_l.append(resets)
+            prev_done = next_ts.last()
+            ts = next_ts
+        self.ts = ts
+        self.hstate = state
+        self.prev_done = prev_done
+        self.buffer.add(
+            {
+                "obs": torch.stack(obs_l, 1),
+                "action": torch.stack(act_l, 1),
+                "reward": torch.stack(rew_l, 1),
+                "discount": torch.stack(disc_l, 1),
+                "resets": torch.stack(reset_l, 1),
+                "hstate": torch.stack(h_l, 1),  # [B, T, H] per-step GRU state
+            }
+        )
+        em = ts.extras["episode_metrics"]
+        final, has = get_final_step_metrics(em)
+        if has:
+            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+
+    def _warmup(self) -> None:
+        steps = max(int(self.sys.sample_sequence_length) + 1, 8)
+        self._rollout_into_buffer(steps, random_actions=True)
+
+    # ---------------------------------------------------------------- update
+
+    def update_step(self) -> Dict[str, Tensor]:
+        self._rollout_into_buffer(self.T)
+        if not self.buffer.can_sample:
+            return {}
+        metrics: Dict[str, Tensor] = {}
+        tau = float(self.sys.tau)
+        for _ in range(int(self.sys.epochs)):
+            batch = self.buffer.sample(self.batch_size)
+            obs = batch["obs"].transpose(0, 1)  # [L, B, D]
+            resets = batch["resets"].transpose(0, 1)
+            act = batch["action"].transpose(0, 1)
+            r = batch["reward"].transpose(0, 1)
+            d = batch["discount"].transpose(0, 1) * self.gamma
+            h0 = [batch["hstate"][:, 0]]  # stored initial GRU state
+
+            bi, ll = self.burn_in, self.learn_len
+            with torch.no_grad():
+                if bi > 0:
+                    _, h_on = self.q_online(obs[:bi], resets[:bi], [h.clone() for h in h0])
+                    _, h_tg = self.q_target(obs[:bi], resets[:bi], [h.clone() for h in h0])
+                else:
+                    h_on = [h.clone() for h in h0]
+                    h_tg = [h.clone() for h in h0]
+                q_tgt_seq, _ = self.q_target(obs[bi:], resets[bi:], [h.clone() for h in h_tg])
+                q_sel_seq, _ = self.q_online(obs[bi:], resets[bi:], [h.clone() for h in h_on])
+
+            q_seq, _ = self.q_online(obs[bi:], resets[bi:], h_on)
+            # learn segment excludes the last step (it only provides bootstrap)
+            q_learn = q_seq[:-1]
+            act_learn = act[bi:-1]
+            q_a = q_learn.gather(-1, act_learn.long().unsqueeze(-1)).squeeze(-1)
+
+            with torch.no_grad():
+                best = q_sel_seq[1:].argmax(dim=-1, keepdim=True)
+                q_boot = q_tgt_seq[1:].gather(-1, best).squeeze(-1)  # [Tl, B]
+                v_boot = signed_parabolic(q_boot)
+                targets_raw = multistep.batch_n_step_bootstrapped_returns(
+                    r[bi:-1], d[bi:-1], v_boot, self.n_step
+                )
+                targets = signed_hyperbolic(targets_raw)
+            td = q_a - targets
+            loss = (0.5 * td**2 * batch["_weights"].unsqueeze(0)).mean()
+
+            self.opt.zero_grad(set_to_none=True)
+            loss.backward()
+            self.reducer.reduce()
+            self.reducer.wait()
+            if getattr(self.sys, "max_grad_norm", None):
+                nn.utils.clip_grad_norm_(self.q_online.parameters(), float(self.sys.max_grad_norm))
+            self.opt.step()
+            with torch.no_grad():
+                abs_td = td.abs()
+                prio = self.eta * abs_td.max(dim=0).values + (1 - self.eta) * abs_td.mean(dim=0)
+                self.buffer.set_priorities(batch["_slots"], prio)
+                for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
+                    pt.mul_(1 - tau).add_(po, alpha=tau)
+            metrics = {"q_loss": loss.detach(), "mean_abs_td": abs_td.mean().detach()}
+        return metrics
+
+    def state_for_checkpoint(self):
+        return {"q_online": dict(self.q_online.state_dict())}
+
+    def snapshot_params(self):
+        return {"q_online": {k: v.clone() for k, v in self.q_online.state_dict().items()}}
+
+    def load_params(self, snap) -> None:
+        self.q_online.load_state_dict(snap["q_online"])
+
+
+def learner_factory(config, env, device) -> R2D2Learner:
+    return R2D2Learner(config, env, device)
+
+
+def run(config) -> float:
+    return run_anakin_experiment(config, learner_factory)
+
+
+def hydra_entry_point(argv=None) -> float:
+    cfg = compose("default/anakin/default_rec_r2d2.yaml", argv if argv is not None else sys.argv[1:])
+    return run(cfg)
+
+
+if __name__ == "__main__":
+    hydra_entry_point()

@@ -66,6 +66,10 @@ SYSTEMS = [
       "system.warmup_steps=16", "system.n_step=3"]),
     ("stoix_amd.systems.ppo.rec_ppo", "default/anakin/default_rec_ppo.yaml",
      ["system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.q_learning.rec_r2d2", "default/anakin/default_rec_r2d2.yaml",
+     ["system.epochs=1", "system.batch_size=8", "system.buffer_size=512",
+      "system.sample_sequence_length=8", "system.burn_in_length=2", "system.n_step=2",
+      "system.rollout_length=10"]),
 ]
 
 
