@@ -1,0 +1,170 @@
+"""Sebulba plumbing: thread lifetimes, rollout pipeline, parameter server,
+async evaluator.
+
+Parity with /root/reference/stoix/utils/sebulba_utils.py: ThreadLifetime
+(:20-45), OnPolicyPipeline with bounded per-actor queues where the learner
+blocks collecting one payload from EVERY actor (:48-96), ParameterServer
+with per-actor queues and None as shutdown sentinel (:99-259),
+AsyncEvaluatorBase tracking best params (:262-367).
+
+MI355X design note: the reference shards each actor's trajectory across the
+learner pmap devices in-process. Here one process drives one learner GPU
+(multi-GPU learners scale via torchrun + RCCL exactly like Anakin), so the
+pipeline carries whole trajectories; H2D staging happens on a dedicated copy
+stream with pinned host buffers (PinnedCopier) so the learner's compute
+overlaps the next payload's transfer.
+"""
+from __future__ import annotations
+
+import queue
+import threading
+from typing import Any, Callable, Dict, List, Optional
+
+import torch
+
+
+class ThreadLifetime:
+    def __init__(self):
+        self._stop = threading.Event()
+
+    def stop(self) -> None:
+        self._stop.set()
+
+    def should_stop(self) -> bool:
+        return self._stop.is_set()
+
+
+class OnPolicyPipeline:
+    """Per-actor bounded queues; collect() waits for one rollout payload from
+    every actor — the synchronisation barrier of the reference design."""
+
+    def __init__(self, num_actors: int, maxsize: int = 1):
+        self.queues: List[queue.Queue] = [queue.Queue(maxsize=maxsize) for _ in range(num_actors)]
+
+    def send_rollout(self, actor_id: int, payload: Any, lifetime: ThreadLifetime) -> None:
+        while not lifetime.should_stop():
+            try:
+                self.queues[actor_id].put(payload, timeout=0.1)
+                return
+            except queue.Full:
+                continue
+
+    def collect_rollouts(self, lifetime: ThreadLifetime) -> Optional[List[Any]]:
+        out: List[Any] = []
+        for q in self.queues:
+            while not lifetime.should_stop():
+                try:
+                    out.append(q.get(timeout=0.1))
+                    break
+                except queue.Empty:
+                    continue
+            else:
+                return None
+        return out
+
+
+class ParameterServer:
+    """Latest-params mailbox per actor (queue of size 1, newest wins)."""
+
+    def __init__(self, num_actors: int):
+        self.queues: List[queue.Queue] = [queue.Queue(maxsize=1) for _ in range(num_actors)]
+
+    def distribute_params(self, params: Dict[str, Any]) -> None:
+        for q in self.queues:
+            # drop the stale copy if the actor has not fetched it yet
+            try:
+                q.get_nowait()
+            except queue.Empty:
+                pass
+            try:
+                q.put_nowait(params)
+            except queue.Full:
+                pass
+
+    def get_params(self, actor_id: int, block: bool = False, timeout: float = 1.0) -> Optional[Dict[str, Any]]:
+        try:
+            return self.queues[actor_id].get(block=block, timeout=timeout if block else None)
+        except queue.Empty:
+            return None
+
+    def shutdown(self) -> None:
+        for q in self.queues:
+            try:
+                q.put_nowait(None)
+            except queue.Full:
+                pass
+
+
+class AsyncEvaluator:
+    """Evaluation thread fed by a queue of (params, t_env); skips rather than
+    blocks on overflow (reference sebulba_utils.py:315-317); tracks best."""
+
+    def __init__(self, evaluate_fn: Callable, lifetime: ThreadLifetime):
+        self.evaluate_fn = evaluate_fn
+        self.lifetime = lifetime
+        self.queue: queue.Queue = queue.Queue(maxsize=2)
+        self.best_return = float("-inf")
+        self.best_params: Optional[Dict] = None
+        self.last_metrics: Dict = {}
+        self.thread = threading.Thread(target=self._run, daemon=True, name="async-evaluator")
+        self.thread.start()
+
+    def submit_evaluation(self, params: Dict, t_env: int) -> bool:
+        try:
+            self.queue.put_nowait((params, t_env))
+            return True
+        except queue.Full:
+            return False
+
+    def _run(self) -> None:
+        while not self.lifetime.should_stop():
+            try:
+                item = self.queue.get(timeout=0.2)
+            except queue.Empty:
+                continue
+            if item is None:
+                return
+            params, t_env = item
+            metrics = self.evaluate_fn(params, t_env)
+            self.last_metrics = metrics
+            mean_ret = float(metrics["episode_return"].mean())
+            if mean_ret >= self.best_return:
+                self.best_return = mean_ret
+                self.best_params = params
+
+    def join(self) -> None:
+        try:
+            self.queue.put_nowait(None)
+        except queue.Full:
+            pass
+        self.thread.join(timeout=30)
+
+
+class PinnedCopier:
+    """Host->device staging through reused pinned buffers on a dedicated HIP
+    stream — the actor->learner trajectory scatter of the reference
+    (device_put_sharded, sebulba ff_ppo.py:263-266) as hipMemcpyAsync
+    overlapped with learner compute."""
+
+    def __init__(self, device: torch.device):
+        self.device = device
+        self.stream = torch.cuda.Stream(device) if device.type == "cuda" else None
+        self._pinned: Dict[str, torch.Tensor] = {}
+
+    def to_device(self, payload: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+        if self.device.type != "cuda":
+            return payload
+        out: Dict[str, torch.Tensor] = {}
+        with torch.cuda.stream(self.stream):
+            for k, v in payload.items():
+                if not isinstance(v, torch.Tensor) or v.is_cuda:
+                    out[k] = v
+                    continue
+                buf = self._pinned.get(k)
+                if buf is None or buf.shape != v.shape or buf.dtype != v.dtype:
+                    buf = torch.empty_like(v, pin_memory=True)
+                    self._pinned[k] = buf
+                buf.copy_(v)
+                out[k] = buf.to(self.device, non_blocking=True)
+        torch.cuda.current_stream(self.device).wait_stream(self.stream)
+        return out
