@@ -70,6 +70,12 @@ SYSTEMS = [
      ["system.epochs=1", "system.batch_size=8", "system.buffer_size=512",
       "system.sample_sequence_length=8", "system.burn_in_length=2", "system.n_step=2",
       "system.rollout_length=10"]),
+    ("stoix_amd.systems.search.ff_az", "default/anakin/default_ff_az.yaml",
+     ["system.num_simulations=6", "system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.search.ff_mz", "default/anakin/default_ff_mz.yaml",
+     ["system.num_simulations=4", "system.epochs=1", "system.unroll_steps=2",
+      "system.n_step=2", "system.batch_size=8", "system.buffer_size=256",
+      "system.rollout_length=8"]),
 ]
 
 
