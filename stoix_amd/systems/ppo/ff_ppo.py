@@ -161,6 +161,12 @@ class PPOLearner:
             if has:
                 self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
+    # ---------------------------------------------------------------- losses
+
+    def policy_loss(self, new_logp: Tensor, old_logp: Tensor, adv: Tensor) -> Tensor:
+        """PPO clipped surrogate; subclasses override (penalty / drift)."""
+        return ppo_clip_loss(new_logp, old_logp, adv, float(self.sys.clip_eps))
+
     # ---------------------------------------------------------------- update
 
     def update_step(self) -> Dict[str, Tensor]:
@@ -224,7 +230,7 @@ class PPOLearner:
                     new_logp = dist.log_prob(flat_action[idx])
                     entropy = dist.entropy().mean()
                     value = self.critic(obs_mb)
-                a_loss = ppo_clip_loss(new_logp, flat_logp[idx], flat_adv[idx], float(self.sys.clip_eps))
+                a_loss = self.policy_loss(new_logp, flat_logp[idx], flat_adv[idx])
                 actor_loss = a_loss - float(self.sys.ent_coef) * entropy
 
                 v_loss = clipped_value_loss(value, flat_value[idx], flat_targets[idx], float(self.sys.clip_eps))

@@ -24,6 +24,12 @@ SYSTEMS = [
      ["system.num_minibatches=2", "system.epochs=1"]),
     ("stoix_amd.systems.ppo.ff_ppo", "default/anakin/default_ff_ppo_continuous.yaml",
      ["system.num_minibatches=2", "system.epochs=1", "env=classic/pendulum"]),
+    ("stoix_amd.systems.ppo.ff_ppo_penalty", "default/anakin/default_ff_ppo_penalty.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.ppo.ff_ppo_penalty", "default/anakin/default_ff_ppo_penalty_continuous.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.ppo.ff_dpo", "default/anakin/default_ff_dpo_continuous.yaml",
+     ["system.num_minibatches=2", "system.epochs=1"]),
     ("stoix_amd.systems.q_learning.ff_dqn", "default/anakin/default_ff_dqn.yaml",
      ["system.epochs=2", "system.batch_size=16", "system.buffer_size=512", "system.warmup_steps=16"]),
     ("stoix_amd.systems.q_learning.ff_ddqn", "default/anakin/default_ff_ddqn.yaml",
