@@ -48,6 +48,38 @@ class HipStepMixin:
     def _hip_action(self, action: torch.Tensor) -> torch.Tensor:
         return action
 
+    def hip_step_into(
+        self,
+        action: torch.Tensor,
+        reward_out: torch.Tensor,
+        discount_out: torch.Tensor,
+        steptype_out: torch.Tensor,
+    ) -> None:
+        """Fused-rollout variant of step(): the kernel writes reward /
+        discount / step-type straight into the caller's rollout storage
+        (e.g. buf_reward[t]) instead of the stable _hb buffers — no
+        per-step copy kernels. obs/next_obs still land in _hb."""
+        hb = self._hb
+        kern = getattr(self._hip, self.HIP_KERNEL)
+        kern(
+            self._state[self.STATE_KEY],
+            self._hip_action(action),
+            self._step_count,
+            self._ep_return,
+            self._ep_length,
+            self._last_ep_return,
+            self._last_ep_length,
+            hb["obs"],
+            hb["next_obs"],
+            reward_out,
+            discount_out,
+            steptype_out,
+            hb["done"],
+            self.max_episode_steps,
+            self._hip_seed,
+            hb["draw"],
+        )
+
     def step(self, action: torch.Tensor) -> TimeStep:  # type: ignore[override]
         if getattr(self, "_hip", None) is None:
             return super().step(action)  # type: ignore[misc]

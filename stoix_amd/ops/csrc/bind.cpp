@@ -31,7 +31,35 @@ void launch_offpolicy_returns(const float*, const float*, const float*,
                               void*);
 void launch_fused_adam(float*, const float*, float*, float*, float*, long*,
                        long, float, float, float, float, float, void*);
+void launch_fused_adam_bf16(float*, const void*, float*, float*, float*,
+                            long*, void*, long, float, float, float, float,
+                            float, float, void*);
 void launch_polyak(const float*, float*, long, float, void*);
+void launch_mfma_probe(const void*, const void*, float*, float*, void*);
+void launch_policy_value_step(const float*, const void*, const float*,
+                              const void*, const float*, const void*,
+                              const float*, const void*, const float*,
+                              const void*, const float*, const void*,
+                              const float*, float*, float*, float*, float*,
+                              const float*, const float*, int, int, int, int,
+                              float, float, float, float, int, uint64_t,
+                              unsigned int*, void*);
+void launch_value_forward(const float*, const void*, const float*,
+                          const void*, const float*, const void*,
+                          const float*, float*, const float*, const float*,
+                          int, int, int, void*);
+void launch_silu_fwd(const void*, void*, long, void*);
+void launch_silu_bwd(const void*, const void*, void*, long, void*);
+void launch_ppo_gather(const long*, int, const float*, int, const float*, int,
+                       const float*, const float*, const float*, const float*,
+                       void*, float*, float*, float*, float*, float*,
+                       const float*, const float*, void*);
+void launch_ppo_head_loss(const void*, const void*, const void*, const float*,
+                          const void*, const float*, const float*,
+                          const float*, const float*, const float*,
+                          const float*, void*, void*, void*, void*, float*,
+                          int, int, int, float, float, float, float, float,
+                          float, float, uint64_t, unsigned int*, void*);
 }
 
 namespace {
@@ -159,6 +187,149 @@ void polyak(torch::Tensor online, torch::Tensor target, double tau) {
                 online.numel(), (float)tau, cur_stream());
 }
 
+void fused_adam_bf16(torch::Tensor param, torch::Tensor grad,
+                     torch::Tensor exp_avg, torch::Tensor exp_avg_sq,
+                     torch::Tensor sqnorm, torch::Tensor step_t,
+                     torch::Tensor param_bf16, double lr, double beta1,
+                     double beta2, double eps, double max_norm,
+                     double grad_scale) {
+  CHK(param, torch::kFloat32);
+  CHK(grad, torch::kBFloat16);
+  void* pbf = param_bf16.numel() > 0 ? param_bf16.data_ptr() : nullptr;
+  launch_fused_adam_bf16(param.data_ptr<float>(), grad.data_ptr(),
+                         exp_avg.data_ptr<float>(),
+                         exp_avg_sq.data_ptr<float>(),
+                         sqnorm.data_ptr<float>(), step_t.data_ptr<long>(),
+                         pbf, param.numel(), (float)lr, (float)beta1,
+                         (float)beta2, (float)eps, (float)max_norm,
+                         (float)grad_scale, cur_stream());
+}
+
+void mfma_probe(torch::Tensor A, torch::Tensor B, torch::Tensor D0,
+                torch::Tensor D1) {
+  CHK(A, torch::kBFloat16);
+  launch_mfma_probe(A.data_ptr(), B.data_ptr(), D0.data_ptr<float>(),
+                    D1.data_ptr<float>(), cur_stream());
+}
+
+const float* fptr_or_null(const torch::Tensor& t) {
+  return t.numel() > 0 ? t.data_ptr<float>() : nullptr;
+}
+
+void policy_value_step(torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
+                       torch::Tensor W2a, torch::Tensor b2a, torch::Tensor Wha,
+                       torch::Tensor bha, torch::Tensor W1c, torch::Tensor b1c,
+                       torch::Tensor W2c, torch::Tensor b2c, torch::Tensor Wvc,
+                       torch::Tensor bvc, torch::Tensor obs_mirror,
+                       torch::Tensor action_out, torch::Tensor logp_out,
+                       torch::Tensor value_out, torch::Tensor nmean,
+                       torch::Tensor nvar, double min_scale, double aff_scale,
+                       double aff_shift, double log_aff_scale, int64_t greedy,
+                       int64_t seed, torch::Tensor draw_buf) {
+  CHK(obs, torch::kFloat32);
+  CHK(W1a, torch::kBFloat16);
+  CHK(W2a, torch::kBFloat16);
+  int B = obs.size(0), OBS = obs.size(1);
+  int HID = W2a.size(0);
+  int ACT = action_out.size(1);
+  TORCH_CHECK(HID == 256 || HID == 128, "fused MLP supports HID 128/256");
+  TORCH_CHECK(ACT <= 8, "fused policy head supports ACT <= 8");
+  TORCH_CHECK(OBS <= 128, "fused MLP supports OBS <= 128");
+  float* om = obs_mirror.numel() > 0 ? obs_mirror.data_ptr<float>() : nullptr;
+  unsigned int* db = draw_buf.numel() > 0
+                         ? (unsigned int*)draw_buf.data_ptr<int>()
+                         : nullptr;
+  launch_policy_value_step(
+      obs.data_ptr<float>(), W1a.data_ptr(), b1a.data_ptr<float>(),
+      W2a.data_ptr(), b2a.data_ptr<float>(), Wha.data_ptr(),
+      bha.data_ptr<float>(), W1c.data_ptr(), b1c.data_ptr<float>(),
+      W2c.data_ptr(), b2c.data_ptr<float>(), Wvc.data_ptr(),
+      bvc.data_ptr<float>(), om, action_out.data_ptr<float>(),
+      logp_out.data_ptr<float>(), value_out.data_ptr<float>(),
+      fptr_or_null(nmean), fptr_or_null(nvar), B, OBS, ACT, HID,
+      (float)min_scale, (float)aff_scale, (float)aff_shift,
+      (float)log_aff_scale, (int)greedy, (uint64_t)seed, db, cur_stream());
+}
+
+void value_forward(torch::Tensor obs, torch::Tensor W1c, torch::Tensor b1c,
+                   torch::Tensor W2c, torch::Tensor b2c, torch::Tensor Wvc,
+                   torch::Tensor bvc, torch::Tensor value_out,
+                   torch::Tensor nmean, torch::Tensor nvar) {
+  CHK(obs, torch::kFloat32);
+  CHK(W1c, torch::kBFloat16);
+  int B = obs.size(0), OBS = obs.size(1);
+  int HID = W2c.size(0);
+  launch_value_forward(obs.data_ptr<float>(), W1c.data_ptr(),
+                       b1c.data_ptr<float>(), W2c.data_ptr(),
+                       b2c.data_ptr<float>(), Wvc.data_ptr(),
+                       bvc.data_ptr<float>(), value_out.data_ptr<float>(),
+                       fptr_or_null(nmean), fptr_or_null(nvar), B, OBS, HID,
+                       cur_stream());
+}
+
+void silu_fwd(torch::Tensor z, torch::Tensor h) {
+  CHK(z, torch::kBFloat16);
+  TORCH_CHECK(z.numel() % 8 == 0, "silu_fwd needs numel % 8 == 0");
+  launch_silu_fwd(z.data_ptr(), h.data_ptr(), z.numel(), cur_stream());
+}
+
+void silu_bwd(torch::Tensor dh, torch::Tensor z, torch::Tensor dz) {
+  CHK(z, torch::kBFloat16);
+  TORCH_CHECK(z.numel() % 8 == 0, "silu_bwd needs numel % 8 == 0");
+  launch_silu_bwd(dh.data_ptr(), z.data_ptr(), dz.data_ptr(), z.numel(),
+                  cur_stream());
+}
+
+void ppo_gather(torch::Tensor idx, torch::Tensor obs, torch::Tensor action,
+                torch::Tensor logp, torch::Tensor value, torch::Tensor adv,
+                torch::Tensor targets, torch::Tensor obs_out,
+                torch::Tensor action_out, torch::Tensor logp_out,
+                torch::Tensor value_out, torch::Tensor adv_out,
+                torch::Tensor targets_out, torch::Tensor nmean,
+                torch::Tensor nvar) {
+  CHK(idx, torch::kInt64);
+  CHK(obs, torch::kFloat32);
+  CHK(obs_out, torch::kBFloat16);
+  int mb = idx.numel();
+  launch_ppo_gather(idx.data_ptr<long>(), mb, obs.data_ptr<float>(),
+                    obs.size(1), action.data_ptr<float>(), action.size(1),
+                    logp.data_ptr<float>(), value.data_ptr<float>(),
+                    adv.data_ptr<float>(), targets.data_ptr<float>(),
+                    obs_out.data_ptr(), action_out.data_ptr<float>(),
+                    logp_out.data_ptr<float>(), value_out.data_ptr<float>(),
+                    adv_out.data_ptr<float>(), targets_out.data_ptr<float>(),
+                    fptr_or_null(nmean), fptr_or_null(nvar), cur_stream());
+}
+
+void ppo_head_loss(torch::Tensor H2a, torch::Tensor H2c, torch::Tensor Wha,
+                   torch::Tensor bha, torch::Tensor Wvc, torch::Tensor bvc,
+                   torch::Tensor action, torch::Tensor old_logp,
+                   torch::Tensor old_value, torch::Tensor adv,
+                   torch::Tensor targets, torch::Tensor dH2a,
+                   torch::Tensor dH2c, torch::Tensor dhead, torch::Tensor dv,
+                   torch::Tensor metrics, double clip_eps, double ent_coef,
+                   double vf_coef, double min_scale, double aff_scale,
+                   double aff_shift, double log_aff_scale, int64_t seed,
+                   torch::Tensor draw_buf) {
+  CHK(H2a, torch::kBFloat16);
+  CHK(action, torch::kFloat32);
+  int B = H2a.size(0), HID = H2a.size(1), ACT = action.size(1);
+  TORCH_CHECK(HID % 64 == 0 && HID <= 512, "ppo_head_loss HID constraint");
+  unsigned int* db = draw_buf.numel() > 0
+                         ? (unsigned int*)draw_buf.data_ptr<int>()
+                         : nullptr;
+  launch_ppo_head_loss(
+      H2a.data_ptr(), H2c.data_ptr(), Wha.data_ptr(), bha.data_ptr<float>(),
+      Wvc.data_ptr(), bvc.data_ptr<float>(), action.data_ptr<float>(),
+      old_logp.data_ptr<float>(), old_value.data_ptr<float>(),
+      adv.data_ptr<float>(), targets.data_ptr<float>(), dH2a.data_ptr(),
+      dH2c.data_ptr(), dhead.data_ptr(), dv.data_ptr(),
+      metrics.data_ptr<float>(), B, HID, ACT, (float)clip_eps,
+      (float)ent_coef, (float)vf_coef, (float)min_scale, (float)aff_scale,
+      (float)aff_shift, (float)log_aff_scale, (uint64_t)seed, db,
+      cur_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -170,5 +341,16 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vtrace", &vtrace, "vtrace errors + pg advantage");
   m.def("offpolicy_returns", &offpolicy_returns, "retrace-style returns");
   m.def("fused_adam", &fused_adam, "fused global-norm-clip + Adam");
+  m.def("fused_adam_bf16", &fused_adam_bf16,
+        "fused clip + Adam, bf16 grads + bf16 param mirror");
   m.def("polyak", &polyak, "polyak target update");
+  m.def("mfma_probe", &mfma_probe, "MFMA 16x16x32 bf16 layout probe");
+  m.def("policy_value_step", &policy_value_step,
+        "fused actor+critic fwd + tanh-normal sample (MFMA)");
+  m.def("value_forward", &value_forward, "fused critic fwd (MFMA)");
+  m.def("silu_fwd", &silu_fwd, "bf16 silu forward");
+  m.def("silu_bwd", &silu_bwd, "bf16 silu backward");
+  m.def("ppo_gather", &ppo_gather, "fused minibatch gather");
+  m.def("ppo_head_loss", &ppo_head_loss,
+        "fused PPO head fwd + losses + analytic head bwd");
 }

@@ -1,0 +1,718 @@
+// Fused MLP kernels for the Anakin hot path (K2/K3/K6/K7 of SURVEY.md §2.9).
+//
+// The reference's XLA fuses scan(env.step ∘ net.apply) into few kernels
+// (/root/reference/stoix/systems/ppo/anakin/ff_ppo.py:118-146); the eager
+// PyTorch equivalent issues ~45 kernels per rollout step. These kernels
+// collapse the canonical actor/critic MLP (obs -> Linear(H) -> SiLU ->
+// Linear(H) -> SiLU -> heads) into:
+//   * policy_value_step_kernel: ONE launch for actor fwd + tanh-normal
+//     sample + log-prob + critic fwd over the whole env batch, using
+//     v_mfma_f32_16x16x32_bf16 matrix cores with LDS-staged activations;
+//   * value_forward_kernel: critic-only fwd (bootstrap values / eval);
+//   * ppo_fused_head_loss_kernel: per-row head forward + PPO losses +
+//     analytic head backward for the update phase (one wave per sample);
+//   * silu fwd/bwd + gather kernels: the epilogue/prologue glue so the
+//     update phase is GEMM (hipBLASLt MFMA) + a handful of fused kernels.
+//
+// Geometry (policy/value forward): one workgroup = 4 waves = 64 batch rows;
+// each wave owns a 16-row M-tile and computes every 16-col N-tile of each
+// layer with mfma_f32_16x16x32_bf16, staging activations in LDS (row-major,
+// +8 col pad -> conflict-free ds_read_b128 A-fragments, §2 of the CDNA4
+// guide). Weights are read straight from L2 (nn.Linear row-major [N,K]
+// bf16 mirrors; they are hot across the 128 sequential rollout steps).
+// Per-wave row blocks are private, so the kernels need NO __syncthreads().
+#include "common.h"
+#include <hip/hip_bf16.h>
+
+typedef __bf16 bf16_t;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
+
+DEV_INLINE float bf2f(bf16_t x) { return (float)x; }
+DEV_INLINE bf16_t f2bf(float x) { return (bf16_t)x; }
+
+DEV_INLINE float silu_f(float x) {
+  float s = 1.0f / (1.0f + __expf(-x));
+  return x * s;
+}
+DEV_INLINE float silu_grad_f(float x) {
+  float s = 1.0f / (1.0f + __expf(-x));
+  return s * (1.0f + x * (1.0f - s));
+}
+DEV_INLINE float softplus_f(float x) {
+  // log1p(exp(x)) stable
+  return (x > 20.0f) ? x : log1pf(__expf(x));
+}
+
+// ---------------------------------------------------------------- probe
+// Validates the assumed lane->element mapping of v_mfma_f32_16x16x32_bf16.
+// D = A[16,32] @ B[32,16]; two candidate A/B layouts:
+//  v0: lane l holds A[m=l&15][k=(l>>4)*8+j], B[k=(l>>4)*8+j][n=l&15]
+//  v1: lane l holds A[m=l&15][k=(l>>4)*4+(j&3)+16*(j>>2)] (split-K halves)
+// C/D (both): D[row=(l>>4)*4+r][col=l&15] from acc[r].
+extern "C" __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
+                                             const bf16_t* __restrict__ B,
+                                             float* __restrict__ D0,
+                                             float* __restrict__ D1) {
+  int l = threadIdx.x;
+  if (l >= 64) return;
+  int m = l & 15, g = l >> 4;
+  bf16x8 a0, b0, a1, b1;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    int k0 = g * 8 + j;
+    a0[j] = A[m * 32 + k0];
+    b0[j] = B[k0 * 16 + m];
+    int k1 = g * 4 + (j & 3) + 16 * (j >> 2);
+    a1[j] = A[m * 32 + k1];
+    b1[j] = B[k1 * 16 + m];
+  }
+  f32x4 z = {0.f, 0.f, 0.f, 0.f};
+  f32x4 d0 = MFMA_BF16_16x16x32(a0, b0, z, 0, 0, 0);
+  f32x4 d1 = MFMA_BF16_16x16x32(a1, b1, z, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    D0[(g * 4 + r) * 16 + m] = d0[r];
+    D1[(g * 4 + r) * 16 + m] = d1[r];
+  }
+}
+
+// ------------------------------------------------- fused policy/value fwd
+//
+// LDS: one __shared__ array (guide §5.5 trap 4a), two regions per wave:
+//   O: obs tile      [4 waves][16 rows][K1P<=128 + 8 pad] bf16
+//   H: activations   [4 waves][16 rows][HID + 8 pad]      bf16
+// A-fragment read (ds_read_b128): lane l -> row l&15, k0=(l>>4)*8.
+// B-fragment read (global, row-major [N,K] bf16): lane l -> n=nt*16+(l&15),
+// k0=ks*32+(l>>4)*8 -> 16B contiguous in K.
+
+#define K1P_MAX 128
+#define OPAD 8
+#define HPAD 8
+
+template <int HID>
+struct MlpLds {
+  bf16_t O[4][16][K1P_MAX + OPAD];
+  bf16_t H[4][16][HID + HPAD];
+};
+
+// Load one B fragment from a row-major [N,K] bf16 weight matrix.
+template <int HID>
+DEV_INLINE bf16x8 load_w_frag(const bf16_t* __restrict__ W, int K, int nt,
+                              int ks, int lane) {
+  int n = nt * 16 + (lane & 15);
+  int k0 = ks * 32 + (lane >> 4) * 8;
+  const bf16x8* p = reinterpret_cast<const bf16x8*>(W + (long)n * K + k0);
+  return *p;
+}
+
+// One MLP layer for a single wave's 16-row tile: reads A fragments from
+// `src` (row-major bf16 with row stride `sstride`), weights W [HID x K],
+// bias fp32 [HID], writes silu(out) to dst (stride dstride) as bf16.
+// KS = K/32 (runtime), NT = HID/16 (compile-time).
+template <int HID, bool ACT_SILU>
+DEV_INLINE void wave_layer(const bf16_t* __restrict__ src, int sstride,
+                           const bf16_t* __restrict__ W,
+                           const float* __restrict__ bias, int K,
+                           bf16_t* __restrict__ dst, int dstride, int lane) {
+  constexpr int NT = HID / 16;
+  const int KS = K / 32;
+  f32x4 acc[NT];
+#pragma unroll
+  for (int nt = 0; nt < NT; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+  const int arow = lane & 15;
+  const int ak0 = (lane >> 4) * 8;
+  for (int ks = 0; ks < KS; ++ks) {
+    const bf16x8 a =
+        *reinterpret_cast<const bf16x8*>(src + arow * sstride + ks * 32 + ak0);
+#pragma unroll
+    for (int nt = 0; nt < NT; ++nt) {
+      bf16x8 b = load_w_frag<HID>(W, K, nt, ks, lane);
+      acc[nt] = MFMA_BF16_16x16x32(a, b, acc[nt], 0, 0, 0);
+    }
+  }
+  // epilogue: bias + activation, write bf16 to dst LDS region
+  const int col = lane & 15;
+  const int g = lane >> 4;
+#pragma unroll
+  for (int nt = 0; nt < NT; ++nt) {
+    float b = bias[nt * 16 + col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float v = acc[nt][r] + b;
+      if (ACT_SILU) v = silu_f(v);
+      dst[(g * 4 + r) * dstride + nt * 16 + col] = f2bf(v);
+    }
+  }
+}
+
+// Stage one wave's 16 obs rows into LDS as bf16 (zero-padded to K1P), with
+// optional Welford normalisation (mean/var fp32, reference
+// running_statistics.py:205-...); optionally mirror the raw fp32 obs into
+// buf_obs[t] (rollout storage) so no separate copy kernel is needed.
+DEV_INLINE void stage_obs(const float* __restrict__ obs, int rbase, int OBS,
+                          int K1P, bf16_t* __restrict__ Orow, int ostride,
+                          const float* __restrict__ nmean,
+                          const float* __restrict__ nvar,
+                          float* __restrict__ obs_mirror, int lane) {
+  for (int idx = lane; idx < 16 * K1P; idx += 64) {
+    int r = idx / K1P, k = idx - r * K1P;
+    float v = 0.0f;
+    if (k < OBS) {
+      v = obs[(long)(rbase + r) * OBS + k];
+      if (obs_mirror) obs_mirror[(long)(rbase + r) * OBS + k] = v;
+      if (nmean) {
+        float sd = sqrtf(fmaxf(nvar[k], 1e-6f));
+        v = (v - nmean[k]) / sd;
+        v = fmaxf(-10.0f, fminf(10.0f, v));
+      }
+    }
+    Orow[r * ostride + k] = f2bf(v);
+  }
+}
+
+// Critic scalar head via VALU: value[row] = dot(H[row], Wv) + bv.
+// 4 lanes per row (lane l: row l>>2, part l&3), butterfly-reduced.
+template <int HID>
+DEV_INLINE void wave_value_head(const bf16_t* __restrict__ H, int hstride,
+                                const bf16_t* __restrict__ Wv, float bv,
+                                int rbase, float* __restrict__ value_out,
+                                int lane) {
+  int row = lane >> 2;   // 0..15
+  int part = lane & 3;   // 0..3
+  float acc = 0.0f;
+  for (int c = part * (HID / 4); c < (part + 1) * (HID / 4); c += 8) {
+    bf16x8 h = *reinterpret_cast<const bf16x8*>(H + row * hstride + c);
+    bf16x8 w = *reinterpret_cast<const bf16x8*>(Wv + c);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) acc += bf2f(h[j]) * bf2f(w[j]);
+  }
+  acc += __shfl_xor(acc, 1);
+  acc += __shfl_xor(acc, 2);
+  if (part == 0 && value_out) value_out[rbase + row] = acc + bv;
+}
+
+// The fused policy step. Writes, for rollout step t:
+//   buf_obs[t]    = obs                    (fp32 mirror, [B,OBS])
+//   action_out    = tanh-normal sample     ([B,ACT] fp32, env input + buffer)
+//   logp_out      = log pi(a|s)            ([B] fp32)
+//   value_out     = V(s)                   ([B] fp32)
+// Head packing: Wh = cat([loc.weight, scale.weight]) [16,HID] bf16,
+// bh = cat([loc.bias, scale.bias]) [16] fp32. ACT <= 8.
+template <int HID>
+__launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
+    const float* __restrict__ obs,        // [B, OBS]
+    const bf16_t* __restrict__ W1a, const float* __restrict__ b1a,
+    const bf16_t* __restrict__ W2a, const float* __restrict__ b2a,
+    const bf16_t* __restrict__ Wha, const float* __restrict__ bha,
+    const bf16_t* __restrict__ W1c, const float* __restrict__ b1c,
+    const bf16_t* __restrict__ W2c, const float* __restrict__ b2c,
+    const bf16_t* __restrict__ Wvc, const float* __restrict__ bvc,  // [HID],[1]
+    float* __restrict__ obs_mirror,       // [B, OBS] or null
+    float* __restrict__ action_out,       // [B, ACT]
+    float* __restrict__ logp_out,         // [B]
+    float* __restrict__ value_out,        // [B]
+    const float* __restrict__ nmean, const float* __restrict__ nvar,
+    int B, int OBS, int ACT, float min_scale, float aff_scale,
+    float aff_shift, float log_aff_scale, int greedy,
+    uint64_t seed, const unsigned int* __restrict__ draw_buf) {
+  __shared__ MlpLds<HID> lds;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int rbase = blockIdx.x * 64 + wid * 16;
+  if (rbase >= B) return;
+  const uint32_t draw = draw_buf ? *draw_buf : 0u;
+  const int K1P = (OBS + 31) & ~31;
+
+  bf16_t* O = &lds.O[wid][0][0];
+  bf16_t* H = &lds.H[wid][0][0];
+  constexpr int OS = K1P_MAX + OPAD;
+  constexpr int HS = HID + HPAD;
+
+  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, obs_mirror, lane);
+  // within-wave LDS write->read ordering is handled by the compiler's
+  // lgkmcnt tracking (no cross-wave sharing anywhere in this kernel).
+
+  // ---- actor torso
+  wave_layer<HID, true>(O, OS, W1a, b1a, K1P, H, HS, lane);
+  wave_layer<HID, true>(H, HS, W2a, b2a, HID, H, HS, lane);
+
+  // ---- actor head: one 16-col N-tile = [loc(0:ACT) | pad | scale(8:8+ACT)]
+  {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15, ak0 = (lane >> 4) * 8;
+    for (int ks = 0; ks < HID / 32; ++ks) {
+      const bf16x8 a =
+          *reinterpret_cast<const bf16x8*>(H + arow * HS + ks * 32 + ak0);
+      bf16x8 b = load_w_frag<HID>(Wha, HID, 0, ks, lane);
+      acc = MFMA_BF16_16x16x32(a, b, acc, 0, 0, 0);
+    }
+    const int col = lane & 15;  // col<8: loc dim, col>=8: scale dim col-8
+    const int g = lane >> 4;
+    float bh = bha[col];
+    float out[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) out[r] = acc[r] + bh;
+    // pair loc (lane c) with scale (lane c+8): shfl within the 16-lane group
+    const int src_lane = (lane & 48) | (((lane & 15) + 8) & 15);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float loc = out[r];
+      float spre = __shfl(out[r], src_lane, 64);
+      float logp = 0.0f;
+      float a_val = 0.0f;
+      if (col < 8) {
+        float sigma = softplus_f(spre) + min_scale;
+        // 4 normals for this (row-block, dim) from one philox block
+        int rowblk = (rbase + g * 4) >> 2;  // global row / 4
+        Rng4 u = philox_uniform4(seed, 2u, (uint32_t)(rowblk * 16 + col), draw);
+        float n[4];
+        box_muller(u.a, u.b, &n[0], &n[1]);
+        box_muller(u.c, u.d, &n[2], &n[3]);
+        float eps = greedy ? 0.0f : n[r];
+        float uu = loc + sigma * eps;
+        a_val = tanhf(uu) * aff_scale + aff_shift;
+        // log N(u;loc,sigma) - log|d a/d u|
+        float log_det =
+            2.0f * (0.6931471805599453f - uu - softplus_f(-2.0f * uu)) +
+            log_aff_scale;
+        logp = -0.5f * eps * eps - __logf(sigma) -
+               0.9189385332046727f - log_det;
+      }
+      if (col >= ACT && col < 8) logp = 0.0f;  // unused dims (ACT<8)
+      // reduce logp over dims (lanes col 0..7 of this group)
+      logp += __shfl_xor(logp, 1);
+      logp += __shfl_xor(logp, 2);
+      logp += __shfl_xor(logp, 4);
+      int grow = rbase + g * 4 + r;
+      if (col < ACT) action_out[(long)grow * ACT + col] = a_val;
+      if (col == 0) logp_out[grow] = logp;
+    }
+  }
+
+  // ---- critic torso + scalar head (H region reuse: actor reads complete)
+  wave_layer<HID, true>(O, OS, W1c, b1c, K1P, H, HS, lane);
+  wave_layer<HID, true>(H, HS, W2c, b2c, HID, H, HS, lane);
+  wave_value_head<HID>(H, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out, lane);
+}
+
+// Critic-only forward (bootstrap values: V(extras["next_obs"]),
+// ff_ppo.py:113-116).
+template <int HID>
+__launch_bounds__(256, 2) __global__ void value_forward_kernel(
+    const float* __restrict__ obs, const bf16_t* __restrict__ W1c,
+    const float* __restrict__ b1c, const bf16_t* __restrict__ W2c,
+    const float* __restrict__ b2c, const bf16_t* __restrict__ Wvc,
+    const float* __restrict__ bvc, float* __restrict__ value_out,
+    const float* __restrict__ nmean, const float* __restrict__ nvar, int B,
+    int OBS) {
+  __shared__ MlpLds<HID> lds;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int rbase = blockIdx.x * 64 + wid * 16;
+  if (rbase >= B) return;
+  const int K1P = (OBS + 31) & ~31;
+  bf16_t* O = &lds.O[wid][0][0];
+  bf16_t* H = &lds.H[wid][0][0];
+  constexpr int OS = K1P_MAX + OPAD;
+  constexpr int HS = HID + HPAD;
+  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, nullptr, lane);
+  wave_layer<HID, true>(O, OS, W1c, b1c, K1P, H, HS, lane);
+  wave_layer<HID, true>(H, HS, W2c, b2c, HID, H, HS, lane);
+  wave_value_head<HID>(H, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out, lane);
+}
+
+// ------------------------------------------------------- update-phase glue
+
+// silu fwd/bwd, bf16, 8-wide vectorised (guide common-mistake #2).
+extern "C" __global__ void silu_fwd_kernel(const bf16_t* __restrict__ z,
+                                           bf16_t* __restrict__ h, long n) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * blockDim.x * 8;
+  for (; i < n; i += stride) {
+    bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + i);
+    bf16x8 hv;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) hv[j] = f2bf(silu_f(bf2f(zv[j])));
+    *reinterpret_cast<bf16x8*>(h + i) = hv;
+  }
+}
+
+extern "C" __global__ void silu_bwd_kernel(const bf16_t* __restrict__ dh,
+                                           const bf16_t* __restrict__ z,
+                                           bf16_t* __restrict__ dz, long n) {
+  long i = ((long)blockIdx.x * blockDim.x + threadIdx.x) * 8;
+  long stride = (long)gridDim.x * blockDim.x * 8;
+  for (; i < n; i += stride) {
+    bf16x8 dv = *reinterpret_cast<const bf16x8*>(dh + i);
+    bf16x8 zv = *reinterpret_cast<const bf16x8*>(z + i);
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j)
+      o[j] = f2bf(bf2f(dv[j]) * silu_grad_f(bf2f(zv[j])));
+    *reinterpret_cast<bf16x8*>(dz + i) = o;
+  }
+}
+
+// Fused minibatch gather: one kernel replaces 6 index_selects. Gathers the
+// permuted rows of the flat rollout storage for one minibatch (obs -> bf16
+// GEMM input; the scalar fields fp32).
+extern "C" __global__ void ppo_gather_kernel(
+    const long* __restrict__ idx, int mb_size,
+    const float* __restrict__ obs, int OBS,
+    const float* __restrict__ action, int ACT,
+    const float* __restrict__ logp, const float* __restrict__ value,
+    const float* __restrict__ adv, const float* __restrict__ targets,
+    bf16_t* __restrict__ obs_out, float* __restrict__ action_out,
+    float* __restrict__ logp_out, float* __restrict__ value_out,
+    float* __restrict__ adv_out, float* __restrict__ targets_out,
+    const float* __restrict__ nmean, const float* __restrict__ nvar) {
+  // one wave per row
+  int row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  int lane = threadIdx.x & 63;
+  if (row >= mb_size) return;
+  long src = idx[row];
+  for (int k = lane; k < OBS; k += 64) {
+    float v = obs[src * OBS + k];
+    if (nmean) {
+      float sd = sqrtf(fmaxf(nvar[k], 1e-6f));
+      v = fmaxf(-10.0f, fminf(10.0f, (v - nmean[k]) / sd));
+    }
+    obs_out[(long)row * OBS + k] = f2bf(v);
+  }
+  if (lane < ACT) action_out[(long)row * ACT + lane] = action[src * ACT + lane];
+  if (lane == 0) {
+    logp_out[row] = logp[src];
+    value_out[row] = value[src];
+    adv_out[row] = adv[src];
+    targets_out[row] = targets[src];
+  }
+}
+
+// ---------------------------------------------- fused PPO head + losses
+//
+// One wave per sample row. Forward: loc/scale/value head dots from H2a/H2c
+// (bf16), tanh-normal log-prob of the stored action, PPO clip loss
+// (loss.py:17-32 equivalent, see stoix_amd/ops/losses.py), clipped value
+// loss, MC entropy (fresh tanh-normal sample). Backward (analytic, verified
+// against autograd in tests/test_fused_math.py): writes dH2a/dH2c (bf16)
+// and per-row head grads (dhead [B,16] = d(loc,scale_pre), dv [B,1]) for
+// the head weight-grad GEMMs. Loss metrics atomicAdd into metrics[3].
+//
+// Gradient scale: d(total_loss)/d* with total = a_loss - ent_coef*entropy
+//                 + vf_coef*v_loss, all means over the minibatch.
+extern "C" __global__ void ppo_head_loss_kernel(
+    const bf16_t* __restrict__ H2a,  // [B, HID]
+    const bf16_t* __restrict__ H2c,  // [B, HID]
+    const bf16_t* __restrict__ Wha,  // [16, HID] cat(loc, scale)
+    const float* __restrict__ bha,   // [16]
+    const bf16_t* __restrict__ Wvc,  // [HID]
+    const float* __restrict__ bvc,   // [1]
+    const float* __restrict__ action,    // [B, ACT]
+    const float* __restrict__ old_logp,  // [B]
+    const float* __restrict__ old_value, // [B]
+    const float* __restrict__ adv,       // [B]
+    const float* __restrict__ targets,   // [B]
+    bf16_t* __restrict__ dH2a, bf16_t* __restrict__ dH2c,
+    bf16_t* __restrict__ dhead,  // [B, 16]
+    bf16_t* __restrict__ dv_out, // [B]
+    float* __restrict__ metrics, // [3]: actor_loss, value_loss, entropy
+    int B, int HID, int ACT, float clip_eps, float ent_coef, float vf_coef,
+    float min_scale, float aff_scale, float aff_shift, float log_aff_scale,
+    float inv_B, uint64_t seed, const unsigned int* __restrict__ draw_buf) {
+  const int row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  const int lane = threadIdx.x & 63;
+  if (row >= B) return;
+  const uint32_t draw = draw_buf ? *draw_buf : 0u;
+  const int PC = HID / 64;  // cols per lane (256 -> 4)
+
+  // ---- load this row of H2a/H2c (lane-contiguous PC-wide slices)
+  float ha[8], hc[8], wv[8];
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    int c = lane * PC + (p & (PC - 1));
+    if (p < PC) {
+      ha[p] = bf2f(H2a[(long)row * HID + c]);
+      hc[p] = bf2f(H2c[(long)row * HID + c]);
+      wv[p] = bf2f(Wvc[c]);
+    }
+  }
+
+  // ---- head dots: loc_j, spre_j (j<8), v — butterfly all-reduce.
+  // Head weights are re-read in the backward instead of cached (16x8 floats
+  // per lane would spill; they are L2-hot).
+  float outs[17];
+#pragma unroll
+  for (int j = 0; j < 16; ++j) {
+    float s = 0.0f;
+#pragma unroll
+    for (int p = 0; p < 8; ++p)
+      if (p < PC) s += ha[p] * bf2f(Wha[(long)j * HID + lane * PC + p]);
+    outs[j] = s;
+  }
+  {
+    float s = 0.0f;
+#pragma unroll
+    for (int p = 0; p < 8; ++p)
+      if (p < PC) s += hc[p] * wv[p];
+    outs[16] = s;
+  }
+#pragma unroll
+  for (int j = 0; j < 17; ++j) {
+#pragma unroll
+    for (int m = 32; m > 0; m >>= 1) outs[j] += __shfl_xor(outs[j], m);
+  }
+  // every lane now has all 17 sums
+  float v_pred = outs[16] + bvc[0];
+
+  // ---- per-dim tanh-normal math (lane j handles dim j, j < ACT)
+  float logp_new = 0.0f, ent = 0.0f;
+  float loc_j = 0.0f, sigma_j = 1.0f, spre_j = 0.0f;
+  float u_j = 0.0f, eps_ent_j = 0.0f, u_ent_j = 0.0f;
+  if (lane < ACT) {
+    loc_j = outs[lane] + bha[lane];
+    spre_j = outs[8 + lane] + bha[8 + lane];
+    sigma_j = softplus_f(spre_j) + min_scale;
+    float a = action[(long)row * ACT + lane];
+    float y = (a - aff_shift) / aff_scale;
+    y = fmaxf(-1.0f + 1e-3f, fminf(1.0f - 1e-3f, y));
+    u_j = atanhf(y);
+    float z = (u_j - loc_j) / sigma_j;
+    float log_det =
+        2.0f * (0.6931471805599453f - u_j - softplus_f(-2.0f * u_j)) +
+        log_aff_scale;
+    logp_new = -0.5f * z * z - __logf(sigma_j) - 0.9189385332046727f - log_det;
+    // MC entropy sample (reference distributions entropy: -log p(u'), u'
+    // = loc + sigma*eps')
+    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 8 + lane), draw);
+    float n1, n2;
+    box_muller(uu.a, uu.b, &n1, &n2);
+    eps_ent_j = n1;
+    u_ent_j = loc_j + sigma_j * eps_ent_j;
+    float log_det_e =
+        2.0f * (0.6931471805599453f - u_ent_j - softplus_f(-2.0f * u_ent_j)) +
+        log_aff_scale;
+    ent = -(-0.5f * eps_ent_j * eps_ent_j - __logf(sigma_j) -
+            0.9189385332046727f - log_det_e);
+  }
+#pragma unroll
+  for (int m = 4; m > 0; m >>= 1) {
+    logp_new += __shfl_xor(logp_new, m);
+    ent += __shfl_xor(ent, m);
+  }
+  logp_new = __shfl(logp_new, 0, 64);
+  ent = __shfl(ent, 0, 64);
+
+  // ---- PPO clip loss (per-sample; losses.py ppo_clip_loss)
+  float A = adv[row];
+  float ratio = __expf(logp_new - old_logp[row]);
+  float r_clip = fmaxf(1.0f - clip_eps, fminf(1.0f + clip_eps, ratio));
+  float l1 = ratio * A, l2 = r_clip * A;
+  float a_loss = -fminf(l1, l2);
+  // d a_loss / d logp_new  (autograd semantics: min picks l1 branch on tie)
+  float dl_dlogp;
+  if (l1 <= l2) {
+    dl_dlogp = -ratio * A;
+  } else {
+    // through clipped ratio: nonzero only when unclipped
+    dl_dlogp = (ratio > 1.0f - clip_eps && ratio < 1.0f + clip_eps)
+                   ? -ratio * A
+                   : 0.0f;
+  }
+  dl_dlogp *= inv_B;
+
+  // ---- clipped value loss (losses.py clipped_value_loss)
+  float ov = old_value[row], tg = targets[row];
+  float v_clip = ov + fmaxf(-clip_eps, fminf(clip_eps, v_pred - ov));
+  float e1 = (v_pred - tg), e2 = (v_clip - tg);
+  float sq1 = e1 * e1, sq2 = e2 * e2;
+  float v_loss = 0.5f * fmaxf(sq1, sq2);
+  float dv;  // d v_loss / d v_pred
+  if (sq1 >= sq2) {
+    dv = e1;
+  } else {
+    dv = (fabsf(v_pred - ov) < clip_eps) ? e2 : 0.0f;
+  }
+  dv *= vf_coef * inv_B;
+
+  // ---- per-dim gradients (lane j): d(total)/dloc_j, d(total)/dspre_j
+  float dloc = 0.0f, dspre = 0.0f;
+  if (lane < ACT) {
+    float z = (u_j - loc_j) / sigma_j;
+    // clip-loss path through logp_new
+    float dlogp_dloc = z / sigma_j;
+    float dlogp_dsig = (z * z - 1.0f) / sigma_j;
+    dloc = dl_dlogp * dlogp_dloc;
+    float dsig = dl_dlogp * dlogp_dsig;
+    // entropy path: ent = -(logN - logdet)(u'), u' = loc + sigma*eps'
+    // d ent/d loc = -2 tanh(u');  d ent/d sigma = 1/sigma - 2 tanh(u')*eps'
+    float th = tanhf(u_ent_j);
+    float dent_dloc = -2.0f * th;
+    float dent_dsig = 1.0f / sigma_j - 2.0f * th * eps_ent_j;
+    float ce = -ent_coef * inv_B;  // total has -ent_coef * entropy
+    dloc += ce * dent_dloc;
+    dsig += ce * dent_dsig;
+    // sigma = softplus(spre) + min_scale
+    dspre = dsig * (1.0f / (1.0f + __expf(-spre_j)));
+  }
+
+  // broadcast dloc/dspre of lanes 0..7 to all lanes, then form dH2a
+  float dlocs[8], dspres[8];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    dlocs[j] = __shfl(dloc, j, 64);
+    dspres[j] = __shfl(dspre, j, 64);
+  }
+#pragma unroll
+  for (int p = 0; p < 8; ++p) {
+    if (p < PC) {
+      float g = 0.0f;
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        g += dlocs[j] * bf2f(Wha[(long)j * HID + lane * PC + p]) +
+             dspres[j] * bf2f(Wha[(long)(8 + j) * HID + lane * PC + p]);
+      dH2a[(long)row * HID + lane * PC + p] = f2bf(g);
+      dH2c[(long)row * HID + lane * PC + p] = f2bf(dv * wv[p]);
+    }
+  }
+  if (lane < 8) {
+    dhead[(long)row * 16 + lane] = f2bf(lane < ACT ? dloc : 0.0f);
+    dhead[(long)row * 16 + 8 + lane] = f2bf(lane < ACT ? dspre : 0.0f);
+  }
+  if (lane == 0) {
+    dv_out[row] = f2bf(dv);
+    atomicAdd(&metrics[0], a_loss * inv_B);
+    atomicAdd(&metrics[1], v_loss * inv_B);
+    atomicAdd(&metrics[2], ent * inv_B);
+  }
+}
+
+// --------------------------------------------------------- host launchers
+
+extern "C" __global__ void bump_u32_kernel2(unsigned int* p) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) (*p)++;
+}
+
+extern "C" void launch_mfma_probe(const void* A, const void* B, float* D0,
+                                  float* D1, void* stream) {
+  hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const bf16_t*)A, (const bf16_t*)B,
+                     D0, D1);
+}
+
+extern "C" void launch_policy_value_step(
+    const float* obs, const void* W1a, const float* b1a, const void* W2a,
+    const float* b2a, const void* Wha, const float* bha, const void* W1c,
+    const float* b1c, const void* W2c, const float* b2c, const void* Wvc,
+    const float* bvc, float* obs_mirror, float* action_out, float* logp_out,
+    float* value_out, const float* nmean, const float* nvar, int B, int OBS,
+    int ACT, int HID, float min_scale, float aff_scale, float aff_shift,
+    float log_aff_scale, int greedy, uint64_t seed, unsigned int* draw_buf,
+    void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((B + 63) / 64), block(256);
+  if (HID == 256) {
+    hipLaunchKernelGGL(policy_value_step_kernel<256>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, min_scale, aff_scale, aff_shift,
+                       log_aff_scale, greedy, seed, draw_buf);
+  } else {
+    hipLaunchKernelGGL(policy_value_step_kernel<128>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, min_scale, aff_scale, aff_shift,
+                       log_aff_scale, greedy, seed, draw_buf);
+  }
+  if (draw_buf)
+    hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
+}
+
+extern "C" void launch_value_forward(const float* obs, const void* W1c,
+                                     const float* b1c, const void* W2c,
+                                     const float* b2c, const void* Wvc,
+                                     const float* bvc, float* value_out,
+                                     const float* nmean, const float* nvar,
+                                     int B, int OBS, int HID, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((B + 63) / 64), block(256);
+  if (HID == 256) {
+    hipLaunchKernelGGL(value_forward_kernel<256>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, value_out, nmean, nvar, B,
+                       OBS);
+  } else {
+    hipLaunchKernelGGL(value_forward_kernel<128>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, value_out, nmean, nvar, B,
+                       OBS);
+  }
+}
+
+extern "C" void launch_silu_fwd(const void* z, void* h, long n, void* stream) {
+  int threads = 256;
+  long want = (n / 8 + threads - 1) / threads;
+  int blocks = (int)(want < 4096 ? (want > 0 ? want : 1) : 4096);
+  hipLaunchKernelGGL(silu_fwd_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, (const bf16_t*)z, (bf16_t*)h, n);
+}
+
+extern "C" void launch_silu_bwd(const void* dh, const void* z, void* dz,
+                                long n, void* stream) {
+  int threads = 256;
+  long want = (n / 8 + threads - 1) / threads;
+  int blocks = (int)(want < 4096 ? (want > 0 ? want : 1) : 4096);
+  hipLaunchKernelGGL(silu_bwd_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, (const bf16_t*)dh, (const bf16_t*)z,
+                     (bf16_t*)dz, n);
+}
+
+extern "C" void launch_ppo_gather(const long* idx, int mb_size,
+                                  const float* obs, int OBS,
+                                  const float* action, int ACT,
+                                  const float* logp, const float* value,
+                                  const float* adv, const float* targets,
+                                  void* obs_out, float* action_out,
+                                  float* logp_out, float* value_out,
+                                  float* adv_out, float* targets_out,
+                                  const float* nmean, const float* nvar,
+                                  void* stream) {
+  int threads = 256;
+  int rows_per_block = threads / 64;
+  int blocks = (mb_size + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL(ppo_gather_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, idx, mb_size, obs, OBS, action, ACT,
+                     logp, value, adv, targets, (bf16_t*)obs_out, action_out,
+                     logp_out, value_out, adv_out, targets_out, nmean, nvar);
+}
+
+extern "C" void launch_ppo_head_loss(
+    const void* H2a, const void* H2c, const void* Wha, const float* bha,
+    const void* Wvc, const float* bvc, const float* action,
+    const float* old_logp, const float* old_value, const float* adv,
+    const float* targets, void* dH2a, void* dH2c, void* dhead, void* dv_out,
+    float* metrics, int B, int HID, int ACT, float clip_eps, float ent_coef,
+    float vf_coef, float min_scale, float aff_scale, float aff_shift,
+    float log_aff_scale, uint64_t seed, unsigned int* draw_buf, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int threads = 256;
+  int rows_per_block = threads / 64;
+  int blocks = (B + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL(ppo_head_loss_kernel, dim3(blocks), dim3(threads), 0, s,
+                     (const bf16_t*)H2a, (const bf16_t*)H2c,
+                     (const bf16_t*)Wha, bha, (const bf16_t*)Wvc, bvc, action,
+                     old_logp, old_value, adv, targets, (bf16_t*)dH2a,
+                     (bf16_t*)dH2c, (bf16_t*)dhead, (bf16_t*)dv_out, metrics,
+                     B, HID, ACT, clip_eps, ent_coef, vf_coef, min_scale,
+                     aff_scale, aff_shift, log_aff_scale, 1.0f / (float)B,
+                     seed, draw_buf);
+  if (draw_buf)
+    hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
+}

@@ -103,6 +103,15 @@ class PPOLearner:
         self.buf_targets = z(self.T, self.B)
         self.perm_buf = torch.arange(self.T * self.B, device=device)
 
+        # fused MI355X path (hand-written MFMA/HIP kernels; see fused.py):
+        # replaces the eager rollout + minibatch update when the network is
+        # the canonical MLP shape. system.fused=false forces eager.
+        self.fused = None
+        if bool(getattr(self.sys, "fused", True)) and use_bf16:
+            from stoix_amd.systems.ppo.fused import FusedPPOEngine
+
+            self.fused = FusedPPOEngine.try_build(self)
+
     # ---------------------------------------------------------------- acting
 
     @torch.no_grad()
@@ -178,7 +187,10 @@ class PPOLearner:
 
     def rollout_phase(self) -> None:
         """Rollout + GAE into stable buffers (hip-graph capturable)."""
-        self._rollout()
+        if self.fused is not None:
+            self.fused.rollout()
+        else:
+            self._rollout()
         gamma = float(self.sys.gamma)
         adv, targets = multistep.batch_truncated_generalized_advantage_estimation(
             self.buf_reward,
@@ -200,6 +212,8 @@ class PPOLearner:
         """One epoch of minibatch updates reading self.perm_buf
         (hip-graph capturable; the permutation is refreshed eagerly between
         replays — randperm is not capture-legal at scale)."""
+        if self.fused is not None:
+            return self.fused.epoch()
         TB = self.T * self.B
         flat_obs = self.buf_obs.reshape(TB, *self.buf_obs.shape[2:])
         flat_action = self.buf_action.reshape(TB, *self.buf_action.shape[2:])
@@ -285,6 +299,8 @@ class PPOLearner:
     def load_params(self, snap) -> None:
         self.actor.load_state_dict(snap["actor"])
         self.critic.load_state_dict(snap["critic"])
+        if self.fused is not None:
+            self.fused.refresh_masters()
 
 
 def learner_factory(config, env, device) -> PPOLearner:

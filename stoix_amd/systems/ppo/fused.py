@@ -1,0 +1,397 @@
+"""Fused MI355X update path for Anakin PPO (continuous, canonical MLP).
+
+The eager path (ff_ppo.py) runs ~45 kernels per rollout step and ~150 per
+minibatch even under hip-graph replay — execution time is dominated by
+thousands of 4 us elementwise kernels (profiles/r01_ppo_ant_SUMMARY.md).
+This engine replaces it with the hand-written CDNA4 kernels of
+stoix_amd/ops/csrc/mlp.hip:
+
+  rollout step:  policy_value_step (MFMA actor+critic fwd + Philox
+                 tanh-normal sample + log-prob, ONE kernel) -> env HIP step
+                 -> value_forward (bootstrap V(next_obs))        [5 launches]
+  minibatch:     ppo_gather -> 4 bf16 GEMMs (hipBLASLt MFMA) + silu kernels
+                 -> ppo_head_loss (head fwd + losses + analytic head bwd)
+                 -> hand backward (6 GEMMs + 2 silu_bwd per net)
+                 -> one flat bf16 RCCL all-reduce
+                 -> fused_adam_bf16 per chain (clip + Adam + bf16 mirror)
+
+Parameters: fp32 masters live in ONE flat buffer per network (the module
+parameters are repointed to views, so evaluator/checkpointing see updates
+for free); a flat bf16 mirror (maintained by the Adam kernel epilogue) is
+what the GEMMs and rollout kernels read; gradients are written by the wgrad
+GEMMs straight into views of one flat bf16 grad buffer (out=), so the whole
+backward produces zero standalone cast/copy kernels.
+
+Eligibility (falls back to the eager path otherwise): CUDA device, MLP torso
+[H,H] with SiLU and no LayerNorm, H in {128,256}, NormalAffineTanh head with
+ACT<=8, ScalarCriticHead, flat obs <= 128 dims, no observation normalisation.
+
+Algorithm semantics match ff_ppo.py exactly (same losses, same GAE buffers,
+same per-minibatch all-reduce + per-chain clip); numerics are bf16-GEMM
+class, verified against the eager fp32 path in tests/test_fused_math.py and
+tests/test_gpu_ops.py.
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+from stoix_amd.networks.heads import NormalAffineTanhDistributionHead, ScalarCriticHead
+from stoix_amd.networks.torso import MLPTorso
+
+Tensor = torch.Tensor
+
+
+def _mlp_linears(torso: nn.Module) -> Optional[List[nn.Linear]]:
+    if not isinstance(torso, MLPTorso):
+        return None
+    lins = [m for m in torso.net if isinstance(m, nn.Linear)]
+    acts = [m for m in torso.net if isinstance(m, nn.SiLU)]
+    if len(lins) != 2 or len(acts) != 2 or len(list(torso.net)) != 4:
+        return None
+    return lins
+
+
+class _Chain:
+    """One flat fp32 master + bf16 mirror + bf16 grad + Adam state."""
+
+    def __init__(self, specs: List[Tuple[str, Tuple[int, ...]]], device, lr: float):
+        self.lr = lr
+        sizes = [int(torch.prod(torch.tensor(s)).item()) for _, s in specs]
+        total = sum(sizes)
+        self.flat = torch.zeros(total, dtype=torch.float32, device=device)
+        self.flat16 = torch.zeros(total, dtype=torch.bfloat16, device=device)
+        self.grad16 = torch.zeros(total, dtype=torch.bfloat16, device=device)
+        self.m = torch.zeros(total, dtype=torch.float32, device=device)
+        self.v = torch.zeros(total, dtype=torch.float32, device=device)
+        self.sqnorm = torch.zeros(1, dtype=torch.float32, device=device)
+        self.step_t = torch.zeros(1, dtype=torch.int64, device=device)
+        self.views: Dict[str, Tensor] = {}
+        self.views16: Dict[str, Tensor] = {}
+        self.gviews16: Dict[str, Tensor] = {}
+        off = 0
+        for (name, shape), n in zip(specs, sizes):
+            self.views[name] = self.flat[off : off + n].view(*shape)
+            self.views16[name] = self.flat16[off : off + n].view(*shape)
+            self.gviews16[name] = self.grad16[off : off + n].view(*shape)
+            off += n
+
+    def sync_mirror(self) -> None:
+        self.flat16.copy_(self.flat)
+
+
+class FusedPPOEngine:
+    """Drives the fused rollout + update for a PPOLearner. Build with
+    :func:`try_build`; returns None when the config isn't eligible."""
+
+    @staticmethod
+    def try_build(learner) -> Optional["FusedPPOEngine"]:
+        if learner.device.type != "cuda":
+            return None
+        if getattr(learner, "normalize_obs", False):
+            return None
+        if learner._discrete:
+            return None
+        if type(learner).policy_loss is not _base_policy_loss_func(learner):
+            # PPO-penalty / DPO override policy_loss; the fused head kernel
+            # implements the clip loss only.
+            return None
+        actor, critic = learner.actor, learner.critic
+        if actor.input_layer is not None or critic.input_layer is not None:
+            return None
+        a_lins = _mlp_linears(actor.torso)
+        c_lins = _mlp_linears(critic.torso)
+        if a_lins is None or c_lins is None:
+            return None
+        head = actor.action_head
+        if not isinstance(head, NormalAffineTanhDistributionHead):
+            return None
+        if not isinstance(critic.critic_head, ScalarCriticHead):
+            return None
+        H = a_lins[0].out_features
+        if H not in (128, 256) or a_lins[1].out_features != H:
+            return None
+        if c_lins[0].out_features != H or c_lins[1].out_features != H:
+            return None
+        OBS = a_lins[0].in_features
+        ACT = head.loc.out_features
+        if OBS > 128 or ACT > 8:
+            return None
+        if getattr(learner.env, "_hip", None) is None:
+            return None
+        from stoix_amd import ops
+
+        if not ops.have_ext():
+            return None
+        try:
+            return FusedPPOEngine(learner, a_lins, c_lins, H, OBS, ACT)
+        except Exception:
+            return None
+
+    def __init__(self, learner, a_lins, c_lins, H: int, OBS: int, ACT: int):
+        from stoix_amd import ops
+
+        self.ext = ops.ext(required=True)
+        self.learner = learner
+        self.device = learner.device
+        self.H, self.OBS, self.ACT = H, OBS, ACT
+        sysc = learner.sys
+        self.clip_eps = float(sysc.clip_eps)
+        self.ent_coef = float(sysc.ent_coef)
+        self.vf_coef = float(sysc.vf_coef)
+        self.max_grad_norm = float(sysc.max_grad_norm)
+        head: NormalAffineTanhDistributionHead = learner.actor.action_head
+        self.min_scale = float(head.min_scale)
+        self.aff_scale = (float(head.maximum) - float(head.minimum)) / 2.0
+        self.aff_shift = (float(head.maximum) + float(head.minimum)) / 2.0
+        self.log_aff_scale = math.log(self.aff_scale)
+        # rank-dependent Philox seed (torch.manual_seed is rank-offset)
+        self.seed = int(torch.initial_seed()) % (2**62) + 101
+
+        dev = self.device
+        self.actor_chain = _Chain(
+            [
+                ("W1", (H, OBS)),
+                ("b1", (H,)),
+                ("W2", (H, H)),
+                ("b2", (H,)),
+                ("Wh", (16, H)),
+                ("bh", (16,)),
+            ],
+            dev,
+            float(sysc.actor_lr),
+        )
+        self.critic_chain = _Chain(
+            [
+                ("W1", (H, OBS)),
+                ("b1", (H,)),
+                ("W2", (H, H)),
+                ("b2", (H,)),
+                ("Wv", (H,)),
+                ("bv", (1,)),
+            ],
+            dev,
+            float(sysc.critic_lr),
+        )
+        self._adopt_params(a_lins, c_lins, head, learner.critic.critic_head)
+        self.actor_chain.sync_mirror()
+        self.critic_chain.sync_mirror()
+
+        # rollout-side buffers
+        B = learner.B
+        self.empty = torch.zeros(0, device=dev)
+        self.draw_policy = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.draw_ent = torch.zeros(1, dtype=torch.int32, device=dev)
+        self.buf_steptype = torch.zeros(learner.T, B, dtype=torch.uint8, device=dev)
+        self.metrics = torch.zeros(3, dtype=torch.float32, device=dev)
+        self.metric_views = {
+            "actor_loss": self.metrics[0],
+            "value_loss": self.metrics[1],
+            "entropy": self.metrics[2],
+        }
+
+        # minibatch workspaces
+        S = (learner.T * B) // int(sysc.num_minibatches)
+        self.S = S
+        z = lambda *s, dtype=torch.bfloat16: torch.zeros(*s, dtype=dtype, device=dev)
+        self.Xmb = z(S, OBS)
+        self.act_mb = z(S, ACT, dtype=torch.float32)
+        self.logp_mb = z(S, dtype=torch.float32)
+        self.val_mb = z(S, dtype=torch.float32)
+        self.adv_mb = z(S, dtype=torch.float32)
+        self.tgt_mb = z(S, dtype=torch.float32)
+        self.Z1a, self.H1a = z(S, H), z(S, H)
+        self.Z2a, self.H2a = z(S, H), z(S, H)
+        self.Z1c, self.H1c = z(S, H), z(S, H)
+        self.Z2c, self.H2c = z(S, H), z(S, H)
+        self.dH2a, self.dH2c = z(S, H), z(S, H)
+        self.dZ = z(S, H)  # scratch for silu_bwd outputs
+        self.dH1 = z(S, H)
+        self.dhead = z(S, 16)
+        self.dv = z(S, 1)
+
+        import torch.distributed as dist
+
+        self.world = dist.get_world_size() if dist.is_initialized() else 1
+
+        # the rollout reads the env's stable obs buffer directly; seed it
+        # with the current observation (reset happened before attach)
+        learner.env._hb["obs"].copy_(learner.cur_obs)
+
+    # ------------------------------------------------------------- params
+
+    def _adopt_params(self, a_lins, c_lins, head, critic_head) -> None:
+        """Copy module params into the flat masters and repoint the module
+        parameters to views (evaluator / checkpointing then see fused
+        updates with no extra copies)."""
+        ac, cc = self.actor_chain, self.critic_chain
+        ACT, H = self.ACT, self.H
+        with torch.no_grad():
+            ac.views["W1"].copy_(a_lins[0].weight)
+            ac.views["b1"].copy_(a_lins[0].bias)
+            ac.views["W2"].copy_(a_lins[1].weight)
+            ac.views["b2"].copy_(a_lins[1].bias)
+            ac.views["Wh"].zero_()
+            ac.views["bh"].zero_()
+            ac.views["Wh"][0:ACT].copy_(head.loc.weight)
+            ac.views["Wh"][8 : 8 + ACT].copy_(head.scale.weight)
+            ac.views["bh"][0:ACT].copy_(head.loc.bias)
+            ac.views["bh"][8 : 8 + ACT].copy_(head.scale.bias)
+            cc.views["W1"].copy_(c_lins[0].weight)
+            cc.views["b1"].copy_(c_lins[0].bias)
+            cc.views["W2"].copy_(c_lins[1].weight)
+            cc.views["b2"].copy_(c_lins[1].bias)
+            cc.views["Wv"].copy_(critic_head.linear.weight.view(-1))
+            cc.views["bv"].copy_(critic_head.linear.bias)
+        a_lins[0].weight.data = ac.views["W1"]
+        a_lins[0].bias.data = ac.views["b1"]
+        a_lins[1].weight.data = ac.views["W2"]
+        a_lins[1].bias.data = ac.views["b2"]
+        head.loc.weight.data = ac.views["Wh"][0:ACT]
+        head.scale.weight.data = ac.views["Wh"][8 : 8 + ACT]
+        head.loc.bias.data = ac.views["bh"][0:ACT]
+        head.scale.bias.data = ac.views["bh"][8 : 8 + ACT]
+        c_lins[0].weight.data = cc.views["W1"]
+        c_lins[0].bias.data = cc.views["b1"]
+        c_lins[1].weight.data = cc.views["W2"]
+        c_lins[1].bias.data = cc.views["b2"]
+        critic_head.linear.weight.data = cc.views["Wv"].view(1, H)
+        critic_head.linear.bias.data = cc.views["bv"]
+
+    def refresh_masters(self) -> None:
+        """After external writes to the module params (checkpoint restore),
+        re-sync the bf16 mirrors."""
+        self.actor_chain.sync_mirror()
+        self.critic_chain.sync_mirror()
+
+    # ------------------------------------------------------------ rollout
+
+    def rollout(self) -> None:
+        L = self.learner
+        env = L.env
+        hb = env._hb
+        ac, cc = self.actor_chain, self.critic_chain
+        a16, c16 = ac.views16, cc.views16
+        ext = self.ext
+        # ensure hb["obs"] holds the current observation (set at engine
+        # attach + maintained by every env step)
+        for t in range(L.T):
+            ext.policy_value_step(
+                hb["obs"],
+                a16["W1"], ac.views["b1"], a16["W2"], ac.views["b2"],
+                a16["Wh"], ac.views["bh"],
+                c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
+                c16["Wv"], cc.views["bv"],
+                L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t], L.buf_value[t],
+                self.empty, self.empty,
+                self.min_scale, self.aff_scale, self.aff_shift,
+                self.log_aff_scale, 0, self.seed, self.draw_policy,
+            )
+            env.hip_step_into(
+                L.buf_action[t],
+                reward_out=L.buf_reward[t],
+                discount_out=L.buf_discount[t],
+                steptype_out=self.buf_steptype[t],
+            )
+            ext.value_forward(
+                hb["next_obs"],
+                c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
+                c16["Wv"], cc.views["bv"],
+                L.buf_bootstrap[t], self.empty, self.empty,
+            )
+        # truncation flags for GAE (StepType.TRUNCATED == 3)
+        L.buf_truncated.copy_(self.buf_steptype == 3)
+
+    # ------------------------------------------------------------- update
+
+    def epoch(self) -> Dict[str, Tensor]:
+        L = self.learner
+        ext = self.ext
+        TB = L.T * L.B
+        n_mb = int(L.sys.num_minibatches)
+        S = self.S
+        OBSd = self.OBS
+        flat_obs = L.buf_obs.view(TB, OBSd)
+        flat_action = L.buf_action.view(TB, self.ACT)
+        flat_logp = L.buf_log_prob.view(TB)
+        flat_value = L.buf_value.view(TB)
+        flat_adv = L.buf_adv.view(TB)
+        flat_tgt = L.buf_targets.view(TB)
+        ac, cc = self.actor_chain, self.critic_chain
+        a16, c16 = ac.views16, cc.views16
+        ag, cg = ac.gviews16, cc.gviews16
+
+        import torch.distributed as dist
+
+        for mb in range(n_mb):
+            idx = L.perm_buf[mb * S : (mb + 1) * S]
+            ext.ppo_gather(
+                idx, flat_obs, flat_action, flat_logp, flat_value, flat_adv,
+                flat_tgt, self.Xmb, self.act_mb, self.logp_mb, self.val_mb,
+                self.adv_mb, self.tgt_mb, self.empty, self.empty,
+            )
+            # ---- forward (bf16 GEMMs on hipBLASLt MFMA + fused silu)
+            torch.addmm(a16["b1"], self.Xmb, a16["W1"].t(), out=self.Z1a)
+            ext.silu_fwd(self.Z1a, self.H1a)
+            torch.addmm(a16["b2"], self.H1a, a16["W2"].t(), out=self.Z2a)
+            ext.silu_fwd(self.Z2a, self.H2a)
+            torch.addmm(c16["b1"], self.Xmb, c16["W1"].t(), out=self.Z1c)
+            ext.silu_fwd(self.Z1c, self.H1c)
+            torch.addmm(c16["b2"], self.H1c, c16["W2"].t(), out=self.Z2c)
+            ext.silu_fwd(self.Z2c, self.H2c)
+            # ---- fused head fwd + losses + analytic head bwd
+            self.metrics.zero_()
+            ext.ppo_head_loss(
+                self.H2a, self.H2c, a16["Wh"], ac.views["bh"], c16["Wv"],
+                cc.views["bv"], self.act_mb, self.logp_mb, self.val_mb,
+                self.adv_mb, self.tgt_mb, self.dH2a, self.dH2c, self.dhead,
+                self.dv, self.metrics, self.clip_eps, self.ent_coef,
+                self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
+                self.log_aff_scale, self.seed, self.draw_ent,
+            )
+            # ---- actor backward (wgrads straight into flat grad views;
+            # dH2a/dH2c already hold the head backward from the kernel)
+            torch.mm(self.dhead.t(), self.H2a, out=ag["Wh"])
+            torch.sum(self.dhead, 0, out=ag["bh"])
+            ext.silu_bwd(self.dH2a, self.Z2a, self.dZ)
+            torch.mm(self.dZ.t(), self.H1a, out=ag["W2"])
+            torch.sum(self.dZ, 0, out=ag["b2"])
+            torch.mm(self.dZ, a16["W2"], out=self.dH1)
+            ext.silu_bwd(self.dH1, self.Z1a, self.dZ)
+            torch.mm(self.dZ.t(), self.Xmb, out=ag["W1"])
+            torch.sum(self.dZ, 0, out=ag["b1"])
+            # ---- critic backward
+            torch.mm(self.dv.t(), self.H2c, out=cg["Wv"].view(1, self.H))
+            torch.sum(self.dv.view(-1), 0, out=cg["bv"].reshape(()))
+            ext.silu_bwd(self.dH2c, self.Z2c, self.dZ)
+            torch.mm(self.dZ.t(), self.H1c, out=cg["W2"])
+            torch.sum(self.dZ, 0, out=cg["b2"])
+            torch.mm(self.dZ, c16["W2"], out=self.dH1)
+            ext.silu_bwd(self.dH1, self.Z1c, self.dZ)
+            torch.mm(self.dZ.t(), self.Xmb, out=cg["W1"])
+            torch.sum(self.dZ, 0, out=cg["b1"])
+            # ---- all-reduce + fused clip/Adam (+ bf16 mirror refresh)
+            if self.world > 1:
+                dist.all_reduce(ac.grad16)
+                dist.all_reduce(cc.grad16)
+            gscale = 1.0 / float(self.world)
+            ext.fused_adam_bf16(
+                ac.flat, ac.grad16, ac.m, ac.v, ac.sqnorm, ac.step_t,
+                ac.flat16, ac.lr, 0.9, 0.999, 1e-5, self.max_grad_norm, gscale,
+            )
+            ext.fused_adam_bf16(
+                cc.flat, cc.grad16, cc.m, cc.v, cc.sqnorm, cc.step_t,
+                cc.flat16, cc.lr, 0.9, 0.999, 1e-5, self.max_grad_norm, gscale,
+            )
+        return {k: v for k, v in self.metric_views.items()}
+
+
+def _base_policy_loss_func(learner):
+    """The PPOLearner base policy_loss function (unbound)."""
+    from stoix_amd.systems.ppo import ff_ppo
+
+    return ff_ppo.PPOLearner.policy_loss

@@ -1,0 +1,384 @@
+"""GPU tests for the fused MLP kernels (stoix_amd/ops/csrc/mlp.hip) against
+plain-PyTorch fp32 references."""
+import math
+
+import pytest
+import torch
+import torch.nn.functional as F
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+
+@pytest.fixture(scope="module")
+def ext():
+    from stoix_amd import ops
+
+    e = ops.ext(required=True)
+    assert e is not None
+    return e
+
+
+@requires_gpu
+def test_mfma_probe_layout(ext):
+    """The fused kernels assume variant-0 operand layout for
+    v_mfma_f32_16x16x32_bf16 (lane l: A[m=l&15][k=(l>>4)*8+j]). The probe
+    computes D = A@B under both candidate layouts; variant 0 must match."""
+    g = torch.Generator().manual_seed(0)
+    A = torch.randn(16, 32, generator=g).bfloat16().cuda()
+    # asymmetric B (guide: symmetric B passes transposed C-writes)
+    B = (torch.randn(32, 16, generator=g) * torch.linspace(0.5, 2.0, 16)).bfloat16().cuda()
+    D0 = torch.zeros(16, 16, device="cuda")
+    D1 = torch.zeros(16, 16, device="cuda")
+    ext.mfma_probe(A, B, D0, D1)
+    ref = (A.float() @ B.float()).cpu()
+    err0 = (D0.cpu() - ref).abs().max().item()
+    err1 = (D1.cpu() - ref).abs().max().item()
+    assert err0 < 0.1, f"variant-0 MFMA layout wrong (err0={err0}, err1={err1})"
+
+
+@requires_gpu
+def test_silu_kernels(ext):
+    g = torch.Generator().manual_seed(1)
+    z = torch.randn(1000, 256, generator=g).bfloat16().cuda()
+    dh = torch.randn(1000, 256, generator=g).bfloat16().cuda()
+    h = torch.empty_like(z)
+    dz = torch.empty_like(z)
+    ext.silu_fwd(z, h)
+    ext.silu_bwd(dh, z, dz)
+    zf = z.float()
+    torch.testing.assert_close(h.float(), F.silu(zf), rtol=1e-2, atol=1e-2)
+    s = torch.sigmoid(zf)
+    ref = dh.float() * (s * (1 + zf * (1 - s)))
+    torch.testing.assert_close(dz.float(), ref, rtol=2e-2, atol=2e-2)
+
+
+def _mk_weights(H, OBS, ACT, device):
+    g = torch.Generator().manual_seed(7)
+    r = lambda *s: torch.randn(*s, generator=g).to(device) * (1.0 / math.sqrt(s[-1]))
+    W1a, b1a = r(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
+    W2a, b2a = r(H, H), torch.randn(H, generator=g).to(device) * 0.1
+    Wha = torch.zeros(16, H, device=device)
+    Wha[0:ACT] = r(ACT, H)
+    Wha[8 : 8 + ACT] = r(ACT, H)
+    bha = torch.zeros(16, device=device)
+    bha[0:ACT] = torch.randn(ACT, generator=g).to(device) * 0.1
+    bha[8 : 8 + ACT] = torch.randn(ACT, generator=g).to(device) * 0.1
+    W1c, b1c = r(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
+    W2c, b2c = r(H, H), torch.randn(H, generator=g).to(device) * 0.1
+    Wvc = r(H)
+    bvc = torch.randn(1, generator=g).to(device) * 0.1
+    return W1a, b1a, W2a, b2a, Wha, bha, W1c, b1c, W2c, b2c, Wvc, bvc
+
+
+def _eager_forward(obs, W1, b1, W2, b2):
+    h = F.silu(F.linear(obs, W1, b1))
+    return F.silu(F.linear(h, W2, b2))
+
+
+@requires_gpu
+@pytest.mark.parametrize("H,OBS,ACT", [(256, 27, 8), (256, 64, 4), (128, 27, 8)])
+def test_policy_value_step_matches_eager(ext, H, OBS, ACT):
+    device = "cuda"
+    B = 256
+    (W1a, b1a, W2a, b2a, Wha, bha, W1c, b1c, W2c, b2c, Wvc, bvc) = _mk_weights(
+        H, OBS, ACT, device
+    )
+    obs = torch.randn(B, OBS, device=device)
+    e = torch.zeros(0, device=device)
+    action = torch.zeros(B, ACT, device=device)
+    logp = torch.zeros(B, device=device)
+    value = torch.zeros(B, device=device)
+    obs_mirror = torch.zeros(B, OBS, device=device)
+    draw = torch.zeros(1, dtype=torch.int32, device=device)
+    bf = lambda t: t.bfloat16()
+    ext.policy_value_step(
+        obs, bf(W1a), b1a, bf(W2a), b2a, bf(Wha), bha, bf(W1c), b1c, bf(W2c),
+        b2c, bf(Wvc), bvc, obs_mirror, action, logp, value, e, e,
+        1e-3, 1.0, 0.0, 0.0, 1, 1234, draw,
+    )
+    torch.cuda.synchronize()
+    assert draw.item() == 1  # counter bumped
+    torch.testing.assert_close(obs_mirror, obs)
+    # greedy=1: action = tanh(loc); logp computed at eps=0 (u = loc)
+    h2a = _eager_forward(obs, W1a, b1a, W2a, b2a)
+    loc = F.linear(h2a, Wha[0:ACT], bha[0:ACT])
+    spre = F.linear(h2a, Wha[8 : 8 + ACT], bha[8 : 8 + ACT])
+    sigma = F.softplus(spre) + 1e-3
+    torch.testing.assert_close(action, torch.tanh(loc), rtol=3e-2, atol=3e-2)
+    log_det = 2.0 * (math.log(2.0) - loc - F.softplus(-2.0 * loc))
+    logp_ref = (-sigma.log() - 0.5 * math.log(2 * math.pi) - log_det).sum(-1)
+    torch.testing.assert_close(logp, logp_ref, rtol=5e-2, atol=0.2)
+    h2c = _eager_forward(obs, W1c, b1c, W2c, b2c)
+    v_ref = F.linear(h2c, Wvc.view(1, -1), bvc).view(-1)
+    torch.testing.assert_close(value, v_ref, rtol=3e-2, atol=5e-2)
+
+    # value_forward agrees with the critic part
+    v2 = torch.zeros(B, device=device)
+    ext.value_forward(obs, bf(W1c), b1c, bf(W2c), b2c, bf(Wvc), bvc, v2, e, e)
+    torch.testing.assert_close(v2, value, rtol=1e-3, atol=1e-3)
+
+
+@requires_gpu
+def test_policy_step_sampling_statistics(ext):
+    """Non-greedy sampling: actions ~ tanh(N(loc, sigma)); mean/std of atanh
+    over many draws must approach loc/sigma, and draws differ per step."""
+    device = "cuda"
+    H, OBS, ACT, B = 256, 27, 8, 4096
+    w = _mk_weights(H, OBS, ACT, device)
+    (W1a, b1a, W2a, b2a, Wha, bha, W1c, b1c, W2c, b2c, Wvc, bvc) = w
+    obs = torch.randn(B, OBS, device=device) * 0.0  # identical rows
+    e = torch.zeros(0, device=device)
+    action = torch.zeros(B, ACT, device=device)
+    logp = torch.zeros(B, device=device)
+    value = torch.zeros(B, device=device)
+    draw = torch.zeros(1, dtype=torch.int32, device=device)
+    bf = lambda t: t.bfloat16()
+    samples = []
+    for _ in range(4):
+        ext.policy_value_step(
+            obs, bf(W1a), b1a, bf(W2a), b2a, bf(Wha), bha, bf(W1c), b1c,
+            bf(W2c), b2c, bf(Wvc), bvc, e, action, logp, value, e, e,
+            1e-3, 1.0, 0.0, 0.0, 0, 99, draw,
+        )
+        samples.append(action.clone())
+    torch.cuda.synchronize()
+    assert draw.item() == 4
+    assert not torch.allclose(samples[0], samples[1])  # fresh draws per step
+    u = torch.atanh(torch.cat(samples).clamp(-0.999999, 0.999999))
+    h2a = _eager_forward(obs[:1], W1a, b1a, W2a, b2a)
+    loc = F.linear(h2a, Wha[0:ACT], bha[0:ACT])[0]
+    sigma = (F.softplus(F.linear(h2a, Wha[8:16], bha[8:16])) + 1e-3)[0]
+    torch.testing.assert_close(u.mean(0), loc, rtol=0.1, atol=0.05 * sigma.max().item() + 0.02)
+    torch.testing.assert_close(u.std(0), sigma, rtol=0.1, atol=0.05)
+
+
+@requires_gpu
+def test_ppo_head_loss_kernel_matches_reference(ext):
+    """Kernel fwd losses + analytic bwd vs the fp32 torch reference
+    (tests/test_fused_math.py validated that reference against autograd)."""
+    from tests.test_fused_math import closed_form_grads, fused_head_reference
+
+    device = "cuda"
+    H, ACT, B = 256, 8, 512
+    g = torch.Generator().manual_seed(3)
+    H2a = torch.randn(B, H, generator=g).to(device).bfloat16()
+    H2c = torch.randn(B, H, generator=g).to(device).bfloat16()
+    Wha = (torch.randn(16, H, generator=g).to(device) / math.sqrt(H)).bfloat16()
+    bha = torch.randn(16, generator=g).to(device) * 0.1
+    Wvc = (torch.randn(H, generator=g).to(device) / math.sqrt(H)).bfloat16()
+    bvc = torch.randn(1, generator=g).to(device) * 0.1
+    action = (torch.rand(B, ACT, generator=g).to(device) * 1.8 - 0.9)
+    old_logp = torch.randn(B, generator=g).to(device)
+    old_value = torch.randn(B, generator=g).to(device)
+    adv = torch.randn(B, generator=g).to(device)
+    targets = torch.randn(B, generator=g).to(device)
+    dH2a = torch.zeros(B, H, device=device, dtype=torch.bfloat16)
+    dH2c = torch.zeros(B, H, device=device, dtype=torch.bfloat16)
+    dhead = torch.zeros(B, 16, device=device, dtype=torch.bfloat16)
+    dv = torch.zeros(B, 1, device=device, dtype=torch.bfloat16)
+    metrics = torch.zeros(3, device=device)
+    draw = torch.zeros(1, dtype=torch.int32, device=device)
+    clip_eps, ent_coef, vf_coef = 0.2, 0.01, 0.5
+    seed = 42
+    ext.ppo_head_loss(
+        H2a, H2c, Wha, bha, Wvc, bvc, action, old_logp, old_value, adv,
+        targets, dH2a, dH2c, dhead, dv, metrics, clip_eps, ent_coef, vf_coef,
+        1e-3, 1.0, 0.0, 0.0, seed, draw,
+    )
+    torch.cuda.synchronize()
+
+    # fp32 reference with the SAME entropy normals (recompute philox draws
+    # on CPU via the kernel's counter scheme is impractical here; instead
+    # check the pieces that don't depend on the entropy sample exactly, and
+    # the entropy-dependent ones statistically)
+    h2a = H2a.float()
+    h2c = H2c.float()
+    loc = F.linear(h2a, Wha.float()[0:ACT], bha[0:ACT])
+    spre = F.linear(h2a, Wha.float()[8:16], bha[8:16])
+    sigma = F.softplus(spre) + 1e-3
+    v_pred = F.linear(h2c, Wvc.float().view(1, -1), bvc).view(-1)
+    y = ((action - 0.0) / 1.0).clamp(-1 + 1e-3, 1 - 1e-3)
+    u = torch.atanh(y)
+    z = (u - loc) / sigma
+    log_det = 2.0 * (math.log(2.0) - u - F.softplus(-2.0 * u))
+    logp_new = (-0.5 * z * z - sigma.log() - 0.5 * math.log(2 * math.pi) - log_det).sum(-1)
+    ratio = torch.exp(logp_new - old_logp)
+    l1, l2 = ratio * adv, ratio.clamp(1 - clip_eps, 1 + clip_eps) * adv
+    a_loss_ref = -torch.minimum(l1, l2).mean()
+    v_clip = old_value + (v_pred - old_value).clamp(-clip_eps, clip_eps)
+    v_loss_ref = 0.5 * torch.maximum((v_pred - targets) ** 2, (v_clip - targets) ** 2).mean()
+    assert abs(metrics[0].item() - a_loss_ref.item()) < 0.05 * (1 + abs(a_loss_ref.item()))
+    assert abs(metrics[1].item() - v_loss_ref.item()) < 0.05 * (1 + abs(v_loss_ref.item()))
+    # entropy of a tanh-normal with these sigmas: MC estimate should be in a
+    # plausible band around the Gaussian-entropy-minus-logdet scale
+    ent = metrics[2].item()
+    base_ent = (0.5 * math.log(2 * math.pi * math.e) + sigma.log()).sum(-1).mean().item()
+    assert ent < base_ent + 1.0 and ent > base_ent - 2.0 * ACT
+
+    # clip-loss + value-loss parts of the gradients (entropy part is
+    # stochastic; use ent_coef=0 in a second launch for the exact check)
+    metrics.zero_()
+    ext.ppo_head_loss(
+        H2a, H2c, Wha, bha, Wvc, bvc, action, old_logp, old_value, adv,
+        targets, dH2a, dH2c, dhead, dv, metrics, clip_eps, 0.0, vf_coef,
+        1e-3, 1.0, 0.0, 0.0, seed, draw,
+    )
+    torch.cuda.synchronize()
+    eps_ent = torch.zeros(B, ACT, device=device)
+    dloc_ref, dspre_ref, dv_ref = closed_form_grads(
+        loc.double(), spre.double(), v_pred.double(), action.double(),
+        old_logp.double(), old_value.double(), adv.double(), targets.double(),
+        eps_ent.double(), clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0)
+    torch.testing.assert_close(dhead.float()[:, 0:ACT], dloc_ref.float(),
+                               rtol=5e-2, atol=2e-4)
+    torch.testing.assert_close(dhead.float()[:, 8 : 8 + ACT], dspre_ref.float(),
+                               rtol=5e-2, atol=2e-4)
+    torch.testing.assert_close(dv.view(-1).float(), dv_ref.float(),
+                               rtol=5e-2, atol=2e-4)
+    # dH2a = dloc @ Wl + dspre @ Ws ; dH2c = dv * Wv
+    dH2a_ref = dloc_ref.float() @ Wha.float()[0:ACT] + dspre_ref.float() @ Wha.float()[8:16]
+    dH2c_ref = dv_ref.float()[:, None] * Wvc.float()[None, :]
+    torch.testing.assert_close(dH2a.float(), dH2a_ref, rtol=6e-2, atol=3e-4)
+    torch.testing.assert_close(dH2c.float(), dH2c_ref, rtol=6e-2, atol=3e-4)
+
+
+@requires_gpu
+def test_fused_adam_bf16_matches_torch(ext):
+    n = 5000
+    g = torch.Generator().manual_seed(5)
+    p32 = torch.randn(n, generator=g).cuda()
+    p_ref = p32.clone()
+    grads = [torch.randn(n, generator=g).cuda() for _ in range(5)]
+    m = torch.zeros(n).cuda()
+    v = torch.zeros(n).cuda()
+    sqn = torch.zeros(1).cuda()
+    st = torch.zeros(1, dtype=torch.int64).cuda()
+    p16 = torch.zeros(n, dtype=torch.bfloat16).cuda()
+    lr, max_norm = 3e-4, 0.5
+    ref_p = torch.nn.Parameter(p_ref)
+    opt = torch.optim.Adam([ref_p], lr=lr, eps=1e-5)
+    for gr in grads:
+        g16 = gr.bfloat16()
+        ext.fused_adam_bf16(p32, g16, m, v, sqn, st, p16, lr, 0.9, 0.999,
+                            1e-5, max_norm, 1.0)
+        ref_p.grad = g16.float()
+        torch.nn.utils.clip_grad_norm_([ref_p], max_norm)
+        opt.step()
+    torch.cuda.synchronize()
+    torch.testing.assert_close(p32, ref_p.detach(), rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(p16.float(), p32, rtol=1e-2, atol=1e-2)
+
+
+@requires_gpu
+def test_fused_ppo_update_step_runs_and_learns_shape(ext):
+    """End-to-end: PPOLearner with the fused engine active — one update
+    step runs, metrics finite, and the fused path is actually attached."""
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo_continuous.yaml",
+        [
+            "env=brax/ant",
+            "arch.total_num_envs=256",
+            "arch.total_timesteps=null",
+            "arch.num_updates=4",
+            "system.rollout_length=16",
+            "system.num_minibatches=4",
+            "system.epochs=2",
+            "system.compute_dtype=bf16",
+            "logger.loggers=[]",
+        ],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    device = torch.device("cuda:0")
+    env = environments.make_single(cfg, 256, device, seed=0)
+    learner = PPOLearner(cfg, env, device)
+    assert learner.fused is not None, "fused engine must attach on ant/bf16"
+    m1 = learner.update_step()
+    m2 = learner.update_step()
+    torch.cuda.synchronize()
+    for k, val in m2.items():
+        assert torch.isfinite(val).all(), f"{k} not finite"
+    # params actually moved
+    w = learner.fused.actor_chain.flat
+    assert w.abs().sum() > 0
+    assert learner.fused.actor_chain.step_t.item() == 2 * 2 * 4  # eps*mb*steps
+
+
+@requires_gpu
+def test_fused_vs_eager_gradients_one_minibatch(ext):
+    """Single-minibatch gradient parity: run the fused epoch (1 epoch, 1
+    minibatch, ent_coef=0) and the eager epoch from identical params and
+    identical rollout data; resulting parameter deltas must agree to bf16
+    tolerance."""
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    def mk(fused: bool):
+        cfg = compose(
+            "default/anakin/default_ff_ppo_continuous.yaml",
+            [
+                "env=brax/ant",
+                "arch.total_num_envs=256",
+                "arch.total_timesteps=null",
+                "arch.num_updates=4",
+                "arch.seed=11",
+                "system.rollout_length=16",
+                "system.num_minibatches=1",
+                "system.epochs=1",
+                "system.ent_coef=0.0",
+                f"system.compute_dtype={'bf16' if fused else 'fp32'}",
+                f"system.fused={str(fused).lower()}",
+                "logger.loggers=[]",
+            ],
+        )
+        cfg.arch.n_devices = 1
+        check_total_timesteps(cfg)
+        device = torch.device("cuda:0")
+        torch.manual_seed(11)
+        env = environments.make_single(cfg, 256, device, seed=3)
+        return PPOLearner(cfg, env, device)
+
+    torch.manual_seed(11)
+    fused = mk(True)
+    assert fused.fused is not None
+    torch.manual_seed(11)
+    eager = mk(False)
+    assert eager.fused is None
+    # identical params (broadcast_module is deterministic per torch seed,
+    # but be explicit)
+    eager.load_params(fused.snapshot_params())
+    before = {k: {n: t.clone() for n, t in d.items()} for k, d in fused.snapshot_params().items()}
+
+    # identical rollout data: run fused rollout, copy its buffers into eager
+    fused.rollout_phase()
+    for name in ["buf_obs", "buf_action", "buf_log_prob", "buf_value",
+                 "buf_bootstrap", "buf_reward", "buf_discount", "buf_adv",
+                 "buf_targets"]:
+        getattr(eager, name).copy_(getattr(fused, name))
+    perm = torch.arange(fused.T * fused.B, device=fused.device)
+    fused.perm_buf.copy_(perm)
+    eager.perm_buf.copy_(perm)
+    fused.epoch_phase()
+    eager.epoch_phase()
+    torch.cuda.synchronize()
+
+    after_f = fused.snapshot_params()
+    after_e = eager.snapshot_params()
+    for part in ["actor", "critic"]:
+        for n in after_f[part]:
+            df = (after_f[part][n].float() - before[part][n].float())
+            de = (after_e[part][n].float() - before[part][n].float())
+            # Adam steps are lr-scale; compare deltas elementwise with a
+            # tolerance proportional to the step size
+            denom = df.abs().max().clamp_min(1e-8)
+            rel = (df - de).abs().max() / denom
+            assert rel < 0.35, f"{part}.{n}: delta mismatch rel={rel:.3f}"
