@@ -50,7 +50,8 @@ void launch_value_forward(const float*, const void*, const float*,
                           int, int, int, void*);
 void launch_silu_fwd(const void*, void*, long, void*);
 void launch_silu_bwd(const void*, const void*, void*, long, void*);
-void launch_ppo_gather(const long*, int, const float*, int, const float*, int,
+void launch_ppo_gather(const long*, int, const float*, int, int,
+                       const float*, int,
                        const float*, const float*, const float*, const float*,
                        void*, float*, float*, float*, float*, float*,
                        const float*, const float*, void*);
@@ -292,7 +293,8 @@ void ppo_gather(torch::Tensor idx, torch::Tensor obs, torch::Tensor action,
   CHK(obs_out, torch::kBFloat16);
   int mb = idx.numel();
   launch_ppo_gather(idx.data_ptr<long>(), mb, obs.data_ptr<float>(),
-                    obs.size(1), action.data_ptr<float>(), action.size(1),
+                    obs.size(1), obs_out.size(1), action.data_ptr<float>(),
+                    action.size(1),
                     logp.data_ptr<float>(), value.data_ptr<float>(),
                     adv.data_ptr<float>(), targets.data_ptr<float>(),
                     obs_out.data_ptr(), action_out.data_ptr<float>(),

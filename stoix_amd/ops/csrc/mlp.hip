@@ -362,7 +362,7 @@ extern "C" __global__ void silu_bwd_kernel(const bf16_t* __restrict__ dh,
 // GEMM input; the scalar fields fp32).
 extern "C" __global__ void ppo_gather_kernel(
     const long* __restrict__ idx, int mb_size,
-    const float* __restrict__ obs, int OBS,
+    const float* __restrict__ obs, int OBS, int OBS_PAD,
     const float* __restrict__ action, int ACT,
     const float* __restrict__ logp, const float* __restrict__ value,
     const float* __restrict__ adv, const float* __restrict__ targets,
@@ -370,18 +370,22 @@ extern "C" __global__ void ppo_gather_kernel(
     float* __restrict__ logp_out, float* __restrict__ value_out,
     float* __restrict__ adv_out, float* __restrict__ targets_out,
     const float* __restrict__ nmean, const float* __restrict__ nvar) {
-  // one wave per row
+  // one wave per row; obs_out is [mb, OBS_PAD] (K padded to the MFMA
+  // K-step, pad columns zero-filled to match the padded W1 layout)
   int row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
   int lane = threadIdx.x & 63;
   if (row >= mb_size) return;
   long src = idx[row];
-  for (int k = lane; k < OBS; k += 64) {
-    float v = obs[src * OBS + k];
-    if (nmean) {
-      float sd = sqrtf(fmaxf(nvar[k], 1e-6f));
-      v = fmaxf(-10.0f, fminf(10.0f, (v - nmean[k]) / sd));
+  for (int k = lane; k < OBS_PAD; k += 64) {
+    float v = 0.0f;
+    if (k < OBS) {
+      v = obs[src * OBS + k];
+      if (nmean) {
+        float sd = sqrtf(fmaxf(nvar[k], 1e-6f));
+        v = fmaxf(-10.0f, fminf(10.0f, (v - nmean[k]) / sd));
+      }
     }
-    obs_out[(long)row * OBS + k] = f2bf(v);
+    obs_out[(long)row * OBS_PAD + k] = f2bf(v);
   }
   if (lane < ACT) action_out[(long)row * ACT + lane] = action[src * ACT + lane];
   if (lane == 0) {
@@ -675,7 +679,7 @@ extern "C" void launch_silu_bwd(const void* dh, const void* z, void* dz,
 }
 
 extern "C" void launch_ppo_gather(const long* idx, int mb_size,
-                                  const float* obs, int OBS,
+                                  const float* obs, int OBS, int OBS_PAD,
                                   const float* action, int ACT,
                                   const float* logp, const float* value,
                                   const float* adv, const float* targets,
@@ -688,9 +692,10 @@ extern "C" void launch_ppo_gather(const long* idx, int mb_size,
   int rows_per_block = threads / 64;
   int blocks = (mb_size + rows_per_block - 1) / rows_per_block;
   hipLaunchKernelGGL(ppo_gather_kernel, dim3(blocks), dim3(threads), 0,
-                     (hipStream_t)stream, idx, mb_size, obs, OBS, action, ACT,
-                     logp, value, adv, targets, (bf16_t*)obs_out, action_out,
-                     logp_out, value_out, adv_out, targets_out, nmean, nvar);
+                     (hipStream_t)stream, idx, mb_size, obs, OBS, OBS_PAD,
+                     action, ACT, logp, value, adv, targets, (bf16_t*)obs_out,
+                     action_out, logp_out, value_out, adv_out, targets_out,
+                     nmean, nvar);
 }
 
 extern "C" void launch_ppo_head_loss(

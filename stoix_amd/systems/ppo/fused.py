@@ -127,9 +127,16 @@ class FusedPPOEngine:
 
         if not ops.have_ext():
             return None
+        import os
+
         try:
             return FusedPPOEngine(learner, a_lins, c_lins, H, OBS, ACT)
-        except Exception:
+        except Exception as e:
+            if os.environ.get("STOIX_FUSED_STRICT"):
+                raise
+            import warnings
+
+            warnings.warn(f"fused PPO engine unavailable, eager path: {e!r}")
             return None
 
     def __init__(self, learner, a_lins, c_lins, H: int, OBS: int, ACT: int):
@@ -151,11 +158,17 @@ class FusedPPOEngine:
         self.log_aff_scale = math.log(self.aff_scale)
         # rank-dependent Philox seed (torch.manual_seed is rank-offset)
         self.seed = int(torch.initial_seed()) % (2**62) + 101
+        # first-layer K padded to the MFMA K-step (32): the rollout kernel
+        # reads W1 with row stride K1P (mlp.hip load_w_frag), and the
+        # update GEMMs use a matching zero-padded obs matrix, so pad
+        # columns carry exact zeros through weights, grads and Adam.
+        self.K1P = (OBS + 31) & ~31
 
         dev = self.device
+        K1P = self.K1P
         self.actor_chain = _Chain(
             [
-                ("W1", (H, OBS)),
+                ("W1", (H, K1P)),
                 ("b1", (H,)),
                 ("W2", (H, H)),
                 ("b2", (H,)),
@@ -167,7 +180,7 @@ class FusedPPOEngine:
         )
         self.critic_chain = _Chain(
             [
-                ("W1", (H, OBS)),
+                ("W1", (H, K1P)),
                 ("b1", (H,)),
                 ("W2", (H, H)),
                 ("b2", (H,)),
@@ -198,7 +211,7 @@ class FusedPPOEngine:
         S = (learner.T * B) // int(sysc.num_minibatches)
         self.S = S
         z = lambda *s, dtype=torch.bfloat16: torch.zeros(*s, dtype=dtype, device=dev)
-        self.Xmb = z(S, OBS)
+        self.Xmb = z(S, self.K1P)  # K-padded GEMM input (zeros past OBS)
         self.act_mb = z(S, ACT, dtype=torch.float32)
         self.logp_mb = z(S, dtype=torch.float32)
         self.val_mb = z(S, dtype=torch.float32)
@@ -230,8 +243,11 @@ class FusedPPOEngine:
         updates with no extra copies)."""
         ac, cc = self.actor_chain, self.critic_chain
         ACT, H = self.ACT, self.H
+        OBS = self.OBS
         with torch.no_grad():
-            ac.views["W1"].copy_(a_lins[0].weight)
+            ac.views["W1"].zero_()
+            cc.views["W1"].zero_()
+            ac.views["W1"][:, :OBS].copy_(a_lins[0].weight)
             ac.views["b1"].copy_(a_lins[0].bias)
             ac.views["W2"].copy_(a_lins[1].weight)
             ac.views["b2"].copy_(a_lins[1].bias)
@@ -241,13 +257,13 @@ class FusedPPOEngine:
             ac.views["Wh"][8 : 8 + ACT].copy_(head.scale.weight)
             ac.views["bh"][0:ACT].copy_(head.loc.bias)
             ac.views["bh"][8 : 8 + ACT].copy_(head.scale.bias)
-            cc.views["W1"].copy_(c_lins[0].weight)
+            cc.views["W1"][:, :OBS].copy_(c_lins[0].weight)
             cc.views["b1"].copy_(c_lins[0].bias)
             cc.views["W2"].copy_(c_lins[1].weight)
             cc.views["b2"].copy_(c_lins[1].bias)
             cc.views["Wv"].copy_(critic_head.linear.weight.view(-1))
             cc.views["bv"].copy_(critic_head.linear.bias)
-        a_lins[0].weight.data = ac.views["W1"]
+        a_lins[0].weight.data = ac.views["W1"][:, :OBS]
         a_lins[0].bias.data = ac.views["b1"]
         a_lins[1].weight.data = ac.views["W2"]
         a_lins[1].bias.data = ac.views["b2"]
@@ -255,7 +271,7 @@ class FusedPPOEngine:
         head.scale.weight.data = ac.views["Wh"][8 : 8 + ACT]
         head.loc.bias.data = ac.views["bh"][0:ACT]
         head.scale.bias.data = ac.views["bh"][8 : 8 + ACT]
-        c_lins[0].weight.data = cc.views["W1"]
+        c_lins[0].weight.data = cc.views["W1"][:, :OBS]
         c_lins[0].bias.data = cc.views["b1"]
         c_lins[1].weight.data = cc.views["W2"]
         c_lins[1].bias.data = cc.views["b2"]

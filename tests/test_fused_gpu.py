@@ -55,9 +55,18 @@ def test_silu_kernels(ext):
 
 
 def _mk_weights(H, OBS, ACT, device):
+    """W1 is zero-padded to the MFMA K-step (32 cols) like the fused engine
+    stores it (mlp.hip reads W1 with row stride K1P)."""
+    K1P = (OBS + 31) & ~31
     g = torch.Generator().manual_seed(7)
     r = lambda *s: torch.randn(*s, generator=g).to(device) * (1.0 / math.sqrt(s[-1]))
-    W1a, b1a = r(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
+
+    def rpad(H_, OBS_):
+        w = torch.zeros(H_, K1P, device=device)
+        w[:, :OBS_] = r(H_, OBS_)
+        return w
+
+    W1a, b1a = rpad(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
     W2a, b2a = r(H, H), torch.randn(H, generator=g).to(device) * 0.1
     Wha = torch.zeros(16, H, device=device)
     Wha[0:ACT] = r(ACT, H)
@@ -65,7 +74,7 @@ def _mk_weights(H, OBS, ACT, device):
     bha = torch.zeros(16, device=device)
     bha[0:ACT] = torch.randn(ACT, generator=g).to(device) * 0.1
     bha[8 : 8 + ACT] = torch.randn(ACT, generator=g).to(device) * 0.1
-    W1c, b1c = r(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
+    W1c, b1c = rpad(H, OBS), torch.randn(H, generator=g).to(device) * 0.1
     W2c, b2c = r(H, H), torch.randn(H, generator=g).to(device) * 0.1
     Wvc = r(H)
     bvc = torch.randn(1, generator=g).to(device) * 0.1
@@ -73,7 +82,7 @@ def _mk_weights(H, OBS, ACT, device):
 
 
 def _eager_forward(obs, W1, b1, W2, b2):
-    h = F.silu(F.linear(obs, W1, b1))
+    h = F.silu(F.linear(obs, W1[:, : obs.shape[1]], b1))
     return F.silu(F.linear(h, W2, b2))
 
 
