@@ -296,6 +296,7 @@ def test_fused_ppo_update_step_runs_and_learns_shape(ext):
             "arch.total_num_envs=256",
             "arch.total_timesteps=null",
             "arch.num_updates=4",
+            "arch.num_evaluation=1",
             "system.rollout_length=16",
             "system.num_minibatches=4",
             "system.epochs=2",
@@ -339,6 +340,7 @@ def test_fused_vs_eager_gradients_one_minibatch(ext):
                 "arch.total_num_envs=256",
                 "arch.total_timesteps=null",
                 "arch.num_updates=4",
+                "arch.num_evaluation=1",
                 "arch.seed=11",
                 "system.rollout_length=16",
                 "system.num_minibatches=1",
