@@ -55,12 +55,11 @@ void launch_ppo_gather(const long*, int, const float*, int, int,
                        const float*, const float*, const float*, const float*,
                        void*, float*, float*, float*, float*, float*,
                        const float*, const float*, void*);
-void launch_ppo_head_loss(const void*, const void*, const void*, const float*,
-                          const void*, const float*, const float*,
+void launch_ppo_head_loss(const void*, const void*, const float*,
                           const float*, const float*, const float*,
-                          const float*, void*, void*, void*, void*, float*,
-                          int, int, int, float, float, float, float, float,
-                          float, float, uint64_t, unsigned int*, void*);
+                          const float*, void*, void*, float*, int, int,
+                          float, float, float, float, float, float, float,
+                          uint64_t, unsigned int*, void*);
 }
 
 namespace {
@@ -303,30 +302,27 @@ void ppo_gather(torch::Tensor idx, torch::Tensor obs, torch::Tensor action,
                     fptr_or_null(nmean), fptr_or_null(nvar), cur_stream());
 }
 
-void ppo_head_loss(torch::Tensor H2a, torch::Tensor H2c, torch::Tensor Wha,
-                   torch::Tensor bha, torch::Tensor Wvc, torch::Tensor bvc,
+void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
                    torch::Tensor action, torch::Tensor old_logp,
                    torch::Tensor old_value, torch::Tensor adv,
-                   torch::Tensor targets, torch::Tensor dH2a,
-                   torch::Tensor dH2c, torch::Tensor dhead, torch::Tensor dv,
-                   torch::Tensor metrics, double clip_eps, double ent_coef,
-                   double vf_coef, double min_scale, double aff_scale,
-                   double aff_shift, double log_aff_scale, int64_t seed,
-                   torch::Tensor draw_buf) {
-  CHK(H2a, torch::kBFloat16);
+                   torch::Tensor targets, torch::Tensor dhead,
+                   torch::Tensor dv, torch::Tensor metrics, double clip_eps,
+                   double ent_coef, double vf_coef, double min_scale,
+                   double aff_scale, double aff_shift, double log_aff_scale,
+                   int64_t seed, torch::Tensor draw_buf) {
+  CHK(heads, torch::kBFloat16);
+  CHK(v_in, torch::kBFloat16);
   CHK(action, torch::kFloat32);
-  int B = H2a.size(0), HID = H2a.size(1), ACT = action.size(1);
-  TORCH_CHECK(HID % 64 == 0 && HID <= 512, "ppo_head_loss HID constraint");
+  int B = heads.size(0), ACT = action.size(1);
+  TORCH_CHECK(heads.size(1) == 16, "heads must be [B,16] (loc|scale packed)");
   unsigned int* db = draw_buf.numel() > 0
                          ? (unsigned int*)draw_buf.data_ptr<int>()
                          : nullptr;
   launch_ppo_head_loss(
-      H2a.data_ptr(), H2c.data_ptr(), Wha.data_ptr(), bha.data_ptr<float>(),
-      Wvc.data_ptr(), bvc.data_ptr<float>(), action.data_ptr<float>(),
+      heads.data_ptr(), v_in.data_ptr(), action.data_ptr<float>(),
       old_logp.data_ptr<float>(), old_value.data_ptr<float>(),
-      adv.data_ptr<float>(), targets.data_ptr<float>(), dH2a.data_ptr(),
-      dH2c.data_ptr(), dhead.data_ptr(), dv.data_ptr(),
-      metrics.data_ptr<float>(), B, HID, ACT, (float)clip_eps,
+      adv.data_ptr<float>(), targets.data_ptr<float>(), dhead.data_ptr(),
+      dv.data_ptr(), metrics.data_ptr<float>(), B, ACT, (float)clip_eps,
       (float)ent_coef, (float)vf_coef, (float)min_scale, (float)aff_scale,
       (float)aff_shift, (float)log_aff_scale, (uint64_t)seed, db,
       cur_stream());

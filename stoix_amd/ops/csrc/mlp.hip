@@ -93,9 +93,14 @@ extern "C" __global__ void mfma_probe_kernel(const bf16_t* __restrict__ A,
 #define OPAD 8
 #define HPAD 8
 
+// One workgroup = 4 waves sharing ONE 16-row M-tile; each wave owns NT/4
+// of the N-tiles per layer (split-N). 4096 envs -> 256 workgroups -> every
+// CU busy (the old 64-row-per-WG layout filled only 64 of 256 CUs).
+// Buffers: obs tile + per-net ping/pong activation tiles (actor 0/1,
+// critic 2/3) in ONE __shared__ struct (guide §5.5 trap 4a).
 template <int HID>
 struct MlpLds {
-  bf16_t O[4][16][K1P_MAX + OPAD];
+  bf16_t O[16][K1P_MAX + OPAD];
   bf16_t H[4][16][HID + HPAD];
 };
 
@@ -109,42 +114,44 @@ DEV_INLINE bf16x8 load_w_frag(const bf16_t* __restrict__ W, int K, int nt,
   return *p;
 }
 
-// One MLP layer for a single wave's 16-row tile: reads A fragments from
-// `src` (row-major bf16 with row stride `sstride`), weights W [HID x K],
-// bias fp32 [HID], writes silu(out) to dst (stride dstride) as bf16.
-// KS = K/32 (runtime), NT = HID/16 (compile-time).
+// One MLP layer over the workgroup's 16-row tile, split-N across the 4
+// waves: wave `wid` computes N-tiles [wid*NTW, (wid+1)*NTW). Reads A
+// fragments from `src` (row-major bf16, stride `sstride`), weights
+// W [HID x K] (row-major, L2-hot), bias fp32 [HID]; writes silu(out) to
+// `dst`. Caller barriers between dependent layers.
 template <int HID, bool ACT_SILU>
-DEV_INLINE void wave_layer(const bf16_t* __restrict__ src, int sstride,
-                           const bf16_t* __restrict__ W,
-                           const float* __restrict__ bias, int K,
-                           bf16_t* __restrict__ dst, int dstride, int lane) {
-  constexpr int NT = HID / 16;
+DEV_INLINE void wg_layer(const bf16_t* __restrict__ src, int sstride,
+                         const bf16_t* __restrict__ W,
+                         const float* __restrict__ bias, int K,
+                         bf16_t* __restrict__ dst, int dstride, int lane,
+                         int wid) {
+  constexpr int NTW = HID / 16 / 4;  // N-tiles per wave
   const int KS = K / 32;
-  f32x4 acc[NT];
+  f32x4 acc[NTW];
 #pragma unroll
-  for (int nt = 0; nt < NT; ++nt) acc[nt] = {0.f, 0.f, 0.f, 0.f};
+  for (int t = 0; t < NTW; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
   const int arow = lane & 15;
   const int ak0 = (lane >> 4) * 8;
+  const int nt0 = wid * NTW;
   for (int ks = 0; ks < KS; ++ks) {
     const bf16x8 a =
         *reinterpret_cast<const bf16x8*>(src + arow * sstride + ks * 32 + ak0);
 #pragma unroll
-    for (int nt = 0; nt < NT; ++nt) {
-      bf16x8 b = load_w_frag<HID>(W, K, nt, ks, lane);
-      acc[nt] = MFMA_BF16_16x16x32(a, b, acc[nt], 0, 0, 0);
+    for (int t = 0; t < NTW; ++t) {
+      bf16x8 b = load_w_frag<HID>(W, K, nt0 + t, ks, lane);
+      acc[t] = MFMA_BF16_16x16x32(a, b, acc[t], 0, 0, 0);
     }
   }
-  // epilogue: bias + activation, write bf16 to dst LDS region
   const int col = lane & 15;
   const int g = lane >> 4;
 #pragma unroll
-  for (int nt = 0; nt < NT; ++nt) {
-    float b = bias[nt * 16 + col];
+  for (int t = 0; t < NTW; ++t) {
+    float b = bias[(nt0 + t) * 16 + col];
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      float v = acc[nt][r] + b;
+      float v = acc[t][r] + b;
       if (ACT_SILU) v = silu_f(v);
-      dst[(g * 4 + r) * dstride + nt * 16 + col] = f2bf(v);
+      dst[(g * 4 + r) * dstride + (nt0 + t) * 16 + col] = f2bf(v);
     }
   }
 }
@@ -157,11 +164,12 @@ DEV_INLINE void stage_obs(const float* __restrict__ obs, int rbase, int OBS,
                           int K1P, bf16_t* __restrict__ Orow, int ostride,
                           const float* __restrict__ nmean,
                           const float* __restrict__ nvar,
-                          float* __restrict__ obs_mirror, int lane) {
-  for (int idx = lane; idx < 16 * K1P; idx += 64) {
+                          float* __restrict__ obs_mirror, int tid,
+                          int nthreads, int nrows_total) {
+  for (int idx = tid; idx < 16 * K1P; idx += nthreads) {
     int r = idx / K1P, k = idx - r * K1P;
     float v = 0.0f;
-    if (k < OBS) {
+    if (k < OBS && rbase + r < nrows_total) {
       v = obs[(long)(rbase + r) * OBS + k];
       if (obs_mirror) obs_mirror[(long)(rbase + r) * OBS + k] = v;
       if (nmean) {
@@ -180,7 +188,7 @@ template <int HID>
 DEV_INLINE void wave_value_head(const bf16_t* __restrict__ H, int hstride,
                                 const bf16_t* __restrict__ Wv, float bv,
                                 int rbase, float* __restrict__ value_out,
-                                int lane) {
+                                int lane, int B_total) {
   int row = lane >> 2;   // 0..15
   int part = lane & 3;   // 0..3
   float acc = 0.0f;
@@ -192,7 +200,8 @@ DEV_INLINE void wave_value_head(const bf16_t* __restrict__ H, int hstride,
   }
   acc += __shfl_xor(acc, 1);
   acc += __shfl_xor(acc, 2);
-  if (part == 0 && value_out) value_out[rbase + row] = acc + bv;
+  if (part == 0 && value_out && rbase + row < B_total)
+    value_out[rbase + row] = acc + bv;
 }
 
 // The fused policy step. Writes, for rollout step t:
@@ -222,31 +231,45 @@ __launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
   __shared__ MlpLds<HID> lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int rbase = blockIdx.x * 64 + wid * 16;
+  const int rbase = blockIdx.x * 16;
   if (rbase >= B) return;
   const uint32_t draw = draw_buf ? *draw_buf : 0u;
   const int K1P = (OBS + 31) & ~31;
 
-  bf16_t* O = &lds.O[wid][0][0];
-  bf16_t* H = &lds.H[wid][0][0];
+  bf16_t* O = &lds.O[0][0];
   constexpr int OS = K1P_MAX + OPAD;
   constexpr int HS = HID + HPAD;
+  bf16_t* Ha0 = &lds.H[0][0][0];
+  bf16_t* Ha1 = &lds.H[1][0][0];
+  bf16_t* Hc0 = &lds.H[2][0][0];
+  bf16_t* Hc1 = &lds.H[3][0][0];
 
-  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, obs_mirror, lane);
-  // within-wave LDS write->read ordering is handled by the compiler's
-  // lgkmcnt tracking (no cross-wave sharing anywhere in this kernel).
+  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, obs_mirror,
+            threadIdx.x, 256, B);
+  __syncthreads();
 
-  // ---- actor torso
-  wave_layer<HID, true>(O, OS, W1a, b1a, K1P, H, HS, lane);
-  wave_layer<HID, true>(H, HS, W2a, b2a, HID, H, HS, lane);
+  // ---- torsos (split-N across waves; ping/pong LDS buffers)
+  wg_layer<HID, true>(O, OS, W1a, b1a, K1P, Ha0, HS, lane, wid);
+  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
+  __syncthreads();
+  wg_layer<HID, true>(Ha0, HS, W2a, b2a, HID, Ha1, HS, lane, wid);
+  wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
+  __syncthreads();
 
-  // ---- actor head: one 16-col N-tile = [loc(0:ACT) | pad | scale(8:8+ACT)]
-  {
+  // ---- critic scalar head on wave 1 (VALU dot)
+  if (wid == 1) {
+    wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out,
+                         lane, B);
+  }
+
+  // ---- actor head + tanh-normal sample on wave 0:
+  // one 16-col N-tile = [loc(0:ACT) | pad | scale(8:8+ACT)]
+  if (wid == 0) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
     const int arow = lane & 15, ak0 = (lane >> 4) * 8;
     for (int ks = 0; ks < HID / 32; ++ks) {
       const bf16x8 a =
-          *reinterpret_cast<const bf16x8*>(H + arow * HS + ks * 32 + ak0);
+          *reinterpret_cast<const bf16x8*>(Ha1 + arow * HS + ks * 32 + ak0);
       bf16x8 b = load_w_frag<HID>(Wha, HID, 0, ks, lane);
       acc = MFMA_BF16_16x16x32(a, b, acc, 0, 0, 0);
     }
@@ -288,15 +311,12 @@ __launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
       logp += __shfl_xor(logp, 2);
       logp += __shfl_xor(logp, 4);
       int grow = rbase + g * 4 + r;
-      if (col < ACT) action_out[(long)grow * ACT + col] = a_val;
-      if (col == 0) logp_out[grow] = logp;
+      if (grow < B) {
+        if (col < ACT) action_out[(long)grow * ACT + col] = a_val;
+        if (col == 0) logp_out[grow] = logp;
+      }
     }
   }
-
-  // ---- critic torso + scalar head (H region reuse: actor reads complete)
-  wave_layer<HID, true>(O, OS, W1c, b1c, K1P, H, HS, lane);
-  wave_layer<HID, true>(H, HS, W2c, b2c, HID, H, HS, lane);
-  wave_value_head<HID>(H, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out, lane);
 }
 
 // Critic-only forward (bootstrap values: V(extras["next_obs"]),
@@ -312,17 +332,25 @@ __launch_bounds__(256, 2) __global__ void value_forward_kernel(
   __shared__ MlpLds<HID> lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int rbase = blockIdx.x * 64 + wid * 16;
+  const int rbase = blockIdx.x * 16;
   if (rbase >= B) return;
   const int K1P = (OBS + 31) & ~31;
-  bf16_t* O = &lds.O[wid][0][0];
-  bf16_t* H = &lds.H[wid][0][0];
+  bf16_t* O = &lds.O[0][0];
+  bf16_t* H0 = &lds.H[0][0][0];
+  bf16_t* H1 = &lds.H[1][0][0];
   constexpr int OS = K1P_MAX + OPAD;
   constexpr int HS = HID + HPAD;
-  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, nullptr, lane);
-  wave_layer<HID, true>(O, OS, W1c, b1c, K1P, H, HS, lane);
-  wave_layer<HID, true>(H, HS, W2c, b2c, HID, H, HS, lane);
-  wave_value_head<HID>(H, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out, lane);
+  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, nullptr, threadIdx.x,
+            256, B);
+  __syncthreads();
+  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, H0, HS, lane, wid);
+  __syncthreads();
+  wg_layer<HID, true>(H0, HS, W2c, b2c, HID, H1, HS, lane, wid);
+  __syncthreads();
+  if (wid == 0) {
+    wave_value_head<HID>(H1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out,
+                         lane, B);
+  }
 }
 
 // ------------------------------------------------------- update-phase glue
@@ -398,143 +426,100 @@ extern "C" __global__ void ppo_gather_kernel(
 
 // ---------------------------------------------- fused PPO head + losses
 //
-// One wave per sample row. Forward: loc/scale/value head dots from H2a/H2c
-// (bf16), tanh-normal log-prob of the stored action, PPO clip loss
-// (loss.py:17-32 equivalent, see stoix_amd/ops/losses.py), clipped value
-// loss, MC entropy (fresh tanh-normal sample). Backward (analytic, verified
-// against autograd in tests/test_fused_math.py): writes dH2a/dH2c (bf16)
-// and per-row head grads (dhead [B,16] = d(loc,scale_pre), dv [B,1]) for
-// the head weight-grad GEMMs. Loss metrics atomicAdd into metrics[3].
+// One THREAD per sample row. The head projections themselves are GEMMs and
+// run on hipBLASLt in the engine (heads = H2a @ Wha^T + bha -> [B,16],
+// v = H2c @ Wvc + bvc -> [B]); this kernel does the remaining per-row
+// scalar math: tanh-normal log-prob of the stored action, PPO clip loss
+// (losses.py ppo_clip_loss), clipped value loss, MC entropy, and the
+// analytic gradients d(total)/d{loc, scale_pre, v} (verified against
+// autograd in tests/test_fused_math.py). dH2a/dH2c are then GEMMs again
+// (dhead @ Wha, dv @ Wvc) in the engine.
 //
-// Gradient scale: d(total_loss)/d* with total = a_loss - ent_coef*entropy
-//                 + vf_coef*v_loss, all means over the minibatch.
+// Gradient scale: total = a_loss - ent_coef*entropy + vf_coef*v_loss,
+// all means over the minibatch (inv_B folded in here).
 extern "C" __global__ void ppo_head_loss_kernel(
-    const bf16_t* __restrict__ H2a,  // [B, HID]
-    const bf16_t* __restrict__ H2c,  // [B, HID]
-    const bf16_t* __restrict__ Wha,  // [16, HID] cat(loc, scale)
-    const float* __restrict__ bha,   // [16]
-    const bf16_t* __restrict__ Wvc,  // [HID]
-    const float* __restrict__ bvc,   // [1]
+    const bf16_t* __restrict__ heads,    // [B, 16] = loc(0:8)|spre(8:16)
+    const bf16_t* __restrict__ v_in,     // [B] critic head output
     const float* __restrict__ action,    // [B, ACT]
     const float* __restrict__ old_logp,  // [B]
     const float* __restrict__ old_value, // [B]
     const float* __restrict__ adv,       // [B]
     const float* __restrict__ targets,   // [B]
-    bf16_t* __restrict__ dH2a, bf16_t* __restrict__ dH2c,
     bf16_t* __restrict__ dhead,  // [B, 16]
     bf16_t* __restrict__ dv_out, // [B]
     float* __restrict__ metrics, // [3]: actor_loss, value_loss, entropy
-    int B, int HID, int ACT, float clip_eps, float ent_coef, float vf_coef,
+    int B, int ACT, float clip_eps, float ent_coef, float vf_coef,
     float min_scale, float aff_scale, float aff_shift, float log_aff_scale,
     float inv_B, uint64_t seed, const unsigned int* __restrict__ draw_buf) {
-  const int row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
-  const int lane = threadIdx.x & 63;
-  if (row >= B) return;
+  const int row = blockIdx.x * blockDim.x + threadIdx.x;
+  const bool active = row < B;  // inactive lanes still join the metric
+                                // shuffle-reduce (contribute zeros)
   const uint32_t draw = draw_buf ? *draw_buf : 0u;
-  const int PC = HID / 64;  // cols per lane (256 -> 4)
 
-  // ---- load this row of H2a/H2c (lane-contiguous PC-wide slices)
-  float ha[8], hc[8], wv[8];
+  // ---- per-dim tanh-normal forward
+  float loc[8], spre[8], sigma[8], u[8], eps_e[8], u_e[8];
+  if (active) {
+    const bf16x8* h8 = reinterpret_cast<const bf16x8*>(heads + (long)row * 16);
+    bf16x8 hl = h8[0], hs = h8[1];
 #pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    int c = lane * PC + (p & (PC - 1));
-    if (p < PC) {
-      ha[p] = bf2f(H2a[(long)row * HID + c]);
-      hc[p] = bf2f(H2c[(long)row * HID + c]);
-      wv[p] = bf2f(Wvc[c]);
+    for (int j = 0; j < 8; ++j) {
+      loc[j] = bf2f(hl[j]);
+      spre[j] = bf2f(hs[j]);
+      sigma[j] = softplus_f(spre[j]) + min_scale;
     }
   }
-
-  // ---- head dots: loc_j, spre_j (j<8), v — butterfly all-reduce.
-  // Head weights are re-read in the backward instead of cached (16x8 floats
-  // per lane would spill; they are L2-hot).
-  float outs[17];
-#pragma unroll
-  for (int j = 0; j < 16; ++j) {
-    float s = 0.0f;
-#pragma unroll
-    for (int p = 0; p < 8; ++p)
-      if (p < PC) s += ha[p] * bf2f(Wha[(long)j * HID + lane * PC + p]);
-    outs[j] = s;
-  }
-  {
-    float s = 0.0f;
-#pragma unroll
-    for (int p = 0; p < 8; ++p)
-      if (p < PC) s += hc[p] * wv[p];
-    outs[16] = s;
-  }
-#pragma unroll
-  for (int j = 0; j < 17; ++j) {
-#pragma unroll
-    for (int m = 32; m > 0; m >>= 1) outs[j] += __shfl_xor(outs[j], m);
-  }
-  // every lane now has all 17 sums
-  float v_pred = outs[16] + bvc[0];
-
-  // ---- per-dim tanh-normal math (lane j handles dim j, j < ACT)
   float logp_new = 0.0f, ent = 0.0f;
-  float loc_j = 0.0f, sigma_j = 1.0f, spre_j = 0.0f;
-  float u_j = 0.0f, eps_ent_j = 0.0f, u_ent_j = 0.0f;
-  if (lane < ACT) {
-    loc_j = outs[lane] + bha[lane];
-    spre_j = outs[8 + lane] + bha[8 + lane];
-    sigma_j = softplus_f(spre_j) + min_scale;
-    float a = action[(long)row * ACT + lane];
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    if (j >= ACT || !active) break;
+    float a = action[(long)row * ACT + j];
     float y = (a - aff_shift) / aff_scale;
     y = fmaxf(-1.0f + 1e-3f, fminf(1.0f - 1e-3f, y));
-    u_j = atanhf(y);
-    float z = (u_j - loc_j) / sigma_j;
+    u[j] = atanhf(y);
+    float z = (u[j] - loc[j]) / sigma[j];
     float log_det =
-        2.0f * (0.6931471805599453f - u_j - softplus_f(-2.0f * u_j)) +
+        2.0f * (0.6931471805599453f - u[j] - softplus_f(-2.0f * u[j])) +
         log_aff_scale;
-    logp_new = -0.5f * z * z - __logf(sigma_j) - 0.9189385332046727f - log_det;
-    // MC entropy sample (reference distributions entropy: -log p(u'), u'
-    // = loc + sigma*eps')
-    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 8 + lane), draw);
+    logp_new +=
+        -0.5f * z * z - __logf(sigma[j]) - 0.9189385332046727f - log_det;
+    // MC entropy sample: u' = loc + sigma*eps'
+    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 8 + j), draw);
     float n1, n2;
     box_muller(uu.a, uu.b, &n1, &n2);
-    eps_ent_j = n1;
-    u_ent_j = loc_j + sigma_j * eps_ent_j;
+    eps_e[j] = n1;
+    u_e[j] = loc[j] + sigma[j] * n1;
     float log_det_e =
-        2.0f * (0.6931471805599453f - u_ent_j - softplus_f(-2.0f * u_ent_j)) +
+        2.0f * (0.6931471805599453f - u_e[j] - softplus_f(-2.0f * u_e[j])) +
         log_aff_scale;
-    ent = -(-0.5f * eps_ent_j * eps_ent_j - __logf(sigma_j) -
-            0.9189385332046727f - log_det_e);
+    ent -= -0.5f * n1 * n1 - __logf(sigma[j]) - 0.9189385332046727f -
+           log_det_e;
   }
-#pragma unroll
-  for (int m = 4; m > 0; m >>= 1) {
-    logp_new += __shfl_xor(logp_new, m);
-    ent += __shfl_xor(ent, m);
-  }
-  logp_new = __shfl(logp_new, 0, 64);
-  ent = __shfl(ent, 0, 64);
 
-  // ---- PPO clip loss (per-sample; losses.py ppo_clip_loss)
-  float A = adv[row];
-  float ratio = __expf(logp_new - old_logp[row]);
+  // ---- PPO clip loss
+  float A = active ? adv[row] : 0.0f;
+  float ratio = __expf(logp_new - (active ? old_logp[row] : 0.0f));
   float r_clip = fmaxf(1.0f - clip_eps, fminf(1.0f + clip_eps, ratio));
   float l1 = ratio * A, l2 = r_clip * A;
   float a_loss = -fminf(l1, l2);
-  // d a_loss / d logp_new  (autograd semantics: min picks l1 branch on tie)
-  float dl_dlogp;
+  float dl_dlogp;  // autograd semantics: min picks l1 branch on tie
   if (l1 <= l2) {
     dl_dlogp = -ratio * A;
   } else {
-    // through clipped ratio: nonzero only when unclipped
     dl_dlogp = (ratio > 1.0f - clip_eps && ratio < 1.0f + clip_eps)
                    ? -ratio * A
                    : 0.0f;
   }
   dl_dlogp *= inv_B;
 
-  // ---- clipped value loss (losses.py clipped_value_loss)
-  float ov = old_value[row], tg = targets[row];
+  // ---- clipped value loss
+  float v_pred = active ? bf2f(v_in[row]) : 0.0f;
+  float ov = active ? old_value[row] : 0.0f;
+  float tg = active ? targets[row] : 0.0f;
   float v_clip = ov + fmaxf(-clip_eps, fminf(clip_eps, v_pred - ov));
   float e1 = (v_pred - tg), e2 = (v_clip - tg);
   float sq1 = e1 * e1, sq2 = e2 * e2;
   float v_loss = 0.5f * fmaxf(sq1, sq2);
-  float dv;  // d v_loss / d v_pred
+  float dv;
   if (sq1 >= sq2) {
     dv = e1;
   } else {
@@ -542,55 +527,45 @@ extern "C" __global__ void ppo_head_loss_kernel(
   }
   dv *= vf_coef * inv_B;
 
-  // ---- per-dim gradients (lane j): d(total)/dloc_j, d(total)/dspre_j
-  float dloc = 0.0f, dspre = 0.0f;
-  if (lane < ACT) {
-    float z = (u_j - loc_j) / sigma_j;
-    // clip-loss path through logp_new
-    float dlogp_dloc = z / sigma_j;
-    float dlogp_dsig = (z * z - 1.0f) / sigma_j;
-    dloc = dl_dlogp * dlogp_dloc;
-    float dsig = dl_dlogp * dlogp_dsig;
-    // entropy path: ent = -(logN - logdet)(u'), u' = loc + sigma*eps'
-    // d ent/d loc = -2 tanh(u');  d ent/d sigma = 1/sigma - 2 tanh(u')*eps'
-    float th = tanhf(u_ent_j);
-    float dent_dloc = -2.0f * th;
-    float dent_dsig = 1.0f / sigma_j - 2.0f * th * eps_ent_j;
-    float ce = -ent_coef * inv_B;  // total has -ent_coef * entropy
-    dloc += ce * dent_dloc;
-    dsig += ce * dent_dsig;
-    // sigma = softplus(spre) + min_scale
-    dspre = dsig * (1.0f / (1.0f + __expf(-spre_j)));
-  }
-
-  // broadcast dloc/dspre of lanes 0..7 to all lanes, then form dH2a
-  float dlocs[8], dspres[8];
+  // ---- analytic head gradients
+  bf16x8 dl8, ds8;
+  const float ce = -ent_coef * inv_B;
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
-    dlocs[j] = __shfl(dloc, j, 64);
-    dspres[j] = __shfl(dspre, j, 64);
-  }
-#pragma unroll
-  for (int p = 0; p < 8; ++p) {
-    if (p < PC) {
-      float g = 0.0f;
-#pragma unroll
-      for (int j = 0; j < 8; ++j)
-        g += dlocs[j] * bf2f(Wha[(long)j * HID + lane * PC + p]) +
-             dspres[j] * bf2f(Wha[(long)(8 + j) * HID + lane * PC + p]);
-      dH2a[(long)row * HID + lane * PC + p] = f2bf(g);
-      dH2c[(long)row * HID + lane * PC + p] = f2bf(dv * wv[p]);
+    float dloc = 0.0f, dspre = 0.0f;
+    if (j < ACT) {
+      float z = (u[j] - loc[j]) / sigma[j];
+      dloc = dl_dlogp * (z / sigma[j]);
+      float dsig = dl_dlogp * ((z * z - 1.0f) / sigma[j]);
+      float th = tanhf(u_e[j]);
+      dloc += ce * (-2.0f * th);
+      dsig += ce * (1.0f / sigma[j] - 2.0f * th * eps_e[j]);
+      dspre = dsig * (1.0f / (1.0f + __expf(-spre[j])));
     }
+    dl8[j] = f2bf(dloc);
+    ds8[j] = f2bf(dspre);
   }
-  if (lane < 8) {
-    dhead[(long)row * 16 + lane] = f2bf(lane < ACT ? dloc : 0.0f);
-    dhead[(long)row * 16 + 8 + lane] = f2bf(lane < ACT ? dspre : 0.0f);
-  }
-  if (lane == 0) {
+  if (active) {
+    bf16x8* out8 = reinterpret_cast<bf16x8*>(dhead + (long)row * 16);
+    out8[0] = dl8;
+    out8[1] = ds8;
     dv_out[row] = f2bf(dv);
-    atomicAdd(&metrics[0], a_loss * inv_B);
-    atomicAdd(&metrics[1], v_loss * inv_B);
-    atomicAdd(&metrics[2], ent * inv_B);
+  }
+
+  // ---- loss metrics (wave-level pre-reduce, one atomic per wave)
+  float m0 = active ? a_loss * inv_B : 0.0f;
+  float m1 = active ? v_loss * inv_B : 0.0f;
+  float m2 = active ? ent * inv_B : 0.0f;
+#pragma unroll
+  for (int off = 32; off > 0; off >>= 1) {
+    m0 += __shfl_down(m0, off);
+    m1 += __shfl_down(m1, off);
+    m2 += __shfl_down(m2, off);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    atomicAdd(&metrics[0], m0);
+    atomicAdd(&metrics[1], m1);
+    atomicAdd(&metrics[2], m2);
   }
 }
 
@@ -617,7 +592,7 @@ extern "C" void launch_policy_value_step(
     float log_aff_scale, int greedy, uint64_t seed, unsigned int* draw_buf,
     void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  dim3 grid((B + 63) / 64), block(256);
+  dim3 grid((B + 15) / 16), block(256);
   if (HID == 256) {
     hipLaunchKernelGGL(policy_value_step_kernel<256>, grid, block, 0, s, obs,
                        (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
@@ -646,7 +621,7 @@ extern "C" void launch_value_forward(const float* obs, const void* W1c,
                                      const float* nmean, const float* nvar,
                                      int B, int OBS, int HID, void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  dim3 grid((B + 63) / 64), block(256);
+  dim3 grid((B + 15) / 16), block(256);
   if (HID == 256) {
     hipLaunchKernelGGL(value_forward_kernel<256>, grid, block, 0, s, obs,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
@@ -699,25 +674,21 @@ extern "C" void launch_ppo_gather(const long* idx, int mb_size,
 }
 
 extern "C" void launch_ppo_head_loss(
-    const void* H2a, const void* H2c, const void* Wha, const float* bha,
-    const void* Wvc, const float* bvc, const float* action,
+    const void* heads, const void* v_in, const float* action,
     const float* old_logp, const float* old_value, const float* adv,
-    const float* targets, void* dH2a, void* dH2c, void* dhead, void* dv_out,
-    float* metrics, int B, int HID, int ACT, float clip_eps, float ent_coef,
-    float vf_coef, float min_scale, float aff_scale, float aff_shift,
-    float log_aff_scale, uint64_t seed, unsigned int* draw_buf, void* stream) {
+    const float* targets, void* dhead, void* dv_out, float* metrics, int B,
+    int ACT, float clip_eps, float ent_coef, float vf_coef, float min_scale,
+    float aff_scale, float aff_shift, float log_aff_scale, uint64_t seed,
+    unsigned int* draw_buf, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int threads = 256;
-  int rows_per_block = threads / 64;
-  int blocks = (B + rows_per_block - 1) / rows_per_block;
+  int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(ppo_head_loss_kernel, dim3(blocks), dim3(threads), 0, s,
-                     (const bf16_t*)H2a, (const bf16_t*)H2c,
-                     (const bf16_t*)Wha, bha, (const bf16_t*)Wvc, bvc, action,
-                     old_logp, old_value, adv, targets, (bf16_t*)dH2a,
-                     (bf16_t*)dH2c, (bf16_t*)dhead, (bf16_t*)dv_out, metrics,
-                     B, HID, ACT, clip_eps, ent_coef, vf_coef, min_scale,
-                     aff_scale, aff_shift, log_aff_scale, 1.0f / (float)B,
-                     seed, draw_buf);
+                     (const bf16_t*)heads, (const bf16_t*)v_in, action,
+                     old_logp, old_value, adv, targets, (bf16_t*)dhead,
+                     (bf16_t*)dv_out, metrics, B, ACT, clip_eps, ent_coef,
+                     vf_coef, min_scale, aff_scale, aff_shift, log_aff_scale,
+                     1.0f / (float)B, seed, draw_buf);
   if (draw_buf)
     hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
 }

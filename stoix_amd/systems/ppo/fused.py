@@ -224,6 +224,8 @@ class FusedPPOEngine:
         self.dH2a, self.dH2c = z(S, H), z(S, H)
         self.dZ = z(S, H)  # scratch for silu_bwd outputs
         self.dH1 = z(S, H)
+        self.heads = z(S, 16)
+        self.vpred = z(S, 1)
         self.dhead = z(S, 16)
         self.dv = z(S, 1)
 
@@ -359,16 +361,21 @@ class FusedPPOEngine:
             ext.silu_fwd(self.Z1c, self.H1c)
             torch.addmm(c16["b2"], self.H1c, c16["W2"].t(), out=self.Z2c)
             ext.silu_fwd(self.Z2c, self.H2c)
-            # ---- fused head fwd + losses + analytic head bwd
+            # ---- heads as GEMMs (hipBLASLt), then the fused per-row
+            # loss + analytic head-backward kernel, then dH2 as GEMMs
+            torch.addmm(a16["bh"], self.H2a, a16["Wh"].t(), out=self.heads)
+            torch.addmm(c16["bv"], self.H2c, c16["Wv"].view(self.H, 1),
+                        out=self.vpred)
             self.metrics.zero_()
             ext.ppo_head_loss(
-                self.H2a, self.H2c, a16["Wh"], ac.views["bh"], c16["Wv"],
-                cc.views["bv"], self.act_mb, self.logp_mb, self.val_mb,
-                self.adv_mb, self.tgt_mb, self.dH2a, self.dH2c, self.dhead,
-                self.dv, self.metrics, self.clip_eps, self.ent_coef,
-                self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
+                self.heads, self.vpred.view(-1), self.act_mb, self.logp_mb,
+                self.val_mb, self.adv_mb, self.tgt_mb, self.dhead, self.dv,
+                self.metrics, self.clip_eps, self.ent_coef, self.vf_coef,
+                self.min_scale, self.aff_scale, self.aff_shift,
                 self.log_aff_scale, self.seed, self.draw_ent,
             )
+            torch.mm(self.dhead, a16["Wh"], out=self.dH2a)
+            torch.mm(self.dv, c16["Wv"].view(1, self.H), out=self.dH2c)
             # ---- actor backward (wgrads straight into flat grad views;
             # dH2a/dH2c already hold the head backward from the kernel)
             torch.mm(self.dhead.t(), self.H2a, out=ag["Wh"])

@@ -166,78 +166,61 @@ def test_policy_step_sampling_statistics(ext):
 @requires_gpu
 def test_ppo_head_loss_kernel_matches_reference(ext):
     """Kernel fwd losses + analytic bwd vs the fp32 torch reference
-    (tests/test_fused_math.py validated that reference against autograd)."""
-    from tests.test_fused_math import closed_form_grads, fused_head_reference
+    (tests/test_fused_math.py validated that reference against autograd).
+    The head projections are GEMMs outside the kernel now; the kernel gets
+    heads [B,16] (loc|spre) and v [B] directly."""
+    from tests.test_fused_math import closed_form_grads
 
     device = "cuda"
-    H, ACT, B = 256, 8, 512
+    ACT, B = 8, 512
     g = torch.Generator().manual_seed(3)
-    H2a = torch.randn(B, H, generator=g).to(device).bfloat16()
-    H2c = torch.randn(B, H, generator=g).to(device).bfloat16()
-    Wha = (torch.randn(16, H, generator=g).to(device) / math.sqrt(H)).bfloat16()
-    bha = torch.randn(16, generator=g).to(device) * 0.1
-    Wvc = (torch.randn(H, generator=g).to(device) / math.sqrt(H)).bfloat16()
-    bvc = torch.randn(1, generator=g).to(device) * 0.1
+    loc = torch.randn(B, ACT, generator=g).to(device) * 0.5
+    spre = torch.randn(B, ACT, generator=g).to(device)
+    v_pred = torch.randn(B, generator=g).to(device)
+    heads = torch.cat([loc, spre], dim=1).bfloat16()
+    v16 = v_pred.bfloat16()
     action = (torch.rand(B, ACT, generator=g).to(device) * 1.8 - 0.9)
     old_logp = torch.randn(B, generator=g).to(device)
     old_value = torch.randn(B, generator=g).to(device)
     adv = torch.randn(B, generator=g).to(device)
     targets = torch.randn(B, generator=g).to(device)
-    dH2a = torch.zeros(B, H, device=device, dtype=torch.bfloat16)
-    dH2c = torch.zeros(B, H, device=device, dtype=torch.bfloat16)
     dhead = torch.zeros(B, 16, device=device, dtype=torch.bfloat16)
     dv = torch.zeros(B, 1, device=device, dtype=torch.bfloat16)
     metrics = torch.zeros(3, device=device)
     draw = torch.zeros(1, dtype=torch.int32, device=device)
-    clip_eps, ent_coef, vf_coef = 0.2, 0.01, 0.5
-    seed = 42
+    clip_eps, vf_coef = 0.2, 0.5
+    # ent_coef=0 so the (stochastic) entropy term has no gradient; the
+    # entropy VALUE is still checked for plausibility below
     ext.ppo_head_loss(
-        H2a, H2c, Wha, bha, Wvc, bvc, action, old_logp, old_value, adv,
-        targets, dH2a, dH2c, dhead, dv, metrics, clip_eps, ent_coef, vf_coef,
-        1e-3, 1.0, 0.0, 0.0, seed, draw,
+        heads, v16, action, old_logp, old_value, adv, targets, dhead, dv,
+        metrics, clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0, 0.0, 42, draw,
     )
     torch.cuda.synchronize()
+    assert draw.item() == 1
 
-    # fp32 reference with the SAME entropy normals (recompute philox draws
-    # on CPU via the kernel's counter scheme is impractical here; instead
-    # check the pieces that don't depend on the entropy sample exactly, and
-    # the entropy-dependent ones statistically)
-    h2a = H2a.float()
-    h2c = H2c.float()
-    loc = F.linear(h2a, Wha.float()[0:ACT], bha[0:ACT])
-    spre = F.linear(h2a, Wha.float()[8:16], bha[8:16])
-    sigma = F.softplus(spre) + 1e-3
-    v_pred = F.linear(h2c, Wvc.float().view(1, -1), bvc).view(-1)
-    y = ((action - 0.0) / 1.0).clamp(-1 + 1e-3, 1 - 1e-3)
+    locf = heads.float()[:, 0:ACT]
+    spref = heads.float()[:, 8 : 8 + ACT]
+    sigma = F.softplus(spref) + 1e-3
+    vf = v16.float()
+    y = action.clamp(-1 + 1e-3, 1 - 1e-3)
     u = torch.atanh(y)
-    z = (u - loc) / sigma
+    z = (u - locf) / sigma
     log_det = 2.0 * (math.log(2.0) - u - F.softplus(-2.0 * u))
     logp_new = (-0.5 * z * z - sigma.log() - 0.5 * math.log(2 * math.pi) - log_det).sum(-1)
     ratio = torch.exp(logp_new - old_logp)
     l1, l2 = ratio * adv, ratio.clamp(1 - clip_eps, 1 + clip_eps) * adv
     a_loss_ref = -torch.minimum(l1, l2).mean()
-    v_clip = old_value + (v_pred - old_value).clamp(-clip_eps, clip_eps)
-    v_loss_ref = 0.5 * torch.maximum((v_pred - targets) ** 2, (v_clip - targets) ** 2).mean()
-    assert abs(metrics[0].item() - a_loss_ref.item()) < 0.05 * (1 + abs(a_loss_ref.item()))
-    assert abs(metrics[1].item() - v_loss_ref.item()) < 0.05 * (1 + abs(v_loss_ref.item()))
-    # entropy of a tanh-normal with these sigmas: MC estimate should be in a
-    # plausible band around the Gaussian-entropy-minus-logdet scale
+    v_clip = old_value + (vf - old_value).clamp(-clip_eps, clip_eps)
+    v_loss_ref = 0.5 * torch.maximum((vf - targets) ** 2, (v_clip - targets) ** 2).mean()
+    assert abs(metrics[0].item() - a_loss_ref.item()) < 0.03 * (1 + abs(a_loss_ref.item()))
+    assert abs(metrics[1].item() - v_loss_ref.item()) < 0.03 * (1 + abs(v_loss_ref.item()))
     ent = metrics[2].item()
     base_ent = (0.5 * math.log(2 * math.pi * math.e) + sigma.log()).sum(-1).mean().item()
     assert ent < base_ent + 1.0 and ent > base_ent - 2.0 * ACT
 
-    # clip-loss + value-loss parts of the gradients (entropy part is
-    # stochastic; use ent_coef=0 in a second launch for the exact check)
-    metrics.zero_()
-    ext.ppo_head_loss(
-        H2a, H2c, Wha, bha, Wvc, bvc, action, old_logp, old_value, adv,
-        targets, dH2a, dH2c, dhead, dv, metrics, clip_eps, 0.0, vf_coef,
-        1e-3, 1.0, 0.0, 0.0, seed, draw,
-    )
-    torch.cuda.synchronize()
     eps_ent = torch.zeros(B, ACT, device=device)
     dloc_ref, dspre_ref, dv_ref = closed_form_grads(
-        loc.double(), spre.double(), v_pred.double(), action.double(),
+        locf.double(), spref.double(), vf.double(), action.double(),
         old_logp.double(), old_value.double(), adv.double(), targets.double(),
         eps_ent.double(), clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0)
     torch.testing.assert_close(dhead.float()[:, 0:ACT], dloc_ref.float(),
@@ -246,11 +229,6 @@ def test_ppo_head_loss_kernel_matches_reference(ext):
                                rtol=5e-2, atol=2e-4)
     torch.testing.assert_close(dv.view(-1).float(), dv_ref.float(),
                                rtol=5e-2, atol=2e-4)
-    # dH2a = dloc @ Wl + dspre @ Ws ; dH2c = dv * Wv
-    dH2a_ref = dloc_ref.float() @ Wha.float()[0:ACT] + dspre_ref.float() @ Wha.float()[8:16]
-    dH2c_ref = dv_ref.float()[:, None] * Wvc.float()[None, :]
-    torch.testing.assert_close(dH2a.float(), dH2a_ref, rtol=6e-2, atol=3e-4)
-    torch.testing.assert_close(dH2c.float(), dH2c_ref, rtol=6e-2, atol=3e-4)
 
 
 @requires_gpu

@@ -120,14 +120,16 @@ def main():
     def k_silu():
         ext.silu_fwd(F.Z1a, F.H1a)
 
+    def k_headgemm():
+        torch.addmm(a16["bh"], F.H2a, a16["Wh"].t(), out=F.heads)
+
     def k_head():
         F.metrics.zero_()
         ext.ppo_head_loss(
-            F.H2a, F.H2c, a16["Wh"], ac.views["bh"], c16["Wv"], cc.views["bv"],
-            F.act_mb, F.logp_mb, F.val_mb, F.adv_mb, F.tgt_mb, F.dH2a, F.dH2c,
-            F.dhead, F.dv, F.metrics, F.clip_eps, F.ent_coef, F.vf_coef,
-            F.min_scale, F.aff_scale, F.aff_shift, F.log_aff_scale, F.seed,
-            F.draw_ent)
+            F.heads, F.vpred.view(-1), F.act_mb, F.logp_mb, F.val_mb,
+            F.adv_mb, F.tgt_mb, F.dhead, F.dv, F.metrics, F.clip_eps,
+            F.ent_coef, F.vf_coef, F.min_scale, F.aff_scale, F.aff_shift,
+            F.log_aff_scale, F.seed, F.draw_ent)
 
     def k_wgrad():
         torch.mm(F.dZ.t(), F.H1a, out=ac.gviews16["W2"])
@@ -147,6 +149,7 @@ def main():
     print(f"gemm fwd L1 (pad):  {timeit(k_gemm_fwd, 50)*1e3:8.1f} us")
     print(f"gemm fwd L2:        {timeit(k_gemm_fwd2, 50)*1e3:8.1f} us")
     print(f"silu_fwd:           {timeit(k_silu, 50)*1e3:8.1f} us")
+    print(f"head gemm:          {timeit(k_headgemm, 50)*1e3:8.1f} us")
     print(f"ppo_head_loss:      {timeit(k_head, 50)*1e3:8.1f} us")
     print(f"wgrad mm:           {timeit(k_wgrad, 50)*1e3:8.1f} us")
     print(f"dgrad mm:           {timeit(k_dgrad, 50)*1e3:8.1f} us")
