@@ -383,7 +383,7 @@ extern "C" void launch_cartpole_step(
     unsigned char* steptype_out, unsigned char* done_out, int B,
     int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
     void* stream) {
-  int threads = 256;
+  int threads = 64;
   int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(cartpole_step_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, state, action, step_count, ep_return,
@@ -401,7 +401,10 @@ extern "C" void launch_ant_step(
     unsigned char* steptype_out, unsigned char* done_out, int B,
     int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
     void* stream) {
-  int threads = 256;
+  // 64-thread blocks: one wave per WG spreads B=4096 envs over 64 CUs
+  // instead of 16 (the per-env physics is register-serial; CU count is the
+  // lever, not waves-per-CU)
+  int threads = 64;
   int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(ant_step_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, state, action, step_count, ep_return,
