@@ -20,6 +20,7 @@ SOURCES = [
     str(CSRC / "scan.hip"),
     str(CSRC / "optim.hip"),
     str(CSRC / "mlp.hip"),
+    str(CSRC / "wgrad.hip"),
 ]
 
 

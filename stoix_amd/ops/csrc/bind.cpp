@@ -57,9 +57,13 @@ void launch_ppo_gather(const long*, int, const float*, int, int,
                        const float*, const float*, void*);
 void launch_ppo_head_loss(const void*, const void*, const float*,
                           const float*, const float*, const float*,
-                          const float*, void*, void*, float*, int, int,
-                          float, float, float, float, float, float, float,
-                          uint64_t, unsigned int*, void*);
+                          const float*, void*, void*, void*, float*, int,
+                          int, float, float, float, float, float, float,
+                          float, uint64_t, unsigned int*, void*);
+void launch_wgrad(const void*, const void*, float*, long, long, long, int,
+                  int, int, int, void*);
+void launch_slab_reduce(float*, void*, long, long, void*);
+void launch_tr16_probe(const void*, float*, int, void*);
 }
 
 namespace {
@@ -306,7 +310,8 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
                    torch::Tensor action, torch::Tensor old_logp,
                    torch::Tensor old_value, torch::Tensor adv,
                    torch::Tensor targets, torch::Tensor dhead,
-                   torch::Tensor dv, torch::Tensor metrics, double clip_eps,
+                   torch::Tensor dv, torch::Tensor dv16,
+                   torch::Tensor metrics, double clip_eps,
                    double ent_coef, double vf_coef, double min_scale,
                    double aff_scale, double aff_shift, double log_aff_scale,
                    int64_t seed, torch::Tensor draw_buf) {
@@ -318,14 +323,43 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
   unsigned int* db = draw_buf.numel() > 0
                          ? (unsigned int*)draw_buf.data_ptr<int>()
                          : nullptr;
+  void* dv16p = dv16.numel() > 0 ? dv16.data_ptr() : nullptr;
   launch_ppo_head_loss(
       heads.data_ptr(), v_in.data_ptr(), action.data_ptr<float>(),
       old_logp.data_ptr<float>(), old_value.data_ptr<float>(),
       adv.data_ptr<float>(), targets.data_ptr<float>(), dhead.data_ptr(),
-      dv.data_ptr(), metrics.data_ptr<float>(), B, ACT, (float)clip_eps,
+      dv.data_ptr(), dv16p, metrics.data_ptr<float>(), B, ACT,
+      (float)clip_eps,
       (float)ent_coef, (float)vf_coef, (float)min_scale, (float)aff_scale,
       (float)aff_shift, (float)log_aff_scale, (uint64_t)seed, db,
       cur_stream());
+}
+
+void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
+           int64_t dW_off, int64_t db_off, int64_t n_valid) {
+  CHK(dZ, torch::kBFloat16);
+  CHK(X, torch::kBFloat16);
+  CHK(slab, torch::kFloat32);
+  int S = dZ.size(0), N_STRIDE = dZ.size(1), K = X.size(1);
+  TORCH_CHECK(X.size(0) == S, "wgrad: dZ/X row mismatch");
+  TORCH_CHECK(S % (16 * 32) == 0, "wgrad: S must divide 512");
+  long stride = slab.size(1);
+  launch_wgrad(dZ.data_ptr(), X.data_ptr(), slab.data_ptr<float>(),
+               (long)dW_off, (long)db_off, stride, S, N_STRIDE, K,
+               (int)n_valid, cur_stream());
+}
+
+void slab_reduce(torch::Tensor slab, torch::Tensor grad16) {
+  CHK(slab, torch::kFloat32);
+  CHK(grad16, torch::kBFloat16);
+  launch_slab_reduce(slab.data_ptr<float>(), grad16.data_ptr(),
+                     slab.size(1), grad16.numel(), cur_stream());
+}
+
+void tr16_probe(torch::Tensor in, torch::Tensor out, int64_t base_mode) {
+  CHK(in, torch::kBFloat16);
+  launch_tr16_probe(in.data_ptr(), out.data_ptr<float>(), (int)base_mode,
+                    cur_stream());
 }
 
 }  // namespace
@@ -351,4 +385,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_gather", &ppo_gather, "fused minibatch gather");
   m.def("ppo_head_loss", &ppo_head_loss,
         "fused PPO head fwd + losses + analytic head bwd");
+  m.def("wgrad", &wgrad, "split-K MFMA weight grad + bias colsum -> slab");
+  m.def("slab_reduce", &slab_reduce, "sum wgrad slabs into flat bf16 grads");
+  m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 semantics probe");
 }

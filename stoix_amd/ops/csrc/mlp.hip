@@ -446,7 +446,8 @@ extern "C" __global__ void ppo_head_loss_kernel(
     const float* __restrict__ adv,       // [B]
     const float* __restrict__ targets,   // [B]
     bf16_t* __restrict__ dhead,  // [B, 16]
-    bf16_t* __restrict__ dv_out, // [B]
+    bf16_t* __restrict__ dv_out, // [B] (for the dH2c outer-product GEMM)
+    bf16_t* __restrict__ dv16_out,  // [B,16] col 0 (wgrad A-operand) or null
     float* __restrict__ metrics, // [3]: actor_loss, value_loss, entropy
     int B, int ACT, float clip_eps, float ent_coef, float vf_coef,
     float min_scale, float aff_scale, float aff_shift, float log_aff_scale,
@@ -550,6 +551,7 @@ extern "C" __global__ void ppo_head_loss_kernel(
     out8[0] = dl8;
     out8[1] = ds8;
     dv_out[row] = f2bf(dv);
+    if (dv16_out) dv16_out[(long)row * 16] = f2bf(dv);
   }
 
   // ---- loss metrics (wave-level pre-reduce, one atomic per wave)
@@ -676,7 +678,8 @@ extern "C" void launch_ppo_gather(const long* idx, int mb_size,
 extern "C" void launch_ppo_head_loss(
     const void* heads, const void* v_in, const float* action,
     const float* old_logp, const float* old_value, const float* adv,
-    const float* targets, void* dhead, void* dv_out, float* metrics, int B,
+    const float* targets, void* dhead, void* dv_out, void* dv16_out,
+    float* metrics, int B,
     int ACT, float clip_eps, float ent_coef, float vf_coef, float min_scale,
     float aff_scale, float aff_shift, float log_aff_scale, uint64_t seed,
     unsigned int* draw_buf, void* stream) {
@@ -686,7 +689,8 @@ extern "C" void launch_ppo_head_loss(
   hipLaunchKernelGGL(ppo_head_loss_kernel, dim3(blocks), dim3(threads), 0, s,
                      (const bf16_t*)heads, (const bf16_t*)v_in, action,
                      old_logp, old_value, adv, targets, (bf16_t*)dhead,
-                     (bf16_t*)dv_out, metrics, B, ACT, clip_eps, ent_coef,
+                     (bf16_t*)dv_out, (bf16_t*)dv16_out, metrics, B, ACT,
+                     clip_eps, ent_coef,
                      vf_coef, min_scale, aff_scale, aff_shift, log_aff_scale,
                      1.0f / (float)B, seed, draw_buf);
   if (draw_buf)

@@ -1,0 +1,210 @@
+// Split-K weight-gradient kernels (the K6/K7 backward of SURVEY.md §2.9).
+//
+// dW[N,K] = dZ^T[N,S] @ X[S,K]  and  db[N] = colsum(dZ),  S = minibatch
+// (32768 for the flagship config). hipBLASLt's best NT kernel for this
+// K-huge/MN-tiny shape runs at ~56 TFLOP/s (tunableop_gfx950.csv:
+// nt_256_256_32768 = 77 us); these kernels split S over 16 wave-slices,
+// each wave MFMA-accumulating its slice into a private fp32 slab, and one
+// slab_reduce kernel sums the 16 slabs straight into the flat bf16 grad
+// buffer (zeroing the slab for the next minibatch in the same pass). The
+// bias column-sum rides along for free from the A-operand fragments.
+//
+// Geometry per wgrad launch: grid (N/16) x (K/(16*KPG)) x 4, block 256
+// (4 waves). wave-slice = blockIdx.z*4 + wid in [0,16); each wave stages
+// 32-deep s-tiles of dZ and X into LDS *transposed* (scatter b16 writes,
+// contiguous ds_read_b128 fragment reads) and issues KPG
+// mfma_f32_16x16x32_bf16 per s-step.
+#include "common.h"
+#include <hip/hip_bf16.h>
+
+typedef __bf16 bf16_t;
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef __bf16 bf16x4 __attribute__((ext_vector_type(4)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+#define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
+
+#define WG_SLICES 16  // 4 z-blocks x 4 waves
+#define SPAD 8        // +8 cols on the 32-wide transposed tiles
+
+// One wave's private staging: dZt [16 n][32+8 s], Xt [KPG*16 k][32+8 s].
+// KPG <= 8 -> per-wave 40*(16+128)*2B = 11.5 KB; 4 waves = 46 KB.
+struct WgradLds {
+  bf16_t dZt[4][16][32 + SPAD];
+  bf16_t Xt[4][8 * 16][32 + SPAD];
+};
+
+template <int KPG>
+__launch_bounds__(256, 2) __global__ void wgrad_kernel(
+    const bf16_t* __restrict__ dZ,  // [S, N_STRIDE]
+    const bf16_t* __restrict__ X,   // [S, K]
+    float* __restrict__ slab,       // [WG_SLICES, slab_stride] fp32
+    long dW_off,                    // element offset of dW[N,K] in a slab
+    long db_off,                    // element offset of db[N] (or -1)
+    long slab_stride, int S, int N_STRIDE, int K, int N_VALID) {
+  __shared__ WgradLds lds;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int nt = blockIdx.x;         // one 16-row n-tile
+  const int k0 = blockIdx.y * KPG * 16;
+  const int slice = blockIdx.z * 4 + wid;
+  const int s_per = S / WG_SLICES;   // S divisible by 16*32 in practice
+  const int s_begin = slice * s_per;
+  const int s_end = s_begin + s_per;
+  const int n0 = nt * 16;
+
+  bf16_t(*dZt)[32 + SPAD] = lds.dZt[wid];
+  bf16_t(*Xt)[32 + SPAD] = lds.Xt[wid];
+
+  f32x4 acc[KPG];
+#pragma unroll
+  for (int t = 0; t < KPG; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
+  float db_acc = 0.0f;  // lane's partial of db[n0 + (lane&15)]
+
+  const int arow = lane & 15;
+  const int ak0 = (lane >> 4) * 8;
+
+  for (int s0 = s_begin; s0 < s_end; s0 += 32) {
+    // ---- stage dZ tile [32 s][16 n] -> dZt [16 n][32 s] (transposed)
+    {
+      // lane l: s-row = l>>1, n-half = (l&1)*8 -> one 16B load, 8 scatter
+      // writes
+      int srow = lane >> 1, nh = (lane & 1) * 8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          dZ + (long)(s0 + srow) * N_STRIDE + n0 + nh);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) dZt[nh + i][srow] = v[i];
+    }
+    // ---- stage X tiles [32 s][KPG*16 k] -> Xt [KPG*16 k][32 s]
+#pragma unroll
+    for (int q = 0; q < KPG; ++q) {
+      // flat chunk id: 64 lanes x KPG iterations cover 32 rows x KPG*2
+      // 8-elem chunks
+      int flat = q * 64 + lane;
+      int srow = flat / (KPG * 2);
+      int kh = (flat % (KPG * 2)) * 8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          X + (long)(s0 + srow) * K + k0 + kh);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) Xt[kh + i][srow] = v[i];
+    }
+    // within-wave LDS write->read ordering is compiler-tracked (lgkmcnt);
+    // no cross-wave sharing, so no barrier.
+    // ---- fragments + MFMA
+    const bf16x8 a = *reinterpret_cast<const bf16x8*>(&dZt[arow][ak0]);
+#pragma unroll
+    for (int t = 0; t < KPG; ++t) {
+      const bf16x8 b =
+          *reinterpret_cast<const bf16x8*>(&Xt[t * 16 + arow][ak0]);
+      acc[t] = MFMA_BF16_16x16x32(a, b, acc[t], 0, 0, 0);
+    }
+    // ---- bias partial from the A fragment (only the kt-group-0 blocks)
+    if (db_off >= 0 && blockIdx.y == 0) {
+#pragma unroll
+      for (int j = 0; j < 8; ++j) db_acc += (float)a[j];
+    }
+  }
+
+  // ---- write this wave's slab slice
+  float* out = slab + (long)slice * slab_stride;
+  const int col = lane & 15;  // k within tile
+  const int g = lane >> 4;
+#pragma unroll
+  for (int t = 0; t < KPG; ++t) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int n = g * 4 + r;  // D row = n (A is dZ^T)
+      if (n < N_VALID)
+        out[dW_off + (long)(n0 + n) * K + k0 + t * 16 + col] = acc[t][r];
+    }
+  }
+  if (db_off >= 0 && blockIdx.y == 0) {
+    // db[n0 + arow] partial: lanes {arow, arow+16, arow+32, arow+48} hold
+    // s-disjoint partials; fold with xor-shuffles over the high bits
+    db_acc += __shfl_xor(db_acc, 16);
+    db_acc += __shfl_xor(db_acc, 32);
+    if ((lane >> 4) == 0 && arow < N_VALID)
+      out[db_off + n0 + arow] = db_acc;
+  }
+}
+
+// Sum the WG_SLICES slabs into the flat bf16 grad buffer and zero them for
+// the next minibatch. One launch covers a whole chain's gradient.
+extern "C" __global__ void slab_reduce_kernel(float* __restrict__ slab,
+                                              __bf16* __restrict__ grad16,
+                                              long slab_stride, long n) {
+  long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  long stride = (long)gridDim.x * blockDim.x;
+  for (; i < n; i += stride) {
+    float s = 0.0f;
+#pragma unroll
+    for (int z = 0; z < WG_SLICES; ++z) {
+      s += slab[(long)z * slab_stride + i];
+      slab[(long)z * slab_stride + i] = 0.0f;
+    }
+    grad16[i] = (__bf16)s;
+  }
+}
+
+// ------------------------------------------------------------- tr probe
+// Empirically maps __builtin_amdgcn_ds_read_tr16_b64_v4bf16: stages 1024
+// known bf16 values linearly in LDS, issues the tr read with per-lane base
+// = base_mode ? lane-pattern : uniform, dumps each lane's 4 elements.
+extern "C" __global__ void tr16_probe_kernel(const bf16_t* __restrict__ in,
+                                             float* __restrict__ out,
+                                             int base_mode) {
+  __shared__ bf16_t l[1024];
+  int t = threadIdx.x;
+  for (int i = t; i < 1024; i += 64) l[i] = in[i];
+  __syncthreads();
+  if (t >= 64) return;
+  int off = 0;
+  if (base_mode == 1) off = (t >> 4) * 64;       // group-strided base
+  else if (base_mode == 2) off = (t >> 4) * 128; // doubled group stride
+  bf16x4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(
+      (__attribute__((address_space(3))) bf16x4*)&l[off]);
+#pragma unroll
+  for (int j = 0; j < 4; ++j) out[t * 4 + j] = (float)v[j];
+}
+
+// --------------------------------------------------------- host launchers
+
+extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
+                             long dW_off, long db_off, long slab_stride,
+                             int S, int N_STRIDE, int K, int N_VALID,
+                             void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int NT = (N_VALID + 15) / 16;
+  int KPG = (K % 128 == 0) ? 8 : ((K % 32 == 0) ? 2 : 1);
+  int KTG = K / (16 * KPG);
+  dim3 grid(NT, KTG, 4), block(256);
+  if (KPG == 8) {
+    hipLaunchKernelGGL(wgrad_kernel<8>, grid, block, 0, s, (const bf16_t*)dZ,
+                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
+                       N_STRIDE, K, N_VALID);
+  } else if (KPG == 2) {
+    hipLaunchKernelGGL(wgrad_kernel<2>, grid, block, 0, s, (const bf16_t*)dZ,
+                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
+                       N_STRIDE, K, N_VALID);
+  } else {
+    hipLaunchKernelGGL(wgrad_kernel<1>, grid, block, 0, s, (const bf16_t*)dZ,
+                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
+                       N_STRIDE, K, N_VALID);
+  }
+}
+
+extern "C" void launch_slab_reduce(float* slab, void* grad16,
+                                   long slab_stride, long n, void* stream) {
+  int threads = 256;
+  long want = (n + threads - 1) / threads;
+  int blocks = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
+  hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, slab, (__bf16*)grad16, slab_stride,
+                     n);
+}
+
+extern "C" void launch_tr16_probe(const void* in, float* out, int base_mode,
+                                  void* stream) {
+  hipLaunchKernelGGL(tr16_probe_kernel, dim3(1), dim3(64), 0,
+                     (hipStream_t)stream, (const bf16_t*)in, out, base_mode);
+}

@@ -134,6 +134,12 @@ def main():
     def k_wgrad():
         torch.mm(F.dZ.t(), F.H1a, out=ac.gviews16["W2"])
 
+    def k_wgrad_custom():
+        ext.wgrad(F.dZ, F.H1a, ac.slab, ac.offsets["W2"], ac.offsets["b2"], F.H)
+
+    def k_slab_reduce():
+        ext.slab_reduce(ac.slab, ac.grad16)
+
     def k_dgrad():
         torch.mm(F.dZ, a16["W2"], out=F.dH1)
 
@@ -152,6 +158,8 @@ def main():
     print(f"head gemm:          {timeit(k_headgemm, 50)*1e3:8.1f} us")
     print(f"ppo_head_loss:      {timeit(k_head, 50)*1e3:8.1f} us")
     print(f"wgrad mm:           {timeit(k_wgrad, 50)*1e3:8.1f} us")
+    print(f"wgrad custom:       {timeit(k_wgrad_custom, 50)*1e3:8.1f} us")
+    print(f"slab_reduce:        {timeit(k_slab_reduce, 50)*1e3:8.1f} us")
     print(f"dgrad mm:           {timeit(k_dgrad, 50)*1e3:8.1f} us")
     print(f"bias colsum:        {timeit(k_bsum, 50)*1e3:8.1f} us")
     print(f"fused_adam_bf16:    {timeit(k_adam, 50)*1e3:8.1f} us")
