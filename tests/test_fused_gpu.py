@@ -191,9 +191,10 @@ def test_ppo_head_loss_kernel_matches_reference(ext):
     clip_eps, vf_coef = 0.2, 0.5
     # ent_coef=0 so the (stochastic) entropy term has no gradient; the
     # entropy VALUE is still checked for plausibility below
+    dv16 = torch.zeros(B, 16, device=device, dtype=torch.bfloat16)
     ext.ppo_head_loss(
         heads, v16, action, old_logp, old_value, adv, targets, dhead, dv,
-        metrics, clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0, 0.0, 42, draw,
+        dv16, metrics, clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0, 0.0, 42, draw,
     )
     torch.cuda.synchronize()
     assert draw.item() == 1

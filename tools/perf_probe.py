@@ -127,7 +127,7 @@ def main():
         F.metrics.zero_()
         ext.ppo_head_loss(
             F.heads, F.vpred.view(-1), F.act_mb, F.logp_mb, F.val_mb,
-            F.adv_mb, F.tgt_mb, F.dhead, F.dv, F.metrics, F.clip_eps,
+            F.adv_mb, F.tgt_mb, F.dhead, F.dv, F.dv16, F.metrics, F.clip_eps,
             F.ent_coef, F.vf_coef, F.min_scale, F.aff_scale, F.aff_shift,
             F.log_aff_scale, F.seed, F.draw_ent)
 
