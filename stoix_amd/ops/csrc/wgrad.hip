@@ -75,14 +75,18 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
 #pragma unroll
       for (int i = 0; i < 8; ++i) dZt[nh + i][srow] = v[i];
     }
-    // ---- stage X tiles [32 s][KPG*16 k] -> Xt [KPG*16 k][32 s]
+    // ---- stage X tiles [32 s][KPG*16 k] -> Xt [KPG*16 k][32 s].
+    // Lane mapping: consecutive lanes take consecutive s-rows of ONE
+    // k-chunk, so the 8 transposed b16 scatter-writes per lane hit
+    // consecutive banks (srow-consecutive) instead of a 16-way conflict
+    // (kh-strided, 160 dwords = bank 0 for every lane). The global loads
+    // become 16B row-strided, but successive q iterations re-touch the
+    // same 32 rows' cache lines, so L1 serves 7/8 of them.
 #pragma unroll
     for (int q = 0; q < KPG; ++q) {
-      // flat chunk id: 64 lanes x KPG iterations cover 32 rows x KPG*2
-      // 8-elem chunks
       int flat = q * 64 + lane;
-      int srow = flat / (KPG * 2);
-      int kh = (flat % (KPG * 2)) * 8;
+      int srow = flat & 31;
+      int kh = (flat >> 5) * 8;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
           X + (long)(s0 + srow) * K + k0 + kh);
 #pragma unroll
