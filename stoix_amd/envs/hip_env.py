@@ -54,6 +54,8 @@ class HipStepMixin:
         reward_out: torch.Tensor,
         discount_out: torch.Tensor,
         steptype_out: torch.Tensor,
+        draw_offset: int = 0,
+        do_bump: bool = True,
     ) -> None:
         """Fused-rollout variant of step(): the kernel writes reward /
         discount / step-type straight into the caller's rollout storage
@@ -78,6 +80,8 @@ class HipStepMixin:
             self.max_episode_steps,
             self._hip_seed,
             hb["draw"],
+            draw_offset,
+            1 if do_bump else 0,
         )
 
     def step(self, action: torch.Tensor) -> TimeStep:  # type: ignore[override]
@@ -102,6 +106,8 @@ class HipStepMixin:
             self.max_episode_steps,
             self._hip_seed,
             hb["draw"],
+            0,
+            1,
         )
         return TimeStep(
             step_type=hb["steptype"],

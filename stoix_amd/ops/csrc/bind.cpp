@@ -14,10 +14,11 @@ extern "C" {
 void launch_cartpole_step(float*, const long*, int*, float*, int*, float*,
                           int*, float*, float*, float*, float*, unsigned char*,
                           unsigned char*, int, int, uint64_t, unsigned int*,
-                          void*);
+                          unsigned int, int, void*);
 void launch_ant_step(float*, const float*, int*, float*, int*, float*, int*,
                      float*, float*, float*, float*, unsigned char*,
-                     unsigned char*, int, int, uint64_t, unsigned int*, void*);
+                     unsigned char*, int, int, uint64_t, unsigned int*,
+                     unsigned int, int, void*);
 void launch_ant_reset(float*, int, uint64_t, uint32_t, void*);
 void launch_gae(const float*, const float*, const float*, const float*,
                 const unsigned char*, float*, float*, int, int, float, void*);
@@ -33,7 +34,7 @@ void launch_fused_adam(float*, const float*, float*, float*, float*, long*,
                        long, float, float, float, float, float, void*);
 void launch_fused_adam_bf16(float*, const void*, float*, float*, float*,
                             long*, void*, long, float, float, float, float,
-                            float, float, void*);
+                            float, float, int, void*);
 void launch_polyak(const float*, float*, long, float, void*);
 void launch_mfma_probe(const void*, const void*, float*, float*, void*);
 void launch_policy_value_step(const float*, const void*, const float*,
@@ -43,7 +44,7 @@ void launch_policy_value_step(const float*, const void*, const float*,
                               const float*, float*, float*, float*, float*,
                               const float*, const float*, int, int, int, int,
                               float, float, float, float, int, uint64_t,
-                              unsigned int*, void*);
+                              unsigned int*, unsigned int, int, void*);
 void launch_value_forward(const float*, const void*, const float*,
                           const void*, const float*, const void*,
                           const float*, float*, const float*, const float*,
@@ -59,10 +60,12 @@ void launch_ppo_head_loss(const void*, const void*, const float*,
                           const float*, const float*, const float*,
                           const float*, void*, void*, void*, float*, int,
                           int, float, float, float, float, float, float,
-                          float, uint64_t, unsigned int*, void*);
+                          float, uint64_t, unsigned int*, unsigned int, int,
+                          void*);
 void launch_wgrad(const void*, const void*, float*, long, long, long, int,
                   int, int, int, void*);
-void launch_slab_reduce(float*, void*, long, long, void*);
+void launch_slab_reduce(float*, void*, long, long, float*, long*, void*);
+void launch_bump_add(unsigned int*, unsigned int, void*);
 void launch_tr16_probe(const void*, float*, int, void*);
 }
 
@@ -84,7 +87,8 @@ void cartpole_step(torch::Tensor state, torch::Tensor action,
                    torch::Tensor next_obs_out, torch::Tensor reward_out,
                    torch::Tensor discount_out, torch::Tensor steptype_out,
                    torch::Tensor done_out, int64_t max_episode_steps,
-                   int64_t seed, torch::Tensor draw_buf) {
+                   int64_t seed, torch::Tensor draw_buf, int64_t draw_offset,
+                   int64_t do_bump) {
   CHK(state, torch::kFloat32);
   CHK(action, torch::kInt64);
   int B = state.size(0);
@@ -96,7 +100,8 @@ void cartpole_step(torch::Tensor state, torch::Tensor action,
       next_obs_out.data_ptr<float>(), reward_out.data_ptr<float>(),
       discount_out.data_ptr<float>(), steptype_out.data_ptr<unsigned char>(),
       done_out.data_ptr<unsigned char>(), B, (int)max_episode_steps,
-      (uint64_t)seed, (unsigned int*)draw_buf.data_ptr<int>(), cur_stream());
+      (uint64_t)seed, (unsigned int*)draw_buf.data_ptr<int>(),
+      (unsigned int)draw_offset, (int)do_bump, cur_stream());
 }
 
 void ant_step(torch::Tensor state, torch::Tensor action,
@@ -106,7 +111,7 @@ void ant_step(torch::Tensor state, torch::Tensor action,
               torch::Tensor next_obs_out, torch::Tensor reward_out,
               torch::Tensor discount_out, torch::Tensor steptype_out,
               torch::Tensor done_out, int64_t max_episode_steps, int64_t seed,
-              torch::Tensor draw_buf) {
+              torch::Tensor draw_buf, int64_t draw_offset, int64_t do_bump) {
   CHK(state, torch::kFloat32);
   CHK(action, torch::kFloat32);
   int B = state.size(0);
@@ -118,7 +123,8 @@ void ant_step(torch::Tensor state, torch::Tensor action,
       next_obs_out.data_ptr<float>(), reward_out.data_ptr<float>(),
       discount_out.data_ptr<float>(), steptype_out.data_ptr<unsigned char>(),
       done_out.data_ptr<unsigned char>(), B, (int)max_episode_steps,
-      (uint64_t)seed, (unsigned int*)draw_buf.data_ptr<int>(), cur_stream());
+      (uint64_t)seed, (unsigned int*)draw_buf.data_ptr<int>(),
+      (unsigned int)draw_offset, (int)do_bump, cur_stream());
 }
 
 void ant_reset(torch::Tensor state, int64_t seed, int64_t draw) {
@@ -196,7 +202,7 @@ void fused_adam_bf16(torch::Tensor param, torch::Tensor grad,
                      torch::Tensor sqnorm, torch::Tensor step_t,
                      torch::Tensor param_bf16, double lr, double beta1,
                      double beta2, double eps, double max_norm,
-                     double grad_scale) {
+                     double grad_scale, int64_t do_prologue) {
   CHK(param, torch::kFloat32);
   CHK(grad, torch::kBFloat16);
   void* pbf = param_bf16.numel() > 0 ? param_bf16.data_ptr() : nullptr;
@@ -206,7 +212,7 @@ void fused_adam_bf16(torch::Tensor param, torch::Tensor grad,
                          sqnorm.data_ptr<float>(), step_t.data_ptr<long>(),
                          pbf, param.numel(), (float)lr, (float)beta1,
                          (float)beta2, (float)eps, (float)max_norm,
-                         (float)grad_scale, cur_stream());
+                         (float)grad_scale, (int)do_prologue, cur_stream());
 }
 
 void mfma_probe(torch::Tensor A, torch::Tensor B, torch::Tensor D0,
@@ -229,7 +235,8 @@ void policy_value_step(torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
                        torch::Tensor value_out, torch::Tensor nmean,
                        torch::Tensor nvar, double min_scale, double aff_scale,
                        double aff_shift, double log_aff_scale, int64_t greedy,
-                       int64_t seed, torch::Tensor draw_buf) {
+                       int64_t seed, torch::Tensor draw_buf,
+                       int64_t draw_offset, int64_t do_bump) {
   CHK(obs, torch::kFloat32);
   CHK(W1a, torch::kBFloat16);
   CHK(W2a, torch::kBFloat16);
@@ -252,7 +259,8 @@ void policy_value_step(torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
       logp_out.data_ptr<float>(), value_out.data_ptr<float>(),
       fptr_or_null(nmean), fptr_or_null(nvar), B, OBS, ACT, HID,
       (float)min_scale, (float)aff_scale, (float)aff_shift,
-      (float)log_aff_scale, (int)greedy, (uint64_t)seed, db, cur_stream());
+      (float)log_aff_scale, (int)greedy, (uint64_t)seed, db,
+      (unsigned int)draw_offset, (int)do_bump, cur_stream());
 }
 
 void value_forward(torch::Tensor obs, torch::Tensor W1c, torch::Tensor b1c,
@@ -314,7 +322,8 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
                    torch::Tensor metrics, double clip_eps,
                    double ent_coef, double vf_coef, double min_scale,
                    double aff_scale, double aff_shift, double log_aff_scale,
-                   int64_t seed, torch::Tensor draw_buf) {
+                   int64_t seed, torch::Tensor draw_buf, int64_t draw_offset,
+                   int64_t do_bump) {
   CHK(heads, torch::kBFloat16);
   CHK(v_in, torch::kBFloat16);
   CHK(action, torch::kFloat32);
@@ -332,7 +341,7 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
       (float)clip_eps,
       (float)ent_coef, (float)vf_coef, (float)min_scale, (float)aff_scale,
       (float)aff_shift, (float)log_aff_scale, (uint64_t)seed, db,
-      cur_stream());
+      (unsigned int)draw_offset, (int)do_bump, cur_stream());
 }
 
 void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
@@ -349,11 +358,19 @@ void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
                (int)n_valid, cur_stream());
 }
 
-void slab_reduce(torch::Tensor slab, torch::Tensor grad16) {
+void slab_reduce(torch::Tensor slab, torch::Tensor grad16,
+                 torch::Tensor sqnorm, torch::Tensor step_t) {
   CHK(slab, torch::kFloat32);
   CHK(grad16, torch::kBFloat16);
+  float* sq = sqnorm.numel() > 0 ? sqnorm.data_ptr<float>() : nullptr;
+  long* st = step_t.numel() > 0 ? step_t.data_ptr<long>() : nullptr;
   launch_slab_reduce(slab.data_ptr<float>(), grad16.data_ptr(),
-                     slab.size(1), grad16.numel(), cur_stream());
+                     slab.size(1), grad16.numel(), sq, st, cur_stream());
+}
+
+void bump_add(torch::Tensor draw_buf, int64_t n) {
+  launch_bump_add((unsigned int*)draw_buf.data_ptr<int>(), (unsigned int)n,
+                  cur_stream());
 }
 
 void tr16_probe(torch::Tensor in, torch::Tensor out, int64_t base_mode) {
@@ -386,6 +403,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_head_loss", &ppo_head_loss,
         "fused PPO head fwd + losses + analytic head bwd");
   m.def("wgrad", &wgrad, "split-K MFMA weight grad + bias colsum -> slab");
-  m.def("slab_reduce", &slab_reduce, "sum wgrad slabs into flat bf16 grads");
+  m.def("slab_reduce", &slab_reduce,
+        "sum wgrad slabs into flat bf16 grads (+ fused Adam prologue)");
+  m.def("bump_add", &bump_add, "add N to a device RNG draw counter");
   m.def("tr16_probe", &tr16_probe, "ds_read_tr16_b64 semantics probe");
 }

@@ -27,10 +27,10 @@ extern "C" __global__ void cartpole_step_kernel(
     unsigned char* __restrict__ steptype_out, // [B]
     unsigned char* __restrict__ done_out,     // [B] is_terminal_step
     int B, int max_episode_steps, uint64_t seed,
-    const unsigned int* __restrict__ draw_buf) {
+    const unsigned int* __restrict__ draw_buf, unsigned int draw_offset) {
   int b = blockIdx.x * blockDim.x + threadIdx.x;
   if (b >= B) return;
-  uint32_t draw = *draw_buf;
+  uint32_t draw = *draw_buf + draw_offset;
 
   float x = state[b * 4 + 0];
   float x_dot = state[b * 4 + 1];
@@ -134,10 +134,10 @@ extern "C" __global__ void ant_step_kernel(
     unsigned char* __restrict__ steptype_out,
     unsigned char* __restrict__ done_out,
     int B, int max_episode_steps, uint64_t seed,
-    const unsigned int* __restrict__ draw_buf) {
+    const unsigned int* __restrict__ draw_buf, unsigned int draw_offset) {
   int b = blockIdx.x * blockDim.x + threadIdx.x;
   if (b >= B) return;
-  uint32_t draw = *draw_buf;
+  uint32_t draw = *draw_buf + draw_offset;
 
   // constants (keep EXACTLY in sync with stoix_amd/envs/ant.py)
   const float TORSO_MASS = 10.0f, TORSO_INERTIA = 0.4f, TORSO_Z0 = 0.55f;
@@ -382,16 +382,18 @@ extern "C" void launch_cartpole_step(
     float* next_obs_out, float* reward_out, float* discount_out,
     unsigned char* steptype_out, unsigned char* done_out, int B,
     int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
-    void* stream) {
+    unsigned int draw_offset, int do_bump, void* stream) {
   int threads = 64;
   int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(cartpole_step_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, state, action, step_count, ep_return,
                      ep_length, last_ep_return, last_ep_length, obs_out,
                      next_obs_out, reward_out, discount_out, steptype_out,
-                     done_out, B, max_episode_steps, seed, draw_buf);
-  hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
-                     (hipStream_t)stream, draw_buf);
+                     done_out, B, max_episode_steps, seed, draw_buf,
+                     draw_offset);
+  if (do_bump)
+    hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
+                       (hipStream_t)stream, draw_buf);
 }
 
 extern "C" void launch_ant_step(
@@ -400,7 +402,7 @@ extern "C" void launch_ant_step(
     float* next_obs_out, float* reward_out, float* discount_out,
     unsigned char* steptype_out, unsigned char* done_out, int B,
     int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
-    void* stream) {
+    unsigned int draw_offset, int do_bump, void* stream) {
   // 64-thread blocks: one wave per WG spreads B=4096 envs over 64 CUs
   // instead of 16 (the per-env physics is register-serial; CU count is the
   // lever, not waves-per-CU)
@@ -410,9 +412,11 @@ extern "C" void launch_ant_step(
                      (hipStream_t)stream, state, action, step_count, ep_return,
                      ep_length, last_ep_return, last_ep_length, obs_out,
                      next_obs_out, reward_out, discount_out, steptype_out,
-                     done_out, B, max_episode_steps, seed, draw_buf);
-  hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
-                     (hipStream_t)stream, draw_buf);
+                     done_out, B, max_episode_steps, seed, draw_buf,
+                     draw_offset);
+  if (do_bump)
+    hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
+                       (hipStream_t)stream, draw_buf);
 }
 
 extern "C" void launch_ant_reset(float* state, int B, uint64_t seed,

@@ -227,13 +227,17 @@ __launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
     const float* __restrict__ nmean, const float* __restrict__ nvar,
     int B, int OBS, int ACT, float min_scale, float aff_scale,
     float aff_shift, float log_aff_scale, int greedy,
-    uint64_t seed, const unsigned int* __restrict__ draw_buf) {
+    uint64_t seed, const unsigned int* __restrict__ draw_buf,
+    unsigned int draw_offset) {
   __shared__ MlpLds<HID> lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int rbase = blockIdx.x * 16;
   if (rbase >= B) return;
-  const uint32_t draw = draw_buf ? *draw_buf : 0u;
+  // draw = base counter + per-graph-node frozen offset: the captured
+  // rollout bumps the base ONCE per replay (bump_add) instead of paying a
+  // 1-thread bump kernel per step (~5 us each, ~2 ms/update-step)
+  const uint32_t draw = (draw_buf ? *draw_buf : 0u) + draw_offset;
   const int K1P = (OBS + 31) & ~31;
 
   bf16_t* O = &lds.O[0][0];
@@ -451,11 +455,12 @@ extern "C" __global__ void ppo_head_loss_kernel(
     float* __restrict__ metrics, // [3]: actor_loss, value_loss, entropy
     int B, int ACT, float clip_eps, float ent_coef, float vf_coef,
     float min_scale, float aff_scale, float aff_shift, float log_aff_scale,
-    float inv_B, uint64_t seed, const unsigned int* __restrict__ draw_buf) {
+    float inv_B, uint64_t seed, const unsigned int* __restrict__ draw_buf,
+    unsigned int draw_offset) {
   const int row = blockIdx.x * blockDim.x + threadIdx.x;
   const bool active = row < B;  // inactive lanes still join the metric
                                 // shuffle-reduce (contribute zeros)
-  const uint32_t draw = draw_buf ? *draw_buf : 0u;
+  const uint32_t draw = (draw_buf ? *draw_buf : 0u) + draw_offset;
 
   // ---- per-dim tanh-normal forward
   float loc[8], spre[8], sigma[8], u[8], eps_e[8], u_e[8];
@@ -577,6 +582,16 @@ extern "C" __global__ void bump_u32_kernel2(unsigned int* p) {
   if (blockIdx.x == 0 && threadIdx.x == 0) (*p)++;
 }
 
+extern "C" __global__ void bump_add_kernel(unsigned int* p, unsigned int n) {
+  if (blockIdx.x == 0 && threadIdx.x == 0) (*p) += n;
+}
+
+extern "C" void launch_bump_add(unsigned int* p, unsigned int n,
+                                void* stream) {
+  hipLaunchKernelGGL(bump_add_kernel, dim3(1), dim3(1), 0,
+                     (hipStream_t)stream, p, n);
+}
+
 extern "C" void launch_mfma_probe(const void* A, const void* B, float* D0,
                                   float* D1, void* stream) {
   hipLaunchKernelGGL(mfma_probe_kernel, dim3(1), dim3(64), 0,
@@ -592,7 +607,7 @@ extern "C" void launch_policy_value_step(
     float* value_out, const float* nmean, const float* nvar, int B, int OBS,
     int ACT, int HID, float min_scale, float aff_scale, float aff_shift,
     float log_aff_scale, int greedy, uint64_t seed, unsigned int* draw_buf,
-    void* stream) {
+    unsigned int draw_offset, int do_bump, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((B + 15) / 16), block(256);
   if (HID == 256) {
@@ -602,7 +617,7 @@ extern "C" void launch_policy_value_step(
                        (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
                        obs_mirror, action_out, logp_out, value_out, nmean,
                        nvar, B, OBS, ACT, min_scale, aff_scale, aff_shift,
-                       log_aff_scale, greedy, seed, draw_buf);
+                       log_aff_scale, greedy, seed, draw_buf, draw_offset);
   } else {
     hipLaunchKernelGGL(policy_value_step_kernel<128>, grid, block, 0, s, obs,
                        (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
@@ -610,9 +625,9 @@ extern "C" void launch_policy_value_step(
                        (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
                        obs_mirror, action_out, logp_out, value_out, nmean,
                        nvar, B, OBS, ACT, min_scale, aff_scale, aff_shift,
-                       log_aff_scale, greedy, seed, draw_buf);
+                       log_aff_scale, greedy, seed, draw_buf, draw_offset);
   }
-  if (draw_buf)
+  if (draw_buf && do_bump)
     hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
 }
 
@@ -682,7 +697,8 @@ extern "C" void launch_ppo_head_loss(
     float* metrics, int B,
     int ACT, float clip_eps, float ent_coef, float vf_coef, float min_scale,
     float aff_scale, float aff_shift, float log_aff_scale, uint64_t seed,
-    unsigned int* draw_buf, void* stream) {
+    unsigned int* draw_buf, unsigned int draw_offset, int do_bump,
+    void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int threads = 256;
   int blocks = (B + threads - 1) / threads;
@@ -692,7 +708,7 @@ extern "C" void launch_ppo_head_loss(
                      (bf16_t*)dv_out, (bf16_t*)dv16_out, metrics, B, ACT,
                      clip_eps, ent_coef,
                      vf_coef, min_scale, aff_scale, aff_shift, log_aff_scale,
-                     1.0f / (float)B, seed, draw_buf);
-  if (draw_buf)
+                     1.0f / (float)B, seed, draw_buf, draw_offset);
+  if (draw_buf && do_bump)
     hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
 }

@@ -152,10 +152,13 @@ extern "C" void launch_fused_adam_bf16(float* param, const void* grad,
                                        void* param_bf16, long n, float lr,
                                        float beta1, float beta2, float eps,
                                        float max_norm, float grad_scale,
-                                       void* stream) {
+                                       int do_prologue, void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  hipLaunchKernelGGL(adam_prologue_kernel, dim3(1), dim3(1), 0, s, sqnorm,
-                     step_t);
+  // the prologue (zero sqnorm, step+1) can ride along in slab_reduce
+  // (wgrad.hip) when that kernel precedes this one in the same stream
+  if (do_prologue)
+    hipLaunchKernelGGL(adam_prologue_kernel, dim3(1), dim3(1), 0, s, sqnorm,
+                       step_t);
   int threads = 256;
   long want = (n / 8 + threads - 1) / threads;
   int blocks = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);

@@ -136,7 +136,15 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
 // the next minibatch. One launch covers a whole chain's gradient.
 extern "C" __global__ void slab_reduce_kernel(float* __restrict__ slab,
                                               __bf16* __restrict__ grad16,
-                                              long slab_stride, long n) {
+                                              long slab_stride, long n,
+                                              float* __restrict__ sqnorm,
+                                              long* __restrict__ step_t) {
+  // fused Adam prologue (zero the norm accumulator, bump the step counter)
+  // rides along: this kernel always runs right before the chain's Adam
+  if (sqnorm && blockIdx.x == 0 && threadIdx.x == 0) {
+    *sqnorm = 0.0f;
+    *step_t += 1;
+  }
   long i = (long)blockIdx.x * blockDim.x + threadIdx.x;
   long stride = (long)gridDim.x * blockDim.x;
   for (; i < n; i += stride) {
@@ -179,7 +187,17 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
                              void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int NT = (N_VALID + 15) / 16;
-  int KPG = (K % 128 == 0) ? 8 : ((K % 32 == 0) ? 2 : 1);
+  // largest KPG whose grid still spans >= 16 workgroups: small-N shapes
+  // (head grads, NT=1) would otherwise run on 8 CUs of 256
+  int KPG = 1;
+  const int cands[3] = {8, 2, 1};
+  for (int ci = 0; ci < 3; ++ci) {
+    int cand = cands[ci];
+    if (K % (16 * cand) == 0 && (cand == 1 || NT * (K / (16 * cand)) >= 16)) {
+      KPG = cand;
+      break;
+    }
+  }
   int KTG = K / (16 * KPG);
   dim3 grid(NT, KTG, 4), block(256);
   if (KPG == 8) {
@@ -198,13 +216,14 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
 }
 
 extern "C" void launch_slab_reduce(float* slab, void* grad16,
-                                   long slab_stride, long n, void* stream) {
+                                   long slab_stride, long n, float* sqnorm,
+                                   long* step_t, void* stream) {
   int threads = 256;
   long want = (n + threads - 1) / threads;
   int blocks = (int)(want < 2048 ? (want > 0 ? want : 1) : 2048);
   hipLaunchKernelGGL(slab_reduce_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, slab, (__bf16*)grad16, slab_stride,
-                     n);
+                     n, sqnorm, step_t);
 }
 
 extern "C" void launch_tr16_probe(const void* in, float* out, int base_mode,

@@ -333,13 +333,15 @@ class FusedPPOEngine:
                 L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t], L.buf_value[t],
                 self.empty, self.empty,
                 self.min_scale, self.aff_scale, self.aff_shift,
-                self.log_aff_scale, 0, self.seed, self.draw_policy,
+                self.log_aff_scale, 0, self.seed, self.draw_policy, t, 0,
             )
             env.hip_step_into(
                 L.buf_action[t],
                 reward_out=L.buf_reward[t],
                 discount_out=L.buf_discount[t],
                 steptype_out=self.buf_steptype[t],
+                draw_offset=t,
+                do_bump=False,
             )
             ext.value_forward(
                 hb["next_obs"],
@@ -347,6 +349,10 @@ class FusedPPOEngine:
                 c16["Wv"], cc.views["bv"],
                 L.buf_bootstrap[t], self.empty, self.empty,
             )
+        # one counter bump per rollout (the per-step draws used frozen
+        # offsets t baked into the graph nodes)
+        ext.bump_add(self.draw_policy, L.T)
+        ext.bump_add(hb["draw"], L.T)
         # truncation flags for GAE (StepType.TRUNCATED == 3)
         L.buf_truncated.copy_(self.buf_steptype == 3)
 
@@ -398,7 +404,7 @@ class FusedPPOEngine:
                 self.val_mb, self.adv_mb, self.tgt_mb, self.dhead, self.dv,
                 self.dv16, self.metrics, self.clip_eps, self.ent_coef,
                 self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
-                self.log_aff_scale, self.seed, self.draw_ent,
+                self.log_aff_scale, self.seed, self.draw_ent, mb, 0,
             )
             torch.mm(self.dhead, a16["Wh"], out=self.dH2a)
             torch.mm(self.dv, c16["Wv"].view(1, self.H), out=self.dH2c)
@@ -419,8 +425,8 @@ class FusedPPOEngine:
             torch.mm(self.dZ, c16["W2"], out=self.dH1)
             ext.silu_bwd(self.dH1, self.Z1c, self.dZ)
             ext.wgrad(self.dZ, self.Xmb, cc.slab, co["W1"], co["b1"], self.H)
-            ext.slab_reduce(ac.slab, ac.grad16)
-            ext.slab_reduce(cc.slab, cc.grad16)
+            ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
+            ext.slab_reduce(cc.slab, cc.grad16, cc.sqnorm, cc.step_t)
             # ---- all-reduce + fused clip/Adam (+ bf16 mirror refresh)
             if self.world > 1:
                 dist.all_reduce(ac.grad16)
@@ -428,12 +434,15 @@ class FusedPPOEngine:
             gscale = 1.0 / float(self.world)
             ext.fused_adam_bf16(
                 ac.flat, ac.grad16, ac.m, ac.v, ac.sqnorm, ac.step_t,
-                ac.flat16, ac.lr, 0.9, 0.999, 1e-5, self.max_grad_norm, gscale,
+                ac.flat16, ac.lr, 0.9, 0.999, 1e-5, self.max_grad_norm,
+                gscale, 0,
             )
             ext.fused_adam_bf16(
                 cc.flat, cc.grad16, cc.m, cc.v, cc.sqnorm, cc.step_t,
-                cc.flat16, cc.lr, 0.9, 0.999, 1e-5, self.max_grad_norm, gscale,
+                cc.flat16, cc.lr, 0.9, 0.999, 1e-5, self.max_grad_norm,
+                gscale, 0,
             )
+        ext.bump_add(self.draw_ent, n_mb)
         return {k: v for k, v in self.metric_views.items()}
 
 

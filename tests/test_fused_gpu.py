@@ -105,7 +105,7 @@ def test_policy_value_step_matches_eager(ext, H, OBS, ACT):
     ext.policy_value_step(
         obs, bf(W1a), b1a, bf(W2a), b2a, bf(Wha), bha, bf(W1c), b1c, bf(W2c),
         b2c, bf(Wvc), bvc, obs_mirror, action, logp, value, e, e,
-        1e-3, 1.0, 0.0, 0.0, 1, 1234, draw,
+        1e-3, 1.0, 0.0, 0.0, 1, 1234, draw, 0, 1,
     )
     torch.cuda.synchronize()
     assert draw.item() == 1  # counter bumped
@@ -149,7 +149,7 @@ def test_policy_step_sampling_statistics(ext):
         ext.policy_value_step(
             obs, bf(W1a), b1a, bf(W2a), b2a, bf(Wha), bha, bf(W1c), b1c,
             bf(W2c), b2c, bf(Wvc), bvc, e, action, logp, value, e, e,
-            1e-3, 1.0, 0.0, 0.0, 0, 99, draw,
+            1e-3, 1.0, 0.0, 0.0, 0, 99, draw, 0, 1,
         )
         samples.append(action.clone())
     torch.cuda.synchronize()
@@ -195,6 +195,7 @@ def test_ppo_head_loss_kernel_matches_reference(ext):
     ext.ppo_head_loss(
         heads, v16, action, old_logp, old_value, adv, targets, dhead, dv,
         dv16, metrics, clip_eps, 0.0, vf_coef, 1e-3, 1.0, 0.0, 0.0, 42, draw,
+        0, 1,
     )
     torch.cuda.synchronize()
     assert draw.item() == 1
@@ -250,7 +251,7 @@ def test_fused_adam_bf16_matches_torch(ext):
     for gr in grads:
         g16 = gr.bfloat16()
         ext.fused_adam_bf16(p32, g16, m, v, sqn, st, p16, lr, 0.9, 0.999,
-                            1e-5, max_norm, 1.0)
+                            1e-5, max_norm, 1.0, 1)
         ref_p.grad = g16.float()
         torch.nn.utils.clip_grad_norm_([ref_p], max_norm)
         opt.step()
@@ -416,8 +417,10 @@ def test_wgrad_kernel_matches_mm(ext, N, K, NV):
     numel = NV * K + NV
     slab = torch.zeros(16, numel, device="cuda")
     grad16 = torch.zeros(numel, dtype=torch.bfloat16, device="cuda")
+    e0 = torch.zeros(0, device="cuda")
+    e1 = torch.zeros(0, dtype=torch.int64, device="cuda")
     ext.wgrad(dZ, X, slab, 0, NV * K, NV)
-    ext.slab_reduce(slab, grad16)
+    ext.slab_reduce(slab, grad16, e0, e1)
     torch.cuda.synchronize()
     dW = grad16[: NV * K].view(NV, K).float()
     db = grad16[NV * K :].float()

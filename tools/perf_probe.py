@@ -80,7 +80,7 @@ def main():
             cc.views["b2"], c16["Wv"], cc.views["bv"], L.buf_obs[0],
             L.buf_action[0], L.buf_log_prob[0], L.buf_value[0], F.empty,
             F.empty, F.min_scale, F.aff_scale, F.aff_shift, F.log_aff_scale,
-            0, F.seed, F.draw_policy)
+            0, F.seed, F.draw_policy, 0, 1)
 
     def k_value():
         ext.value_forward(hb["next_obs"], c16["W1"], cc.views["b1"],
@@ -89,7 +89,7 @@ def main():
 
     def k_env():
         env.hip_step_into(L.buf_action[0], L.buf_reward[0], L.buf_discount[0],
-                          F.buf_steptype[0])
+                          F.buf_steptype[0], 0, True)
 
     print(f"policy_value_step:  {timeit(k_policy, 50)*1e3:8.1f} us")
     print(f"value_forward:      {timeit(k_value, 50)*1e3:8.1f} us")
@@ -129,7 +129,7 @@ def main():
             F.heads, F.vpred.view(-1), F.act_mb, F.logp_mb, F.val_mb,
             F.adv_mb, F.tgt_mb, F.dhead, F.dv, F.dv16, F.metrics, F.clip_eps,
             F.ent_coef, F.vf_coef, F.min_scale, F.aff_scale, F.aff_shift,
-            F.log_aff_scale, F.seed, F.draw_ent)
+            F.log_aff_scale, F.seed, F.draw_ent, 0, 1)
 
     def k_wgrad():
         torch.mm(F.dZ.t(), F.H1a, out=ac.gviews16["W2"])
@@ -138,7 +138,7 @@ def main():
         ext.wgrad(F.dZ, F.H1a, ac.slab, ac.offsets["W2"], ac.offsets["b2"], F.H)
 
     def k_slab_reduce():
-        ext.slab_reduce(ac.slab, ac.grad16)
+        ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
 
     def k_dgrad():
         torch.mm(F.dZ, a16["W2"], out=F.dH1)
@@ -149,7 +149,7 @@ def main():
     def k_adam():
         ext.fused_adam_bf16(ac.flat, ac.grad16, ac.m, ac.v, ac.sqnorm,
                             ac.step_t, ac.flat16, ac.lr, 0.9, 0.999, 1e-5,
-                            F.max_grad_norm, 1.0)
+                            F.max_grad_norm, 1.0, 1)
 
     print(f"ppo_gather:         {timeit(k_gather, 50)*1e3:8.1f} us")
     print(f"gemm fwd L1 (pad):  {timeit(k_gemm_fwd, 50)*1e3:8.1f} us")
