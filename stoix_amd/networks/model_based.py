@@ -117,3 +117,71 @@ class RewardBasedWorldModel(nn.Module):
         pol = self.policy_head(self.policy_torso(h))
         val, val_logits = self.value_head(self.value_torso(h))
         return ModelOutput(h, new_state, reward, reward_logits, pol, val, val_logits)
+
+
+class ContinuousModelOutput(NamedTuple):
+    hidden: Tensor
+    rnn_state: list
+    reward: Tensor
+    reward_logits: Tensor
+    policy_loc: Tensor
+    policy_scale: Tensor
+    value: Tensor
+    value_logits: Tensor
+
+
+class ContinuousRewardBasedWorldModel(nn.Module):
+    """Continuous-action world model for Sampled MuZero (reference
+    ff_sampled_mz.py uses the same RewardBasedWorldModel with an
+    action-sampled search; here the dynamics consume the raw action vector
+    and the policy head is a tanh-normal parameterisation)."""
+
+    def __init__(
+        self,
+        obs_dim: int,
+        action_dim: int,
+        hidden_dim: int = 128,
+        repr_layers=(128,),
+        head_layers=(64,),
+        rnn_layers: int = 1,
+        vmin: float = -50.0,
+        vmax: float = 50.0,
+        num_atoms: int = 51,
+        min_scale: float = 1e-3,
+    ):
+        super().__init__()
+        self.action_dim = action_dim
+        self.hidden_dim = hidden_dim
+        self.min_scale = min_scale
+        self.repr_net = MLPTorso(obs_dim, (*repr_layers, hidden_dim))
+        self.dynamics = StackedRNN(action_dim, hidden_dim, num_layers=rnn_layers, cell_type="gru")
+        self.reward_head = CategoricalValueHead(hidden_dim, vmin, vmax, num_atoms)
+        self.policy_torso = MLPTorso(hidden_dim, head_layers)
+        self.loc_head = orthogonal_init(nn.Linear(self.policy_torso.output_dim, action_dim), scale=0.01)
+        self.scale_head = orthogonal_init(nn.Linear(self.policy_torso.output_dim, action_dim), scale=0.01)
+        self.value_torso = MLPTorso(hidden_dim, head_layers)
+        self.value_head = CategoricalValueHead(self.value_torso.output_dim, vmin, vmax, num_atoms)
+
+    def _policy(self, h: Tensor) -> Tuple[Tensor, Tensor]:
+        p = self.policy_torso(h)
+        loc = self.loc_head(p)
+        scale = F.softplus(self.scale_head(p)) + self.min_scale
+        return loc, scale
+
+    def initial_inference(self, obs: Tensor) -> ContinuousModelOutput:
+        h = RewardBasedWorldModel._normalize_hidden(self.repr_net(obs))
+        rnn_state = [h.clone() for _ in self.dynamics.initial_state(obs.shape[0], obs.device)]
+        loc, scale = self._policy(h)
+        val, val_logits = self.value_head(self.value_torso(h))
+        zero_r = torch.zeros_like(val)
+        zero_rl = torch.zeros_like(val_logits)
+        return ContinuousModelOutput(h, rnn_state, zero_r, zero_rl, loc, scale, val, val_logits)
+
+    def recurrent_inference(self, rnn_state: list, action: Tensor) -> ContinuousModelOutput:
+        h, new_state = self.dynamics(action.to(torch.float32), rnn_state)
+        h = RewardBasedWorldModel._normalize_hidden(h)
+        new_state = [RewardBasedWorldModel._normalize_hidden(s) if isinstance(s, Tensor) else s for s in new_state]
+        reward, reward_logits = self.reward_head(h)
+        loc, scale = self._policy(h)
+        val, val_logits = self.value_head(self.value_torso(h))
+        return ContinuousModelOutput(h, new_state, reward, reward_logits, loc, scale, val, val_logits)

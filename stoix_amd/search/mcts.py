@@ -188,3 +188,160 @@ def mcts_search(
         gnoise = -torch.log((-torch.log(u.clamp(min=1e-10))).clamp(min=1e-10))
         action = (logits + gnoise).argmax(dim=-1)
     return SearchOutput(action=action, action_weights=weights, search_value=search_value)
+
+
+class SampledSearchOutput(NamedTuple):
+    action: Tensor  # [B, act_dim] chosen continuous action
+    sampled_actions: Tensor  # [B, K, act_dim] root candidate actions
+    action_weights: Tensor  # [B, K] normalised root visit counts
+    search_value: Tensor  # [B] root value estimate
+
+
+def sampled_mcts_search(
+    root_obs: Tensor,
+    root_embedding: Dict[str, Tensor],
+    root_candidates: Tensor,  # [B, K, act_dim] sampled from the root policy
+    root_value: Tensor,
+    recurrent_fn: Callable,
+    num_simulations: int,
+    c_puct: float = 1.25,
+    temperature: float = 1.0,
+    generator: Optional[torch.Generator] = None,
+) -> SampledSearchOutput:
+    """Sampled MCTS for continuous actions (parity surface: mctx's sampled
+    policies as used by /root/reference/stoix/systems/search/ff_sampled_az.py
+    and ff_sampled_mz.py).
+
+    Each node holds K candidate actions drawn from the node's policy; the
+    tree then searches over the K discrete arms. Because arms are i.i.d.
+    samples from the current policy, the sampled prior over arms is uniform
+    (the policy density is already represented by the sampling itself —
+    the Sampled MuZero beta-correction cancels for beta = pi).
+
+    ``recurrent_fn(embedding, action[B, act_dim]) ->
+        (embedding, reward, discount, candidates[B, K, act_dim], value)``
+    """
+    device = root_obs.device
+    B, K, act_dim = root_candidates.shape
+    N = num_simulations + 1
+    arena = _Arena(B, N, K, device, root_value)
+    bidx = arena.batch
+    arena.prior[:, 0] = 1.0 / K
+    arena.discount[:, 0] = 1.0
+    cand = torch.zeros((B, N, K, act_dim), device=device)
+    cand[:, 0] = root_candidates
+
+    emb_arena = {
+        k: torch.zeros((B, N, *v.shape[1:]), dtype=v.dtype, device=device)
+        for k, v in root_embedding.items()
+    }
+    for k, v in root_embedding.items():
+        emb_arena[k][:, 0] = v
+
+    arena.visit[:, 0] = 1.0
+    arena.value_sum[:, 0] = root_value
+    arena.min_q = torch.minimum(arena.min_q, root_value)
+    arena.max_q = torch.maximum(arena.max_q, root_value)
+
+    for sim in range(1, num_simulations + 1):
+        node = torch.zeros(B, dtype=torch.long, device=device)
+        descending = torch.ones(B, dtype=torch.bool, device=device)
+        sel_parent = torch.zeros(B, dtype=torch.long, device=device)
+        sel_action = torch.zeros(B, dtype=torch.long, device=device)
+        for _ in range(sim):
+            child_idx = arena.children[bidx, node]
+            child_visit = torch.where(
+                child_idx >= 0,
+                arena.visit[bidx.unsqueeze(1), child_idx.clamp(min=0)],
+                torch.zeros_like(child_idx, dtype=torch.float32),
+            )
+            child_q = torch.where(
+                child_idx >= 0,
+                arena.value_sum[bidx.unsqueeze(1), child_idx.clamp(min=0)] / child_visit.clamp(min=1),
+                torch.zeros_like(child_visit),
+            )
+            child_r = torch.where(
+                child_idx >= 0,
+                arena.reward[bidx.unsqueeze(1), child_idx.clamp(min=0)],
+                torch.zeros_like(child_visit),
+            )
+            child_g = torch.where(
+                child_idx >= 0,
+                arena.discount[bidx.unsqueeze(1), child_idx.clamp(min=0)],
+                torch.zeros_like(child_visit),
+            )
+            q_edge = child_r + child_g * child_q
+            span = (arena.max_q - arena.min_q).clamp(min=1e-3).unsqueeze(1)
+            q_norm = torch.where(
+                child_visit > 0,
+                (q_edge - arena.min_q.unsqueeze(1)) / span,
+                torch.zeros_like(q_edge),
+            )
+            parent_visit = arena.visit[bidx, node].unsqueeze(1)
+            ucb = q_norm + c_puct * arena.prior[bidx, node] * torch.sqrt(
+                parent_visit.clamp(min=1)
+            ) / (1.0 + child_visit)
+            best_a = ucb.argmax(dim=-1)
+            has_child = child_idx.gather(1, best_a.unsqueeze(1)).squeeze(1) >= 0
+            stop_here = descending & ~has_child
+            sel_parent = torch.where(stop_here, node, sel_parent)
+            sel_action = torch.where(stop_here, best_a, sel_action)
+            step_child = child_idx.gather(1, best_a.unsqueeze(1)).squeeze(1).clamp(min=0)
+            node = torch.where(descending & has_child, step_child, node)
+            descending = descending & has_child
+        sel_parent = torch.where(descending, node, sel_parent)
+        if bool(descending.any()):
+            best_a = arena.prior[bidx, node].argmax(dim=-1)
+            sel_action = torch.where(descending, best_a, sel_action)
+
+        parent_emb = {k: v[bidx, sel_parent] for k, v in emb_arena.items()}
+        action_cont = cand[bidx, sel_parent, sel_action]  # [B, act_dim]
+        new_emb, reward, discount, new_cand, value = recurrent_fn(parent_emb, action_cont)
+        new_node = sim
+        for k, v in new_emb.items():
+            emb_arena[k][:, new_node] = v
+        arena.reward[:, new_node] = reward
+        arena.discount[:, new_node] = discount
+        arena.prior[:, new_node] = 1.0 / K
+        cand[:, new_node] = new_cand
+        arena.parent[:, new_node] = sel_parent
+        arena.act_from_parent[:, new_node] = sel_action
+        arena.children[bidx, sel_parent, sel_action] = new_node
+
+        g = value.clone()
+        cur = torch.full((B,), new_node, dtype=torch.long, device=device)
+        alive = torch.ones(B, dtype=torch.bool, device=device)
+        for _ in range(sim + 1):
+            af = alive.to(torch.float32)
+            arena.visit[bidx, cur] += af
+            arena.value_sum[bidx, cur] += g * af
+            q_here = arena.value_sum[bidx, cur] / arena.visit[bidx, cur].clamp(min=1)
+            arena.min_q = torch.where(alive, torch.minimum(arena.min_q, q_here), arena.min_q)
+            arena.max_q = torch.where(alive, torch.maximum(arena.max_q, q_here), arena.max_q)
+            g = arena.reward[bidx, cur] + arena.discount[bidx, cur] * g
+            nxt = arena.parent[bidx, cur]
+            alive = alive & (nxt >= 0)
+            cur = nxt.clamp(min=0)
+
+    root_children = arena.children[:, 0]
+    counts = torch.where(
+        root_children >= 0,
+        arena.visit[bidx.unsqueeze(1), root_children.clamp(min=0)],
+        torch.zeros_like(root_children, dtype=torch.float32),
+    )
+    weights = counts / counts.sum(-1, keepdim=True).clamp(min=1e-9)
+    search_value = arena.value_sum[:, 0] / arena.visit[:, 0].clamp(min=1)
+    if temperature <= 0:
+        arm = weights.argmax(dim=-1)
+    else:
+        logits = torch.log(weights.clamp(min=1e-9)) / temperature
+        u = torch.rand(logits.shape, device=device, generator=generator)
+        gnoise = -torch.log((-torch.log(u.clamp(min=1e-10))).clamp(min=1e-10))
+        arm = (logits + gnoise).argmax(dim=-1)
+    action = root_candidates[bidx, arm]
+    return SampledSearchOutput(
+        action=action,
+        sampled_actions=root_candidates,
+        action_weights=weights,
+        search_value=search_value,
+    )
