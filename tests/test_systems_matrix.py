@@ -85,6 +85,12 @@ SYSTEMS = [
     ("stoix_amd.systems.search.ff_sampled_az", "default/anakin/default_ff_sampled_az.yaml",
      ["system.num_simulations=4", "system.num_sampled_actions=4",
       "system.num_minibatches=2", "system.epochs=1"]),
+    ("stoix_amd.systems.spo.ff_spo", "default/anakin/default_ff_spo.yaml",
+     ["system.num_particles=4", "system.search_depth=2", "system.num_minibatches=2",
+      "system.epochs=1"]),
+    ("stoix_amd.systems.spo.ff_spo", "default/anakin/default_ff_spo_continuous.yaml",
+     ["system.num_particles=4", "system.search_depth=2", "system.num_minibatches=2",
+      "system.epochs=1"]),
     ("stoix_amd.systems.search.ff_sampled_mz", "default/anakin/default_ff_sampled_mz.yaml",
      ["system.num_simulations=4", "system.num_sampled_actions=4", "system.epochs=1",
       "system.unroll_steps=2", "system.n_step=2", "system.batch_size=8",
