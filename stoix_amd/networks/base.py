@@ -162,3 +162,19 @@ class ChainedTorsos(nn.Module):
         for t in self.torsos:
             x = t(x)
         return x
+
+
+class SharedPolicyValueNetwork(nn.Module):
+    """One torso feeding a joint PolicyValueHead (IMPALA shared-torso /
+    AZ-style nets; reference base.py + heads.py:220-232). forward returns
+    ``(Distribution, value)``."""
+
+    def __init__(self, torso: nn.Module, action_head: nn.Module):
+        super().__init__()
+        from stoix_amd.networks.heads import PolicyValueHead
+
+        self.torso = torso
+        self.head = PolicyValueHead(torso.output_dim, action_head)
+
+    def forward(self, obs: Tensor):
+        return self.head(self.torso(obs))

@@ -93,3 +93,20 @@ def test_sebulba_impala_end_to_end():
     )
     r = run(cfg)
     assert r == r
+
+
+def test_sebulba_impala_shared_torso_end_to_end():
+    from stoix_amd.systems.impala.sebulba_ff_impala_shared_torso import run
+
+    cfg = compose(
+        "default/sebulba/default_ff_impala_shared_torso.yaml",
+        [
+            "arch.total_num_envs=8", "arch.total_timesteps=null", "arch.num_updates=3",
+            "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2",
+            "logger.loggers=[]", "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r == r
