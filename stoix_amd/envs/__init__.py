@@ -35,9 +35,29 @@ def _classic(name: str):
 
 
 def _brax(name: str):
-    table = {"ant": Ant}
+    from stoix_amd.envs.humanoid import Humanoid
+
+    table = {"ant": Ant, "humanoid": Humanoid}
     if name not in table:
         raise ValueError(f"unknown brax-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
+def _jumanji(name: str):
+    from stoix_amd.envs.snake import Snake
+
+    table = {"snake": Snake, "Snake-v1": Snake}
+    if name not in table:
+        raise ValueError(f"unknown jumanji-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
+def _envpool(name: str):
+    from stoix_amd.envs.breakout import Breakout
+
+    table = {"breakout": Breakout, "Breakout-v5": Breakout}
+    if name not in table:
+        raise ValueError(f"unknown envpool-suite env '{name}' (have {list(table)})")
     return table[name]
 
 
@@ -51,6 +71,8 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "classic": _classic,
     "gymnax": _classic,  # alias: the reference's gymnax suite maps to classic control here
     "brax": _brax,
+    "jumanji": _jumanji,
+    "envpool": _envpool,
     "debug": _debug,
 }
 

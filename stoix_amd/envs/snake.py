@@ -1,0 +1,122 @@
+"""Snake-v1 grid game (vectorised; Jumanji-class suite).
+
+Fills the role of Jumanji Snake-v1 in the reference's configs
+(/root/reference/stoix/configs/env/jumanji/snake.yaml; BASELINE.json config
+#5: Anakin Rainbow-DQN). Jumanji is JAX-only; this is an original, fully
+tensorised torch implementation (every op batched over B boards, so it runs
+on CPU and on the GPU device tensors without per-env Python).
+
+Board 12x12. Action in {0: up, 1: right, 2: down, 3: left}. Reward +1 per
+fruit. Terminates on wall hit, self-collision, or the step limit. The body
+is stored as a countdown grid: cell value = steps until that segment
+vanishes (head = current length); eating a fruit skips the decrement
+(the tail stays put -> the snake grows).
+
+Observation [12, 12, 5] float32 channels: body mask, head one-hot, tail
+one-hot, fruit one-hot, normalised body order (countdown / length) — the
+Jumanji Snake observation layout.
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from stoix_amd.envs.env import State, StatefulVecEnv, Tensor
+from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
+
+ROWS, COLS = 12, 12
+# action -> (drow, dcol)
+_DR = [-1, 0, 1, 0]
+_DC = [0, 1, 0, -1]
+
+
+class Snake(StatefulVecEnv):
+    max_episode_steps = 4000
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((ROWS, COLS, 5), 0.0, 1.0)
+        self.action_space = DiscreteSpace(4)
+        self._dr = torch.tensor(_DR, device=self.device)
+        self._dc = torch.tensor(_DC, device=self.device)
+
+    # ------------------------------------------------------------ state ops
+
+    def _spawn_fruit(self, grid: Tensor, n: int) -> Tuple[Tensor, Tensor]:
+        """Uniform fruit position over empty cells per board: [n] rows, cols."""
+        empty = grid <= 0  # [n, R, C]
+        flat = empty.reshape(n, ROWS * COLS).float()
+        # multinomial over empty cells (there is always at least one until
+        # the board is full, at which point position is irrelevant)
+        probs = flat.clamp(min=1e-9)
+        idx = torch.multinomial(probs, 1, generator=self.gen).squeeze(-1)
+        return idx // COLS, idx % COLS
+
+    def _reset_fn(self, n: int) -> State:
+        grid = torch.zeros(n, ROWS, COLS, dtype=torch.int32, device=self.device)
+        hr = torch.full((n,), ROWS // 2, dtype=torch.long, device=self.device)
+        hc = torch.full((n,), COLS // 2, dtype=torch.long, device=self.device)
+        bidx = torch.arange(n, device=self.device)
+        grid[bidx, hr, hc] = 1  # length-1 snake
+        fr, fc = self._spawn_fruit(grid, n)
+        return {
+            "grid": grid,
+            "head_r": hr,
+            "head_c": hc,
+            "fruit_r": fr,
+            "fruit_c": fc,
+            "length": torch.ones(n, dtype=torch.int32, device=self.device),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        grid = state["grid"]
+        n = grid.shape[0]
+        bidx = torch.arange(n, device=self.device)
+        length = state["length"].clamp(min=1).float()
+        body = (grid > 0).float()
+        head = torch.zeros_like(body)
+        head[bidx, state["head_r"], state["head_c"]] = 1.0
+        tail = (grid == 1).float()
+        fruit = torch.zeros_like(body)
+        fruit[bidx, state["fruit_r"], state["fruit_c"]] = 1.0
+        order = grid.float() / length.view(-1, 1, 1)
+        return torch.stack([body, head, tail, fruit, order], dim=-1)
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        grid = state["grid"].clone()
+        n = grid.shape[0]
+        bidx = torch.arange(n, device=self.device)
+        a = action.long().clamp(0, 3)
+        nr = state["head_r"] + self._dr[a]
+        nc = state["head_c"] + self._dc[a]
+        hit_wall = (nr < 0) | (nr >= ROWS) | (nc < 0) | (nc >= COLS)
+        nr_s = nr.clamp(0, ROWS - 1)
+        nc_s = nc.clamp(0, COLS - 1)
+        ate = (nr_s == state["fruit_r"]) & (nc_s == state["fruit_c"]) & ~hit_wall
+        # decrement body countdown unless the snake grew
+        dec = (~ate).int().view(-1, 1, 1)
+        grid = torch.where(grid > 0, grid - dec, grid)
+        # self-collision: new head lands on a still-occupied cell
+        hit_self = grid[bidx, nr_s, nc_s] > 0
+        terminated = hit_wall | hit_self
+        length = state["length"] + ate.int()
+        grid[bidx, nr_s, nc_s] = torch.where(
+            terminated, grid[bidx, nr_s, nc_s], length
+        )
+        # respawn fruit where eaten
+        fr, fc = state["fruit_r"].clone(), state["fruit_c"].clone()
+        if bool(ate.any()):
+            nfr, nfc = self._spawn_fruit(grid, n)
+            fr = torch.where(ate, nfr, fr)
+            fc = torch.where(ate, nfc, fc)
+        reward = ate.float()
+        new_state = {
+            "grid": grid,
+            "head_r": torch.where(terminated, state["head_r"], nr_s),
+            "head_c": torch.where(terminated, state["head_c"], nc_s),
+            "fruit_r": fr,
+            "fruit_c": fc,
+            "length": length,
+        }
+        return new_state, reward, terminated

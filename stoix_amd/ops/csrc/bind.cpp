@@ -19,6 +19,10 @@ void launch_ant_step(float*, const float*, int*, float*, int*, float*, int*,
                      float*, float*, float*, float*, unsigned char*,
                      unsigned char*, int, int, uint64_t, unsigned int*,
                      unsigned int, int, void*);
+void launch_humanoid_step(float*, const float*, int*, float*, int*, float*,
+                          int*, float*, float*, float*, float*,
+                          unsigned char*, unsigned char*, int, int, uint64_t,
+                          unsigned int*, unsigned int, int, void*);
 void launch_ant_reset(float*, int, uint64_t, uint32_t, void*);
 void launch_gae(const float*, const float*, const float*, const float*,
                 const unsigned char*, float*, float*, int, int, float, void*);
@@ -116,6 +120,30 @@ void ant_step(torch::Tensor state, torch::Tensor action,
   CHK(action, torch::kFloat32);
   int B = state.size(0);
   launch_ant_step(
+      state.data_ptr<float>(), action.data_ptr<float>(),
+      step_count.data_ptr<int>(), ep_return.data_ptr<float>(),
+      ep_length.data_ptr<int>(), last_ep_return.data_ptr<float>(),
+      last_ep_length.data_ptr<int>(), obs_out.data_ptr<float>(),
+      next_obs_out.data_ptr<float>(), reward_out.data_ptr<float>(),
+      discount_out.data_ptr<float>(), steptype_out.data_ptr<unsigned char>(),
+      done_out.data_ptr<unsigned char>(), B, (int)max_episode_steps,
+      (uint64_t)seed, (unsigned int*)draw_buf.data_ptr<int>(),
+      (unsigned int)draw_offset, (int)do_bump, cur_stream());
+}
+
+void humanoid_step(torch::Tensor state, torch::Tensor action,
+                   torch::Tensor step_count, torch::Tensor ep_return,
+                   torch::Tensor ep_length, torch::Tensor last_ep_return,
+                   torch::Tensor last_ep_length, torch::Tensor obs_out,
+                   torch::Tensor next_obs_out, torch::Tensor reward_out,
+                   torch::Tensor discount_out, torch::Tensor steptype_out,
+                   torch::Tensor done_out, int64_t max_episode_steps,
+                   int64_t seed, torch::Tensor draw_buf, int64_t draw_offset,
+                   int64_t do_bump) {
+  CHK(state, torch::kFloat32);
+  CHK(action, torch::kFloat32);
+  int B = state.size(0);
+  launch_humanoid_step(
       state.data_ptr<float>(), action.data_ptr<float>(),
       step_count.data_ptr<int>(), ep_return.data_ptr<float>(),
       ep_length.data_ptr<int>(), last_ep_return.data_ptr<float>(),
@@ -385,6 +413,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("cartpole_step", &cartpole_step, "fused CartPole env step");
   m.def("ant_step", &ant_step, "fused Ant env step");
   m.def("ant_reset", &ant_reset, "Ant reset");
+  m.def("humanoid_step", &humanoid_step, "fused Humanoid env step");
   m.def("gae", &gae, "truncation-aware GAE reverse scan");
   m.def("lambda_returns", &lambda_returns, "lambda returns reverse scan");
   m.def("vtrace", &vtrace, "vtrace errors + pg advantage");

@@ -340,6 +340,215 @@ extern "C" __global__ void ant_step_kernel(
   done_out[b] = done ? 1 : 0;
 }
 
+
+// ----------------------------------------------------- humanoid (biped)
+// Mirrors stoix_amd/envs/humanoid.py EXACTLY (17 damped joints with soft
+// limits; two feet from hip-pitch/roll + knee; penalty contacts; forward
+// reward). One env per thread; the 47 state floats live in registers.
+
+#define HUM_STATE 47
+#define HUM_OBS 45
+#define HUM_ACT 17
+
+extern "C" __global__ void humanoid_step_kernel(
+    float* __restrict__ state,          // [B, 47]
+    const float* __restrict__ action,   // [B, 17]
+    int* __restrict__ step_count,
+    float* __restrict__ ep_return,
+    int* __restrict__ ep_length,
+    float* __restrict__ last_ep_return,
+    int* __restrict__ last_ep_length,
+    float* __restrict__ obs_out,        // [B, 45]
+    float* __restrict__ next_obs_out,   // [B, 45]
+    float* __restrict__ reward_out,
+    float* __restrict__ discount_out,
+    unsigned char* __restrict__ steptype_out,
+    unsigned char* __restrict__ done_out,
+    int B, int max_episode_steps, uint64_t seed,
+    const unsigned int* __restrict__ draw_buf, unsigned int draw_offset) {
+  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  if (b >= B) return;
+  uint32_t draw = *draw_buf + draw_offset;
+
+  // constants (keep EXACTLY in sync with stoix_amd/envs/humanoid.py)
+  const float TORSO_MASS = 40.0f, TORSO_INERTIA = 2.0f, TORSO_Z0 = 1.3f;
+  const float HIP_SEP = 0.12f, LU = 0.45f, LL = 0.45f;
+  const float JOINT_INERTIA = 0.12f, JOINT_DAMPING = 2.0f, GEAR = 60.0f;
+  const float LIMIT = 1.2f, KNEE_LO = 0.02f, KNEE_HI = 2.0f, LIMIT_K = 80.0f;
+  const float KN = 1.2e4f, KD = 300.0f, FRICTION = 1.0f, GRAV = -9.81f;
+  const float DT = 0.015f;
+  const int SUBSTEPS = 4;
+  const float CTRL_COST = 0.1f, HEALTHY = 5.0f, FORWARD_W = 1.25f;
+  const float Z_MIN = 0.8f, Z_MAX = 2.1f;
+  const int R_HIP_X = 3, R_HIP_Y = 5, R_KNEE = 6;
+  const int L_HIP_X = 7, L_HIP_Y = 9, L_KNEE = 10;
+
+  float s[HUM_STATE];
+#pragma unroll
+  for (int i = 0; i < HUM_STATE; ++i) s[i] = state[b * HUM_STATE + i];
+  float a[HUM_ACT];
+#pragma unroll
+  for (int i = 0; i < HUM_ACT; ++i)
+    a[i] = fminf(fmaxf(action[b * HUM_ACT + i], -1.0f), 1.0f);
+
+  float x_before = s[0];
+  const float dt = DT / SUBSTEPS;
+
+  for (int sub = 0; sub < SUBSTEPS; ++sub) {
+    float* pos = s + 0;
+    float* quat = s + 3;
+    float* linvel = s + 7;
+    float* angvel = s + 10;
+    float* qpos = s + 13;   // 17
+    float* qvel = s + 30;   // 17
+
+    // ---- joints first (humanoid.py updates qpos before the foot calc)
+#pragma unroll
+    for (int j = 0; j < HUM_ACT; ++j) {
+      float lo = -LIMIT, hi = LIMIT;
+      if (j == R_KNEE || j == L_KNEE) { lo = KNEE_LO; hi = KNEE_HI; }
+      float limit_tau = -LIMIT_K * (fmaxf(qpos[j] - hi, 0.0f) - fmaxf(lo - qpos[j], 0.0f));
+      float qacc = (GEAR * a[j] - JOINT_DAMPING * qvel[j] + limit_tau) / JOINT_INERTIA;
+      qvel[j] += dt * qacc;
+      qpos[j] += dt * qvel[j];
+    }
+
+    // ---- feet from NEW joint angles, torso pose still old
+    V3 total_f = {0.f, 0.f, 0.f}, total_tau = {0.f, 0.f, 0.f};
+#pragma unroll
+    for (int side = 0; side < 2; ++side) {
+      int hx = side == 0 ? R_HIP_X : L_HIP_X;
+      int hy = side == 0 ? R_HIP_Y : L_HIP_Y;
+      int kn = side == 0 ? R_KNEE : L_KNEE;
+      float hp = qpos[hy], hr = qpos[hx], k = qpos[kn];
+      V3 body_off = {LU * sinf(hp) + LL * sinf(hp + k),
+                     (LU + LL) * sinf(hr) + (side == 0 ? HIP_SEP : -HIP_SEP),
+                     -(LU * cosf(hp) + LL * cosf(hp + k))};
+      V3 r = quat_rot(quat, body_off);
+      V3 foot_w = {pos[0] + r.x, pos[1] + r.y, pos[2] + r.z};
+      V3 av = {angvel[0], angvel[1], angvel[2]};
+      V3 fv = add3(v3(linvel[0], linvel[1], linvel[2]), cross3(av, r));
+      float pen = fmaxf(-foot_w.z, 0.0f);
+      float fn = fmaxf(KN * pen - KD * fv.z, 0.0f);
+      if (pen <= 0.0f) fn = 0.0f;
+      V3 cf = {-FRICTION * fn * tanhf(4.0f * fv.x),
+               -FRICTION * fn * tanhf(4.0f * fv.y), fn};
+      total_f = add3(total_f, cf);
+      total_tau = add3(total_tau, cross3(r, cf));
+    }
+
+    // ---- torso integration
+    linvel[0] += dt * (total_f.x / TORSO_MASS);
+    linvel[1] += dt * (total_f.y / TORSO_MASS);
+    linvel[2] += dt * (total_f.z / TORSO_MASS + GRAV);
+    pos[0] += dt * linvel[0];
+    pos[1] += dt * linvel[1];
+    pos[2] += dt * linvel[2];
+    angvel[0] += dt * (total_tau.x / TORSO_INERTIA - 0.5f * angvel[0]);
+    angvel[1] += dt * (total_tau.y / TORSO_INERTIA - 0.5f * angvel[1]);
+    angvel[2] += dt * (total_tau.z / TORSO_INERTIA - 0.5f * angvel[2]);
+    // quaternion integration (matches ant.py quat_integrate)
+    {
+      float w = quat[0], x = quat[1], y = quat[2], z = quat[3];
+      float ox = angvel[0], oy = angvel[1], oz = angvel[2];
+      float dw = 0.5f * (-x * ox - y * oy - z * oz);
+      float dx = 0.5f * (w * ox + y * oz - z * oy);
+      float dy = 0.5f * (w * oy + z * ox - x * oz);
+      float dz = 0.5f * (w * oz + x * oy - y * ox);
+      float nw = w + dt * dw, nx = x + dt * dx, ny = y + dt * dy, nz = z + dt * dz;
+      float nrm = sqrtf(fmaxf(nw * nw + nx * nx + ny * ny + nz * nz, 1e-16f));
+      quat[0] = nw / nrm; quat[1] = nx / nrm; quat[2] = ny / nrm; quat[3] = nz / nrm;
+    }
+  }
+
+  // ---- reward / termination
+  float forward_vel = (s[0] - x_before) / DT;
+  float ctrl = 0.0f;
+#pragma unroll
+  for (int i = 0; i < HUM_ACT; ++i) ctrl += a[i] * a[i];
+  float z = s[2];
+  bool finite = true;
+#pragma unroll
+  for (int i = 0; i < HUM_STATE; ++i) finite = finite && isfinite(s[i]);
+  bool healthy = (z > Z_MIN) && (z < Z_MAX) && finite;
+  float reward = FORWARD_W * forward_vel + HEALTHY - CTRL_COST * ctrl;
+  bool terminated = !healthy;
+  if (!finite) {
+#pragma unroll
+    for (int i = 0; i < HUM_STATE; ++i) s[i] = 0.0f;
+    reward = isfinite(reward) ? reward : 0.0f;
+  }
+
+  int sc = step_count[b] + 1;
+  bool truncated = (!terminated) && (sc >= max_episode_steps);
+  bool done = terminated || truncated;
+  float ret = ep_return[b] + reward;
+  int len = ep_length[b] + 1;
+  if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
+
+  // true final obs
+  {
+    float* o = next_obs_out + b * HUM_OBS;
+    o[0] = s[2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) o[1 + i] = s[3 + i];
+#pragma unroll
+    for (int i = 0; i < 3; ++i) o[5 + i] = s[7 + i];
+#pragma unroll
+    for (int i = 0; i < 3; ++i) o[8 + i] = s[10 + i];
+#pragma unroll
+    for (int i = 0; i < 17; ++i) o[11 + i] = s[13 + i];
+#pragma unroll
+    for (int i = 0; i < 17; ++i) o[28 + i] = s[30 + i];
+  }
+
+  if (done) {
+    // autoreset, mirroring Humanoid._reset_fn (philox stream 4)
+    float u[36];
+#pragma unroll
+    for (int blk = 0; blk < 9; ++blk) {
+      Rng4 r = philox_uniform4(seed, 4u, (uint32_t)b, draw * 16u + blk);
+      u[blk * 4 + 0] = r.a; u[blk * 4 + 1] = r.b;
+      u[blk * 4 + 2] = r.c; u[blk * 4 + 3] = r.d;
+    }
+#pragma unroll
+    for (int i = 0; i < HUM_STATE; ++i) s[i] = 0.0f;
+    s[2] = TORSO_Z0;
+    s[3] = 1.0f;
+#pragma unroll
+    for (int i = 0; i < 17; ++i) s[13 + i] = -0.03f + 0.06f * u[i];
+    s[13 + R_KNEE] += 0.15f;
+    s[13 + L_KNEE] += 0.15f;
+#pragma unroll
+    for (int i = 0; i < 17; ++i) s[30 + i] = -0.02f + 0.04f * u[17 + i];
+    sc = 0; ret = 0.0f; len = 0;
+  }
+
+#pragma unroll
+  for (int i = 0; i < HUM_STATE; ++i) state[b * HUM_STATE + i] = s[i];
+  {
+    float* o = obs_out + b * HUM_OBS;
+    o[0] = s[2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) o[1 + i] = s[3 + i];
+#pragma unroll
+    for (int i = 0; i < 3; ++i) o[5 + i] = s[7 + i];
+#pragma unroll
+    for (int i = 0; i < 3; ++i) o[8 + i] = s[10 + i];
+#pragma unroll
+    for (int i = 0; i < 17; ++i) o[11 + i] = s[13 + i];
+#pragma unroll
+    for (int i = 0; i < 17; ++i) o[28 + i] = s[30 + i];
+  }
+  step_count[b] = sc;
+  ep_return[b] = ret;
+  ep_length[b] = len;
+  reward_out[b] = reward;
+  discount_out[b] = terminated ? 0.0f : 1.0f;
+  steptype_out[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+  done_out[b] = done ? 1 : 0;
+}
+
 // ----------------------------------------------------------- ant reset
 extern "C" __global__ void ant_reset_kernel(
     float* __restrict__ state, int B, uint64_t seed, uint32_t draw) {
@@ -409,6 +618,26 @@ extern "C" void launch_ant_step(
   int threads = 64;
   int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(ant_step_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, state, action, step_count, ep_return,
+                     ep_length, last_ep_return, last_ep_length, obs_out,
+                     next_obs_out, reward_out, discount_out, steptype_out,
+                     done_out, B, max_episode_steps, seed, draw_buf,
+                     draw_offset);
+  if (do_bump)
+    hipLaunchKernelGGL(bump_u32_kernel, dim3(1), dim3(1), 0,
+                       (hipStream_t)stream, draw_buf);
+}
+
+extern "C" void launch_humanoid_step(
+    float* state, const float* action, int* step_count, float* ep_return,
+    int* ep_length, float* last_ep_return, int* last_ep_length, float* obs_out,
+    float* next_obs_out, float* reward_out, float* discount_out,
+    unsigned char* steptype_out, unsigned char* done_out, int B,
+    int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
+    unsigned int draw_offset, int do_bump, void* stream) {
+  int threads = 64;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(humanoid_step_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, state, action, step_count, ep_return,
                      ep_length, last_ep_return, last_ep_length, obs_out,
                      next_obs_out, reward_out, discount_out, steptype_out,

@@ -285,3 +285,31 @@ def test_ppo_graph_capture(ext):
         learner.update_step()
     torch.cuda.synchronize()
     assert not torch.equal(before, p.detach())
+
+
+@requires_gpu
+def test_humanoid_kernel_matches_torch_env(ext):
+    """HIP humanoid_step vs the torch reference physics for identical
+    states/actions (single non-terminal steps)."""
+    from stoix_amd.envs.humanoid import Humanoid
+
+    torch.manual_seed(0)
+    B = 64
+    env_g = Humanoid(num_envs=B, device="cuda", seed=5)
+    ts = env_g.reset()
+    env_c = Humanoid(num_envs=B, device="cpu", seed=5)
+    env_c.reset()
+    g = torch.Generator().manual_seed(1)
+    for i in range(5):
+        # keep CPU state mirror synchronised with the GPU state
+        env_c._state = {"s": env_g._state["s"].detach().cpu().clone()}
+        a = (torch.rand(B, 17, generator=g) * 2 - 1)
+        state_c, reward_c, term_c = env_c._step_fn({"s": env_c._state["s"]}, a)
+        ts = env_g.step(a.cuda())
+        torch.cuda.synchronize()
+        # compare PRE-autoreset next_obs against the CPU physics
+        obs_c = env_c._obs_fn(state_c)
+        torch.testing.assert_close(
+            ts.extras["next_obs"].cpu(), obs_c, rtol=2e-4, atol=2e-4
+        )
+        torch.testing.assert_close(ts.reward.cpu(), reward_c, rtol=2e-4, atol=2e-3)

@@ -110,3 +110,24 @@ def test_sebulba_impala_shared_torso_end_to_end():
     )
     r = run(cfg)
     assert r == r
+
+
+def test_sebulba_ppo_breakout_pixels():
+    """Config #4 shape: Sebulba PPO on the Breakout-class pixel env with the
+    CNN network (CPU envs + CPU 'learner devices' here; the GPU split is
+    exercised on the GPU box)."""
+    from stoix_amd.systems.ppo.sebulba_ff_ppo import run
+
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        [
+            "env=envpool/breakout", "network=cnn",
+            "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
+            "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+            "logger.loggers=[]", "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r == r
