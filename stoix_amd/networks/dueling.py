@@ -29,6 +29,8 @@ class DuelingQNetwork(nn.Module):
         self.epsilon = epsilon
 
     def q_values(self, x: Tensor) -> Tensor:
+        if x.dim() > 2:
+            x = x.reshape(x.shape[0], -1)
         v = self.value(x)
         a = self.advantage(x)
         return v + a - a.mean(dim=-1, keepdim=True)
@@ -64,6 +66,8 @@ class DistributionalDuelingQNetwork(nn.Module):
         self.register_buffer("atoms", torch.linspace(vmin, vmax, num_atoms))
 
     def forward(self, x: Tensor) -> C51Output:
+        if x.dim() > 2:
+            x = x.reshape(x.shape[0], -1)  # flatten grid/pixel observations
         v = self.value(x).view(*x.shape[:-1], 1, self.num_atoms)
         a = self.advantage(x).view(*x.shape[:-1], self.num_actions, self.num_atoms)
         logits = v + a - a.mean(dim=-2, keepdim=True)

@@ -39,9 +39,12 @@ class MLPTorso(nn.Module):
                 mods.append(nn.LayerNorm(h))
             mods.append(act())
             in_d = h
+        self._in_dim = input_dim
         self.net = nn.Sequential(*mods)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() > 2 and x.shape[-1] != self._in_dim:
+            x = x.reshape(x.shape[0], -1)  # flatten grid/pixel observations
         return self.net(x)
 
 
@@ -67,9 +70,12 @@ class NoisyMLPTorso(nn.Module):
                 mods.append(nn.LayerNorm(h))
             mods.append(act())
             in_d = h
+        self._in_dim = input_dim
         self.net = nn.Sequential(*mods)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        if x.dim() > 2 and x.shape[-1] != self._in_dim:
+            x = x.reshape(x.shape[0], -1)
         return self.net(x)
 
 

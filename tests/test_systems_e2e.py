@@ -78,3 +78,42 @@ def test_ff_ppo_learns_identity_game():
     r = run(cfg)
     # optimal = 10 (episode length), random = 10/4 = 2.5
     assert r > 7.0, f"PPO failed to learn identity game: return={r}"
+
+
+def test_rainbow_on_snake_grid_obs():
+    """BASELINE config #5 shape: Rainbow on the Snake grid env (multi-dim
+    observation flattened by the MLP/dueling torsos)."""
+    from stoix_amd.config import compose
+    from stoix_amd.systems.q_learning.ff_rainbow import run
+
+    cfg = compose(
+        "default/anakin/default_ff_rainbow.yaml",
+        [
+            "env=jumanji/snake", "arch.total_num_envs=8", "arch.total_timesteps=null",
+            "arch.num_updates=2", "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "system.rollout_length=4", "system.batch_size=8", "system.buffer_size=1024",
+            "system.warmup_steps=8", "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r == r
+
+
+def test_sac_on_humanoid():
+    """BASELINE config #3 shape: SAC on the Humanoid-class biped."""
+    from stoix_amd.config import compose
+    from stoix_amd.systems.sac.ff_sac import run
+
+    cfg = compose(
+        "default/anakin/default_ff_sac.yaml",
+        [
+            "env=brax/humanoid", "arch.total_num_envs=8", "arch.total_timesteps=null",
+            "arch.num_updates=2", "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "system.rollout_length=4", "system.batch_size=8", "system.buffer_size=1024",
+            "system.warmup_steps=8", "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r == r
