@@ -1,0 +1,41 @@
+"""Aux-subsystem tests: SLURM launcher script generation + sweep runner."""
+import torch
+
+from stoix_amd.config import compose
+
+
+def test_slurm_launcher_writes_script(tmp_path):
+    from stoix_amd.slurm_launcher import main
+
+    rc = main([
+        "--entry", "stoix_amd.systems.ppo.ff_ppo", "--nodes", "1",
+        "--gpus-per-node", "8", "--logdir", str(tmp_path), "--dry-run",
+        "--", "env=brax/ant", "arch.seed=0,1",
+    ])
+    assert rc == 0
+    script = (tmp_path / "stoix_amd.sbatch").read_text()
+    assert "--array=0-1" in script
+    assert "torch.distributed.run" in script
+    assert "--nproc-per-node=8" in script or "--nproc-per-node 8" in script
+    assert "arch.seed=0" in script and "arch.seed=1" in script
+
+
+def test_sweep_random_finds_best():
+    from stoix_amd.utils.sweep import Choice, LogUniform, Sweep
+
+    sweep = Sweep(
+        entry="stoix_amd.systems.ppo.ff_ppo",
+        default="default/anakin/default_ff_ppo.yaml",
+        space={"system.actor_lr": LogUniform(1e-4, 1e-3),
+               "system.num_minibatches": Choice([1, 2])},
+        base_overrides=[
+            "env=debug/identity", "arch.total_num_envs=4", "arch.total_timesteps=null",
+            "arch.num_updates=2", "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "system.rollout_length=4", "system.epochs=1", "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    best = sweep.run_random(num_trials=2, seed=0)
+    assert best is not None and best.value == best.value
+    assert len(sweep.trials) == 2
+    assert sweep.summary()
