@@ -178,3 +178,20 @@ class SharedPolicyValueNetwork(nn.Module):
 
     def forward(self, obs: Tensor):
         return self.head(self.torso(obs))
+
+
+def chained_torsos(*torsos: nn.Module) -> nn.Module:
+    """Compose torsos sequentially (reference base.py:225-252)."""
+
+    class _Chained(nn.Module):
+        def __init__(self):
+            super().__init__()
+            self.torsos = nn.ModuleList(torsos)
+            self.output_dim = getattr(torsos[-1], "output_dim", None)
+
+        def forward(self, x: Tensor) -> Tensor:
+            for t in self.torsos:
+                x = t(x)
+            return x
+
+    return _Chained()

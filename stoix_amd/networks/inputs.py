@@ -33,3 +33,17 @@ class EmbeddingActionOnehotInput(nn.Module):
     def forward(self, obs: Tensor, action: Tensor) -> Tensor:
         onehot = torch.nn.functional.one_hot(action.long(), self.num_actions).to(obs.dtype)
         return torch.cat([obs, onehot], dim=-1)
+
+
+class FeatureInput(nn.Module):
+    """Select one named feature from a dict observation (reference
+    inputs.py FeatureInput: structured-observation envs expose a dict; the
+    network consumes a single flat field)."""
+
+    def __init__(self, feature: str = "obs"):
+        super().__init__()
+        self.feature = feature
+
+    def forward(self, obs) -> Tensor:
+        x = obs[self.feature] if isinstance(obs, dict) else obs
+        return x.flatten(1) if x.dim() > 2 else x

@@ -36,3 +36,25 @@ class ExplorationNoisePostProcessor:
     def __call__(self, action: Tensor, generator: Optional[torch.Generator] = None) -> Tensor:
         noise = torch.randn(action.shape, device=action.device, generator=generator) * self.sigma
         return (action + noise).clamp(self.minimum, self.maximum)
+
+
+class PostProcessedDistribution:
+    """Wrap a distribution so sampled/greedy actions pass through a
+    postprocessor fn (reference postprocessors.py:13-30); log-prob/entropy
+    delegate to the base distribution."""
+
+    def __init__(self, base, postprocessor):
+        self.base = base
+        self.post = postprocessor
+
+    def sample(self, generator: Optional[torch.Generator] = None) -> Tensor:
+        return self.post(self.base.sample(generator))
+
+    def mode(self) -> Tensor:
+        return self.post(self.base.mode())
+
+    def log_prob(self, value: Tensor) -> Tensor:
+        return self.base.log_prob(value)
+
+    def entropy(self, *a, **kw) -> Tensor:
+        return self.base.entropy(*a, **kw)
