@@ -53,6 +53,16 @@ void launch_value_forward(const float*, const void*, const float*,
                           const void*, const float*, const void*,
                           const float*, float*, const float*, const float*,
                           int, int, int, void*);
+void launch_rollout_step_ant(float*, float*, int*, float*, int*, float*,
+                             int*, const void*, const float*, const void*,
+                             const float*, const void*, const float*,
+                             const void*, const float*, const void*,
+                             const float*, const void*, const float*, float*,
+                             float*, float*, float*, float*, float*, float*,
+                             unsigned char*, int, int, int, int, int, float,
+                             float, float, float, uint64_t, uint64_t,
+                             unsigned int*, unsigned int*, unsigned int,
+                             void*);
 void launch_silu_fwd(const void*, void*, long, void*);
 void launch_silu_bwd(const void*, const void*, void*, long, void*);
 void launch_ppo_gather(const long*, int, const float*, int, int,
@@ -307,6 +317,48 @@ void value_forward(torch::Tensor obs, torch::Tensor W1c, torch::Tensor b1c,
                        cur_stream());
 }
 
+void rollout_step_ant(torch::Tensor obs_io, torch::Tensor env_state,
+                      torch::Tensor step_count, torch::Tensor ep_return,
+                      torch::Tensor ep_length, torch::Tensor last_ep_return,
+                      torch::Tensor last_ep_length, torch::Tensor W1a,
+                      torch::Tensor b1a, torch::Tensor W2a, torch::Tensor b2a,
+                      torch::Tensor Wha, torch::Tensor bha, torch::Tensor W1c,
+                      torch::Tensor b1c, torch::Tensor W2c, torch::Tensor b2c,
+                      torch::Tensor Wvc, torch::Tensor bvc,
+                      torch::Tensor buf_obs, torch::Tensor buf_action,
+                      torch::Tensor buf_logp, torch::Tensor buf_value,
+                      torch::Tensor buf_bootstrap, torch::Tensor buf_reward,
+                      torch::Tensor buf_discount, torch::Tensor buf_steptype,
+                      int64_t max_episode_steps, double min_scale,
+                      double aff_scale, double aff_shift,
+                      double log_aff_scale, int64_t policy_seed,
+                      int64_t env_seed, torch::Tensor policy_draw,
+                      torch::Tensor env_draw, int64_t draw_offset) {
+  CHK(obs_io, torch::kFloat32);
+  CHK(env_state, torch::kFloat32);
+  CHK(W1a, torch::kBFloat16);
+  int B = obs_io.size(0), OBS = obs_io.size(1);
+  int HID = W2a.size(0), ACT = buf_action.size(1);
+  launch_rollout_step_ant(
+      obs_io.data_ptr<float>(), env_state.data_ptr<float>(),
+      step_count.data_ptr<int>(), ep_return.data_ptr<float>(),
+      ep_length.data_ptr<int>(), last_ep_return.data_ptr<float>(),
+      last_ep_length.data_ptr<int>(), W1a.data_ptr(), b1a.data_ptr<float>(),
+      W2a.data_ptr(), b2a.data_ptr<float>(), Wha.data_ptr(),
+      bha.data_ptr<float>(), W1c.data_ptr(), b1c.data_ptr<float>(),
+      W2c.data_ptr(), b2c.data_ptr<float>(), Wvc.data_ptr(),
+      bvc.data_ptr<float>(), buf_obs.data_ptr<float>(),
+      buf_action.data_ptr<float>(), buf_logp.data_ptr<float>(),
+      buf_value.data_ptr<float>(), buf_bootstrap.data_ptr<float>(),
+      buf_reward.data_ptr<float>(), buf_discount.data_ptr<float>(),
+      buf_steptype.data_ptr<unsigned char>(), B, OBS, ACT, HID,
+      (int)max_episode_steps, (float)min_scale, (float)aff_scale,
+      (float)aff_shift, (float)log_aff_scale, (uint64_t)policy_seed,
+      (uint64_t)env_seed, (unsigned int*)policy_draw.data_ptr<int>(),
+      (unsigned int*)env_draw.data_ptr<int>(), (unsigned int)draw_offset,
+      cur_stream());
+}
+
 void silu_fwd(torch::Tensor z, torch::Tensor h) {
   CHK(z, torch::kBFloat16);
   TORCH_CHECK(z.numel() % 8 == 0, "silu_fwd needs numel % 8 == 0");
@@ -426,6 +478,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("policy_value_step", &policy_value_step,
         "fused actor+critic fwd + tanh-normal sample (MFMA)");
   m.def("value_forward", &value_forward, "fused critic fwd (MFMA)");
+  m.def("rollout_step_ant", &rollout_step_ant,
+        "fused rollout step: policy + Ant physics + bootstrap, one launch");
   m.def("silu_fwd", &silu_fwd, "bf16 silu forward");
   m.def("silu_bwd", &silu_bwd, "bf16 silu backward");
   m.def("ppo_gather", &ppo_gather, "fused minibatch gather");

@@ -22,6 +22,7 @@
 // bf16 mirrors; they are hot across the 128 sequential rollout steps).
 // Per-wave row blocks are private, so the kernels need NO __syncthreads().
 #include "common.h"
+#include "ant_core.h"
 #include <hip/hip_bf16.h>
 
 typedef __bf16 bf16_t;
@@ -102,6 +103,7 @@ template <int HID>
 struct MlpLds {
   bf16_t O[16][K1P_MAX + OPAD];
   bf16_t H[4][16][HID + HPAD];
+  float act[16][8];  // sampled actions for the in-kernel env step
 };
 
 // Load one B fragment from a row-major [N,K] bf16 weight matrix.
@@ -354,6 +356,228 @@ __launch_bounds__(256, 2) __global__ void value_forward_kernel(
   if (wid == 0) {
     wave_value_head<HID>(H1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out,
                          lane, B);
+  }
+}
+
+// --------------------------------------- fused rollout step (Ant)
+//
+// ONE launch per rollout step: actor fwd + tanh-normal sample + critic fwd
+// (MFMA, all 4 waves) -> Ant physics + autoreset + episode metrics (16 env
+// threads of the workgroup; everything is per-env-row independent, so no
+// grid-wide sync is needed) -> critic fwd on the pre-reset next_obs
+// (bootstrap). Replaces the 3-kernel (policy/env/value) chain: fewer
+// launches, 4x more CUs on the physics (256 WGs vs 64), and next_obs never
+// round-trips through HBM (it stays in LDS for the bootstrap pass).
+template <int HID>
+__launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
+    float* __restrict__ obs_io,          // [B, 27] env obs buffer (in/out)
+    float* __restrict__ env_state,       // [B, 29]
+    int* __restrict__ step_count, float* __restrict__ ep_return,
+    int* __restrict__ ep_length, float* __restrict__ last_ep_return,
+    int* __restrict__ last_ep_length,
+    const bf16_t* __restrict__ W1a, const float* __restrict__ b1a,
+    const bf16_t* __restrict__ W2a, const float* __restrict__ b2a,
+    const bf16_t* __restrict__ Wha, const float* __restrict__ bha,
+    const bf16_t* __restrict__ W1c, const float* __restrict__ b1c,
+    const bf16_t* __restrict__ W2c, const float* __restrict__ b2c,
+    const bf16_t* __restrict__ Wvc, const float* __restrict__ bvc,
+    float* __restrict__ buf_obs,         // [B, 27] rollout storage row t
+    float* __restrict__ buf_action,      // [B, 8]
+    float* __restrict__ buf_logp,        // [B]
+    float* __restrict__ buf_value,       // [B]
+    float* __restrict__ buf_bootstrap,   // [B]
+    float* __restrict__ buf_reward, float* __restrict__ buf_discount,
+    unsigned char* __restrict__ buf_steptype,
+    int B, int OBS, int ACT, int max_episode_steps, float min_scale,
+    float aff_scale, float aff_shift, float log_aff_scale,
+    uint64_t policy_seed, uint64_t env_seed,
+    const unsigned int* __restrict__ policy_draw,
+    const unsigned int* __restrict__ env_draw, unsigned int draw_offset) {
+  __shared__ MlpLds<HID> lds;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int rbase = blockIdx.x * 16;
+  if (rbase >= B) return;
+  const uint32_t pdraw = *policy_draw + draw_offset;
+  const uint32_t edraw = *env_draw + draw_offset;
+  const int K1P = (OBS + 31) & ~31;
+
+  bf16_t* O = &lds.O[0][0];
+  constexpr int OS = K1P_MAX + OPAD;
+  constexpr int HS = HID + HPAD;
+  bf16_t* Ha0 = &lds.H[0][0][0];
+  bf16_t* Ha1 = &lds.H[1][0][0];
+  bf16_t* Hc0 = &lds.H[2][0][0];
+  bf16_t* Hc1 = &lds.H[3][0][0];
+
+  stage_obs(obs_io, rbase, OBS, K1P, O, OS, nullptr, nullptr, buf_obs,
+            threadIdx.x, 256, B);
+  __syncthreads();
+
+  // ---- policy + value (same structure as policy_value_step_kernel)
+  wg_layer<HID, true>(O, OS, W1a, b1a, K1P, Ha0, HS, lane, wid);
+  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
+  __syncthreads();
+  wg_layer<HID, true>(Ha0, HS, W2a, b2a, HID, Ha1, HS, lane, wid);
+  wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
+  __syncthreads();
+
+  if (wid == 1) {
+    wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, buf_value,
+                         lane, B);
+  }
+  if (wid == 0) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15, ak0 = (lane >> 4) * 8;
+    for (int ks = 0; ks < HID / 32; ++ks) {
+      const bf16x8 aa =
+          *reinterpret_cast<const bf16x8*>(Ha1 + arow * HS + ks * 32 + ak0);
+      bf16x8 bb = load_w_frag<HID>(Wha, HID, 0, ks, lane);
+      acc = MFMA_BF16_16x16x32(aa, bb, acc, 0, 0, 0);
+    }
+    const int col = lane & 15;
+    const int g = lane >> 4;
+    float bh = bha[col];
+    float out[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) out[r] = acc[r] + bh;
+    const int src_lane = (lane & 48) | (((lane & 15) + 8) & 15);
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float loc = out[r];
+      float spre = __shfl(out[r], src_lane, 64);
+      float logp = 0.0f;
+      float a_val = 0.0f;
+      if (col < 8) {
+        float sigma = softplus_f(spre) + min_scale;
+        int rowblk = (rbase + g * 4) >> 2;
+        Rng4 u = philox_uniform4(policy_seed, 2u, (uint32_t)(rowblk * 16 + col),
+                                 pdraw);
+        float n[4];
+        box_muller(u.a, u.b, &n[0], &n[1]);
+        box_muller(u.c, u.d, &n[2], &n[3]);
+        float eps = n[r];
+        float uu = loc + sigma * eps;
+        a_val = tanhf(uu) * aff_scale + aff_shift;
+        float log_det =
+            2.0f * (0.6931471805599453f - uu - softplus_f(-2.0f * uu)) +
+            log_aff_scale;
+        logp = -0.5f * eps * eps - __logf(sigma) - 0.9189385332046727f -
+               log_det;
+      }
+      if (col >= ACT && col < 8) logp = 0.0f;
+      logp += __shfl_xor(logp, 1);
+      logp += __shfl_xor(logp, 2);
+      logp += __shfl_xor(logp, 4);
+      int grow = rbase + g * 4 + r;
+      if (grow < B) {
+        if (col < ACT) {
+          buf_action[(long)grow * ACT + col] = a_val;
+          lds.act[g * 4 + r][col] = a_val;
+        }
+        if (col == 0) buf_logp[grow] = logp;
+      }
+    }
+  }
+  __syncthreads();
+
+  // ---- env step: 16 threads, one per row (Ant physics in registers)
+  if (threadIdx.x < 16) {
+    int b = rbase + threadIdx.x;
+    if (b < B) {
+      float es[ANT_STATE];
+#pragma unroll
+      for (int i = 0; i < ANT_STATE; ++i) es[i] = env_state[b * ANT_STATE + i];
+      float ea[ANT_ACT];
+#pragma unroll
+      for (int i = 0; i < ANT_ACT; ++i)
+        ea[i] = fminf(fmaxf(lds.act[threadIdx.x][i], -1.0f), 1.0f);
+      float reward;
+      bool terminated;
+      ant_physics_step(es, ea, &reward, &terminated);
+
+      int sc = step_count[b] + 1;
+      bool truncated = (sc >= max_episode_steps) && !terminated;
+      bool done = terminated || truncated;
+      float ret = ep_return[b] + reward;
+      int len = ep_length[b] + 1;
+      if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
+
+      // pre-reset next_obs -> LDS obs tile (bootstrap input; never HBM)
+      float nobs[ANT_OBS];
+      ant_write_obs(es, nobs);
+#pragma unroll
+      for (int k = 0; k < ANT_OBS; ++k)
+        lds.O[threadIdx.x][k] = f2bf(nobs[k]);
+
+      if (done) {
+        ant_reset_state(es, env_seed, (uint32_t)b, edraw);
+        sc = 0; ret = 0.0f; len = 0;
+      }
+#pragma unroll
+      for (int i = 0; i < ANT_STATE; ++i) env_state[b * ANT_STATE + i] = es[i];
+      ant_write_obs(es, obs_io + b * ANT_OBS);  // post-reset obs for t+1
+      step_count[b] = sc;
+      ep_return[b] = ret;
+      ep_length[b] = len;
+      buf_reward[b] = reward;
+      buf_discount[b] = terminated ? 0.0f : 1.0f;
+      buf_steptype[b] =
+          terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+    }
+  }
+  __syncthreads();
+
+  // ---- bootstrap: critic fwd on the pre-reset next_obs (LDS-resident)
+  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
+  __syncthreads();
+  wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
+  __syncthreads();
+  if (wid == 0) {
+    wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase,
+                         buf_bootstrap, lane, B);
+  }
+}
+
+extern "C" void launch_rollout_step_ant(
+    float* obs_io, float* env_state, int* step_count, float* ep_return,
+    int* ep_length, float* last_ep_return, int* last_ep_length,
+    const void* W1a, const float* b1a, const void* W2a, const float* b2a,
+    const void* Wha, const float* bha, const void* W1c, const float* b1c,
+    const void* W2c, const float* b2c, const void* Wvc, const float* bvc,
+    float* buf_obs, float* buf_action, float* buf_logp, float* buf_value,
+    float* buf_bootstrap, float* buf_reward, float* buf_discount,
+    unsigned char* buf_steptype, int B, int OBS, int ACT, int HID,
+    int max_episode_steps, float min_scale, float aff_scale, float aff_shift,
+    float log_aff_scale, uint64_t policy_seed, uint64_t env_seed,
+    unsigned int* policy_draw, unsigned int* env_draw,
+    unsigned int draw_offset, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((B + 15) / 16), block(256);
+  if (HID == 256) {
+    hipLaunchKernelGGL(rollout_step_ant_kernel<256>, grid, block, 0, s,
+                       obs_io, env_state, step_count, ep_return, ep_length,
+                       last_ep_return, last_ep_length, (const bf16_t*)W1a,
+                       b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_steptype, B, OBS, ACT, max_episode_steps,
+                       min_scale, aff_scale, aff_shift, log_aff_scale,
+                       policy_seed, env_seed, policy_draw, env_draw,
+                       draw_offset);
+  } else {
+    hipLaunchKernelGGL(rollout_step_ant_kernel<128>, grid, block, 0, s,
+                       obs_io, env_state, step_count, ep_return, ep_length,
+                       last_ep_return, last_ep_length, (const bf16_t*)W1a,
+                       b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_steptype, B, OBS, ACT, max_episode_steps,
+                       min_scale, aff_scale, aff_shift, log_aff_scale,
+                       policy_seed, env_seed, policy_draw, env_draw,
+                       draw_offset);
   }
 }
 

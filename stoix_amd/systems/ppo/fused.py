@@ -321,8 +321,29 @@ class FusedPPOEngine:
         ac, cc = self.actor_chain, self.critic_chain
         a16, c16 = ac.views16, cc.views16
         ext = self.ext
-        # ensure hb["obs"] holds the current observation (set at engine
-        # attach + maintained by every env step)
+        if getattr(env, "HIP_KERNEL", "") == "ant_step":
+            # megakernel path: ONE launch per rollout step (policy + Ant
+            # physics + bootstrap critic; mlp.hip rollout_step_ant_kernel)
+            for t in range(L.T):
+                ext.rollout_step_ant(
+                    hb["obs"], env._state["s"], env._step_count,
+                    env._ep_return, env._ep_length, env._last_ep_return,
+                    env._last_ep_length,
+                    a16["W1"], ac.views["b1"], a16["W2"], ac.views["b2"],
+                    a16["Wh"], ac.views["bh"],
+                    c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
+                    c16["Wv"], cc.views["bv"],
+                    L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t],
+                    L.buf_value[t], L.buf_bootstrap[t], L.buf_reward[t],
+                    L.buf_discount[t], self.buf_steptype[t],
+                    env.max_episode_steps, self.min_scale, self.aff_scale,
+                    self.aff_shift, self.log_aff_scale, self.seed,
+                    env._hip_seed, self.draw_policy, hb["draw"], t,
+                )
+            ext.bump_add(self.draw_policy, L.T)
+            ext.bump_add(hb["draw"], L.T)
+            L.buf_truncated.copy_(self.buf_steptype == 3)
+            return
         for t in range(L.T):
             ext.policy_value_step(
                 hb["obs"],

@@ -429,3 +429,44 @@ def test_wgrad_kernel_matches_mm(ext, N, K, NV):
     torch.testing.assert_close(dW, ref_w, rtol=3e-2, atol=3e-2)
     torch.testing.assert_close(db, ref_b, rtol=3e-2, atol=3e-1)
     assert (slab == 0).all()  # reduce re-zeroes for the next minibatch
+
+
+@requires_gpu
+def test_fused_rollout_megakernel_invariants(ext):
+    """rollout_step_ant megakernel: stored log-probs/values must agree with
+    the eager modules evaluated on the stored observations/actions (the
+    module params are fp32 views of the fused masters)."""
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo_continuous.yaml",
+        [
+            "env=brax/ant", "arch.total_num_envs=256", "arch.total_timesteps=null",
+            "arch.num_updates=4", "arch.num_evaluation=1",
+            "system.rollout_length=8", "system.num_minibatches=2",
+            "system.epochs=1", "system.compute_dtype=bf16", "logger.loggers=[]",
+        ],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    device = torch.device("cuda:0")
+    env = environments.make_single(cfg, 256, device, seed=0)
+    learner = PPOLearner(cfg, env, device)
+    assert learner.fused is not None
+    learner.rollout_phase()
+    torch.cuda.synchronize()
+    for t in [0, 3, 7]:
+        obs = learner.buf_obs[t]
+        act = learner.buf_action[t]
+        with torch.no_grad():
+            dist = learner.actor(obs)
+            logp = dist.log_prob(act)
+            v = learner.critic(obs)
+        torch.testing.assert_close(learner.buf_log_prob[t], logp, rtol=5e-2, atol=0.15)
+        torch.testing.assert_close(learner.buf_value[t], v, rtol=5e-2, atol=5e-2)
+    # rewards/discounts sane; steptypes valid
+    assert torch.isfinite(learner.buf_reward).all()
+    assert ((learner.buf_discount == 0) | (learner.buf_discount == 1)).all()
