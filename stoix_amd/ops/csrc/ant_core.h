@@ -45,6 +45,9 @@ DEV_INLINE void ant_physics_step(float* __restrict__ s,
   const float SQ2 = 0.70710678118654752f;
   const float hdx[4] = {SQ2, -SQ2, -SQ2, SQ2};
   const float hdy[4] = {SQ2, SQ2, -SQ2, -SQ2};
+  // atan2f(hdy, hdx) is constant per leg: pi/4 + leg*pi/2
+  const float base_angs[4] = {0.78539816339744831f, 2.35619449019234493f,
+                              -2.35619449019234493f, -0.78539816339744831f};
 
   float x_before = s[0];
   float contact_mag = 0.0f;
@@ -62,11 +65,10 @@ DEV_INLINE void ant_physics_step(float* __restrict__ s,
     V3 r_arm[4], foot_w[4];
 #pragma unroll
     for (int leg = 0; leg < 4; ++leg) {
-      float base_ang = atan2f(hdy[leg], hdx[leg]);
-      float leg_ang = base_ang + qpos[leg];
-      float ca = cosf(leg_ang), sa = sinf(leg_ang);
-      float knee = qpos[4 + leg];
-      float ck = cosf(knee), sk = sinf(knee);
+      float leg_ang = base_angs[leg] + qpos[leg];
+      float ca, sa, ck, sk;
+      __sincosf(leg_ang, &sa, &ca);
+      __sincosf(qpos[4 + leg], &sk, &ck);
       V3 body_off = {hdx[leg] * HIP_RADIUS + ca * L1 + ca * L2 * ck,
                      hdy[leg] * HIP_RADIUS + sa * L1 + sa * L2 * ck,
                      -L2 * sk};

@@ -385,7 +385,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
     float* __restrict__ buf_action,      // [B, 8]
     float* __restrict__ buf_logp,        // [B]
     float* __restrict__ buf_value,       // [B]
-    float* __restrict__ buf_bootstrap,   // [B]
+    float* __restrict__ next_obs_buf,    // [B, 27] pre-reset next obs row t
     float* __restrict__ buf_reward, float* __restrict__ buf_discount,
     unsigned char* __restrict__ buf_steptype,
     int B, int OBS, int ACT, int max_episode_steps, float min_scale,
@@ -503,12 +503,10 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       int len = ep_length[b] + 1;
       if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
 
-      // pre-reset next_obs -> LDS obs tile (bootstrap input; never HBM)
-      float nobs[ANT_OBS];
-      ant_write_obs(es, nobs);
-#pragma unroll
-      for (int k = 0; k < ANT_OBS; ++k)
-        lds.O[threadIdx.x][k] = f2bf(nobs[k]);
+      // pre-reset next_obs -> global row t (the bootstrap critic runs
+      // ONCE over all T*B rows after the rollout: one big MFMA-efficient
+      // launch instead of 128 small 2-layer passes)
+      ant_write_obs(es, next_obs_buf + (long)b * ANT_OBS);
 
       if (done) {
         ant_reset_state(es, env_seed, (uint32_t)b, edraw);
@@ -526,17 +524,6 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
           terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
     }
   }
-  __syncthreads();
-
-  // ---- bootstrap: critic fwd on the pre-reset next_obs (LDS-resident)
-  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
-  __syncthreads();
-  wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
-  __syncthreads();
-  if (wid == 0) {
-    wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase,
-                         buf_bootstrap, lane, B);
-  }
 }
 
 extern "C" void launch_rollout_step_ant(
@@ -546,7 +533,7 @@ extern "C" void launch_rollout_step_ant(
     const void* Wha, const float* bha, const void* W1c, const float* b1c,
     const void* W2c, const float* b2c, const void* Wvc, const float* bvc,
     float* buf_obs, float* buf_action, float* buf_logp, float* buf_value,
-    float* buf_bootstrap, float* buf_reward, float* buf_discount,
+    float* next_obs_buf, float* buf_reward, float* buf_discount,
     unsigned char* buf_steptype, int B, int OBS, int ACT, int HID,
     int max_episode_steps, float min_scale, float aff_scale, float aff_shift,
     float log_aff_scale, uint64_t policy_seed, uint64_t env_seed,
@@ -561,7 +548,7 @@ extern "C" void launch_rollout_step_ant(
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
                        (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
-                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_value, next_obs_buf, buf_reward, buf_discount,
                        buf_steptype, B, OBS, ACT, max_episode_steps,
                        min_scale, aff_scale, aff_shift, log_aff_scale,
                        policy_seed, env_seed, policy_draw, env_draw,
@@ -573,7 +560,7 @@ extern "C" void launch_rollout_step_ant(
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
                        (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
-                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_value, next_obs_buf, buf_reward, buf_discount,
                        buf_steptype, B, OBS, ACT, max_episode_steps,
                        min_scale, aff_scale, aff_shift, log_aff_scale,
                        policy_seed, env_seed, policy_draw, env_draw,
