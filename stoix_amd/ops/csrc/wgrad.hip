@@ -25,41 +25,14 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
 #define WG_SLICES 16  // 4 z-blocks x 4 waves
+#define SPAD 8        // +8 cols on the 32-wide transposed tiles
 
-// Per-wave staging, ROW-MAJOR [32 s][16 x] subtiles: global loads and the
-// LDS writes are both 16B-contiguous (no transposed scatter), and the MFMA
-// fragments are read with the gfx950 hardware transpose-read
-// ds_read_tr16_b64 (semantics decoded empirically, tools/tr16_analyze.py:
-// within each quad of lanes, lane 4q+i element j = 16-bit element i at
-// lane 4q+j's address -> a 4x4 transpose per quad). For an operand tile
-// [32 s][16 x], per-lane address row (l>>4)*8+(l&3)+s_half*4, column
-// 4*((l>>2)&3) yields exactly the MFMA fragment x = l&15, s = (l>>4)*8+j.
+// One wave's private staging: dZt [16 n][32+8 s], Xt [KPG*16 k][32+8 s].
+// KPG <= 8 -> per-wave 40*(16+128)*2B = 11.5 KB; 4 waves = 46 KB.
 struct WgradLds {
-  bf16_t dZt[4][32][16];
-  bf16_t Xt[4][8][32][16];
+  bf16_t dZt[4][16][32 + SPAD];
+  bf16_t Xt[4][8 * 16][32 + SPAD];
 };
-
-typedef __attribute__((address_space(3))) bf16x4 lds_bf16x4;
-
-DEV_INLINE bf16x4 tr_read(const bf16_t* base) {
-  return __builtin_amdgcn_ds_read_tr16_b64_v4bf16((lds_bf16x4*)base);
-}
-
-// Read one [32 s] x [16 x] row-major LDS tile as an MFMA A/B fragment
-// (lane l -> x = l&15, s = (l>>4)*8 + j, j = 0..7) via two tr reads.
-DEV_INLINE bf16x8 frag_from_tile(const bf16_t (*tile)[16], int lane) {
-  int row = (lane >> 4) * 8 + (lane & 3);
-  int col = 4 * ((lane >> 2) & 3);
-  bf16x4 lo = tr_read(&tile[row][col]);
-  bf16x4 hi = tr_read(&tile[row + 4][col]);
-  bf16x8 out;
-#pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    out[j] = lo[j];
-    out[4 + j] = hi[j];
-  }
-  return out;
-}
 
 template <int KPG>
 __launch_bounds__(256, 2) __global__ void wgrad_kernel(
@@ -72,70 +45,90 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
   __shared__ WgradLds lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int nt = blockIdx.x;
+  const int nt = blockIdx.x;         // one 16-row n-tile
   const int k0 = blockIdx.y * KPG * 16;
   const int slice = blockIdx.z * 4 + wid;
-  const int s_per = S / WG_SLICES;
+  const int s_per = S / WG_SLICES;   // S divisible by 16*32 in practice
   const int s_begin = slice * s_per;
   const int s_end = s_begin + s_per;
   const int n0 = nt * 16;
 
-  bf16_t(*dZt)[16] = lds.dZt[wid];
-  bf16_t(*Xt)[32][16] = lds.Xt[wid];
+  bf16_t(*dZt)[32 + SPAD] = lds.dZt[wid];
+  bf16_t(*Xt)[32 + SPAD] = lds.Xt[wid];
 
   f32x4 acc[KPG];
 #pragma unroll
   for (int t = 0; t < KPG; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
-  float db_acc = 0.0f;
+  float db_acc = 0.0f;  // lane's partial of db[n0 + (lane&15)]
+
+  const int arow = lane & 15;
+  const int ak0 = (lane >> 4) * 8;
 
   for (int s0 = s_begin; s0 < s_end; s0 += 32) {
-    // ---- stage dZ tile [32 s][16 n], row-major (1 x 16B load+write/lane)
+    // ---- stage dZ tile [32 s][16 n] -> dZt [16 n][32 s] (transposed)
     {
+      // lane l: s-row = l>>1, n-half = (l&1)*8 -> one 16B load, 8 scatter
+      // writes
       int srow = lane >> 1, nh = (lane & 1) * 8;
-      *reinterpret_cast<bf16x8*>(&dZt[srow][nh]) =
-          *reinterpret_cast<const bf16x8*>(dZ + (long)(s0 + srow) * N_STRIDE +
-                                           n0 + nh);
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          dZ + (long)(s0 + srow) * N_STRIDE + n0 + nh);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) dZt[nh + i][srow] = v[i];
     }
-    // ---- stage X subtiles [KPG][32 s][16 k], row-major
+    // ---- stage X tiles [32 s][KPG*16 k] -> Xt [KPG*16 k][32 s].
+    // Lane mapping: consecutive lanes take consecutive s-rows of ONE
+    // k-chunk, so the 8 transposed b16 scatter-writes per lane hit
+    // consecutive banks (srow-consecutive) instead of a 16-way conflict
+    // (kh-strided, 160 dwords = bank 0 for every lane). The global loads
+    // become 16B row-strided, but successive q iterations re-touch the
+    // same 32 rows' cache lines, so L1 serves 7/8 of them.
 #pragma unroll
     for (int q = 0; q < KPG; ++q) {
       int flat = q * 64 + lane;
-      int srow = flat / (KPG * 2);
-      int hc = flat % (KPG * 2);
-      *reinterpret_cast<bf16x8*>(&Xt[hc >> 1][srow][(hc & 1) * 8]) =
-          *reinterpret_cast<const bf16x8*>(X + (long)(s0 + srow) * K + k0 +
-                                           hc * 8);
+      int srow = flat & 31;
+      int kh = (flat >> 5) * 8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          X + (long)(s0 + srow) * K + k0 + kh);
+#pragma unroll
+      for (int i = 0; i < 8; ++i) Xt[kh + i][srow] = v[i];
     }
-    // within-wave LDS write->read ordering is compiler-tracked (lgkmcnt)
-    const bf16x8 a = frag_from_tile(dZt, lane);
+    // within-wave LDS write->read ordering is compiler-tracked (lgkmcnt);
+    // no cross-wave sharing, so no barrier.
+    // ---- fragments + MFMA
+    const bf16x8 a = *reinterpret_cast<const bf16x8*>(&dZt[arow][ak0]);
 #pragma unroll
     for (int t = 0; t < KPG; ++t) {
-      const bf16x8 b = frag_from_tile(Xt[t], lane);
+      const bf16x8 b =
+          *reinterpret_cast<const bf16x8*>(&Xt[t * 16 + arow][ak0]);
       acc[t] = MFMA_BF16_16x16x32(a, b, acc[t], 0, 0, 0);
     }
+    // ---- bias partial from the A fragment (only the kt-group-0 blocks)
     if (db_off >= 0 && blockIdx.y == 0) {
 #pragma unroll
       for (int j = 0; j < 8; ++j) db_acc += (float)a[j];
     }
   }
 
+  // ---- write this wave's slab slice
   float* out = slab + (long)slice * slab_stride;
-  const int col = lane & 15;
+  const int col = lane & 15;  // k within tile
   const int g = lane >> 4;
 #pragma unroll
   for (int t = 0; t < KPG; ++t) {
 #pragma unroll
     for (int r = 0; r < 4; ++r) {
-      int n = g * 4 + r;
+      int n = g * 4 + r;  // D row = n (A is dZ^T)
       if (n < N_VALID)
         out[dW_off + (long)(n0 + n) * K + k0 + t * 16 + col] = acc[t][r];
     }
   }
   if (db_off >= 0 && blockIdx.y == 0) {
+    // db[n0 + arow] partial: lanes {arow, arow+16, arow+32, arow+48} hold
+    // s-disjoint partials; fold with xor-shuffles over the high bits
     db_acc += __shfl_xor(db_acc, 16);
     db_acc += __shfl_xor(db_acc, 32);
-    if ((lane >> 4) == 0 && (lane & 15) < N_VALID)
-      out[db_off + n0 + (lane & 15)] = db_acc;
+    if ((lane >> 4) == 0 && arow < N_VALID)
+      out[db_off + n0 + arow] = db_acc;
   }
 }
 
