@@ -34,15 +34,28 @@ class HipStepMixin:
         self._hip = ops.ext(required=True)
         B = self.num_envs
         dev = self.device
-        self._hb: Dict[str, torch.Tensor] = {
-            "obs": torch.zeros(B, self.OBS_DIM, device=dev),
-            "next_obs": torch.zeros(B, self.OBS_DIM, device=dev),
-            "reward": torch.zeros(B, device=dev),
-            "discount": torch.zeros(B, device=dev),
-            "steptype": torch.zeros(B, dtype=torch.uint8, device=dev),
-            "done": torch.zeros(B, dtype=torch.uint8, device=dev),
-            "draw": torch.zeros(1, dtype=torch.int32, device=dev),
-        }
+
+        def mk() -> Dict[str, torch.Tensor]:
+            return {
+                "obs": torch.zeros(B, self.OBS_DIM, device=dev),
+                "next_obs": torch.zeros(B, self.OBS_DIM, device=dev),
+                "reward": torch.zeros(B, device=dev),
+                "discount": torch.zeros(B, device=dev),
+                "steptype": torch.zeros(B, dtype=torch.uint8, device=dev),
+                "done": torch.zeros(B, dtype=torch.uint8, device=dev),
+            }
+
+        # The kernels write into ONE stable buffer set, but step() returns
+        # CLONES: returning the buffers themselves silently aliased — a
+        # learner doing ``obs = ts.observation; env.step(a); buf[t] = obs``
+        # stored the NEXT step's observation (stream-ordered overwrite),
+        # putting actions and observations off by one. That broke PPO
+        # learning on GPU while every single-step numerics test passed
+        # (tools/envdebug*.py forensics). The clones cost ~6 small copies
+        # per eager step; the fused rollout path (FusedPPOEngine) bypasses
+        # step() entirely and keeps the zero-copy pipeline.
+        self._hb: Dict[str, torch.Tensor] = mk()
+        self._hb["draw"] = torch.zeros(1, dtype=torch.int32, device=dev)
         self._hip_seed = int(torch.randint(0, 2**31 - 1, (1,), generator=self.gen, device=dev).item())
 
     def _hip_action(self, action: torch.Tensor) -> torch.Tensor:
@@ -110,14 +123,14 @@ class HipStepMixin:
             1,
         )
         return TimeStep(
-            step_type=hb["steptype"],
-            reward=hb["reward"],
-            discount=hb["discount"],
-            observation=hb["obs"],
+            step_type=hb["steptype"].clone(),
+            reward=hb["reward"].clone(),
+            discount=hb["discount"].clone(),
+            observation=hb["obs"].clone(),
             extras={
-                "next_obs": hb["next_obs"],
+                "next_obs": hb["next_obs"].clone(),
                 "episode_metrics": {
-                    "episode_return": self._last_ep_return,
+                    "episode_return": self._last_ep_return.clone(),
                     "episode_length": self._last_ep_length.to(torch.float32),
                     "is_terminal_step": hb["done"].bool(),
                 },

@@ -15,7 +15,7 @@ sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 import torch
 
 
-def run_ppo(env_override, net, updates, num_envs, extra=()):
+def run_ppo(env_override, net, updates, num_envs, extra=(), graphs=True):
     from stoix_amd import envs as environments
     from stoix_amd.config import compose
     from stoix_amd.systems.ppo.ff_ppo import PPOLearner
@@ -35,7 +35,7 @@ def run_ppo(env_override, net, updates, num_envs, extra=()):
     torch.manual_seed(3)
     env = environments.make_single(cfg, num_envs, device, seed=3)
     learner = PPOLearner(cfg, env, device)
-    if device.type == "cuda":
+    if device.type == "cuda" and graphs:
         try_enable_graphs(learner)
     curve = []
     t0 = time.time()
@@ -46,11 +46,21 @@ def run_ppo(env_override, net, updates, num_envs, extra=()):
             m = learner.episode_metrics
             r = float(m.get("episode_return", torch.tensor(float("nan"))))
             curve.append(round(r, 2))
-    return {"env": env_override, "curve": curve, "wall_s": round(time.time() - t0, 1),
+    return {"env": env_override, "graphs": graphs, "curve": curve,
+            "wall_s": round(time.time() - t0, 1),
             "fused": learner.fused is not None}
 
 
 def main():
+    if os.environ.get("LEARNCHECK_AB"):
+        for g in (False, True):
+            out = run_ppo("env=classic/cartpole", "disc", 60, 256,
+                          ("system.compute_dtype=fp32",), graphs=g)
+            print(json.dumps(out))
+        out = run_ppo("env=brax/ant", "cont", 60, 2048,
+                      ("system.compute_dtype=bf16",), graphs=False)
+        print(json.dumps(out))
+        return
     out = run_ppo("env=classic/cartpole", "disc", 60, 256,
                   ("system.compute_dtype=fp32",))
     print(json.dumps(out))
