@@ -313,3 +313,25 @@ def test_humanoid_kernel_matches_torch_env(ext):
             ts.extras["next_obs"].cpu(), obs_c, rtol=2e-4, atol=2e-4
         )
         torch.testing.assert_close(ts.reward.cpu(), reward_c, rtol=2e-4, atol=2e-3)
+
+
+@requires_gpu
+def test_hip_env_step_outputs_are_stable_across_steps(ext):
+    """Regression (GPU-env aliasing): the TimeStep returned by step t must
+    stay unchanged after step t+1 runs — learners hold these tensors
+    across subsequent env steps (some for the whole rollout)."""
+    from stoix_amd.envs.classic import CartPole
+
+    env = CartPole(num_envs=32, device="cuda:0", seed=0)
+    env.reset()
+    a = torch.zeros(32, dtype=torch.long, device="cuda:0")
+    ts1 = env.step(a)
+    obs1 = ts1.observation.clone()
+    r1 = ts1.reward.clone()
+    n1 = ts1.extras["next_obs"].clone()
+    for _ in range(3):
+        env.step(a)
+    torch.cuda.synchronize()
+    torch.testing.assert_close(ts1.observation, obs1, rtol=0, atol=0)
+    torch.testing.assert_close(ts1.reward, r1, rtol=0, atol=0)
+    torch.testing.assert_close(ts1.extras["next_obs"], n1, rtol=0, atol=0)
