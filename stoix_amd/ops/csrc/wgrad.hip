@@ -24,7 +24,7 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
-#define WG_SLICES 16  // 4 z-blocks x 4 waves
+#define WG_SLICES 32  // 8 z-blocks x 4 waves
 #define SPAD 8        // +8 cols on the 32-wide transposed tiles
 
 // One wave's private staging: dZt [16 n][32+8 s], Xt [KPG*16 k][32+8 s].
@@ -47,7 +47,7 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
   const int wid = threadIdx.x >> 6;
   const int nt = blockIdx.x;         // one 16-row n-tile
   const int k0 = blockIdx.y * KPG * 16;
-  const int slice = blockIdx.z * 4 + wid;
+  const int slice = blockIdx.z * 4 + wid;  // [0, WG_SLICES)
   const int s_per = S / WG_SLICES;   // S divisible by 16*32 in practice
   const int s_begin = slice * s_per;
   const int s_end = s_begin + s_per;
@@ -199,7 +199,7 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
     }
   }
   int KTG = K / (16 * KPG);
-  dim3 grid(NT, KTG, 4), block(256);
+  dim3 grid(NT, KTG, WG_SLICES / 4), block(256);
   if (KPG == 8) {
     hipLaunchKernelGGL(wgrad_kernel<8>, grid, block, 0, s, (const bf16_t*)dZ,
                        (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,

@@ -82,9 +82,10 @@ class _Chain:
             self.offsets[name] = off
             off += n
         self.numel = total
-        # split-K wgrad slab: 16 fp32 partial-gradient images of the whole
-        # chain (one per wave-slice); slab_reduce sums them into grad16
-        self.slab = torch.zeros(16, total, dtype=torch.float32, device=device)
+        # split-K wgrad slab: 32 fp32 partial-gradient images of the whole
+        # chain (one per wave-slice; wgrad.hip WG_SLICES); slab_reduce sums
+        # them into grad16
+        self.slab = torch.zeros(32, total, dtype=torch.float32, device=device)
 
     def sync_mirror(self) -> None:
         self.flat16.copy_(self.flat)

@@ -415,7 +415,7 @@ def test_wgrad_kernel_matches_mm(ext, N, K, NV):
     dZ = (torch.randn(S, N, generator=g) * 0.1).bfloat16().cuda()
     X = torch.randn(S, K, generator=g).bfloat16().cuda()
     numel = NV * K + NV
-    slab = torch.zeros(16, numel, device="cuda")
+    slab = torch.zeros(32, numel, device="cuda")
     grad16 = torch.zeros(numel, dtype=torch.bfloat16, device="cuda")
     e0 = torch.zeros(0, device="cuda")
     e1 = torch.zeros(0, dtype=torch.int64, device="cuda")
