@@ -431,7 +431,7 @@ void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
   CHK(slab, torch::kFloat32);
   int S = dZ.size(0), N_STRIDE = dZ.size(1), K = X.size(1);
   TORCH_CHECK(X.size(0) == S, "wgrad: dZ/X row mismatch");
-  TORCH_CHECK(S % (32 * 32) == 0, "wgrad: S must be a multiple of 1024");
+  TORCH_CHECK(S % (4 * 32) == 0, "wgrad: S must be a multiple of 128");
   long stride = slab.size(1);
   launch_wgrad(dZ.data_ptr(), X.data_ptr(), slab.data_ptr<float>(),
                (long)dW_off, (long)db_off, stride, S, N_STRIDE, K,

@@ -24,7 +24,8 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
-#define WG_SLICES 32  // 8 z-blocks x 4 waves
+#define WG_SLICES 64  // max wave-slices; the launcher picks the
+// largest power of two <= 64 that divides S/32 (small minibatches use fewer)
 #define SPAD 8        // +8 cols on the 32-wide transposed tiles
 
 // One wave's private staging: dZt [16 n][32+8 s], Xt [KPG*16 k][32+8 s].
@@ -41,14 +42,15 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
     float* __restrict__ slab,       // [WG_SLICES, slab_stride] fp32
     long dW_off,                    // element offset of dW[N,K] in a slab
     long db_off,                    // element offset of db[N] (or -1)
-    long slab_stride, int S, int N_STRIDE, int K, int N_VALID) {
+    long slab_stride, int S, int N_STRIDE, int K, int N_VALID,
+    int n_slices) {
   __shared__ WgradLds lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
   const int nt = blockIdx.x;         // one 16-row n-tile
   const int k0 = blockIdx.y * KPG * 16;
-  const int slice = blockIdx.z * 4 + wid;  // [0, WG_SLICES)
-  const int s_per = S / WG_SLICES;   // S divisible by 16*32 in practice
+  const int slice = blockIdx.z * 4 + wid;  // [0, n_slices)
+  const int s_per = S / n_slices;
   const int s_begin = slice * s_per;
   const int s_end = s_begin + s_per;
   const int n0 = nt * 16;
@@ -199,19 +201,21 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
     }
   }
   int KTG = K / (16 * KPG);
-  dim3 grid(NT, KTG, WG_SLICES / 4), block(256);
+  int n_slices = WG_SLICES;
+  while (n_slices > 4 && (S % (n_slices * 32)) != 0) n_slices >>= 1;
+  dim3 grid(NT, KTG, n_slices / 4), block(256);
   if (KPG == 8) {
     hipLaunchKernelGGL(wgrad_kernel<8>, grid, block, 0, s, (const bf16_t*)dZ,
                        (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID);
+                       N_STRIDE, K, N_VALID, n_slices);
   } else if (KPG == 2) {
     hipLaunchKernelGGL(wgrad_kernel<2>, grid, block, 0, s, (const bf16_t*)dZ,
                        (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID);
+                       N_STRIDE, K, N_VALID, n_slices);
   } else {
     hipLaunchKernelGGL(wgrad_kernel<1>, grid, block, 0, s, (const bf16_t*)dZ,
                        (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID);
+                       N_STRIDE, K, N_VALID, n_slices);
   }
 }
 

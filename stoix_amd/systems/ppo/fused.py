@@ -82,10 +82,10 @@ class _Chain:
             self.offsets[name] = off
             off += n
         self.numel = total
-        # split-K wgrad slab: 32 fp32 partial-gradient images of the whole
+        # split-K wgrad slab: 64 fp32 partial-gradient images of the whole
         # chain (one per wave-slice; wgrad.hip WG_SLICES); slab_reduce sums
         # them into grad16
-        self.slab = torch.zeros(32, total, dtype=torch.float32, device=device)
+        self.slab = torch.zeros(64, total, dtype=torch.float32, device=device)
 
     def sync_mirror(self) -> None:
         self.flat16.copy_(self.flat)

@@ -278,7 +278,7 @@ def test_fused_ppo_update_step_runs_and_learns_shape(ext):
             "arch.num_updates=4",
             "arch.num_evaluation=1",
             "system.rollout_length=16",
-            "system.num_minibatches=4",
+            "system.num_minibatches=2",
             "system.epochs=2",
             "system.compute_dtype=bf16",
             "logger.loggers=[]",
@@ -410,12 +410,12 @@ def test_fused_vs_eager_gradients_one_minibatch(ext):
 def test_wgrad_kernel_matches_mm(ext, N, K, NV):
     """Split-K wgrad kernel (wgrad.hip): dW = dZ^T @ X and db = colsum(dZ)
     through the slab + slab_reduce path, vs torch fp32 reference."""
-    S = 1024
+    S = 2048
     g = torch.Generator().manual_seed(9)
     dZ = (torch.randn(S, N, generator=g) * 0.1).bfloat16().cuda()
     X = torch.randn(S, K, generator=g).bfloat16().cuda()
     numel = NV * K + NV
-    slab = torch.zeros(32, numel, device="cuda")
+    slab = torch.zeros(64, numel, device="cuda")
     grad16 = torch.zeros(numel, dtype=torch.bfloat16, device="cuda")
     e0 = torch.zeros(0, device="cuda")
     e1 = torch.zeros(0, dtype=torch.int64, device="cuda")
