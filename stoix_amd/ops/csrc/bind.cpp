@@ -63,6 +63,8 @@ void launch_rollout_step_ant(float*, float*, int*, float*, int*, float*,
                              float, float, float, uint64_t, uint64_t,
                              unsigned int*, unsigned int*, unsigned int,
                              void*);
+void launch_linear_silu(const void*, const void*, const float*, void*,
+                        void*, int, int, int, int, void*);
 void launch_silu_fwd(const void*, void*, long, void*);
 void launch_silu_bwd(const void*, const void*, void*, long, void*);
 void launch_ppo_gather(const long*, int, const float*, int, int,
@@ -359,6 +361,19 @@ void rollout_step_ant(torch::Tensor obs_io, torch::Tensor env_state,
       cur_stream());
 }
 
+void linear_silu(torch::Tensor X, torch::Tensor W, torch::Tensor bias,
+                 torch::Tensor Z, torch::Tensor H, int64_t do_silu) {
+  CHK(X, torch::kBFloat16);
+  CHK(W, torch::kBFloat16);
+  CHK(bias, torch::kFloat32);
+  int S = X.size(0), K = X.size(1), N = W.size(0);
+  TORCH_CHECK(S % 128 == 0 && N % 128 == 0 && K % 32 == 0,
+              "linear_silu tile constraints");
+  void* h = H.numel() > 0 ? H.data_ptr() : nullptr;
+  launch_linear_silu(X.data_ptr(), W.data_ptr(), bias.data_ptr<float>(),
+                     Z.data_ptr(), h, S, K, N, (int)do_silu, cur_stream());
+}
+
 void silu_fwd(torch::Tensor z, torch::Tensor h) {
   CHK(z, torch::kBFloat16);
   TORCH_CHECK(z.numel() % 8 == 0, "silu_fwd needs numel % 8 == 0");
@@ -480,6 +495,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("value_forward", &value_forward, "fused critic fwd (MFMA)");
   m.def("rollout_step_ant", &rollout_step_ant,
         "fused rollout step: policy + Ant physics + bootstrap, one launch");
+  m.def("linear_silu", &linear_silu,
+        "fused Linear(+bias)+SiLU forward, MFMA tiled");
   m.def("silu_fwd", &silu_fwd, "bf16 silu forward");
   m.def("silu_bwd", &silu_bwd, "bf16 silu backward");
   m.def("ppo_gather", &ppo_gather, "fused minibatch gather");

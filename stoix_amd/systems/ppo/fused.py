@@ -414,15 +414,12 @@ class FusedPPOEngine:
                 flat_tgt, self.Xmb, self.act_mb, self.logp_mb, self.val_mb,
                 self.adv_mb, self.tgt_mb, self.empty, self.empty,
             )
-            # ---- forward (bf16 GEMMs on hipBLASLt MFMA + fused silu)
-            torch.addmm(a16["b1"], self.Xmb, a16["W1"].t(), out=self.Z1a)
-            ext.silu_fwd(self.Z1a, self.H1a)
-            torch.addmm(a16["b2"], self.H1a, a16["W2"].t(), out=self.Z2a)
-            ext.silu_fwd(self.Z2a, self.H2a)
-            torch.addmm(c16["b1"], self.Xmb, c16["W1"].t(), out=self.Z1c)
-            ext.silu_fwd(self.Z1c, self.H1c)
-            torch.addmm(c16["b2"], self.H1c, c16["W2"].t(), out=self.Z2c)
-            ext.silu_fwd(self.Z2c, self.H2c)
+            # ---- forward: custom fused Linear+SiLU MFMA kernels (one
+            # launch per layer instead of GemmAndBias + silu)
+            ext.linear_silu(self.Xmb, a16["W1"], ac.views["b1"], self.Z1a, self.H1a, 1)
+            ext.linear_silu(self.H1a, a16["W2"], ac.views["b2"], self.Z2a, self.H2a, 1)
+            ext.linear_silu(self.Xmb, c16["W1"], cc.views["b1"], self.Z1c, self.H1c, 1)
+            ext.linear_silu(self.H1c, c16["W2"], cc.views["b2"], self.Z2c, self.H2c, 1)
             # ---- heads as GEMMs (hipBLASLt), then the fused per-row
             # loss + analytic head-backward kernel, then dH2 as GEMMs
             torch.addmm(a16["bh"], self.H2a, a16["Wh"].t(), out=self.heads)

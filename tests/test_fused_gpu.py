@@ -471,3 +471,20 @@ def test_fused_rollout_megakernel_invariants(ext):
     # rewards/discounts sane; steptypes valid
     assert torch.isfinite(learner.buf_reward).all()
     assert ((learner.buf_discount == 0) | (learner.buf_discount == 1)).all()
+
+
+@requires_gpu
+@pytest.mark.parametrize("K", [32, 256])
+def test_linear_silu_kernel_matches_torch(ext, K):
+    S, N = 2048, 256
+    g = torch.Generator().manual_seed(12)
+    X = (torch.randn(S, K, generator=g) / math.sqrt(K)).bfloat16().cuda()
+    W = (torch.randn(N, K, generator=g) / math.sqrt(K)).bfloat16().cuda()
+    b = (torch.randn(N, generator=g) * 0.1).cuda()
+    Z = torch.zeros(S, N, dtype=torch.bfloat16, device="cuda")
+    H = torch.zeros(S, N, dtype=torch.bfloat16, device="cuda")
+    ext.linear_silu(X, W, b, Z, H, 1)
+    torch.cuda.synchronize()
+    Zref = X.float() @ W.float().t() + b
+    torch.testing.assert_close(Z.float(), Zref, rtol=3e-2, atol=3e-2)
+    torch.testing.assert_close(H.float(), F.silu(Zref), rtol=3e-2, atol=3e-2)

@@ -120,6 +120,9 @@ def main():
     def k_silu():
         ext.silu_fwd(F.Z1a, F.H1a)
 
+    def k_linsilu():
+        ext.linear_silu(F.H1a, a16["W2"], ac.views["b2"], F.Z2a, F.H2a, 1)
+
     def k_headgemm():
         torch.addmm(a16["bh"], F.H2a, a16["Wh"].t(), out=F.heads)
 
@@ -155,6 +158,7 @@ def main():
     print(f"gemm fwd L1 (pad):  {timeit(k_gemm_fwd, 50)*1e3:8.1f} us")
     print(f"gemm fwd L2:        {timeit(k_gemm_fwd2, 50)*1e3:8.1f} us")
     print(f"silu_fwd:           {timeit(k_silu, 50)*1e3:8.1f} us")
+    print(f"linear_silu fused:  {timeit(k_linsilu, 50)*1e3:8.1f} us")
     print(f"head gemm:          {timeit(k_headgemm, 50)*1e3:8.1f} us")
     print(f"ppo_head_loss:      {timeit(k_head, 50)*1e3:8.1f} us")
     print(f"wgrad mm:           {timeit(k_wgrad, 50)*1e3:8.1f} us")
