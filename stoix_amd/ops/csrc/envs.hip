@@ -117,7 +117,10 @@ extern "C" __global__ void ant_step_kernel(
     unsigned char* __restrict__ done_out,
     int B, int max_episode_steps, uint64_t seed,
     const unsigned int* __restrict__ draw_buf, unsigned int draw_offset) {
-  int b = blockIdx.x * blockDim.x + threadIdx.x;
+  // FOUR lanes per env (lane quad, one leg each; ant_core.h x4 physics)
+  int tid = blockIdx.x * blockDim.x + threadIdx.x;
+  int b = tid >> 2;
+  int leg = tid & 3;
   if (b >= B) return;
   uint32_t draw = *draw_buf + draw_offset;
 
@@ -131,7 +134,8 @@ extern "C" __global__ void ant_step_kernel(
 
   float reward;
   bool terminated;
-  ant_physics_step(s, a, &reward, &terminated);
+  ant_physics_step_x4(s, a, leg, &reward, &terminated);
+  if (leg != 0) return;  // lane 0 of each quad does the bookkeeping
 
   // ---- wrapper semantics (autoreset + episode metrics)
   int sc = step_count[b] + 1;
@@ -411,11 +415,9 @@ extern "C" void launch_ant_step(
     unsigned char* steptype_out, unsigned char* done_out, int B,
     int max_episode_steps, uint64_t seed, unsigned int* draw_buf,
     unsigned int draw_offset, int do_bump, void* stream) {
-  // 64-thread blocks: one wave per WG spreads B=4096 envs over 64 CUs
-  // instead of 16 (the per-env physics is register-serial; CU count is the
-  // lever, not waves-per-CU)
+  // 4 threads per env (quad-cooperative physics), 64-thread blocks
   int threads = 64;
-  int blocks = (B + threads - 1) / threads;
+  int blocks = (B * 4 + threads - 1) / threads;
   hipLaunchKernelGGL(ant_step_kernel, dim3(blocks), dim3(threads), 0,
                      (hipStream_t)stream, state, action, step_count, ep_return,
                      ep_length, last_ep_return, last_ep_length, obs_out,

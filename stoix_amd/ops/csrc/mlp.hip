@@ -481,9 +481,12 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
   }
   __syncthreads();
 
-  // ---- env step: 16 threads, one per row (Ant physics in registers)
-  if (threadIdx.x < 16) {
-    int b = rbase + threadIdx.x;
+  // ---- env step: wave 0's 64 lanes, FOUR lanes per env (one leg each;
+  // ant_physics_step_x4 quad-cooperative physics)
+  if (threadIdx.x < 64) {
+    int row = threadIdx.x >> 2;
+    int leg = threadIdx.x & 3;
+    int b = rbase + row;
     if (b < B) {
       float es[ANT_STATE];
 #pragma unroll
@@ -491,11 +494,11 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       float ea[ANT_ACT];
 #pragma unroll
       for (int i = 0; i < ANT_ACT; ++i)
-        ea[i] = fminf(fmaxf(lds.act[threadIdx.x][i], -1.0f), 1.0f);
+        ea[i] = fminf(fmaxf(lds.act[row][i], -1.0f), 1.0f);
       float reward;
       bool terminated;
-      ant_physics_step(es, ea, &reward, &terminated);
-
+      ant_physics_step_x4(es, ea, leg, &reward, &terminated);
+      if (leg == 0) {  // quad lane 0 does all the bookkeeping/writes
       int sc = step_count[b] + 1;
       bool truncated = (sc >= max_episode_steps) && !terminated;
       bool done = terminated || truncated;
@@ -504,8 +507,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
 
       // pre-reset next_obs -> global row t (the bootstrap critic runs
-      // ONCE over all T*B rows after the rollout: one big MFMA-efficient
-      // launch instead of 128 small 2-layer passes)
+      // ONCE over all T*B rows after the rollout)
       ant_write_obs(es, next_obs_buf + (long)b * ANT_OBS);
 
       if (done) {
@@ -522,6 +524,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       buf_discount[b] = terminated ? 0.0f : 1.0f;
       buf_steptype[b] =
           terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+      }  // leg == 0
     }
   }
 }
