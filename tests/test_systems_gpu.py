@@ -1,0 +1,69 @@
+"""GPU smoke tests for the non-flagship systems (the driver re-runs
+`pytest -m gpu` on a real MI355X at round end; these pin the off-policy /
+search tiers to the HIP env kernels + device-resident buffers)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+requires_gpu = pytest.mark.skipif(not torch.cuda.is_available(), reason="no GPU")
+
+TINY = [
+    "arch.total_timesteps=null", "arch.num_updates=2", "arch.num_evaluation=1",
+    "arch.num_eval_episodes=4", "logger.loggers=[]",
+    "logger.checkpointing.save_model=false",
+]
+
+
+def _run(module, default, extra):
+    import importlib
+
+    from stoix_amd.config import compose
+
+    mod = importlib.import_module(module)
+    cfg = compose(default, TINY + extra)
+    r = mod.run(cfg)
+    assert r == r
+
+
+@requires_gpu
+def test_gpu_sac_humanoid():
+    _run("stoix_amd.systems.sac.ff_sac", "default/anakin/default_ff_sac.yaml",
+         ["env=brax/humanoid", "arch.total_num_envs=128",
+          "system.rollout_length=4", "system.batch_size=128",
+          "system.buffer_size=8192", "system.warmup_steps=16"])
+
+
+@requires_gpu
+def test_gpu_rainbow_snake():
+    _run("stoix_amd.systems.q_learning.ff_rainbow",
+         "default/anakin/default_ff_rainbow.yaml",
+         ["env=jumanji/snake", "arch.total_num_envs=64",
+          "system.rollout_length=4", "system.batch_size=64",
+          "system.buffer_size=4096", "system.warmup_steps=16",
+          "system.n_step=3", "system.epochs=2"])
+
+
+@requires_gpu
+def test_gpu_dqn_cartpole():
+    _run("stoix_amd.systems.q_learning.ff_dqn",
+         "default/anakin/default_ff_dqn.yaml",
+         ["env=classic/cartpole", "arch.total_num_envs=128",
+          "system.rollout_length=4", "system.batch_size=128",
+          "system.buffer_size=8192", "system.warmup_steps=16"])
+
+
+@requires_gpu
+def test_gpu_td3_ant():
+    _run("stoix_amd.systems.ddpg.ff_td3", "default/anakin/default_ff_td3.yaml",
+         ["env=brax/ant", "arch.total_num_envs=128",
+          "system.rollout_length=4", "system.batch_size=128",
+          "system.buffer_size=8192", "system.warmup_steps=16"])
+
+
+@requires_gpu
+def test_gpu_az_cartpole():
+    _run("stoix_amd.systems.search.ff_az", "default/anakin/default_ff_az.yaml",
+         ["env=classic/cartpole", "arch.total_num_envs=32",
+          "system.rollout_length=4", "system.num_simulations=6",
+          "system.num_minibatches=2", "system.epochs=1"])
