@@ -28,14 +28,18 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 // largest power of two <= 64 that divides S/32 (small minibatches use fewer)
 #define SPAD 8        // +8 cols on the 32-wide transposed tiles
 
-// One wave's private staging: dZt [16 n][32+8 s], Xt [KPG*16 k][32+8 s].
-// KPG <= 8 -> per-wave 40*(16+128)*2B = 11.5 KB; 4 waves = 46 KB.
+// One wave's private staging: dZt [NTB*16 n][32+8 s], Xt [KPG*16 k][32+8 s].
+// NTB n-tiles per workgroup cut the X read-amplification (every n-tile
+// group re-reads the whole X k-slice; at NTB=1 a 256x256 wgrad re-reads
+// the 16 MB X sixteen times = 256 MB -> the kernel is HBM/L2-traffic
+// bound). NTB=4, KPG=8: per-wave 4*16*40*2 + 8*16*40*2 = 15.3 KB; x4
+// waves = 61 KB.
 struct WgradLds {
-  bf16_t dZt[4][16][32 + SPAD];
+  bf16_t dZt[4][4 * 16][32 + SPAD];
   bf16_t Xt[4][8 * 16][32 + SPAD];
 };
 
-template <int KPG>
+template <int KPG, int NTB>
 __launch_bounds__(256, 2) __global__ void wgrad_kernel(
     const bf16_t* __restrict__ dZ,  // [S, N_STRIDE]
     const bf16_t* __restrict__ X,   // [S, K]
@@ -47,7 +51,7 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
   __shared__ WgradLds lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int nt = blockIdx.x;         // one 16-row n-tile
+  const int nt = blockIdx.x * NTB;   // NTB 16-row n-tiles per workgroup
   const int k0 = blockIdx.y * KPG * 16;
   const int slice = blockIdx.z * 4 + wid;  // [0, n_slices)
   const int s_per = S / n_slices;
@@ -58,24 +62,27 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
   bf16_t(*dZt)[32 + SPAD] = lds.dZt[wid];
   bf16_t(*Xt)[32 + SPAD] = lds.Xt[wid];
 
-  f32x4 acc[KPG];
+  f32x4 acc[NTB][KPG];
 #pragma unroll
-  for (int t = 0; t < KPG; ++t) acc[t] = {0.f, 0.f, 0.f, 0.f};
-  float db_acc = 0.0f;  // lane's partial of db[n0 + (lane&15)]
+  for (int u = 0; u < NTB; ++u)
+#pragma unroll
+    for (int t = 0; t < KPG; ++t) acc[u][t] = {0.f, 0.f, 0.f, 0.f};
+  float db_acc[NTB];
+#pragma unroll
+  for (int u = 0; u < NTB; ++u) db_acc[u] = 0.0f;
 
   const int arow = lane & 15;
   const int ak0 = (lane >> 4) * 8;
 
   for (int s0 = s_begin; s0 < s_end; s0 += 32) {
-    // ---- stage dZ tile [32 s][16 n] -> dZt [16 n][32 s] (transposed)
-    {
-      // lane l: s-row = l>>1, n-half = (l&1)*8 -> one 16B load, 8 scatter
-      // writes
+    // ---- stage dZ tiles [32 s][NTB*16 n] -> dZt [NTB*16 n][32 s]
+#pragma unroll
+    for (int u = 0; u < NTB; ++u) {
       int srow = lane >> 1, nh = (lane & 1) * 8;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
-          dZ + (long)(s0 + srow) * N_STRIDE + n0 + nh);
+          dZ + (long)(s0 + srow) * N_STRIDE + n0 + u * 16 + nh);
 #pragma unroll
-      for (int i = 0; i < 8; ++i) dZt[nh + i][srow] = v[i];
+      for (int i = 0; i < 8; ++i) dZt[u * 16 + nh + i][srow] = v[i];
     }
     // ---- stage X tiles [32 s][KPG*16 k] -> Xt [KPG*16 k][32 s].
     // Lane mapping: consecutive lanes take consecutive s-rows of ONE
@@ -96,18 +103,27 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
     }
     // within-wave LDS write->read ordering is compiler-tracked (lgkmcnt);
     // no cross-wave sharing, so no barrier.
-    // ---- fragments + MFMA
-    const bf16x8 a = *reinterpret_cast<const bf16x8*>(&dZt[arow][ak0]);
+    // ---- fragments + MFMA (B fragments shared by the NTB A tiles)
 #pragma unroll
     for (int t = 0; t < KPG; ++t) {
       const bf16x8 b =
           *reinterpret_cast<const bf16x8*>(&Xt[t * 16 + arow][ak0]);
-      acc[t] = MFMA_BF16_16x16x32(a, b, acc[t], 0, 0, 0);
+#pragma unroll
+      for (int u = 0; u < NTB; ++u) {
+        const bf16x8 a =
+            *reinterpret_cast<const bf16x8*>(&dZt[u * 16 + arow][ak0]);
+        acc[u][t] = MFMA_BF16_16x16x32(a, b, acc[u][t], 0, 0, 0);
+      }
     }
-    // ---- bias partial from the A fragment (only the kt-group-0 blocks)
+    // ---- bias partials from the A fragments (kt-group-0 blocks only)
     if (db_off >= 0 && blockIdx.y == 0) {
 #pragma unroll
-      for (int j = 0; j < 8; ++j) db_acc += (float)a[j];
+      for (int u = 0; u < NTB; ++u) {
+        const bf16x8 a =
+            *reinterpret_cast<const bf16x8*>(&dZt[u * 16 + arow][ak0]);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) db_acc[u] += (float)a[j];
+      }
     }
   }
 
@@ -116,21 +132,23 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
   const int col = lane & 15;  // k within tile
   const int g = lane >> 4;
 #pragma unroll
-  for (int t = 0; t < KPG; ++t) {
+  for (int u = 0; u < NTB; ++u) {
 #pragma unroll
-    for (int r = 0; r < 4; ++r) {
-      int n = g * 4 + r;  // D row = n (A is dZ^T)
-      if (n < N_VALID)
-        out[dW_off + (long)(n0 + n) * K + k0 + t * 16 + col] = acc[t][r];
+    for (int t = 0; t < KPG; ++t) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        int n = u * 16 + g * 4 + r;  // D row = n (A is dZ^T)
+        if (n0 + n < N_VALID)
+          out[dW_off + (long)(n0 + n) * K + k0 + t * 16 + col] = acc[u][t][r];
+      }
     }
-  }
-  if (db_off >= 0 && blockIdx.y == 0) {
-    // db[n0 + arow] partial: lanes {arow, arow+16, arow+32, arow+48} hold
-    // s-disjoint partials; fold with xor-shuffles over the high bits
-    db_acc += __shfl_xor(db_acc, 16);
-    db_acc += __shfl_xor(db_acc, 32);
-    if ((lane >> 4) == 0 && arow < N_VALID)
-      out[db_off + n0 + arow] = db_acc;
+    if (db_off >= 0 && blockIdx.y == 0) {
+      float d = db_acc[u];
+      d += __shfl_xor(d, 16);
+      d += __shfl_xor(d, 32);
+      if ((lane >> 4) == 0 && n0 + u * 16 + arow < N_VALID)
+        out[db_off + n0 + u * 16 + arow] = d;
+    }
   }
 }
 
@@ -189,13 +207,16 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
                              void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int NT = (N_VALID + 15) / 16;
-  // largest KPG whose grid still spans >= 16 workgroups: small-N shapes
-  // (head grads, NT=1) would otherwise run on 8 CUs of 256
+  // NTB n-tiles per workgroup (X-amplification = NT/NTB); largest KPG whose
+  // grid still spans >= 16 workgroups (small-N head grads would otherwise
+  // run on 8 CUs of 256)
+  int NTB = (NT % 4 == 0) ? 4 : 1;
   int KPG = 1;
   const int cands[3] = {8, 2, 1};
   for (int ci = 0; ci < 3; ++ci) {
     int cand = cands[ci];
-    if (K % (16 * cand) == 0 && (cand == 1 || NT * (K / (16 * cand)) >= 16)) {
+    if (K % (16 * cand) == 0 &&
+        (cand == 1 || (NT / NTB) * (K / (16 * cand)) >= 2)) {
       KPG = cand;
       break;
     }
@@ -203,20 +224,21 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
   int KTG = K / (16 * KPG);
   int n_slices = WG_SLICES;
   while (n_slices > 4 && (S % (n_slices * 32)) != 0) n_slices >>= 1;
-  dim3 grid(NT, KTG, n_slices / 4), block(256);
-  if (KPG == 8) {
-    hipLaunchKernelGGL(wgrad_kernel<8>, grid, block, 0, s, (const bf16_t*)dZ,
-                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID, n_slices);
-  } else if (KPG == 2) {
-    hipLaunchKernelGGL(wgrad_kernel<2>, grid, block, 0, s, (const bf16_t*)dZ,
-                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID, n_slices);
+  dim3 grid(NT / NTB, KTG, n_slices / 4), block(256);
+#define WGRAD_LAUNCH(KPGV, NTBV)                                            \
+  hipLaunchKernelGGL((wgrad_kernel<KPGV, NTBV>), grid, block, 0, s,         \
+                     (const bf16_t*)dZ, (const bf16_t*)X, slab, dW_off,     \
+                     db_off, slab_stride, S, N_STRIDE, K, N_VALID, n_slices)
+  if (NTB == 4) {
+    if (KPG == 8) WGRAD_LAUNCH(8, 4);
+    else if (KPG == 2) WGRAD_LAUNCH(2, 4);
+    else WGRAD_LAUNCH(1, 4);
   } else {
-    hipLaunchKernelGGL(wgrad_kernel<1>, grid, block, 0, s, (const bf16_t*)dZ,
-                       (const bf16_t*)X, slab, dW_off, db_off, slab_stride, S,
-                       N_STRIDE, K, N_VALID, n_slices);
+    if (KPG == 8) WGRAD_LAUNCH(8, 1);
+    else if (KPG == 2) WGRAD_LAUNCH(2, 1);
+    else WGRAD_LAUNCH(1, 1);
   }
+#undef WGRAD_LAUNCH
 }
 
 extern "C" void launch_slab_reduce(float* slab, void* grad16,
