@@ -367,7 +367,7 @@ void linear_silu(torch::Tensor X, torch::Tensor W, torch::Tensor bias,
   CHK(W, torch::kBFloat16);
   CHK(bias, torch::kFloat32);
   int S = X.size(0), K = X.size(1), N = W.size(0);
-  TORCH_CHECK(S % 128 == 0 && N % 128 == 0 && K % 32 == 0 && K <= 256,
+  TORCH_CHECK(S % 128 == 0 && N % 128 == 0 && K % 32 == 0,
               "linear_silu tile constraints");
   void* h = H.numel() > 0 ? H.data_ptr() : nullptr;
   launch_linear_silu(X.data_ptr(), W.data_ptr(), bias.data_ptr<float>(),

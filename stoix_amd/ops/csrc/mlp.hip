@@ -579,13 +579,11 @@ extern "C" void launch_rollout_step_ant(
 // 64x64 each); X staged in LDS double-buffered, W fragments straight from
 // L2 (row-major [N,K] bf16 mirrors, hot across minibatches).
 #define LSF_BM 128
-#define LSF_KMAX 256
+#define LSF_BK 32
 #define LSF_APAD 8
 
-// whole K staged once (K <= 256: 128x264 bf16 = 66 KB, 2 WGs/CU) -> ONE
-// barrier per workgroup instead of one per 32-deep k-step
 struct LinSiluLds {
-  bf16_t A[LSF_BM][LSF_KMAX + LSF_APAD];
+  bf16_t A[2][LSF_BM][LSF_BK + LSF_APAD];
 };
 
 template <bool SILU>
@@ -604,7 +602,7 @@ __launch_bounds__(256, 2) __global__ void linear_silu_kernel(
   // wave -> 64x64 quadrant
   const int wm = (wid >> 1) * 64;  // 0 or 64 (row offset in tile)
   const int wn = (wid & 1) * 64;   // 0 or 64 (col offset)
-  const int KS = K / 32;
+  const int KS = K / LSF_BK;
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -612,29 +610,35 @@ __launch_bounds__(256, 2) __global__ void linear_silu_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = {0.f, 0.f, 0.f, 0.f};
 
-  // stage the whole [128 x K] X tile (256 threads, 16B pieces, coalesced)
-  {
-    int pieces = LSF_BM * K / 8;
-    for (int flat = threadIdx.x; flat < pieces; flat += 256) {
-      int row = flat / (K / 8);
-      int kh = (flat % (K / 8)) * 8;
-      *reinterpret_cast<bf16x8*>(&lds.A[row][kh]) =
-          *reinterpret_cast<const bf16x8*>(X + (long)(m0 + row) * K + kh);
+  // stage k-step `ks` of the X tile into buffer pp: 128x32 bf16 = 8 KB,
+  // 256 threads x 2 16B pieces, fully coalesced both sides
+  auto stage = [&](int ks, int pp) {
+#pragma unroll
+    for (int q = 0; q < 2; ++q) {
+      int flat = q * 256 + threadIdx.x;  // 512 chunks of 8 elems
+      int row = flat >> 2;
+      int kh = (flat & 3) * 8;
+      *reinterpret_cast<bf16x8*>(&lds.A[pp][row][kh]) =
+          *reinterpret_cast<const bf16x8*>(X + (long)(m0 + row) * K +
+                                           ks * LSF_BK + kh);
     }
-  }
-  __syncthreads();
+  };
+
+  stage(0, 0);
   const int arow_base = wm + (lane & 15);
   const int ak0 = (lane >> 4) * 8;
   for (int ks = 0; ks < KS; ++ks) {
+    __syncthreads();  // staged tile visible to all waves
+    if (ks + 1 < KS) stage(ks + 1, (ks + 1) & 1);
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          &lds.A[arow_base + mi * 16][ks * 32 + ak0]);
+          &lds.A[ks & 1][arow_base + mi * 16][ak0]);
 #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
         int n = n_blk + wn + nj * 16 + (lane & 15);
         const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            W + (long)n * K + ks * 32 + ak0);
+            W + (long)n * K + ks * LSF_BK + ak0);
         acc[mi][nj] = MFMA_BF16_16x16x32(a, b, acc[mi][nj], 0, 0, 0);
       }
     }
