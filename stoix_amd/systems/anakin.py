@@ -48,6 +48,16 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
 
     learner = learner_factory(config, train_env, device)
 
+    # hip-graph capture of the update loop (rollout graph + epoch graph)
+    # for capture-shaped learners on GPU (ops/graph.py); eager otherwise
+    if device.type == "cuda" and hasattr(learner, "rollout_phase") and hasattr(learner, "epoch_phase"):
+        try:
+            from stoix_amd.ops.graph import try_enable_graphs
+
+            try_enable_graphs(learner)
+        except Exception as e:
+            print(f"[anakin] hip-graph capture unavailable ({e!r}); running eager")
+
     eval_fn, absolute_eval_fn = evaluator_setup(eval_env, config)
     logger = StoixLogger(config) if ctx.is_main else None
     checkpointer = None
