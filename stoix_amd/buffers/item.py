@@ -17,14 +17,21 @@ Tensor = torch.Tensor
 
 
 class ItemBuffer:
+    """The write cursor and size live in DEVICE tensors so that add() and
+    sample() are hip-graph-capturable: a captured add advances the cursor
+    on-device every replay (a Python-int cursor would bake the capture-time
+    slots into the graph and silently overwrite them forever)."""
+
     def __init__(self, capacity: int, device: torch.device | str = "cpu", seed: int = 0):
         self.capacity = int(capacity)
         self.device = torch.device(device)
         self.gen = torch.Generator(device=self.device)
         self.gen.manual_seed(seed)
         self.storage: Dict[str, Tensor] = {}
-        self.ptr = 0
-        self.size = 0
+        self._ptr = torch.zeros(1, dtype=torch.int64, device=self.device)
+        self._size = torch.zeros(1, dtype=torch.int64, device=self.device)
+        # graph mode switches sampling to the default (graph-aware) RNG
+        self.graph_safe_rng = False
 
     def _alloc(self, example: Dict[str, Tensor]) -> None:
         for k, v in example.items():
@@ -32,23 +39,33 @@ class ItemBuffer:
                 (self.capacity, *v.shape[1:]), dtype=v.dtype, device=self.device
             )
 
+    @property
+    def ptr(self) -> int:
+        return int(self._ptr.item())
+
+    @property
+    def size(self) -> int:
+        return int(self._size.item())
+
     @torch.no_grad()
     def add(self, batch: Dict[str, Tensor]) -> None:
-        """Add a batch of items (leading dim = batch)."""
+        """Add a batch of items (leading dim = batch); pure device ops."""
         if not self.storage:
             self._alloc(batch)
         b = next(iter(batch.values())).shape[0]
-        idx = (torch.arange(b, device=self.device) + self.ptr) % self.capacity
+        idx = (torch.arange(b, device=self.device) + self._ptr) % self.capacity
         for k, v in batch.items():
             self.storage[k][idx] = v.to(self.device)
-        self.ptr = (self.ptr + b) % self.capacity
-        self.size = min(self.size + b, self.capacity)
+        self._ptr.add_(b).remainder_(self.capacity)
+        self._size.add_(b).clamp_(max=self.capacity)
 
     @torch.no_grad()
     def sample(self, batch_size: int) -> Dict[str, Tensor]:
-        idx = torch.randint(0, self.size, (batch_size,), device=self.device, generator=self.gen)
+        gen = None if self.graph_safe_rng else self.gen
+        u = torch.rand(batch_size, device=self.device, generator=gen)
+        idx = (u * self._size.to(torch.float32)).long().clamp_(max=self.capacity - 1)
         return {k: v[idx] for k, v in self.storage.items()}
 
     @property
     def can_sample(self) -> bool:
-        return self.size > 0
+        return bool((self._size > 0).item())

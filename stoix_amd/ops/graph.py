@@ -73,3 +73,40 @@ def try_enable_graphs(learner) -> bool:
     learner._graphs = (g_rollout, g_epoch)
     learner.update_step = update_step
     return True
+
+
+def try_enable_update_graph(learner) -> bool:
+    """Capture an off-policy learner's whole update_step (rollout into the
+    device-resident replay buffer + epochs of sample/loss/backward/Adam/
+    polyak) into ONE hip graph. Requires: HIP env (no host-side autoreset
+    branch), device-cursor ItemBuffer, capturable Adam. The learner opts in
+    via ``graph_capturable`` and provides prepare_for_graph_capture /
+    after_graph_replay hooks (metrics read eagerly from the env's latched
+    buffers after each replay)."""
+    device = learner.device
+    if device.type != "cuda":
+        return False
+    if getattr(learner.env, "_hip", None) is None:
+        raise RuntimeError("update-graph capture needs the HIP env step path")
+    learner.prepare_for_graph_capture()
+
+    side = torch.cuda.Stream(device)
+    side.wait_stream(torch.cuda.current_stream(device))
+    with torch.cuda.stream(side):
+        for _ in range(2):
+            learner.update_step()
+    torch.cuda.current_stream(device).wait_stream(side)
+    torch.cuda.synchronize(device)
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        static_metrics = learner.update_step()
+
+    def update_step() -> Dict[str, torch.Tensor]:
+        g.replay()
+        learner.after_graph_replay()
+        return static_metrics
+
+    learner._graphs = (g,)
+    learner.update_step = update_step
+    return True

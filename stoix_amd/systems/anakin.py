@@ -57,6 +57,13 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             try_enable_graphs(learner)
         except Exception as e:
             print(f"[anakin] hip-graph capture unavailable ({e!r}); running eager")
+    elif device.type == "cuda" and getattr(learner, "graph_capturable", False):
+        try:
+            from stoix_amd.ops.graph import try_enable_update_graph
+
+            try_enable_update_graph(learner)
+        except Exception as e:
+            print(f"[anakin] update-graph capture unavailable ({e!r}); running eager")
 
     eval_fn, absolute_eval_fn = evaluator_setup(eval_env, config)
     logger = StoixLogger(config) if ctx.is_main else None

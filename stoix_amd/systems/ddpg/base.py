@@ -51,8 +51,8 @@ class DDPGFamilyLearner:
         for p in list(self.actor_target.parameters()) + list(self.q_target.parameters()):
             p.requires_grad_(False)
 
-        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr))
-        self.q_opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr))
+        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr), capturable=device.type == "cuda")
+        self.q_opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), capturable=device.type == "cuda")
         self.reducer = FlatGradReducer(
             list(self.actor.parameters()) + list(self.q_online.parameters()), device
         )
@@ -67,6 +67,7 @@ class DDPGFamilyLearner:
 
         self.ts = env.reset()
         self.episode_metrics: Dict[str, Tensor] = {}
+        self.collect_metrics = True
         self._warmup()
 
     # --------------------------------------------------------------- acting
@@ -106,10 +107,11 @@ class DDPGFamilyLearner:
             )
             ts = next_ts
         self.ts = ts
-        em = ts.extras["episode_metrics"]
-        final, has = get_final_step_metrics(em)
-        if has:
-            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+        if self.collect_metrics:
+            em = ts.extras["episode_metrics"]
+            final, has = get_final_step_metrics(em)
+            if has:
+                self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
     def _warmup(self) -> None:
         steps = max(1, int(getattr(self.sys, "warmup_steps", 64)) // self.B + 1)
@@ -166,6 +168,32 @@ class DDPGFamilyLearner:
                     pt.mul_(1.0 - tau).add_(po, alpha=tau)
             self.update_count += 1
         return metrics
+
+
+    # ------------------------------------------------------- graph support
+
+    @property
+    def graph_capturable(self) -> bool:
+        # the delayed-actor branch (update_count % policy_frequency) varies
+        # per call; a single captured graph is only valid at frequency 1
+        return (
+            getattr(self.env, "_hip", None) is not None
+            and int(getattr(self.sys, "policy_frequency", 1)) == 1
+        )
+
+    def prepare_for_graph_capture(self) -> None:
+        """Capture-safe modes: default (graph-aware) CUDA RNG, inline
+        all-reduce, no host-side metric reads, graph-safe buffer RNG."""
+        self.gen = None
+        self.collect_metrics = False
+        self.reducer._stream = None
+        self.buffer.graph_safe_rng = True
+
+    def after_graph_replay(self) -> None:
+        self.episode_metrics = {
+            "episode_return": self.env._last_ep_return,
+            "episode_length": self.env._last_ep_length.to(torch.float32),
+        }
 
     # ------------------------------------------------------------ checkpoint
 

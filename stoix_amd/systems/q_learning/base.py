@@ -43,7 +43,7 @@ class OffPolicyQLearner:
         for p in self.q_target.parameters():
             p.requires_grad_(False)
 
-        self.opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), eps=1e-5)
+        self.opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), eps=1e-5, capturable=device.type == "cuda")
         self.reducer = FlatGradReducer(self.q_online.parameters(), device)
         self.gen = torch.Generator(device=device)
         self.gen.manual_seed(int(config.arch.seed) * 7919 + 41)
@@ -57,6 +57,7 @@ class OffPolicyQLearner:
 
         self.ts = env.reset()
         self.episode_metrics: Dict[str, Tensor] = {}
+        self.collect_metrics = True
         self._warmup()
 
     # ------------------------------------------------------------- networks
@@ -116,10 +117,11 @@ class OffPolicyQLearner:
             )
             ts = next_ts
         self.ts = ts
-        em = ts.extras["episode_metrics"]
-        final, has = get_final_step_metrics(em)
-        if has:
-            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+        if self.collect_metrics:
+            em = ts.extras["episode_metrics"]
+            final, has = get_final_step_metrics(em)
+            if has:
+                self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
     def _warmup(self) -> None:
         steps = max(1, int(getattr(self.sys, "warmup_steps", 16)) // self.B + 1)
@@ -158,6 +160,27 @@ class OffPolicyQLearner:
                 bt.mul_(1.0 - tau).add_(bo, alpha=tau)
             else:
                 bt.copy_(bo)
+
+
+    # ------------------------------------------------------- graph support
+
+    @property
+    def graph_capturable(self) -> bool:
+        return getattr(self.env, "_hip", None) is not None
+
+    def prepare_for_graph_capture(self) -> None:
+        """Capture-safe modes: default (graph-aware) CUDA RNG, inline
+        all-reduce, no host-side metric reads, graph-safe buffer RNG."""
+        self.gen = None
+        self.collect_metrics = False
+        self.reducer._stream = None
+        self.buffer.graph_safe_rng = True
+
+    def after_graph_replay(self) -> None:
+        self.episode_metrics = {
+            "episode_return": self.env._last_ep_return,
+            "episode_length": self.env._last_ep_length.to(torch.float32),
+        }
 
     # ------------------------------------------------------------ checkpoint
 

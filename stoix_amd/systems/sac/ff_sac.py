@@ -58,9 +58,9 @@ class SACLearner:
         )
         self.autotune = bool(getattr(self.sys, "autotune", True))
 
-        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr))
-        self.q_opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr))
-        self.alpha_opt = torch.optim.Adam([self.log_alpha], lr=float(self.sys.alpha_lr))
+        self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr), capturable=device.type == "cuda")
+        self.q_opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), capturable=device.type == "cuda")
+        self.alpha_opt = torch.optim.Adam([self.log_alpha], lr=float(self.sys.alpha_lr), capturable=device.type == "cuda")
         self.reducer = FlatGradReducer(
             list(self.actor.parameters()) + list(self.q_online.parameters()) + [self.log_alpha],
             device,
@@ -74,6 +74,7 @@ class SACLearner:
 
         self.ts = env.reset()
         self.episode_metrics: Dict[str, Tensor] = {}
+        self.collect_metrics = True
         self._warmup()
 
     # --------------------------------------------------------------- acting
@@ -106,10 +107,11 @@ class SACLearner:
             )
             ts = next_ts
         self.ts = ts
-        em = ts.extras["episode_metrics"]
-        final, has = get_final_step_metrics(em)
-        if has:
-            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+        if self.collect_metrics:
+            em = ts.extras["episode_metrics"]
+            final, has = get_final_step_metrics(em)
+            if has:
+                self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
     def _warmup(self) -> None:
         steps = max(1, int(getattr(self.sys, "warmup_steps", 64)) // self.B + 1)
@@ -172,6 +174,29 @@ class SACLearner:
                 "entropy": -logp_new.mean().detach(),
             }
         return metrics
+
+    # ------------------------------------------------------------ checkpoint
+
+
+    # ------------------------------------------------------- graph support
+
+    @property
+    def graph_capturable(self) -> bool:
+        return getattr(self.env, "_hip", None) is not None
+
+    def prepare_for_graph_capture(self) -> None:
+        """Capture-safe modes: default (graph-aware) CUDA RNG, inline
+        all-reduce, no host-side metric reads, graph-safe buffer RNG."""
+        self.gen = None
+        self.collect_metrics = False
+        self.reducer._stream = None
+        self.buffer.graph_safe_rng = True
+
+    def after_graph_replay(self) -> None:
+        self.episode_metrics = {
+            "episode_return": self.env._last_ep_return,
+            "episode_length": self.env._last_ep_length.to(torch.float32),
+        }
 
     # ------------------------------------------------------------ checkpoint
 

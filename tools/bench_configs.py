@@ -77,6 +77,10 @@ def bench_sac_humanoid(quick):
     device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     env = environments.make_single(cfg, B, device, seed=0)
     learner = SACLearner(cfg, env, device)
+    if device.type == "cuda":
+        from stoix_amd.ops.graph import try_enable_update_graph
+
+        try_enable_update_graph(learner)
     sps, ms = time_anakin(learner, 8, B, 5 if quick else 20, 2)
     # HBM residency of the replay buffer
     buf_bytes = sum(v.numel() * v.element_size() for v in learner.buffer.storage.values()) \
