@@ -51,7 +51,7 @@ class HipStepMixin:
         # stored the NEXT step's observation (stream-ordered overwrite),
         # putting actions and observations off by one. That broke PPO
         # learning on GPU while every single-step numerics test passed
-        # (tools/envdebug*.py forensics). The clones cost ~6 small copies
+        # (forensics recorded in profiles/r01_learning_curves.md). The clones cost ~6 small copies
         # per eager step; the fused rollout path (FusedPPOEngine) bypasses
         # step() entirely and keeps the zero-copy pipeline.
         self._hb: Dict[str, torch.Tensor] = mk()

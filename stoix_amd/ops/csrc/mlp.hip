@@ -630,16 +630,22 @@ __launch_bounds__(256, 2) __global__ void linear_silu_kernel(
   for (int ks = 0; ks < KS; ++ks) {
     __syncthreads();  // staged tile visible to all waves
     if (ks + 1 < KS) stage(ks + 1, (ks + 1) & 1);
+    // B fragments hoisted out of the mi loop (they depend only on nj, ks;
+    // reloading them per mi quadrupled the L2 W-traffic)
+    bf16x8 b[4];
+#pragma unroll
+    for (int nj = 0; nj < 4; ++nj) {
+      int n = n_blk + wn + nj * 16 + (lane & 15);
+      b[nj] = *reinterpret_cast<const bf16x8*>(W + (long)n * K +
+                                               ks * LSF_BK + ak0);
+    }
 #pragma unroll
     for (int mi = 0; mi < 4; ++mi) {
       const bf16x8 a = *reinterpret_cast<const bf16x8*>(
           &lds.A[ks & 1][arow_base + mi * 16][ak0]);
 #pragma unroll
       for (int nj = 0; nj < 4; ++nj) {
-        int n = n_blk + wn + nj * 16 + (lane & 15);
-        const bf16x8 b = *reinterpret_cast<const bf16x8*>(
-            W + (long)n * K + ks * LSF_BK + ak0);
-        acc[mi][nj] = MFMA_BF16_16x16x32(a, b, acc[mi][nj], 0, 0, 0);
+        acc[mi][nj] = MFMA_BF16_16x16x32(a, b[nj], acc[mi][nj], 0, 0, 0);
       }
     }
   }
