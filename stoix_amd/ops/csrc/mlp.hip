@@ -104,6 +104,8 @@ struct MlpLds {
   bf16_t O[16][K1P_MAX + OPAD];
   bf16_t H[4][16][HID + HPAD];
   float act[16][8];  // sampled actions for the in-kernel env step
+  int done[16];      // per-row episode-end flags from the env phase
+  int any_done;      // OR of done[] (gates the bootstrap critic pass)
 };
 
 // Load one B fragment from a row-major [N,K] bf16 weight matrix.
@@ -385,7 +387,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
     float* __restrict__ buf_action,      // [B, 8]
     float* __restrict__ buf_logp,        // [B]
     float* __restrict__ buf_value,       // [B]
-    float* __restrict__ next_obs_buf,    // [B, 27] pre-reset next obs row t
+    float* __restrict__ buf_bootstrap,   // [B] V(next_obs), DONE rows only
     float* __restrict__ buf_reward, float* __restrict__ buf_discount,
     unsigned char* __restrict__ buf_steptype,
     int B, int OBS, int ACT, int max_episode_steps, float min_scale,
@@ -410,6 +412,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
   bf16_t* Hc0 = &lds.H[2][0][0];
   bf16_t* Hc1 = &lds.H[3][0][0];
 
+  if (threadIdx.x == 0) lds.any_done = 0;
   stage_obs(obs_io, rbase, OBS, K1P, O, OS, nullptr, nullptr, buf_obs,
             threadIdx.x, 256, B);
   __syncthreads();
@@ -506,9 +509,18 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       int len = ep_length[b] + 1;
       if (done) { last_ep_return[b] = ret; last_ep_length[b] = len; }
 
-      // pre-reset next_obs -> global row t (the bootstrap critic runs
-      // ONCE over all T*B rows after the rollout)
-      ant_write_obs(es, next_obs_buf + (long)b * ANT_OBS);
+      // pre-reset next_obs -> LDS obs tile; the in-kernel bootstrap
+      // critic below runs ONLY for workgroups with done rows: for every
+      // non-done row V(next_obs) == V(obs_{t+1}) == next step's buf_value,
+      // which the engine fills in with one shifted masked copy
+      {
+        float nobs[ANT_OBS];
+        ant_write_obs(es, nobs);
+#pragma unroll
+        for (int k = 0; k < ANT_OBS; ++k) lds.O[row][k] = f2bf(nobs[k]);
+        lds.done[row] = done ? 1 : 0;
+        if (done) lds.any_done = 1;
+      }
 
       if (done) {
         ant_reset_state(es, env_seed, (uint32_t)b, edraw);
@@ -527,6 +539,32 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       }  // leg == 0
     }
   }
+  __syncthreads();
+
+  // ---- bootstrap critic, only when this WG saw an episode end
+  if (lds.any_done) {
+    wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
+    __syncthreads();
+    wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
+    __syncthreads();
+    if (wid == 0) {
+      // value head, but store only the done rows (others get the shifted
+      // next-step value from the engine)
+      int row = lane >> 2;
+      int part = lane & 3;
+      float acc = 0.0f;
+      for (int c = part * (HID / 4); c < (part + 1) * (HID / 4); c += 8) {
+        bf16x8 h = *reinterpret_cast<const bf16x8*>(Hc1 + row * HS + c);
+        bf16x8 w = *reinterpret_cast<const bf16x8*>(Wvc + c);
+#pragma unroll
+        for (int j = 0; j < 8; ++j) acc += bf2f(h[j]) * bf2f(w[j]);
+      }
+      acc += __shfl_xor(acc, 1);
+      acc += __shfl_xor(acc, 2);
+      if (part == 0 && rbase + row < B && lds.done[row])
+        buf_bootstrap[rbase + row] = acc + (bvc ? *bvc : 0.0f);
+    }
+  }
 }
 
 extern "C" void launch_rollout_step_ant(
@@ -536,7 +574,7 @@ extern "C" void launch_rollout_step_ant(
     const void* Wha, const float* bha, const void* W1c, const float* b1c,
     const void* W2c, const float* b2c, const void* Wvc, const float* bvc,
     float* buf_obs, float* buf_action, float* buf_logp, float* buf_value,
-    float* next_obs_buf, float* buf_reward, float* buf_discount,
+    float* buf_bootstrap, float* buf_reward, float* buf_discount,
     unsigned char* buf_steptype, int B, int OBS, int ACT, int HID,
     int max_episode_steps, float min_scale, float aff_scale, float aff_shift,
     float log_aff_scale, uint64_t policy_seed, uint64_t env_seed,
@@ -551,7 +589,7 @@ extern "C" void launch_rollout_step_ant(
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
                        (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
-                       buf_value, next_obs_buf, buf_reward, buf_discount,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
                        buf_steptype, B, OBS, ACT, max_episode_steps,
                        min_scale, aff_scale, aff_shift, log_aff_scale,
                        policy_seed, env_seed, policy_draw, env_draw,
@@ -563,7 +601,7 @@ extern "C" void launch_rollout_step_ant(
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
                        (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
-                       buf_value, next_obs_buf, buf_reward, buf_discount,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
                        buf_steptype, B, OBS, ACT, max_episode_steps,
                        min_scale, aff_scale, aff_shift, log_aff_scale,
                        policy_seed, env_seed, policy_draw, env_draw,

@@ -226,7 +226,7 @@ class FusedPPOEngine:
         self.draw_policy = torch.zeros(1, dtype=torch.int32, device=dev)
         self.draw_ent = torch.zeros(1, dtype=torch.int32, device=dev)
         self.buf_steptype = torch.zeros(learner.T, B, dtype=torch.uint8, device=dev)
-        self.buf_next_obs = torch.zeros(learner.T, B, self.OBS, device=dev)
+        self.vT = torch.zeros(B, device=dev)  # V(obs_T) for the last step
         self.metrics = torch.zeros(3, dtype=torch.float32, device=dev)
         self.metric_views = {
             "actor_loss": self.metrics[0],
@@ -336,19 +336,27 @@ class FusedPPOEngine:
                     c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
                     c16["Wv"], cc.views["bv"],
                     L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t],
-                    L.buf_value[t], self.buf_next_obs[t], L.buf_reward[t],
+                    L.buf_value[t], L.buf_bootstrap[t], L.buf_reward[t],
                     L.buf_discount[t], self.buf_steptype[t],
                     env.max_episode_steps, self.min_scale, self.aff_scale,
                     self.aff_shift, self.log_aff_scale, self.seed,
                     env._hip_seed, self.draw_policy, hb["draw"], t,
                 )
-            # ONE batched bootstrap pass over all T*B next-obs rows
+            # bootstrap fill: the megakernel wrote V(next_obs) for DONE
+            # rows only; for every non-done row V(next_obs) == the next
+            # step's stored value (same weights, identical observation) —
+            # one shifted masked copy + a single B-row critic pass for the
+            # final step's non-done rows
             ext.value_forward(
-                self.buf_next_obs.view(L.T * L.B, self.OBS),
-                c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
-                c16["Wv"], cc.views["bv"],
-                L.buf_bootstrap.view(L.T * L.B), self.empty, self.empty,
+                hb["obs"], c16["W1"], cc.views["b1"], c16["W2"],
+                cc.views["b2"], c16["Wv"], cc.views["bv"], self.vT,
+                self.empty, self.empty,
             )
+            done = self.buf_steptype >= 2  # TERMINATED(2) | TRUNCATED(3)
+            torch.where(done[:-1], L.buf_bootstrap[:-1], L.buf_value[1:],
+                        out=L.buf_bootstrap[:-1])
+            torch.where(done[-1], L.buf_bootstrap[-1], self.vT,
+                        out=L.buf_bootstrap[-1])
             ext.bump_add(self.draw_policy, L.T)
             ext.bump_add(hb["draw"], L.T)
             L.buf_truncated.copy_(self.buf_steptype == 3)

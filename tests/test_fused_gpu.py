@@ -471,6 +471,15 @@ def test_fused_rollout_megakernel_invariants(ext):
     # rewards/discounts sane; steptypes valid
     assert torch.isfinite(learner.buf_reward).all()
     assert ((learner.buf_discount == 0) | (learner.buf_discount == 1)).all()
+    # bootstrap identity: for non-done rows V(next_obs) == next step's value
+    done = learner.fused.buf_steptype >= 2
+    nd = ~done[:-1]
+    torch.testing.assert_close(
+        learner.buf_bootstrap[:-1][nd], learner.buf_value[1:][nd],
+        rtol=0, atol=0)
+    # for done rows the in-kernel critic supplied V(pre-reset next_obs):
+    # finite and in the value range
+    assert torch.isfinite(learner.buf_bootstrap).all()
 
 
 @requires_gpu
