@@ -89,3 +89,35 @@ def test_mcts_multistep_credit():
         temperature=0.0,
     )
     assert (out.action == 0).all(), out.action_weights
+
+
+def test_sampled_mcts_prefers_rewarding_arm():
+    """Continuous sampled search: candidates near +1 yield reward; the
+    search must concentrate visits on (and select) high-value candidates."""
+    import torch
+    from stoix_amd.search.mcts import sampled_mcts_search
+
+    B, K, AD = 6, 6, 1
+    torch.manual_seed(0)
+    # candidates: fixed spread in [-1, 1]
+    cand = torch.linspace(-1, 1, K).view(1, K, 1).repeat(B, 1, 1)
+
+    def recurrent_fn(emb, action):
+        # reward = action value; terminal-free chain with 0 value tail
+        r = action.squeeze(-1)
+        new_cand = torch.linspace(-1, 1, K).view(1, K, 1).repeat(B, 1, 1)
+        return emb, r, torch.full_like(r, 0.9), new_cand, torch.zeros_like(r)
+
+    out = sampled_mcts_search(
+        root_obs=torch.zeros(B, 1),
+        root_embedding={"h": torch.zeros(B, 1)},
+        root_candidates=cand,
+        root_value=torch.zeros(B),
+        recurrent_fn=recurrent_fn,
+        num_simulations=48,
+        temperature=0.0,
+    )
+    assert out.action_weights.shape == (B, K)
+    # best arm (action +1) must dominate the visit distribution
+    assert (out.action_weights.argmax(-1) == K - 1).float().mean() > 0.8
+    assert (out.action.squeeze(-1) > 0.5).float().mean() > 0.8

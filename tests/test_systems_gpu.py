@@ -67,3 +67,28 @@ def test_gpu_az_cartpole():
          ["env=classic/cartpole", "arch.total_num_envs=32",
           "system.rollout_length=4", "system.num_simulations=6",
           "system.num_minibatches=2", "system.epochs=1"])
+
+
+@requires_gpu
+def test_gpu_sebulba_ppo_breakout():
+    """Config #4 shape on one GPU: CPU Breakout-pixel actor threads feeding
+    a CUDA learner (inference + V-trace-free PPO update on device)."""
+    import importlib
+
+    from stoix_amd.config import compose
+
+    mod = importlib.import_module("stoix_amd.systems.ppo.sebulba_ff_ppo")
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        [
+            "env=envpool/breakout", "network=cnn",
+            "arch.total_num_envs=8", "arch.total_timesteps=null",
+            "arch.num_updates=2", "arch.num_evaluation=1",
+            "arch.num_eval_episodes=2", "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2",
+            "system.epochs=1", "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = mod.run(cfg)
+    assert r == r
