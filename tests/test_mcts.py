@@ -103,10 +103,11 @@ def test_sampled_mcts_prefers_rewarding_arm():
     cand = torch.linspace(-1, 1, K).view(1, K, 1).repeat(B, 1, 1)
 
     def recurrent_fn(emb, action):
-        # reward = action value; terminal-free chain with 0 value tail
+        # one-step bandit: reward = action value, episode ends (discount 0)
+        # so arm quality is exactly the immediate reward
         r = action.squeeze(-1)
         new_cand = torch.linspace(-1, 1, K).view(1, K, 1).repeat(B, 1, 1)
-        return emb, r, torch.full_like(r, 0.9), new_cand, torch.zeros_like(r)
+        return emb, r, torch.zeros_like(r), new_cand, torch.zeros_like(r)
 
     out = sampled_mcts_search(
         root_obs=torch.zeros(B, 1),
