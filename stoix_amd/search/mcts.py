@@ -117,8 +117,15 @@ def mcts_search(
             )
             q_edge = child_r + child_g * child_q
             span = (arena.max_q - arena.min_q).clamp(min=1e-3).unsqueeze(1)
+            # unvisited children score as the PARENT's normalised value
+            # (mctx qtransform_by_parent_and_siblings): scoring them as the
+            # global minimum lets one mediocre visited arm starve its
+            # unvisited siblings for hundreds of simulations
+            parent_q = (arena.q_value(node) - arena.min_q) / span.squeeze(1)
             q_norm = torch.where(
-                child_visit > 0, (q_edge - arena.min_q.unsqueeze(1)) / span, torch.zeros_like(q_edge)
+                child_visit > 0,
+                (q_edge - arena.min_q.unsqueeze(1)) / span,
+                parent_q.unsqueeze(1).expand_as(q_edge),
             )
             parent_visit = arena.visit[bidx, node].unsqueeze(1)
             ucb = q_norm + c_puct * arena.prior[bidx, node] * torch.sqrt(parent_visit.clamp(min=1)) / (
@@ -272,10 +279,15 @@ def sampled_mcts_search(
             )
             q_edge = child_r + child_g * child_q
             span = (arena.max_q - arena.min_q).clamp(min=1e-3).unsqueeze(1)
+            # unvisited children score as the PARENT's normalised value
+            # (mctx qtransform_by_parent_and_siblings): scoring them as the
+            # global minimum lets one mediocre visited arm starve its
+            # unvisited siblings for hundreds of simulations
+            parent_q = (arena.q_value(node) - arena.min_q) / span.squeeze(1)
             q_norm = torch.where(
                 child_visit > 0,
                 (q_edge - arena.min_q.unsqueeze(1)) / span,
-                torch.zeros_like(q_edge),
+                parent_q.unsqueeze(1).expand_as(q_edge),
             )
             parent_visit = arena.visit[bidx, node].unsqueeze(1)
             ucb = q_norm + c_puct * arena.prior[bidx, node] * torch.sqrt(
