@@ -142,7 +142,8 @@ class AZLearner:
             for i in range(n_mb):
                 idx = perm[i * mb : (i + 1) * mb]
                 dist = self.actor(flat_obs[idx])
-                ce = -(flat_w[idx] * dist.logits).sum(-1).mean()
+                log_pi = torch.log_softmax(dist.logits, dim=-1)
+                ce = -(flat_w[idx] * log_pi).sum(-1).mean()
                 entropy = dist.entropy().mean()
                 actor_loss = ce - float(self.sys.ent_coef) * entropy
                 v = self.critic(flat_obs[idx])

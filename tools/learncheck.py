@@ -51,6 +51,38 @@ def run_ppo(env_override, net, updates, num_envs, extra=(), graphs=True):
             "fused": learner.fused is not None}
 
 
+def run_dqn_cartpole(updates=80):
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.q_learning.ff_dqn import learner_factory
+    from stoix_amd.ops.graph import try_enable_update_graph
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_dqn.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=128",
+         "arch.total_timesteps=null", f"arch.num_updates={updates}",
+         "arch.num_evaluation=1", "system.rollout_length=4",
+         "system.batch_size=256", "system.buffer_size=100000",
+         "system.warmup_steps=256", "system.epochs=4", "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    device = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
+    torch.manual_seed(5)
+    env = environments.make_single(cfg, 128, device, seed=5)
+    learner = learner_factory(cfg, env, device)
+    if device.type == "cuda":
+        try_enable_update_graph(learner)
+    curve = []
+    for u in range(updates):
+        learner.update_step()
+        if (u + 1) % max(1, updates // 10) == 0:
+            m = learner.episode_metrics
+            curve.append(round(float(m.get("episode_return", torch.tensor(float("nan")))), 1))
+    return {"env": "dqn/cartpole", "curve": curve}
+
+
 def main():
     if os.environ.get("LEARNCHECK_AB"):
         for g in (False, True):
@@ -65,6 +97,9 @@ def main():
                   ("system.compute_dtype=fp32",))
     print(json.dumps(out))
     assert out["curve"][-1] > 400, f"CartPole not solved: {out['curve']}"
+    out = run_dqn_cartpole()
+    print(json.dumps(out))
+    assert max(out["curve"][-3:]) > 80, f"DQN did not learn: {out['curve']}"
     out = run_ppo("env=brax/ant", "cont", 150, 2048,
                   ("system.compute_dtype=bf16",))
     print(json.dumps(out))
