@@ -79,7 +79,7 @@ def run_dqn_cartpole(updates=80):
         learner.update_step()
         if (u + 1) % max(1, updates // 10) == 0:
             m = learner.episode_metrics
-                        r = m.get("episode_return", torch.tensor(float("nan")))
+            r = m.get("episode_return", torch.tensor(float("nan")))
             curve.append(round(float(r.float().mean()), 1))
     return {"env": "dqn/cartpole", "curve": curve}
 
