@@ -127,7 +127,9 @@ class PPOLearner:
     def _rollout(self) -> None:
         obs_raw = self.cur_obs
         for t in range(self.T):
-            obs = obs_raw
+            obs = obs_raw  # buf stores RAW obs (reference ff_ppo.py:148-162:
+            # the trajectory keeps raw observations; normalisation for the
+            # update happens post-rollout with PRE-update statistics)
             if self.normalize_obs:
                 obs = rs.normalize(obs, self.obs_stats)
             if self.amp is not None:
@@ -151,7 +153,7 @@ class PPOLearner:
             else:
                 bootstrap_value = self.critic(next_obs)
 
-            self.buf_obs[t] = obs
+            self.buf_obs[t] = obs_raw
             self.buf_action[t] = action
             self.buf_log_prob[t] = log_prob.float()
             self.buf_value[t] = value.float()
@@ -163,7 +165,12 @@ class PPOLearner:
             last_ts = next_ts
         self.cur_obs.copy_(obs_raw)
         if self.normalize_obs:
+            # normalise the stored trajectory with the PRE-update statistics,
+            # THEN update the statistics from the raw observations
+            # (reference ff_ppo.py:150-162 ordering)
+            normalized = rs.normalize(self.buf_obs, self.obs_stats)
             self.obs_stats = rs.update(self.obs_stats, self.buf_obs, all_reduce=True)
+            self.buf_obs.copy_(normalized)
         if self.collect_metrics:
             em = last_ts.extras["episode_metrics"]
             final, has = get_final_step_metrics(em)

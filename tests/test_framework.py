@@ -165,3 +165,40 @@ def test_network_factory_builds_and_runs():
         obs,
     )
     assert critic(torch.randn(5, 8)).shape == (5,)
+
+
+def test_ppo_obs_normalization_reference_semantics():
+    """Running statistics must be estimated from RAW observations (the
+    trajectory is normalised post-rollout with pre-update stats; reference
+    ff_ppo.py:148-162). A feedback loop on normalised obs would drive the
+    stats to (0,1) regardless of the data."""
+    import torch
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.ops import running_statistics as rs
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo_continuous.yaml",
+        ["env=brax/ant", "arch.total_num_envs=8", "arch.total_timesteps=null",
+         "arch.num_updates=4", "arch.num_evaluation=1",
+         "system.rollout_length=8", "system.num_minibatches=2",
+         "system.epochs=1", "system.normalize_observations=true",
+         "system.fused=false", "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    torch.manual_seed(0)
+    env = environments.make_single(cfg, 8, "cpu", seed=0)
+    learner = PPOLearner(cfg, env, torch.device("cpu"))
+    for _ in range(4):
+        learner.update_step()
+    # Ant's first obs dim is the torso z-height (~0.5, clearly non-zero):
+    # raw-data statistics must capture it
+    mean0 = float(learner.obs_stats.mean[0])
+    assert 0.2 < mean0 < 0.9, f"stats look collapsed/raw-less: mean[0]={mean0}"
+    # the stored trajectory is the NORMALISED one after rollout_phase
+    learner.rollout_phase()
+    assert float(learner.buf_obs.mean().abs()) < 1.0
