@@ -78,8 +78,8 @@ void launch_ppo_head_loss(const void*, const void*, const float*,
                           int, float, float, float, float, float, float,
                           float, uint64_t, unsigned int*, unsigned int, int,
                           void*);
-void launch_wgrad(const void*, const void*, void*, const void*, float*,
-                  long, long, long, int, int, int, int, void*);
+void launch_wgrad(const void*, const void*, float*, long, long, long, int,
+                  int, int, int, void*);
 void launch_slab_reduce(float*, void*, long, long, float*, long*, void*);
 void launch_bump_add(unsigned int*, unsigned int, void*);
 void launch_tr16_probe(const void*, float*, int, void*);
@@ -448,26 +448,9 @@ void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
   TORCH_CHECK(X.size(0) == S, "wgrad: dZ/X row mismatch");
   TORCH_CHECK(S % (4 * 32) == 0, "wgrad: S must be a multiple of 128");
   long stride = slab.size(1);
-  launch_wgrad(dZ.data_ptr(), nullptr, nullptr, X.data_ptr(),
-               slab.data_ptr<float>(), (long)dW_off, (long)db_off, stride, S,
-               N_STRIDE, K, (int)n_valid, cur_stream());
-}
-
-void wgrad_silu(torch::Tensor dH, torch::Tensor Zpre, torch::Tensor dZ_out,
-                torch::Tensor X, torch::Tensor slab, int64_t dW_off,
-                int64_t db_off, int64_t n_valid) {
-  CHK(dH, torch::kBFloat16);
-  CHK(Zpre, torch::kBFloat16);
-  CHK(X, torch::kBFloat16);
-  CHK(slab, torch::kFloat32);
-  int S = dH.size(0), N_STRIDE = dH.size(1), K = X.size(1);
-  TORCH_CHECK(X.size(0) == S, "wgrad_silu: dH/X row mismatch");
-  TORCH_CHECK(S % (4 * 32) == 0, "wgrad_silu: S must be a multiple of 128");
-  long stride = slab.size(1);
-  launch_wgrad(dH.data_ptr(), Zpre.data_ptr(), dZ_out.data_ptr(),
-               X.data_ptr(), slab.data_ptr<float>(), (long)dW_off,
-               (long)db_off, stride, S, N_STRIDE, K, (int)n_valid,
-               cur_stream());
+  launch_wgrad(dZ.data_ptr(), X.data_ptr(), slab.data_ptr<float>(),
+               (long)dW_off, (long)db_off, stride, S, N_STRIDE, K,
+               (int)n_valid, cur_stream());
 }
 
 void slab_reduce(torch::Tensor slab, torch::Tensor grad16,
@@ -520,8 +503,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_head_loss", &ppo_head_loss,
         "fused PPO head fwd + losses + analytic head bwd");
   m.def("wgrad", &wgrad, "split-K MFMA weight grad + bias colsum -> slab");
-  m.def("wgrad_silu", &wgrad_silu,
-        "wgrad with fused SiLU backward (dH,Z -> dZ inline + dZ emission)");
   m.def("slab_reduce", &slab_reduce,
         "sum wgrad slabs into flat bf16 grads (+ fused Adam prologue)");
   m.def("bump_add", &bump_add, "add N to a device RNG draw counter");

@@ -24,11 +24,6 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 #define MFMA_BF16_16x16x32 __builtin_amdgcn_mfma_f32_16x16x32_bf16
 
-DEV_INLINE float wg_silu_grad(float x) {
-  float sg = 1.0f / (1.0f + __expf(-x));
-  return sg * (1.0f + x * (1.0f - sg));
-}
-
 #define WG_SLICES 64  // max wave-slices; the launcher picks the
 // largest power of two <= 64 that divides S/32 (small minibatches use fewer)
 #define SPAD 8        // +8 cols on the 32-wide transposed tiles
@@ -44,16 +39,9 @@ struct WgradLds {
   bf16_t Xt[4][8 * 16][32 + SPAD];
 };
 
-// SILU mode fuses the activation backward into the staging pass: the
-// incoming gradient is dH and the pre-activation Z; the staged A operand
-// becomes dZ = dH * silu'(Z), and the blockIdx.y==0 workgroups also write
-// dZ out for the following dgrad GEMM (one kernel replaces
-// silu_bwd + wgrad).
-template <int KPG, int NTB, bool SILU>
+template <int KPG, int NTB>
 __launch_bounds__(256, 2) __global__ void wgrad_kernel(
-    const bf16_t* __restrict__ dZ,  // [S, N_STRIDE] (dH when SILU)
-    const bf16_t* __restrict__ Zpre,  // [S, N_STRIDE] pre-activations (SILU)
-    bf16_t* __restrict__ dZ_out,    // [S, N_STRIDE] dZ emission (SILU)
+    const bf16_t* __restrict__ dZ,  // [S, N_STRIDE]
     const bf16_t* __restrict__ X,   // [S, K]
     float* __restrict__ slab,       // [WG_SLICES, slab_stride] fp32
     long dW_off,                    // element offset of dW[N,K] in a slab
@@ -93,16 +81,6 @@ __launch_bounds__(256, 2) __global__ void wgrad_kernel(
       int srow = lane >> 1, nh = (lane & 1) * 8;
       bf16x8 v = *reinterpret_cast<const bf16x8*>(
           dZ + (long)(s0 + srow) * N_STRIDE + n0 + u * 16 + nh);
-      if (SILU) {
-        bf16x8 zp = *reinterpret_cast<const bf16x8*>(
-            Zpre + (long)(s0 + srow) * N_STRIDE + n0 + u * 16 + nh);
-#pragma unroll
-        for (int i = 0; i < 8; ++i)
-          v[i] = (bf16_t)((float)v[i] * wg_silu_grad((float)zp[i]));
-        if (blockIdx.y == 0)
-          *reinterpret_cast<bf16x8*>(dZ_out + (long)(s0 + srow) * N_STRIDE +
-                                     n0 + u * 16 + nh) = v;
-      }
 #pragma unroll
       for (int i = 0; i < 8; ++i) dZt[u * 16 + nh + i][srow] = v[i];
     }
@@ -223,12 +201,10 @@ extern "C" __global__ void tr16_probe_kernel(const bf16_t* __restrict__ in,
 
 // --------------------------------------------------------- host launchers
 
-extern "C" void launch_wgrad(const void* dZ, const void* Zpre, void* dZ_out,
-                             const void* X, float* slab,
+extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
                              long dW_off, long db_off, long slab_stride,
                              int S, int N_STRIDE, int K, int N_VALID,
                              void* stream) {
-  const int SILU = Zpre != nullptr;
   hipStream_t s = (hipStream_t)stream;
   int NT = (N_VALID + 15) / 16;
   // NTB n-tiles per workgroup (X-amplification = NT/NTB); largest KPG whose
@@ -249,31 +225,18 @@ extern "C" void launch_wgrad(const void* dZ, const void* Zpre, void* dZ_out,
   int n_slices = WG_SLICES;
   while (n_slices > 4 && (S % (n_slices * 32)) != 0) n_slices >>= 1;
   dim3 grid(NT / NTB, KTG, n_slices / 4), block(256);
-#define WGRAD_LAUNCH(KPGV, NTBV, SILUV)                                     \
-  hipLaunchKernelGGL((wgrad_kernel<KPGV, NTBV, SILUV>), grid, block, 0, s,  \
-                     (const bf16_t*)dZ, (const bf16_t*)Zpre,                \
-                     (bf16_t*)dZ_out, (const bf16_t*)X, slab, dW_off,       \
+#define WGRAD_LAUNCH(KPGV, NTBV)                                            \
+  hipLaunchKernelGGL((wgrad_kernel<KPGV, NTBV>), grid, block, 0, s,         \
+                     (const bf16_t*)dZ, (const bf16_t*)X, slab, dW_off,     \
                      db_off, slab_stride, S, N_STRIDE, K, N_VALID, n_slices)
-  if (SILU) {
-    if (NTB == 4) {
-      if (KPG == 8) WGRAD_LAUNCH(8, 4, true);
-      else if (KPG == 2) WGRAD_LAUNCH(2, 4, true);
-      else WGRAD_LAUNCH(1, 4, true);
-    } else {
-      if (KPG == 8) WGRAD_LAUNCH(8, 1, true);
-      else if (KPG == 2) WGRAD_LAUNCH(2, 1, true);
-      else WGRAD_LAUNCH(1, 1, true);
-    }
+  if (NTB == 4) {
+    if (KPG == 8) WGRAD_LAUNCH(8, 4);
+    else if (KPG == 2) WGRAD_LAUNCH(2, 4);
+    else WGRAD_LAUNCH(1, 4);
   } else {
-    if (NTB == 4) {
-      if (KPG == 8) WGRAD_LAUNCH(8, 4, false);
-      else if (KPG == 2) WGRAD_LAUNCH(2, 4, false);
-      else WGRAD_LAUNCH(1, 4, false);
-    } else {
-      if (KPG == 8) WGRAD_LAUNCH(8, 1, false);
-      else if (KPG == 2) WGRAD_LAUNCH(2, 1, false);
-      else WGRAD_LAUNCH(1, 1, false);
-    }
+    if (KPG == 8) WGRAD_LAUNCH(8, 1);
+    else if (KPG == 2) WGRAD_LAUNCH(2, 1);
+    else WGRAD_LAUNCH(1, 1);
   }
 #undef WGRAD_LAUNCH
 }

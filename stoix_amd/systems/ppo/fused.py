@@ -450,19 +450,16 @@ class FusedPPOEngine:
             ao, co = ac.offsets, cc.offsets
             ext.wgrad(self.dhead, self.H2a, ac.slab, ao["Wh"], ao["bh"], 16)
             ext.wgrad(self.dv16, self.H2c, cc.slab, co["Wv"], co["bv"], 1)
-            # wgrad_silu fuses the SiLU backward into the wgrad staging and
-            # emits dZ for the dgrad GEMM (one kernel per layer instead of
-            # silu_bwd + wgrad)
-            ext.wgrad_silu(self.dH2a, self.Z2a, self.dZ, self.H1a, ac.slab,
-                           ao["W2"], ao["b2"], self.H)
+            ext.silu_bwd(self.dH2a, self.Z2a, self.dZ)
+            ext.wgrad(self.dZ, self.H1a, ac.slab, ao["W2"], ao["b2"], self.H)
             torch.mm(self.dZ, a16["W2"], out=self.dH1)
-            ext.wgrad_silu(self.dH1, self.Z1a, self.dZ, self.Xmb, ac.slab,
-                           ao["W1"], ao["b1"], self.H)
-            ext.wgrad_silu(self.dH2c, self.Z2c, self.dZ, self.H1c, cc.slab,
-                           co["W2"], co["b2"], self.H)
+            ext.silu_bwd(self.dH1, self.Z1a, self.dZ)
+            ext.wgrad(self.dZ, self.Xmb, ac.slab, ao["W1"], ao["b1"], self.H)
+            ext.silu_bwd(self.dH2c, self.Z2c, self.dZ)
+            ext.wgrad(self.dZ, self.H1c, cc.slab, co["W2"], co["b2"], self.H)
             torch.mm(self.dZ, c16["W2"], out=self.dH1)
-            ext.wgrad_silu(self.dH1, self.Z1c, self.dZ, self.Xmb, cc.slab,
-                           co["W1"], co["b1"], self.H)
+            ext.silu_bwd(self.dH1, self.Z1c, self.dZ)
+            ext.wgrad(self.dZ, self.Xmb, cc.slab, co["W1"], co["b1"], self.H)
             ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
             ext.slab_reduce(cc.slab, cc.grad16, cc.sqnorm, cc.step_t)
             # ---- all-reduce + fused clip/Adam (+ bf16 mirror refresh)
