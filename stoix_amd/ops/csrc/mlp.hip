@@ -857,19 +857,17 @@ extern "C" __global__ void ppo_head_loss_kernel(
         log_aff_scale;
     logp_new +=
         -0.5f * z * z - __logf(sigma[j]) - 0.9189385332046727f - log_det;
-    // MC entropy sample: u' = loc + sigma*eps' (one philox block serves
-    // 4 dims: 4 uniforms -> 4 normals via two Box-Muller pairs)
-    if ((j & 3) == 0) {
-      Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 2 + (j >> 2)), draw);
-      box_muller(uu.a, uu.b, &eps_e[j], &eps_e[j + 1]);
-      box_muller(uu.c, uu.d, &eps_e[j + 2], &eps_e[j + 3]);
-    }
-    u_e[j] = loc[j] + sigma[j] * eps_e[j];
+    // MC entropy sample: u' = loc + sigma*eps'
+    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 8 + j), draw);
+    float n1, n2;
+    box_muller(uu.a, uu.b, &n1, &n2);
+    eps_e[j] = n1;
+    u_e[j] = loc[j] + sigma[j] * n1;
     float log_det_e =
         2.0f * (0.6931471805599453f - u_e[j] - softplus_f(-2.0f * u_e[j])) +
         log_aff_scale;
-    ent -= -0.5f * eps_e[j] * eps_e[j] - __logf(sigma[j]) -
-           0.9189385332046727f - log_det_e;
+    ent -= -0.5f * n1 * n1 - __logf(sigma[j]) - 0.9189385332046727f -
+           log_det_e;
   }
 
   // ---- PPO clip loss
@@ -1071,9 +1069,7 @@ extern "C" void launch_ppo_head_loss(
     unsigned int* draw_buf, unsigned int draw_offset, int do_bump,
     void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  // 128-thread blocks double the workgroup count (>=256 for the flagship
-  // minibatch) so the per-row transcendental chain spreads over every CU
-  int threads = 128;
+  int threads = 256;
   int blocks = (B + threads - 1) / threads;
   hipLaunchKernelGGL(ppo_head_loss_kernel, dim3(blocks), dim3(threads), 0, s,
                      (const bf16_t*)heads, (const bf16_t*)v_in, action,
