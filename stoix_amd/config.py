@@ -15,7 +15,6 @@ framework needs:
 """
 from __future__ import annotations
 
-import copy
 import os
 from pathlib import Path
 from typing import Any, Dict, Iterable, List, Optional

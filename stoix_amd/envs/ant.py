@@ -28,7 +28,7 @@ registers.
 from __future__ import annotations
 
 import math
-from typing import Dict, Tuple
+from typing import Tuple
 
 import torch
 

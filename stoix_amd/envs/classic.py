@@ -9,7 +9,7 @@ the textbook definitions, batched over torch tensors.
 from __future__ import annotations
 
 import math
-from typing import Dict, Tuple
+from typing import Tuple
 
 import torch
 

@@ -21,7 +21,6 @@ kernel-side, one env per lane).
 """
 from __future__ import annotations
 
-import math
 from typing import Tuple
 
 import torch

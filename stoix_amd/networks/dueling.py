@@ -3,7 +3,7 @@ DuelingQNetwork :15-47, DistributionalDuelingQNetwork :50-87,
 NoisyDistributionalDuelingQNetwork :90-124)."""
 from __future__ import annotations
 
-from typing import NamedTuple, Sequence
+from typing import Sequence
 
 import torch
 import torch.nn as nn

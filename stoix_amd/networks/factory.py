@@ -9,7 +9,7 @@ unlike Flax's lazy shape inference).
 """
 from __future__ import annotations
 
-from typing import Any, Optional
+from typing import Optional
 
 import torch.nn as nn
 

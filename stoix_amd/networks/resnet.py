@@ -7,7 +7,6 @@ from typing import List, Sequence
 
 import torch
 import torch.nn as nn
-import torch.nn.functional as F
 
 from stoix_amd.networks.torso import MLPTorso, orthogonal_init
 from stoix_amd.networks.utils import get_activation

@@ -2,7 +2,7 @@
 MLPTorso :12-33, NoisyMLPTorso :36-57, CNNTorso :60-108)."""
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import List, Sequence
 
 import torch
 import torch.nn as nn

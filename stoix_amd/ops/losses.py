@@ -14,7 +14,7 @@ these references (tests/test_losses.py).
 """
 from __future__ import annotations
 
-from typing import Optional, Tuple
+from typing import Tuple
 
 import torch
 import torch.nn.functional as F

@@ -18,12 +18,11 @@ from __future__ import annotations
 
 import argparse
 import itertools
-import os
 import shutil
 import subprocess
 import sys
 from pathlib import Path
-from typing import Dict, List
+from typing import List
 
 
 SBATCH_TEMPLATE = """#!/bin/bash
