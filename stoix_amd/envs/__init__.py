@@ -36,8 +36,9 @@ def _classic(name: str):
 
 def _brax(name: str):
     from stoix_amd.envs.humanoid import Humanoid
+    from stoix_amd.envs.planar import HalfCheetah, Hopper
 
-    table = {"ant": Ant, "humanoid": Humanoid}
+    table = {"ant": Ant, "humanoid": Humanoid, "halfcheetah": HalfCheetah, "hopper": Hopper}
     if name not in table:
         raise ValueError(f"unknown brax-suite env '{name}' (have {list(table)})")
     return table[name]

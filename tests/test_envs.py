@@ -223,3 +223,37 @@ def test_breakout_plays():
     assert torch.isfinite(ts.observation).all()
     # ball bouncing into the brick band should have scored something
     assert float(total_r.sum()) > 0
+
+
+def test_halfcheetah_steps_no_early_termination():
+    from stoix_amd.envs.planar import HalfCheetah
+
+    env = HalfCheetah(num_envs=8, device="cpu", seed=2)
+    ts = env.reset()
+    assert ts.observation.shape == (8, 17)
+    g = torch.Generator().manual_seed(1)
+    for _ in range(40):
+        a = torch.rand(8, 6, generator=g) * 2 - 1
+        ts = env.step(a)
+        assert torch.isfinite(ts.observation).all()
+        assert torch.isfinite(ts.reward).all()
+        # cheetah never terminates early (MuJoCo contract) -> discount stays 1
+        assert (ts.discount == 1.0).all()
+
+
+def test_hopper_steps_and_terminates():
+    from stoix_amd.envs.planar import Hopper
+
+    env = Hopper(num_envs=16, device="cpu", seed=4)
+    ts = env.reset()
+    assert ts.observation.shape == (16, 11)
+    g = torch.Generator().manual_seed(2)
+    term_seen = False
+    for _ in range(120):
+        a = torch.rand(16, 3, generator=g) * 2 - 1
+        ts = env.step(a)
+        assert torch.isfinite(ts.observation).all()
+        assert torch.isfinite(ts.reward).all()
+        if bool(ts.extras["episode_metrics"]["is_terminal_step"].any()):
+            term_seen = True
+    assert term_seen, "random-torque hopper must fall over within 120 steps"
