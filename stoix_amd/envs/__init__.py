@@ -3,7 +3,8 @@
 Parity with /root/reference/stoix/utils/make_env.py: an ``ENV_MAKERS``
 dispatch over suites (:420-433) and ``make(config) -> (train_env, eval_env)``
 (:436-466). Suites available offline: ``classic`` (gymnax-equivalent
-CartPole/Pendulum/MountainCar/Acrobot), ``brax`` (Ant-class physics), and
+CartPole/Pendulum/MountainCar/Acrobot), ``brax`` (Ant/Humanoid/HalfCheetah/
+Hopper-class physics), and
 ``debug`` (five diagnostic games). External-binding suites (jumanji, envpool
 Atari, ...) require packages absent from this image and raise a clear error.
 """
