@@ -46,9 +46,16 @@ def _brax(name: str):
 
 
 def _jumanji(name: str):
+    from stoix_amd.envs.game2048 import Game2048
     from stoix_amd.envs.snake import Snake
 
-    table = {"snake": Snake, "Snake-v1": Snake}
+    table = {
+        "snake": Snake,
+        "Snake-v1": Snake,
+        "game_2048": Game2048,
+        "2048": Game2048,
+        "Game2048-v1": Game2048,
+    }
     if name not in table:
         raise ValueError(f"unknown jumanji-suite env '{name}' (have {list(table)})")
     return table[name]

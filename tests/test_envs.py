@@ -257,3 +257,42 @@ def test_hopper_steps_and_terminates():
         if bool(ts.extras["episode_metrics"]["is_terminal_step"].any()):
             term_seen = True
     assert term_seen, "random-torque hopper must fall over within 120 steps"
+
+
+def test_game2048_slide_merge_rules():
+    from stoix_amd.envs.game2048 import Game2048
+
+    env = Game2048(num_envs=1, device="cpu", seed=0)
+    b = torch.tensor([[[1, 1, 1, 1],
+                       [2, 0, 2, 0],
+                       [0, 3, 0, 0],
+                       [1, 2, 2, 3]]], dtype=torch.int32)
+    nb, r = env._slide_left(b)
+    # [2,2,2,2]->[4,4,.,.]; [4,0,4,0]->[8,.,.,.]; [.,8,.,.]->[8,.,.,.];
+    # [2,4,4,8]->[2,8,8] (only the middle pair merges)
+    assert nb[0].tolist() == [[2, 2, 0, 0], [3, 0, 0, 0], [3, 0, 0, 0], [1, 3, 3, 0]]
+    # merged tile values: 4+4 + 8 + 8 = 24
+    assert float(r[0]) == 24.0
+    # no chain merge: [2,2,4] must give [4,4], not [8]
+    b2 = torch.tensor([[[1, 1, 2, 0]] * 4], dtype=torch.int32)
+    nb2, _ = env._slide_left(b2)
+    assert nb2[0, 0].tolist() == [2, 2, 0, 0]
+
+
+def test_game2048_steps_and_spawns():
+    from stoix_amd.envs.game2048 import Game2048
+
+    env = Game2048(num_envs=8, device="cpu", seed=3)
+    ts = env.reset()
+    assert ts.observation.shape == (8, 4, 4, 16)
+    # fresh boards carry exactly two tiles
+    assert (ts.observation[..., 0].sum(dim=(1, 2)) == 14).all()
+    g = torch.Generator().manual_seed(0)
+    total_r = torch.zeros(8)
+    for _ in range(60):
+        a = torch.randint(0, 4, (8,), generator=g)
+        ts = env.step(a)
+        total_r += ts.reward
+        assert torch.isfinite(ts.reward).all()
+    # an hour of random play earns merge reward on most boards
+    assert (total_r > 0).sum() >= 6
