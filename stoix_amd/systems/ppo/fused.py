@@ -57,15 +57,37 @@ def _mlp_linears(torso: nn.Module) -> Optional[List[nn.Linear]]:
 
 
 class _Chain:
-    """One flat fp32 master + bf16 mirror + bf16 grad + Adam state."""
+    """One flat fp32 master + bf16 mirror + bf16 grad + Adam state.
 
-    def __init__(self, specs: List[Tuple[str, Tuple[int, ...]]], device, lr: float):
+    ``flat16``/``grad16`` may be preallocated slices of a buffer shared
+    with the sibling chain: the engine lays actor and critic out
+    back-to-back (actor ends with W2, critic starts with W2) so the two
+    W2 blocks are ADJACENT — a zero-copy [2, H, H] view batches the dH1
+    backward GEMMs into one bmm — and the whole gradient goes out in ONE
+    RCCL all-reduce. Adam/clip state stays per chain (separate lr and
+    per-network global-norm clip, matching the reference's per-network
+    optax chains)."""
+
+    def __init__(
+        self,
+        specs: List[Tuple[str, Tuple[int, ...]]],
+        device,
+        lr: float,
+        flat16: Optional[Tensor] = None,
+        grad16: Optional[Tensor] = None,
+    ):
         self.lr = lr
         sizes = [int(torch.prod(torch.tensor(s)).item()) for _, s in specs]
         total = sum(sizes)
         self.flat = torch.zeros(total, dtype=torch.float32, device=device)
-        self.flat16 = torch.zeros(total, dtype=torch.bfloat16, device=device)
-        self.grad16 = torch.zeros(total, dtype=torch.bfloat16, device=device)
+        self.flat16 = (
+            flat16 if flat16 is not None
+            else torch.zeros(total, dtype=torch.bfloat16, device=device)
+        )
+        self.grad16 = (
+            grad16 if grad16 is not None
+            else torch.zeros(total, dtype=torch.bfloat16, device=device)
+        )
         self.m = torch.zeros(total, dtype=torch.float32, device=device)
         self.v = torch.zeros(total, dtype=torch.float32, device=device)
         self.sqnorm = torch.zeros(1, dtype=torch.float32, device=device)
@@ -192,30 +214,38 @@ class FusedPPOEngine:
 
         dev = self.device
         K1P = self.K1P
+        # layout: actor ENDS with W2, critic STARTS with W2 -> adjacent in
+        # the shared bf16 buffers (see _Chain docstring)
+        a_specs = [
+            ("W1", (H, K1P)),
+            ("b1", (H,)),
+            ("Wh", (16, H)),
+            ("bh", (16,)),
+            ("b2", (H,)),
+            ("W2", (H, H)),
+        ]
+        c_specs = [
+            ("W2", (H, H)),
+            ("W1", (H, K1P)),
+            ("b1", (H,)),
+            ("b2", (H,)),
+            ("Wv", (H,)),
+            ("bv", (1,)),
+        ]
+        nA = sum(int(torch.prod(torch.tensor(s)).item()) for _, s in a_specs)
+        nC = sum(int(torch.prod(torch.tensor(s)).item()) for _, s in c_specs)
+        self.big16 = torch.zeros(nA + nC, dtype=torch.bfloat16, device=dev)
+        self.biggrad16 = torch.zeros(nA + nC, dtype=torch.bfloat16, device=dev)
         self.actor_chain = _Chain(
-            [
-                ("W1", (H, K1P)),
-                ("b1", (H,)),
-                ("W2", (H, H)),
-                ("b2", (H,)),
-                ("Wh", (16, H)),
-                ("bh", (16,)),
-            ],
-            dev,
-            float(sysc.actor_lr),
+            a_specs, dev, float(sysc.actor_lr),
+            flat16=self.big16[:nA], grad16=self.biggrad16[:nA],
         )
         self.critic_chain = _Chain(
-            [
-                ("W1", (H, K1P)),
-                ("b1", (H,)),
-                ("W2", (H, H)),
-                ("b2", (H,)),
-                ("Wv", (H,)),
-                ("bv", (1,)),
-            ],
-            dev,
-            float(sysc.critic_lr),
+            c_specs, dev, float(sysc.critic_lr),
+            flat16=self.big16[nA:], grad16=self.biggrad16[nA:],
         )
+        # zero-copy [2, H, H] view over [actor W2 | critic W2]
+        self.W2pair = self.big16[nA - H * H : nA + H * H].view(2, H, H)
         self._adopt_params(a_lins, c_lins, head, learner.critic.critic_head)
         self.actor_chain.sync_mirror()
         self.critic_chain.sync_mirror()
@@ -244,18 +274,25 @@ class FusedPPOEngine:
         self.val_mb = z(S, dtype=torch.float32)
         self.adv_mb = z(S, dtype=torch.float32)
         self.tgt_mb = z(S, dtype=torch.float32)
-        self.Z1a, self.H1a = z(S, H), z(S, H)
-        self.Z2a, self.H2a = z(S, H), z(S, H)
-        self.Z1c, self.H1c = z(S, H), z(S, H)
-        self.Z2c, self.H2c = z(S, H), z(S, H)
-        self.dH2a, self.dH2c = z(S, H), z(S, H)
-        self.dZ = z(S, H)  # scratch for silu_bwd outputs
-        self.dH1 = z(S, H)
+        # stacked [net, S, H] activation buffers: index 0 = actor, 1 =
+        # critic; lets silu_bwd run ONCE over both nets and dH1 run as one
+        # batched bmm against W2pair
+        self.Z1, self.H1 = z(2, S, H), z(2, S, H)
+        self.Z2, self.H2 = z(2, S, H), z(2, S, H)
+        self.dH2 = z(2, S, H)
+        self.dZ2 = z(2, S, H)
+        self.dH1 = z(2, S, H)
+        self.dZ1 = z(2, S, H)
         self.heads = z(S, 16)
         self.vpred = z(S, 1)
         self.dhead = z(S, 16)
         self.dv = z(S, 1)
         self.dv16 = z(S, 16)  # col 0 = dv (wgrad A-operand; cols 1-15 zero)
+
+        import os as _os
+
+        # batched dH1 GEMM (one bmm vs two mm); A/B knob for measurement
+        self.use_bmm = _os.environ.get("STOIX_FUSED_BMM", "1") != "0"
 
         import torch.distributed as dist
 
@@ -424,14 +461,14 @@ class FusedPPOEngine:
             )
             # ---- forward: custom fused Linear+SiLU MFMA kernels (one
             # launch per layer instead of GemmAndBias + silu)
-            ext.linear_silu(self.Xmb, a16["W1"], ac.views["b1"], self.Z1a, self.H1a, 1)
-            ext.linear_silu(self.H1a, a16["W2"], ac.views["b2"], self.Z2a, self.H2a, 1)
-            ext.linear_silu(self.Xmb, c16["W1"], cc.views["b1"], self.Z1c, self.H1c, 1)
-            ext.linear_silu(self.H1c, c16["W2"], cc.views["b2"], self.Z2c, self.H2c, 1)
+            ext.linear_silu(self.Xmb, a16["W1"], ac.views["b1"], self.Z1[0], self.H1[0], 1)
+            ext.linear_silu(self.H1[0], a16["W2"], ac.views["b2"], self.Z2[0], self.H2[0], 1)
+            ext.linear_silu(self.Xmb, c16["W1"], cc.views["b1"], self.Z1[1], self.H1[1], 1)
+            ext.linear_silu(self.H1[1], c16["W2"], cc.views["b2"], self.Z2[1], self.H2[1], 1)
             # ---- heads as GEMMs (hipBLASLt), then the fused per-row
             # loss + analytic head-backward kernel, then dH2 as GEMMs
-            torch.addmm(a16["bh"], self.H2a, a16["Wh"].t(), out=self.heads)
-            torch.addmm(c16["bv"], self.H2c, c16["Wv"].view(self.H, 1),
+            torch.addmm(a16["bh"], self.H2[0], a16["Wh"].t(), out=self.heads)
+            torch.addmm(c16["bv"], self.H2[1], c16["Wv"].view(self.H, 1),
                         out=self.vpred)
             self.metrics.zero_()
             ext.ppo_head_loss(
@@ -441,31 +478,34 @@ class FusedPPOEngine:
                 self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
                 self.log_aff_scale, self.seed, self.draw_ent, mb, 0,
             )
-            torch.mm(self.dhead, a16["Wh"], out=self.dH2a)
-            torch.mm(self.dv, c16["Wv"].view(1, self.H), out=self.dH2c)
+            torch.mm(self.dhead, a16["Wh"], out=self.dH2[0])
+            torch.mm(self.dv, c16["Wv"].view(1, self.H), out=self.dH2[1])
             # ---- backward: wgrads + bias colsums via the split-K MFMA
             # wgrad kernel (wgrad.hip; hipBLASLt NT at K=32768 is ~3x
             # slower), partial sums land in the chain slab, one
-            # slab_reduce per chain folds them into the flat bf16 grads
+            # slab_reduce per chain folds them into the flat bf16 grads.
+            # silu_bwd runs once over the stacked [2, S, H] buffers and the
+            # two dH1 GEMMs go as one bmm against the adjacent-W2 view.
             ao, co = ac.offsets, cc.offsets
-            ext.wgrad(self.dhead, self.H2a, ac.slab, ao["Wh"], ao["bh"], 16)
-            ext.wgrad(self.dv16, self.H2c, cc.slab, co["Wv"], co["bv"], 1)
-            ext.silu_bwd(self.dH2a, self.Z2a, self.dZ)
-            ext.wgrad(self.dZ, self.H1a, ac.slab, ao["W2"], ao["b2"], self.H)
-            torch.mm(self.dZ, a16["W2"], out=self.dH1)
-            ext.silu_bwd(self.dH1, self.Z1a, self.dZ)
-            ext.wgrad(self.dZ, self.Xmb, ac.slab, ao["W1"], ao["b1"], self.H)
-            ext.silu_bwd(self.dH2c, self.Z2c, self.dZ)
-            ext.wgrad(self.dZ, self.H1c, cc.slab, co["W2"], co["b2"], self.H)
-            torch.mm(self.dZ, c16["W2"], out=self.dH1)
-            ext.silu_bwd(self.dH1, self.Z1c, self.dZ)
-            ext.wgrad(self.dZ, self.Xmb, cc.slab, co["W1"], co["b1"], self.H)
+            ext.wgrad(self.dhead, self.H2[0], ac.slab, ao["Wh"], ao["bh"], 16)
+            ext.wgrad(self.dv16, self.H2[1], cc.slab, co["Wv"], co["bv"], 1)
+            ext.silu_bwd(self.dH2, self.Z2, self.dZ2)
+            ext.wgrad(self.dZ2[0], self.H1[0], ac.slab, ao["W2"], ao["b2"], self.H)
+            ext.wgrad(self.dZ2[1], self.H1[1], cc.slab, co["W2"], co["b2"], self.H)
+            if self.use_bmm:
+                torch.bmm(self.dZ2, self.W2pair, out=self.dH1)
+            else:
+                torch.mm(self.dZ2[0], a16["W2"], out=self.dH1[0])
+                torch.mm(self.dZ2[1], c16["W2"], out=self.dH1[1])
+            ext.silu_bwd(self.dH1, self.Z1, self.dZ1)
+            ext.wgrad(self.dZ1[0], self.Xmb, ac.slab, ao["W1"], ao["b1"], self.H)
+            ext.wgrad(self.dZ1[1], self.Xmb, cc.slab, co["W1"], co["b1"], self.H)
             ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
             ext.slab_reduce(cc.slab, cc.grad16, cc.sqnorm, cc.step_t)
-            # ---- all-reduce + fused clip/Adam (+ bf16 mirror refresh)
+            # ---- ONE all-reduce over both chains' shared grad buffer,
+            # then fused clip/Adam per chain (+ bf16 mirror refresh)
             if self.world > 1:
-                dist.all_reduce(ac.grad16)
-                dist.all_reduce(cc.grad16)
+                dist.all_reduce(self.biggrad16)
             gscale = 1.0 / float(self.world)
             ext.fused_adam_bf16(
                 ac.flat, ac.grad16, ac.m, ac.v, ac.sqnorm, ac.step_t,

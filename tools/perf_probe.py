@@ -112,19 +112,19 @@ def main():
                        F.val_mb, F.adv_mb, F.tgt_mb, F.empty, F.empty)
 
     def k_gemm_fwd():
-        torch.addmm(a16["b1"], F.Xmb, a16["W1"].t(), out=F.Z1a)
+        torch.addmm(a16["b1"], F.Xmb, a16["W1"].t(), out=F.Z1[0])
 
     def k_gemm_fwd2():
-        torch.addmm(a16["b2"], F.H1a, a16["W2"].t(), out=F.Z2a)
+        torch.addmm(a16["b2"], F.H1[0], a16["W2"].t(), out=F.Z2[0])
 
     def k_silu():
-        ext.silu_fwd(F.Z1a, F.H1a)
+        ext.silu_fwd(F.Z1[0], F.H1[0])
 
     def k_linsilu():
-        ext.linear_silu(F.H1a, a16["W2"], ac.views["b2"], F.Z2a, F.H2a, 1)
+        ext.linear_silu(F.H1[0], a16["W2"], ac.views["b2"], F.Z2[0], F.H2[0], 1)
 
     def k_headgemm():
-        torch.addmm(a16["bh"], F.H2a, a16["Wh"].t(), out=F.heads)
+        torch.addmm(a16["bh"], F.H2[0], a16["Wh"].t(), out=F.heads)
 
     def k_head():
         F.metrics.zero_()
@@ -135,19 +135,25 @@ def main():
             F.log_aff_scale, F.seed, F.draw_ent, 0, 1)
 
     def k_wgrad():
-        torch.mm(F.dZ.t(), F.H1a, out=ac.gviews16["W2"])
+        torch.mm(F.dZ2[0].t(), F.H1[0], out=ac.gviews16["W2"])
 
     def k_wgrad_custom():
-        ext.wgrad(F.dZ, F.H1a, ac.slab, ac.offsets["W2"], ac.offsets["b2"], F.H)
+        ext.wgrad(F.dZ2[0], F.H1[0], ac.slab, ac.offsets["W2"], ac.offsets["b2"], F.H)
 
     def k_slab_reduce():
         ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
 
     def k_dgrad():
-        torch.mm(F.dZ, a16["W2"], out=F.dH1)
+        torch.mm(F.dZ2[0], a16["W2"], out=F.dH1[0])
+
+    def k_dgrad_bmm():
+        torch.bmm(F.dZ2, F.W2pair, out=F.dH1)
+
+    def k_silu_bwd_stacked():
+        ext.silu_bwd(F.dH2, F.Z2, F.dZ2)
 
     def k_bsum():
-        torch.sum(F.dZ, 0, out=ac.gviews16["b2"])
+        torch.sum(F.dZ2[0], 0, out=ac.gviews16["b2"])
 
     def k_adam():
         ext.fused_adam_bf16(ac.flat, ac.grad16, ac.m, ac.v, ac.sqnorm,
@@ -165,6 +171,8 @@ def main():
     print(f"wgrad custom:       {timeit(k_wgrad_custom, 50)*1e3:8.1f} us")
     print(f"slab_reduce:        {timeit(k_slab_reduce, 50)*1e3:8.1f} us")
     print(f"dgrad mm:           {timeit(k_dgrad, 50)*1e3:8.1f} us")
+    print(f"dgrad bmm stacked:  {timeit(k_dgrad_bmm, 50)*1e3:8.1f} us")
+    print(f"silu_bwd stacked:   {timeit(k_silu_bwd_stacked, 50)*1e3:8.1f} us")
     print(f"bias colsum:        {timeit(k_bsum, 50)*1e3:8.1f} us")
     print(f"fused_adam_bf16:    {timeit(k_adam, 50)*1e3:8.1f} us")
 
