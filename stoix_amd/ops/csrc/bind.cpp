@@ -428,11 +428,12 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
                          ? (unsigned int*)draw_buf.data_ptr<int>()
                          : nullptr;
   void* dv16p = dv16.numel() > 0 ? dv16.data_ptr() : nullptr;
+  float* mp = metrics.numel() > 0 ? metrics.data_ptr<float>() : nullptr;
   launch_ppo_head_loss(
       heads.data_ptr(), v_in.data_ptr(), action.data_ptr<float>(),
       old_logp.data_ptr<float>(), old_value.data_ptr<float>(),
       adv.data_ptr<float>(), targets.data_ptr<float>(), dhead.data_ptr(),
-      dv.data_ptr(), dv16p, metrics.data_ptr<float>(), B, ACT,
+      dv.data_ptr(), dv16p, mp, B, ACT,
       (float)clip_eps,
       (float)ent_coef, (float)vf_coef, (float)min_scale, (float)aff_scale,
       (float)aff_shift, (float)log_aff_scale, (uint64_t)seed, db,

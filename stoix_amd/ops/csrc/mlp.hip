@@ -47,6 +47,22 @@ DEV_INLINE float softplus_f(float x) {
   // log1p(exp(x)) stable
   return (x > 20.0f) ? x : log1pf(__expf(x));
 }
+// Fast-math variants for the per-row loss kernel: built from the
+// hardware-accelerated __expf/__logf so the 8-dim unrolled loop pipelines
+// instead of serialising on branchy libm calls (atanhf/tanhf/log1pf).
+DEV_INLINE float softplus_fast(float x) {
+  // |err| vs log1pf path is ~1 ulp of __logf; below x=-15 the result is
+  // ~e^x < 3e-7, negligible against min_scale
+  return (x > 20.0f) ? x : __logf(1.0f + __expf(x));
+}
+DEV_INLINE float atanh_fast(float y) {
+  // y is pre-clamped to +-(1 - 1e-3)
+  return 0.5f * __logf((1.0f + y) / (1.0f - y));
+}
+DEV_INLINE float tanh_fast(float x) {
+  // saturates correctly: __expf overflows to +inf for large 2x -> 1.0
+  return 1.0f - 2.0f / (__expf(2.0f * x) + 1.0f);
+}
 
 // ---------------------------------------------------------------- probe
 // Validates the assumed lane->element mapping of v_mfma_f32_16x16x32_bf16.
@@ -833,40 +849,65 @@ extern "C" __global__ void ppo_head_loss_kernel(
 
   // ---- per-dim tanh-normal forward
   float loc[8], spre[8], sigma[8], u[8], eps_e[8], u_e[8];
+  float a_row[8];
   if (active) {
     const bf16x8* h8 = reinterpret_cast<const bf16x8*>(heads + (long)row * 16);
     bf16x8 hl = h8[0], hs = h8[1];
+    // issue the action-row loads early (two float4 vectors when ACT fills
+    // them); at 0.5 resident waves/SIMD there is no other wave to hide
+    // their latency, so overlap them with the Philox/softplus ALU below
+    const float* arow_p = action + (long)row * ACT;
+#pragma unroll
+    for (int j4 = 0; j4 < 8; j4 += 4) {
+      if (j4 < ACT && (ACT & 3) == 0) {
+        const f32x4 av = *reinterpret_cast<const f32x4*>(arow_p + j4);
+        a_row[j4 + 0] = av[0]; a_row[j4 + 1] = av[1];
+        a_row[j4 + 2] = av[2]; a_row[j4 + 3] = av[3];
+      }
+    }
+    if ((ACT & 3) != 0) {
+      for (int j = 0; j < ACT; ++j) a_row[j] = arow_p[j];
+    }
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       loc[j] = bf2f(hl[j]);
       spre[j] = bf2f(hs[j]);
-      sigma[j] = softplus_f(spre[j]) + min_scale;
+      sigma[j] = softplus_fast(spre[j]) + min_scale;
     }
   }
   float logp_new = 0.0f, ent = 0.0f;
+  // one Philox call serves 4 dims (4 uniforms -> 4 normals via two
+  // box_muller pairs) instead of one call per dim discarding half its draws
+  float nrm[8];
+#pragma unroll
+  for (int q = 0; q < 2; ++q) {
+    if (q * 4 >= ACT || !active) break;
+    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 2 + q), draw);
+    box_muller(uu.a, uu.b, &nrm[q * 4 + 0], &nrm[q * 4 + 1]);
+    box_muller(uu.c, uu.d, &nrm[q * 4 + 2], &nrm[q * 4 + 3]);
+  }
 #pragma unroll
   for (int j = 0; j < 8; ++j) {
     if (j >= ACT || !active) break;
-    float a = action[(long)row * ACT + j];
+    float a = a_row[j];
     float y = (a - aff_shift) / aff_scale;
     y = fmaxf(-1.0f + 1e-3f, fminf(1.0f - 1e-3f, y));
-    u[j] = atanhf(y);
+    u[j] = atanh_fast(y);
     float z = (u[j] - loc[j]) / sigma[j];
+    float log_sigma = __logf(sigma[j]);
     float log_det =
-        2.0f * (0.6931471805599453f - u[j] - softplus_f(-2.0f * u[j])) +
+        2.0f * (0.6931471805599453f - u[j] - softplus_fast(-2.0f * u[j])) +
         log_aff_scale;
     logp_new +=
-        -0.5f * z * z - __logf(sigma[j]) - 0.9189385332046727f - log_det;
+        -0.5f * z * z - log_sigma - 0.9189385332046727f - log_det;
     // MC entropy sample: u' = loc + sigma*eps'
-    Rng4 uu = philox_uniform4(seed, 3u, (uint32_t)(row * 8 + j), draw);
-    float n1, n2;
-    box_muller(uu.a, uu.b, &n1, &n2);
+    float n1 = nrm[j];
     eps_e[j] = n1;
     u_e[j] = loc[j] + sigma[j] * n1;
     float log_det_e =
-        2.0f * (0.6931471805599453f - u_e[j] - softplus_f(-2.0f * u_e[j])) +
+        2.0f * (0.6931471805599453f - u_e[j] - softplus_fast(-2.0f * u_e[j])) +
         log_aff_scale;
-    ent -= -0.5f * n1 * n1 - __logf(sigma[j]) - 0.9189385332046727f -
+    ent -= -0.5f * n1 * n1 - log_sigma - 0.9189385332046727f -
            log_det_e;
   }
 
@@ -912,7 +953,7 @@ extern "C" __global__ void ppo_head_loss_kernel(
       float z = (u[j] - loc[j]) / sigma[j];
       dloc = dl_dlogp * (z / sigma[j]);
       float dsig = dl_dlogp * ((z * z - 1.0f) / sigma[j]);
-      float th = tanhf(u_e[j]);
+      float th = tanh_fast(u_e[j]);
       dloc += ce * (-2.0f * th);
       dsig += ce * (1.0f / sigma[j] - 2.0f * th * eps_e[j]);
       dspre = dsig * (1.0f / (1.0f + __expf(-spre[j])));
@@ -928,20 +969,25 @@ extern "C" __global__ void ppo_head_loss_kernel(
     if (dv16_out) dv16_out[(long)row * 16] = f2bf(dv);
   }
 
-  // ---- loss metrics (wave-level pre-reduce, one atomic per wave)
-  float m0 = active ? a_loss * inv_B : 0.0f;
-  float m1 = active ? v_loss * inv_B : 0.0f;
-  float m2 = active ? ent * inv_B : 0.0f;
+  // ---- loss metrics (wave-level pre-reduce, one atomic per wave).
+  // Optional: 512 waves funnelling atomics into the same 3 words
+  // serialise across all XCDs, so the engine requests metrics only on
+  // the minibatch it actually reports (the last one).
+  if (metrics) {
+    float m0 = active ? a_loss * inv_B : 0.0f;
+    float m1 = active ? v_loss * inv_B : 0.0f;
+    float m2 = active ? ent * inv_B : 0.0f;
 #pragma unroll
-  for (int off = 32; off > 0; off >>= 1) {
-    m0 += __shfl_down(m0, off);
-    m1 += __shfl_down(m1, off);
-    m2 += __shfl_down(m2, off);
-  }
-  if ((threadIdx.x & 63) == 0) {
-    atomicAdd(&metrics[0], m0);
-    atomicAdd(&metrics[1], m1);
-    atomicAdd(&metrics[2], m2);
+    for (int off = 32; off > 0; off >>= 1) {
+      m0 += __shfl_down(m0, off);
+      m1 += __shfl_down(m1, off);
+      m2 += __shfl_down(m2, off);
+    }
+    if ((threadIdx.x & 63) == 0) {
+      atomicAdd(&metrics[0], m0);
+      atomicAdd(&metrics[1], m1);
+      atomicAdd(&metrics[2], m2);
+    }
   }
 }
 

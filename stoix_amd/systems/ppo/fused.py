@@ -258,6 +258,7 @@ class FusedPPOEngine:
         self.buf_steptype = torch.zeros(learner.T, B, dtype=torch.uint8, device=dev)
         self.vT = torch.zeros(B, device=dev)  # V(obs_T) for the last step
         self.metrics = torch.zeros(3, dtype=torch.float32, device=dev)
+        self.metrics_none = torch.zeros(0, dtype=torch.float32, device=dev)
         self.metric_views = {
             "actor_loss": self.metrics[0],
             "value_loss": self.metrics[1],
@@ -470,11 +471,16 @@ class FusedPPOEngine:
             torch.addmm(a16["bh"], self.H2[0], a16["Wh"].t(), out=self.heads)
             torch.addmm(c16["bv"], self.H2[1], c16["Wv"].view(self.H, 1),
                         out=self.vpred)
-            self.metrics.zero_()
+            # metrics only on the LAST minibatch (the one epoch() reports):
+            # 512 waves of atomics into 3 words serialise across XCDs
+            last = mb == n_mb - 1
+            if last:
+                self.metrics.zero_()
             ext.ppo_head_loss(
                 self.heads, self.vpred.view(-1), self.act_mb, self.logp_mb,
                 self.val_mb, self.adv_mb, self.tgt_mb, self.dhead, self.dv,
-                self.dv16, self.metrics, self.clip_eps, self.ent_coef,
+                self.dv16, self.metrics if last else self.metrics_none,
+                self.clip_eps, self.ent_coef,
                 self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
                 self.log_aff_scale, self.seed, self.draw_ent, mb, 0,
             )
