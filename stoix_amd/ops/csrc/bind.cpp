@@ -79,7 +79,7 @@ void launch_ppo_head_loss(const void*, const void*, const float*,
                           float, uint64_t, unsigned int*, unsigned int, int,
                           void*);
 void launch_wgrad(const void*, const void*, float*, long, long, long, int,
-                  int, int, int, void*);
+                  int, int, int, int, int, void*);
 void launch_slab_reduce(float*, void*, long, long, float*, long*, void*);
 void launch_bump_add(unsigned int*, unsigned int, void*);
 void launch_tr16_probe(const void*, float*, int, void*);
@@ -441,7 +441,8 @@ void ppo_head_loss(torch::Tensor heads, torch::Tensor v_in,
 }
 
 void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
-           int64_t dW_off, int64_t db_off, int64_t n_valid) {
+           int64_t dW_off, int64_t db_off, int64_t n_valid,
+           int64_t kpg = 0, int64_t ntb = 0) {
   CHK(dZ, torch::kBFloat16);
   CHK(X, torch::kBFloat16);
   CHK(slab, torch::kFloat32);
@@ -451,7 +452,7 @@ void wgrad(torch::Tensor dZ, torch::Tensor X, torch::Tensor slab,
   long stride = slab.size(1);
   launch_wgrad(dZ.data_ptr(), X.data_ptr(), slab.data_ptr<float>(),
                (long)dW_off, (long)db_off, stride, S, N_STRIDE, K,
-               (int)n_valid, cur_stream());
+               (int)n_valid, (int)kpg, (int)ntb, cur_stream());
 }
 
 void slab_reduce(torch::Tensor slab, torch::Tensor grad16,
@@ -503,7 +504,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ppo_gather", &ppo_gather, "fused minibatch gather");
   m.def("ppo_head_loss", &ppo_head_loss,
         "fused PPO head fwd + losses + analytic head bwd");
-  m.def("wgrad", &wgrad, "split-K MFMA weight grad + bias colsum -> slab");
+  m.def("wgrad", &wgrad, "split-K MFMA weight grad + bias colsum -> slab",
+        py::arg("dZ"), py::arg("X"), py::arg("slab"), py::arg("dW_off"),
+        py::arg("db_off"), py::arg("n_valid"), py::arg("kpg") = 0,
+        py::arg("ntb") = 0);
   m.def("slab_reduce", &slab_reduce,
         "sum wgrad slabs into flat bf16 grads (+ fused Adam prologue)");
   m.def("bump_add", &bump_add, "add N to a device RNG draw counter");

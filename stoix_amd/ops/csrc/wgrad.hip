@@ -204,22 +204,29 @@ extern "C" __global__ void tr16_probe_kernel(const bf16_t* __restrict__ in,
 extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
                              long dW_off, long db_off, long slab_stride,
                              int S, int N_STRIDE, int K, int N_VALID,
+                             int kpg_override, int ntb_override,
                              void* stream) {
   hipStream_t s = (hipStream_t)stream;
   int NT = (N_VALID + 15) / 16;
-  // NTB n-tiles per workgroup (X-amplification = NT/NTB); largest KPG whose
-  // grid still spans >= 16 workgroups (small-N head grads would otherwise
-  // run on 8 CUs of 256)
+  // NTB n-tiles per workgroup (X-amplification = NT/NTB); KPG picked per
+  // shape so the grid spans the whole 256-CU chip (swept on MI355X,
+  // tools/wgrad_sweep.py): fill beats per-WG tile depth at these shapes.
   int NTB = (NT % 4 == 0) ? 4 : 1;
   int KPG = 1;
-  const int cands[3] = {8, 2, 1};
-  for (int ci = 0; ci < 3; ++ci) {
-    int cand = cands[ci];
-    if (K % (16 * cand) == 0 &&
-        (cand == 1 || (NT / NTB) * (K / (16 * cand)) >= 2)) {
-      KPG = cand;
-      break;
-    }
+  if (ntb_override > 0) NTB = ntb_override;
+  if (kpg_override > 0) {
+    KPG = kpg_override;
+  } else if (NT <= 1) {
+    // head grads (N<=16): swept best is KPG=4 x 16 slices = 64 WGs
+    // (per-WG k-depth beats chip fill at this tiny N)
+    KPG = (K % 64 == 0) ? 4 : ((K % 32 == 0) ? 2 : 1);
+  } else if (K % 64 == 0 && (NT / NTB) * (K / 64) >= 16) {
+    // big-K layer grads: KPG=4 puts exactly 256 WGs on the 256 CUs
+    // (KPG=8 halves the grid; KPG<=2 doubles it past the fill point) --
+    // 27.4 -> 20.6 us for the 256x256/S=32768 shape (tools/wgrad_sweep.py)
+    KPG = 4;
+  } else if (K % 32 == 0) {
+    KPG = 2;
   }
   int KTG = K / (16 * KPG);
   int n_slices = WG_SLICES;
@@ -231,10 +238,12 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
                      db_off, slab_stride, S, N_STRIDE, K, N_VALID, n_slices)
   if (NTB == 4) {
     if (KPG == 8) WGRAD_LAUNCH(8, 4);
+    else if (KPG == 4) WGRAD_LAUNCH(4, 4);
     else if (KPG == 2) WGRAD_LAUNCH(2, 4);
     else WGRAD_LAUNCH(1, 4);
   } else {
     if (KPG == 8) WGRAD_LAUNCH(8, 1);
+    else if (KPG == 4) WGRAD_LAUNCH(4, 1);
     else if (KPG == 2) WGRAD_LAUNCH(2, 1);
     else WGRAD_LAUNCH(1, 1);
   }
