@@ -460,12 +460,16 @@ class FusedPPOEngine:
                 flat_tgt, self.Xmb, self.act_mb, self.logp_mb, self.val_mb,
                 self.adv_mb, self.tgt_mb, self.empty, self.empty,
             )
-            # ---- forward: custom fused Linear+SiLU MFMA kernels (one
-            # launch per layer instead of GemmAndBias + silu)
+            # ---- forward. Layer 1 (K=32): the custom fused Linear+SiLU
+            # MFMA kernel wins (14.4 us vs 11.2 GEMM + silu pass). Layer 2
+            # (K=256): hipBLASLt's tuned GemmAndBias (14.4 us) beats the
+            # custom kernel (24.6 us), so run both nets' GEMMs then ONE
+            # silu over the stacked [2, S, H] preacts.
             ext.linear_silu(self.Xmb, a16["W1"], ac.views["b1"], self.Z1[0], self.H1[0], 1)
-            ext.linear_silu(self.H1[0], a16["W2"], ac.views["b2"], self.Z2[0], self.H2[0], 1)
             ext.linear_silu(self.Xmb, c16["W1"], cc.views["b1"], self.Z1[1], self.H1[1], 1)
-            ext.linear_silu(self.H1[1], c16["W2"], cc.views["b2"], self.Z2[1], self.H2[1], 1)
+            torch.addmm(a16["b2"], self.H1[0], a16["W2"].t(), out=self.Z2[0])
+            torch.addmm(c16["b2"], self.H1[1], c16["W2"].t(), out=self.Z2[1])
+            ext.silu_fwd(self.Z2, self.H2)
             # ---- heads as GEMMs (hipBLASLt), then the fused per-row
             # loss + analytic head-backward kernel, then dH2 as GEMMs
             torch.addmm(a16["bh"], self.H2[0], a16["Wh"].t(), out=self.heads)
