@@ -15,6 +15,7 @@
 // contiguous ds_read_b128 fragment reads) and issues KPG
 // mfma_f32_16x16x32_bf16 per s-step.
 #include "common.h"
+#include <cstdlib>
 #include <hip/hip_bf16.h>
 
 typedef __bf16 bf16_t;
@@ -230,6 +231,10 @@ extern "C" void launch_wgrad(const void* dZ, const void* X, float* slab,
   }
   int KTG = K / (16 * KPG);
   int n_slices = WG_SLICES;
+  if (const char* e = getenv("STOIX_WGRAD_SLICES")) {
+    int v = atoi(e);
+    if (v >= 4 && v <= WG_SLICES) n_slices = v;
+  }
   while (n_slices > 4 && (S % (n_slices * 32)) != 0) n_slices >>= 1;
   dim3 grid(NT / NTB, KTG, n_slices / 4), block(256);
 #define WGRAD_LAUNCH(KPGV, NTBV)                                            \

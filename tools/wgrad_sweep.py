@@ -35,8 +35,11 @@ def main():
     for name, NV, K, NSTRIDE in shapes:
         dZ = torch.randn(S, NSTRIDE, device=dev, dtype=bf)
         X = torch.randn(S, K, device=dev, dtype=bf)
-        numel = NV * K + NV
-        slab = torch.zeros(64, numel, device=dev, dtype=torch.float32)
+        # realistic slab: the engine's chain slab has stride ~74.5K floats
+        # and this dW sits at a mid-chain offset
+        stride = 74512
+        off = 8448
+        slab = torch.zeros(64, stride, device=dev, dtype=torch.float32)
         NT = (NV + 15) // 16
         results = []
         for kpg in (1, 2, 4, 8):
@@ -46,7 +49,7 @@ def main():
                 if NT % ntb != 0:
                     continue
                 wgs = (NT // ntb) * (K // (16 * kpg)) * 16
-                t = timeit(lambda: ext.wgrad(dZ, X, slab, 0, NV * K, NV, kpg, ntb))
+                t = timeit(lambda: ext.wgrad(dZ, X, slab, off, off + NV * K, NV, kpg, ntb))
                 results.append((t, kpg, ntb, wgs))
         results.sort()
         best = results[0]
