@@ -23,12 +23,19 @@ __all__ = ["make", "make_factory", "StatefulVecEnv", "get_final_step_metrics", "
 
 
 def _classic(name: str):
+    from stoix_amd.envs.minatar import Freeway, SpaceInvaders
+
     table = {
         "CartPole-v1": CartPole,
         "Pendulum-v1": Pendulum,
         "MountainCar-v0": MountainCar,
         "MountainCarContinuous-v0": lambda **kw: MountainCar(continuous=True, **kw),
         "Acrobot-v1": Acrobot,
+        # MinAtar-class grid games (the reference's gymnax suite)
+        "Freeway-MinAtar": Freeway,
+        "freeway": Freeway,
+        "SpaceInvaders-MinAtar": SpaceInvaders,
+        "space_invaders": SpaceInvaders,
     }
     if name not in table:
         raise ValueError(f"unknown classic env '{name}' (have {list(table)})")

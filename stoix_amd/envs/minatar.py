@@ -1,0 +1,229 @@
+"""MinAtar-class 10x10 grid games: Freeway and SpaceInvaders (vectorised).
+
+Fill the roles of gymnax's MinAtar suite in the reference's configs
+(/root/reference/stoix/configs/env/gymnax/{freeway,space_invaders}.yaml;
+SURVEY §8.8 — breakout is covered by envs/breakout.py). gymnax/MinAtar are
+JAX/numpy-only; these are original, fully tensorised torch implementations
+(every rule batched over B boards) with the MinAtar contracts: 10x10
+multi-channel binary observations, small discrete action sets, +1-per-event
+rewards.
+
+Freeway: the chicken starts at the bottom (row 9, col 4) and must cross 8
+lanes of traffic (rows 1-8, random per-episode direction and speed). Reaching
+the top row earns +1 and resets the chicken; a car hit knocks it back to the
+start (no negative reward). Fixed 2500-step episodes (MinAtar timer).
+
+SpaceInvaders: a cannon on the bottom row moves left/right and fires; a 4x6
+alien grid marches across and down, accelerating as it thins; aliens drop
+bombs. +1 per alien destroyed; terminates when a bomb hits the cannon or the
+aliens reach the cannon row; a cleared wave respawns faster.
+"""
+from __future__ import annotations
+
+from typing import Tuple
+
+import torch
+
+from stoix_amd.envs.env import State, StatefulVecEnv, Tensor
+from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
+
+G = 10  # grid side
+
+
+class Freeway(StatefulVecEnv):
+    max_episode_steps = 2500
+    CHICKEN_COL = 4
+    MOVE_COOLDOWN = 3
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((G, G, 4), -1.0, 1.0)
+        self.action_space = DiscreteSpace(3)  # noop / up / down
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "chicken": torch.full((n,), G - 1, dtype=torch.long, device=dev),
+            "cooldown": torch.zeros(n, dtype=torch.long, device=dev),
+            # 8 lanes on rows 1..8
+            "car_pos": self.randint(G, n, 8),
+            "car_dir": self.randint(2, n, 8) * 2 - 1,
+            "car_period": self.randint(4, n, 8) + 1,  # 1..4
+            "car_timer": torch.zeros(n, 8, dtype=torch.long, device=dev),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        n = state["chicken"].shape[0]
+        dev = self.device
+        obs = torch.zeros(n, G, G, 4, device=dev)
+        bidx = torch.arange(n, device=dev)
+        obs[bidx, state["chicken"], self.CHICKEN_COL, 0] = 1.0
+        lanes = torch.arange(1, 9, device=dev)  # car rows
+        bb = bidx.unsqueeze(1).expand(-1, 8)
+        ll = lanes.unsqueeze(0).expand(n, -1)
+        obs[bb, ll, state["car_pos"], 1] = 1.0
+        obs[bb, ll, state["car_pos"], 2] = state["car_dir"].float()
+        obs[bb, ll, state["car_pos"], 3] = state["car_period"].float() / 4.0
+        return obs
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long()
+        chicken = state["chicken"]
+        cooldown = (state["cooldown"] - 1).clamp(min=0)
+        can_move = cooldown == 0
+        move = torch.where(a == 1, -1, torch.where(a == 2, 1, 0))
+        moved = can_move & (move != 0)
+        chicken = (chicken + torch.where(moved, move, 0)).clamp(0, G - 1)
+        cooldown = torch.where(moved, torch.full_like(cooldown, self.MOVE_COOLDOWN), cooldown)
+
+        timer = state["car_timer"] + 1
+        advance = timer >= state["car_period"]
+        pos = (state["car_pos"] + torch.where(advance, state["car_dir"], 0)) % G
+        timer = torch.where(advance, torch.zeros_like(timer), timer)
+
+        # collision: a car in the chicken's lane (rows 1..8) at its column
+        in_lane = (chicken >= 1) & (chicken <= 8)
+        lane_idx = (chicken - 1).clamp(0, 7)
+        bidx = torch.arange(chicken.shape[0], device=self.device)
+        hit = in_lane & (pos[bidx, lane_idx] == self.CHICKEN_COL)
+        scored = chicken == 0
+        reward = scored.float()
+        chicken = torch.where(hit | scored, torch.full_like(chicken, G - 1), chicken)
+
+        terminated = torch.zeros_like(scored)  # time-limit only (MinAtar timer)
+        return (
+            {
+                "chicken": chicken,
+                "cooldown": cooldown,
+                "car_pos": pos,
+                "car_dir": state["car_dir"],
+                "car_period": state["car_period"],
+                "car_timer": timer,
+            },
+            reward,
+            terminated,
+        )
+
+
+class SpaceInvaders(StatefulVecEnv):
+    max_episode_steps = 1000
+    ROWS_A, COLS_A = 4, 6  # alien grid
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((G, G, 4), 0.0, 1.0)
+        self.action_space = DiscreteSpace(4)  # noop / left / right / fire
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "cannon": torch.full((n,), G // 2, dtype=torch.long, device=dev),
+            "alive": torch.ones(n, self.ROWS_A, self.COLS_A, dtype=torch.bool, device=dev),
+            "ax": torch.full((n,), 2, dtype=torch.long, device=dev),  # left edge
+            "ay": torch.zeros(n, dtype=torch.long, device=dev),
+            "adir": torch.ones(n, dtype=torch.long, device=dev),
+            "atimer": torch.zeros(n, dtype=torch.long, device=dev),
+            "wave": torch.zeros(n, dtype=torch.long, device=dev),
+            # friendly shot (moves up) and alien bomb (moves down); y<0 or
+            # y>=G means inactive
+            "fx": torch.zeros(n, dtype=torch.long, device=dev),
+            "fy": torch.full((n,), -1, dtype=torch.long, device=dev),
+            "ex": torch.zeros(n, dtype=torch.long, device=dev),
+            "ey": torch.full((n,), G, dtype=torch.long, device=dev),
+        }
+
+    def _alien_cells(self, state: State) -> Tuple[Tensor, Tensor, Tensor]:
+        """(rows [B,4,6], cols [B,4,6], alive mask) of alien world cells."""
+        r = torch.arange(self.ROWS_A, device=self.device).view(1, -1, 1)
+        c = torch.arange(self.COLS_A, device=self.device).view(1, 1, -1)
+        rows = state["ay"].view(-1, 1, 1) + r
+        cols = state["ax"].view(-1, 1, 1) + c
+        return rows, cols, state["alive"]
+
+    def _obs_fn(self, state: State) -> Tensor:
+        n = state["cannon"].shape[0]
+        dev = self.device
+        obs = torch.zeros(n, G, G, 4, device=dev)
+        bidx = torch.arange(n, device=dev)
+        obs[bidx, G - 1, state["cannon"], 0] = 1.0
+        rows, cols, alive = self._alien_cells(state)
+        rr = rows.clamp(0, G - 1)
+        cc = cols.clamp(0, G - 1)
+        ali = torch.zeros(n, G * G, device=dev)
+        ali.scatter_add_(1, (rr * G + cc).reshape(n, -1), alive.float().reshape(n, -1))
+        obs[:, :, :, 1] = ali.view(n, G, G).clamp(max=1.0)
+        f_act = (state["fy"] >= 0) & (state["fy"] < G)
+        obs[bidx[f_act], state["fy"][f_act], state["fx"][f_act], 2] = 1.0
+        e_act = (state["ey"] >= 0) & (state["ey"] < G)
+        obs[bidx[e_act], state["ey"][e_act], state["ex"][e_act], 3] = 1.0
+        return obs
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        dev = self.device
+        a = action.long()
+        n = a.shape[0]
+        bidx = torch.arange(n, device=dev)
+        s = {k: v.clone() for k, v in state.items()}
+
+        s["cannon"] = (s["cannon"] + torch.where(a == 1, -1, 0) + torch.where(a == 2, 1, 0)).clamp(0, G - 1)
+        # fire: only when no friendly shot in flight
+        fire = (a == 3) & (s["fy"] < 0)
+        s["fx"] = torch.where(fire, s["cannon"], s["fx"])
+        s["fy"] = torch.where(fire, torch.full_like(s["fy"], G - 2), s["fy"])
+
+        # move shots
+        s["fy"] = torch.where(s["fy"] >= 0, s["fy"] - 1, s["fy"])
+        s["ey"] = torch.where(s["ey"] < G, s["ey"] + 1, s["ey"])
+
+        # friendly shot vs aliens
+        rows, cols, alive = self._alien_cells(s)
+        f_act = s["fy"] >= 0
+        hit = (
+            alive
+            & f_act.view(-1, 1, 1)
+            & (rows == s["fy"].view(-1, 1, 1))
+            & (cols == s["fx"].view(-1, 1, 1))
+        )
+        any_hit = hit.any(dim=(-1, -2))
+        reward = any_hit.float()
+        s["alive"] = s["alive"] & ~hit
+        s["fy"] = torch.where(any_hit, torch.full_like(s["fy"], -1), s["fy"])
+
+        # alien march: period shrinks as the wave thins / waves advance
+        n_alive = s["alive"].sum(dim=(-1, -2))
+        period = (1 + n_alive // 8 - s["wave"]).clamp(min=1)
+        s["atimer"] = s["atimer"] + 1
+        do_move = s["atimer"] >= period
+        s["atimer"] = torch.where(do_move, torch.zeros_like(s["atimer"]), s["atimer"])
+        nx = s["ax"] + torch.where(do_move, s["adir"], 0)
+        at_wall = (nx < 0) | (nx > G - self.COLS_A)
+        s["adir"] = torch.where(do_move & at_wall, -s["adir"], s["adir"])
+        s["ay"] = s["ay"] + (do_move & at_wall).long()
+        s["ax"] = torch.where(do_move & ~at_wall, nx, s["ax"])
+
+        # alien bomb: lowest alive alien in a uniformly chosen alive column
+        e_free = s["ey"] >= G
+        col_has = s["alive"].any(dim=1)  # [B, 6]
+        probs = col_has.float().clamp(min=1e-9)
+        shooter_col = torch.multinomial(probs, 1, generator=self.gen).squeeze(-1)
+        col_alive = s["alive"][bidx, :, shooter_col]  # [B, 4]
+        # lowest alive row index in that column
+        row_idx = torch.arange(self.ROWS_A, device=dev).view(1, -1)
+        low_row = torch.where(col_alive, row_idx, torch.full_like(row_idx, -1)).max(dim=1).values
+        can_bomb = e_free & col_has.any(dim=-1) & (low_row >= 0)
+        s["ex"] = torch.where(can_bomb, (s["ax"] + shooter_col).clamp(0, G - 1), s["ex"])
+        s["ey"] = torch.where(can_bomb, (s["ay"] + low_row + 1).clamp(0, G - 1), s["ey"])
+
+        # terminal: bomb reaches the cannon row at its column, or aliens low
+        bomb_hit = (s["ey"] == G - 1) & (s["ex"] == s["cannon"])
+        rows2, _, alive2 = self._alien_cells(s)
+        aliens_low = (alive2 & (rows2 >= G - 1)).any(dim=(-1, -2))
+        terminated = bomb_hit | aliens_low
+
+        # cleared wave -> respawn, one step faster
+        cleared = s["alive"].sum(dim=(-1, -2)) == 0
+        s["alive"] = torch.where(cleared.view(-1, 1, 1), torch.ones_like(s["alive"]), s["alive"])
+        s["ax"] = torch.where(cleared, torch.full_like(s["ax"], 2), s["ax"])
+        s["ay"] = torch.where(cleared, torch.zeros_like(s["ay"]), s["ay"])
+        s["wave"] = s["wave"] + cleared.long()
+        return s, reward, terminated
