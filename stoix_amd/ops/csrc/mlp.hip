@@ -312,7 +312,7 @@ __launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
       float logp = 0.0f;
       float a_val = 0.0f;
       if (col < 8) {
-        float sigma = softplus_f(spre) + min_scale;
+        float sigma = softplus_fast(spre) + min_scale;
         // 4 normals for this (row-block, dim) from one philox block
         int rowblk = (rbase + g * 4) >> 2;  // global row / 4
         Rng4 u = philox_uniform4(seed, 2u, (uint32_t)(rowblk * 16 + col), draw);
@@ -321,10 +321,10 @@ __launch_bounds__(256, 2) __global__ void policy_value_step_kernel(
         box_muller(u.c, u.d, &n[2], &n[3]);
         float eps = greedy ? 0.0f : n[r];
         float uu = loc + sigma * eps;
-        a_val = tanhf(uu) * aff_scale + aff_shift;
+        a_val = tanh_fast(uu) * aff_scale + aff_shift;
         // log N(u;loc,sigma) - log|d a/d u|
         float log_det =
-            2.0f * (0.6931471805599453f - uu - softplus_f(-2.0f * uu)) +
+            2.0f * (0.6931471805599453f - uu - softplus_fast(-2.0f * uu)) +
             log_aff_scale;
         logp = -0.5f * eps * eps - __logf(sigma) -
                0.9189385332046727f - log_det;
@@ -468,7 +468,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       float logp = 0.0f;
       float a_val = 0.0f;
       if (col < 8) {
-        float sigma = softplus_f(spre) + min_scale;
+        float sigma = softplus_fast(spre) + min_scale;
         int rowblk = (rbase + g * 4) >> 2;
         Rng4 u = philox_uniform4(policy_seed, 2u, (uint32_t)(rowblk * 16 + col),
                                  pdraw);
@@ -477,9 +477,9 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
         box_muller(u.c, u.d, &n[2], &n[3]);
         float eps = n[r];
         float uu = loc + sigma * eps;
-        a_val = tanhf(uu) * aff_scale + aff_shift;
+        a_val = tanh_fast(uu) * aff_scale + aff_shift;
         float log_det =
-            2.0f * (0.6931471805599453f - uu - softplus_f(-2.0f * uu)) +
+            2.0f * (0.6931471805599453f - uu - softplus_fast(-2.0f * uu)) +
             log_aff_scale;
         logp = -0.5f * eps * eps - __logf(sigma) - 0.9189385332046727f -
                log_det;
