@@ -1,5 +1,6 @@
 """Environment-contract tests: the load-bearing autoreset/truncation
 semantics of SURVEY.md §8.7, plus per-env sanity."""
+import pytest
 import torch
 
 from stoix_amd.envs.ant import Ant
@@ -352,3 +353,28 @@ def test_space_invaders_terminates_eventually():
             term = True
             break
     assert term, "random play must eventually get bombed or overrun"
+
+
+@pytest.mark.gpu
+def test_new_envs_step_on_gpu():
+    """Planar/2048/MinAtar envs run the generic torch path as pure device
+    tensor work (no HIP kernel tier for these; the registry must still build
+    and step them on cuda without host round-trips erroring)."""
+    from stoix_amd.envs.game2048 import Game2048
+    from stoix_amd.envs.minatar import Freeway, SpaceInvaders
+    from stoix_amd.envs.planar import HalfCheetah, Hopper
+
+    dev = "cuda:0"
+    for cls, act in [
+        (HalfCheetah, lambda g: torch.rand(8, 6, device=dev) * 2 - 1),
+        (Hopper, lambda g: torch.rand(8, 3, device=dev) * 2 - 1),
+        (Game2048, lambda g: torch.randint(0, 4, (8,), device=dev)),
+        (Freeway, lambda g: torch.randint(0, 3, (8,), device=dev)),
+        (SpaceInvaders, lambda g: torch.randint(0, 4, (8,), device=dev)),
+    ]:
+        env = cls(num_envs=8, device=dev, seed=0)
+        ts = env.reset()
+        for _ in range(10):
+            ts = env.step(act(None))
+        assert ts.observation.is_cuda
+        assert torch.isfinite(ts.reward).all()
