@@ -71,7 +71,17 @@ def _jumanji(name: str):
 def _envpool(name: str):
     from stoix_amd.envs.breakout import Breakout
 
-    table = {"breakout": Breakout, "Breakout-v5": Breakout}
+    def breakout_maker(**kw):
+        # envpool parity: prefer the native C++ batched CPU engine for
+        # CPU actor threads (Sebulba); torch-ops path on CUDA / unbuilt ext
+        if torch.device(kw.get("device", "cpu")).type == "cpu":
+            from stoix_amd.envs.envpool_cpu import BreakoutCpu, envpool_ext
+
+            if envpool_ext() is not None:
+                return BreakoutCpu(**kw)
+        return Breakout(**kw)
+
+    table = {"breakout": breakout_maker, "Breakout-v5": breakout_maker}
     if name not in table:
         raise ValueError(f"unknown envpool-suite env '{name}' (have {list(table)})")
     return table[name]

@@ -1,0 +1,206 @@
+// Native batched CPU environment engine (envpool-equivalent).
+//
+// The reference's envpool suite is a C++ batched env pool feeding Sebulba's
+// CPU actor threads (/root/reference/stoix/utils/make_env.py envpool rows;
+// SURVEY.md §2.2/§8.6). This is the MI355X-framework counterpart: the whole
+// per-step pipeline — physics, termination/truncation, episode metrics,
+// autoreset, and FRAME RENDERING — runs fused in one C++ call, parallelised
+// over envs with at::parallel_for. The torch-ops Breakout
+// (stoix_amd/envs/breakout.py) renders via an einsum over 72 brick masks
+// (~0.5 MFLOP/env/step plus ~30 kernel dispatches); this path draws the
+// ~1.1k changed pixels directly (~10 us/env single-thread), which is what
+// makes CPU actors keep a GPU learner fed (BASELINE config #4).
+//
+// Game rules mirror stoix_amd/envs/breakout.py exactly (84x84 grayscale,
+// 4 actions, +1/brick, terminate on miss or clear).
+#include <torch/extension.h>
+#include <ATen/Parallel.h>
+
+#include <cmath>
+#include <cstdint>
+
+namespace {
+
+constexpr int H = 84, W = 84;
+constexpr int BRICK_ROWS = 6, BRICK_COLS = 12;
+constexpr int BRICK_W = W / BRICK_COLS, BRICK_H = 3;
+constexpr int BRICK_TOP = 12;
+constexpr int PADDLE_W = 12, PADDLE_Y = 80;
+constexpr float PADDLE_SPEED = 3.0f, BALL_SPEED = 1.8f;
+constexpr int NBRICK = BRICK_ROWS * BRICK_COLS;
+// state row: paddle_x, ball_x, ball_y, ball_vx, ball_vy, pad, bricks[72]
+constexpr int SOFF = 6;
+constexpr int SDIM = SOFF + NBRICK;
+
+constexpr uint8_t ST_MID = 1, ST_TERMINATED = 2, ST_TRUNCATED = 3;
+
+// counter-based RNG (splitmix64) -> uniform [0,1)
+inline float hash_uniform(uint64_t seed, uint64_t a, uint64_t b) {
+  uint64_t x = seed + 0x9e3779b97f4a7c15ull * (a + 1) + 0xbf58476d1ce4e5b9ull * (b + 1);
+  x ^= x >> 30; x *= 0xbf58476d1ce4e5b9ull;
+  x ^= x >> 27; x *= 0x94d049bb133111ebull;
+  x ^= x >> 31;
+  return (float)(x >> 40) * (1.0f / 16777216.0f);
+}
+
+inline void reset_env(float* s, uint64_t seed, uint64_t env, uint64_t draw) {
+  s[0] = W / 2.0f;
+  s[1] = W * 0.3f + hash_uniform(seed, env * 2, draw) * (W * 0.4f);
+  s[2] = 46.0f;
+  s[3] = (hash_uniform(seed, env * 2 + 1, draw) > 0.5f) ? BALL_SPEED * 0.7f
+                                                        : -BALL_SPEED * 0.7f;
+  s[4] = BALL_SPEED;
+  s[5] = 0.0f;
+  for (int k = 0; k < NBRICK; ++k) s[SOFF + k] = 1.0f;
+}
+
+inline void render(const float* s, float* obs) {
+  std::memset(obs, 0, sizeof(float) * H * W);
+  // bricks (value 0.6; each brick leaves a 1px gutter right/bottom)
+  for (int r = 0; r < BRICK_ROWS; ++r) {
+    int y0 = BRICK_TOP + r * BRICK_H;
+    for (int c = 0; c < BRICK_COLS; ++c) {
+      if (s[SOFF + r * BRICK_COLS + c] <= 0.0f) continue;
+      int x0 = c * BRICK_W;
+      for (int y = y0; y < y0 + BRICK_H - 1; ++y)
+        for (int x = x0; x < x0 + BRICK_W - 1; ++x) obs[y * W + x] = 0.6f;
+    }
+  }
+  // paddle (2 rows)
+  int px = (int)s[0];
+  if (px < PADDLE_W / 2) px = PADDLE_W / 2;
+  if (px > W - 1 - PADDLE_W / 2) px = W - 1 - PADDLE_W / 2;
+  for (int dx = -(PADDLE_W / 2); dx < PADDLE_W / 2; ++dx) {
+    int x = px + dx;
+    if (x < 0) x = 0;
+    if (x > W - 1) x = W - 1;
+    obs[PADDLE_Y * W + x] = 1.0f;
+    obs[(PADDLE_Y + 1) * W + x] = 1.0f;
+  }
+  // ball (2x2)
+  int by = (int)s[2], bx = (int)s[1];
+  if (by < 0) by = 0;
+  if (by > H - 2) by = H - 2;
+  if (bx < 0) bx = 0;
+  if (bx > W - 2) bx = W - 2;
+  for (int dy = 0; dy < 2; ++dy)
+    for (int dx = 0; dx < 2; ++dx) obs[(by + dy) * W + bx + dx] = 1.0f;
+}
+
+}  // namespace
+
+void breakout_cpu_reset(torch::Tensor state, torch::Tensor obs,
+                        int64_t seed, int64_t draw) {
+  TORCH_CHECK(state.size(1) == SDIM, "state must be [B, ", SDIM, "]");
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  float* op = obs.data_ptr<float>();
+  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+    for (int64_t b = lo; b < hi; ++b) {
+      reset_env(sp + b * SDIM, (uint64_t)seed, (uint64_t)b, (uint64_t)draw);
+      render(sp + b * SDIM, op + b * H * W);
+    }
+  });
+}
+
+void breakout_cpu_step(torch::Tensor state, torch::Tensor action,
+                       torch::Tensor step_count, torch::Tensor ep_return,
+                       torch::Tensor ep_length, torch::Tensor last_ep_return,
+                       torch::Tensor last_ep_length, torch::Tensor obs,
+                       torch::Tensor next_obs, torch::Tensor reward,
+                       torch::Tensor discount, torch::Tensor steptype,
+                       torch::Tensor done, int64_t max_episode_steps,
+                       int64_t seed, torch::Tensor draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  const int64_t* ap = action.data_ptr<int64_t>();
+  int32_t* scp = step_count.data_ptr<int32_t>();
+  float* erp = ep_return.data_ptr<float>();
+  int32_t* elp = ep_length.data_ptr<int32_t>();
+  float* lerp = last_ep_return.data_ptr<float>();
+  int32_t* lelp = last_ep_length.data_ptr<int32_t>();
+  float* op = obs.data_ptr<float>();
+  float* nop = next_obs.data_ptr<float>();
+  float* rp = reward.data_ptr<float>();
+  float* dp = discount.data_ptr<float>();
+  uint8_t* stp = steptype.data_ptr<uint8_t>();
+  uint8_t* dnp = done.data_ptr<uint8_t>();
+  int32_t dr = *draw.data_ptr<int32_t>();
+
+  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+    for (int64_t b = lo; b < hi; ++b) {
+      float* s = sp + b * SDIM;
+      int a = (int)ap[b];
+      if (a < 0) a = 0;
+      if (a > 3) a = 3;
+      float px = s[0] + PADDLE_SPEED * ((a == 2) - (a == 3));
+      if (px < PADDLE_W / 2.0f) px = PADDLE_W / 2.0f;
+      if (px > W - PADDLE_W / 2.0f) px = W - PADDLE_W / 2.0f;
+      float bx = s[1] + s[3], by = s[2] + s[4];
+      float vx = s[3], vy = s[4];
+      if (bx < 1.0f || bx > W - 2.0f) vx = -vx;
+      if (bx < 1.0f) bx = 1.0f;
+      if (bx > W - 2.0f) bx = W - 2.0f;
+      if (by < 1.0f) { vy = std::fabs(vy); by = 1.0f; }
+      bool on_paddle = by >= PADDLE_Y - 1 && by <= PADDLE_Y + 1 &&
+                       std::fabs(bx - px) <= PADDLE_W / 2.0f && vy > 0;
+      if (on_paddle) {
+        vx += 0.4f * (bx - px) / (PADDLE_W / 2.0f);
+        vy = -std::fabs(vy);
+      }
+      float rew = 0.0f;
+      if (by >= BRICK_TOP && by < BRICK_TOP + BRICK_ROWS * BRICK_H) {
+        int br = (int)((by - BRICK_TOP) / BRICK_H);
+        if (br < 0) br = 0;
+        if (br > BRICK_ROWS - 1) br = BRICK_ROWS - 1;
+        int bc = (int)(bx / BRICK_W);
+        if (bc < 0) bc = 0;
+        if (bc > BRICK_COLS - 1) bc = BRICK_COLS - 1;
+        int k = br * BRICK_COLS + bc;
+        if (s[SOFF + k] > 0.0f) {
+          s[SOFF + k] = 0.0f;
+          vy = -vy;
+          rew = 1.0f;
+        }
+      }
+      if (vx > 2.5f) vx = 2.5f;
+      if (vx < -2.5f) vx = -2.5f;
+      bool missed = by > H - 2.0f;
+      float bricks_left = 0.0f;
+      for (int k = 0; k < NBRICK; ++k) bricks_left += s[SOFF + k];
+      bool terminated = missed || bricks_left <= 0.0f;
+      s[0] = px; s[1] = bx;
+      s[2] = by < 0.0f ? 0.0f : (by > H - 1.0f ? (float)(H - 1) : by);
+      s[3] = vx; s[4] = vy;
+
+      int sc = scp[b] + 1;
+      bool truncated = (sc >= max_episode_steps) && !terminated;
+      bool dn = terminated || truncated;
+      float ret = erp[b] + rew;
+      int len = elp[b] + 1;
+      if (dn) { lerp[b] = ret; lelp[b] = len; }
+      render(s, nop + b * H * W);  // true final obs (extras["next_obs"])
+      if (dn) {
+        reset_env(s, (uint64_t)seed, (uint64_t)b * 977 + 13, (uint64_t)dr);
+        sc = 0; ret = 0.0f; len = 0;
+        render(s, op + b * H * W);
+      } else {
+        std::memcpy(op + b * H * W, nop + b * H * W, sizeof(float) * H * W);
+      }
+      scp[b] = sc; erp[b] = ret; elp[b] = len;
+      rp[b] = rew;
+      dp[b] = terminated ? 0.0f : 1.0f;
+      stp[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+      dnp[b] = dn ? 1 : 0;
+    }
+  });
+  *draw.data_ptr<int32_t>() = dr + 1;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("breakout_reset", &breakout_cpu_reset,
+        "batched Breakout reset + render (CPU, threaded)");
+  m.def("breakout_step", &breakout_cpu_step,
+        "batched Breakout fused step: physics + metrics + autoreset + render");
+  m.attr("STATE_DIM") = SDIM;
+}

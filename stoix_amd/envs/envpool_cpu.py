@@ -1,0 +1,109 @@
+"""Python wrapper over the native batched CPU env engine (envpool-class).
+
+``BreakoutCpu`` exposes the same TimeStep contract as every other env here,
+but the entire step — physics, termination/truncation, episode metrics,
+autoreset, and frame rendering — is ONE C++ call parallelised over envs
+(stoix_amd/envs/csrc/envpool_cpu.cpp), the same fused-step shape the HIP
+env kernels use on GPU. This is the suite the Sebulba actor threads drive
+(reference envpool parity: C++ batched CPU envs).
+
+The registry's envpool suite uses this class automatically on CPU devices
+when the extension is built, falling back to the torch-ops Breakout
+otherwise (and always on CUDA, where the torch path runs on-device).
+"""
+from __future__ import annotations
+
+import torch
+
+from stoix_amd.envs.env import StatefulVecEnv, Tensor
+from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
+from stoix_amd.types import StepType, TimeStep
+
+H, W = 84, 84
+
+_EXT = None
+
+
+def envpool_ext(required: bool = False):
+    """Load the CPU envpool extension built in-tree (envs/build/)."""
+    global _EXT
+    if _EXT is None:
+        try:
+            from stoix_amd.envs.build_envpool import load_built
+
+            _EXT = load_built()
+        except Exception:
+            if required:
+                raise
+            _EXT = False
+    return _EXT or None
+
+
+class BreakoutCpu(StatefulVecEnv):
+    """Native-engine Breakout; drop-in for envs/breakout.py on CPU."""
+
+    max_episode_steps = 3000
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        assert self.device.type == "cpu", "BreakoutCpu is the CPU actor engine"
+        self._ext = envpool_ext(required=True)
+        self.observation_space = BoxSpace((H, W, 1), 0.0, 1.0)
+        self.action_space = DiscreteSpace(4)
+        self.seed = int(seed)
+        B = self.num_envs
+        self._s = torch.zeros(B, int(self._ext.STATE_DIM), dtype=torch.float32)
+        self._obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
+        self._next_obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
+        self._reward = torch.zeros(B, dtype=torch.float32)
+        self._discount = torch.zeros(B, dtype=torch.float32)
+        self._steptype = torch.zeros(B, dtype=torch.uint8)
+        self._done = torch.zeros(B, dtype=torch.uint8)
+        self._draw = torch.zeros(1, dtype=torch.int32)
+
+    def reset(self) -> TimeStep:
+        self._step_count.zero_()
+        self._ep_return.zero_()
+        self._ep_length.zero_()
+        self._draw += 1
+        self._ext.breakout_reset(self._s, self._obs.view(self.num_envs, -1),
+                                 self.seed, int(self._draw.item()))
+        B = self.num_envs
+        return TimeStep(
+            step_type=torch.full((B,), StepType.FIRST, dtype=torch.uint8),
+            reward=torch.zeros(B, dtype=torch.float32),
+            discount=torch.ones(B, dtype=torch.float32),
+            observation=self._obs.clone(),
+            extras={
+                "next_obs": self._obs.clone(),
+                "episode_metrics": {
+                    "episode_return": self._last_ep_return.clone(),
+                    "episode_length": self._last_ep_length.to(torch.float32),
+                    "is_terminal_step": torch.zeros(B, dtype=torch.bool),
+                },
+            },
+        )
+
+    def step(self, action: Tensor) -> TimeStep:
+        self._ext.breakout_step(
+            self._s, action.to(torch.int64).contiguous(), self._step_count,
+            self._ep_return, self._ep_length, self._last_ep_return,
+            self._last_ep_length, self._obs.view(self.num_envs, -1),
+            self._next_obs.view(self.num_envs, -1), self._reward,
+            self._discount, self._steptype, self._done,
+            self.max_episode_steps, self.seed, self._draw,
+        )
+        return TimeStep(
+            step_type=self._steptype.clone(),
+            reward=self._reward.clone(),
+            discount=self._discount.clone(),
+            observation=self._obs.clone(),
+            extras={
+                "next_obs": self._next_obs.clone(),
+                "episode_metrics": {
+                    "episode_return": self._last_ep_return.clone(),
+                    "episode_length": self._last_ep_length.to(torch.float32),
+                    "is_terminal_step": self._done.bool(),
+                },
+            },
+        )

@@ -378,3 +378,70 @@ def test_new_envs_step_on_gpu():
             ts = env.step(act(None))
         assert ts.observation.is_cuda
         assert torch.isfinite(ts.reward).all()
+
+
+def test_native_breakout_contract_and_registry():
+    """The C++ batched pool must honour the TimeStep contract and be what
+    the envpool suite serves on CPU (torch fallback on CUDA)."""
+    pytest.importorskip("stoix_amd.envs.build_envpool")
+    from stoix_amd.envs.envpool_cpu import BreakoutCpu, envpool_ext
+
+    if envpool_ext() is None:
+        pytest.skip("envpool extension not built")
+    env = BreakoutCpu(num_envs=4, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (4, 84, 84, 1)
+    assert (ts.observation[:, 80, :, 0] == 1.0).any()  # paddle row drawn
+    term_seen = False
+    for _ in range(400):
+        ts = env.step(torch.randint(0, 4, (4,)))
+        assert torch.isfinite(ts.reward).all()
+        if bool(ts.extras["episode_metrics"]["is_terminal_step"].any()):
+            term_seen = True
+            # autoreset: returned obs is the fresh board (full brick rows),
+            # true final obs is in extras
+            break
+    assert term_seen, "random play must miss the ball within 400 steps"
+
+    # registry serves the native pool on cpu
+    from stoix_amd.envs import ENV_REGISTRY
+
+    maker = ENV_REGISTRY["envpool"]("breakout")
+    e = maker(num_envs=2, device="cpu", seed=1)
+    assert type(e).__name__ == "BreakoutCpu"
+
+
+def test_native_breakout_matches_torch_rules():
+    """Native physics mirrors the torch-ops Breakout: from the same state,
+    one step with the same action gives the same ball/paddle/bricks."""
+    pytest.importorskip("stoix_amd.envs.build_envpool")
+    from stoix_amd.envs.breakout import Breakout
+    from stoix_amd.envs.envpool_cpu import BreakoutCpu, envpool_ext
+
+    if envpool_ext() is None:
+        pytest.skip("envpool extension not built")
+    tor = Breakout(num_envs=1, seed=0)
+    tor.reset()
+    nat = BreakoutCpu(num_envs=1, seed=0)
+    nat.reset()
+    # force identical mid-flight state
+    tor._state["paddle_x"][:] = 40.0
+    tor._state["ball_x"][:] = 41.0
+    tor._state["ball_y"][:] = 78.5
+    tor._state["ball_vx"][:] = 0.5
+    tor._state["ball_vy"][:] = 1.8
+    nat._s[0, 0] = 40.0
+    nat._s[0, 1] = 41.0
+    nat._s[0, 2] = 78.5
+    nat._s[0, 3] = 0.5
+    nat._s[0, 4] = 1.8
+    for i in range(30):
+        a = torch.tensor([i % 4])
+        t1 = tor.step(a)
+        t2 = nat.step(a)
+        assert abs(float(tor._state["ball_x"][0]) - float(nat._s[0, 1])) < 1e-4, i
+        assert abs(float(tor._state["ball_y"][0]) - float(nat._s[0, 2])) < 1e-4, i
+        assert abs(float(tor._state["paddle_x"][0]) - float(nat._s[0, 0])) < 1e-4, i
+        assert float(t1.reward[0]) == float(t2.reward[0]), i
+        if bool(t1.extras["episode_metrics"]["is_terminal_step"][0]):
+            break
