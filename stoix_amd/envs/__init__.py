@@ -53,6 +53,7 @@ def _brax(name: str):
 
 
 def _jumanji(name: str):
+    from stoix_amd.envs.connector import Connector
     from stoix_amd.envs.game2048 import Game2048
     from stoix_amd.envs.snake import Snake
 
@@ -62,6 +63,8 @@ def _jumanji(name: str):
         "game_2048": Game2048,
         "2048": Game2048,
         "Game2048-v1": Game2048,
+        "connector": Connector,
+        "Connector-v2": Connector,
     }
     if name not in table:
         raise ValueError(f"unknown jumanji-suite env '{name}' (have {list(table)})")

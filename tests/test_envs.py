@@ -445,3 +445,56 @@ def test_native_breakout_matches_torch_rules():
         assert float(t1.reward[0]) == float(t2.reward[0]), i
         if bool(t1.extras["episode_metrics"]["is_terminal_step"][0]):
             break
+
+
+def test_connector_rules():
+    from stoix_amd.envs.connector import Connector
+
+    env = Connector(num_envs=1, device="cpu", seed=0)
+    env.reset()
+    # craft a known position: agent0 head at (0,0), target at (0,2);
+    # agent1 head (5,5), target (5,3)
+    import torch as T
+
+    grid = T.zeros(1, 6, 6, dtype=T.long)
+    grid[0, 0, 0] = 2   # head a0
+    grid[0, 0, 2] = 3   # target a0
+    grid[0, 5, 5] = 5   # head a1
+    grid[0, 5, 3] = 6   # target a1
+    env._state = {
+        "grid": grid,
+        "heads": T.tensor([[[0, 0], [5, 5]]]),
+        "targets": T.tensor([[[0, 2], [5, 3]]]),
+        "connected": T.zeros(1, 2, dtype=T.bool),
+    }
+    # both move right/left toward their targets: a0 right (2), a1 left (4)
+    ts = env.step(T.tensor([[2, 4]]))
+    assert abs(float(ts.reward[0]) - (-0.03)) < 1e-6  # not connected yet
+    assert env._state["grid"][0, 0, 1] == 2  # new head
+    assert env._state["grid"][0, 0, 0] == 1  # trail left behind
+    ts = env.step(T.tensor([[2, 4]]))
+    # both reach targets simultaneously: +2 reward, episode terminates
+    assert abs(float(ts.reward[0]) - 2.0) < 1e-6
+    assert bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+
+
+def test_connector_blocked_move_is_noop():
+    from stoix_amd.envs.connector import Connector
+    import torch as T
+
+    env = Connector(num_envs=1, device="cpu", seed=1)
+    env.reset()
+    grid = T.zeros(1, 6, 6, dtype=T.long)
+    grid[0, 2, 2] = 2  # head a0
+    grid[0, 2, 3] = 5  # head a1 right next to it
+    grid[0, 0, 0] = 3
+    grid[0, 5, 5] = 6
+    env._state = {
+        "grid": grid,
+        "heads": T.tensor([[[2, 2], [2, 3]]]),
+        "targets": T.tensor([[[0, 0], [5, 5]]]),
+        "connected": T.zeros(1, 2, dtype=T.bool),
+    }
+    ts = env.step(T.tensor([[2, 0]]))  # a0 tries to move onto a1's head
+    assert env._state["heads"][0, 0].tolist() == [2, 2]  # blocked -> no move
+    assert env._state["grid"][0, 2, 2] == 2  # still a head, no trail
