@@ -121,22 +121,26 @@ def bench_sebulba_breakout(quick):
     from stoix_amd.config import compose
     from stoix_amd.systems.ppo.sebulba_ff_ppo import run_experiment
 
-    n_updates = 4 if quick else 12
+    n_updates = 4 if quick else 24
+    n_envs = 32 if quick else 256
+    rollout = 32 if quick else 64
+    actors = 2 if quick else 4
     cfg = compose(
         "default/sebulba/default_ff_ppo.yaml",
         ["env=envpool/breakout", "network=cnn",
-         "arch.total_num_envs=32", "arch.total_timesteps=null",
+         f"arch.total_num_envs={n_envs}", "arch.total_timesteps=null",
          f"arch.num_updates={n_updates}", "arch.num_evaluation=1",
-         "arch.num_eval_episodes=4", "arch.actor.actor_per_device=2",
-         "system.rollout_length=32", "system.num_minibatches=2", "system.epochs=1",
+         "arch.num_eval_episodes=4", f"arch.actor.actor_per_device={actors}",
+         f"system.rollout_length={rollout}", "system.num_minibatches=2", "system.epochs=1",
          "logger.loggers=[]", "logger.checkpointing.save_model=false"],
     )
     t0 = time.perf_counter()
     run_experiment(cfg)
     dt = time.perf_counter() - t0
-    steps = 32 * 32 * n_updates
+    steps = n_envs * rollout * n_updates
     return {"config": "sebulba_ppo_breakout_pixels", "steps_per_second": steps / dt,
-            "wall_s": dt, "device": "cpu-actors + learner device"}
+            "wall_s": dt, "envs": n_envs, "actors": actors,
+            "device": "cpu-actors + learner device"}
 
 
 def main():
