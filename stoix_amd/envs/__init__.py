@@ -56,6 +56,7 @@ def _jumanji(name: str):
     from stoix_amd.envs.connector import Connector
     from stoix_amd.envs.game2048 import Game2048
     from stoix_amd.envs.snake import Snake
+    from stoix_amd.envs.sokoban import Sokoban
 
     table = {
         "snake": Snake,
@@ -65,6 +66,8 @@ def _jumanji(name: str):
         "Game2048-v1": Game2048,
         "connector": Connector,
         "Connector-v2": Connector,
+        "sokoban": Sokoban,
+        "Sokoban-v0": Sokoban,
     }
     if name not in table:
         raise ValueError(f"unknown jumanji-suite env '{name}' (have {list(table)})")

@@ -498,3 +498,62 @@ def test_connector_blocked_move_is_noop():
     ts = env.step(T.tensor([[2, 0]]))  # a0 tries to move onto a1's head
     assert env._state["heads"][0, 0].tolist() == [2, 2]  # blocked -> no move
     assert env._state["grid"][0, 2, 2] == 2  # still a head, no trail
+
+
+def test_sokoban_push_rules():
+    from stoix_amd.envs.sokoban import Sokoban
+    import torch as T
+
+    env = Sokoban(num_envs=1, device="cpu", seed=0)
+    env.reset()
+    # agent at (5,2), box at (5,3), target at (5,4): pushing right solves
+    # one box; park other boxes on their targets already
+    env._state = {
+        "agent": T.tensor([[5, 2]]),
+        "boxes": T.tensor([[[5, 3], [1, 1], [1, 3], [1, 5]]]),
+        "targets": T.tensor([[[5, 4], [1, 1], [1, 3], [1, 5]]]),
+    }
+    ts = env.step(T.tensor([1]))  # right
+    # push succeeded: box on target, all 4 on target -> solved (+1 +10 -0.1)
+    assert abs(float(ts.reward[0]) - 10.9) < 1e-5
+    assert bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+
+
+def test_sokoban_blocked_push():
+    from stoix_amd.envs.sokoban import Sokoban
+    import torch as T
+
+    env = Sokoban(num_envs=1, device="cpu", seed=0)
+    env.reset()
+    env._state = {
+        "agent": T.tensor([[5, 2]]),
+        "boxes": T.tensor([[[5, 3], [5, 4], [1, 3], [1, 5]]]),  # two in a row
+        "targets": T.tensor([[[5, 6], [5, 7], [1, 3], [1, 5]]]),
+    }
+    ts = env.step(T.tensor([1]))  # right: box behind box -> blocked
+    assert env._state["agent"][0].tolist() == [5, 2]
+    assert env._state["boxes"][0, 0].tolist() == [5, 3]
+    assert abs(float(ts.reward[0]) + 0.1) < 1e-5
+
+
+def test_sokoban_generated_levels_valid():
+    from stoix_amd.envs.sokoban import Sokoban
+    import torch as T
+
+    env = Sokoban(num_envs=64, device="cpu", seed=2)
+    ts = env.reset()
+    assert ts.observation.shape == (64, 10, 10, 7)
+    s = env._state
+    # boxes distinct and interior
+    for k in range(4):
+        assert (s["boxes"][:, k, 0] >= 1).all() and (s["boxes"][:, k, 0] <= 8).all()
+    flat = s["boxes"][:, :, 0] * 10 + s["boxes"][:, :, 1]
+    assert (flat.sort(-1).values.diff(dim=-1) > 0).all(), "boxes overlap"
+    # agent not on a box
+    af = s["agent"][:, 0] * 10 + s["agent"][:, 1]
+    assert not (flat == af.unsqueeze(1)).any()
+    # steps run
+    g = T.Generator().manual_seed(0)
+    for _ in range(30):
+        ts = env.step(T.randint(0, 4, (64,), generator=g))
+        assert T.isfinite(ts.reward).all()
