@@ -6,14 +6,25 @@ thousands of 4 us elementwise kernels (profiles/r01_ppo_ant_SUMMARY.md).
 This engine replaces it with the hand-written CDNA4 kernels of
 stoix_amd/ops/csrc/mlp.hip:
 
-  rollout step:  policy_value_step (MFMA actor+critic fwd + Philox
-                 tanh-normal sample + log-prob, ONE kernel) -> env HIP step
-                 -> value_forward (bootstrap V(next_obs))        [5 launches]
-  minibatch:     ppo_gather -> 4 bf16 GEMMs (hipBLASLt MFMA) + silu kernels
-                 -> ppo_head_loss (head fwd + losses + analytic head bwd)
-                 -> hand backward (6 GEMMs + 2 silu_bwd per net)
-                 -> one flat bf16 RCCL all-reduce
-                 -> fused_adam_bf16 per chain (clip + Adam + bf16 mirror)
+  rollout step:  rollout_step_ant (megakernel: actor fwd + Philox
+                 tanh-normal sample + log-prob + critic fwd + quad-lane
+                 physics + autoreset + bootstrap critic, ONE launch); or
+                 policy_value_step -> env HIP step -> value_forward for
+                 non-Ant HIP envs
+  minibatch:     ppo_gather -> layer-1 fused linear_silu per net (custom
+                 MFMA wins at K=32) -> layer-2 tuned hipBLASLt GemmAndBias
+                 per net + ONE stacked silu (library wins at K=256)
+                 -> head GEMMs -> ppo_head_loss (per-row losses + analytic
+                 head bwd; metrics only on the reported minibatch)
+                 -> stacked backward over [2, S, H] buffers: one silu_bwd
+                 per layer pair, dH1 as one batched bmm against the
+                 adjacent-W2 [2, H, H] view, split-K wgrads into per-chain
+                 slabs -> slab_reduce per chain
+                 -> ONE bf16 RCCL all-reduce over both chains' shared
+                 grad buffer
+                 -> fused_adam_bf16 per chain (clip + Adam + bf16 mirror;
+                 per-chain lr and global-norm clip match the reference's
+                 per-network optax chains)
 
 Parameters: fp32 masters live in ONE flat buffer per network (the module
 parameters are repointed to views, so evaluator/checkpointing see updates
@@ -29,7 +40,8 @@ ACT<=8, ScalarCriticHead, flat obs <= 128 dims, no observation normalisation.
 Algorithm semantics match ff_ppo.py exactly (same losses, same GAE buffers,
 same per-minibatch all-reduce + per-chain clip); numerics are bf16-GEMM
 class, verified against the eager fp32 path in tests/test_fused_math.py and
-tests/test_gpu_ops.py.
+tests/test_fused_gpu.py, and learning curves are re-validated after every
+numerics change (tools/learncheck.py; profiles/r01_learning_curves.md).
 """
 from __future__ import annotations
 
