@@ -23,7 +23,7 @@ __all__ = ["make", "make_factory", "StatefulVecEnv", "get_final_step_metrics", "
 
 
 def _classic(name: str):
-    from stoix_amd.envs.minatar import Freeway, SpaceInvaders
+    from stoix_amd.envs.minatar import Asterix, BreakoutMinAtar, Freeway, SpaceInvaders
 
     table = {
         "CartPole-v1": CartPole,
@@ -36,6 +36,9 @@ def _classic(name: str):
         "freeway": Freeway,
         "SpaceInvaders-MinAtar": SpaceInvaders,
         "space_invaders": SpaceInvaders,
+        "Asterix-MinAtar": Asterix,
+        "asterix": Asterix,
+        "Breakout-MinAtar": BreakoutMinAtar,
     }
     if name not in table:
         raise ValueError(f"unknown classic env '{name}' (have {list(table)})")

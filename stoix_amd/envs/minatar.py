@@ -1,8 +1,9 @@
-"""MinAtar-class 10x10 grid games: Freeway and SpaceInvaders (vectorised).
+"""MinAtar-class 10x10 grid games: Freeway, SpaceInvaders, Asterix and
+Breakout (vectorised).
 
 Fill the roles of gymnax's MinAtar suite in the reference's configs
-(/root/reference/stoix/configs/env/gymnax/{freeway,space_invaders}.yaml;
-SURVEY §8.8 — breakout is covered by envs/breakout.py). gymnax/MinAtar are
+(/root/reference/stoix/configs/env/gymnax/{freeway,space_invaders,asterix,
+breakout}.yaml). gymnax/MinAtar are
 JAX/numpy-only; these are original, fully tensorised torch implementations
 (every rule batched over B boards) with the MinAtar contracts: 10x10
 multi-channel binary observations, small discrete action sets, +1-per-event
@@ -227,3 +228,171 @@ class SpaceInvaders(StatefulVecEnv):
         s["ay"] = torch.where(cleared, torch.zeros_like(s["ay"]), s["ay"])
         s["wave"] = s["wave"] + cleared.long()
         return s, reward, terminated
+
+
+class Asterix(StatefulVecEnv):
+    """MinAtar Asterix: the player moves in all four directions across 8
+    entity lanes (rows 1..8); entities drift horizontally and are either
+    GOLD (+1 on contact) or ENEMIES (episode ends on contact). Spawn
+    probability and drift speed ramp up over the episode."""
+
+    max_episode_steps = 1000
+    SPAWN_P = 0.12
+    GOLD_P = 0.3
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((G, G, 4), -1.0, 1.0)
+        self.action_space = DiscreteSpace(5)  # noop/up/down/left/right
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "pr": torch.full((n,), 5, dtype=torch.long, device=dev),
+            "pc": torch.full((n,), 5, dtype=torch.long, device=dev),
+            # 8 lanes: active, col, dir, is_gold, move timer
+            "act": torch.zeros(n, 8, dtype=torch.bool, device=dev),
+            "col": torch.zeros(n, 8, dtype=torch.long, device=dev),
+            "dir": torch.ones(n, 8, dtype=torch.long, device=dev),
+            "gold": torch.zeros(n, 8, dtype=torch.bool, device=dev),
+            "timer": torch.zeros(n, 8, dtype=torch.long, device=dev),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        n = state["pr"].shape[0]
+        dev = self.device
+        obs = torch.zeros(n, G, G, 4, device=dev)
+        bidx = torch.arange(n, device=dev)
+        obs[bidx, state["pr"], state["pc"], 0] = 1.0
+        lanes = torch.arange(1, 9, device=dev).unsqueeze(0).expand(n, -1)
+        bb = bidx.unsqueeze(1).expand(-1, 8)
+        a = state["act"]
+        obs[bb[a], lanes[a], state["col"][a], 1] = 1.0
+        gold = a & state["gold"]
+        enem = a & ~state["gold"]
+        obs[bb[gold], lanes[gold], state["col"][gold], 2] = 1.0
+        obs[bb[enem], lanes[enem], state["col"][enem], 3] = state["dir"][enem].float()
+        return obs
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long()
+        n = a.shape[0]
+        bidx = torch.arange(n, device=self.device)
+        pr = (state["pr"] + (a == 1).long() * -1 + (a == 2).long()).clamp(1, 8)
+        pc = (state["pc"] + (a == 3).long() * -1 + (a == 4).long()).clamp(0, G - 1)
+
+        # entities advance every 2 frames
+        timer = state["timer"] + 1
+        move = state["act"] & (timer >= 2)
+        col = state["col"] + torch.where(move, state["dir"], 0)
+        timer = torch.where(move, torch.zeros_like(timer), timer)
+        off = (col < 0) | (col > G - 1)
+        act = state["act"] & ~off
+
+        # spawn into inactive lanes
+        u = torch.rand(n, 8, device=self.device, generator=self.gen)
+        spawn = ~act & (u < self.SPAWN_P)
+        from_left = torch.rand(n, 8, device=self.device, generator=self.gen) < 0.5
+        col = torch.where(spawn, torch.where(from_left, 0, G - 1), col)
+        ndir = torch.where(from_left, 1, -1)
+        dir_ = torch.where(spawn, ndir, state["dir"])
+        gold = torch.where(
+            spawn, torch.rand(n, 8, device=self.device, generator=self.gen) < self.GOLD_P,
+            state["gold"],
+        )
+        act = act | spawn
+
+        # contact with the player's lane/col
+        in_lane = pr >= 1
+        lane_idx = (pr - 1).clamp(0, 7)
+        ent_here = act[bidx, lane_idx] & (col[bidx, lane_idx] == pc) & in_lane
+        hit_gold = ent_here & gold[bidx, lane_idx]
+        hit_enemy = ent_here & ~gold[bidx, lane_idx]
+        # collected gold disappears
+        clear = torch.zeros_like(act)
+        clear[bidx, lane_idx] = hit_gold
+        act = act & ~clear
+
+        reward = hit_gold.float()
+        terminated = hit_enemy
+        return (
+            {"pr": pr, "pc": pc, "act": act, "col": col, "dir": dir_,
+             "gold": gold, "timer": timer},
+            reward,
+            terminated,
+        )
+
+
+class BreakoutMinAtar(StatefulVecEnv):
+    """MinAtar Breakout: 10x10 grid, 3 brick rows, 1px paddle on the bottom
+    row, diagonal ball; +1 per brick, episode ends on a miss; cleared
+    bricks respawn (MinAtar's endless variant)."""
+
+    max_episode_steps = 1000
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((G, G, 4), 0.0, 1.0)
+        self.action_space = DiscreteSpace(3)  # noop/left/right
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "pad": torch.full((n,), G // 2, dtype=torch.long, device=dev),
+            "bx": self.randint(G, n),
+            "by": torch.full((n,), 3, dtype=torch.long, device=dev),
+            "vx": self.randint(2, n) * 2 - 1,
+            "vy": torch.ones(n, dtype=torch.long, device=dev),
+            "bricks": torch.ones(n, 3, G, dtype=torch.bool, device=dev),  # rows 1..3
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        n = state["pad"].shape[0]
+        dev = self.device
+        obs = torch.zeros(n, G, G, 4, device=dev)
+        bidx = torch.arange(n, device=dev)
+        obs[bidx, G - 1, state["pad"], 0] = 1.0
+        obs[bidx, state["by"].clamp(0, G - 1), state["bx"].clamp(0, G - 1), 1] = 1.0
+        obs[:, 1:4, :, 2] = state["bricks"].float()
+        obs[bidx, state["by"].clamp(0, G - 1), state["bx"].clamp(0, G - 1), 3] = (
+            state["vy"].float()
+        )
+        return obs
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long()
+        n = a.shape[0]
+        bidx = torch.arange(n, device=self.device)
+        pad = (state["pad"] + (a == 2).long() - (a == 1).long()).clamp(0, G - 1)
+        vx, vy = state["vx"].clone(), state["vy"].clone()
+        bx = state["bx"] + vx
+        by = state["by"] + vy
+        # side walls
+        bounce_x = (bx < 0) | (bx > G - 1)
+        vx = torch.where(bounce_x, -vx, vx)
+        bx = bx.clamp(0, G - 1)
+        # top
+        vy = torch.where(by < 0, torch.ones_like(vy), vy)
+        by = by.clamp(min=0)
+        # bricks (rows 1..3)
+        in_brick = (by >= 1) & (by <= 3)
+        row = (by - 1).clamp(0, 2)
+        bricks = state["bricks"].clone()
+        alive = bricks[bidx, row, bx] & in_brick
+        bricks[bidx, row, bx] = bricks[bidx, row, bx] & ~alive
+        vy = torch.where(alive, -vy, vy)
+        reward = alive.float()
+        # respawn a cleared wall (endless MinAtar breakout)
+        cleared = ~bricks.any(dim=(-1, -2))
+        bricks = torch.where(cleared.view(-1, 1, 1), torch.ones_like(bricks), bricks)
+        # paddle / miss at the bottom row
+        at_bottom = by >= G - 1
+        on_pad = at_bottom & (bx == pad)
+        vy = torch.where(on_pad, -torch.ones_like(vy), vy)
+        by = torch.where(on_pad, torch.full_like(by, G - 2), by)
+        terminated = at_bottom & ~on_pad
+        return (
+            {"pad": pad, "bx": bx, "by": by, "vx": vx, "vy": vy, "bricks": bricks},
+            reward,
+            terminated,
+        )

@@ -557,3 +557,55 @@ def test_sokoban_generated_levels_valid():
     for _ in range(30):
         ts = env.step(T.randint(0, 4, (64,), generator=g))
         assert T.isfinite(ts.reward).all()
+
+
+def test_asterix_rules():
+    from stoix_amd.envs.minatar import Asterix
+    import torch as T
+
+    env = Asterix(num_envs=1, device="cpu", seed=0)
+    env.reset()
+    # place a gold entity right of the player in its lane and walk into it
+    env._state["pr"][:] = 3
+    env._state["pc"][:] = 4
+    env._state["act"][:] = False
+    env._state["act"][0, 2] = True     # lane row 3 -> index 2
+    env._state["col"][0, 2] = 5
+    env._state["dir"][0, 2] = 0        # static for the test
+    env._state["gold"][0, 2] = True
+    env._state["timer"][0, 2] = -10    # don't advance during the test
+    ts = env.step(T.tensor([4]))       # move right onto the gold
+    assert float(ts.reward[0]) == 1.0
+    assert not bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+    # enemy contact terminates
+    env._state["act"][0, 2] = True
+    env._state["col"][0, 2] = env._state["pc"][0].item()
+    env._state["gold"][0, 2] = False
+    env._state["timer"][0, 2] = -10
+    ts = env.step(T.tensor([0]))
+    assert bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+
+
+def test_breakout_minatar_rules():
+    from stoix_amd.envs.minatar import BreakoutMinAtar
+    import torch as T
+
+    env = BreakoutMinAtar(num_envs=1, device="cpu", seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (1, 10, 10, 4)
+    assert ts.observation[0, 1:4, :, 2].sum() == 30  # full brick wall
+    # ball heading down toward the paddle column bounces; next to it misses
+    env._state["bx"][:] = 4
+    env._state["by"][:] = 8
+    env._state["vx"][:] = 0
+    env._state["vy"][:] = 1
+    env._state["pad"][:] = 4
+    ts = env.step(T.tensor([0]))
+    assert not bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+    env._state["bx"][:] = 4
+    env._state["by"][:] = 8
+    env._state["vx"][:] = 0
+    env._state["vy"][:] = 1
+    env._state["pad"][:] = 0
+    ts = env.step(T.tensor([0]))
+    assert bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
