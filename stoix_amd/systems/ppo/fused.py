@@ -277,16 +277,34 @@ class FusedPPOEngine:
             "entropy": self.metrics[2],
         }
 
-        # minibatch workspaces
+        # minibatch workspaces: TWO sets (ping/pong) so the next
+        # minibatch's gather can prefetch on a side stream while the
+        # current one is in backward (the gather is latency-bound random
+        # reads — ~17 us/mb of otherwise dead time)
         S = (learner.T * B) // int(sysc.num_minibatches)
         self.S = S
         z = lambda *s, dtype=torch.bfloat16: torch.zeros(*s, dtype=dtype, device=dev)
-        self.Xmb = z(S, self.K1P)  # K-padded GEMM input (zeros past OBS)
-        self.act_mb = z(S, ACT, dtype=torch.float32)
-        self.logp_mb = z(S, dtype=torch.float32)
-        self.val_mb = z(S, dtype=torch.float32)
-        self.adv_mb = z(S, dtype=torch.float32)
-        self.tgt_mb = z(S, dtype=torch.float32)
+
+        class _Ws:
+            def __init__(ws):
+                ws.Xmb = z(S, self.K1P)  # K-padded GEMM input (zeros past OBS)
+                ws.act = z(S, ACT, dtype=torch.float32)
+                ws.logp = z(S, dtype=torch.float32)
+                ws.val = z(S, dtype=torch.float32)
+                ws.adv = z(S, dtype=torch.float32)
+                ws.tgt = z(S, dtype=torch.float32)
+
+        self.ws = [_Ws(), _Ws()]
+        # single-set aliases (perf probes / tests reach these)
+        self.Xmb = self.ws[0].Xmb
+        self.act_mb = self.ws[0].act
+        self.logp_mb = self.ws[0].logp
+        self.val_mb = self.ws[0].val
+        self.adv_mb = self.ws[0].adv
+        self.tgt_mb = self.ws[0].tgt
+        self.gather_stream = torch.cuda.Stream(dev)
+        self.ev_ready = [torch.cuda.Event(), torch.cuda.Event()]
+        self.ev_consumed = [torch.cuda.Event(), torch.cuda.Event()]
         # stacked [net, S, H] activation buffers: index 0 = actor, 1 =
         # critic; lets silu_bwd run ONCE over both nets and dH1 run as one
         # batched bmm against W2pair
@@ -306,6 +324,11 @@ class FusedPPOEngine:
 
         # batched dH1 GEMM (one bmm vs two mm); A/B knob for measurement
         self.use_bmm = _os.environ.get("STOIX_FUSED_BMM", "1") != "0"
+        # side-stream gather prefetch. Measured SLOWER (28.0 vs 26.8
+        # ms/step): the backward kernels already fill all 256 CUs, so the
+        # concurrent gather steals bandwidth instead of hiding latency.
+        # Kept behind a knob as a documented negative result.
+        self.prefetch = _os.environ.get("STOIX_FUSED_PREFETCH", "0") == "1"
 
         import torch.distributed as dist
 
@@ -465,20 +488,41 @@ class FusedPPOEngine:
 
         import torch.distributed as dist
 
-        for mb in range(n_mb):
+        def gather_into(mb: int, w) -> None:
             idx = L.perm_buf[mb * S : (mb + 1) * S]
             ext.ppo_gather(
                 idx, flat_obs, flat_action, flat_logp, flat_value, flat_adv,
-                flat_tgt, self.Xmb, self.act_mb, self.logp_mb, self.val_mb,
-                self.adv_mb, self.tgt_mb, self.empty, self.empty,
+                flat_tgt, w.Xmb, w.act, w.logp, w.val, w.adv, w.tgt,
+                self.empty, self.empty,
             )
+
+        # gather prefetch: minibatch mb+1's gather runs on a side stream
+        # while mb's backward is on the main stream; double-buffered
+        # workspaces + events make the dependency explicit (capture-legal:
+        # record/wait become hipGraph edges)
+        gs = self.gather_stream
+        cur = torch.cuda.current_stream(self.device)
+        prefetch = self.prefetch
+        if prefetch:
+            gs.wait_stream(cur)
+            with torch.cuda.stream(gs):
+                gather_into(0, self.ws[0])
+                self.ev_ready[0].record(gs)
+
+        for mb in range(n_mb):
+            p = mb & 1 if prefetch else 0
+            w = self.ws[p]
+            if prefetch:
+                cur.wait_event(self.ev_ready[p])
+            else:
+                gather_into(mb, w)
             # ---- forward. Layer 1 (K=32): the custom fused Linear+SiLU
             # MFMA kernel wins (14.4 us vs 11.2 GEMM + silu pass). Layer 2
             # (K=256): hipBLASLt's tuned GemmAndBias (14.4 us) beats the
             # custom kernel (24.6 us), so run both nets' GEMMs then ONE
             # silu over the stacked [2, S, H] preacts.
-            ext.linear_silu(self.Xmb, a16["W1"], ac.views["b1"], self.Z1[0], self.H1[0], 1)
-            ext.linear_silu(self.Xmb, c16["W1"], cc.views["b1"], self.Z1[1], self.H1[1], 1)
+            ext.linear_silu(w.Xmb, a16["W1"], ac.views["b1"], self.Z1[0], self.H1[0], 1)
+            ext.linear_silu(w.Xmb, c16["W1"], cc.views["b1"], self.Z1[1], self.H1[1], 1)
             torch.addmm(a16["b2"], self.H1[0], a16["W2"].t(), out=self.Z2[0])
             torch.addmm(c16["b2"], self.H1[1], c16["W2"].t(), out=self.Z2[1])
             ext.silu_fwd(self.Z2, self.H2)
@@ -493,8 +537,8 @@ class FusedPPOEngine:
             if last:
                 self.metrics.zero_()
             ext.ppo_head_loss(
-                self.heads, self.vpred.view(-1), self.act_mb, self.logp_mb,
-                self.val_mb, self.adv_mb, self.tgt_mb, self.dhead, self.dv,
+                self.heads, self.vpred.view(-1), w.act, w.logp,
+                w.val, w.adv, w.tgt, self.dhead, self.dv,
                 self.dv16, self.metrics if last else self.metrics_none,
                 self.clip_eps, self.ent_coef,
                 self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
@@ -520,8 +564,19 @@ class FusedPPOEngine:
                 torch.mm(self.dZ2[0], a16["W2"], out=self.dH1[0])
                 torch.mm(self.dZ2[1], c16["W2"], out=self.dH1[1])
             ext.silu_bwd(self.dH1, self.Z1, self.dZ1)
-            ext.wgrad(self.dZ1[0], self.Xmb, ac.slab, ao["W1"], ao["b1"], self.H)
-            ext.wgrad(self.dZ1[1], self.Xmb, cc.slab, co["W1"], co["b1"], self.H)
+            ext.wgrad(self.dZ1[0], w.Xmb, ac.slab, ao["W1"], ao["b1"], self.H)
+            ext.wgrad(self.dZ1[1], w.Xmb, cc.slab, co["W1"], co["b1"], self.H)
+            if prefetch:
+                # all reads of workspace set p are done (the W1 wgrads were
+                # the last); let the side stream refill it for mb+2, and
+                # kick off mb+1's gather now
+                self.ev_consumed[p].record(cur)
+                if mb + 1 < n_mb:
+                    q = (mb + 1) & 1
+                    gs.wait_event(self.ev_consumed[q])
+                    with torch.cuda.stream(gs):
+                        gather_into(mb + 1, self.ws[q])
+                        self.ev_ready[q].record(gs)
             ext.slab_reduce(ac.slab, ac.grad16, ac.sqnorm, ac.step_t)
             ext.slab_reduce(cc.slab, cc.grad16, cc.sqnorm, cc.step_t)
             # ---- ONE all-reduce over both chains' shared grad buffer,
@@ -539,6 +594,8 @@ class FusedPPOEngine:
                 cc.flat16, cc.lr, 0.9, 0.999, 1e-5, self.max_grad_norm,
                 gscale, 0,
             )
+        if prefetch:
+            cur.wait_stream(gs)
         ext.bump_add(self.draw_ent, n_mb)
         return {k: v for k, v in self.metric_views.items()}
 
