@@ -77,3 +77,28 @@ def test_prioritised_buffer_sample_and_writeback():
     buf.set_priorities(s["_slots"], torch.rand(8) * 5)
     s2 = buf.sample(8)
     assert s2["x"].shape == (8, 4, 3)
+
+
+def test_trajectory_branch_free_wrap():
+    """After wrapping, window starts must lie in the valid region behind
+    the write pointer (branch-free formula: oldest = (ptr - filled) % t_max)."""
+    buf = TrajectoryBuffer(add_batch_size=2, max_length_time_axis=16,
+                           sample_sequence_length=4, seed=0)
+    import torch as T
+
+    for i in range(5):  # 5 adds of 6 steps = 30 > 16 -> wrapped
+        buf.add({"x": T.full((2, 6, 1), float(i))})
+    assert buf.t_filled == 16
+    assert buf.t_ptr == 30 % 16
+    out = buf.sample(64)
+    assert out["x"].shape == (64, 4, 1)
+    # windows must be monotone in write order: along the time axis the
+    # values (add indices) never decrease by more than 0 (consecutive
+    # windows span one or two adds)
+    v = out["x"][..., 0]
+    assert (v.diff(dim=1) >= 0).all()
+    # and never include data older than the valid region (adds 0-1 were
+    # overwritten: values 0 can survive only at slots not yet rewritten)
+    # oldest surviving step is at (ptr - filled) % 16
+    t0 = out["_t0"]
+    assert ((t0 >= 0) & (t0 < 16)).all()

@@ -360,9 +360,11 @@ def test_new_envs_step_on_gpu():
     """Planar/2048/MinAtar envs run the generic torch path as pure device
     tensor work (no HIP kernel tier for these; the registry must still build
     and step them on cuda without host round-trips erroring)."""
+    from stoix_amd.envs.connector import Connector
     from stoix_amd.envs.game2048 import Game2048
-    from stoix_amd.envs.minatar import Freeway, SpaceInvaders
+    from stoix_amd.envs.minatar import Asterix, BreakoutMinAtar, Freeway, SpaceInvaders
     from stoix_amd.envs.planar import HalfCheetah, Hopper
+    from stoix_amd.envs.sokoban import Sokoban
 
     dev = "cuda:0"
     for cls, act in [
@@ -371,6 +373,10 @@ def test_new_envs_step_on_gpu():
         (Game2048, lambda g: torch.randint(0, 4, (8,), device=dev)),
         (Freeway, lambda g: torch.randint(0, 3, (8,), device=dev)),
         (SpaceInvaders, lambda g: torch.randint(0, 4, (8,), device=dev)),
+        (Asterix, lambda g: torch.randint(0, 5, (8,), device=dev)),
+        (BreakoutMinAtar, lambda g: torch.randint(0, 3, (8,), device=dev)),
+        (Connector, lambda g: torch.randint(0, 5, (8, 2), device=dev)),
+        (Sokoban, lambda g: torch.randint(0, 4, (8,), device=dev)),
     ]:
         env = cls(num_envs=8, device=dev, seed=0)
         ts = env.reset()

@@ -149,3 +149,35 @@ def test_fused_reference_matches_eager_losses():
     assert torch.allclose(a_loss, a_ref, atol=1e-9)
     assert torch.allclose(v_loss, v_ref, atol=1e-9)
     assert torch.allclose(entropy, ent_ref, atol=1e-9)
+
+
+def test_chain_shared_buffer_w2_adjacency():
+    """The fused engine's actor/critic chain layout: actor ends with W2,
+    critic starts with W2, both flat16/grad16 slices of one buffer -> the
+    [2, H, H] view over the boundary aliases exactly both W2 mirrors
+    (zero-copy bmm operand) and one all-reduce covers both grads."""
+    import torch
+
+    from stoix_amd.systems.ppo.fused import _Chain
+
+    H, K1P = 8, 32
+    a_specs = [("W1", (H, K1P)), ("b1", (H,)), ("Wh", (16, H)), ("bh", (16,)),
+               ("b2", (H,)), ("W2", (H, H))]
+    c_specs = [("W2", (H, H)), ("W1", (H, K1P)), ("b1", (H,)), ("b2", (H,)),
+               ("Wv", (H,)), ("bv", (1,))]
+    nA = sum(int(torch.tensor(s).prod()) for _, s in a_specs)
+    nC = sum(int(torch.tensor(s).prod()) for _, s in c_specs)
+    big16 = torch.zeros(nA + nC, dtype=torch.bfloat16)
+    biggrad = torch.zeros(nA + nC, dtype=torch.bfloat16)
+    ac = _Chain(a_specs, "cpu", 1e-3, flat16=big16[:nA], grad16=biggrad[:nA])
+    cc = _Chain(c_specs, "cpu", 1e-3, flat16=big16[nA:], grad16=biggrad[nA:])
+    pair = big16[nA - H * H : nA + H * H].view(2, H, H)
+    ac.views16["W2"].copy_(torch.full((H, H), 2.0, dtype=torch.bfloat16))
+    cc.views16["W2"].copy_(torch.full((H, H), 3.0, dtype=torch.bfloat16))
+    assert (pair[0] == 2.0).all() and (pair[1] == 3.0).all()
+    # writing through the pair view is visible to the chains (same storage)
+    pair[0, 0, 0] = 7.0
+    assert float(ac.views16["W2"][0, 0]) == 7.0
+    # one flat grad buffer covers both chains
+    cc.gviews16["Wv"].fill_(1.0)
+    assert float(biggrad.sum()) == H
