@@ -386,7 +386,11 @@ __launch_bounds__(256, 2) __global__ void value_forward_kernel(
 // (bootstrap). Replaces the 3-kernel (policy/env/value) chain: fewer
 // launches, 4x more CUs on the physics (256 WGs vs 64), and next_obs never
 // round-trips through HBM (it stays in LDS for the bootstrap pass).
-template <int HID>
+// RPW = batch rows (envs) per workgroup (16 = full MFMA M-tile; the 8-row
+// variant exists for the occupancy experiment documented in the launcher
+// -- it lost, because a half-empty tile still issues the full instruction
+// stream).
+template <int HID, int RPW>
 __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
     float* __restrict__ obs_io,          // [B, 27] env obs buffer (in/out)
     float* __restrict__ env_state,       // [B, 29]
@@ -414,8 +418,11 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
   __shared__ MlpLds<HID> lds;
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int rbase = blockIdx.x * 16;
+  const int rbase = blockIdx.x * RPW;
   if (rbase >= B) return;
+  // rows [RPW, 16) of the tile belong to the NEXT workgroup at RPW=8:
+  // cap every per-row bound so they are staged as zeros and never written
+  const int Bcap = min(B, rbase + RPW);
   const uint32_t pdraw = *policy_draw + draw_offset;
   const uint32_t edraw = *env_draw + draw_offset;
   const int K1P = (OBS + 31) & ~31;
@@ -430,7 +437,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
 
   if (threadIdx.x == 0) lds.any_done = 0;
   stage_obs(obs_io, rbase, OBS, K1P, O, OS, nullptr, nullptr, buf_obs,
-            threadIdx.x, 256, B);
+            threadIdx.x, 256, Bcap);
   __syncthreads();
 
   // ---- policy + value (same structure as policy_value_step_kernel)
@@ -443,7 +450,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
 
   if (wid == 1) {
     wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, buf_value,
-                         lane, B);
+                         lane, Bcap);
   }
   if (wid == 0) {
     f32x4 acc = {0.f, 0.f, 0.f, 0.f};
@@ -489,7 +496,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       logp += __shfl_xor(logp, 2);
       logp += __shfl_xor(logp, 4);
       int grow = rbase + g * 4 + r;
-      if (grow < B) {
+      if (grow < Bcap) {
         if (col < ACT) {
           buf_action[(long)grow * ACT + col] = a_val;
           lds.act[g * 4 + r][col] = a_val;
@@ -506,7 +513,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
     int row = threadIdx.x >> 2;
     int leg = threadIdx.x & 3;
     int b = rbase + row;
-    if (b < B) {
+    if (b < Bcap) {
       float es[ANT_STATE];
 #pragma unroll
       for (int i = 0; i < ANT_STATE; ++i) es[i] = env_state[b * ANT_STATE + i];
@@ -577,7 +584,7 @@ __launch_bounds__(256, 2) __global__ void rollout_step_ant_kernel(
       }
       acc += __shfl_xor(acc, 1);
       acc += __shfl_xor(acc, 2);
-      if (part == 0 && rbase + row < B && lds.done[row])
+      if (part == 0 && rbase + row < Bcap && lds.done[row])
         buf_bootstrap[rbase + row] = acc + (bvc ? *bvc : 0.0f);
     }
   }
@@ -597,9 +604,38 @@ extern "C" void launch_rollout_step_ant(
     unsigned int* policy_draw, unsigned int* env_draw,
     unsigned int draw_offset, void* stream) {
   hipStream_t s = (hipStream_t)stream;
-  dim3 grid((B + 15) / 16), block(256);
-  if (HID == 256) {
-    hipLaunchKernelGGL(rollout_step_ant_kernel<256>, grid, block, 0, s,
+  // RPW=8 was tried to double the grid for stall overlap at B=4096
+  // (1 WG/CU): measured WORSE (rollout 4.0 -> 5.4 ms) because each WG
+  // still issues the full 16-row tile instruction stream -- halving the
+  // rows doubles total issue. 16 stays.
+  int rpw = 16;
+  dim3 grid((B + rpw - 1) / rpw), block(256);
+  if (HID == 256 && rpw == 8) {
+    hipLaunchKernelGGL((rollout_step_ant_kernel<256, 8>), grid, block, 0, s,
+                       obs_io, env_state, step_count, ep_return, ep_length,
+                       last_ep_return, last_ep_length, (const bf16_t*)W1a,
+                       b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_steptype, B, OBS, ACT, max_episode_steps,
+                       min_scale, aff_scale, aff_shift, log_aff_scale,
+                       policy_seed, env_seed, policy_draw, env_draw,
+                       draw_offset);
+  } else if (HID == 256) {
+    hipLaunchKernelGGL((rollout_step_ant_kernel<256, 16>), grid, block, 0, s,
+                       obs_io, env_state, step_count, ep_return, ep_length,
+                       last_ep_return, last_ep_length, (const bf16_t*)W1a,
+                       b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, buf_obs, buf_action, buf_logp,
+                       buf_value, buf_bootstrap, buf_reward, buf_discount,
+                       buf_steptype, B, OBS, ACT, max_episode_steps,
+                       min_scale, aff_scale, aff_shift, log_aff_scale,
+                       policy_seed, env_seed, policy_draw, env_draw,
+                       draw_offset);
+  } else if (rpw == 8) {
+    hipLaunchKernelGGL((rollout_step_ant_kernel<128, 8>), grid, block, 0, s,
                        obs_io, env_state, step_count, ep_return, ep_length,
                        last_ep_return, last_ep_length, (const bf16_t*)W1a,
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
@@ -611,7 +647,7 @@ extern "C" void launch_rollout_step_ant(
                        policy_seed, env_seed, policy_draw, env_draw,
                        draw_offset);
   } else {
-    hipLaunchKernelGGL(rollout_step_ant_kernel<128>, grid, block, 0, s,
+    hipLaunchKernelGGL((rollout_step_ant_kernel<128, 16>), grid, block, 0, s,
                        obs_io, env_state, step_count, ep_return, ep_length,
                        last_ep_return, last_ep_length, (const bf16_t*)W1a,
                        b1a, (const bf16_t*)W2a, b2a, (const bf16_t*)Wha, bha,
