@@ -59,6 +59,7 @@ def _jumanji(name: str):
     from stoix_amd.envs.connector import Connector
     from stoix_amd.envs.game2048 import Game2048
     from stoix_amd.envs.snake import Snake
+    from stoix_amd.envs.rware import RobotWarehouse
     from stoix_amd.envs.sokoban import Sokoban
 
     table = {
@@ -71,6 +72,9 @@ def _jumanji(name: str):
         "Connector-v2": Connector,
         "sokoban": Sokoban,
         "Sokoban-v0": Sokoban,
+        "robot_warehouse": RobotWarehouse,
+        "RobotWarehouse-v0": RobotWarehouse,
+        "rware": RobotWarehouse,
     }
     if name not in table:
         raise ValueError(f"unknown jumanji-suite env '{name}' (have {list(table)})")

@@ -615,3 +615,57 @@ def test_breakout_minatar_rules():
     env._state["pad"][:] = 0
     ts = env.step(T.tensor([0]))
     assert bool(ts.extras["episode_metrics"]["is_terminal_step"][0])
+
+
+def test_rware_pickup_and_delivery():
+    from stoix_amd.envs.rware import RobotWarehouse
+    import torch as T
+
+    env = RobotWarehouse(num_envs=1, device="cpu", seed=0)
+    env.reset()
+    s = env._state
+    # put agent 0 on shelf 0's cell (2,1), facing down, others parked away
+    s["agents"][0] = T.tensor([[2, 1], [0, 7], [0, 8], [0, 9]])
+    s["dir"][0] = T.tensor([2, 0, 0, 0])
+    s["carry"][0] = T.tensor([-1, -1, -1, -1])
+    s["requested"][0] = T.zeros(12, dtype=T.bool)
+    s["requested"][0, 0] = True
+    toggle = T.tensor([[4, 0, 0, 0]])  # agent 0 toggles load
+    ts = env.step(toggle)
+    assert int(env._state["carry"][0, 0]) == 0, "pickup failed"
+    # the row-6 rack blocks a laden agent in shelf columns: route through
+    # the aisle — down to row 5, right to col 3, down to row 9, right to
+    # the goal at (9,4)
+    fwd = T.tensor([[1, 0, 0, 0]])
+    tl = T.tensor([[2, 0, 0, 0]])  # counter-clockwise
+    tr = T.tensor([[3, 0, 0, 0]])  # clockwise
+    for _ in range(3):
+        env.step(fwd)  # (2,1) -> (5,1)
+    assert env._state["agents"][0, 0].tolist() == [5, 1]
+    env.step(tl)  # down -> right
+    for _ in range(2):
+        env.step(fwd)  # -> (5,3)
+    env.step(tr)  # right -> down
+    for _ in range(4):
+        env.step(fwd)  # -> (9,3)
+    assert env._state["agents"][0, 0].tolist() == [9, 3]
+    env.step(tl)  # down -> right
+    ts = env.step(fwd)  # -> goal (9,4)
+    assert float(ts.reward[0]) == 1.0, "delivery on the goal cell must pay +1"
+    # the delivered shelf is un-requested; a fresh request replaced it
+    assert not bool(env._state["requested"][0, 0])
+    assert int(env._state["requested"][0].sum()) == 1
+
+
+def test_rware_agent_blocking():
+    from stoix_amd.envs.rware import RobotWarehouse
+    import torch as T
+
+    env = RobotWarehouse(num_envs=1, device="cpu", seed=1)
+    env.reset()
+    s = env._state
+    s["agents"][0] = T.tensor([[5, 5], [5, 6], [0, 0], [0, 9]])
+    s["dir"][0] = T.tensor([1, 3, 0, 0])  # facing each other
+    s["carry"][0] = T.tensor([-1, -1, -1, -1])
+    ts = env.step(T.tensor([[1, 0, 0, 0]]))  # agent 0 forward into agent 1
+    assert env._state["agents"][0, 0].tolist() == [5, 5], "must be blocked"
