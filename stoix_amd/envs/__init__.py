@@ -94,7 +94,18 @@ def _envpool(name: str):
                 return BreakoutCpu(**kw)
         return Breakout(**kw)
 
-    table = {"breakout": breakout_maker, "Breakout-v5": breakout_maker}
+    def pong_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import PongCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return PongCpu(**kw)
+
+    table = {
+        "breakout": breakout_maker,
+        "Breakout-v5": breakout_maker,
+        "pong": pong_maker,
+        "Pong-v5": pong_maker,
+    }
     if name not in table:
         raise ValueError(f"unknown envpool-suite env '{name}' (have {list(table)})")
     return table[name]

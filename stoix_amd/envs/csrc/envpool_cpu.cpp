@@ -11,8 +11,9 @@
 // ~1.1k changed pixels directly (~10 us/env single-thread), which is what
 // makes CPU actors keep a GPU learner fed (BASELINE config #4).
 //
-// Game rules mirror stoix_amd/envs/breakout.py exactly (84x84 grayscale,
-// 4 actions, +1/brick, terminate on miss or clear).
+// Games: Breakout (rules mirror stoix_amd/envs/breakout.py exactly:
+// 84x84 grayscale, 4 actions, +1/brick, terminate on miss or clear) and
+// Pong (scripted tracking opponent, +-1 per point, first to 21).
 #include <torch/extension.h>
 #include <ATen/Parallel.h>
 
@@ -197,7 +198,167 @@ void breakout_cpu_step(torch::Tensor state, torch::Tensor action,
   *draw.data_ptr<int32_t>() = dr + 1;
 }
 
+// ----------------------------------------------------------------- Pong
+// Atari-class Pong: player paddle on the right, scripted opponent on the
+// left (tracks the ball with capped speed), +1 when the opponent misses,
+// -1 when the player misses; first to 21 points ends the episode.
+namespace pong {
+
+constexpr int PADDLE_H = 14;
+constexpr float PLAYER_X = 80.0f, OPP_X = 3.0f;
+constexpr float PADDLE_SPEED = 2.5f, OPP_SPEED = 1.6f, BSPEED = 1.6f;
+// state row: player_y, opp_y, ball_x, ball_y, vx, vy, p_score, o_score
+constexpr int SDIM = 8;
+
+inline void reset_point(float* s, uint64_t seed, uint64_t env, uint64_t d) {
+  s[2] = W / 2.0f;
+  s[3] = 12.0f + hash_uniform(seed, env * 3, d) * 60.0f;
+  s[4] = (hash_uniform(seed, env * 3 + 1, d) > 0.5f) ? BSPEED : -BSPEED;
+  s[5] = (hash_uniform(seed, env * 3 + 2, d) - 0.5f) * 2.0f;
+}
+
+inline void reset_env(float* s, uint64_t seed, uint64_t env, uint64_t d) {
+  s[0] = H / 2.0f;
+  s[1] = H / 2.0f;
+  s[6] = 0.0f;
+  s[7] = 0.0f;
+  reset_point(s, seed, env, d);
+}
+
+inline void render(const float* s, float* obs) {
+  std::memset(obs, 0, sizeof(float) * H * W);
+  for (int y = 0; y < H; y += 4) obs[y * W + W / 2] = 0.3f;  // net
+  auto draw_paddle = [&](float py, int x) {
+    int y0 = (int)py - PADDLE_H / 2;
+    for (int y = y0; y < y0 + PADDLE_H; ++y) {
+      if (y < 0 || y > H - 1) continue;
+      obs[y * W + x] = 1.0f;
+      obs[y * W + x + 1] = 1.0f;
+    }
+  };
+  draw_paddle(s[0], (int)PLAYER_X);
+  draw_paddle(s[1], (int)OPP_X);
+  int by = (int)s[3], bx = (int)s[2];
+  if (by < 0) by = 0;
+  if (by > H - 2) by = H - 2;
+  if (bx < 0) bx = 0;
+  if (bx > W - 2) bx = W - 2;
+  for (int dy = 0; dy < 2; ++dy)
+    for (int dx = 0; dx < 2; ++dx) obs[(by + dy) * W + bx + dx] = 1.0f;
+}
+
+}  // namespace pong
+
+void pong_cpu_reset(torch::Tensor state, torch::Tensor obs, int64_t seed,
+                    int64_t draw) {
+  TORCH_CHECK(state.size(1) == pong::SDIM, "state must be [B, 8]");
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  float* op = obs.data_ptr<float>();
+  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+    for (int64_t b = lo; b < hi; ++b) {
+      pong::reset_env(sp + b * pong::SDIM, (uint64_t)seed, (uint64_t)b,
+                      (uint64_t)draw);
+      pong::render(sp + b * pong::SDIM, op + b * H * W);
+    }
+  });
+}
+
+void pong_cpu_step(torch::Tensor state, torch::Tensor action,
+                   torch::Tensor step_count, torch::Tensor ep_return,
+                   torch::Tensor ep_length, torch::Tensor last_ep_return,
+                   torch::Tensor last_ep_length, torch::Tensor obs,
+                   torch::Tensor next_obs, torch::Tensor reward,
+                   torch::Tensor discount, torch::Tensor steptype,
+                   torch::Tensor done, int64_t max_episode_steps,
+                   int64_t seed, torch::Tensor draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  const int64_t* ap = action.data_ptr<int64_t>();
+  int32_t* scp = step_count.data_ptr<int32_t>();
+  float* erp = ep_return.data_ptr<float>();
+  int32_t* elp = ep_length.data_ptr<int32_t>();
+  float* lerp = last_ep_return.data_ptr<float>();
+  int32_t* lelp = last_ep_length.data_ptr<int32_t>();
+  float* op = obs.data_ptr<float>();
+  float* nop = next_obs.data_ptr<float>();
+  float* rp = reward.data_ptr<float>();
+  float* dp = discount.data_ptr<float>();
+  uint8_t* stp = steptype.data_ptr<uint8_t>();
+  uint8_t* dnp = done.data_ptr<uint8_t>();
+  int32_t dr = *draw.data_ptr<int32_t>();
+
+  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+    for (int64_t b = lo; b < hi; ++b) {
+      float* s = sp + b * pong::SDIM;
+      int a = (int)ap[b];  // 0 noop, 1 up, 2 down
+      float py = s[0] + pong::PADDLE_SPEED * ((a == 2) - (a == 1));
+      if (py < pong::PADDLE_H / 2.0f) py = pong::PADDLE_H / 2.0f;
+      if (py > H - pong::PADDLE_H / 2.0f) py = H - pong::PADDLE_H / 2.0f;
+      // opponent tracks the ball with capped speed
+      float oy = s[1];
+      float track = s[3] - oy;
+      if (track > pong::OPP_SPEED) track = pong::OPP_SPEED;
+      if (track < -pong::OPP_SPEED) track = -pong::OPP_SPEED;
+      oy += track;
+      float bx = s[2] + s[4], by = s[3] + s[5];
+      float vx = s[4], vy = s[5];
+      if (by < 1.0f) { vy = std::fabs(vy); by = 1.0f; }
+      if (by > H - 2.0f) { vy = -std::fabs(vy); by = H - 2.0f; }
+      // paddle bounces (with english)
+      if (vx > 0 && bx >= pong::PLAYER_X - 1 && bx <= pong::PLAYER_X + 2 &&
+          std::fabs(by - py) <= pong::PADDLE_H / 2.0f + 1) {
+        vx = -std::fabs(vx) * 1.02f;
+        vy += 0.25f * (by - py) / (pong::PADDLE_H / 2.0f);
+        bx = pong::PLAYER_X - 1;
+      }
+      if (vx < 0 && bx <= pong::OPP_X + 2 && bx >= pong::OPP_X - 1 &&
+          std::fabs(by - oy) <= pong::PADDLE_H / 2.0f + 1) {
+        vx = std::fabs(vx);
+        vy += 0.25f * (by - oy) / (pong::PADDLE_H / 2.0f);
+        bx = pong::OPP_X + 2;
+      }
+      if (vx > 2.8f) vx = 2.8f;
+      if (vx < -2.8f) vx = -2.8f;
+      float rew = 0.0f;
+      bool point = false;
+      if (bx > W - 1.0f) { rew = -1.0f; s[7] += 1.0f; point = true; }
+      if (bx < 0.0f) { rew = 1.0f; s[6] += 1.0f; point = true; }
+      s[0] = py; s[1] = oy; s[2] = bx; s[3] = by; s[4] = vx; s[5] = vy;
+      bool terminated = s[6] >= 21.0f || s[7] >= 21.0f;
+      if (point && !terminated) pong::reset_point(s, (uint64_t)seed,
+                                            (uint64_t)b * 131 + 7,
+                                            (uint64_t)dr + (uint64_t)scp[b]);
+      int sc = scp[b] + 1;
+      bool truncated = (sc >= max_episode_steps) && !terminated;
+      bool dn = terminated || truncated;
+      float ret = erp[b] + rew;
+      int len = elp[b] + 1;
+      if (dn) { lerp[b] = ret; lelp[b] = len; }
+      pong::render(s, nop + b * H * W);
+      if (dn) {
+        pong::reset_env(s, (uint64_t)seed, (uint64_t)b * 977 + 13, (uint64_t)dr);
+        sc = 0; ret = 0.0f; len = 0;
+        pong::render(s, op + b * H * W);
+      } else {
+        std::memcpy(op + b * H * W, nop + b * H * W, sizeof(float) * H * W);
+      }
+      scp[b] = sc; erp[b] = ret; elp[b] = len;
+      rp[b] = rew;
+      dp[b] = terminated ? 0.0f : 1.0f;
+      stp[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+      dnp[b] = dn ? 1 : 0;
+    }
+  });
+  *draw.data_ptr<int32_t>() = dr + 1;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("pong_reset", &pong_cpu_reset,
+        "batched Pong reset + render (CPU, threaded)");
+  m.def("pong_step", &pong_cpu_step,
+        "batched Pong fused step: physics + opponent AI + metrics + autoreset + render");
+  m.attr("PONG_STATE_DIM") = pong::SDIM;
   m.def("breakout_reset", &breakout_cpu_reset,
         "batched Breakout reset + render (CPU, threaded)");
   m.def("breakout_step", &breakout_cpu_step,

@@ -1,6 +1,7 @@
 """Python wrapper over the native batched CPU env engine (envpool-class).
 
-``BreakoutCpu`` exposes the same TimeStep contract as every other env here,
+``BreakoutCpu``/``PongCpu`` expose the same TimeStep contract as every other
+env here,
 but the entire step — physics, termination/truncation, episode metrics,
 autoreset, and frame rendering — is ONE C++ call parallelised over envs
 (stoix_amd/envs/csrc/envpool_cpu.cpp), the same fused-step shape the HIP
@@ -39,20 +40,26 @@ def envpool_ext(required: bool = False):
     return _EXT or None
 
 
-class BreakoutCpu(StatefulVecEnv):
-    """Native-engine Breakout; drop-in for envs/breakout.py on CPU."""
+class _PoolEnvBase(StatefulVecEnv):
+    """Shared wrapper over a native pool game: preallocated output buffers,
+    one fused C++ call per step, TimeStep contract identical to the HIP
+    envs (clone-on-return, next_obs in extras)."""
 
-    max_episode_steps = 3000
+    STATE_DIM_ATTR = ""
+    RESET_FN = ""
+    STEP_FN = ""
+    N_ACTIONS = 4
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
-        assert self.device.type == "cpu", "BreakoutCpu is the CPU actor engine"
+        assert self.device.type == "cpu", "native pool envs are the CPU actor engine"
         self._ext = envpool_ext(required=True)
         self.observation_space = BoxSpace((H, W, 1), 0.0, 1.0)
-        self.action_space = DiscreteSpace(4)
+        self.action_space = DiscreteSpace(self.N_ACTIONS)
         self.seed = int(seed)
         B = self.num_envs
-        self._s = torch.zeros(B, int(self._ext.STATE_DIM), dtype=torch.float32)
+        sdim = int(getattr(self._ext, self.STATE_DIM_ATTR))
+        self._s = torch.zeros(B, sdim, dtype=torch.float32)
         self._obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
         self._next_obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
         self._reward = torch.zeros(B, dtype=torch.float32)
@@ -66,8 +73,9 @@ class BreakoutCpu(StatefulVecEnv):
         self._ep_return.zero_()
         self._ep_length.zero_()
         self._draw += 1
-        self._ext.breakout_reset(self._s, self._obs.view(self.num_envs, -1),
-                                 self.seed, int(self._draw.item()))
+        getattr(self._ext, self.RESET_FN)(
+            self._s, self._obs.view(self.num_envs, -1), self.seed,
+            int(self._draw.item()))
         B = self.num_envs
         return TimeStep(
             step_type=torch.full((B,), StepType.FIRST, dtype=torch.uint8),
@@ -85,7 +93,7 @@ class BreakoutCpu(StatefulVecEnv):
         )
 
     def step(self, action: Tensor) -> TimeStep:
-        self._ext.breakout_step(
+        getattr(self._ext, self.STEP_FN)(
             self._s, action.to(torch.int64).contiguous(), self._step_count,
             self._ep_return, self._ep_length, self._last_ep_return,
             self._last_ep_length, self._obs.view(self.num_envs, -1),
@@ -107,3 +115,24 @@ class BreakoutCpu(StatefulVecEnv):
                 },
             },
         )
+
+
+class PongCpu(_PoolEnvBase):
+    """Native-engine Atari-class Pong (scripted tracking opponent, first to
+    21; reward +-1 per point)."""
+
+    max_episode_steps = 5000
+    STATE_DIM_ATTR = "PONG_STATE_DIM"
+    RESET_FN = "pong_reset"
+    STEP_FN = "pong_step"
+    N_ACTIONS = 3  # noop / up / down
+
+
+class BreakoutCpu(_PoolEnvBase):
+    """Native-engine Breakout; drop-in for envs/breakout.py on CPU."""
+
+    max_episode_steps = 3000
+    STATE_DIM_ATTR = "STATE_DIM"
+    RESET_FN = "breakout_reset"
+    STEP_FN = "breakout_step"
+    N_ACTIONS = 4

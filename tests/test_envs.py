@@ -669,3 +669,24 @@ def test_rware_agent_blocking():
     s["carry"][0] = T.tensor([-1, -1, -1, -1])
     ts = env.step(T.tensor([[1, 0, 0, 0]]))  # agent 0 forward into agent 1
     assert env._state["agents"][0, 0].tolist() == [5, 5], "must be blocked"
+
+
+def test_native_pong_rules():
+    pytest.importorskip("stoix_amd.envs.build_envpool")
+    from stoix_amd.envs.envpool_cpu import PongCpu, envpool_ext
+
+    if envpool_ext() is None:
+        pytest.skip("envpool extension not built")
+    env = PongCpu(num_envs=4, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (4, 84, 84, 1)
+    # both paddles rendered (columns 3-4 and 80-81 have pixels)
+    assert (ts.observation[:, :, 80, 0] == 1.0).any()
+    assert (ts.observation[:, :, 3, 0] == 1.0).any()
+    # random play: the tracking opponent scores -> negative total reward
+    tot = torch.zeros(4)
+    g = torch.Generator().manual_seed(1)
+    for _ in range(600):
+        ts = env.step(torch.randint(0, 3, (4,), generator=g))
+        tot += ts.reward
+    assert float(tot.sum()) < 0, "opponent must out-score random play"
