@@ -105,6 +105,9 @@ def _envpool(name: str):
         "Breakout-v5": breakout_maker,
         "pong": pong_maker,
         "Pong-v5": pong_maker,
+        # classic-control names envpool also serves
+        "cartpole": CartPole,
+        "CartPole-v1": CartPole,
     }
     if name not in table:
         raise ValueError(f"unknown envpool-suite env '{name}' (have {list(table)})")
