@@ -94,6 +94,8 @@ def _envpool(name: str):
                 return BreakoutCpu(**kw)
         return Breakout(**kw)
 
+    from stoix_amd.envs.lunarlander import LunarLander
+
     def pong_maker(**kw):
         from stoix_amd.envs.envpool_cpu import PongCpu, envpool_ext
 
@@ -108,6 +110,8 @@ def _envpool(name: str):
         # classic-control names envpool also serves
         "cartpole": CartPole,
         "CartPole-v1": CartPole,
+        "lunarlander": LunarLander,
+        "LunarLander-v2": LunarLander,
     }
     if name not in table:
         raise ValueError(f"unknown envpool-suite env '{name}' (have {list(table)})")

@@ -690,3 +690,36 @@ def test_native_pong_rules():
         ts = env.step(torch.randint(0, 3, (4,), generator=g))
         tot += ts.reward
     assert float(tot.sum()) < 0, "opponent must out-score random play"
+
+
+def test_lunarlander_rules():
+    from stoix_amd.envs.lunarlander import LunarLander
+    import torch as T
+
+    env = LunarLander(num_envs=8, device="cpu", seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (8, 8)
+    # free fall (noop) must end in a crash (big negative terminal reward)
+    crashed = False
+    for _ in range(120):
+        ts = env.step(T.zeros(8, dtype=T.long))
+        assert T.isfinite(ts.reward).all()
+        if bool(ts.extras["episode_metrics"]["is_terminal_step"].any()):
+            crashed = True
+            assert float(ts.reward.min()) < -50.0
+            break
+    assert crashed, "free fall must crash"
+
+    # a gentle on-pad touchdown scores the +100 landing bonus
+    env2 = LunarLander(num_envs=1, device="cpu", seed=1)
+    env2.reset()
+    s = env2._state["s"]
+    s[0] = T.tensor([0.0, 0.12, 0.0, -0.01, 0.0, 0.0])
+    landed = False
+    for _ in range(10):  # settle onto the pad within a few steps
+        ts = env2.step(T.zeros(1, dtype=T.long))
+        if bool(ts.extras["episode_metrics"]["is_terminal_step"][0]):
+            landed = True
+            assert float(ts.reward[0]) > 50.0, "gentle pad touchdown must pay +100"
+            break
+    assert landed
