@@ -104,6 +104,12 @@ class LunarLander(StatefulVecEnv):
 
         ly, ry = self._legs(s)
         lc, rc = ly <= 0.0, ry <= 0.0
+        # shaping potential BEFORE the contact clamps: zeroing the descent
+        # velocity on touchdown is a discontinuous state change, and
+        # evaluating the potential after it would hand out a free
+        # +100*|vy| for every leg tap (a farmable reward exploit a CPU
+        # PPO run actually found: +2874-return hover-tap cycles)
+        phi1 = self._potential(s, lc, rc)
         # leg contacts: hold the lander up (simple support: zero downward
         # motion, damp horizontal drift)
         supported = lc & rc
@@ -124,7 +130,6 @@ class LunarLander(StatefulVecEnv):
             & (s[:, 2].abs() < 0.05) & (s[:, 3].abs() < 0.05)
             & (s[:, 4].abs() < 0.2) & (s[:, 5].abs() < 0.1)
         )
-        phi1 = self._potential(s, lc, rc)
         reward = (
             (phi1 - phi0)
             - MAIN_COST * main - SIDE_COST * (left + right)
