@@ -13,7 +13,7 @@ run() {
 }
 
 run stoix_amd.systems.ppo.ff_ppo              "" env=classic/cartpole system.num_minibatches=2 system.epochs=1
-run stoix_amd.systems.ppo.ff_ppo              "" env=classic/pendulum system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.ff_ppo              "" env=classic/pendulum network=mlp_continuous system=ppo/ff_ppo_continuous system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo_penalty      "" system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_dpo              "" system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.rec_ppo             "" system.num_minibatches=2 system.epochs=1
