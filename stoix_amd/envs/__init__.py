@@ -2,11 +2,20 @@
 
 Parity with /root/reference/stoix/utils/make_env.py: an ``ENV_MAKERS``
 dispatch over suites (:420-433) and ``make(config) -> (train_env, eval_env)``
-(:436-466). Suites available offline: ``classic`` (gymnax-equivalent
-CartPole/Pendulum/MountainCar/Acrobot), ``brax`` (Ant/Humanoid/HalfCheetah/
-Hopper-class physics), and
-``debug`` (five diagnostic games). External-binding suites (jumanji, envpool
-Atari, ...) require packages absent from this image and raise a clear error.
+(:436-466). Suites implemented offline:
+
+  * ``classic``/``gymnax`` — CartPole/Pendulum/MountainCar(+continuous)/
+    Acrobot plus the MinAtar grid games (Freeway, SpaceInvaders, Asterix,
+    Breakout-MinAtar);
+  * ``brax`` — Ant / Humanoid / HalfCheetah / Hopper-class physics;
+  * ``jumanji`` — Snake, Game2048, Connector, Sokoban, RobotWarehouse;
+  * ``envpool`` — Breakout + Pong via the native C++ batched CPU pool
+    (envs/csrc/envpool_cpu.cpp), LunarLander, and the cartpole alias;
+  * ``debug`` — five diagnostic games.
+
+JAX-only external suites with no offline equivalent (craftax, xland,
+navix, kinetix, popjym, playground, jaxarc) raise a clear error naming
+what is available.
 """
 from __future__ import annotations
 
