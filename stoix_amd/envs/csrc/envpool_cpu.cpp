@@ -354,14 +354,22 @@ void pong_cpu_step(torch::Tensor state, torch::Tensor action,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  // gil_scoped_release: the whole fused step runs WITHOUT the GIL so
+  // Sebulba's learner thread and sibling actor threads keep running
+  // while a batch of envs steps (the reference's envpool releases the
+  // GIL for exactly this reason)
   m.def("pong_reset", &pong_cpu_reset,
-        "batched Pong reset + render (CPU, threaded)");
+        "batched Pong reset + render (CPU, threaded)",
+        py::call_guard<py::gil_scoped_release>());
   m.def("pong_step", &pong_cpu_step,
-        "batched Pong fused step: physics + opponent AI + metrics + autoreset + render");
+        "batched Pong fused step: physics + opponent AI + metrics + autoreset + render",
+        py::call_guard<py::gil_scoped_release>());
   m.attr("PONG_STATE_DIM") = pong::SDIM;
   m.def("breakout_reset", &breakout_cpu_reset,
-        "batched Breakout reset + render (CPU, threaded)");
+        "batched Breakout reset + render (CPU, threaded)",
+        py::call_guard<py::gil_scoped_release>());
   m.def("breakout_step", &breakout_cpu_step,
-        "batched Breakout fused step: physics + metrics + autoreset + render");
+        "batched Breakout fused step: physics + metrics + autoreset + render",
+        py::call_guard<py::gil_scoped_release>());
   m.attr("STATE_DIM") = SDIM;
 }
