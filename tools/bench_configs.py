@@ -121,6 +121,9 @@ def bench_sebulba_breakout(quick):
     from stoix_amd.config import compose
     from stoix_amd.systems.ppo.sebulba_ff_ppo import run_experiment
 
+    # operating point swept on MI355X: 256 envs x 4 actor threads = 9.0K
+    # SPS; 512 x 8 drops to 5.1K (python-thread thrash), 32 x 2 to 1.2K
+    # (startup-dominated)
     n_updates = 4 if quick else 24
     n_envs = 32 if quick else 256
     rollout = 32 if quick else 64
