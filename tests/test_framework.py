@@ -202,3 +202,43 @@ def test_ppo_obs_normalization_reference_semantics():
     # the stored trajectory is the NORMALISED one after rollout_phase
     learner.rollout_phase()
     assert float(learner.buf_obs.mean().abs()) < 1.0
+
+
+def test_learner_checkpoint_roundtrip(tmp_path):
+    """Full save -> fresh learner -> restore cycle: the restored learner's
+    greedy actions match the original exactly (BASELINE checkpoint-format
+    parity: the recovery story is restart-from-checkpoint)."""
+    import torch
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.checkpointing import Checkpointer
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8", "arch.total_timesteps=null",
+         "arch.num_updates=2", "arch.num_evaluation=1", "system.rollout_length=8",
+         "system.num_minibatches=2", "system.epochs=1", "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    torch.manual_seed(0)
+    env = environments.make_single(cfg, 8, torch.device("cpu"), seed=0)
+    learner = PPOLearner(cfg, env, torch.device("cpu"))
+    for _ in range(2):
+        learner.update_step()
+    ckpt = Checkpointer("rt", {"algo": "ppo"}, directory=str(tmp_path))
+    ckpt.save(16, learner.state_for_checkpoint(), metric_value=1.0)
+
+    torch.manual_seed(123)  # different init
+    env2 = environments.make_single(cfg, 8, torch.device("cpu"), seed=0)
+    learner2 = PPOLearner(cfg, env2, torch.device("cpu"))
+    restored = ckpt.restore_params(learner2.state_for_checkpoint())
+    learner2.load_params(restored)
+
+    obs = torch.randn(16, 4)
+    a1 = learner.act_fn(obs, True)
+    a2 = learner2.act_fn(obs, True)
+    torch.testing.assert_close(a1, a2)
