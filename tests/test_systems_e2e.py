@@ -169,3 +169,32 @@ def test_ff_reinforce_learns_identity_game():
     r = run(cfg)
     # REINFORCE is high-variance: 200 updates reach ~9.8 (optimal 10)
     assert r > 6.0, f"REINFORCE failed to learn identity game: return={r}"
+
+
+def test_rec_ppo_learns_memory_game():
+    """Recurrent PPO must solve the sequence debug game (reward only for
+    recalling the first-step symbol at the end: pure memory — feed-forward
+    policies cap at chance; validates the RNN + done-masked hidden reset
+    path end-to-end)."""
+    from stoix_amd.systems.ppo.rec_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_rec_ppo.yaml",
+        [
+            "env=debug/sequence",
+            "arch.total_num_envs=128",
+            "arch.total_timesteps=null",
+            "arch.num_updates=120",
+            "arch.num_evaluation=1",
+            "arch.num_eval_episodes=32",
+            "arch.absolute_metric=false",
+            "system.rollout_length=16",
+            "system.num_minibatches=4",
+            "system.epochs=4",
+            "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    # optimal 1.0, chance ~0.25; measured 0.97 at this budget
+    assert r > 0.8, f"rec_ppo failed the memory game: return={r}"
