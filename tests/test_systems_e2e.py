@@ -198,3 +198,30 @@ def test_rec_ppo_learns_memory_game():
     r = run(cfg)
     # optimal 1.0, chance ~0.25; measured 0.97 at this budget
     assert r > 0.8, f"rec_ppo failed the memory game: return={r}"
+
+
+def test_ff_ppo_discount_sensitivity():
+    """With gamma=0.99 the delayed +1.0 beats the immediate +0.6: the agent
+    must learn to WAIT (verifies GAE/discount plumbing is not silently
+    myopic). With gamma=0.5 the immediate arm would win instead."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        [
+            "env=debug/discount_sensitive",
+            "arch.total_num_envs=64",
+            "arch.total_timesteps=null",
+            "arch.num_updates=40",
+            "arch.num_evaluation=1",
+            "arch.num_eval_episodes=32",
+            "arch.absolute_metric=false",
+            "system.rollout_length=16",
+            "system.num_minibatches=4",
+            "system.epochs=4",
+            "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r > 0.9, f"agent should wait for the delayed +1.0 at gamma=0.99: {r}"
