@@ -63,7 +63,11 @@ class ItemBuffer:
     def sample(self, batch_size: int) -> Dict[str, Tensor]:
         gen = None if self.graph_safe_rng else self.gen
         u = torch.rand(batch_size, device=self.device, generator=gen)
-        idx = (u * self._size.to(torch.float32)).long().clamp_(max=self.capacity - 1)
+        # defensive bound against size-1 (not capacity-1): an index in
+        # [size, capacity) would read unwritten zero slots pre-fill
+        idx = torch.minimum(
+            (u * self._size.to(torch.float32)).long(), self._size - 1
+        )
         return {k: v[idx] for k, v in self.storage.items()}
 
     @property

@@ -52,6 +52,10 @@ class StatefulVecEnv:
         self._ep_length = torch.zeros(self.num_envs, dtype=torch.int32, device=self.device)
         self._last_ep_return = torch.zeros(self.num_envs, dtype=torch.float32, device=self.device)
         self._last_ep_length = torch.zeros(self.num_envs, dtype=torch.int32, device=self.device)
+        # running count of completed episodes (device scalar; graph-legal):
+        # lets graph-replay learners tell fresh latched metrics from stale
+        # ones without a host sync
+        self._done_count = torch.zeros((), dtype=torch.long, device=self.device)
 
     # ------------------------------------------------------- subclass hooks
 
@@ -91,6 +95,7 @@ class StatefulVecEnv:
 
         self._ep_return += reward
         self._ep_length += 1
+        self._done_count += done.sum()
         # Latch completed-episode metrics at the terminal step.
         self._last_ep_return = torch.where(done, self._ep_return, self._last_ep_return)
         self._last_ep_length = torch.where(done, self._ep_length, self._last_ep_length)
@@ -151,6 +156,23 @@ class StatefulVecEnv:
 
     def randint(self, high: int, *shape) -> Tensor:
         return torch.randint(0, high, shape, device=self.device, generator=self.gen)
+
+
+def latched_episode_metrics(env: "StatefulVecEnv", learner) -> Dict[str, Tensor]:
+    """Episode metrics for graph-replay learners, read from the env's
+    latched device buffers with a freshness flag: ``has_final`` is a device
+    bool that is True only when at least one episode completed since the
+    previous call (no host sync; the flag is resolved host-side at log
+    time). Closes the stale-latched-metrics gap from ADVICE r1."""
+    dc = env._done_count.clone()
+    prev = getattr(learner, "_dc_prev", None)
+    fresh = (dc > prev) if prev is not None else (dc > 0)
+    learner._dc_prev = dc
+    return {
+        "episode_return": env._last_ep_return,
+        "episode_length": env._last_ep_length.to(torch.float32),
+        "has_final": fresh,
+    }
 
 
 def get_final_step_metrics(metrics: Dict[str, Tensor]) -> Tuple[Dict[str, Tensor], bool]:

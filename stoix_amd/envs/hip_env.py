@@ -96,6 +96,7 @@ class HipStepMixin:
             draw_offset,
             1 if do_bump else 0,
         )
+        self._done_count += hb["done"].sum()
 
     def step(self, action: torch.Tensor) -> TimeStep:  # type: ignore[override]
         if getattr(self, "_hip", None) is None:
@@ -122,6 +123,7 @@ class HipStepMixin:
             0,
             1,
         )
+        self._done_count += hb["done"].sum()
         return TimeStep(
             step_type=hb["steptype"].clone(),
             reward=hb["reward"].clone(),

@@ -110,23 +110,27 @@ class LunarLander(StatefulVecEnv):
         # +100*|vy| for every leg tap (a farmable reward exploit a CPU
         # PPO run actually found: +2874-return hover-tap cycles)
         phi1 = self._potential(s, lc, rc)
+        # crash condition on the PRE-clamp impact state: the support clamp
+        # below zeroes vy for two-leg-supported rows, so checking velocity
+        # after it would score a full-speed slam as a soft landing (another
+        # farmable exploit of the same family as the potential-order one)
+        impact_vy, impact_ang = s[:, 3], s[:, 4]
+        supported = lc & rc
+        body_down = s[:, 1] <= 0.02
+        crash = (
+            (body_down & ~supported)
+            | ((lc | rc) & ((impact_vy.abs() > 0.6) | (impact_ang.abs() > 0.6)))
+            | (s[:, 0].abs() > X_LIM)
+        )
         # leg contacts: hold the lander up (simple support: zero downward
         # motion, damp horizontal drift)
-        supported = lc & rc
         s[:, 3] = torch.where(supported & (s[:, 3] < 0), torch.zeros_like(s[:, 3]), s[:, 3])
         s[:, 2] = torch.where(supported, s[:, 2] * 0.7, s[:, 2])
         s[:, 5] = torch.where(supported, s[:, 5] * 0.7, s[:, 5])
         s[:, 1] = torch.where(supported & (s[:, 1] < 0.1), torch.full_like(s[:, 1], 0.1), s[:, 1])
-
-        body_down = s[:, 1] <= 0.02
-        crash = (
-            (body_down & ~supported)
-            | ((lc | rc) & ((s[:, 3].abs() > 0.6) | (s[:, 4].abs() > 0.6)))
-            | (s[:, 0].abs() > X_LIM)
-        )
         on_pad = s[:, 0].abs() <= PAD_W
         rest = (
-            supported & on_pad
+            supported & on_pad & ~crash
             & (s[:, 2].abs() < 0.05) & (s[:, 3].abs() < 0.05)
             & (s[:, 4].abs() < 0.2) & (s[:, 5].abs() < 0.1)
         )

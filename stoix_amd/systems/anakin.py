@@ -47,6 +47,10 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
     )
 
     learner = learner_factory(config, train_env, device)
+    # search-based systems (AZ / sampled-AZ / SPO) evaluate with search
+    # acting when an env handle is available (reference
+    # stoix/systems/search/evaluator.py); plain systems ignore this.
+    learner._eval_env_ref = eval_env
 
     # hip-graph capture of the update loop (rollout graph + epoch graph)
     # for capture-shaped learners on GPU (ops/graph.py); eager otherwise
@@ -103,7 +107,10 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             logger.log({"steps_per_second": sps, **train_metrics}, t_env, eval_idx, LogEvent.TRAIN)
             ep_metrics = getattr(learner, "episode_metrics", None)
             if ep_metrics:
-                logger.log(ep_metrics, t_env, eval_idx, LogEvent.ACT)
+                ep_metrics = dict(ep_metrics)
+                has_final = ep_metrics.pop("has_final", None)
+                if has_final is None or bool(has_final):
+                    logger.log(ep_metrics, t_env, eval_idx, LogEvent.ACT)
 
             eval_metrics = eval_fn(learner.act_fn, rnn=rnn)
             mean_return = float(eval_metrics["episode_return"].mean())

@@ -283,11 +283,11 @@ class PPOLearner:
         self.reducer._stream = None
 
     def after_graph_replay(self) -> None:
-        # episode metrics read eagerly from the env's latched buffers
-        self.episode_metrics = {
-            "episode_return": self.env._last_ep_return.mean(),
-            "episode_length": self.env._last_ep_length.float().mean(),
-        }
+        # episode metrics read eagerly from the env's latched buffers, with
+        # a device freshness flag (resolved at log time)
+        from stoix_amd.envs.env import latched_episode_metrics
+
+        self.episode_metrics = latched_episode_metrics(self.env, self)
 
     # ------------------------------------------------------------ checkpoint
 

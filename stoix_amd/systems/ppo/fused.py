@@ -433,6 +433,7 @@ class FusedPPOEngine:
             ext.bump_add(self.draw_policy, L.T)
             ext.bump_add(hb["draw"], L.T)
             L.buf_truncated.copy_(self.buf_steptype == 3)
+            env._done_count += done.sum()
             return
         for t in range(L.T):
             ext.policy_value_step(

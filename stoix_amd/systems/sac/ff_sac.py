@@ -193,10 +193,9 @@ class SACLearner:
         self.buffer.graph_safe_rng = True
 
     def after_graph_replay(self) -> None:
-        self.episode_metrics = {
-            "episode_return": self.env._last_ep_return,
-            "episode_length": self.env._last_ep_length.to(torch.float32),
-        }
+        from stoix_amd.envs.env import latched_episode_metrics
+
+        self.episode_metrics = latched_episode_metrics(self.env, self)
 
     # ------------------------------------------------------------ checkpoint
 

@@ -104,6 +104,11 @@ class SPOLearner:
         pobs = obs.repeat_interleave(P, dim=0)
         v = self.critic(pobs)  # [B*P]
         logw = torch.zeros(B, P, device=self.device)
+        # un-reset advantage accumulator for the temperature dual: follows
+        # particle lineage through resampling but is NOT zeroed by it (the
+        # dual must see the whole search's advantages, not the post-resample
+        # tail — ADVICE r1)
+        adv_acc = torch.zeros(B, P, device=self.device)
         alive = torch.ones(B * P, device=self.device)
         root_action = None
 
@@ -117,6 +122,7 @@ class SPOLearner:
             v_next = self.critic(pobs)
             td = (reward + self.gamma * v_next * (~terminated).float() - v) * alive
             logw = logw + (td / eta).view(B, P)
+            adv_acc = adv_acc + td.view(B, P)
             alive = alive * (~terminated).float()
             v = v_next
             # ESS-triggered systematic resampling (per env)
@@ -134,6 +140,9 @@ class SPOLearner:
                 pobs = torch.where(_expand(sel, pobs), pobs[flat_idx], pobs)
                 v = torch.where(sel, v[flat_idx], v)
                 alive = torch.where(sel, alive[flat_idx], alive)
+                adv_acc = torch.where(
+                    need.unsqueeze(1), adv_acc.view(-1)[flat_idx].view(B, P), adv_acc
+                )
                 ra_flat = root_action.view(B * P, *root_action.shape[2:])
                 ra_new = torch.where(_expand(sel, ra_flat), ra_flat[flat_idx], ra_flat)
                 root_action = ra_new.view_as(root_action)
@@ -143,8 +152,9 @@ class SPOLearner:
         # particle-weighted search value at the root: V(s) + weighted
         # advantage signal is approximated by the weighted bootstrap values
         search_value = (weights * v.view(B, P)).sum(-1)
-        # record advantages for the temperature dual (pre-normalised)
-        self._adv_buffer = (logw * eta).detach()
+        # record the full search's advantages for the temperature dual
+        # (lineage-tracked accumulator, unaffected by resample resets)
+        self._adv_buffer = adv_acc.detach()
         return root_action, weights, search_value
 
     @torch.no_grad()
@@ -173,12 +183,14 @@ class SPOLearner:
     def update_step(self) -> Dict[str, Tensor]:
         T, B, P = self.T, self.B, self.P
         obs_l, ra_l, w_l, sv_l, rew_l, disc_l, trunc_l = [], [], [], [], [], [], []
+        adv_l = []
         ts = self.ts
         with torch.no_grad():
             for _ in range(T):
                 obs = ts.observation
                 root_state = {k: v.clone() for k, v in self.env._state.items()}
                 action, root_actions, weights, search_value = self._search_root(obs, root_state)
+                adv_l.append(self._adv_buffer)
                 next_ts = self.env.step(action)
                 obs_l.append(obs.clone())
                 ra_l.append(root_actions)
@@ -207,10 +219,11 @@ class SPOLearner:
             r_t, d_t, float(self.sys.gae_lambda), sv_t, sv_next, truncation_t=trunc_t
         )
 
-        # ---- temperature dual step on the search advantages
-        if self._adv_buffer is not None:
+        # ---- temperature dual step on the search advantages, aggregated
+        # over ALL T per-step searches of the rollout (not just the last)
+        if adv_l:
             eta = self.log_eta.exp()
-            A = self._adv_buffer  # [B, P] raw advantage sums
+            A = torch.cat(adv_l, dim=0)  # [T*B, P] raw advantage sums
             dual = eta * self.kl_epsilon + eta * torch.logsumexp(
                 A / eta - math.log(A.shape[-1]), dim=-1
             ).mean()
