@@ -24,21 +24,55 @@ outside the graphs.
 """
 from __future__ import annotations
 
+import os
 from typing import Dict
 
 import torch
 
 
+def _world_size() -> int:
+    import torch.distributed as dist
+
+    return dist.get_world_size() if dist.is_initialized() else 1
+
+
+def _warm_collectives(device: torch.device) -> None:
+    """Run one eager all-reduce before any capture: the RCCL communicator
+    (ring setup over xGMI) must be created OUTSIDE graph capture — the
+    first collective builds it, and building it inside a capture fails.
+    Also doubles as a fail-fast connectivity check at bench start."""
+    import torch.distributed as dist
+
+    if dist.is_initialized() and dist.get_world_size() > 1:
+        t = torch.ones(1, device=device)
+        dist.all_reduce(t)
+        if device.type == "cuda":
+            # warm the bf16 path too (the fused engine all-reduces bf16)
+            tb = torch.ones(8, dtype=torch.bfloat16, device=device)
+            dist.all_reduce(tb)
+            torch.cuda.synchronize(device)
+
+
 def try_enable_graphs(learner) -> bool:
     """Capture learner's rollout/epoch phases into hip graphs and
-    monkey-patch update_step to replay them. Returns True on success."""
+    monkey-patch update_step to replay them. Returns True on success.
+
+    Knobs (multi-GPU first-try safety):
+      STOIX_NO_GRAPH=1             — skip capture entirely (pure eager).
+      STOIX_FUSED_EAGER_ALLREDUCE=1 — at world>1, capture only the rollout
+        graph and run the epoch phase eagerly, so the per-minibatch RCCL
+        all-reduce is an ordinary eager call instead of a graph node.
+    """
     device = learner.device
     if device.type != "cuda":
+        return False
+    if os.environ.get("STOIX_NO_GRAPH"):
         return False
     if getattr(learner.env, "_hip", None) is None:
         raise RuntimeError("env has no HIP step kernel; graph capture needs it")
 
     learner.prepare_for_graph_capture()
+    _warm_collectives(device)
 
     # eager warmup on a side stream (rocBLAS/hipBLASLt workspaces, autotune)
     side = torch.cuda.Stream(device)
@@ -55,12 +89,28 @@ def try_enable_graphs(learner) -> bool:
     with torch.cuda.graph(g_rollout):
         learner.rollout_phase()
 
+    epochs = int(learner.sys.epochs)
+    eager_epoch = (
+        _world_size() > 1 and os.environ.get("STOIX_FUSED_EAGER_ALLREDUCE") == "1"
+    )
+    if eager_epoch:
+
+        def update_step() -> Dict[str, torch.Tensor]:
+            g_rollout.replay()
+            for _ in range(epochs):
+                learner._new_perm()
+                metrics = learner.epoch_phase()
+            learner.after_graph_replay()
+            return metrics
+
+        learner._graphs = (g_rollout,)
+        learner.update_step = update_step
+        return True
+
     learner._new_perm()
     g_epoch = torch.cuda.CUDAGraph()
     with torch.cuda.graph(g_epoch):
         static_metrics = learner.epoch_phase()
-
-    epochs = int(learner.sys.epochs)
 
     def update_step() -> Dict[str, torch.Tensor]:
         g_rollout.replay()
@@ -86,9 +136,16 @@ def try_enable_update_graph(learner) -> bool:
     device = learner.device
     if device.type != "cuda":
         return False
+    if os.environ.get("STOIX_NO_GRAPH"):
+        return False
+    if _world_size() > 1 and os.environ.get("STOIX_FUSED_EAGER_ALLREDUCE") == "1":
+        # whole-update capture would put the RCCL all-reduce inside the
+        # graph; the knob demands eager collectives, so skip capture
+        return False
     if getattr(learner.env, "_hip", None) is None:
         raise RuntimeError("update-graph capture needs the HIP env step path")
     learner.prepare_for_graph_capture()
+    _warm_collectives(device)
 
     side = torch.cuda.Stream(device)
     side.wait_stream(torch.cuda.current_stream(device))
