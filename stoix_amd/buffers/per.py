@@ -1,13 +1,17 @@
-"""Prioritised replay with a vectorised device-resident sum-tree.
+"""Prioritised replay with a device-resident sum-tree.
 
 Functional parity with flashbax ``make_prioritised_trajectory_buffer`` as
 used by Rainbow / R2D2 (/root/reference/stoix/systems/q_learning/
 ff_rainbow.py:433-444 incl. ``set_priorities``, rec_r2d2 sequences).
 
-The sum-tree is a flat level-order array of size 2*cap; batched updates are
-scatter-adds per level and batched sampling is a log2(cap)-step descent done
-with gathers — both fully vectorised device ops (the same structure the HIP
-kernel version uses; each level touch is one coalesced gather).
+The sum-tree is a flat level-order array of size 2*cap. Every operation is
+capture-legal (no torch.unique, no host syncs, device max-priority cursor),
+so the whole Rainbow update — sample, loss, priority writeback — can be
+captured into ONE hip graph (ops/graph.try_enable_update_graph). On GPU the
+update and the stratified sampling descent run as hand-written HIP kernels
+(ops/csrc/per.hip: scatter + idempotent per-level ancestor repair, one
+launch for batch-sized updates); the torch fallback does the identical
+per-level repair with strided tensor ops (CPU tests + numerics parity).
 """
 from __future__ import annotations
 
@@ -23,25 +27,42 @@ Tensor = torch.Tensor
 class SumTree:
     def __init__(self, capacity: int, device: torch.device | str = "cpu"):
         self.capacity = 1
+        self.depth = 0
         while self.capacity < capacity:
             self.capacity *= 2
+            self.depth += 1
         self.n_items = capacity
         self.device = torch.device(device)
         self.tree = torch.zeros(2 * self.capacity, device=self.device)
+        self._hip = None
+        if self.device.type == "cuda":
+            from stoix_amd import ops
+
+            self._hip = ops.ext(required=True)
 
     @torch.no_grad()
     def set(self, idx: Tensor, priority: Tensor) -> None:
-        """Set priorities at item indices (batched; duplicate idx keep the
-        last write via index_put)."""
-        leaf = idx.long() + self.capacity
+        """Set priorities at item indices (batched, capture-legal).
+
+        Duplicate indices: an arbitrary scatter winner, then an idempotent
+        per-level ancestor recompute (parent = sum of children read fresh)
+        — never double-counts, unlike atomic delta propagation."""
+        idx = idx.long()
+        priority = priority.float()
+        if self._hip is not None:
+            self._hip.sumtree_update(self.tree, idx.contiguous(),
+                                     priority.contiguous(), self.capacity,
+                                     self.depth)
+            return
+        leaf = idx + self.capacity
         self.tree[leaf] = priority
-        # rebuild ancestors of touched leaves level by level
-        nodes = torch.unique(leaf // 2)
-        while nodes.numel() > 0 and nodes[0] >= 1:
-            self.tree[nodes] = self.tree[2 * nodes] + self.tree[2 * nodes + 1]
-            nodes = torch.unique(nodes // 2)
-            if nodes.numel() == 1 and nodes[0] == 0:
-                break
+        node = leaf
+        for _ in range(self.depth):
+            node = node >> 1
+            # gather children first, then scatter: duplicate nodes write
+            # identical values, and reads (level l-1) never alias writes
+            # (level l)
+            self.tree[node] = self.tree[2 * node] + self.tree[2 * node + 1]
 
     @property
     def total(self) -> Tensor:
@@ -50,19 +71,22 @@ class SumTree:
     @torch.no_grad()
     def sample(self, batch_size: int, generator=None) -> Tensor:
         """Stratified proportional sampling: batched tree descent."""
-        seg = self.total / batch_size
         u = torch.rand(batch_size, device=self.device, generator=generator)
+        if self._hip is not None:
+            out = torch.empty(batch_size, dtype=torch.long, device=self.device)
+            self._hip.sumtree_sample(self.tree, u, out, self.capacity,
+                                     self.depth, self.n_items)
+            return out
+        seg = self.total / batch_size
         mass = (torch.arange(batch_size, device=self.device, dtype=torch.float32) + u) * seg
         node = torch.ones(batch_size, dtype=torch.long, device=self.device)
-        depth = int(torch.log2(torch.tensor(float(self.capacity))).item())
-        for _ in range(depth):
+        for _ in range(self.depth):
             left = 2 * node
             left_sum = self.tree[left]
             go_right = mass >= left_sum
             mass = torch.where(go_right, mass - left_sum, mass)
             node = torch.where(go_right, left + 1, left)
-        item = (node - self.capacity).clamp(0, self.n_items - 1)
-        return item
+        return (node - self.capacity).clamp(0, self.n_items - 1)
 
     def get(self, idx: Tensor) -> Tensor:
         return self.tree[idx.long() + self.capacity]
@@ -70,7 +94,8 @@ class SumTree:
 
 class PrioritisedBuffer(TrajectoryBuffer):
     """Prioritised sequence buffer: proportional sampling with alpha-powered
-    priorities and IS weights (1/(N p))^beta / max."""
+    priorities and IS weights (1/(N p))^beta / max. Fully device-resident
+    cursors (graph-capturable end to end)."""
 
     def __init__(
         self,
@@ -87,7 +112,8 @@ class PrioritisedBuffer(TrajectoryBuffer):
         # one priority per (row, t0) start slot
         self.n_slots = self.rows * self.t_max
         self.tree = SumTree(self.n_slots, device)
-        self._max_priority = 1.0
+        self._max_priority = torch.ones((), device=self.device)
+        self._rowss = torch.arange(self.rows, device=self.device)
 
     def _slot(self, rows: Tensor, t0: Tensor) -> Tensor:
         return rows * self.t_max + t0
@@ -95,26 +121,23 @@ class PrioritisedBuffer(TrajectoryBuffer):
     @torch.no_grad()
     def add(self, batch: Dict[str, Tensor]) -> None:
         t_block = next(iter(batch.values())).shape[1]
-        t_start = self.t_ptr
+        t_start = self._t_ptr.clone()  # device cursor, pre-advance
         super().add(batch)
         # new items get max priority so they are sampled at least once
         offs = (torch.arange(t_block, device=self.device) + t_start) % self.t_max
-        rows = torch.arange(self.rows, device=self.device)
-        slots = (rows.unsqueeze(1) * self.t_max + offs.unsqueeze(0)).reshape(-1)
-        # only starts with a full valid window ahead are sampleable; priority
-        # zero marks unsampleable slots. A start is valid when its whole
-        # window lies in filled data; approximate by marking slots older than
-        # seq_len behind the pointer valid (exact masking at sample()).
-        self.tree.set(slots, torch.full((slots.numel(),), self._max_priority**self.alpha, device=self.device))
+        slots = (self._rowss.unsqueeze(1) * self.t_max + offs.unsqueeze(0)).reshape(-1)
+        fill = (self._max_priority ** self.alpha).expand(slots.numel()).contiguous()
+        self.tree.set(slots, fill)
         # invalidate the seq_len-1 slots straight behind the new pointer
         # (their windows would cross the write head)
-        inv = (torch.arange(self.seq_len - 1, device=self.device) + self.t_ptr - (self.seq_len - 1)) % self.t_max
-        inv_slots = (rows.unsqueeze(1) * self.t_max + inv.unsqueeze(0)).reshape(-1)
+        inv = (torch.arange(self.seq_len - 1, device=self.device) + self._t_ptr - (self.seq_len - 1)) % self.t_max
+        inv_slots = (self._rowss.unsqueeze(1) * self.t_max + inv.unsqueeze(0)).reshape(-1)
         self.tree.set(inv_slots, torch.zeros(inv_slots.numel(), device=self.device))
 
     @torch.no_grad()
-    def sample(self, batch_size: int, importance_sampling_exponent: float = 0.4) -> Dict[str, Tensor]:
-        slots = self.tree.sample(batch_size, self.gen)
+    def sample(self, batch_size: int, importance_sampling_exponent=0.4) -> Dict[str, Tensor]:
+        gen = None if self.graph_safe_rng else self.gen
+        slots = self.tree.sample(batch_size, gen)
         rows = slots // self.t_max
         t0 = slots % self.t_max
         offs = torch.arange(self.seq_len, device=self.device)
@@ -123,7 +146,10 @@ class PrioritisedBuffer(TrajectoryBuffer):
         pr = self.tree.get(slots)
         probs = pr / self.tree.total.clamp(min=1e-12)
         n = (self.tree.tree[self.tree.capacity :] > 0).sum().clamp(min=1)
-        weights = (1.0 / (probs * n).clamp(min=1e-12)) ** importance_sampling_exponent
+        beta = importance_sampling_exponent
+        if not torch.is_tensor(beta):
+            beta = torch.as_tensor(float(beta), device=self.device)
+        weights = (1.0 / (probs * n).clamp(min=1e-12)) ** beta
         weights = weights / weights.max().clamp(min=1e-12)
         out["_rows"] = rows
         out["_t0"] = t0
@@ -134,7 +160,10 @@ class PrioritisedBuffer(TrajectoryBuffer):
     @torch.no_grad()
     def set_priorities(self, slots: Tensor, priorities: Tensor) -> None:
         priorities = priorities.abs().clamp(min=1e-6)
-        self._max_priority = max(self._max_priority, float(priorities.max()))
+        # in-place update of the STABLE max-priority buffer: add() reads this
+        # tensor's address inside the captured graph, so rebinding the
+        # attribute to a fresh tensor would leave replays reading stale data
+        self._max_priority.copy_(torch.maximum(self._max_priority, priorities.max()))
         self.tree.set(slots, priorities**self.alpha)
 
     @property

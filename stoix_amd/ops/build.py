@@ -21,6 +21,7 @@ SOURCES = [
     str(CSRC / "optim.hip"),
     str(CSRC / "mlp.hip"),
     str(CSRC / "wgrad.hip"),
+    str(CSRC / "per.hip"),
 ]
 
 

@@ -83,6 +83,10 @@ void launch_wgrad(const void*, const void*, float*, long, long, long, int,
 void launch_slab_reduce(float*, void*, long, long, float*, long*, void*);
 void launch_bump_add(unsigned int*, unsigned int, void*);
 void launch_tr16_probe(const void*, float*, int, void*);
+void launch_sumtree_update(float*, const long*, const float*, long, int, int,
+                           void*);
+void launch_sumtree_sample(const float*, const float*, long*, long, int, int,
+                           int, void*);
 }
 
 namespace {
@@ -478,7 +482,31 @@ void tr16_probe(torch::Tensor in, torch::Tensor out, int64_t base_mode) {
 
 }  // namespace
 
+void sumtree_update(torch::Tensor tree, torch::Tensor idx,
+                    torch::Tensor prio, int64_t cap, int64_t depth) {
+  CHK(tree, torch::kFloat32);
+  CHK(idx, torch::kInt64);
+  CHK(prio, torch::kFloat32);
+  launch_sumtree_update(tree.data_ptr<float>(), idx.data_ptr<long>(),
+                        prio.data_ptr<float>(), idx.numel(), (int)cap,
+                        (int)depth, cur_stream());
+}
+
+void sumtree_sample(torch::Tensor tree, torch::Tensor u, torch::Tensor out,
+                    int64_t cap, int64_t depth, int64_t n_items) {
+  CHK(tree, torch::kFloat32);
+  CHK(u, torch::kFloat32);
+  CHK(out, torch::kInt64);
+  launch_sumtree_sample(tree.data_ptr<float>(), u.data_ptr<float>(),
+                        out.data_ptr<long>(), u.numel(), (int)cap, (int)depth,
+                        (int)n_items, cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sumtree_update", &sumtree_update,
+        "scatter leaf priorities + repair ancestors (device sum-tree)");
+  m.def("sumtree_sample", &sumtree_sample,
+        "stratified proportional sum-tree descent");
   m.def("cartpole_step", &cartpole_step, "fused CartPole env step");
   m.def("ant_step", &ant_step, "fused Ant env step");
   m.def("ant_reset", &ant_reset, "Ant reset");

@@ -52,7 +52,10 @@ class RainbowLearner:
         for p in self.q_target.parameters():
             p.requires_grad_(False)
 
-        self.opt = torch.optim.Adam(self.q_online.parameters(), lr=float(self.sys.q_lr), eps=1e-5)
+        self.opt = torch.optim.Adam(
+            self.q_online.parameters(), lr=float(self.sys.q_lr), eps=1e-5,
+            capturable=device.type == "cuda",
+        )
         self.reducer = FlatGradReducer(self.q_online.parameters(), device)
         self.gen = torch.Generator(device=device)
         self.gen.manual_seed(int(config.arch.seed) * 7919 + 503)
@@ -68,9 +71,12 @@ class RainbowLearner:
         self.batch_size = int(self.sys.batch_size)
         self.beta0 = float(self.sys.importance_sampling_exponent)
         self.total_updates = max(1, int(config.arch.num_updates) * int(self.sys.epochs))
-        self.update_count = 0
+        # device update counter: the beta anneal is computed on-device so
+        # the whole update step stays hip-graph capturable
+        self._update_count = torch.zeros((), device=device)
 
         self.ts = env.reset()
+        self.collect_metrics = True
         self.episode_metrics: Dict[str, Tensor] = {}
         self._warmup()
 
@@ -112,10 +118,11 @@ class RainbowLearner:
                 "next_obs": torch.stack(next_l, 1),
             }
         )
-        em = ts.extras["episode_metrics"]
-        final, has = get_final_step_metrics(em)
-        if has:
-            self.episode_metrics = {k: v.mean() for k, v in final.items()}
+        if self.collect_metrics:
+            em = ts.extras["episode_metrics"]
+            final, has = get_final_step_metrics(em)
+            if has:
+                self.episode_metrics = {k: v.mean() for k, v in final.items()}
 
     def _warmup(self) -> None:
         steps = max(self.n_step + 1, int(getattr(self.sys, "warmup_steps", 64)) // self.B + 1)
@@ -128,9 +135,9 @@ class RainbowLearner:
         metrics: Dict[str, Tensor] = {}
         tau = float(self.sys.tau)
         for _ in range(int(self.sys.epochs)):
-            beta = self.beta0 + (1.0 - self.beta0) * min(
-                1.0, self.update_count / self.total_updates
-            )
+            beta = self.beta0 + (1.0 - self.beta0) * (
+                self._update_count / self.total_updates
+            ).clamp(max=1.0)
             batch = self.buffer.sample(self.batch_size, importance_sampling_exponent=beta)
             # n-step reward over the window; bootstrap from the window end
             r = batch["reward"]  # [B, n]
@@ -174,9 +181,30 @@ class RainbowLearner:
             with torch.no_grad():
                 for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
                     pt.mul_(1 - tau).add_(po, alpha=tau)
-            self.update_count += 1
-            metrics = {"q_loss": loss.detach(), "beta": torch.tensor(beta)}
+            self._update_count += 1
+            metrics = {"q_loss": loss.detach(), "beta": beta.detach()}
         return metrics
+
+    # ------------------------------------------------------- graph support
+
+    @property
+    def graph_capturable(self) -> bool:
+        return getattr(self.env, "_hip", None) is not None
+
+    def prepare_for_graph_capture(self) -> None:
+        """Capture-safe modes: default (graph-aware) CUDA RNG for noisy-net
+        resampling and buffer sampling, inline all-reduce, no host-side
+        metric reads. The sum-tree update/sample and the beta anneal are
+        already device-resident (buffers/per.py, ops/csrc/per.hip)."""
+        self.gen = None
+        self.collect_metrics = False
+        self.reducer._stream = None
+        self.buffer.graph_safe_rng = True
+
+    def after_graph_replay(self) -> None:
+        from stoix_amd.envs.env import latched_episode_metrics
+
+        self.episode_metrics = latched_episode_metrics(self.env, self)
 
     def state_for_checkpoint(self):
         return {"q_online": dict(self.q_online.state_dict())}
