@@ -40,6 +40,11 @@ class StatefulVecEnv:
     action_space: Space
     max_episode_steps: int = 10**9
     solved_return_threshold: Optional[float] = None
+    # subclass opt-in: _step_fn/_reset_fn/_obs_fn are free of host-synced
+    # branches, so step() can run under hip-graph capture (graph_mode takes
+    # the unconditional-autoreset path and the default graph-aware RNG)
+    capture_safe: bool = False
+    graph_mode: bool = False
 
     def __init__(self, num_envs: int, device: torch.device | str = "cpu", seed: int = 0):
         self.num_envs = int(num_envs)
@@ -104,7 +109,9 @@ class StatefulVecEnv:
 
         # Autoreset: regenerate state for done envs; observation returned is
         # the RESET obs there, the true final obs goes to extras["next_obs"].
-        if bool(done.any()):
+        # graph_mode takes this branch unconditionally (a host-synced
+        # bool(done.any()) is not capture-legal).
+        if self.graph_mode or bool(done.any()):
             fresh = self._reset_fn(self.num_envs)
             mask = done
             for k in self._state:
@@ -144,6 +151,13 @@ class StatefulVecEnv:
                 "is_terminal_step": done.clone(),
             },
         }
+
+    def prepare_for_graph_capture(self) -> None:
+        """Switch to capture-safe modes: unconditional autoreset and the
+        default (graph-aware) CUDA generator. Only valid when the subclass
+        declares ``capture_safe`` (or has a HIP step kernel)."""
+        self.graph_mode = True
+        self.gen = None
 
     # ------------------------------------------------------------- utility
 

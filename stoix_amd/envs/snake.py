@@ -33,6 +33,7 @@ _DC = [0, 1, 0, -1]
 
 class Snake(StatefulVecEnv):
     max_episode_steps = 4000
+    capture_safe = True  # no host-synced branches in graph_mode
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
@@ -44,13 +45,18 @@ class Snake(StatefulVecEnv):
     # ------------------------------------------------------------ state ops
 
     def _spawn_fruit(self, grid: Tensor, n: int) -> Tuple[Tensor, Tensor]:
-        """Uniform fruit position over empty cells per board: [n] rows, cols."""
-        empty = grid <= 0  # [n, R, C]
-        flat = empty.reshape(n, ROWS * COLS).float()
-        # multinomial over empty cells (there is always at least one until
-        # the board is full, at which point position is irrelevant)
-        probs = flat.clamp(min=1e-9)
-        idx = torch.multinomial(probs, 1, generator=self.gen).squeeze(-1)
+        """Uniform fruit position over empty cells per board: [n] rows, cols.
+
+        Gumbel-max over the empty mask (capture-legal, unlike
+        torch.multinomial): argmax of Gumbel noise restricted to empty
+        cells samples uniformly among them. There is always at least one
+        empty cell until the board is full, at which point position is
+        irrelevant."""
+        empty = grid.reshape(n, ROWS * COLS) <= 0  # [n, R*C]
+        u = torch.rand(n, ROWS * COLS, device=self.device, generator=self.gen)
+        gumbel = -torch.log(-torch.log(u.clamp(min=1e-12)).clamp(min=1e-12))
+        scores = torch.where(empty, gumbel, torch.full_like(gumbel, -torch.inf))
+        idx = scores.argmax(dim=-1)
         return idx // COLS, idx % COLS
 
     def _reset_fn(self, n: int) -> State:
@@ -106,7 +112,7 @@ class Snake(StatefulVecEnv):
         )
         # respawn fruit where eaten
         fr, fc = state["fruit_r"].clone(), state["fruit_c"].clone()
-        if bool(ate.any()):
+        if self.graph_mode or bool(ate.any()):
             nfr, nfc = self._spawn_fruit(grid, n)
             fr = torch.where(ate, nfr, fr)
             fc = torch.where(ate, nfc, fc)

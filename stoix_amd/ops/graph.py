@@ -142,8 +142,14 @@ def try_enable_update_graph(learner) -> bool:
         # whole-update capture would put the RCCL all-reduce inside the
         # graph; the knob demands eager collectives, so skip capture
         return False
-    if getattr(learner.env, "_hip", None) is None:
-        raise RuntimeError("update-graph capture needs the HIP env step path")
+    env = learner.env
+    if getattr(env, "_hip", None) is None and not getattr(env, "capture_safe", False):
+        raise RuntimeError(
+            "update-graph capture needs a HIP env step kernel or a "
+            "capture_safe torch env step"
+        )
+    if hasattr(env, "prepare_for_graph_capture"):
+        env.prepare_for_graph_capture()
     learner.prepare_for_graph_capture()
     _warm_collectives(device)
 

@@ -178,8 +178,8 @@ class DDPGFamilyLearner:
         # per call; a single captured graph is only valid at frequency 1
         return (
             getattr(self.env, "_hip", None) is not None
-            and int(getattr(self.sys, "policy_frequency", 1)) == 1
-        )
+            or getattr(self.env, "capture_safe", False)
+        ) and int(getattr(self.sys, "policy_frequency", 1)) == 1
 
     def prepare_for_graph_capture(self) -> None:
         """Capture-safe modes: default (graph-aware) CUDA RNG, inline

@@ -182,7 +182,10 @@ class SACLearner:
 
     @property
     def graph_capturable(self) -> bool:
-        return getattr(self.env, "_hip", None) is not None
+        return (
+            getattr(self.env, "_hip", None) is not None
+            or getattr(self.env, "capture_safe", False)
+        )
 
     def prepare_for_graph_capture(self) -> None:
         """Capture-safe modes: default (graph-aware) CUDA RNG, inline
