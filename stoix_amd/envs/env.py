@@ -57,6 +57,11 @@ class StatefulVecEnv:
         self._ep_length = torch.zeros(self.num_envs, dtype=torch.int32, device=self.device)
         self._last_ep_return = torch.zeros(self.num_envs, dtype=torch.float32, device=self.device)
         self._last_ep_length = torch.zeros(self.num_envs, dtype=torch.int32, device=self.device)
+        # device-resident step-type constants: creating them per step would
+        # be a pageable H2D copy — illegal inside hip-graph capture
+        self._st_terminated = torch.tensor(StepType.TERMINATED, dtype=torch.uint8, device=self.device)
+        self._st_truncated = torch.tensor(StepType.TRUNCATED, dtype=torch.uint8, device=self.device)
+        self._st_mid = torch.tensor(StepType.MID, dtype=torch.uint8, device=self.device)
         # running count of completed episodes (device scalar; graph-legal):
         # lets graph-replay learners tell fresh latched metrics from stale
         # ones without a host sync
@@ -126,12 +131,8 @@ class StatefulVecEnv:
 
         step_type = torch.where(
             terminated,
-            torch.tensor(StepType.TERMINATED, dtype=torch.uint8, device=self.device),
-            torch.where(
-                truncated,
-                torch.tensor(StepType.TRUNCATED, dtype=torch.uint8, device=self.device),
-                torch.tensor(StepType.MID, dtype=torch.uint8, device=self.device),
-            ),
+            self._st_terminated,
+            torch.where(truncated, self._st_truncated, self._st_mid),
         )
         discount = torch.where(terminated, 0.0, 1.0).to(torch.float32)
         return TimeStep(

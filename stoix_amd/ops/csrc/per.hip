@@ -29,16 +29,22 @@
 extern "C" __global__ void sumtree_update_single_kernel(
     float* __restrict__ tree, const long* __restrict__ idx,
     const float* __restrict__ prio, int n, int cap, int depth) {
+  // NB: s_barrier alone does NOT drain outstanding global stores on CDNA
+  // (it waits lgkmcnt, not vmcnt) — each level's writes must be fenced
+  // with a workgroup-scope fence (s_waitcnt vmcnt(0)) before the barrier
+  // or the next level reads stale children.
   for (int i = threadIdx.x; i < n; i += blockDim.x) {
     long leaf = (long)cap + idx[i];
     tree[leaf] = prio[i];
   }
+  __threadfence_block();
   __syncthreads();
   for (int l = 1; l <= depth; ++l) {
     for (int i = threadIdx.x; i < n; i += blockDim.x) {
       long node = ((long)cap + idx[i]) >> l;
       tree[node] = tree[2 * node] + tree[2 * node + 1];
     }
+    __threadfence_block();
     __syncthreads();
   }
 }
