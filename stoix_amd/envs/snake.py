@@ -41,6 +41,11 @@ class Snake(StatefulVecEnv):
         self.action_space = DiscreteSpace(4)
         self._dr = torch.tensor(_DR, device=self.device)
         self._dc = torch.tensor(_DC, device=self.device)
+        # device scalar constants: `t[bidx, r, c] = 1.0` with a python
+        # scalar wraps it in a CPU tensor and does a pageable H2D copy per
+        # call — illegal under hip-graph capture
+        self._one_f = torch.ones((), device=self.device)
+        self._one_i = torch.ones((), dtype=torch.int32, device=self.device)
 
     # ------------------------------------------------------------ state ops
 
@@ -64,7 +69,7 @@ class Snake(StatefulVecEnv):
         hr = torch.full((n,), ROWS // 2, dtype=torch.long, device=self.device)
         hc = torch.full((n,), COLS // 2, dtype=torch.long, device=self.device)
         bidx = torch.arange(n, device=self.device)
-        grid[bidx, hr, hc] = 1  # length-1 snake
+        grid[bidx, hr, hc] = self._one_i  # length-1 snake
         fr, fc = self._spawn_fruit(grid, n)
         return {
             "grid": grid,
@@ -82,10 +87,10 @@ class Snake(StatefulVecEnv):
         length = state["length"].clamp(min=1).float()
         body = (grid > 0).float()
         head = torch.zeros_like(body)
-        head[bidx, state["head_r"], state["head_c"]] = 1.0
+        head[bidx, state["head_r"], state["head_c"]] = self._one_f
         tail = (grid == 1).float()
         fruit = torch.zeros_like(body)
-        fruit[bidx, state["fruit_r"], state["fruit_c"]] = 1.0
+        fruit[bidx, state["fruit_r"], state["fruit_c"]] = self._one_f
         order = grid.float() / length.view(-1, 1, 1)
         return torch.stack([body, head, tail, fruit, order], dim=-1)
 

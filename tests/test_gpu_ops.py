@@ -351,10 +351,12 @@ def test_sumtree_hip_update_matches_cpu_reference(ext):
         cpu = SumTree(n_items, "cpu")
         gpu = SumTree(n_items, "cuda")
         g = torch.Generator().manual_seed(7)
-        idx = torch.randint(0, n_items, (n_upd,), generator=g)
-        # force duplicates with identical priorities (the defined case)
-        idx[: n_upd // 4] = idx[n_upd // 4 : n_upd // 2]
+        # UNIQUE base indices (duplicate indices with different priorities
+        # have an intentionally arbitrary scatter winner), plus controlled
+        # duplicates with identical priorities (the defined case)
+        idx = torch.randperm(n_items, generator=g)[:n_upd].contiguous()
         prio = torch.rand(n_upd, generator=g) + 0.01
+        idx[: n_upd // 4] = idx[n_upd // 4 : n_upd // 2]
         prio[: n_upd // 4] = prio[n_upd // 4 : n_upd // 2]
         cpu.set(idx, prio)
         gpu.set(idx.cuda(), prio.cuda())
