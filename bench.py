@@ -211,6 +211,7 @@ def build_rainbow(args, ctx, dtype):
                 "system.buffer_size=500000",
                 "system.warmup_steps=32",
                 "system.n_step=3",
+                f"system.compute_dtype={dtype}",
                 "logger.loggers=[]",
                 "logger.checkpointing.save_model=false",
             ],
@@ -222,7 +223,7 @@ def build_rainbow(args, ctx, dtype):
         "metric": "env steps/sec, Anakin Rainbow-DQN on Snake, prioritised replay",
         "steps_per_update": T * B,
         "graph": "update",
-        "dtype": "fp32",
+        "dtype": None,
         "config": {
             "model": "rainbow_noisy_dueling_c51",
             "global_batch": B * n,

@@ -121,18 +121,40 @@ class PrioritisedBuffer(TrajectoryBuffer):
     @torch.no_grad()
     def add(self, batch: Dict[str, Tensor]) -> None:
         t_block = next(iter(batch.values())).shape[1]
-        t_start = self._t_ptr.clone()  # device cursor, pre-advance
+        t_start = self._t_ptr.clone()  # device cursors, pre-advance
+        t_filled_pre = self._t_filled.clone()
         super().add(batch)
-        # new items get max priority so they are sampled at least once
-        offs = (torch.arange(t_block, device=self.device) + t_start) % self.t_max
+        S = self.seq_len - 1
+        # ONE tree update per add covering [t_start - S, t_start + t_block):
+        #   * the S columns BEHIND the old head are re-validated — their
+        #     windows now continue into the fresh data (skipping this, as
+        #     round 1 did, permanently starved every window straddling a
+        #     block boundary of sampling until the ring wrapped);
+        #     guarded by t_filled so unwritten slots stay at priority 0;
+        #   * the new columns get max priority so they are sampled at
+        #     least once;
+        #   * the S columns straight behind the NEW head get priority 0
+        #     (their windows would cross the write head).
+        # All shapes are static and all values device-computed, so the call
+        # is hip-graph capturable.
+        if S + t_block > self.t_max:
+            raise ValueError(
+                f"add block ({t_block}) + seq overlap ({S}) exceed the ring "
+                f"length ({self.t_max}); priorities would alias"
+            )
+        rel = torch.arange(-S, t_block, device=self.device)  # [S + t_block]
+        offs = (rel + t_start) % self.t_max
+        maxp = self._max_priority ** self.alpha
+        col_prio = torch.where(
+            rel >= t_block - S,  # the S columns behind the NEW head
+            torch.zeros((), device=self.device),
+            maxp,
+        )
+        # behind-columns only valid where data exists (rel >= -t_filled_pre)
+        col_prio = torch.where(rel < -t_filled_pre, torch.zeros((), device=self.device), col_prio)
         slots = (self._rowss.unsqueeze(1) * self.t_max + offs.unsqueeze(0)).reshape(-1)
-        fill = (self._max_priority ** self.alpha).expand(slots.numel()).contiguous()
-        self.tree.set(slots, fill)
-        # invalidate the seq_len-1 slots straight behind the new pointer
-        # (their windows would cross the write head)
-        inv = (torch.arange(self.seq_len - 1, device=self.device) + self._t_ptr - (self.seq_len - 1)) % self.t_max
-        inv_slots = (self._rowss.unsqueeze(1) * self.t_max + inv.unsqueeze(0)).reshape(-1)
-        self.tree.set(inv_slots, torch.zeros(inv_slots.numel(), device=self.device))
+        prio = col_prio.unsqueeze(0).expand(self.rows, -1).reshape(-1).contiguous()
+        self.tree.set(slots, prio)
 
     @torch.no_grad()
     def sample(self, batch_size: int, importance_sampling_exponent=0.4) -> Dict[str, Tensor]:

@@ -40,9 +40,28 @@ extern "C" __global__ void sumtree_update_single_kernel(
   __threadfence_block();
   __syncthreads();
   for (int l = 1; l <= depth; ++l) {
-    for (int i = threadIdx.x; i < n; i += blockDim.x) {
-      long node = ((long)cap + idx[i]) >> l;
-      tree[node] = tree[2 * node] + tree[2 * node + 1];
+    // two-phase per level: gather all child sums into registers FIRST,
+    // then store. A fused load-add-store loop serialises on the compiler's
+    // may-alias assumption (stores to `tree` vs the next iteration's
+    // loads of `tree`), costing one full memory latency per item; the
+    // split keeps all gathers of a level in flight together.
+    float vals[8];
+    long nodes[8];
+    for (int base = threadIdx.x; base < n; base += blockDim.x * 8) {
+      int cnt = 0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        int i = base + j * blockDim.x;
+        if (i < n) {
+          long node = ((long)cap + idx[i]) >> l;
+          nodes[cnt] = node;
+          vals[cnt] = tree[2 * node] + tree[2 * node + 1];
+          ++cnt;
+        }
+      }
+#pragma unroll
+      for (int j = 0; j < 8; ++j)
+        if (j < cnt) tree[nodes[j]] = vals[j];
     }
     __threadfence_block();
     __syncthreads();
@@ -94,7 +113,7 @@ extern "C" void launch_sumtree_update(float* tree, const long* idx,
                                       int depth, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   if (n <= 0) return;
-  if (n <= 4096) {
+  if (n <= 8192) {  // <= 8 items/thread/level: one in-flight gather chunk
     int threads = (int)((n < 1024) ? ((n + 63) / 64) * 64 : 1024);
     if (threads < 64) threads = 64;
     hipLaunchKernelGGL(sumtree_update_single_kernel, dim3(1), dim3(threads),

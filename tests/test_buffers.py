@@ -102,3 +102,27 @@ def test_trajectory_branch_free_wrap():
     # oldest surviving step is at (ptr - filled) % 16
     t0 = out["_t0"]
     assert ((t0 >= 0) & (t0 < 16)).all()
+
+
+def test_prioritised_add_revalidates_block_boundary_windows():
+    """Windows straddling an add-block boundary must become sampleable on
+    the NEXT add (round-1 regression: slots invalidated behind the write
+    head were never re-validated until the ring wrapped, starving half the
+    data at small block sizes)."""
+    buf = PrioritisedBuffer(
+        add_batch_size=2, max_length_time_axis=32, sample_sequence_length=3,
+        device="cpu", seed=0,
+    )
+    mk = lambda: {"x": torch.randn(2, 4, 1)}
+    buf.add(mk())
+    leaves = buf.tree.tree[buf.tree.capacity : buf.tree.capacity + buf.n_slots].view(2, 32)
+    # after add 1 (cols 0-3): starts 0,1 valid; 2,3 would cross the head
+    assert (leaves[:, 0:2] > 0).all() and (leaves[:, 2:4] == 0).all()
+    buf.add(mk())
+    leaves = buf.tree.tree[buf.tree.capacity : buf.tree.capacity + buf.n_slots].view(2, 32)
+    # after add 2 (cols 4-7): the boundary starts 2,3 are re-validated,
+    # 4,5 valid, 6,7 cross the new head
+    assert (leaves[:, 0:6] > 0).all(), leaves[0, :10]
+    assert (leaves[:, 6:8] == 0).all()
+    # unwritten region stays unsampleable
+    assert (leaves[:, 8:] == 0).all()
