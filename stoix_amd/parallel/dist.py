@@ -137,3 +137,11 @@ def all_reduce_mean_scalar(x: torch.Tensor) -> torch.Tensor:
         dist.all_reduce(x)
         x = x / dist.get_world_size()
     return x
+
+
+def polyak_update(online, target, tau: float) -> None:
+    """Fused polyak target update (2 _foreach kernels instead of 2 per
+    parameter tensor; K9 of SURVEY.md §2.9)."""
+    tgt = list(target)
+    torch._foreach_mul_(tgt, 1.0 - tau)
+    torch._foreach_add_(tgt, list(online), alpha=tau)

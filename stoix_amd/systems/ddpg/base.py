@@ -158,14 +158,14 @@ class DDPGFamilyLearner:
                 nn.utils.clip_grad_norm_(self.q_online.parameters(), float(self.sys.max_grad_norm))
                 nn.utils.clip_grad_norm_(self.actor.parameters(), float(self.sys.max_grad_norm))
             self.q_opt.step()
+            from stoix_amd.parallel.dist import polyak_update
+
             if do_actor:
                 self.actor_opt.step()
                 with torch.no_grad():
-                    for po, pt in zip(self.actor.parameters(), self.actor_target.parameters()):
-                        pt.mul_(1.0 - tau).add_(po, alpha=tau)
+                    polyak_update(self.actor.parameters(), self.actor_target.parameters(), tau)
             with torch.no_grad():
-                for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
-                    pt.mul_(1.0 - tau).add_(po, alpha=tau)
+                polyak_update(self.q_online.parameters(), self.q_target.parameters(), tau)
             self.update_count += 1
         return metrics
 

@@ -309,10 +309,10 @@ class MPOLearner:
             self.actor_opt.step()
             self.dual_opt.step()
             with torch.no_grad():
-                for po, pt in zip(self.q.parameters(), self.q_target.parameters()):
-                    pt.mul_(1 - tau).add_(po, alpha=tau)
-                for po, pt in zip(self.actor.parameters(), self.actor_target.parameters()):
-                    pt.mul_(1 - tau).add_(po, alpha=tau)
+                from stoix_amd.parallel.dist import polyak_update
+
+                polyak_update(self.q.parameters(), self.q_target.parameters(), tau)
+                polyak_update(self.actor.parameters(), self.actor_target.parameters(), tau)
             metrics = {"q_loss": q_loss, **pm}
         return metrics
 

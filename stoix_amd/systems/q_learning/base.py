@@ -153,8 +153,9 @@ class OffPolicyQLearner:
 
     @torch.no_grad()
     def _polyak(self, tau: float) -> None:
-        for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
-            pt.mul_(1.0 - tau).add_(po, alpha=tau)
+        from stoix_amd.parallel.dist import polyak_update
+
+        polyak_update(self.q_online.parameters(), self.q_target.parameters(), tau)
         for bo, bt in zip(self.q_online.buffers(), self.q_target.buffers()):
             if bt.dtype.is_floating_point:
                 bt.mul_(1.0 - tau).add_(bo, alpha=tau)

@@ -164,26 +164,30 @@ class RainbowLearner:
 
             self.noise_online.resample(self.gen)
             self.noise_target.resample(self.gen)
+            # ONE online forward over [obs0; obs_n] (the double-Q selector
+            # shares weights and noise with the TD forward; batching the
+            # rows halves the online-net kernel count)
+            obs_cat = torch.cat([obs0, obs_n], dim=0)
+            Bn = obs0.shape[0]
             if self._use_amp:
                 with torch.autocast("cuda", torch.bfloat16):
-                    out_tm1 = self.q_online(obs0)
+                    out_all = self.q_online(obs_cat)
                     with torch.no_grad():
                         out_t = self.q_target(obs_n)
-                        sel = self.q_online(obs_n).q_values
             else:
-                out_tm1 = self.q_online(obs0)
+                out_all = self.q_online(obs_cat)
                 with torch.no_grad():
                     out_t = self.q_target(obs_n)
-                    sel = self.q_online(obs_n).q_values
+            out_tm1 = out_all
             with torch.no_grad():
-                best_a = sel.argmax(dim=-1)
+                best_a = out_all.q_values[Bn:].argmax(dim=-1)
                 probs_t = F.softmax(out_t.q_logits.float(), dim=-1)
                 p_best = probs_t.gather(
                     1, best_a.view(-1, 1, 1).expand(-1, 1, probs_t.shape[-1])
                 ).squeeze(1)
                 target_z = n_step_reward.unsqueeze(-1) + bootstrap_disc.unsqueeze(-1) * out_t.atoms
                 target = categorical_l2_project(target_z, p_best, out_tm1.atoms)
-            logits_a = out_tm1.q_logits.float().gather(
+            logits_a = out_tm1.q_logits[:Bn].float().gather(
                 1, act0.view(-1, 1, 1).expand(-1, 1, out_tm1.q_logits.shape[-1])
             ).squeeze(1)
             ce = -(target * F.log_softmax(logits_a, dim=-1)).sum(-1)  # [B]
@@ -199,8 +203,9 @@ class RainbowLearner:
             # priority writeback: per-sample TD error magnitude
             self.buffer.set_priorities(batch["_slots"], ce.detach())
             with torch.no_grad():
-                for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
-                    pt.mul_(1 - tau).add_(po, alpha=tau)
+                tgt = list(self.q_target.parameters())
+                torch._foreach_mul_(tgt, 1 - tau)
+                torch._foreach_add_(tgt, list(self.q_online.parameters()), alpha=tau)
             self._update_count += 1
             metrics = {"q_loss": loss.detach(), "beta": beta.detach()}
         return metrics

@@ -241,8 +241,9 @@ class R2D2Learner:
                 abs_td = td.abs()
                 prio = self.eta * abs_td.max(dim=0).values + (1 - self.eta) * abs_td.mean(dim=0)
                 self.buffer.set_priorities(batch["_slots"], prio)
-                for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
-                    pt.mul_(1 - tau).add_(po, alpha=tau)
+                from stoix_amd.parallel.dist import polyak_update
+
+                polyak_update(self.q_online.parameters(), self.q_target.parameters(), tau)
             metrics = {"q_loss": loss.detach(), "mean_abs_td": abs_td.mean().detach()}
         return metrics
 

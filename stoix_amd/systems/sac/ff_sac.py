@@ -165,8 +165,9 @@ class SACLearner:
                 self.alpha_opt.step()
 
             with torch.no_grad():
-                for po, pt in zip(self.q_online.parameters(), self.q_target.parameters()):
-                    pt.mul_(1.0 - tau).add_(po, alpha=tau)
+                from stoix_amd.parallel.dist import polyak_update
+
+                polyak_update(self.q_online.parameters(), self.q_target.parameters(), tau)
             metrics = {
                 "q_loss": q_loss.detach(),
                 "actor_loss": actor_loss.detach(),
