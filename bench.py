@@ -60,6 +60,7 @@ def _anakin_learner(ctx, compose_args, learner_cls, num_envs, seed_mul=31):
 def build_ppo(args, ctx, dtype):
     from stoix_amd.systems.ppo.ff_ppo import PPOLearner
 
+    args.num_envs = args.num_envs or 4096
     n = ctx.world_size
     cfg, learner = _anakin_learner(
         ctx,
@@ -103,7 +104,7 @@ def build_ppo(args, ctx, dtype):
 def build_cartpole(args, ctx, dtype):
     from stoix_amd.systems.ppo.ff_ppo import PPOLearner
 
-    B = args.num_envs if args.num_envs != 4096 else 4
+    B = args.num_envs or 4
     n = ctx.world_size
     cfg, learner = _anakin_learner(
         ctx,
@@ -146,7 +147,7 @@ def build_cartpole(args, ctx, dtype):
 def build_sac(args, ctx, dtype):
     from stoix_amd.systems.sac.ff_sac import SACLearner
 
-    B = args.num_envs if args.num_envs != 4096 else 2048
+    B = args.num_envs or 2048
     T = 8
     n = ctx.world_size
     cfg, learner = _anakin_learner(
@@ -192,7 +193,7 @@ def build_sac(args, ctx, dtype):
 def build_rainbow(args, ctx, dtype):
     from stoix_amd.systems.q_learning.ff_rainbow import RainbowLearner
 
-    B = args.num_envs if args.num_envs != 4096 else 1024
+    B = args.num_envs or 1024
     T = 4
     n = ctx.world_size
     cfg, learner = _anakin_learner(
@@ -256,7 +257,7 @@ def bench_sebulba(args, ctx) -> None:
     from stoix_amd.config import compose
     from stoix_amd.systems.ppo.sebulba_ff_ppo import run_experiment
 
-    n_envs = args.num_envs if args.num_envs != 4096 else 256
+    n_envs = args.num_envs or 256
     rollout = min(args.rollout_length, 64)
     updates = args.steps + args.warmup
     actors = 4
@@ -317,7 +318,7 @@ def main() -> None:
     p.add_argument("--warmup", type=int, default=3)
     p.add_argument("--config", type=str, default="ppo",
                    choices=["ppo", "cartpole", "sac", "rainbow", "sebulba"])
-    p.add_argument("--num-envs", type=int, default=4096, help="envs per GPU")
+    p.add_argument("--num-envs", type=int, default=None, help="envs per GPU (default: per-config)")
     p.add_argument("--rollout-length", type=int, default=128)
     p.add_argument("--env", type=str, default="brax/ant")
     p.add_argument("--dtype", type=str, default="bf16", choices=["bf16", "fp32"])
