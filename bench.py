@@ -193,7 +193,7 @@ def build_sac(args, ctx, dtype):
 def build_rainbow(args, ctx, dtype):
     from stoix_amd.systems.q_learning.ff_rainbow import RainbowLearner
 
-    B = args.num_envs or 1024
+    B = args.num_envs or 4096  # MI355X sizing; 1024 matches the round-1 table
     T = 4
     n = ctx.world_size
     cfg, learner = _anakin_learner(
