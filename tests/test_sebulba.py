@@ -131,3 +131,84 @@ def test_sebulba_ppo_breakout_pixels():
     )
     r = run(cfg)
     assert r == r
+
+
+# ---------------------------------------------------- gymnasium adapter
+
+
+class _MockVecGym:
+    """Duck-typed gymnasium VectorEnv: 2 envs, env 0 TERMINATES at step 3,
+    env 1 TRUNCATES at step 5; autoreset returns the reset obs and stashes
+    the true final obs under info['final_observation']."""
+
+    class _Sp:
+        def __init__(self, shape=None, low=None, high=None, n=None):
+            self.shape, self.low, self.high, self.n = shape, low, high, n
+
+    def __init__(self):
+        self.num_envs = 2
+        self.single_observation_space = self._Sp(shape=(3,), low=-1.0, high=1.0)
+        self.single_action_space = self._Sp(n=2)
+        self._t = None
+
+    def reset(self, seed=None):
+        import numpy as np
+
+        self._t = np.zeros(2, dtype=np.int64)
+        return np.zeros((2, 3), dtype=np.float32), {}
+
+    def step(self, actions):
+        import numpy as np
+
+        self._t += 1
+        obs = np.tile(self._t[:, None].astype(np.float32), (1, 3))
+        reward = np.ones(2, dtype=np.float32)
+        terminated = np.array([self._t[0] == 3, False])
+        truncated = np.array([False, self._t[1] == 5])
+        done = terminated | truncated
+        finals = [None, None]
+        for i in range(2):
+            if done[i]:
+                finals[i] = obs[i].copy()
+                obs[i] = 0.0  # autoreset observation
+                self._t[i] = 0
+        info = {"final_observation": finals} if any(done) else {}
+        return obs, reward, terminated, truncated, info
+
+
+def test_vecgym_adapter_timestep_contract():
+    """VecGymToStoa reconstructs the §8.7 TimeStep semantics from the
+    gymnasium vector API (reference wrappers/gymnasium.py:12)."""
+    import torch
+
+    from stoix_amd.envs.gymnasium_adapter import VecGymToStoa
+    from stoix_amd.types import StepType
+
+    env = VecGymToStoa(_MockVecGym(), seed=0)
+    assert env.observation_space.shape == (3,)
+    assert env.action_space.num_values == 2
+    ts = env.reset()
+    assert ts.step_type.tolist() == [StepType.FIRST] * 2
+
+    for t in range(1, 3):
+        ts = env.step(torch.zeros(2, dtype=torch.long))
+        assert ts.step_type.tolist() == [StepType.MID] * 2
+
+    # step 3: env 0 terminates -> discount 0, autoreset obs, final in extras
+    ts = env.step(torch.zeros(2, dtype=torch.long))
+    assert ts.step_type[0] == StepType.TERMINATED
+    assert float(ts.discount[0]) == 0.0 and float(ts.discount[1]) == 1.0
+    assert torch.all(ts.observation[0] == 0.0)          # reset obs
+    assert torch.all(ts.extras["next_obs"][0] == 3.0)   # true final obs
+    em = ts.extras["episode_metrics"]
+    assert bool(em["is_terminal_step"][0]) and not bool(em["is_terminal_step"][1])
+    assert float(em["episode_return"][0]) == 3.0
+    assert float(em["episode_length"][0]) == 3.0
+
+    # env 1 truncates at its step 5 -> discount stays 1
+    ts = env.step(torch.zeros(2, dtype=torch.long))  # env1 t=4
+    ts = env.step(torch.zeros(2, dtype=torch.long))  # env1 t=5 truncates
+    assert ts.step_type[1] == StepType.TRUNCATED
+    assert float(ts.discount[1]) == 1.0
+    assert torch.all(ts.extras["next_obs"][1] == 5.0)
+    assert float(ts.extras["episode_metrics"]["episode_return"][1]) == 5.0
