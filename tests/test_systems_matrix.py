@@ -111,3 +111,25 @@ def test_system_smoke(module, default, extra):
     cfg = compose(default, TINY + extra)
     r = mod.run(cfg)
     assert r == r, f"{module} returned NaN"
+
+
+def test_disco103_smoke():
+    """DisCo-103 runs end-to-end with the random-init meta-network
+    fallback (reference ff_disco103.py; meta-params download is offline-
+    gated — docstring of systems/disco_rl/ff_disco103.py)."""
+    from stoix_amd.config import compose
+    from stoix_amd.systems.disco_rl import ff_disco103
+
+    cfg = compose(
+        "default/anakin/default_ff_disco103.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8",
+         "arch.total_timesteps=null", "arch.num_updates=2",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+         "system.rollout_length=8", "system.num_minibatches=2",
+         "system.epochs=1", "system.disco_rule.num_bins=21",
+         "system.disco_rule.max_abs_value=50.0",
+         "system.disco_rule.net.prediction_size=16",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = ff_disco103.run(cfg)
+    assert r == r  # finite float out of the experiment
