@@ -87,6 +87,23 @@ void launch_sumtree_update(float*, const long*, const float*, long, int, int,
                            void*);
 void launch_sumtree_sample(const float*, const float*, long*, long, int, int,
                            int, void*);
+void launch_policy_value_step_disc(const float*, const void*, const float*,
+                                   const void*, const float*, const void*,
+                                   const float*, const void*, const float*,
+                                   const void*, const float*, const void*,
+                                   const float*, float*, long*, float*,
+                                   float*, const float*, const float*, int,
+                                   int, int, int, int, uint64_t,
+                                   unsigned int*, unsigned int, int, void*);
+void launch_ppo_head_loss_disc(const void*, const void*, const long*,
+                               const float*, const float*, const float*,
+                               const float*, void*, void*, void*, float*,
+                               int, int, float, float, float, void*);
+void launch_ppo_gather_disc(const long*, int, const float*, int, int,
+                            const long*, const float*, const float*,
+                            const float*, const float*, void*, long*, float*,
+                            float*, float*, float*, const float*,
+                            const float*, void*);
 }
 
 namespace {
@@ -307,6 +324,84 @@ void policy_value_step(torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
       (unsigned int)draw_offset, (int)do_bump, cur_stream());
 }
 
+void policy_value_step_disc(
+    torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
+    torch::Tensor W2a, torch::Tensor b2a, torch::Tensor Wha,
+    torch::Tensor bha, torch::Tensor W1c, torch::Tensor b1c,
+    torch::Tensor W2c, torch::Tensor b2c, torch::Tensor Wvc,
+    torch::Tensor bvc, torch::Tensor obs_mirror, torch::Tensor action_out,
+    torch::Tensor logp_out, torch::Tensor value_out, torch::Tensor nmean,
+    torch::Tensor nvar, int64_t num_actions, int64_t greedy, int64_t seed,
+    torch::Tensor draw_buf, int64_t draw_offset, int64_t do_bump) {
+  CHK(obs, torch::kFloat32);
+  CHK(W1a, torch::kBFloat16);
+  CHK(action_out, torch::kInt64);
+  int B = obs.size(0), OBS = obs.size(1);
+  int HID = W2a.size(0);
+  int ACT = (int)num_actions;
+  TORCH_CHECK(HID == 256 || HID == 128, "fused MLP supports HID 128/256");
+  TORCH_CHECK(ACT <= 16, "fused categorical head supports ACT <= 16");
+  TORCH_CHECK(OBS <= 128, "fused MLP supports OBS <= 128");
+  float* om = obs_mirror.numel() > 0 ? obs_mirror.data_ptr<float>() : nullptr;
+  unsigned int* db = draw_buf.numel() > 0
+                         ? (unsigned int*)draw_buf.data_ptr<int>()
+                         : nullptr;
+  launch_policy_value_step_disc(
+      obs.data_ptr<float>(), W1a.data_ptr(), b1a.data_ptr<float>(),
+      W2a.data_ptr(), b2a.data_ptr<float>(), Wha.data_ptr(),
+      bha.data_ptr<float>(), W1c.data_ptr(), b1c.data_ptr<float>(),
+      W2c.data_ptr(), b2c.data_ptr<float>(), Wvc.data_ptr(),
+      bvc.data_ptr<float>(), om, action_out.data_ptr<long>(),
+      logp_out.data_ptr<float>(), value_out.data_ptr<float>(),
+      fptr_or_null(nmean), fptr_or_null(nvar), B, OBS, ACT, HID, (int)greedy,
+      (uint64_t)seed, db, (unsigned int)draw_offset, (int)do_bump,
+      cur_stream());
+}
+
+void ppo_head_loss_disc(torch::Tensor heads, torch::Tensor v_in,
+                        torch::Tensor action, torch::Tensor old_logp,
+                        torch::Tensor old_value, torch::Tensor adv,
+                        torch::Tensor targets, torch::Tensor dhead,
+                        torch::Tensor dv, torch::Tensor dv16,
+                        torch::Tensor metrics, int64_t num_actions,
+                        double clip_eps, double ent_coef, double vf_coef) {
+  CHK(heads, torch::kBFloat16);
+  CHK(action, torch::kInt64);
+  int B = heads.size(0);
+  launch_ppo_head_loss_disc(
+      heads.data_ptr(), v_in.data_ptr(), action.data_ptr<long>(),
+      old_logp.data_ptr<float>(), old_value.data_ptr<float>(),
+      adv.data_ptr<float>(), targets.data_ptr<float>(), dhead.data_ptr(),
+      dv.data_ptr(), dv16.numel() > 0 ? dv16.data_ptr() : nullptr,
+      metrics.numel() > 0 ? metrics.data_ptr<float>() : nullptr, B,
+      (int)num_actions, (float)clip_eps, (float)ent_coef, (float)vf_coef,
+      cur_stream());
+}
+
+void ppo_gather_disc(torch::Tensor idx, torch::Tensor obs,
+                     torch::Tensor action, torch::Tensor logp,
+                     torch::Tensor value, torch::Tensor adv,
+                     torch::Tensor targets, torch::Tensor obs_out,
+                     torch::Tensor action_out, torch::Tensor logp_out,
+                     torch::Tensor value_out, torch::Tensor adv_out,
+                     torch::Tensor targets_out, torch::Tensor nmean,
+                     torch::Tensor nvar) {
+  CHK(idx, torch::kInt64);
+  CHK(obs, torch::kFloat32);
+  CHK(obs_out, torch::kBFloat16);
+  CHK(action, torch::kInt64);
+  int mb = idx.numel();
+  launch_ppo_gather_disc(
+      idx.data_ptr<long>(), mb, obs.data_ptr<float>(), obs.size(1),
+      obs_out.size(1), action.data_ptr<long>(), logp.data_ptr<float>(),
+      value.data_ptr<float>(), adv.data_ptr<float>(),
+      targets.data_ptr<float>(), obs_out.data_ptr(),
+      action_out.data_ptr<long>(), logp_out.data_ptr<float>(),
+      value_out.data_ptr<float>(), adv_out.data_ptr<float>(),
+      targets_out.data_ptr<float>(), fptr_or_null(nmean), fptr_or_null(nvar),
+      cur_stream());
+}
+
 void value_forward(torch::Tensor obs, torch::Tensor W1c, torch::Tensor b1c,
                    torch::Tensor W2c, torch::Tensor b2c, torch::Tensor Wvc,
                    torch::Tensor bvc, torch::Tensor value_out,
@@ -503,6 +598,12 @@ void sumtree_sample(torch::Tensor tree, torch::Tensor u, torch::Tensor out,
 }
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("policy_value_step_disc", &policy_value_step_disc,
+        "fused actor+critic fwd, categorical Gumbel-max sample (MFMA)");
+  m.def("ppo_head_loss_disc", &ppo_head_loss_disc,
+        "categorical PPO losses + analytic logits backward");
+  m.def("ppo_gather_disc", &ppo_gather_disc,
+        "fused minibatch gather (int64 actions)");
   m.def("sumtree_update", &sumtree_update,
         "scatter leaf priorities + repair ancestors (device sum-tree)");
   m.def("sumtree_sample", &sumtree_sample,

@@ -1163,3 +1163,325 @@ extern "C" void launch_ppo_head_loss(
   if (draw_buf && do_bump)
     hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
 }
+
+// ===================================================================
+// Discrete (categorical) fused PPO path — BASELINE config #1 class
+// (discrete CartPole / MinAtar heads; reference heads.py:30-41
+// CategoricalHead). Same torso machinery as the tanh-normal kernels;
+// the epilogues differ: Gumbel-max sampling + log-softmax in the
+// rollout, exact categorical entropy + softmax-jacobian head backward
+// in the loss kernel. ACT <= 16 (one MFMA N-tile of logits).
+// ===================================================================
+
+template <int HID>
+__launch_bounds__(256, 2) __global__ void policy_value_step_disc_kernel(
+    const float* __restrict__ obs,        // [B, OBS]
+    const bf16_t* __restrict__ W1a, const float* __restrict__ b1a,
+    const bf16_t* __restrict__ W2a, const float* __restrict__ b2a,
+    const bf16_t* __restrict__ Wha, const float* __restrict__ bha,
+    const bf16_t* __restrict__ W1c, const float* __restrict__ b1c,
+    const bf16_t* __restrict__ W2c, const float* __restrict__ b2c,
+    const bf16_t* __restrict__ Wvc, const float* __restrict__ bvc,
+    float* __restrict__ obs_mirror,       // [B, OBS] or null
+    long* __restrict__ action_out,        // [B] int64
+    float* __restrict__ logp_out,         // [B]
+    float* __restrict__ value_out,        // [B]
+    const float* __restrict__ nmean, const float* __restrict__ nvar,
+    int B, int OBS, int ACT, int greedy,
+    uint64_t seed, const unsigned int* __restrict__ draw_buf,
+    unsigned int draw_offset) {
+  __shared__ MlpLds<HID> lds;
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int rbase = blockIdx.x * 16;
+  if (rbase >= B) return;
+  const uint32_t draw = (draw_buf ? *draw_buf : 0u) + draw_offset;
+  const int K1P = (OBS + 31) & ~31;
+
+  bf16_t* O = &lds.O[0][0];
+  constexpr int OS = K1P_MAX + OPAD;
+  constexpr int HS = HID + HPAD;
+  bf16_t* Ha0 = &lds.H[0][0][0];
+  bf16_t* Ha1 = &lds.H[1][0][0];
+  bf16_t* Hc0 = &lds.H[2][0][0];
+  bf16_t* Hc1 = &lds.H[3][0][0];
+
+  stage_obs(obs, rbase, OBS, K1P, O, OS, nmean, nvar, obs_mirror,
+            threadIdx.x, 256, B);
+  __syncthreads();
+
+  wg_layer<HID, true>(O, OS, W1a, b1a, K1P, Ha0, HS, lane, wid);
+  wg_layer<HID, true>(O, OS, W1c, b1c, K1P, Hc0, HS, lane, wid);
+  __syncthreads();
+  wg_layer<HID, true>(Ha0, HS, W2a, b2a, HID, Ha1, HS, lane, wid);
+  wg_layer<HID, true>(Hc0, HS, W2c, b2c, HID, Hc1, HS, lane, wid);
+  __syncthreads();
+
+  if (wid == 1) {
+    wave_value_head<HID>(Hc1, HS, Wvc, bvc ? *bvc : 0.0f, rbase, value_out,
+                         lane, B);
+  }
+
+  // ---- actor head: logits tile [16 rows x 16 cols] -> per-row
+  // log-softmax + Gumbel-max sample (wave 0)
+  if (wid == 0) {
+    f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+    const int arow = lane & 15, ak0 = (lane >> 4) * 8;
+    for (int ks = 0; ks < HID / 32; ++ks) {
+      const bf16x8 a =
+          *reinterpret_cast<const bf16x8*>(Ha1 + arow * HS + ks * 32 + ak0);
+      bf16x8 b = load_w_frag<HID>(Wha, HID, 0, ks, lane);
+      acc = MFMA_BF16_16x16x32(a, b, acc, 0, 0, 0);
+    }
+    const int col = lane & 15;
+    const int g = lane >> 4;
+    const bool valid = col < ACT;
+    float bh = bha[col];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int grow = rbase + g * 4 + r;
+      float logit = valid ? acc[r] + bh : -3.0e38f;
+      // log-sum-exp over the 16-lane column group
+      float m = logit;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) m = fmaxf(m, __shfl_xor(m, off));
+      float e = valid ? __expf(logit - m) : 0.0f;
+      float sum = e;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) sum += __shfl_xor(sum, off);
+      float lse = m + __logf(sum);
+      // Gumbel-max sample (one philox block serves 4 rows per col)
+      float score = logit;
+      if (!greedy && valid) {
+        int rowblk = (rbase + g * 4) >> 2;
+        Rng4 u = philox_uniform4(seed, 2u, (uint32_t)(rowblk * 16 + col), draw);
+        float uu = (r == 0) ? u.a : (r == 1) ? u.b : (r == 2) ? u.c : u.d;
+        uu = fmaxf(uu, 1e-12f);
+        score = logit - __logf(-__logf(uu));
+      }
+      if (!valid) score = -3.0e38f;
+      // argmax over the group with index
+      float s = score;
+      int arg = col;
+#pragma unroll
+      for (int off = 1; off < 16; off <<= 1) {
+        float so = __shfl_xor(s, off);
+        int ao = __shfl_xor(arg, off);
+        if (so > s || (so == s && ao < arg)) { s = so; arg = ao; }
+      }
+      float logit_a = __shfl(logit, (lane & 48) | arg, 64);
+      if (grow < B && col == 0) {
+        action_out[grow] = (long)arg;
+        logp_out[grow] = logit_a - lse;
+      }
+    }
+  }
+}
+
+// Per-row categorical PPO head loss + analytic backward (see the
+// tanh-normal ppo_head_loss_kernel above for the shared clip/value math;
+// gradients here go through the softmax jacobian:
+// d logp_a / d logits_j = delta_aj - p_j;
+// dH / d logits_j = -p_j (logp_j + H)).
+extern "C" __global__ void ppo_head_loss_disc_kernel(
+    const bf16_t* __restrict__ heads,    // [B, 16] logits (cols >= ACT pad)
+    const bf16_t* __restrict__ v_in,     // [B]
+    const long* __restrict__ action,     // [B] int64
+    const float* __restrict__ old_logp,  // [B]
+    const float* __restrict__ old_value, // [B]
+    const float* __restrict__ adv,       // [B]
+    const float* __restrict__ targets,   // [B]
+    bf16_t* __restrict__ dhead,          // [B, 16]
+    bf16_t* __restrict__ dv_out,         // [B]
+    bf16_t* __restrict__ dv16_out,       // [B,16] col 0 or null
+    float* __restrict__ metrics,         // [3] or null
+    int B, int ACT, float clip_eps, float ent_coef, float vf_coef,
+    float inv_B) {
+  const int row = blockIdx.x * blockDim.x + threadIdx.x;
+  const bool active = row < B;
+
+  float l[16], p[16];
+  float lse = 0.0f, Hent = 0.0f, logp_new = 0.0f;
+  int a = 0;
+  if (active) {
+    const bf16x8* h8 = reinterpret_cast<const bf16x8*>(heads + (long)row * 16);
+    bf16x8 h0 = h8[0], h1 = h8[1];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) { l[j] = bf2f(h0[j]); l[8 + j] = bf2f(h1[j]); }
+    a = (int)action[row];
+    float m = -3.0e38f;
+    for (int j = 0; j < ACT; ++j) m = fmaxf(m, l[j]);
+    float sum = 0.0f;
+    for (int j = 0; j < ACT; ++j) sum += __expf(l[j] - m);
+    lse = m + __logf(sum);
+    for (int j = 0; j < 16; ++j) p[j] = (j < ACT) ? __expf(l[j] - lse) : 0.0f;
+    for (int j = 0; j < ACT; ++j) Hent -= p[j] * (l[j] - lse);
+    logp_new = l[a] - lse;
+  }
+
+  // ---- PPO clip loss (identical to the continuous kernel)
+  float A = active ? adv[row] : 0.0f;
+  float ratio = __expf(logp_new - (active ? old_logp[row] : 0.0f));
+  float r_clip = fmaxf(1.0f - clip_eps, fminf(1.0f + clip_eps, ratio));
+  float l1 = ratio * A, l2 = r_clip * A;
+  float a_loss = -fminf(l1, l2);
+  float dl_dlogp;
+  if (l1 <= l2) {
+    dl_dlogp = -ratio * A;
+  } else {
+    dl_dlogp = (ratio > 1.0f - clip_eps && ratio < 1.0f + clip_eps)
+                   ? -ratio * A
+                   : 0.0f;
+  }
+  dl_dlogp *= inv_B;
+
+  // ---- clipped value loss (identical)
+  float v_pred = active ? bf2f(v_in[row]) : 0.0f;
+  float ov = active ? old_value[row] : 0.0f;
+  float tg = active ? targets[row] : 0.0f;
+  float v_clip = ov + fmaxf(-clip_eps, fminf(clip_eps, v_pred - ov));
+  float e1 = (v_pred - tg), e2 = (v_clip - tg);
+  float sq1 = e1 * e1, sq2 = e2 * e2;
+  float v_loss = 0.5f * fmaxf(sq1, sq2);
+  float dv;
+  if (sq1 >= sq2) {
+    dv = e1;
+  } else {
+    dv = (fabsf(v_pred - ov) < clip_eps) ? e2 : 0.0f;
+  }
+  dv *= vf_coef * inv_B;
+
+  // ---- analytic logits gradient
+  if (active) {
+    const float ce = ent_coef * inv_B;
+    bf16x8 d0, d1;
+#pragma unroll
+    for (int j = 0; j < 16; ++j) {
+      float g = 0.0f;
+      if (j < ACT) {
+        float delta = (j == a) ? 1.0f : 0.0f;
+        g = dl_dlogp * (delta - p[j]);
+        g += ce * p[j] * ((l[j] - lse) + Hent);
+      }
+      if (j < 8) d0[j] = f2bf(g); else d1[j - 8] = f2bf(g);
+    }
+    bf16x8* out8 = reinterpret_cast<bf16x8*>(dhead + (long)row * 16);
+    out8[0] = d0;
+    out8[1] = d1;
+    dv_out[row] = f2bf(dv);
+    if (dv16_out) dv16_out[(long)row * 16] = f2bf(dv);
+  }
+
+  if (metrics) {
+    float m0 = active ? a_loss * inv_B : 0.0f;
+    float m1 = active ? v_loss * inv_B : 0.0f;
+    float m2 = active ? Hent * inv_B : 0.0f;
+#pragma unroll
+    for (int off = 32; off > 0; off >>= 1) {
+      m0 += __shfl_down(m0, off);
+      m1 += __shfl_down(m1, off);
+      m2 += __shfl_down(m2, off);
+    }
+    if ((threadIdx.x & 63) == 0) {
+      atomicAdd(&metrics[0], m0);
+      atomicAdd(&metrics[1], m1);
+      atomicAdd(&metrics[2], m2);
+    }
+  }
+}
+
+// Minibatch gather for the discrete path: int64 actions, no per-dim loop.
+extern "C" __global__ void ppo_gather_disc_kernel(
+    const long* __restrict__ idx, int mb_size,
+    const float* __restrict__ obs, int OBS, int OBS_PAD,
+    const long* __restrict__ action,
+    const float* __restrict__ logp, const float* __restrict__ value,
+    const float* __restrict__ adv, const float* __restrict__ targets,
+    bf16_t* __restrict__ obs_out, long* __restrict__ action_out,
+    float* __restrict__ logp_out, float* __restrict__ value_out,
+    float* __restrict__ adv_out, float* __restrict__ targets_out,
+    const float* __restrict__ nmean, const float* __restrict__ nvar) {
+  int row = (blockIdx.x * blockDim.x + threadIdx.x) >> 6;
+  int lane = threadIdx.x & 63;
+  if (row >= mb_size) return;
+  long src = idx[row];
+  for (int k = lane; k < OBS_PAD; k += 64) {
+    float v = 0.0f;
+    if (k < OBS) {
+      v = obs[src * OBS + k];
+      if (nmean) {
+        float sd = sqrtf(fmaxf(nvar[k], 1e-6f));
+        v = fmaxf(-10.0f, fminf(10.0f, (v - nmean[k]) / sd));
+      }
+    }
+    obs_out[(long)row * OBS_PAD + k] = f2bf(v);
+  }
+  if (lane == 0) {
+    action_out[row] = action[src];
+    logp_out[row] = logp[src];
+    value_out[row] = value[src];
+    adv_out[row] = adv[src];
+    targets_out[row] = targets[src];
+  }
+}
+
+extern "C" void launch_policy_value_step_disc(
+    const float* obs, const void* W1a, const float* b1a, const void* W2a,
+    const float* b2a, const void* Wha, const float* bha, const void* W1c,
+    const float* b1c, const void* W2c, const float* b2c, const void* Wvc,
+    const float* bvc, float* obs_mirror, long* action_out, float* logp_out,
+    float* value_out, const float* nmean, const float* nvar, int B, int OBS,
+    int ACT, int HID, int greedy, uint64_t seed, unsigned int* draw_buf,
+    unsigned int draw_offset, int do_bump, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  dim3 grid((B + 15) / 16), block(256);
+  if (HID == 256) {
+    hipLaunchKernelGGL(policy_value_step_disc_kernel<256>, grid, block, 0, s,
+                       obs, (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, greedy, seed, draw_buf, draw_offset);
+  } else {
+    hipLaunchKernelGGL(policy_value_step_disc_kernel<128>, grid, block, 0, s,
+                       obs, (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, greedy, seed, draw_buf, draw_offset);
+  }
+  if (draw_buf && do_bump)
+    hipLaunchKernelGGL(bump_u32_kernel2, dim3(1), dim3(1), 0, s, draw_buf);
+}
+
+extern "C" void launch_ppo_head_loss_disc(
+    const void* heads, const void* v_in, const long* action,
+    const float* old_logp, const float* old_value, const float* adv,
+    const float* targets, void* dhead, void* dv_out, void* dv16_out,
+    float* metrics, int B, int ACT, float clip_eps, float ent_coef,
+    float vf_coef, void* stream) {
+  hipStream_t s = (hipStream_t)stream;
+  int threads = 256;
+  int blocks = (B + threads - 1) / threads;
+  hipLaunchKernelGGL(ppo_head_loss_disc_kernel, dim3(blocks), dim3(threads),
+                     0, s, (const bf16_t*)heads, (const bf16_t*)v_in, action,
+                     old_logp, old_value, adv, targets, (bf16_t*)dhead,
+                     (bf16_t*)dv_out, (bf16_t*)dv16_out, metrics, B, ACT,
+                     clip_eps, ent_coef, vf_coef, 1.0f / (float)B);
+}
+
+extern "C" void launch_ppo_gather_disc(
+    const long* idx, int mb_size, const float* obs, int OBS, int OBS_PAD,
+    const long* action, const float* logp, const float* value,
+    const float* adv, const float* targets, void* obs_out, long* action_out,
+    float* logp_out, float* value_out, float* adv_out, float* targets_out,
+    const float* nmean, const float* nvar, void* stream) {
+  int threads = 256;
+  int rows_per_block = threads / 64;
+  int blocks = (mb_size + rows_per_block - 1) / rows_per_block;
+  hipLaunchKernelGGL(ppo_gather_disc_kernel, dim3(blocks), dim3(threads), 0,
+                     (hipStream_t)stream, idx, mb_size, obs, OBS, OBS_PAD,
+                     action, logp, value, adv, targets, (bf16_t*)obs_out,
+                     action_out, logp_out, value_out, adv_out, targets_out,
+                     nmean, nvar);
+}

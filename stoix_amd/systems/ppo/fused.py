@@ -35,7 +35,10 @@ backward produces zero standalone cast/copy kernels.
 
 Eligibility (falls back to the eager path otherwise): CUDA device, MLP torso
 [H,H] with SiLU and no LayerNorm, H in {128,256}, NormalAffineTanh head with
-ACT<=8, ScalarCriticHead, flat obs <= 128 dims, no observation normalisation.
+ACT<=8 OR Categorical head with ACT<=16 (round 2: Gumbel-max sampling +
+log-softmax in policy_value_step_disc, exact-entropy categorical loss +
+softmax-jacobian backward in ppo_head_loss_disc), ScalarCriticHead, flat
+obs <= 128 dims, no observation normalisation.
 
 Algorithm semantics match ff_ppo.py exactly (same losses, same GAE buffers,
 same per-minibatch all-reduce + per-chain clip); numerics are bf16-GEMM
@@ -135,8 +138,6 @@ class FusedPPOEngine:
             return None
         if getattr(learner, "normalize_obs", False):
             return None
-        if learner._discrete:
-            return None
         if type(learner).policy_loss is not _base_policy_loss_func(learner):
             # PPO-penalty / DPO override policy_loss; the fused head kernel
             # implements the clip loss only.
@@ -149,8 +150,21 @@ class FusedPPOEngine:
         if a_lins is None or c_lins is None:
             return None
         head = actor.action_head
-        if not isinstance(head, NormalAffineTanhDistributionHead):
-            return None
+        discrete = learner._discrete
+        if discrete:
+            from stoix_amd.networks.heads import CategoricalHead
+
+            if not isinstance(head, CategoricalHead):
+                return None
+            ACT = head.linear.out_features
+            if ACT > 16:
+                return None
+        else:
+            if not isinstance(head, NormalAffineTanhDistributionHead):
+                return None
+            ACT = head.loc.out_features
+            if ACT > 8:
+                return None
         if not isinstance(critic.critic_head, ScalarCriticHead):
             return None
         H = a_lins[0].out_features
@@ -159,8 +173,7 @@ class FusedPPOEngine:
         if c_lins[0].out_features != H or c_lins[1].out_features != H:
             return None
         OBS = a_lins[0].in_features
-        ACT = head.loc.out_features
-        if OBS > 128 or ACT > 8:
+        if OBS > 128:
             return None
         if getattr(learner.env, "_hip", None) is None:
             return None
@@ -206,16 +219,23 @@ class FusedPPOEngine:
         self.learner = learner
         self.device = learner.device
         self.H, self.OBS, self.ACT = H, OBS, ACT
+        self.discrete = bool(learner._discrete)
         sysc = learner.sys
         self.clip_eps = float(sysc.clip_eps)
         self.ent_coef = float(sysc.ent_coef)
         self.vf_coef = float(sysc.vf_coef)
         self.max_grad_norm = float(sysc.max_grad_norm)
-        head: NormalAffineTanhDistributionHead = learner.actor.action_head
-        self.min_scale = float(head.min_scale)
-        self.aff_scale = (float(head.maximum) - float(head.minimum)) / 2.0
-        self.aff_shift = (float(head.maximum) + float(head.minimum)) / 2.0
-        self.log_aff_scale = math.log(self.aff_scale)
+        head = learner.actor.action_head
+        if self.discrete:
+            self.min_scale = 0.0
+            self.aff_scale = 1.0
+            self.aff_shift = 0.0
+            self.log_aff_scale = 0.0
+        else:
+            self.min_scale = float(head.min_scale)
+            self.aff_scale = (float(head.maximum) - float(head.minimum)) / 2.0
+            self.aff_shift = (float(head.maximum) + float(head.minimum)) / 2.0
+            self.log_aff_scale = math.log(self.aff_scale)
         # rank-dependent Philox seed (torch.manual_seed is rank-offset)
         self.seed = int(torch.initial_seed()) % (2**62) + 101
         # first-layer K padded to the MFMA K-step (32): the rollout kernel
@@ -288,7 +308,10 @@ class FusedPPOEngine:
         class _Ws:
             def __init__(ws):
                 ws.Xmb = z(S, self.K1P)  # K-padded GEMM input (zeros past OBS)
-                ws.act = z(S, ACT, dtype=torch.float32)
+                if self.discrete:
+                    ws.act = torch.zeros(S, dtype=torch.long, device=dev)
+                else:
+                    ws.act = z(S, ACT, dtype=torch.float32)
                 ws.logp = z(S, dtype=torch.float32)
                 ws.val = z(S, dtype=torch.float32)
                 ws.adv = z(S, dtype=torch.float32)
@@ -356,10 +379,15 @@ class FusedPPOEngine:
             ac.views["b2"].copy_(a_lins[1].bias)
             ac.views["Wh"].zero_()
             ac.views["bh"].zero_()
-            ac.views["Wh"][0:ACT].copy_(head.loc.weight)
-            ac.views["Wh"][8 : 8 + ACT].copy_(head.scale.weight)
-            ac.views["bh"][0:ACT].copy_(head.loc.bias)
-            ac.views["bh"][8 : 8 + ACT].copy_(head.scale.bias)
+            if self.discrete:
+                # categorical head: logits rows 0..A-1 of the 16-row tile
+                ac.views["Wh"][0:ACT].copy_(head.linear.weight)
+                ac.views["bh"][0:ACT].copy_(head.linear.bias)
+            else:
+                ac.views["Wh"][0:ACT].copy_(head.loc.weight)
+                ac.views["Wh"][8 : 8 + ACT].copy_(head.scale.weight)
+                ac.views["bh"][0:ACT].copy_(head.loc.bias)
+                ac.views["bh"][8 : 8 + ACT].copy_(head.scale.bias)
             cc.views["W1"][:, :OBS].copy_(c_lins[0].weight)
             cc.views["b1"].copy_(c_lins[0].bias)
             cc.views["W2"].copy_(c_lins[1].weight)
@@ -370,10 +398,14 @@ class FusedPPOEngine:
         a_lins[0].bias.data = ac.views["b1"]
         a_lins[1].weight.data = ac.views["W2"]
         a_lins[1].bias.data = ac.views["b2"]
-        head.loc.weight.data = ac.views["Wh"][0:ACT]
-        head.scale.weight.data = ac.views["Wh"][8 : 8 + ACT]
-        head.loc.bias.data = ac.views["bh"][0:ACT]
-        head.scale.bias.data = ac.views["bh"][8 : 8 + ACT]
+        if self.discrete:
+            head.linear.weight.data = ac.views["Wh"][0:ACT]
+            head.linear.bias.data = ac.views["bh"][0:ACT]
+        else:
+            head.loc.weight.data = ac.views["Wh"][0:ACT]
+            head.scale.weight.data = ac.views["Wh"][8 : 8 + ACT]
+            head.loc.bias.data = ac.views["bh"][0:ACT]
+            head.scale.bias.data = ac.views["bh"][8 : 8 + ACT]
         c_lins[0].weight.data = cc.views["W1"][:, :OBS]
         c_lins[0].bias.data = cc.views["b1"]
         c_lins[1].weight.data = cc.views["W2"]
@@ -436,17 +468,29 @@ class FusedPPOEngine:
             env._done_count += done.sum()
             return
         for t in range(L.T):
-            ext.policy_value_step(
-                hb["obs"],
-                a16["W1"], ac.views["b1"], a16["W2"], ac.views["b2"],
-                a16["Wh"], ac.views["bh"],
-                c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
-                c16["Wv"], cc.views["bv"],
-                L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t], L.buf_value[t],
-                self.empty, self.empty,
-                self.min_scale, self.aff_scale, self.aff_shift,
-                self.log_aff_scale, 0, self.seed, self.draw_policy, t, 0,
-            )
+            if self.discrete:
+                ext.policy_value_step_disc(
+                    hb["obs"],
+                    a16["W1"], ac.views["b1"], a16["W2"], ac.views["b2"],
+                    a16["Wh"], ac.views["bh"],
+                    c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
+                    c16["Wv"], cc.views["bv"],
+                    L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t],
+                    L.buf_value[t], self.empty, self.empty,
+                    self.ACT, 0, self.seed, self.draw_policy, t, 0,
+                )
+            else:
+                ext.policy_value_step(
+                    hb["obs"],
+                    a16["W1"], ac.views["b1"], a16["W2"], ac.views["b2"],
+                    a16["Wh"], ac.views["bh"],
+                    c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
+                    c16["Wv"], cc.views["bv"],
+                    L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t], L.buf_value[t],
+                    self.empty, self.empty,
+                    self.min_scale, self.aff_scale, self.aff_shift,
+                    self.log_aff_scale, 0, self.seed, self.draw_policy, t, 0,
+                )
             env.hip_step_into(
                 L.buf_action[t],
                 reward_out=L.buf_reward[t],
@@ -478,7 +522,10 @@ class FusedPPOEngine:
         S = self.S
         OBSd = self.OBS
         flat_obs = L.buf_obs.view(TB, OBSd)
-        flat_action = L.buf_action.view(TB, self.ACT)
+        flat_action = (
+            L.buf_action.view(TB) if self.discrete
+            else L.buf_action.view(TB, self.ACT)
+        )
         flat_logp = L.buf_log_prob.view(TB)
         flat_value = L.buf_value.view(TB)
         flat_adv = L.buf_adv.view(TB)
@@ -489,9 +536,11 @@ class FusedPPOEngine:
 
         import torch.distributed as dist
 
+        gather_fn = ext.ppo_gather_disc if self.discrete else ext.ppo_gather
+
         def gather_into(mb: int, w) -> None:
             idx = L.perm_buf[mb * S : (mb + 1) * S]
-            ext.ppo_gather(
+            gather_fn(
                 idx, flat_obs, flat_action, flat_logp, flat_value, flat_adv,
                 flat_tgt, w.Xmb, w.act, w.logp, w.val, w.adv, w.tgt,
                 self.empty, self.empty,
@@ -537,14 +586,22 @@ class FusedPPOEngine:
             last = mb == n_mb - 1
             if last:
                 self.metrics.zero_()
-            ext.ppo_head_loss(
-                self.heads, self.vpred.view(-1), w.act, w.logp,
-                w.val, w.adv, w.tgt, self.dhead, self.dv,
-                self.dv16, self.metrics if last else self.metrics_none,
-                self.clip_eps, self.ent_coef,
-                self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
-                self.log_aff_scale, self.seed, self.draw_ent, mb, 0,
-            )
+            if self.discrete:
+                ext.ppo_head_loss_disc(
+                    self.heads, self.vpred.view(-1), w.act, w.logp,
+                    w.val, w.adv, w.tgt, self.dhead, self.dv,
+                    self.dv16, self.metrics if last else self.metrics_none,
+                    self.ACT, self.clip_eps, self.ent_coef, self.vf_coef,
+                )
+            else:
+                ext.ppo_head_loss(
+                    self.heads, self.vpred.view(-1), w.act, w.logp,
+                    w.val, w.adv, w.tgt, self.dhead, self.dv,
+                    self.dv16, self.metrics if last else self.metrics_none,
+                    self.clip_eps, self.ent_coef,
+                    self.vf_coef, self.min_scale, self.aff_scale, self.aff_shift,
+                    self.log_aff_scale, self.seed, self.draw_ent, mb, 0,
+                )
             torch.mm(self.dhead, a16["Wh"], out=self.dH2[0])
             torch.mm(self.dv, c16["Wv"].view(1, self.H), out=self.dH2[1])
             # ---- backward: wgrads + bias colsums via the split-K MFMA
