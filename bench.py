@@ -119,6 +119,7 @@ def build_cartpole(args, ctx, dtype):
                 f"system.rollout_length={args.rollout_length}",
                 "system.epochs=4",
                 "system.num_minibatches=2",
+                f"system.compute_dtype={dtype}",
                 "logger.loggers=[]",
                 "logger.checkpointing.save_model=false",
             ],
@@ -130,7 +131,7 @@ def build_cartpole(args, ctx, dtype):
         "metric": "env steps/sec, Anakin PPO on CartPole (4 envs, plumbing config)",
         "steps_per_update": args.rollout_length * B,
         "graph": "phases",
-        "dtype": "fp32",
+        "dtype": None,
         "config": {
             "model": "ppo_mlp_256x2_categorical",
             "global_batch": B * n,

@@ -508,13 +508,15 @@ def test_policy_value_step_disc_matches_eager(ext, H, OBS, ACT):
     """Fused categorical rollout kernel vs plain fp32 eager: value head,
     greedy action = argmax(logits), logp = log_softmax(logits)[a]."""
     dev = "cuda"
-    W1a, b1a, W2a, b2a, Wha, bha, W1c, b1c, W2c, b2c, Wvc, bvc = _mk_weights(
-        H, OBS, ACT, dev
+    W1a, b1a, W2a, b2a, _, _, W1c, b1c, W2c, b2c, Wvc, bvc = _mk_weights(
+        H, OBS, min(ACT, 8), dev
     )
-    # categorical packing: logits rows 0..ACT-1 only
-    Wha[ACT:] = 0
-    bha[ACT:] = 0
+    # categorical packing: logits rows 0..ACT-1 of the 16-row tile
     g = torch.Generator().manual_seed(11)
+    Wha = torch.zeros(16, H, device=dev)
+    Wha[0:ACT] = torch.randn(ACT, H, generator=g).to(dev) / math.sqrt(H)
+    bha = torch.zeros(16, device=dev)
+    bha[0:ACT] = torch.randn(ACT, generator=g).to(dev) * 0.1
     B = 977
     obs = torch.randn(B, OBS, generator=g).to(dev)
     action = torch.zeros(B, dtype=torch.long, device=dev)
