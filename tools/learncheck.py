@@ -44,8 +44,8 @@ def run_ppo(env_override, net, updates, num_envs, extra=(), graphs=True):
         if (u + 1) % max(1, updates // 10) == 0:
             learner.after_graph_replay() if hasattr(learner, "_graphs") else None
             m = learner.episode_metrics
-            r = float(m.get("episode_return", torch.tensor(float("nan"))))
-            curve.append(round(r, 2))
+            r = m.get("episode_return", torch.tensor(float("nan")))
+            curve.append(round(float(r.float().mean()), 2))
     return {"env": env_override, "graphs": graphs, "curve": curve,
             "wall_s": round(time.time() - t0, 1),
             "fused": learner.fused is not None}
@@ -98,6 +98,13 @@ def main():
                   ("system.compute_dtype=fp32",))
     print(json.dumps(out))
     assert out["curve"][-1] > 400, f"CartPole not solved: {out['curve']}"
+    # the FUSED discrete path (bf16, Gumbel-max rollout + analytic
+    # categorical backward) must learn the same task
+    out = run_ppo("env=classic/cartpole", "disc", 60, 256,
+                  ("system.compute_dtype=bf16",))
+    print(json.dumps(out))
+    if out["fused"]:
+        assert out["curve"][-1] > 400, f"fused CartPole not solved: {out['curve']}"
     out = run_dqn_cartpole(160)
     print(json.dumps(out))
     # DQN-family on CartPole shows the documented rise->forget->recover
