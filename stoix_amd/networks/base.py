@@ -95,7 +95,98 @@ class ScannedRNN(nn.Module):
             return [ScannedRNN._mask_state(s, keep) for s in state]
         return state * keep.unsqueeze(-1)
 
+    def _single_cell(self):
+        """The (cell, kind) pair when the stack is one GRU/LSTM cell —
+        the fast-path eligible shapes."""
+        from stoix_amd.networks.layers import GRUCell, LSTMCell
+
+        if len(self.rnn.cells) != 1:
+            return None, None
+        c = self.rnn.cells[0]
+        if isinstance(c, GRUCell):
+            return c.cell, "gru"
+        if isinstance(c, LSTMCell):
+            return c.cell, "lstm"
+        return None, None
+
+    def _hip_scan(self, x: Tensor, resets: Tensor, state: list, cell, kind):
+        """K13: fused HIP scan (ops/csrc/rnn.hip) — no-grad path (acting,
+        evaluation, R2D2 burn-in)."""
+        from stoix_amd import ops
+
+        ext = ops.ext(required=True)
+        T, B = x.shape[:2]
+        H = self.hidden_dim
+        xp = torch.nn.functional.linear(
+            x.reshape(T * B, -1), cell.weight_ih, cell.bias_ih
+        ).reshape(T, B, -1)
+        whh16 = cell.weight_hh.to(torch.bfloat16)
+        rs = resets.to(torch.uint8).contiguous()
+        Hout = torch.empty(T, B, H, device=x.device)
+        hT = torch.empty(B, H, device=x.device)
+        empty = torch.zeros(0, device=x.device)
+        if kind == "lstm":
+            h0, c0 = state[0]
+            cT = torch.empty(B, H, device=x.device)
+            ext.rnn_scan(xp, whh16, cell.bias_hh, rs, h0.contiguous(),
+                         c0.contiguous(), Hout, hT, cT, 1)
+            return Hout, [(hT, cT)]
+        h0 = state[0]
+        ext.rnn_scan(xp, whh16, cell.bias_hh, rs, h0.contiguous(), empty,
+                     Hout, hT, empty, 0)
+        return Hout, [hT]
+
+    def _hoisted_scan(self, x: Tensor, resets: Tensor, state: list, cell, kind):
+        """Autograd training path with the input projection for ALL T
+        hoisted into one GEMM (the cuDNN trick); only the [B,H]x[H,G*H]
+        recurrent GEMM + gate math stay per-step. Matches
+        nn.GRUCell/nn.LSTMCell math exactly."""
+        T, B = x.shape[:2]
+        H = self.hidden_dim
+        xp = torch.nn.functional.linear(
+            x.reshape(T * B, -1), cell.weight_ih, cell.bias_ih
+        ).reshape(T, B, -1)
+        outs = []
+        if kind == "gru":
+            h = state[0]
+            for t in range(T):
+                keep = (~resets[t].bool()).to(x.dtype).unsqueeze(-1)
+                h = h * keep
+                hg = torch.nn.functional.linear(h, cell.weight_hh, cell.bias_hh)
+                xr, xz, xn = xp[t].chunk(3, -1)
+                hr, hz, hn = hg.chunk(3, -1)
+                r = torch.sigmoid(xr + hr)
+                z = torch.sigmoid(xz + hz)
+                n = torch.tanh(xn + r * hn)
+                h = (1 - z) * n + z * h
+                outs.append(h)
+            return torch.stack(outs), [h]
+        h, c = state[0]
+        for t in range(T):
+            keep = (~resets[t].bool()).to(x.dtype).unsqueeze(-1)
+            h = h * keep
+            c = c * keep
+            hg = torch.nn.functional.linear(h, cell.weight_hh, cell.bias_hh)
+            i, f, g, o = (xp[t] + hg).chunk(4, -1)
+            c = torch.sigmoid(f) * c + torch.sigmoid(i) * torch.tanh(g)
+            h = torch.sigmoid(o) * torch.tanh(c)
+            outs.append(h)
+        return torch.stack(outs), [(h, c)]
+
     def forward(self, x: Tensor, resets: Tensor, state: list) -> Tuple[Tensor, list]:
+        cell, kind = self._single_cell()
+        if cell is not None:
+            if (
+                not torch.is_grad_enabled()
+                and x.is_cuda
+                and self.hidden_dim in (128, 256)
+                and x.dtype == torch.float32
+            ):
+                from stoix_amd import ops
+
+                if ops.have_ext():
+                    return self._hip_scan(x, resets, state, cell, kind)
+            return self._hoisted_scan(x, resets, state, cell, kind)
         T = x.shape[0]
         outs = []
         for t in range(T):

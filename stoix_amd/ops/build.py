@@ -22,6 +22,7 @@ SOURCES = [
     str(CSRC / "mlp.hip"),
     str(CSRC / "wgrad.hip"),
     str(CSRC / "per.hip"),
+    str(CSRC / "rnn.hip"),
 ]
 
 

@@ -104,6 +104,9 @@ void launch_ppo_gather_disc(const long*, int, const float*, int, int,
                             const float*, const float*, void*, long*, float*,
                             float*, float*, float*, const float*,
                             const float*, void*);
+void launch_rnn_scan(const float*, const void*, const float*,
+                     const unsigned char*, const float*, const float*,
+                     float*, float*, float*, int, int, int, int, void*);
 }
 
 namespace {
@@ -597,7 +600,30 @@ void sumtree_sample(torch::Tensor tree, torch::Tensor u, torch::Tensor out,
                         (int)n_items, cur_stream());
 }
 
+void rnn_scan(torch::Tensor Xp, torch::Tensor Whh, torch::Tensor bhh,
+              torch::Tensor resets, torch::Tensor h0, torch::Tensor c0,
+              torch::Tensor Hout, torch::Tensor hT, torch::Tensor cT,
+              int64_t lstm) {
+  CHK(Xp, torch::kFloat32);
+  CHK(Whh, torch::kBFloat16);
+  CHK(resets, torch::kUInt8);
+  CHK(h0, torch::kFloat32);
+  int T = Xp.size(0), B = Xp.size(1);
+  int H = h0.size(1);
+  TORCH_CHECK(H == 128 || H == 256, "rnn_scan supports H 128/256");
+  TORCH_CHECK(Xp.size(2) == (lstm ? 4 : 3) * H, "Xp gate dim mismatch");
+  launch_rnn_scan(
+      Xp.data_ptr<float>(), Whh.data_ptr(), bhh.data_ptr<float>(),
+      resets.data_ptr<unsigned char>(), h0.data_ptr<float>(),
+      c0.numel() > 0 ? c0.data_ptr<float>() : nullptr,
+      Hout.data_ptr<float>(), hT.data_ptr<float>(),
+      cT.numel() > 0 ? cT.data_ptr<float>() : nullptr, T, B, H, (int)lstm,
+      cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("rnn_scan", &rnn_scan,
+        "fused done-masked GRU/LSTM sequence scan (K13)");
   m.def("policy_value_step_disc", &policy_value_step_disc,
         "fused actor+critic fwd, categorical Gumbel-max sample (MFMA)");
   m.def("ppo_head_loss_disc", &ppo_head_loss_disc,

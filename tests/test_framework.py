@@ -242,3 +242,31 @@ def test_learner_checkpoint_roundtrip(tmp_path):
     a1 = learner.act_fn(obs, True)
     a2 = learner2.act_fn(obs, True)
     torch.testing.assert_close(a1, a2)
+
+
+def test_scanned_rnn_hoisted_matches_per_step():
+    """The hoisted-GEMM training scan (input projection for all T in one
+    GEMM) must match the generic per-step cell path bit-closely for GRU
+    and LSTM, including done-masked resets (K13 torch side)."""
+    import torch
+
+    from stoix_amd.networks.base import ScannedRNN
+
+    torch.manual_seed(0)
+    for kind in ("gru", "lstm"):
+        rnn = ScannedRNN(12, 32, cell_type=kind)
+        T, B = 9, 7
+        x = torch.randn(T, B, 12)
+        resets = torch.rand(T, B) < 0.3
+        st0 = rnn.initial_state(B, "cpu")
+        out_fast, st_fast = rnn(x, resets, list(st0))
+        orig = rnn._single_cell
+        rnn._single_cell = lambda: (None, None)  # force generic path
+        out_ref, st_ref = rnn(x, resets, list(st0))
+        rnn._single_cell = orig
+        torch.testing.assert_close(out_fast, out_ref, rtol=1e-5, atol=1e-6)
+        if kind == "gru":
+            torch.testing.assert_close(st_fast[0], st_ref[0], rtol=1e-5, atol=1e-6)
+        else:
+            torch.testing.assert_close(st_fast[0][0], st_ref[0][0], rtol=1e-5, atol=1e-6)
+            torch.testing.assert_close(st_fast[0][1], st_ref[0][1], rtol=1e-5, atol=1e-6)
