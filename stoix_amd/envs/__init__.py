@@ -13,9 +13,15 @@ dispatch over suites (:420-433) and ``make(config) -> (train_env, eval_env)``
     (envs/csrc/envpool_cpu.cpp), LunarLander, and the cartpole alias;
   * ``debug`` — five diagnostic games.
 
-JAX-only external suites with no offline equivalent (craftax, xland,
-navix, kinetix, popjym, playground, jaxarc) raise a clear error naming
-what is available.
+  * ``xland_minigrid`` — goal-conditioned procedural gridworld
+    (envs/xland.py; capability-class equivalent of the reference's
+    JAX-only xminigrid suite);
+  * ``craftax`` — crafting/achievement-chain world (envs/crafting.py;
+    capability-class equivalent of the JAX-only craftax suite).
+
+JAX-only external suites with no offline equivalent (navix, kinetix,
+popjym, playground, jaxarc) raise a clear error naming what is
+available.
 """
 from __future__ import annotations
 
@@ -133,6 +139,24 @@ def _debug(name: str):
     return DEBUG_ENVIRONMENTS[name]
 
 
+def _xland(name: str):
+    from stoix_amd.envs.xland import XLandGrid
+
+    table = {"goal_grid": XLandGrid, "XLand-MiniGrid-R1": XLandGrid}
+    if name not in table:
+        raise ValueError(f"unknown xland-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
+def _craftax(name: str):
+    from stoix_amd.envs.crafting import Crafting
+
+    table = {"crafting": Crafting, "Craftax-Symbolic-v1": Crafting}
+    if name not in table:
+        raise ValueError(f"unknown craftax-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
 def _gymnasium(name: str):
     # the gymnasium suite is factory-only (stateful CPU envs for Sebulba,
     # reference utils/env_factory.py:71-86); Anakin's make_single cannot
@@ -154,6 +178,9 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "jumanji": _jumanji,
     "envpool": _envpool,
     "gymnasium": _gymnasium,
+    "xland_minigrid": _xland,
+    "xland": _xland,
+    "craftax": _craftax,
     "debug": _debug,
 }
 

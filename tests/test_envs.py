@@ -723,3 +723,152 @@ def test_lunarlander_rules():
             assert float(ts.reward[0]) > 50.0, "gentle pad touchdown must pay +100"
             break
     assert landed
+
+
+# ----------------------------------------- xland-class goal-conditioned grid
+
+
+def test_xland_goal_reward_and_termination():
+    """Reaching the GOAL-coloured object gives +1 and terminates; a wrong
+    object gives -0.1, is consumed, and the episode continues."""
+    import torch
+
+    from stoix_amd.envs.xland import N, XLandGrid
+
+    env = XLandGrid(4, seed=0)
+    env.reset()
+    s = env._state
+    # place a deterministic scene: agent at (4,4), goal colour 0 at (4,5),
+    # wrong colour 1 at (4,3); clear interior walls around them
+    s["walls"][:] = 0.0
+    s["walls"][:, 0, :] = s["walls"][:, -1, :] = 1.0
+    s["walls"][:, :, 0] = s["walls"][:, :, -1] = 1.0
+    s["agent"][:] = float(4 * N + 4)
+    s["goal"][:] = 0.0
+    s["obj_pos"][:, 0] = float(4 * N + 5)
+    s["obj_pos"][:, 1] = float(4 * N + 3)
+    s["obj_pos"][:, 2] = float(1 * N + 1)
+    s["obj_pos"][:, 3] = float(1 * N + 2)
+    s["obj_alive"][:] = 1.0
+
+    # move RIGHT (action 1) onto the goal object
+    ts = env.step(torch.full((4,), 1, dtype=torch.long))
+    assert torch.all(ts.reward == 1.0)
+    assert torch.all(ts.discount == 0.0)  # terminated
+
+    # fresh scene; move LEFT onto the wrong object
+    env.reset()
+    s = env._state
+    s["walls"][:] = 0.0
+    s["walls"][:, 0, :] = s["walls"][:, -1, :] = 1.0
+    s["walls"][:, :, 0] = s["walls"][:, :, -1] = 1.0
+    s["agent"][:] = float(4 * N + 4)
+    s["goal"][:] = 0.0
+    s["obj_pos"][:, 0] = float(4 * N + 6)
+    s["obj_pos"][:, 1] = float(4 * N + 3)
+    s["obj_pos"][:, 2] = float(1 * N + 1)
+    s["obj_pos"][:, 3] = float(1 * N + 2)
+    s["obj_alive"][:] = 1.0
+    ts = env.step(torch.full((4,), 3, dtype=torch.long))  # LEFT
+    assert torch.allclose(ts.reward, torch.full((4,), -0.1))
+    assert torch.all(ts.discount == 1.0)  # not terminal
+    assert torch.all(env._state["obj_alive"][:, 1] == 0.0)  # consumed
+
+
+def test_xland_walls_block_and_goal_visible_in_obs():
+    import torch
+
+    from stoix_amd.envs.xland import N, NUM_COLORS, XLandGrid
+
+    env = XLandGrid(2, seed=1)
+    ts = env.reset()
+    assert ts.observation.shape == (2, N, N, 3 + NUM_COLORS)
+    # goal plane: exactly one colour plane carries the +0.5 broadcast
+    goal = env._state["goal"].long()
+    obs = ts.observation
+    for b in range(2):
+        g = int(goal[b])
+        # off-object cells of the goal plane are exactly 0.5
+        plane = obs[b, :, :, 3 + g]
+        assert torch.isclose(plane.min(), torch.tensor(0.5))
+    # stepping into a border wall keeps the agent in place
+    env._state["agent"][:] = float(1 * N + 1)
+    env._state["walls"][:, 0, 1] = 1.0
+    before = env._state["agent"].clone()
+    env.step(torch.zeros(2, dtype=torch.long))  # UP into the border
+    assert torch.all(env._state["agent"] == before)
+
+
+# ------------------------------------------------- craftax-class crafting
+
+
+def test_crafting_achievement_chain():
+    """wood -> table -> pickaxe -> stone: each first-time achievement pays
+    +1 exactly once; completing all four terminates (return 4.0)."""
+    import torch
+
+    from stoix_amd.envs.crafting import M, Crafting
+
+    env = Crafting(2, seed=0)
+    env.reset()
+    s = env._state
+    # deterministic scene: agent at (5,5) facing UP; tree above, stone two
+    # to the right, empty elsewhere
+    s["grid"][:] = 0.0
+    pos = 5 * M + 5
+    s["agent"][:] = float(pos)
+    s["facing"][:] = 0.0
+    s["grid"][:, pos - M] = 1.0  # tree at (4,5)
+    s["grid"][:, pos + 2] = 2.0  # stone at (5,7)
+    for k in ("wood", "stone", "pickaxe", "ach_wood", "ach_table", "ach_pick", "ach_stone"):
+        s[k][:] = 0.0
+
+    up = torch.zeros(2, dtype=torch.long)
+    interact = torch.full((2,), 4, dtype=torch.long)
+    craft = torch.full((2,), 5, dtype=torch.long)
+
+    ts = env.step(interact)  # chop the faced tree
+    assert torch.all(ts.reward == 1.0)  # collect_wood achievement
+    assert torch.all(env._state["wood"] == 1.0)
+    ts = env.step(interact)  # nothing left to chop
+    assert torch.all(ts.reward == 0.0)
+
+    env._state["wood"][:] = 2.0  # give enough for a table
+    ts = env.step(craft)  # place table on the faced empty cell
+    assert torch.all(ts.reward == 1.0)  # place_table achievement
+    assert torch.all(env._state["grid"][:, pos - M] == 3.0)
+    assert torch.all(env._state["wood"] == 0.0)
+
+    env._state["wood"][:] = 1.0
+    ts = env.step(craft)  # next to the table now -> pickaxe
+    assert torch.all(ts.reward == 1.0)  # make_pickaxe achievement
+    assert torch.all(env._state["pickaxe"] == 1.0)
+
+    # face the stone: move right twice is blocked by... cells are empty;
+    # walk right once, then interact on the stone at (5,7)
+    right = torch.full((2,), 1, dtype=torch.long)
+    env.step(right)  # agent to (5,6), facing right
+    ts = env.step(interact)  # mine stone with the pickaxe
+    assert torch.all(ts.reward == 1.0)  # collect_stone achievement
+    assert torch.all(ts.discount == 0.0)  # all 4 achievements -> done
+    em = ts.extras["episode_metrics"]
+    assert torch.all(em["episode_return"] == 4.0)
+
+
+def test_crafting_stone_needs_pickaxe():
+    import torch
+
+    from stoix_amd.envs.crafting import M, Crafting
+
+    env = Crafting(1, seed=0)
+    env.reset()
+    s = env._state
+    s["grid"][:] = 0.0
+    pos = 5 * M + 5
+    s["agent"][:] = float(pos)
+    s["facing"][:] = 1.0  # facing right
+    s["grid"][:, pos + 1] = 2.0  # stone
+    s["pickaxe"][:] = 0.0
+    ts = env.step(torch.full((1,), 4, dtype=torch.long))
+    assert torch.all(ts.reward == 0.0)
+    assert torch.all(env._state["grid"][:, pos + 1] == 2.0)  # still there
