@@ -96,7 +96,7 @@ void breakout_cpu_reset(torch::Tensor state, torch::Tensor obs,
   int64_t B = state.size(0);
   float* sp = state.data_ptr<float>();
   float* op = obs.data_ptr<float>();
-  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+  at::parallel_for(0, B, std::max<int64_t>(1, B / 64), [&](int64_t lo, int64_t hi) {
     for (int64_t b = lo; b < hi; ++b) {
       reset_env(sp + b * SDIM, (uint64_t)seed, (uint64_t)b, (uint64_t)draw);
       render(sp + b * SDIM, op + b * H * W);
@@ -128,7 +128,7 @@ void breakout_cpu_step(torch::Tensor state, torch::Tensor action,
   uint8_t* dnp = done.data_ptr<uint8_t>();
   int32_t dr = *draw.data_ptr<int32_t>();
 
-  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+  at::parallel_for(0, B, std::max<int64_t>(1, B / 64), [&](int64_t lo, int64_t hi) {
     for (int64_t b = lo; b < hi; ++b) {
       float* s = sp + b * SDIM;
       int a = (int)ap[b];
@@ -255,7 +255,7 @@ void pong_cpu_reset(torch::Tensor state, torch::Tensor obs, int64_t seed,
   int64_t B = state.size(0);
   float* sp = state.data_ptr<float>();
   float* op = obs.data_ptr<float>();
-  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+  at::parallel_for(0, B, std::max<int64_t>(1, B / 64), [&](int64_t lo, int64_t hi) {
     for (int64_t b = lo; b < hi; ++b) {
       pong::reset_env(sp + b * pong::SDIM, (uint64_t)seed, (uint64_t)b,
                       (uint64_t)draw);
@@ -288,7 +288,7 @@ void pong_cpu_step(torch::Tensor state, torch::Tensor action,
   uint8_t* dnp = done.data_ptr<uint8_t>();
   int32_t dr = *draw.data_ptr<int32_t>();
 
-  at::parallel_for(0, B, 1, [&](int64_t lo, int64_t hi) {
+  at::parallel_for(0, B, std::max<int64_t>(1, B / 64), [&](int64_t lo, int64_t hi) {
     for (int64_t b = lo; b < hi; ++b) {
       float* s = sp + b * pong::SDIM;
       int a = (int)ap[b];  // 0 noop, 1 up, 2 down

@@ -59,9 +59,14 @@ class _PoolEnvBase(StatefulVecEnv):
         self.seed = int(seed)
         B = self.num_envs
         sdim = int(getattr(self._ext, self.STATE_DIM_ATTR))
+        pin = torch.cuda.is_available()
         self._s = torch.zeros(B, sdim, dtype=torch.float32)
-        self._obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
-        self._next_obs = torch.zeros(B, H, W, 1, dtype=torch.float32)
+        # pinned output buffers: the Sebulba actor transfers the CURRENT
+        # obs to its device for inference straight from these (pinned H2D
+        # DMA ~50 GB/s vs ~3.5 GB/s pageable — measured 8.2 -> <1 ms per
+        # 1024-env step), then keeps a CPU clone for the payload
+        self._obs = torch.zeros(B, H, W, 1, dtype=torch.float32, pin_memory=pin)
+        self._next_obs = torch.zeros(B, H, W, 1, dtype=torch.float32, pin_memory=pin)
         self._reward = torch.zeros(B, dtype=torch.float32)
         self._discount = torch.zeros(B, dtype=torch.float32)
         self._steptype = torch.zeros(B, dtype=torch.uint8)

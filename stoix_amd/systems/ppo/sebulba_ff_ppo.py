@@ -86,12 +86,35 @@ def actor_thread_fn(
             critic_net.load_state_dict(params["critic"])
 
         obs_l, act_l, logp_l, val_l, rew_l, disc_l, trunc_l = [], [], [], [], [], [], []
+        # pinned fast path: native-pool envs expose their pinned CURRENT-obs
+        # buffer; transfer THAT for inference (DMA) and keep the clone the
+        # env returned for the payload. bf16 autocast for the actor nets
+        # (inference only; the learner trains its own copy).
+        pinned_src = getattr(env, "_obs", None)
+        use_pin = (
+            actor_device.type == "cuda"
+            and pinned_src is not None
+            and pinned_src.is_pinned()
+        )
+        amp = (
+            torch.autocast("cuda", torch.bfloat16)
+            if actor_device.type == "cuda"
+            else None
+        )
         with torch.no_grad():
             for _ in range(T):
-                obs_dev = ts.observation.to(actor_device)
+                if use_pin:
+                    obs_dev = pinned_src.to(actor_device, non_blocking=True)
+                else:
+                    obs_dev = ts.observation.to(actor_device)
                 with timers.time("inference"):
-                    dist = actor_net(obs_dev)
-                    value = critic_net(obs_dev)
+                    if amp is not None:
+                        with amp:
+                            dist = actor_net(obs_dev)
+                            value = critic_net(obs_dev)
+                    else:
+                        dist = actor_net(obs_dev)
+                        value = critic_net(obs_dev)
                     action = dist.sample(gen)
                     logp = dist.log_prob(action)
                 cpu_action = action.cpu()
@@ -99,14 +122,18 @@ def actor_thread_fn(
                     next_ts = env.step(cpu_action)
                 obs_l.append(ts.observation)
                 act_l.append(cpu_action)
-                logp_l.append(logp.cpu())
-                val_l.append(value.cpu())
+                logp_l.append(logp.float().cpu())
+                val_l.append(value.float().cpu())
                 rew_l.append(next_ts.reward)
                 disc_l.append(next_ts.discount)
                 trunc_l.append(next_ts.truncated())
                 ts = next_ts
             # bootstrap value for the trace
-            last_val = critic_net(ts.observation.to(actor_device)).cpu()
+            if amp is not None:
+                with amp:
+                    last_val = critic_net(ts.observation.to(actor_device)).float().cpu()
+            else:
+                last_val = critic_net(ts.observation.to(actor_device)).cpu()
 
         payload = {
             "obs": torch.stack(obs_l),
@@ -213,6 +240,12 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     config.arch.n_devices = ctx.world_size
     check_total_timesteps(config)
     torch.manual_seed(int(config.arch.seed) + ctx.rank)
+    # cap the intra-op pool: on big hosts (256 cores) torch defaults to
+    # half the cores and the per-task overhead of batched env stepping
+    # collapses (measured 23.8K SPS at 128 threads vs 503K at 4)
+    import os as _os
+
+    torch.set_num_threads(min(8, _os.cpu_count() or 8))
 
     actor_device_ids = list(config.arch.actor.device_ids)
     per_dev = int(config.arch.actor.actor_per_device)
