@@ -323,6 +323,15 @@ def run_experiment(config, force_cpu: bool = False) -> float:
                 logger.log({k: v.mean() for k, v in final.items()}, t_env, update, LogEvent.ACT)
             async_eval.submit_evaluation(learner.cpu_params(), t_env)
 
+    # train-phase throughput, recorded BEFORE the evaluator drain: the
+    # async evaluator plays full episodes to completion at shutdown, and a
+    # better policy means a LONGER drain — wall-clock around the whole
+    # experiment anti-correlates with training quality (bench/probe read
+    # this attribute instead)
+    train_elapsed = time.perf_counter() - t0
+    run_experiment.last_sps = t_env / max(train_elapsed, 1e-9)
+    run_experiment.last_env_steps = t_env
+
     lifetime.stop()
     for th in threads:
         th.join(timeout=10)

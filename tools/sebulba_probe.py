@@ -96,7 +96,8 @@ def probe_e2e(n_envs, actors, rollout=64, updates=20):
     run_experiment(cfg)
     dt = time.perf_counter() - t0
     return {"probe": "e2e", "n_envs": n_envs, "actors": actors,
-            "rollout": rollout, "sps": n_envs * rollout * updates / dt,
+            "rollout": rollout,
+            "sps": getattr(run_experiment, "last_sps", n_envs * rollout * updates / dt),
             "wall_s": dt}
 
 
