@@ -246,6 +246,11 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     import os as _os
 
     torch.set_num_threads(min(8, _os.cpu_count() or 8))
+    # MIOpen exhaustive find on a fresh box costs ~minutes for the CNN
+    # shapes before the first step (measured: first in-process run 8.8K
+    # SPS over a 250 s wall, later runs 70K); immediate-find mode trades
+    # a few % conv throughput for a fast first launch
+    _os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
 
     actor_device_ids = list(config.arch.actor.device_ids)
     per_dev = int(config.arch.actor.actor_per_device)
