@@ -246,11 +246,11 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     import os as _os
 
     torch.set_num_threads(min(8, _os.cpu_count() or 8))
-    # MIOpen exhaustive find on a fresh box costs ~minutes for the CNN
-    # shapes before the first step (measured: first in-process run 8.8K
-    # SPS over a 250 s wall, later runs 70K); immediate-find mode trades
-    # a few % conv throughput for a fast first launch
-    _os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+    # NOTE: the FIRST run on a fresh box pays MIOpen conv auto-tuning for
+    # the CNN shapes (~minutes, one-off, cached afterwards): measured 8.8K
+    # SPS on the tuning run vs 68-73K steady-state. MIOPEN_FIND_MODE=FAST
+    # was tried and measured WORSE (fallback-path pathology on gfx950);
+    # leave MIOpen's default find in place.
 
     actor_device_ids = list(config.arch.actor.device_ids)
     per_dev = int(config.arch.actor.actor_per_device)
