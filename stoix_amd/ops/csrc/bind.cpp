@@ -307,7 +307,8 @@ void policy_value_step(torch::Tensor obs, torch::Tensor W1a, torch::Tensor b1a,
   int B = obs.size(0), OBS = obs.size(1);
   int HID = W2a.size(0);
   int ACT = action_out.size(1);
-  TORCH_CHECK(HID == 256 || HID == 128, "fused MLP supports HID 128/256");
+  TORCH_CHECK(HID == 512 || HID == 256 || HID == 128,
+              "fused MLP supports HID 128/256/512");
   TORCH_CHECK(ACT <= 8, "fused policy head supports ACT <= 8");
   TORCH_CHECK(OBS <= 128, "fused MLP supports OBS <= 128");
   float* om = obs_mirror.numel() > 0 ? obs_mirror.data_ptr<float>() : nullptr;
@@ -342,7 +343,8 @@ void policy_value_step_disc(
   int B = obs.size(0), OBS = obs.size(1);
   int HID = W2a.size(0);
   int ACT = (int)num_actions;
-  TORCH_CHECK(HID == 256 || HID == 128, "fused MLP supports HID 128/256");
+  TORCH_CHECK(HID == 512 || HID == 256 || HID == 128,
+              "fused MLP supports HID 128/256/512");
   TORCH_CHECK(ACT <= 16, "fused categorical head supports ACT <= 16");
   TORCH_CHECK(OBS <= 128, "fused MLP supports OBS <= 128");
   float* om = obs_mirror.numel() > 0 ? obs_mirror.data_ptr<float>() : nullptr;

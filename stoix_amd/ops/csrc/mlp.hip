@@ -1061,7 +1061,15 @@ extern "C" void launch_policy_value_step(
     unsigned int draw_offset, int do_bump, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((B + 15) / 16), block(256);
-  if (HID == 256) {
+  if (HID == 512) {
+    hipLaunchKernelGGL(policy_value_step_kernel<512>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, min_scale, aff_scale, aff_shift,
+                       log_aff_scale, greedy, seed, draw_buf, draw_offset);
+  } else if (HID == 256) {
     hipLaunchKernelGGL(policy_value_step_kernel<256>, grid, block, 0, s, obs,
                        (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
                        (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
@@ -1090,7 +1098,12 @@ extern "C" void launch_value_forward(const float* obs, const void* W1c,
                                      int B, int OBS, int HID, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((B + 15) / 16), block(256);
-  if (HID == 256) {
+  if (HID == 512) {
+    hipLaunchKernelGGL(value_forward_kernel<512>, grid, block, 0, s, obs,
+                       (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
+                       (const bf16_t*)Wvc, bvc, value_out, nmean, nvar, B,
+                       OBS);
+  } else if (HID == 256) {
     hipLaunchKernelGGL(value_forward_kernel<256>, grid, block, 0, s, obs,
                        (const bf16_t*)W1c, b1c, (const bf16_t*)W2c, b2c,
                        (const bf16_t*)Wvc, bvc, value_out, nmean, nvar, B,
@@ -1435,7 +1448,14 @@ extern "C" void launch_policy_value_step_disc(
     unsigned int draw_offset, int do_bump, void* stream) {
   hipStream_t s = (hipStream_t)stream;
   dim3 grid((B + 15) / 16), block(256);
-  if (HID == 256) {
+  if (HID == 512) {
+    hipLaunchKernelGGL(policy_value_step_disc_kernel<512>, grid, block, 0, s,
+                       obs, (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
+                       (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,
+                       (const bf16_t*)W2c, b2c, (const bf16_t*)Wvc, bvc,
+                       obs_mirror, action_out, logp_out, value_out, nmean,
+                       nvar, B, OBS, ACT, greedy, seed, draw_buf, draw_offset);
+  } else if (HID == 256) {
     hipLaunchKernelGGL(policy_value_step_disc_kernel<256>, grid, block, 0, s,
                        obs, (const bf16_t*)W1a, b1a, (const bf16_t*)W2a, b2a,
                        (const bf16_t*)Wha, bha, (const bf16_t*)W1c, b1c,

@@ -71,6 +71,25 @@ def update(
     return RunningStatisticsState(count=new_count, mean=new_mean, summed_variance=new_m2, std=std)
 
 
+@torch.no_grad()
+def update_(
+    state: RunningStatisticsState,
+    batch: Tensor,
+    std_min_value: float = 1e-6,
+    std_max_value: float = 1e6,
+    all_reduce: bool = False,
+) -> RunningStatisticsState:
+    """In-place Welford update: writes into the state's STABLE tensors so
+    kernels (and hip graphs) holding their addresses see fresh statistics
+    without rebinding (fused PPO obs-norm path)."""
+    new = update(state, batch, std_min_value, std_max_value, all_reduce)
+    state.count.copy_(new.count)
+    state.mean.copy_(new.mean)
+    state.summed_variance.copy_(new.summed_variance)
+    state.std.copy_(new.std)
+    return state
+
+
 def normalize(batch: Tensor, state: RunningStatisticsState, max_abs_value: Optional[float] = None) -> Tensor:
     out = (batch - state.mean) / state.std
     if max_abs_value is not None:

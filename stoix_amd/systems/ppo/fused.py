@@ -136,8 +136,6 @@ class FusedPPOEngine:
     def try_build(learner) -> Optional["FusedPPOEngine"]:
         if learner.device.type != "cuda":
             return None
-        if getattr(learner, "normalize_obs", False):
-            return None
         if type(learner).policy_loss is not _base_policy_loss_func(learner):
             # PPO-penalty / DPO override policy_loss; the fused head kernel
             # implements the clip loss only.
@@ -168,7 +166,7 @@ class FusedPPOEngine:
         if not isinstance(critic.critic_head, ScalarCriticHead):
             return None
         H = a_lins[0].out_features
-        if H not in (128, 256) or a_lins[1].out_features != H:
+        if H not in (128, 256, 512) or a_lins[1].out_features != H:
             return None
         if c_lins[0].out_features != H or c_lins[1].out_features != H:
             return None
@@ -357,6 +355,21 @@ class FusedPPOEngine:
 
         self.world = dist.get_world_size() if dist.is_initialized() else 1
 
+        # observation normalisation (reference ff_ppo.py:90-162): the
+        # kernels read STABLE mean/var buffers (stage_obs / ppo_gather
+        # nmean/nvar args, plumbed since round 1); the Welford update runs
+        # in place after each rollout. Semantics match the reference
+        # ordering: act with CURRENT stats, normalise the stored
+        # trajectory for the update with the PRE-update stats (gmean/gvar
+        # snapshot), THEN update the stats from the raw trajectory.
+        self.norm = bool(getattr(learner, "normalize_obs", False))
+        if self.norm:
+            st = learner.obs_stats
+            self.nmean = st.mean  # stable (updated in place by rs.update_)
+            self.nvar = (st.std ** 2).clone()
+            self.gmean = st.mean.clone()
+            self.gvar = self.nvar.clone()
+
         # the rollout reads the env's stable obs buffer directly; seed it
         # with the current observation (reset happened before attach)
         learner.env._hb["obs"].copy_(learner.cur_obs)
@@ -428,7 +441,9 @@ class FusedPPOEngine:
         ac, cc = self.actor_chain, self.critic_chain
         a16, c16 = ac.views16, cc.views16
         ext = self.ext
-        if getattr(env, "HIP_KERNEL", "") == "ant_step":
+        nmean = self.nmean if self.norm else self.empty
+        nvar = self.nvar if self.norm else self.empty
+        if getattr(env, "HIP_KERNEL", "") == "ant_step" and not self.norm:
             # megakernel path: ONE launch per rollout step (policy + Ant
             # physics + bootstrap critic; mlp.hip rollout_step_ant_kernel)
             for t in range(L.T):
@@ -476,7 +491,7 @@ class FusedPPOEngine:
                     c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
                     c16["Wv"], cc.views["bv"],
                     L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t],
-                    L.buf_value[t], self.empty, self.empty,
+                    L.buf_value[t], nmean, nvar,
                     self.ACT, 0, self.seed, self.draw_policy, t, 0,
                 )
             else:
@@ -487,7 +502,7 @@ class FusedPPOEngine:
                     c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
                     c16["Wv"], cc.views["bv"],
                     L.buf_obs[t], L.buf_action[t], L.buf_log_prob[t], L.buf_value[t],
-                    self.empty, self.empty,
+                    nmean, nvar,
                     self.min_scale, self.aff_scale, self.aff_shift,
                     self.log_aff_scale, 0, self.seed, self.draw_policy, t, 0,
                 )
@@ -503,7 +518,7 @@ class FusedPPOEngine:
                 hb["next_obs"],
                 c16["W1"], cc.views["b1"], c16["W2"], cc.views["b2"],
                 c16["Wv"], cc.views["bv"],
-                L.buf_bootstrap[t], self.empty, self.empty,
+                L.buf_bootstrap[t], nmean, nvar,
             )
         # one counter bump per rollout (the per-step draws used frozen
         # offsets t baked into the graph nodes)
@@ -511,6 +526,20 @@ class FusedPPOEngine:
         ext.bump_add(hb["draw"], L.T)
         # truncation flags for GAE (StepType.TRUNCATED == 3)
         L.buf_truncated.copy_(self.buf_steptype == 3)
+        if self.norm:
+            self._update_obs_stats()
+
+    def _update_obs_stats(self) -> None:
+        """Snapshot the PRE-update stats for the epoch gathers, then run
+        the in-place Welford update over the raw stored trajectory
+        (reference ff_ppo.py:150-162 ordering)."""
+        from stoix_amd.ops import running_statistics as rs
+
+        L = self.learner
+        self.gmean.copy_(self.nmean)
+        self.gvar.copy_(self.nvar)
+        rs.update_(L.obs_stats, L.buf_obs, all_reduce=self.world > 1)
+        torch.pow(L.obs_stats.std, 2, out=self.nvar)
 
     # ------------------------------------------------------------- update
 
@@ -537,13 +566,15 @@ class FusedPPOEngine:
         import torch.distributed as dist
 
         gather_fn = ext.ppo_gather_disc if self.discrete else ext.ppo_gather
+        gmean = self.gmean if self.norm else self.empty
+        gvar = self.gvar if self.norm else self.empty
 
         def gather_into(mb: int, w) -> None:
             idx = L.perm_buf[mb * S : (mb + 1) * S]
             gather_fn(
                 idx, flat_obs, flat_action, flat_logp, flat_value, flat_adv,
                 flat_tgt, w.Xmb, w.act, w.logp, w.val, w.adv, w.tgt,
-                self.empty, self.empty,
+                gmean, gvar,
             )
 
         # gather prefetch: minibatch mb+1's gather runs on a side stream

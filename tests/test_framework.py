@@ -270,3 +270,25 @@ def test_scanned_rnn_hoisted_matches_per_step():
         else:
             torch.testing.assert_close(st_fast[0][0], st_ref[0][0], rtol=1e-5, atol=1e-6)
             torch.testing.assert_close(st_fast[0][1], st_ref[0][1], rtol=1e-5, atol=1e-6)
+
+
+def test_running_statistics_update_inplace_matches_functional():
+    """rs.update_ must leave the state's ORIGINAL tensors holding the same
+    values as the functional update (the fused engine's kernels hold those
+    addresses across hip-graph replays)."""
+    import torch
+
+    from stoix_amd.ops import running_statistics as rs
+
+    torch.manual_seed(0)
+    s1 = rs.init_state((5,))
+    s2 = rs.init_state((5,))
+    mean_ptr = s2.mean.data_ptr()
+    for _ in range(3):
+        batch = torch.randn(64, 5) * 3 + 1
+        s1 = rs.update(s1, batch)
+        rs.update_(s2, batch)
+    assert s2.mean.data_ptr() == mean_ptr  # stable address
+    torch.testing.assert_close(s1.mean, s2.mean)
+    torch.testing.assert_close(s1.std, s2.std)
+    torch.testing.assert_close(s1.count, s2.count)

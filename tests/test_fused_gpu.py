@@ -653,3 +653,42 @@ def test_fused_discrete_cartpole_update_runs(ext):
         assert torch.isfinite(v).all()
     w_after = learner.actor.action_head.linear.weight.detach()
     assert (w_after - w_before).abs().max() > 0
+
+
+@requires_gpu
+def test_fused_obs_norm_cartpole(ext):
+    """Fused engine with observation normalisation: stats update in place
+    each rollout, the trajectory is normalised with PRE-update stats, and
+    the update stays finite (reference ff_ppo.py:90-162 path)."""
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=1024",
+         "arch.total_timesteps=null", "arch.num_updates=4",
+         "arch.num_evaluation=1", "system.rollout_length=16",
+         "system.num_minibatches=4", "system.epochs=2",
+         "system.compute_dtype=bf16", "system.normalize_observations=true",
+         "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    dev = torch.device("cuda:0")
+    env = environments.make_single(cfg, 1024, dev, seed=3)
+    learner = PPOLearner(cfg, env, dev)
+    assert learner.normalize_obs
+    assert learner.fused is not None and learner.fused.norm
+    c0 = float(learner.obs_stats.count)
+    for _ in range(3):
+        m = learner.update_step()
+    torch.cuda.synchronize()
+    for v in m.values():
+        assert torch.isfinite(v).all()
+    # Welford state advanced by T*B per rollout, in place
+    assert float(learner.obs_stats.count) == c0 + 3 * 16 * 1024
+    assert float(learner.obs_stats.std.min()) > 0
+    # stats are no longer the init values (mean moved)
+    assert float(learner.obs_stats.mean.abs().sum()) > 0
