@@ -87,7 +87,7 @@ def _eager_forward(obs, W1, b1, W2, b2):
 
 
 @requires_gpu
-@pytest.mark.parametrize("H,OBS,ACT", [(256, 27, 8), (256, 64, 4), (128, 27, 8)])
+@pytest.mark.parametrize("H,OBS,ACT", [(256, 27, 8), (256, 64, 4), (128, 27, 8), (512, 27, 8)])
 def test_policy_value_step_matches_eager(ext, H, OBS, ACT):
     device = "cuda"
     B = 256
@@ -503,7 +503,7 @@ def test_linear_silu_kernel_matches_torch(ext, K):
 
 
 @requires_gpu
-@pytest.mark.parametrize("H,OBS,ACT", [(256, 4, 2), (256, 27, 6), (128, 10, 16)])
+@pytest.mark.parametrize("H,OBS,ACT", [(256, 4, 2), (256, 27, 6), (128, 10, 16), (512, 4, 2)])
 def test_policy_value_step_disc_matches_eager(ext, H, OBS, ACT):
     """Fused categorical rollout kernel vs plain fp32 eager: value head,
     greedy action = argmax(logits), logp = log_softmax(logits)[a]."""
