@@ -117,11 +117,27 @@ def _envpool(name: str):
         if envpool_ext(required=True) is not None:
             return PongCpu(**kw)
 
+    def spaceinv_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import SpaceInvadersCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return SpaceInvadersCpu(**kw)
+
+    def qbert_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import QbertCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return QbertCpu(**kw)
+
     table = {
         "breakout": breakout_maker,
         "Breakout-v5": breakout_maker,
         "pong": pong_maker,
         "Pong-v5": pong_maker,
+        "space_invaders": spaceinv_maker,
+        "SpaceInvaders-v5": spaceinv_maker,
+        "qbert": qbert_maker,
+        "Qbert-v5": qbert_maker,
         # classic-control names envpool also serves
         "cartpole": CartPole,
         "CartPole-v1": CartPole,

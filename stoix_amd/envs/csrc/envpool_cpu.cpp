@@ -377,6 +377,349 @@ void pong_cpu_step(torch::Tensor state, torch::Tensor action,
   *draw.data_ptr<int32_t>() = dr + 1;
 }
 
+
+// --------------------------------------------------- SpaceInvaders-class
+// Alien grid marches and descends; the cannon fires one shot at a time;
+// a random alive alien drops bombs. +1 per alien; clearing the wave pays
+// +5 and terminates; a bomb hit or aliens reaching the cannon row ends
+// the episode.
+namespace spaceinv {
+constexpr int AR = 5, AC = 8, NA = AR * AC;
+// state: player_x, cooldown, shot_x, shot_y, shot_alive,
+//        bomb_x, bomb_y, bomb_alive, adir, aoffx, aoffy, aliens[NA]
+constexpr int SOFF = 11;
+constexpr int SDIM = SOFF + NA;
+constexpr float PSPEED = 2.5f, SHOT_V = 3.0f, BOMB_V = 1.2f;
+constexpr int PW = 8, PY = 80;
+
+inline void reset_env(float* s, uint64_t seed, uint64_t env, uint64_t draw) {
+  s[0] = W / 2.0f;
+  s[1] = 0.0f;
+  s[2] = s[3] = 0.0f; s[4] = 0.0f;
+  s[5] = s[6] = 0.0f; s[7] = 0.0f;
+  s[8] = (hash_uniform(seed, env * 3, draw) > 0.5f) ? 1.0f : -1.0f;
+  s[9] = 6.0f;
+  s[10] = 8.0f;
+  for (int k = 0; k < NA; ++k) s[SOFF + k] = 1.0f;
+}
+
+inline void render(const float* s, float* obs) {
+  std::memset(obs, 0, sizeof(float) * H * W);
+  // aliens: 5x3 blocks on a 9x7 lattice from (aoffx, aoffy)
+  for (int r = 0; r < AR; ++r)
+    for (int c = 0; c < AC; ++c) {
+      if (s[SOFF + r * AC + c] <= 0.0f) continue;
+      int y0 = (int)s[10] + r * 7, x0 = (int)s[9] + c * 9;
+      for (int y = y0; y < y0 + 3; ++y)
+        for (int x = x0; x < x0 + 5; ++x)
+          if (y >= 0 && y < H && x >= 0 && x < W) obs[y * W + x] = 0.6f;
+    }
+  // player cannon
+  int px = (int)s[0];
+  for (int dx = -(PW / 2); dx < PW / 2; ++dx) {
+    int x = px + dx;
+    if (x < 0) x = 0;
+    if (x > W - 1) x = W - 1;
+    obs[PY * W + x] = 1.0f;
+    obs[(PY + 1) * W + x] = 1.0f;
+  }
+  obs[(PY - 1) * W + (px < 0 ? 0 : (px > W - 1 ? W - 1 : px))] = 1.0f;
+  if (s[4] > 0.0f) {  // shot, 1x2
+    int x = (int)s[2], y = (int)s[3];
+    if (x >= 0 && x < W && y >= 1 && y < H) {
+      obs[y * W + x] = 0.9f;
+      obs[(y - 1) * W + x] = 0.9f;
+    }
+  }
+  if (s[7] > 0.0f) {  // bomb, 2x2
+    int x = (int)s[5], y = (int)s[6];
+    if (x >= 0 && x < W - 1 && y >= 0 && y < H - 1)
+      for (int dy = 0; dy < 2; ++dy)
+        for (int dx = 0; dx < 2; ++dx) obs[(y + dy) * W + x + dx] = 0.8f;
+  }
+}
+
+void spaceinv_cpu_reset(torch::Tensor state, torch::Tensor obs,
+                        int64_t seed, int64_t draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  float* op = obs.data_ptr<float>();
+  for (int64_t b = 0; b < B; ++b) {
+    reset_env(sp + b * SDIM, (uint64_t)seed, (uint64_t)b, (uint64_t)draw);
+    render(sp + b * SDIM, op + b * H * W);
+  }
+}
+
+void spaceinv_cpu_step(torch::Tensor state, torch::Tensor action,
+                       torch::Tensor step_count, torch::Tensor ep_return,
+                       torch::Tensor ep_length, torch::Tensor last_ep_return,
+                       torch::Tensor last_ep_length, torch::Tensor obs,
+                       torch::Tensor next_obs, torch::Tensor reward,
+                       torch::Tensor discount, torch::Tensor steptype,
+                       torch::Tensor done, int64_t max_episode_steps,
+                       int64_t seed, torch::Tensor draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  const int64_t* ap = action.data_ptr<int64_t>();
+  int32_t* scp = step_count.data_ptr<int32_t>();
+  float* erp = ep_return.data_ptr<float>();
+  int32_t* elp = ep_length.data_ptr<int32_t>();
+  float* lerp = last_ep_return.data_ptr<float>();
+  int32_t* lelp = last_ep_length.data_ptr<int32_t>();
+  float* op = obs.data_ptr<float>();
+  float* nop = next_obs.data_ptr<float>();
+  float* rp = reward.data_ptr<float>();
+  float* dp = discount.data_ptr<float>();
+  uint8_t* stp = steptype.data_ptr<uint8_t>();
+  uint8_t* dnp = done.data_ptr<uint8_t>();
+  int32_t dr = *draw.data_ptr<int32_t>();
+  for (int64_t b = 0; b < B; ++b) {
+    float* s = sp + b * SDIM;
+    int a = (int)ap[b];
+    if (a < 0) a = 0;
+    if (a > 3) a = 3;
+    float px = s[0] + PSPEED * ((a == 2) - (a == 1));
+    if (px < PW / 2.0f) px = PW / 2.0f;
+    if (px > W - PW / 2.0f) px = W - PW / 2.0f;
+    float cool = s[1] > 0.0f ? s[1] - 1.0f : 0.0f;
+    float rew = 0.0f;
+    // fire
+    if (a == 3 && s[4] <= 0.0f && cool <= 0.0f) {
+      s[2] = px; s[3] = PY - 2.0f; s[4] = 1.0f; cool = 6.0f;
+    }
+    // alien march
+    float adir = s[8], aoffx = s[9], aoffy = s[10];
+    aoffx += adir * 0.5f;
+    float span = AC * 9.0f;
+    if (aoffx < 2.0f || aoffx + span > W - 2.0f) { adir = -adir; aoffy += 3.0f; }
+    // shot flight + alien hit
+    if (s[4] > 0.0f) {
+      s[3] -= SHOT_V;
+      if (s[3] < 1.0f) s[4] = 0.0f;
+      int c = (int)((s[2] - aoffx) / 9.0f);
+      int r = (int)((s[3] - aoffy) / 7.0f);
+      if (s[4] > 0.0f && r >= 0 && r < AR && c >= 0 && c < AC) {
+        float rel_x = s[2] - (aoffx + c * 9.0f);
+        float rel_y = s[3] - (aoffy + r * 7.0f);
+        if (rel_x >= 0 && rel_x < 5.0f && rel_y >= 0 && rel_y < 3.0f &&
+            s[SOFF + r * AC + c] > 0.0f) {
+          s[SOFF + r * AC + c] = 0.0f;
+          s[4] = 0.0f;
+          rew += 1.0f;
+        }
+      }
+    }
+    // bombs
+    bool player_hit = false;
+    if (s[7] > 0.0f) {
+      s[6] += BOMB_V;
+      if (s[6] >= PY - 1.0f && std::fabs(s[5] - px) <= PW / 2.0f) {
+        player_hit = true;
+        s[7] = 0.0f;
+      } else if (s[6] > H - 2.0f) {
+        s[7] = 0.0f;
+      }
+    } else if (hash_uniform((uint64_t)seed, (uint64_t)b * 31 + 7,
+                            (uint64_t)(dr + scp[b])) < 0.06f) {
+      // drop from a random alive alien
+      int pick = (int)(hash_uniform((uint64_t)seed, (uint64_t)b * 31 + 8,
+                                    (uint64_t)(dr + scp[b])) * NA);
+      for (int k = 0; k < NA; ++k) {
+        int idx = (pick + k) % NA;
+        if (s[SOFF + idx] > 0.0f) {
+          int r = idx / AC, c = idx % AC;
+          s[5] = aoffx + c * 9.0f + 2.0f;
+          s[6] = aoffy + r * 7.0f + 3.0f;
+          s[7] = 1.0f;
+          break;
+        }
+      }
+    }
+    float alive = 0.0f;
+    for (int k = 0; k < NA; ++k) alive += s[SOFF + k];
+    bool landed = aoffy + AR * 7.0f >= PY - 2.0f;
+    bool cleared = alive <= 0.0f;
+    if (cleared) rew += 5.0f;
+    bool terminated = player_hit || landed || cleared;
+    s[0] = px; s[1] = cool; s[8] = adir; s[9] = aoffx; s[10] = aoffy;
+
+    int sc = scp[b] + 1;
+    bool truncated = (sc >= max_episode_steps) && !terminated;
+    bool dn = terminated || truncated;
+    float ret = erp[b] + rew;
+    int len = elp[b] + 1;
+    if (dn) { lerp[b] = ret; lelp[b] = len; }
+    render(s, nop + b * H * W);
+    if (dn) {
+      reset_env(s, (uint64_t)seed, (uint64_t)b * 977 + 13, (uint64_t)dr);
+      sc = 0; ret = 0.0f; len = 0;
+      render(s, op + b * H * W);
+    } else {
+      std::memcpy(op + b * H * W, nop + b * H * W, sizeof(float) * H * W);
+    }
+    scp[b] = sc; erp[b] = ret; elp[b] = len;
+    rp[b] = rew;
+    dp[b] = terminated ? 0.0f : 1.0f;
+    stp[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+    dnp[b] = dn ? 1 : 0;
+  }
+  *draw.data_ptr<int32_t>() = dr + 1;
+}
+}  // namespace spaceinv
+
+// --------------------------------------------------------- Qbert-class
+// Triangular pyramid of cubes; diagonal hops colour cubes (+1 first
+// visit); colouring every cube pays +5 and terminates; hopping off the
+// pyramid or colliding with the bouncing ball ends the episode.
+namespace qbert {
+constexpr int ROWS = 7, NCUBE = ROWS * (ROWS + 1) / 2;
+// state: agent_r, agent_c, ball_r, ball_c, ball_alive, ball_timer,
+//        colored[NCUBE]
+constexpr int SOFF = 6;
+constexpr int SDIM = SOFF + NCUBE;
+
+inline int cube_index(int r, int c) { return r * (r + 1) / 2 + c; }
+
+inline void reset_env(float* s, uint64_t seed, uint64_t env, uint64_t draw) {
+  (void)seed; (void)env; (void)draw;
+  s[0] = 0.0f; s[1] = 0.0f;
+  s[2] = 0.0f; s[3] = 0.0f; s[4] = 0.0f;
+  s[5] = 10.0f;
+  for (int k = 0; k < NCUBE; ++k) s[SOFF + k] = 0.0f;
+  s[SOFF + 0] = 1.0f;  // start cube coloured
+}
+
+inline void render(const float* s, float* obs) {
+  std::memset(obs, 0, sizeof(float) * H * W);
+  for (int r = 0; r < ROWS; ++r)
+    for (int c = 0; c <= r; ++c) {
+      int cx = W / 2 + (2 * c - r) * 6;
+      int cy = 8 + r * 10;
+      float v = s[SOFF + cube_index(r, c)] > 0.0f ? 0.8f : 0.35f;
+      for (int y = cy; y < cy + 7; ++y)
+        for (int x = cx - 4; x < cx + 4; ++x)
+          if (y >= 0 && y < H && x >= 0 && x < W) obs[y * W + x] = v;
+    }
+  int ar = (int)s[0], ac = (int)s[1];
+  int axc = W / 2 + (2 * ac - ar) * 6, ayc = 5 + ar * 10;
+  for (int y = ayc; y < ayc + 4; ++y)
+    for (int x = axc - 2; x < axc + 2; ++x)
+      if (y >= 0 && y < H && x >= 0 && x < W) obs[y * W + x] = 1.0f;
+  if (s[4] > 0.0f) {
+    int br = (int)s[2], bc = (int)s[3];
+    int bxc = W / 2 + (2 * bc - br) * 6, byc = 5 + br * 10;
+    for (int y = byc; y < byc + 3; ++y)
+      for (int x = bxc - 2; x < bxc + 1; ++x)
+        if (y >= 0 && y < H && x >= 0 && x < W) obs[y * W + x] = 0.55f;
+  }
+}
+
+void qbert_cpu_reset(torch::Tensor state, torch::Tensor obs, int64_t seed,
+                     int64_t draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  float* op = obs.data_ptr<float>();
+  for (int64_t b = 0; b < B; ++b) {
+    reset_env(sp + b * SDIM, (uint64_t)seed, (uint64_t)b, (uint64_t)draw);
+    render(sp + b * SDIM, op + b * H * W);
+  }
+}
+
+void qbert_cpu_step(torch::Tensor state, torch::Tensor action,
+                    torch::Tensor step_count, torch::Tensor ep_return,
+                    torch::Tensor ep_length, torch::Tensor last_ep_return,
+                    torch::Tensor last_ep_length, torch::Tensor obs,
+                    torch::Tensor next_obs, torch::Tensor reward,
+                    torch::Tensor discount, torch::Tensor steptype,
+                    torch::Tensor done, int64_t max_episode_steps,
+                    int64_t seed, torch::Tensor draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  const int64_t* ap = action.data_ptr<int64_t>();
+  int32_t* scp = step_count.data_ptr<int32_t>();
+  float* erp = ep_return.data_ptr<float>();
+  int32_t* elp = ep_length.data_ptr<int32_t>();
+  float* lerp = last_ep_return.data_ptr<float>();
+  int32_t* lelp = last_ep_length.data_ptr<int32_t>();
+  float* op = obs.data_ptr<float>();
+  float* nop = next_obs.data_ptr<float>();
+  float* rp = reward.data_ptr<float>();
+  float* dp = discount.data_ptr<float>();
+  uint8_t* stp = steptype.data_ptr<uint8_t>();
+  uint8_t* dnp = done.data_ptr<uint8_t>();
+  int32_t dr = *draw.data_ptr<int32_t>();
+  for (int64_t b = 0; b < B; ++b) {
+    float* s = sp + b * SDIM;
+    int a = (int)ap[b];
+    if (a < 0) a = 0;
+    if (a > 3) a = 3;
+    int r = (int)s[0], c = (int)s[1];
+    // hops: 0 down-left, 1 down-right, 2 up-left, 3 up-right
+    int nr = r + ((a < 2) ? 1 : -1);
+    int nc = c + ((a == 1) ? 1 : 0) - ((a == 2) ? 1 : 0);
+    float rew = 0.0f;
+    bool fell = nr < 0 || nr >= ROWS || nc < 0 || nc > nr;
+    bool cleared = false;
+    if (!fell) {
+      int k = cube_index(nr, nc);
+      if (s[SOFF + k] <= 0.0f) {
+        s[SOFF + k] = 1.0f;
+        rew += 1.0f;
+      }
+      float colored = 0.0f;
+      for (int q = 0; q < NCUBE; ++q) colored += s[SOFF + q];
+      if (colored >= (float)NCUBE) { cleared = true; rew += 5.0f; }
+      s[0] = (float)nr; s[1] = (float)nc;
+    }
+    // ball: spawns at the top after a countdown, hops down randomly
+    bool caught = false;
+    if (!fell && !cleared) {
+      if (s[4] <= 0.0f) {
+        s[5] -= 1.0f;
+        if (s[5] <= 0.0f) { s[2] = 0.0f; s[3] = 0.0f; s[4] = 1.0f; }
+      } else {
+        int br = (int)s[2], bc = (int)s[3];
+        int bnr = br + 1;
+        int bnc = bc + ((hash_uniform((uint64_t)seed, (uint64_t)b * 53 + 5,
+                                      (uint64_t)(dr + scp[b])) > 0.5f) ? 1 : 0);
+        if (bnc > bnr) bnc = bnr;
+        if (bnr >= ROWS) {
+          s[4] = 0.0f;
+          s[5] = 8.0f + 8.0f * hash_uniform((uint64_t)seed,
+                                            (uint64_t)b * 53 + 6,
+                                            (uint64_t)(dr + scp[b]));
+        } else {
+          s[2] = (float)bnr; s[3] = (float)bnc;
+        }
+      }
+      caught = s[4] > 0.0f && (int)s[0] == (int)s[2] && (int)s[1] == (int)s[3];
+    }
+    bool terminated = fell || cleared || caught;
+
+    int sc = scp[b] + 1;
+    bool truncated = (sc >= max_episode_steps) && !terminated;
+    bool dn = terminated || truncated;
+    float ret = erp[b] + rew;
+    int len = elp[b] + 1;
+    if (dn) { lerp[b] = ret; lelp[b] = len; }
+    render(s, nop + b * H * W);
+    if (dn) {
+      reset_env(s, (uint64_t)seed, (uint64_t)b * 977 + 13, (uint64_t)dr);
+      sc = 0; ret = 0.0f; len = 0;
+      render(s, op + b * H * W);
+    } else {
+      std::memcpy(op + b * H * W, nop + b * H * W, sizeof(float) * H * W);
+    }
+    scp[b] = sc; erp[b] = ret; elp[b] = len;
+    rp[b] = rew;
+    dp[b] = terminated ? 0.0f : 1.0f;
+    stp[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+    dnp[b] = dn ? 1 : 0;
+  }
+  *draw.data_ptr<int32_t>() = dr + 1;
+}
+}  // namespace qbert
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // gil_scoped_release: the whole fused step runs WITHOUT the GIL so
   // Sebulba's learner thread and sibling actor threads keep running
@@ -396,4 +739,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched Breakout fused step: physics + metrics + autoreset + render",
         py::call_guard<py::gil_scoped_release>());
   m.attr("STATE_DIM") = SDIM;
+  m.def("spaceinv_reset", &spaceinv::spaceinv_cpu_reset,
+        "batched SpaceInvaders-class reset + render",
+        py::call_guard<py::gil_scoped_release>());
+  m.def("spaceinv_step", &spaceinv::spaceinv_cpu_step,
+        "batched SpaceInvaders-class fused step",
+        py::call_guard<py::gil_scoped_release>());
+  m.attr("SPACEINV_STATE_DIM") = spaceinv::SDIM;
+  m.def("qbert_reset", &qbert::qbert_cpu_reset,
+        "batched Qbert-class reset + render",
+        py::call_guard<py::gil_scoped_release>());
+  m.def("qbert_step", &qbert::qbert_cpu_step, "batched Qbert-class fused step",
+        py::call_guard<py::gil_scoped_release>());
+  m.attr("QBERT_STATE_DIM") = qbert::SDIM;
 }

@@ -133,6 +133,29 @@ class PongCpu(_PoolEnvBase):
     N_ACTIONS = 3  # noop / up / down
 
 
+class SpaceInvadersCpu(_PoolEnvBase):
+    """Native-engine SpaceInvaders-class: marching alien grid, one shot
+    at a time, alien bombs; +1 per alien, +5 for clearing the wave."""
+
+    max_episode_steps = 3000
+    STATE_DIM_ATTR = "SPACEINV_STATE_DIM"
+    RESET_FN = "spaceinv_reset"
+    STEP_FN = "spaceinv_step"
+    N_ACTIONS = 4  # noop / left / right / fire
+
+
+class QbertCpu(_PoolEnvBase):
+    """Native-engine Qbert-class: diagonal hops colour the pyramid's
+    cubes (+1 first visit, +5 clear); falling off or the bouncing ball
+    ends the episode."""
+
+    max_episode_steps = 2000
+    STATE_DIM_ATTR = "QBERT_STATE_DIM"
+    RESET_FN = "qbert_reset"
+    STEP_FN = "qbert_step"
+    N_ACTIONS = 4  # down-left / down-right / up-left / up-right
+
+
 class BreakoutCpu(_PoolEnvBase):
     """Native-engine Breakout; drop-in for envs/breakout.py on CPU."""
 

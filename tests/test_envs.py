@@ -872,3 +872,64 @@ def test_crafting_stone_needs_pickaxe():
     ts = env.step(torch.full((1,), 4, dtype=torch.long))
     assert torch.all(ts.reward == 0.0)
     assert torch.all(env._state["grid"][:, pos + 1] == 2.0)  # still there
+
+
+# -------------------------------------------- native pool: new game rules
+
+
+def _pool_available():
+    from stoix_amd.envs.envpool_cpu import envpool_ext
+
+    return envpool_ext() is not None
+
+
+def test_spaceinvaders_pool_rules():
+    import pytest as _pytest
+    import torch
+
+    if not _pool_available():
+        _pytest.skip("native pool ext not built")
+    from stoix_amd.envs.envpool_cpu import SpaceInvadersCpu
+
+    env = SpaceInvadersCpu(8, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (8, 84, 84, 1)
+    assert float(ts.observation.max()) == 1.0  # cannon rendered
+    # fire straight up until something happens; aliens above the cannon
+    # column should eventually be hit (+1) or a bomb ends the episode
+    got_reward = False
+    for _ in range(400):
+        ts = env.step(torch.full((8,), 3, dtype=torch.long))
+        if float(ts.reward.max()) >= 1.0:
+            got_reward = True
+            break
+    assert got_reward, "firing never hit an alien"
+    # contract invariants
+    assert set(ts.step_type.unique().tolist()) <= {1, 2, 3}
+    assert torch.all((ts.discount == 0) | (ts.discount == 1))
+
+
+def test_qbert_pool_rules():
+    import pytest as _pytest
+    import torch
+
+    if not _pool_available():
+        _pytest.skip("native pool ext not built")
+    from stoix_amd.envs.envpool_cpu import QbertCpu
+
+    env = QbertCpu(4, seed=0)
+    ts = env.reset()
+    # hop down-left onto an uncoloured cube: +1
+    ts = env.step(torch.zeros(4, dtype=torch.long))
+    assert torch.all(ts.reward == 1.0)
+    assert torch.all(ts.discount == 1.0)
+    # hop up-right back to the start cube (already coloured): no reward
+    ts = env.step(torch.full((4,), 3, dtype=torch.long))
+    assert torch.all(ts.reward == 0.0)
+    # hop up-right AGAIN: off the pyramid -> terminated, autoreset
+    ts = env.step(torch.full((4,), 3, dtype=torch.long))
+    assert torch.all(ts.reward == 0.0)
+    assert torch.all(ts.discount == 0.0)
+    em = ts.extras["episode_metrics"]
+    assert torch.all(em["is_terminal_step"])
+    assert torch.all(em["episode_return"] == 1.0)
