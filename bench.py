@@ -309,7 +309,7 @@ def bench_sebulba(args, ctx) -> None:
                     "parallelism": f"{actors}-actor-threads+1-learner",
                     "env": "envpool/breakout",
                 },
-                "note": "wall-clock includes thread startup + eval (whole experiment)",
+                "note": "train-phase SPS (MIOpen prewarm + async-eval drain excluded)",
             }
         )
     )
