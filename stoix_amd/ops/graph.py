@@ -68,8 +68,12 @@ def try_enable_graphs(learner) -> bool:
         return False
     if os.environ.get("STOIX_NO_GRAPH"):
         return False
-    if getattr(learner.env, "_hip", None) is None:
-        raise RuntimeError("env has no HIP step kernel; graph capture needs it")
+    env = learner.env
+    if getattr(env, "_hip", None) is None and not getattr(env, "capture_safe", False):
+        raise RuntimeError(
+            "env has no HIP step kernel and no capture_safe torch step; "
+            "graph capture needs one"
+        )
 
     learner.prepare_for_graph_capture()
     _warm_collectives(device)
