@@ -27,6 +27,10 @@ def bench_rec_ppo(B=1024, T=16, steps=20, warmup=3):
     dev = torch.device("cuda:0" if torch.cuda.is_available() else "cpu")
     env = environments.make_single(cfg, B, dev, seed=0)
     learner = RecPPOLearner(cfg, env, dev)
+    if dev.type == "cuda":
+        from stoix_amd.ops.graph import try_enable_graphs
+
+        try_enable_graphs(learner)
     for _ in range(warmup):
         learner.update_step()
     if dev.type == "cuda":
