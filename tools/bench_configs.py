@@ -1,4 +1,10 @@
-"""Measure the non-flagship BASELINE.json configs (steps/sec) on one GPU.
+"""SUPERSEDED (round 2): every BASELINE config is now driver-runnable
+through the repo-root bench.py contract —
+    python bench.py --config {cartpole,ppo,sac,sebulba,rainbow}
+(one JSON line each, warmup + barrier-bracketed timing).
+This tool is kept for the round-1 measurement provenance.
+
+Measure the non-flagship BASELINE.json configs (steps/sec) on one GPU.
 
 The driver's bench.py covers config #2 (Anakin PPO / Ant). This tool times
 configs #1, #3, #4, #5 with the same steps_per_second definition
