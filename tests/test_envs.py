@@ -933,3 +933,50 @@ def test_qbert_pool_rules():
     em = ts.extras["episode_metrics"]
     assert torch.all(em["is_terminal_step"])
     assert torch.all(em["episode_return"] == 1.0)
+
+
+# ------------------------------------------------ popjym-class POMDP envs
+
+
+def test_stateless_cartpole_masks_velocities():
+    """The POMDP wrapper hides velocity, appends the start flag and the
+    previous-action one-hot (reference AddStartFlagAndPrevAction
+    semantics), and preserves the inner dynamics."""
+    import torch
+
+    from stoix_amd.envs.classic import CartPole
+    from stoix_amd.envs.pomdp import StatelessCartPole
+
+    env = StatelessCartPole(8, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (8, 2 + 1 + 2)
+    assert torch.all(ts.observation[:, 2] == 1.0)  # start flag set
+    assert torch.all(ts.observation[:, 3] == 1.0)  # prev action one-hot(0)
+    # full env from identical state must agree on the visible components
+    full = CartPole(8, seed=0)
+    full._hip = None
+    full.reset()
+    full._state = {"s": env._state["s"].clone()}
+    a = torch.ones(8, dtype=torch.long)
+    ts_m = env.step(a)
+    ts_f = full.step(a)
+    assert torch.all(ts_m.observation[:, 2] == 0.0)  # start flag cleared
+    assert torch.all(ts_m.observation[:, 4] == 1.0)  # prev action = 1
+    torch.testing.assert_close(ts_m.observation[:, 0], ts_f.observation[:, 0])
+    torch.testing.assert_close(ts_m.observation[:, 1], ts_f.observation[:, 2])
+    torch.testing.assert_close(ts_m.reward, ts_f.reward)
+    torch.testing.assert_close(ts_m.discount, ts_f.discount)
+
+
+def test_stateless_pendulum_continuous_prev_action():
+    import torch
+
+    from stoix_amd.envs.pomdp import StatelessPendulum
+
+    env = StatelessPendulum(4, seed=0)
+    ts = env.reset()
+    adim = env.action_space.shape[0]
+    assert ts.observation.shape == (4, 2 + 1 + adim)
+    a = torch.full((4, adim), 0.7)
+    ts = env.step(a)
+    torch.testing.assert_close(ts.observation[:, -adim:], a)
