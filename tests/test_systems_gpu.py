@@ -92,3 +92,63 @@ def test_gpu_sebulba_ppo_breakout():
     )
     r = mod.run(cfg)
     assert r == r
+
+
+@requires_gpu
+def test_gpu_ppo_xland_goal_grid():
+    """Round-2 env family on device: goal-conditioned gridworld steps as
+    CUDA tensors under PPO."""
+    _run("stoix_amd.systems.ppo.ff_ppo", "default/anakin/default_ff_ppo.yaml",
+         ["env=xland_minigrid/goal_grid", "arch.total_num_envs=256",
+          "system.rollout_length=8", "system.num_minibatches=2",
+          "system.epochs=1"])
+
+
+@requires_gpu
+def test_gpu_ppo_crafting():
+    _run("stoix_amd.systems.ppo.ff_ppo", "default/anakin/default_ff_ppo.yaml",
+         ["env=craftax/crafting", "arch.total_num_envs=256",
+          "system.rollout_length=8", "system.num_minibatches=2",
+          "system.epochs=1"])
+
+
+@requires_gpu
+def test_gpu_rec_ppo_pomdp():
+    _run("stoix_amd.systems.ppo.rec_ppo", "default/anakin/default_rec_ppo.yaml",
+         ["env=popjym/stateless_cartpole", "arch.total_num_envs=128",
+          "system.rollout_length=8", "system.num_minibatches=2",
+          "system.epochs=1"])
+
+
+@requires_gpu
+def test_gpu_dqn_capture_on_snake_capture_safe_env():
+    """The capture_safe torch-env path (no HIP kernel) must graph-capture a
+    whole off-policy update end to end (widened gate, round 2)."""
+    import torch as _t
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.ops.graph import try_enable_update_graph
+    from stoix_amd.systems.q_learning.ff_dqn import DQNLearner
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_dqn.yaml",
+        ["env=jumanji/snake", "arch.total_num_envs=64",
+         "arch.total_timesteps=null", "arch.num_updates=4",
+         "arch.num_evaluation=1", "system.rollout_length=2",
+         "system.batch_size=64", "system.buffer_size=4096",
+         "system.warmup_steps=16", "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    dev = _t.device("cuda:0")
+    env = environments.make_single(cfg, 64, dev, seed=0)
+    learner = DQNLearner(cfg, env, dev)
+    assert learner.graph_capturable
+    ok = try_enable_update_graph(learner)
+    assert ok
+    for _ in range(3):
+        m = learner.update_step()
+    _t.cuda.synchronize()
+    assert _t.isfinite(m["q_loss"]).all()
