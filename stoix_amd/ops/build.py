@@ -23,6 +23,7 @@ SOURCES = [
     str(CSRC / "wgrad.hip"),
     str(CSRC / "per.hip"),
     str(CSRC / "rnn.hip"),
+    str(CSRC / "snake.hip"),
 ]
 
 

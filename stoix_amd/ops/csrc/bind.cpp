@@ -107,6 +107,11 @@ void launch_ppo_gather_disc(const long*, int, const float*, int, int,
 void launch_rnn_scan(const float*, const void*, const float*,
                      const unsigned char*, const float*, const float*,
                      float*, float*, float*, int, int, int, int, void*);
+void launch_snake_step(int*, long*, long*, long*, long*, int*, const long*,
+                       int*, float*, int*, float*, int*, float*, float*,
+                       float*, float*, unsigned char*, unsigned char*, int,
+                       int, uint64_t, unsigned int*, unsigned int, int,
+                       void*);
 }
 
 namespace {
@@ -623,7 +628,44 @@ void rnn_scan(torch::Tensor Xp, torch::Tensor Whh, torch::Tensor bhh,
       cur_stream());
 }
 
+void snake_step(torch::Tensor grid, torch::Tensor head_r,
+                torch::Tensor head_c, torch::Tensor fruit_r,
+                torch::Tensor fruit_c, torch::Tensor length,
+                torch::Tensor action, torch::Tensor step_count,
+                torch::Tensor ep_return, torch::Tensor ep_length,
+                torch::Tensor last_ep_return, torch::Tensor last_ep_length,
+                torch::Tensor obs_out, torch::Tensor next_obs_out,
+                torch::Tensor reward_out, torch::Tensor discount_out,
+                torch::Tensor steptype_out, torch::Tensor done_out,
+                int64_t max_episode_steps, int64_t seed,
+                torch::Tensor draw_buf, int64_t draw_offset,
+                int64_t do_bump) {
+  CHK(grid, torch::kInt32);
+  CHK(head_r, torch::kInt64);
+  CHK(action, torch::kInt64);
+  CHK(length, torch::kInt32);
+  int B = grid.size(0);
+  unsigned int* db = draw_buf.numel() > 0
+                         ? (unsigned int*)draw_buf.data_ptr<int>()
+                         : nullptr;
+  launch_snake_step(
+      grid.data_ptr<int>(), head_r.data_ptr<long>(), head_c.data_ptr<long>(),
+      fruit_r.data_ptr<long>(), fruit_c.data_ptr<long>(),
+      length.data_ptr<int>(), action.data_ptr<long>(),
+      step_count.data_ptr<int>(), ep_return.data_ptr<float>(),
+      ep_length.data_ptr<int>(), last_ep_return.data_ptr<float>(),
+      last_ep_length.data_ptr<int>(), obs_out.data_ptr<float>(),
+      next_obs_out.data_ptr<float>(), reward_out.data_ptr<float>(),
+      discount_out.data_ptr<float>(),
+      steptype_out.data_ptr<unsigned char>(),
+      done_out.data_ptr<unsigned char>(), B, (int)max_episode_steps,
+      (uint64_t)seed, db, (unsigned int)draw_offset, (int)do_bump,
+      cur_stream());
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("snake_step", &snake_step,
+        "fused Snake env step (dynamics + metrics + autoreset + render)");
   m.def("rnn_scan", &rnn_scan,
         "fused done-masked GRU/LSTM sequence scan (K13)");
   m.def("policy_value_step_disc", &policy_value_step_disc,
