@@ -15,7 +15,7 @@ per-level repair with strided tensor ops (CPU tests + numerics parity).
 """
 from __future__ import annotations
 
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 

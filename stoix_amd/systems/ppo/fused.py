@@ -1,4 +1,5 @@
-"""Fused MI355X update path for Anakin PPO (continuous, canonical MLP).
+"""Fused MI355X update path for Anakin PPO (continuous tanh-normal AND
+discrete categorical heads on the canonical MLP torsos).
 
 The eager path (ff_ppo.py) runs ~45 kernels per rollout step and ~150 per
 minibatch even under hip-graph replay — execution time is dominated by

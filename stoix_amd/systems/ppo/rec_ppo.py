@@ -8,7 +8,7 @@ bootstrap from the value trace's final value (:165-176).
 from __future__ import annotations
 
 import sys
-from typing import Any, Dict, List
+from typing import Any, Dict
 
 import torch
 import torch.nn as nn

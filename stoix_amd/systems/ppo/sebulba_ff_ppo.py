@@ -31,7 +31,6 @@ from stoix_amd.networks.factory import build_actor, build_critic
 from stoix_amd.ops import multistep
 from stoix_amd.ops.losses import clipped_value_loss, ppo_clip_loss
 from stoix_amd.parallel.dist import FlatGradReducer, broadcast_module, get_dist_context
-from stoix_amd.utils.checkpointing import Checkpointer
 from stoix_amd.utils.logger import LogEvent, StoixLogger
 from stoix_amd.utils.sebulba import (
     AsyncEvaluator,

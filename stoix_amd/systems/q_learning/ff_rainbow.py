@@ -18,7 +18,6 @@ import torch.nn.functional as F
 from stoix_amd.buffers import PrioritisedBuffer
 from stoix_amd.config import compose
 from stoix_amd.envs.env import StatefulVecEnv, get_final_step_metrics
-from stoix_amd.networks.distributions import EpsilonGreedy
 from stoix_amd.networks.dueling import DistributionalDuelingQNetwork
 from stoix_amd.networks.layers import NoiseBank, set_noise_enabled
 from stoix_amd.ops.losses import categorical_l2_project
