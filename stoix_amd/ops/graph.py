@@ -68,6 +68,11 @@ def try_enable_graphs(learner) -> bool:
         return False
     if os.environ.get("STOIX_NO_GRAPH"):
         return False
+    if getattr(learner, "lr_decay", None) is not None:
+        # a python-float lr would be frozen into the captured optimiser
+        # step; the linear-decay path stays eager (reference parity:
+        # utils/training.py decay is opt-in and off in every shipped config)
+        return False
     env = learner.env
     if getattr(env, "_hip", None) is None and not getattr(env, "capture_safe", False):
         raise RuntimeError(
@@ -145,6 +150,8 @@ def try_enable_update_graph(learner) -> bool:
     if _world_size() > 1 and os.environ.get("STOIX_FUSED_EAGER_ALLREDUCE") == "1":
         # whole-update capture would put the RCCL all-reduce inside the
         # graph; the knob demands eager collectives, so skip capture
+        return False
+    if getattr(learner, "lr_decay", None) is not None:
         return False
     env = learner.env
     if getattr(env, "_hip", None) is None and not getattr(env, "capture_safe", False):

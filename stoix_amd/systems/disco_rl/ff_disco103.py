@@ -115,6 +115,9 @@ class DiscoLearner:
             p.requires_grad_(False)
 
         self.opt = torch.optim.Adam(self.net.parameters(), lr=float(self.sys.lr), eps=1e-5)
+        from stoix_amd.utils.training import maybe_lr_decay
+
+        self.lr_decay = maybe_lr_decay(config, self.opt)
         self.reducer = FlatGradReducer(self.net.parameters(), device)
         self.gen = torch.Generator(device=device)
         self.gen.manual_seed(int(config.arch.seed) * 7919 + 1031)
@@ -214,6 +217,8 @@ class DiscoLearner:
                 with torch.no_grad():
                     polyak_update(self.net.parameters(), self.target_net.parameters(),
                                   1.0 - self.target_coeff)
+        if self.lr_decay is not None:
+            self.lr_decay.step()
         return metrics
 
     def _minibatch_loss(self, obs, act, rew, done, beh: DiscoAgentOutput):

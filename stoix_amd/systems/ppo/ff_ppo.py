@@ -103,11 +103,16 @@ class PPOLearner:
         self.buf_targets = z(self.T, self.B)
         self.perm_buf = torch.arange(self.T * self.B, device=device)
 
+        # linear LR decay by update count (reference utils/training.py:6-49)
+        from stoix_amd.utils.training import maybe_lr_decay
+
+        self.lr_decay = maybe_lr_decay(config, self.actor_opt, self.critic_opt)
+
         # fused MI355X path (hand-written MFMA/HIP kernels; see fused.py):
         # replaces the eager rollout + minibatch update when the network is
         # the canonical MLP shape. system.fused=false forces eager.
         self.fused = None
-        if bool(getattr(self.sys, "fused", True)) and use_bf16:
+        if bool(getattr(self.sys, "fused", True)) and use_bf16 and self.lr_decay is None:
             from stoix_amd.systems.ppo.fused import FusedPPOEngine
 
             self.fused = FusedPPOEngine.try_build(self)
@@ -190,6 +195,8 @@ class PPOLearner:
         for _ in range(int(self.sys.epochs)):
             self._new_perm()
             metrics = self.epoch_phase()
+        if self.lr_decay is not None:
+            self.lr_decay.step()
         return metrics
 
     def rollout_phase(self) -> None:

@@ -78,6 +78,9 @@ class RecPPOLearner:
         )
         self.gen = torch.Generator(device=device)
         self.gen.manual_seed(int(config.arch.seed) * 7919 + 601)
+        from stoix_amd.utils.training import maybe_lr_decay
+
+        self.lr_decay = maybe_lr_decay(config, self.actor_opt, self.critic_opt)
 
         self.ts = env.reset()
         # STABLE hidden-state / carry buffers (hip-graph capturable: the
@@ -125,6 +128,8 @@ class RecPPOLearner:
         for _ in range(int(self.sys.epochs)):
             self._new_perm()
             metrics = self.epoch_phase()
+        if self.lr_decay is not None:
+            self.lr_decay.step()
         return metrics
 
     @torch.no_grad()

@@ -69,3 +69,41 @@ def test_sweep_tpe_beats_random_on_quadratic():
     assert t is not None and r is not None
     assert t.value >= r.value - 1e-9
     assert t.value > -0.05  # near the optimum (max 0.5)
+
+
+def test_linear_lr_decay_matches_reference_schedule():
+    """lr(update) = init * (1 - update/num_updates), per param group
+    (reference utils/training.py:24-28 at epoch*minibatch granularity
+    collapsed to update granularity)."""
+    import torch
+
+    from stoix_amd.utils.training import LinearLRDecay
+
+    p1 = torch.nn.Parameter(torch.zeros(3))
+    p2 = torch.nn.Parameter(torch.zeros(3))
+    o1 = torch.optim.Adam([p1], lr=1e-3)
+    o2 = torch.optim.Adam([p2], lr=5e-4)
+    dec = LinearLRDecay([o1, o2], num_updates=10)
+    for u in range(1, 11):
+        dec.step()
+        assert abs(o1.param_groups[0]["lr"] - 1e-3 * (1 - u / 10)) < 1e-12
+        assert abs(o2.param_groups[0]["lr"] - 5e-4 * (1 - u / 10)) < 1e-12
+    dec.step()  # past the horizon: clamps at 0
+    assert o1.param_groups[0]["lr"] == 0.0
+
+
+def test_ppo_with_lr_decay_runs():
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8",
+         "arch.total_timesteps=null", "arch.num_updates=3",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+         "system.rollout_length=8", "system.num_minibatches=2",
+         "system.epochs=1", "system.decay_learning_rates=true",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r == r
