@@ -292,3 +292,43 @@ def test_running_statistics_update_inplace_matches_functional():
     torch.testing.assert_close(s1.mean, s2.mean)
     torch.testing.assert_close(s1.std, s2.std)
     torch.testing.assert_close(s1.count, s2.count)
+
+
+def test_all_network_presets_build():
+    """Every shipped network preset yaml builds an actor (and runs a
+    forward) against an appropriate space — mirrors the reference's config
+    surface (configs/network/*)."""
+    import os
+
+    import torch
+
+    from stoix_amd.config import CONFIG_ROOT, compose
+    from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
+    from stoix_amd.networks.factory import build_actor
+
+    pixel = {"cnn", "visual_resnet"}
+    skip = {"world_model", "mlp_rainbow", "mlp_dueling_dqn"}  # built by their systems directly
+    rnn = {"rnn", "rnn_dqn", "rnn_dueling_dqn"}
+    cont = {"mlp_continuous", "mlp_sac", "mlp_ddpg", "mlp_d4pg", "mlp_mpo_continuous"}
+    for f in sorted(os.listdir(CONFIG_ROOT / "network")):
+        name = f[:-5]
+        if name in skip:
+            continue
+        cfg = compose("default/anakin/default_ff_ppo.yaml", [f"network={name}"])
+        obs = BoxSpace((84, 84, 1) if name in pixel else (8,), 0.0, 1.0)
+        if name in cont:
+            act = BoxSpace((3,), -1.0, 1.0)
+        else:
+            act = DiscreteSpace(4)
+        if name in rnn:
+            from stoix_amd.networks.factory import build_recurrent_actor
+
+            a = build_recurrent_actor(cfg.network.actor_network, obs, act)
+            st = a.initial_state(2, "cpu")
+            x = torch.zeros(3, 2, *obs.shape)
+            resets = torch.zeros(3, 2, dtype=torch.bool)
+            out, _ = a(x, resets, st)
+        else:
+            a = build_actor(cfg.network.actor_network, obs, act)
+            out = a(torch.zeros(2, *obs.shape))
+        assert out is not None, name
