@@ -39,6 +39,10 @@ run stoix_amd.systems.search.ff_sampled_az    "" system.num_simulations=4 system
 run stoix_amd.systems.search.ff_mz            "" system.num_simulations=4 system.epochs=1 system.unroll_steps=2 system.n_step=2 system.batch_size=8 system.buffer_size=256
 run stoix_amd.systems.search.ff_sampled_mz    "" system.num_simulations=4 system.num_sampled_actions=4 system.epochs=1 system.unroll_steps=2 system.n_step=2 system.batch_size=8 system.buffer_size=256
 run stoix_amd.systems.spo.ff_spo              "" system.num_particles=4 system.search_depth=2 system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.disco_rl.ff_disco103    "" system.num_minibatches=2 system.epochs=1 system.disco_rule.num_bins=21 system.disco_rule.net.prediction_size=16
+run stoix_amd.systems.ppo.ff_ppo              "xland" env=xland_minigrid/goal_grid system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.ff_ppo              "craftax" env=craftax/crafting system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.rec_ppo             "popjym" env=popjym/stateless_cartpole system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.impala.sebulba_ff_impala "" arch.actor.actor_per_device=2 system.num_minibatches=2
 run stoix_amd.systems.ppo.sebulba_ff_ppo      "" arch.actor.actor_per_device=2 system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.impala.sebulba_ff_impala_shared_torso "" arch.actor.actor_per_device=2 system.num_minibatches=2
