@@ -180,3 +180,49 @@ def test_two_rank_ppo_learns_identity():
     r0, _ = res[0]
     # rank 0 evaluates; optimal = 10, random = 2.5
     assert r0 > 7.0, f"2-rank DP PPO failed to learn: return={r0}"
+
+
+def _reducer4_worker(rank: int, port: int, outdir: str):
+    try:
+        os.environ["MASTER_ADDR"] = "127.0.0.1"
+        os.environ["MASTER_PORT"] = str(port)
+        os.environ["RANK"] = str(rank)
+        os.environ["WORLD_SIZE"] = "4"
+        os.environ["LOCAL_RANK"] = str(rank)
+        from stoix_amd.parallel.dist import FlatGradReducer, get_dist_context, reset_dist_context
+
+        reset_dist_context()
+        ctx = get_dist_context(force_cpu=True)
+        torch.manual_seed(100 + rank)
+        lin = torch.nn.Linear(8, 4)
+        loss = lin(torch.randn(16, 8)).pow(2).mean()
+        loss.backward()
+        red = FlatGradReducer(list(lin.parameters()), ctx.device)
+        red.reduce()
+        red.wait()
+        torch.save([p.grad.clone() for p in lin.parameters()],
+                   os.path.join(outdir, f"r{rank}.pt"))
+    except Exception as e:  # pragma: no cover
+        torch.save(("ERROR", repr(e)), os.path.join(outdir, f"r{rank}.pt"))
+
+
+def test_flat_grad_reducer_four_ranks():
+    """4-rank mean all-reduce: closer to the driver's 4/8-GPU topology
+    than the world=2 tests (averaging over >2 contributions)."""
+    import tempfile
+
+    ctx = mp.get_context("spawn")
+    with tempfile.TemporaryDirectory() as outdir:
+        procs = [ctx.Process(target=_reducer4_worker, args=(r, 29623, outdir)) for r in range(4)]
+        for p in procs:
+            p.start()
+        for p in procs:
+            p.join(timeout=120)
+        grads = []
+        for r in range(4):
+            g = torch.load(os.path.join(outdir, f"r{r}.pt"), weights_only=False)
+            assert not (isinstance(g, tuple) and g[0] == "ERROR"), g
+            grads.append(g)
+    for r in range(1, 4):
+        for a, b in zip(grads[0], grads[r]):
+            torch.testing.assert_close(a, b)
