@@ -129,8 +129,16 @@ def _envpool(name: str):
         if envpool_ext(required=True) is not None:
             return QbertCpu(**kw)
 
+    def vizdoom_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import VizdoomBasicCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return VizdoomBasicCpu(**kw)
+
     table = {
         "breakout": breakout_maker,
+        "vizdoom_basic": vizdoom_maker,
+        "VizdoomBasic-v1": vizdoom_maker,
         "Breakout-v5": breakout_maker,
         "pong": pong_maker,
         "Pong-v5": pong_maker,

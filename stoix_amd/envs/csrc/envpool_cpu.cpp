@@ -720,6 +720,158 @@ void qbert_cpu_step(torch::Tensor state, torch::Tensor action,
 }
 }  // namespace qbert
 
+
+// --------------------------------------------------- VizDoom-basic-class
+// First-person raycast shooter, the "basic" scenario: a rectangular room,
+// a stationary monster at a random position along the far wall, the
+// player strafes along the near wall and shoots. Rewards follow the
+// vizdoom_basic shape: +101 kill, -5 per missed shot, -1 living penalty
+// per step; episode ends on the kill or the step cap. The 84x84 obs is a
+// true perspective render: per-column wall raycast (box walls, distance
+// shading) with the monster as a distance-scaled billboard.
+namespace vizdoom {
+constexpr float ROOM_W = 8.0f, ROOM_D = 6.0f;
+constexpr float PLAYER_Y = 0.5f, MONSTER_Y = 5.5f;
+constexpr float FOV = 1.57079632679f;  // 90 degrees
+constexpr float STRAFE = 0.35f, AIM_HALF = 0.06f;
+constexpr float MONSTER_HALF_W = 0.35f;
+// state: player_x, monster_x, monster_alive, cooldown
+constexpr int SDIM = 4;
+
+inline void reset_env(float* s, uint64_t seed, uint64_t env, uint64_t draw) {
+  s[0] = ROOM_W / 2.0f;
+  s[1] = 1.0f + hash_uniform(seed, env * 7 + 3, draw) * (ROOM_W - 2.0f);
+  s[2] = 1.0f;
+  s[3] = 0.0f;
+}
+
+inline void render(const float* s, float* obs) {
+  const float px = s[0];
+  const bool alive = s[2] > 0.0f;
+  const float mx = s[1];
+  // monster angular extent (from the player, facing +y)
+  float mdx0 = (mx - MONSTER_HALF_W) - px, mdx1 = (mx + MONSTER_HALF_W) - px;
+  float mdy = MONSTER_Y - PLAYER_Y;
+  float mang0 = atan2f(mdx0, mdy), mang1 = atan2f(mdx1, mdy);
+  float mdist = mdy;  // billboard depth (player faces +y)
+  for (int c = 0; c < W; ++c) {
+    float ang = ((float)c / (float)(W - 1) - 0.5f) * FOV;
+    float dx = sinf(ang), dy = cosf(ang);
+    // box-wall intersection from (px, PLAYER_Y) along (dx, dy), dy > 0
+    float t_far = (ROOM_D - PLAYER_Y) / dy;
+    float t_side = 1e9f;
+    if (dx > 1e-6f) t_side = (ROOM_W - px) / dx;
+    else if (dx < -1e-6f) t_side = -px / dx;
+    float t = t_far < t_side ? t_far : t_side;
+    float dist = t * dy;  // perpendicular distance (no fisheye)
+    if (dist < 0.3f) dist = 0.3f;
+    int wall_h = (int)(60.0f / dist);
+    if (wall_h > H) wall_h = H;
+    int w0 = H / 2 - wall_h / 2, w1 = H / 2 + wall_h / 2;
+    float shade = 0.55f - 0.05f * dist;
+    if (shade < 0.15f) shade = 0.15f;
+    bool monster_col = alive && ang >= mang0 && ang <= mang1 && mdist < dist;
+    int m_h = (int)(50.0f / mdist);
+    int m0 = H / 2 - m_h / 2, m1 = H / 2 + m_h / 2;
+    for (int r = 0; r < H; ++r) {
+      float v;
+      if (monster_col && r >= m0 && r < m1) v = 0.95f;
+      else if (r >= w0 && r < w1) v = shade;
+      else if (r >= w1) v = 0.25f;  // floor
+      else v = 0.05f;               // ceiling
+      obs[r * W + c] = v;
+    }
+    // crosshair
+  }
+  obs[(H / 2) * W + W / 2] = 1.0f;
+  obs[(H / 2 + 1) * W + W / 2] = 1.0f;
+}
+
+void vizdoom_cpu_reset(torch::Tensor state, torch::Tensor obs, int64_t seed,
+                       int64_t draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  float* op = obs.data_ptr<float>();
+  for (int64_t b = 0; b < B; ++b) {
+    reset_env(sp + b * SDIM, (uint64_t)seed, (uint64_t)b, (uint64_t)draw);
+    render(sp + b * SDIM, op + b * H * W);
+  }
+}
+
+void vizdoom_cpu_step(torch::Tensor state, torch::Tensor action,
+                      torch::Tensor step_count, torch::Tensor ep_return,
+                      torch::Tensor ep_length, torch::Tensor last_ep_return,
+                      torch::Tensor last_ep_length, torch::Tensor obs,
+                      torch::Tensor next_obs, torch::Tensor reward,
+                      torch::Tensor discount, torch::Tensor steptype,
+                      torch::Tensor done, int64_t max_episode_steps,
+                      int64_t seed, torch::Tensor draw) {
+  int64_t B = state.size(0);
+  float* sp = state.data_ptr<float>();
+  const int64_t* ap = action.data_ptr<int64_t>();
+  int32_t* scp = step_count.data_ptr<int32_t>();
+  float* erp = ep_return.data_ptr<float>();
+  int32_t* elp = ep_length.data_ptr<int32_t>();
+  float* lerp = last_ep_return.data_ptr<float>();
+  int32_t* lelp = last_ep_length.data_ptr<int32_t>();
+  float* op = obs.data_ptr<float>();
+  float* nop = next_obs.data_ptr<float>();
+  float* rp = reward.data_ptr<float>();
+  float* dp = discount.data_ptr<float>();
+  uint8_t* stp = steptype.data_ptr<uint8_t>();
+  uint8_t* dnp = done.data_ptr<uint8_t>();
+  int32_t dr = *draw.data_ptr<int32_t>();
+  for (int64_t b = 0; b < B; ++b) {
+    float* s = sp + b * SDIM;
+    int a = (int)ap[b];
+    if (a < 0) a = 0;
+    if (a > 3) a = 3;
+    float px = s[0] + STRAFE * ((a == 2) - (a == 1));
+    if (px < 0.7f) px = 0.7f;
+    if (px > ROOM_W - 0.7f) px = ROOM_W - 0.7f;
+    float cool = s[3] > 0.0f ? s[3] - 1.0f : 0.0f;
+    float rew = -1.0f;  // living penalty
+    bool killed = false;
+    if (a == 3 && cool <= 0.0f) {
+      cool = 3.0f;
+      // hit if the monster centre is inside the aim cone (facing +y)
+      float aim = atan2f(s[1] - px, MONSTER_Y - PLAYER_Y);
+      if (s[2] > 0.0f && fabsf(aim) <= AIM_HALF) {
+        killed = true;
+        s[2] = 0.0f;
+        rew += 101.0f;
+      } else {
+        rew -= 5.0f;  // missed shot
+      }
+    }
+    s[0] = px;
+    s[3] = cool;
+    bool terminated = killed;
+
+    int sc = scp[b] + 1;
+    bool truncated = (sc >= max_episode_steps) && !terminated;
+    bool dn = terminated || truncated;
+    float ret = erp[b] + rew;
+    int len = elp[b] + 1;
+    if (dn) { lerp[b] = ret; lelp[b] = len; }
+    render(s, nop + b * H * W);
+    if (dn) {
+      reset_env(s, (uint64_t)seed, (uint64_t)b * 977 + 13, (uint64_t)dr);
+      sc = 0; ret = 0.0f; len = 0;
+      render(s, op + b * H * W);
+    } else {
+      std::memcpy(op + b * H * W, nop + b * H * W, sizeof(float) * H * W);
+    }
+    scp[b] = sc; erp[b] = ret; elp[b] = len;
+    rp[b] = rew;
+    dp[b] = terminated ? 0.0f : 1.0f;
+    stp[b] = terminated ? ST_TERMINATED : (truncated ? ST_TRUNCATED : ST_MID);
+    dnp[b] = dn ? 1 : 0;
+  }
+  *draw.data_ptr<int32_t>() = dr + 1;
+}
+}  // namespace vizdoom
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // gil_scoped_release: the whole fused step runs WITHOUT the GIL so
   // Sebulba's learner thread and sibling actor threads keep running
@@ -752,4 +904,11 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("qbert_step", &qbert::qbert_cpu_step, "batched Qbert-class fused step",
         py::call_guard<py::gil_scoped_release>());
   m.attr("QBERT_STATE_DIM") = qbert::SDIM;
+  m.def("vizdoom_reset", &vizdoom::vizdoom_cpu_reset,
+        "batched VizDoom-basic-class reset + raycast render",
+        py::call_guard<py::gil_scoped_release>());
+  m.def("vizdoom_step", &vizdoom::vizdoom_cpu_step,
+        "batched VizDoom-basic-class fused step",
+        py::call_guard<py::gil_scoped_release>());
+  m.attr("VIZDOOM_STATE_DIM") = vizdoom::SDIM;
 }

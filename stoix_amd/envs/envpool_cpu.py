@@ -156,6 +156,20 @@ class QbertCpu(_PoolEnvBase):
     N_ACTIONS = 4  # down-left / down-right / up-left / up-right
 
 
+class VizdoomBasicCpu(_PoolEnvBase):
+    """Native-engine VizDoom-basic-class: first-person raycast room, a
+    monster at a random far-wall position; strafe and shoot. Rewards:
+    +101 kill, -5 missed shot, -1 living penalty (the vizdoom_basic
+    shape). Obs is a true perspective render (per-column wall raycast +
+    distance-scaled monster billboard)."""
+
+    max_episode_steps = 300
+    STATE_DIM_ATTR = "VIZDOOM_STATE_DIM"
+    RESET_FN = "vizdoom_reset"
+    STEP_FN = "vizdoom_step"
+    N_ACTIONS = 4  # noop / left / right / shoot
+
+
 class BreakoutCpu(_PoolEnvBase):
     """Native-engine Breakout; drop-in for envs/breakout.py on CPU."""
 

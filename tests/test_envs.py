@@ -980,3 +980,38 @@ def test_stateless_pendulum_continuous_prev_action():
     a = torch.full((4, adim), 0.7)
     ts = env.step(a)
     torch.testing.assert_close(ts.observation[:, -adim:], a)
+
+
+def test_vizdoom_pool_rules():
+    import pytest as _pytest
+    import torch
+
+    if not _pool_available():
+        _pytest.skip("native pool ext not built")
+    from stoix_amd.envs.envpool_cpu import VizdoomBasicCpu
+
+    env = VizdoomBasicCpu(4, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (4, 84, 84, 1)
+    obs = ts.observation[..., 0]
+    # perspective render: ceiling darker than floor, monster billboard
+    # bright, walls mid
+    assert float(obs[:, 2, :].max()) <= 0.06       # ceiling
+    assert float(obs[:, 80, :].min()) >= 0.2       # floor
+    assert float(obs.max()) >= 0.9                 # monster or crosshair
+    # living penalty every step
+    ts = env.step(torch.zeros(4, dtype=torch.long))
+    assert torch.all(ts.reward == -1.0)
+    # missed shot costs -5 on top (aim the monster away first)
+    env._s[:, 1] = 7.3  # monster far right
+    env._s[:, 0] = 0.7  # player far left
+    ts = env.step(torch.full((4,), 3, dtype=torch.long))
+    assert torch.all(ts.reward == -6.0)
+    # point-blank aligned shot kills: +101 - 1 = +100
+    env._s[:, 1] = 4.0
+    env._s[:, 0] = 4.0
+    env._s[:, 2] = 1.0
+    env._s[:, 3] = 0.0  # cooldown clear
+    ts = env.step(torch.full((4,), 3, dtype=torch.long))
+    assert torch.all(ts.reward == 100.0)
+    assert torch.all(ts.discount == 0.0)  # kill terminates
