@@ -295,3 +295,24 @@ def test_rec_ppo_learns_stateless_cartpole():
     )
     r = run(cfg)
     assert r > 100.0, f"rec_ppo failed the POMDP cartpole: return={r}"
+
+
+@pytest.mark.slow
+def test_ppo_learns_vizdoom_basic():
+    """PPO+CNN on the first-person raycast shooter (vizdoom_basic reward
+    shape): must learn to centre the monster and shoot. Random play times
+    out around -300; measured 96 at this budget (near-optimal: quick kill,
+    few penalties)."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=envpool/vizdoom_basic", "network=cnn", "arch.total_num_envs=16",
+         "arch.total_timesteps=null", "arch.num_updates=40",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=8",
+         "arch.absolute_metric=false", "system.rollout_length=32",
+         "system.num_minibatches=4", "system.epochs=2", "system.ent_coef=0.02",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 50.0, f"vizdoom_basic not learned: return={r}"
