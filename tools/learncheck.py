@@ -105,6 +105,14 @@ def main():
     print(json.dumps(out))
     if out["fused"]:
         assert out["curve"][-1] > 400, f"fused CartPole not solved: {out['curve']}"
+    # fused + OBSERVATION NORMALISATION (round-2 path: Welford stats into
+    # the kernels' stable buffers; reference ff_ppo.py:90-162 ordering)
+    out = run_ppo("env=classic/cartpole", "disc", 60, 256,
+                  ("system.compute_dtype=bf16",
+                   "system.normalize_observations=true"))
+    print(json.dumps(out))
+    if out["fused"]:
+        assert out["curve"][-1] > 400, f"fused+obsnorm CartPole not solved: {out['curve']}"
     out = run_dqn_cartpole(160)
     print(json.dumps(out))
     # DQN-family on CartPole shows the documented rise->forget->recover
