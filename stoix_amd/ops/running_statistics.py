@@ -48,7 +48,10 @@ def update(
     running_statistics.py:62-70, 297-310)."""
     obs_ndim = state.mean.dim()
     flat = batch.reshape(-1, *state.mean.shape) if obs_ndim else batch.reshape(-1)
-    n = torch.tensor(float(flat.shape[0]), device=batch.device, dtype=state.mean.dtype)
+    # torch.full (device fill kernel), NOT torch.tensor(scalar, device=...):
+    # the latter is a pageable H2D copy — illegal inside hip-graph capture
+    # (the fused obs-norm path runs this inside the rollout graph)
+    n = torch.full((), float(flat.shape[0]), device=batch.device, dtype=state.mean.dtype)
     s = flat.sum(dim=0)
     if all_reduce and dist.is_initialized() and dist.get_world_size() > 1:
         packed = torch.cat([n.reshape(1), s.reshape(-1)])
