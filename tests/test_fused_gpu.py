@@ -681,6 +681,12 @@ def test_fused_obs_norm_cartpole(ext):
     learner = PPOLearner(cfg, env, dev)
     assert learner.normalize_obs
     assert learner.fused is not None and learner.fused.norm
+    # capture the phases too: the Welford update runs INSIDE the rollout
+    # graph (a capture-illegal op here once shipped behind the uncaptured
+    # version of this test)
+    from stoix_amd.ops.graph import try_enable_graphs
+
+    assert try_enable_graphs(learner)
     c0 = float(learner.obs_stats.count)
     for _ in range(3):
         m = learner.update_step()
