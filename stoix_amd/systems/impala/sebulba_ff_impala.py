@@ -192,6 +192,10 @@ def run_experiment(config, force_cpu: bool = False) -> float:
             logger.log({"steps_per_second": sps, **train_metrics, **timers.summary()}, t_env, update, LogEvent.TRAIN)
             async_eval.submit_evaluation(learner.cpu_params(), t_env)
 
+    # train-phase throughput (same contract as sebulba_ff_ppo: recorded
+    # before the evaluator drain)
+    run_experiment.last_sps = t_env / max(time.perf_counter() - t0, 1e-9)
+
     lifetime.stop()
     for th in threads:
         th.join(timeout=10)

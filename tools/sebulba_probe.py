@@ -112,3 +112,26 @@ if __name__ == "__main__":
     if which in ("all", "e2e"):
         for n_envs, actors in ((256, 4), (1024, 4), (2048, 8)):
             print(json.dumps(probe_e2e(n_envs, actors)))
+
+
+def probe_impala(n_envs=2048, actors=8, rollout=64, updates=16):
+    from stoix_amd.config import compose
+    from stoix_amd.systems.impala.sebulba_ff_impala import run_experiment
+
+    cfg = compose(
+        "default/sebulba/default_ff_impala.yaml",
+        ["env=envpool/breakout", "network=cnn",
+         f"arch.total_num_envs={n_envs}", "arch.total_timesteps=null",
+         f"arch.num_updates={updates}", "arch.num_evaluation=1",
+         "arch.num_eval_episodes=4", f"arch.actor.actor_per_device={actors}",
+         f"system.rollout_length={rollout}", "system.num_minibatches=2",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    import time as _t
+
+    t0 = _t.perf_counter()
+    run_experiment(cfg)
+    dt = _t.perf_counter() - t0
+    return {"probe": "impala_e2e", "n_envs": n_envs, "actors": actors,
+            "sps": getattr(run_experiment, "last_sps", n_envs * rollout * updates / dt),
+            "wall_s": dt}
