@@ -146,6 +146,13 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     eval_gen = torch.Generator(device=eval_device)
     eval_gen.manual_seed(int(config.arch.seed) + 99_999)
 
+    from stoix_amd.utils.sebulba import prewarm_convs
+
+    prewarm_convs(config, learner.actor, learner.critic, obs_space.shape, learner_device)
+    import os as _os
+
+    torch.set_num_threads(min(8, _os.cpu_count() or 8))
+
     def evaluate_snapshot(params: Dict, t_env: int) -> Dict:
         from stoix_amd.evaluator import evaluate
 

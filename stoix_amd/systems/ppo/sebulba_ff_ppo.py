@@ -290,33 +290,9 @@ def run_experiment(config, force_cpu: bool = False) -> float:
 
     async_eval = AsyncEvaluator(evaluate_snapshot, lifetime)
 
-    # MIOpen conv-find prewarm: on a fresh box the FIRST use of each conv
-    # shape pays auto-tuning (minutes for the CNN shapes). Run one pass of
-    # every shape the run will use — actor-batch bf16 inference, eval
-    # batch, learner-minibatch fp32 forward+backward — BEFORE the actors
-    # and the throughput clock start. One-off, disk-cached afterwards.
-    if learner_device.type == "cuda":
-        with torch.random.fork_rng(devices=[learner_device]):
-            T_ = int(config.system.rollout_length)
-            bs_actor = int(config.arch.num_envs_per_actor)
-            mb_rows = max(
-                1,
-                T_ * int(config.arch.total_num_envs) // int(config.system.num_minibatches),
-            )
-            shp = obs_space.shape
-            with torch.no_grad(), torch.autocast("cuda", torch.bfloat16):
-                for bs in {bs_actor, int(config.arch.num_eval_episodes)}:
-                    x = torch.zeros(bs, *shp, device=learner_device)
-                    learner.actor(x)
-                    learner.critic(x)
-            x = torch.zeros(mb_rows, *shp, device=learner_device)
-            d = learner.actor(x)
-            probe = d.entropy().sum() + learner.critic(x).sum()
-            probe.backward()
-            learner.actor_opt.zero_grad(set_to_none=True)
-            learner.critic_opt.zero_grad(set_to_none=True)
-            del x, d, probe
-            torch.cuda.synchronize(learner_device)
+    from stoix_amd.utils.sebulba import prewarm_convs
+
+    prewarm_convs(config, learner.actor, learner.critic, obs_space.shape, learner_device)
 
     num_updates = int(config.arch.num_updates)
     threads = []
