@@ -204,6 +204,32 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     eval_gen = torch.Generator(device=eval_device)
     eval_gen.manual_seed(int(config.arch.seed) + 99_999)
 
+    # MIOpen conv prewarm for the shared net's shapes (see
+    # utils/sebulba.prewarm_convs; the shared-torso net returns
+    # (dist, value) from one call so the helper's actor/critic split
+    # doesn't apply)
+    if learner_device.type == "cuda":
+        with torch.random.fork_rng(devices=[learner_device]):
+            mb_rows = max(
+                1,
+                int(config.system.rollout_length) * int(config.arch.total_num_envs)
+                // int(config.system.num_minibatches),
+            )
+            with torch.no_grad(), torch.autocast("cuda", torch.bfloat16):
+                for bs in {int(config.arch.num_envs_per_actor),
+                           int(config.arch.num_eval_episodes)}:
+                    learner.net(torch.zeros(bs, *obs_space.shape, device=learner_device))
+            x = torch.zeros(mb_rows, *obs_space.shape, device=learner_device)
+            d, v = learner.net(x)
+            (d.entropy().sum() + v.sum()).backward()
+            for p_ in learner.net.parameters():
+                p_.grad = None
+            del x, d, v
+            torch.cuda.synchronize(learner_device)
+    import os as _os
+
+    torch.set_num_threads(min(8, _os.cpu_count() or 8))
+
     def evaluate_snapshot(params: Dict, t_env: int) -> Dict:
         from stoix_amd.evaluator import evaluate
 
