@@ -284,10 +284,14 @@ class PPOLearner:
 
     def prepare_for_graph_capture(self) -> None:
         """Switch to capture-safe modes: default CUDA RNG (graph-aware),
-        inline all-reduce on the capture stream, no host-side metric reads."""
+        inline all-reduce on the capture stream, no host-side metric
+        reads; capture-safe torch envs switch to unconditional-autoreset
+        graph mode."""
         self.gen = None
         self.collect_metrics = False
         self.reducer._stream = None
+        if hasattr(self.env, "prepare_for_graph_capture"):
+            self.env.prepare_for_graph_capture()
 
     def after_graph_replay(self) -> None:
         # episode metrics read eagerly from the env's latched buffers, with
