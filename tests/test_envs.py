@@ -1015,3 +1015,42 @@ def test_vizdoom_pool_rules():
     ts = env.step(torch.full((4,), 3, dtype=torch.long))
     assert torch.all(ts.reward == 100.0)
     assert torch.all(ts.discount == 0.0)  # kill terminates
+
+
+def test_graph_mode_step_equivalence_capture_safe_envs():
+    """graph_mode (unconditional-autoreset capture path) must produce the
+    SAME transition as the eager path for non-done envs — the only
+    difference is that the reset branch always executes (overwriting done
+    rows only). Pinned for every capture_safe env family."""
+    import torch
+
+    from stoix_amd.envs.crafting import Crafting
+    from stoix_amd.envs.snake import Snake
+    from stoix_amd.envs.xland import XLandGrid
+
+    for cls, n_act in ((Snake, 4), (XLandGrid, 4), (Crafting, 6)):
+        eager = cls(8, seed=3)
+        graph = cls(8, seed=3)
+        eager.reset()
+        graph.reset()
+        graph.graph_mode = True
+        g = torch.Generator().manual_seed(1)
+        for i in range(10):
+            # keep the two envs' states identical at step entry
+            for k in eager._state:
+                graph._state[k] = eager._state[k].clone()
+            graph._step_count.copy_(eager._step_count)
+            a = torch.randint(0, n_act, (8,), generator=g)
+            ts_e = eager.step(a)
+            ts_g = graph.step(a)
+            torch.testing.assert_close(ts_g.reward, ts_e.reward)
+            torch.testing.assert_close(ts_g.discount, ts_e.discount)
+            torch.testing.assert_close(ts_g.step_type, ts_e.step_type)
+            alive = ~ts_e.extras["episode_metrics"]["is_terminal_step"]
+            torch.testing.assert_close(
+                ts_g.observation[alive], ts_e.observation[alive],
+                msg=f"{cls.__name__} step {i}",
+            )
+            torch.testing.assert_close(
+                ts_g.extras["next_obs"], ts_e.extras["next_obs"]
+            )
