@@ -34,6 +34,7 @@ CH = 16
 
 class Game2048(StatefulVecEnv):
     max_episode_steps = 10000
+    capture_safe = True  # Gumbel-max spawns, no host-synced branches
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
@@ -109,8 +110,12 @@ class Game2048(StatefulVecEnv):
         board where mask is True."""
         B = board.shape[0]
         flat = board.reshape(B, N * N)
-        empty = (flat == 0).float().clamp(min=1e-9)
-        idx = torch.multinomial(empty, 1, generator=self.gen).squeeze(-1)
+        # Gumbel-max over the empty mask: uniform over empty cells and
+        # capture-legal (torch.multinomial is not graph-capturable)
+        empty = flat == 0
+        u = torch.rand(B, N * N, device=self.device, generator=self.gen)
+        gum = -torch.log(-torch.log(u.clamp(min=1e-12)).clamp(min=1e-12))
+        idx = torch.where(empty, gum, torch.full_like(gum, -torch.inf)).argmax(dim=-1)
         val = torch.where(
             torch.rand(B, device=self.device, generator=self.gen) < 0.9, 1, 2
         ).to(board.dtype)

@@ -202,8 +202,11 @@ class RobotWarehouse(StatefulVecEnv):
             )
             unreq = ~requested
             unreq[bidx, cid2] = unreq[bidx, cid2] & ~deliver  # not the one just delivered? keep eligible
-            probs = unreq.float().clamp(min=1e-9)
-            new_req = torch.multinomial(probs, 1, generator=self.gen).squeeze(-1)
+            # Gumbel-max over the un-requested mask (capture-legal
+            # uniform choice; torch.multinomial is not graph-capturable)
+            u = torch.rand(unreq.shape, device=dev, generator=self.gen)
+            gum = -torch.log(-torch.log(u.clamp(min=1e-12)).clamp(min=1e-12))
+            new_req = torch.where(unreq, gum, torch.full_like(gum, -torch.inf)).argmax(dim=-1)
             requested[bidx, new_req] = torch.where(
                 deliver, torch.ones_like(deliver), requested[bidx, new_req]
             )

@@ -1025,10 +1025,11 @@ def test_graph_mode_step_equivalence_capture_safe_envs():
     import torch
 
     from stoix_amd.envs.crafting import Crafting
+    from stoix_amd.envs.game2048 import Game2048
     from stoix_amd.envs.snake import Snake
     from stoix_amd.envs.xland import XLandGrid
 
-    for cls, n_act in ((Snake, 4), (XLandGrid, 4), (Crafting, 6)):
+    for cls, n_act in ((Snake, 4), (XLandGrid, 4), (Crafting, 6), (Game2048, 4)):
         eager = cls(8, seed=3)
         graph = cls(8, seed=3)
         eager.reset()
