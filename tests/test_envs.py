@@ -1037,10 +1037,13 @@ def test_graph_mode_step_equivalence_capture_safe_envs():
         graph.graph_mode = True
         g = torch.Generator().manual_seed(1)
         for i in range(10):
-            # keep the two envs' states identical at step entry
+            # keep the two envs' states AND RNG streams identical at step
+            # entry (graph mode consumes extra reset draws AFTER the
+            # transition draws, which only touch done rows)
             for k in eager._state:
                 graph._state[k] = eager._state[k].clone()
             graph._step_count.copy_(eager._step_count)
+            graph.gen.set_state(eager.gen.get_state())
             a = torch.randint(0, n_act, (8,), generator=g)
             ts_e = eager.step(a)
             ts_g = graph.step(a)
