@@ -164,9 +164,15 @@ def _debug(name: str):
 
 
 def _xland(name: str):
-    from stoix_amd.envs.xland import XLandGrid
+    from stoix_amd.envs.xland import DoorKeyGrid, XLandGrid
 
-    table = {"goal_grid": XLandGrid, "XLand-MiniGrid-R1": XLandGrid}
+    table = {
+        "goal_grid": XLandGrid,
+        "XLand-MiniGrid-R1": XLandGrid,
+        # navix/MiniGrid-class tasks share the gridworld machinery
+        "doorkey": DoorKeyGrid,
+        "Navix-DoorKey-8x8-v0": DoorKeyGrid,
+    }
     if name not in table:
         raise ValueError(f"unknown xland-suite env '{name}' (have {list(table)})")
     return table[name]
@@ -224,6 +230,7 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "gymnasium": _gymnasium,
     "xland_minigrid": _xland,
     "xland": _xland,
+    "navix": _xland,
     "craftax": _craftax,
     "popjym": _popjym,
     "debug": _debug,

@@ -142,3 +142,114 @@ class XLandGrid(StatefulVecEnv):
             reward,
             terminated,
         )
+
+
+class DoorKeyGrid(StatefulVecEnv):
+    """DoorKey (navix/MiniGrid-class): a wall splits the room; the agent
+    must pick up the KEY, open the DOOR, and reach the GOAL. Sparse
+    terminal reward 1 - 0.9 * t/T (the MiniGrid shaping), sub-rewards 0.
+    Restores the navix suite's capability class (the reference's navix is
+    JAX-only, make_env.py). Vectorised torch; capture-safe.
+
+    Observation [9, 9, 6]: walls, agent, key (if not held), door (closed),
+    goal, held-key indicator plane (broadcast 0.5 when carrying).
+    Actions: 0-3 move (pickup/open happen by walking into the cell).
+    """
+
+    max_episode_steps = 200
+    capture_safe = True
+    solved_return_threshold = 0.5
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.observation_space = BoxSpace((N, N, 6), 0.0, 1.5)
+        self.action_space = DiscreteSpace(4)
+        self._dr = torch.tensor(_DR, device=self.device)
+        self._dc = torch.tensor(_DC, device=self.device)
+        self._one_f = torch.ones((), device=self.device)
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        # dividing wall at a random column 3..5 with a door at a random row
+        wall_col = torch.randint(3, 6, (n,), device=dev, generator=self.gen)
+        door_row = torch.randint(1, N - 1, (n,), device=dev, generator=self.gen)
+        cols = torch.arange(N, device=dev)
+        rows = torch.arange(N, device=dev)
+        walls = torch.zeros(n, N, N, dtype=torch.bool, device=dev)
+        walls[:, 0, :] = walls[:, -1, :] = True
+        walls[:, :, 0] = walls[:, :, -1] = True
+        walls |= cols.view(1, 1, N) == wall_col.view(n, 1, 1)
+        # carve the door cell out of the wall mask (it is tracked separately)
+        bidx = torch.arange(n, device=dev)
+        walls[bidx, door_row, wall_col] = False
+        # key on the LEFT side, goal on the RIGHT side, agent LEFT
+        u = torch.rand(n, 4, device=dev, generator=self.gen)
+        key_r = 1 + (u[:, 0] * (N - 2)).long().clamp(max=N - 3)
+        key_c = 1 + (u[:, 1] * (wall_col.float() - 1.0)).long().clamp(min=0)
+        key_c = torch.minimum(key_c, wall_col - 1).clamp(min=1)
+        agent_r = 1 + (u[:, 2] * (N - 2)).long().clamp(max=N - 3)
+        agent_c = torch.ones(n, dtype=torch.long, device=dev)
+        goal_r = door_row  # reachable by construction
+        goal_c = torch.full((n,), N - 2, dtype=torch.long, device=dev)
+        return {
+            "walls": walls.float(),
+            "door_r": door_row.float(), "door_c": wall_col.float(),
+            "door_open": torch.zeros(n, device=dev),
+            "key_r": key_r.float(), "key_c": key_c.float(),
+            "has_key": torch.zeros(n, device=dev),
+            "agent_r": agent_r.float(), "agent_c": agent_c.float(),
+            "goal_r": goal_r.float(), "goal_c": goal_c.float(),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        n = state["agent_r"].shape[0]
+        dev = self.device
+        bidx = torch.arange(n, device=dev)
+        obs = torch.zeros(n, N, N, 6, device=dev)
+        obs[..., 0] = state["walls"]
+        obs[bidx, state["agent_r"].long(), state["agent_c"].long(), 1] = self._one_f
+        key_vis = (state["has_key"] < 0.5).float()
+        obs[bidx, state["key_r"].long(), state["key_c"].long(), 2] = key_vis
+        door_closed = (state["door_open"] < 0.5).float()
+        obs[bidx, state["door_r"].long(), state["door_c"].long(), 3] = door_closed
+        obs[bidx, state["goal_r"].long(), state["goal_c"].long(), 4] = self._one_f
+        obs[..., 5] = 0.5 * state["has_key"].view(n, 1, 1)
+        return obs
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        n = state["agent_r"].shape[0]
+        dev = self.device
+        a = action.long().clamp(0, 3)
+        r, c = state["agent_r"].long(), state["agent_c"].long()
+        nr = (r + self._dr[a]).clamp(0, N - 1)
+        nc = (c + self._dc[a]).clamp(0, N - 1)
+        bidx = torch.arange(n, device=dev)
+        walls = state["walls"] > 0.5
+        at_door = (nr == state["door_r"].long()) & (nc == state["door_c"].long())
+        door_open = state["door_open"] > 0.5
+        has_key = state["has_key"] > 0.5
+        # a closed door blocks unless the agent carries the key (walking
+        # into it with the key OPENS it and moves through)
+        blocked = walls[bidx, nr, nc] | (at_door & ~door_open & ~has_key)
+        nr = torch.where(blocked, r, nr)
+        nc = torch.where(blocked, c, nc)
+        new_door_open = door_open | (at_door & has_key & ~blocked)
+        on_key = (nr == state["key_r"].long()) & (nc == state["key_c"].long()) & ~has_key
+        new_has_key = has_key | on_key
+        at_goal = (nr == state["goal_r"].long()) & (nc == state["goal_c"].long())
+        # MiniGrid terminal shaping: 1 - 0.9 * t / T
+        frac = self._step_count.float() / float(self.max_episode_steps)
+        reward = at_goal.float() * (1.0 - 0.9 * frac)
+        return (
+            {
+                "walls": state["walls"],
+                "door_r": state["door_r"], "door_c": state["door_c"],
+                "door_open": new_door_open.float(),
+                "key_r": state["key_r"], "key_c": state["key_c"],
+                "has_key": new_has_key.float(),
+                "agent_r": nr.float(), "agent_c": nc.float(),
+                "goal_r": state["goal_r"], "goal_c": state["goal_c"],
+            },
+            reward,
+            at_goal,
+        )

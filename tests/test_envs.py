@@ -1058,3 +1058,56 @@ def test_graph_mode_step_equivalence_capture_safe_envs():
             torch.testing.assert_close(
                 ts_g.extras["next_obs"], ts_e.extras["next_obs"]
             )
+
+
+def test_doorkey_rules():
+    """navix/MiniGrid-class DoorKey: closed door blocks without the key;
+    walking the key->door->goal chain terminates with the shaped reward."""
+    import torch
+
+    from stoix_amd.envs.xland import N, DoorKeyGrid
+
+    env = DoorKeyGrid(2, seed=0)
+    env.reset()
+    s = env._state
+    # deterministic scene: wall col 4, door at row 4; agent left of door
+    s["walls"][:] = 0.0
+    s["walls"][:, 0, :] = s["walls"][:, -1, :] = 1.0
+    s["walls"][:, :, 0] = s["walls"][:, :, -1] = 1.0
+    s["walls"][:, :, 4] = 1.0
+    s["walls"][:, 4, 4] = 0.0  # door cell carved out
+    s["door_r"][:] = 4.0
+    s["door_c"][:] = 4.0
+    s["door_open"][:] = 0.0
+    s["key_r"][:] = 4.0
+    s["key_c"][:] = 2.0
+    s["has_key"][:] = 0.0
+    s["agent_r"][:] = 4.0
+    s["agent_c"][:] = 3.0
+    s["goal_r"][:] = 4.0
+    s["goal_c"][:] = 7.0
+    right = torch.full((2,), 1, dtype=torch.long)
+    left = torch.full((2,), 3, dtype=torch.long)
+
+    # walking into the closed door WITHOUT the key: blocked
+    env.step(right)
+    assert torch.all(env._state["agent_c"] == 3.0)
+    assert torch.all(env._state["door_open"] == 0.0)
+    # grab the key (walk left onto it)
+    env.step(left)
+    assert torch.all(env._state["has_key"] == 1.0)
+    # key plane disappears, held-key plane lights up
+    ts_obs = env._obs_fn(env._state)
+    assert torch.all(ts_obs[..., 2] == 0.0)
+    assert torch.all(ts_obs[..., 5] == 0.5)
+    # back to the door and through it (opens with the key)
+    env.step(right)
+    env.step(right)
+    assert torch.all(env._state["door_open"] == 1.0)
+    assert torch.all(env._state["agent_c"] == 4.0)
+    # walk to the goal: reward = 1 - 0.9 * t/T and termination
+    env.step(right)  # col 5
+    env.step(right)  # col 6
+    ts = env.step(right)  # col 7 == goal
+    assert torch.all(ts.discount == 0.0)
+    assert torch.all(ts.reward > 0.9)  # 7 steps of 200 -> ~0.97

@@ -338,3 +338,24 @@ def test_sac_learns_pendulum():
     )
     r = run(cfg)
     assert r > -1200.0, f"SAC did not learn pendulum: return={r}"
+
+
+@pytest.mark.slow
+def test_ppo_learns_doorkey():
+    """PPO on the navix/MiniGrid-class DoorKey chain (key -> door -> goal,
+    sparse terminal-only reward): measured 0.96 at this budget (random
+    exploration rarely completes the chain inside 200 steps)."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=navix/doorkey", "arch.total_num_envs=256",
+         "arch.total_timesteps=null", "arch.num_updates=80",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=64",
+         "system.num_minibatches=4", "system.epochs=4",
+         "system.ent_coef=0.03", "logger.loggers=[]",
+         "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 0.5, f"DoorKey not learned: return={r}"
