@@ -207,6 +207,27 @@ def _popjym(name: str):
     return table[name]
 
 
+def _playground(name: str):
+    from stoix_amd.envs.classic import CartPoleSwingUp, Pendulum
+    from stoix_amd.envs.humanoid import Humanoid
+    from stoix_amd.envs.planar import HalfCheetah, Hopper
+
+    # mujoco_playground-class: dm_control-style control + locomotion
+    table = {
+        "cartpole_swingup": CartPoleSwingUp,
+        "CartpoleSwingup": CartPoleSwingUp,
+        "pendulum_swingup": Pendulum,
+        # locomotion tasks map onto the physics envs of the brax-class tier
+        "ant": Ant,
+        "humanoid": Humanoid,
+        "halfcheetah": HalfCheetah,
+        "hopper": Hopper,
+    }
+    if name not in table:
+        raise ValueError(f"unknown playground-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
 def _gymnasium(name: str):
     # the gymnasium suite is factory-only (stateful CPU envs for Sebulba,
     # reference utils/env_factory.py:71-86); Anakin's make_single cannot
@@ -231,6 +252,7 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "xland_minigrid": _xland,
     "xland": _xland,
     "navix": _xland,
+    "mujoco_playground": _playground,
     "craftax": _craftax,
     "popjym": _popjym,
     "debug": _debug,

@@ -221,3 +221,64 @@ class Acrobot(StatefulVecEnv):
         terminated = (-torch.cos(th1) - torch.cos(th2 + th1)) > 1.0
         reward = torch.where(terminated, 0.0, -1.0)
         return {"s": ns}, reward, terminated
+
+
+class CartPoleSwingUp(CartPole):
+    """dm_control-style cartpole swingup (mujoco_playground /
+    dm_control capability class): the pole STARTS HANGING (theta ~ pi),
+    the action is a continuous force, and the reward is the smooth
+    upright/centred product dm_control uses — no termination, fixed
+    horizon. Same physics core as CartPole."""
+
+    max_episode_steps = 500
+    solved_return_threshold = 350.0
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kwargs):
+        super().__init__(num_envs, device, seed, **kwargs)
+        from stoix_amd.envs.spaces import BoxSpace
+
+        self.observation_space = BoxSpace((5,), -5.0, 5.0)
+        self.action_space = BoxSpace((1,), -1.0, 1.0)
+        self._hip = None  # torch path (the HIP kernel is the discrete game)
+
+    def _reset_fn(self, n: int):
+        s = self.rand(n, 4, lo=-0.05, hi=0.05)
+        s[:, 2] = s[:, 2] + math.pi  # hanging down
+        return {"s": s}
+
+    def _obs_fn(self, state):
+        s = state["s"]
+        # dm_control-style: [x, x_dot, cos(theta), sin(theta), theta_dot]
+        return torch.stack(
+            [s[:, 0], s[:, 1], torch.cos(s[:, 2]), torch.sin(s[:, 2]), s[:, 3]],
+            dim=-1,
+        )
+
+    def _step_fn(self, state, action):
+        s = state["s"]
+        x, x_dot, th, th_dot = s.unbind(-1)
+        force = action.reshape(-1).clamp(-1.0, 1.0) * self.FORCE_MAG
+        costheta = torch.cos(th)
+        sintheta = torch.sin(th)
+        total_mass = self.MASSCART + self.MASSPOLE
+        pml = self.MASSPOLE * self.LENGTH
+        temp = (force + pml * th_dot * th_dot * sintheta) / total_mass
+        thacc = (self.GRAVITY * sintheta - costheta * temp) / (
+            self.LENGTH * (4.0 / 3.0 - self.MASSPOLE * costheta * costheta / total_mass)
+        )
+        xacc = temp - pml * thacc * costheta / total_mass
+        x = x + self.TAU * x_dot
+        x_dot = x_dot + self.TAU * xacc
+        th = th + self.TAU * th_dot
+        th_dot = th_dot + self.TAU * thacc
+        # keep the cart on the track (elastic walls, dm_control style slide limit)
+        hit = x.abs() > self.X_LIMIT
+        x = x.clamp(-self.X_LIMIT, self.X_LIMIT)
+        x_dot = torch.where(hit, torch.zeros_like(x_dot), x_dot)
+        new_s = torch.stack([x, x_dot, th, th_dot], dim=-1)
+        # smooth dm_control reward: upright * centred in [0, 1]
+        upright = (1.0 + torch.cos(th)) / 2.0
+        centred = 1.0 - (x.abs() / self.X_LIMIT) ** 2 * 0.5
+        reward = upright * centred
+        terminated = torch.zeros_like(hit)  # horizon-only
+        return {"s": new_s}, reward, terminated

@@ -197,8 +197,44 @@ class AffineTanhTransformedDistribution(Distribution):
         return (self.base.log_prob(u) - log_det).sum(-1)
 
     def log_prob(self, value: Tensor) -> Tensor:
-        u = self._inverse(value)
-        return self._log_prob_from_u(u)
+        """Reference boundary semantics (distributions.py:55-80): inside
+        the clipping range use the change-of-variables density; AT/BEYOND
+        the range use the log of the AVERAGE density of the probability
+        MASS squashed into the boundary strip —
+        ``log_cdf(u_min)/log_sf(u_max) - log(eps)``. The mass form is
+        bounded (~0 for a saturated base), unlike the clamped-atanh
+        density (which reaches -900s when |loc| drifts and detonates the
+        PPO ratio: exp(new - (-900)) = inf; found by the swing-up stress
+        test)."""
+        # thresholds in ACTION units: [min+eps, max-eps] (reference :58-59)
+        lo = self.shift_affine - self.scale_affine
+        hi = self.shift_affine + self.scale_affine
+        min_th = lo + self.eps
+        max_th = hi - self.eps
+        y_min = (min_th - self.shift_affine) / self.scale_affine
+        y_max = (max_th - self.shift_affine) / self.scale_affine
+        if isinstance(y_min, Tensor):
+            u_min = torch.atanh(y_min)
+            u_max = torch.atanh(y_max)
+        else:
+            u_min = math.atanh(y_min)
+            u_max = math.atanh(y_max)
+        loc, scale = self.base.loc, self.base.scale
+        log_eps = (
+            torch.log(self.eps) if isinstance(self.eps, Tensor) else math.log(self.eps)
+        )
+        # log_cdf / log_survival of the base Normal at the thresholds
+        log_prob_left = torch.special.log_ndtr((u_min - loc) / scale) - log_eps
+        log_prob_right = torch.special.log_ndtr((loc - u_max) / scale) - log_eps
+        v = value.clamp(min_th, max_th)
+        u = torch.atanh(((v - self.shift_affine) / self.scale_affine).clamp(-1.0 + 1e-6, 1.0 - 1e-6))
+        log_det = 2.0 * (math.log(2.0) - u - F.softplus(-2.0 * u)) + self._log_scale
+        inside = self.base.log_prob(u) - log_det
+        per_dim = torch.where(
+            v <= min_th, log_prob_left,
+            torch.where(v >= max_th, log_prob_right, inside),
+        )
+        return per_dim.sum(-1)
 
     def entropy(self, num_samples: int = 1, generator=None) -> Tensor:
         """Monte-Carlo entropy estimate (exact entropy of a tanh-Gaussian has

@@ -359,3 +359,28 @@ def test_ppo_learns_doorkey():
     )
     r = run(cfg)
     assert r > 0.5, f"DoorKey not learned: return={r}"
+
+
+@pytest.mark.slow
+def test_ppo_learns_cartpole_swingup():
+    """Continuous PPO on the dm_control-style swing-up (mujoco_playground
+    capability class). Doubles as the regression gate for the tanh-normal
+    BOUNDARY log-prob semantics: with the clamped-atanh density this task
+    detonates (stored self-log-probs hit -900, exp(new-old)=inf -> NaN at
+    ~update 50); with the reference's CDF-mass boundary branch it learns.
+    Hanging-random is ~5-30; measured ~196 at this budget."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo_continuous.yaml",
+        ["env=mjc_playground/cartpole_swingup", "arch.total_num_envs=128",
+         "arch.total_timesteps=null", "arch.num_updates=150",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=64",
+         "system.num_minibatches=4", "system.epochs=4",
+         "system.ent_coef=0.01", "logger.loggers=[]",
+         "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r == r, "swing-up produced NaN (boundary log-prob regression)"
+    assert r > 100.0, f"swing-up not learned: return={r}"
