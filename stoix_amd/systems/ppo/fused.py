@@ -42,8 +42,13 @@ softmax-jacobian backward in ppo_head_loss_disc), ScalarCriticHead, flat
 obs <= 128 dims, no observation normalisation.
 
 Algorithm semantics match ff_ppo.py exactly (same losses, same GAE buffers,
-same per-minibatch all-reduce + per-chain clip); numerics are bf16-GEMM
-class, verified against the eager fp32 path in tests/test_fused_math.py and
+same per-minibatch all-reduce + per-chain clip) with ONE documented
+divergence: at the tanh BOUNDARY (|y| >= 1-eps) the kernel's log-prob uses
+the clamped-atanh density while the eager distribution uses the
+reference's CDF-mass branch (distributions.py log_prob). The fused
+eligibility regime (Ant-class, healthy sigma) keeps actions interior where
+the two coincide; boundary-saturating tasks (e.g. the swing-up stress
+test) run the eager path. Numerics are bf16-GEMM class, verified against the eager fp32 path in tests/test_fused_math.py and
 tests/test_fused_gpu.py, and learning curves are re-validated after every
 numerics change (tools/learncheck.py; profiles/r01_learning_curves.md).
 """
