@@ -228,6 +228,15 @@ def _playground(name: str):
     return table[name]
 
 
+def _kinetix(name: str):
+    from stoix_amd.envs.reacher import ProceduralReacher
+
+    table = {"reacher": ProceduralReacher, "Kinetix-Reacher-v1": ProceduralReacher}
+    if name not in table:
+        raise ValueError(f"unknown kinetix-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
 def _gymnasium(name: str):
     # the gymnasium suite is factory-only (stateful CPU envs for Sebulba,
     # reference utils/env_factory.py:71-86); Anakin's make_single cannot
@@ -253,6 +262,7 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "xland": _xland,
     "navix": _xland,
     "mujoco_playground": _playground,
+    "kinetix": _kinetix,
     "craftax": _craftax,
     "popjym": _popjym,
     "debug": _debug,

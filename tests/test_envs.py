@@ -1111,3 +1111,39 @@ def test_doorkey_rules():
     ts = env.step(right)  # col 7 == goal
     assert torch.all(ts.discount == 0.0)
     assert torch.all(ts.reward > 0.9)  # 7 steps of 200 -> ~0.97
+
+
+def test_procedural_reacher_rules():
+    """kinetix-class reacher: forward kinematics, per-episode procedural
+    link lengths exposed in the obs, touch bonus + termination."""
+    import math
+
+    import torch
+
+    from stoix_amd.envs.reacher import GOAL_R, ProceduralReacher
+
+    env = ProceduralReacher(4, seed=0)
+    ts = env.reset()
+    assert ts.observation.shape == (4, 12)
+    # obs carries the procedural lengths
+    torch.testing.assert_close(ts.observation[:, 10:12], env._state["len"])
+    # forward kinematics: straight arm along +x
+    env._state["q"][:] = 0.0
+    env._state["dq"][:] = 0.0
+    tip = env._tip(env._state["q"], env._state["len"])
+    torch.testing.assert_close(tip[:, 0], env._state["len"].sum(-1))
+    torch.testing.assert_close(tip[:, 1], torch.zeros(4), atol=1e-6, rtol=0)
+    # place the goal AT the tip: immediate touch bonus + termination
+    env._state["goal"] = tip.clone()
+    ts = env.step(torch.zeros(4, 2))
+    assert torch.all(ts.reward > 4.0)  # +5 bonus minus tiny drift distance
+    assert torch.all(ts.discount == 0.0)
+    # far goal: negative distance shaping, no termination
+    env.reset()
+    env._state["q"][:] = 0.0
+    env._state["dq"][:] = 0.0
+    env._state["goal"][:, 0] = -env._state["len"].sum(-1)
+    env._state["goal"][:, 1] = 0.0
+    ts = env.step(torch.zeros(4, 2))
+    assert torch.all(ts.reward < -2 * GOAL_R)
+    assert torch.all(ts.discount == 1.0)
