@@ -111,7 +111,6 @@ class Crafting(StatefulVecEnv):
         agent = state["agent"].long()
         facing = state["facing"].long().clamp(0, 3)
         grid = state["grid"].long().clone()
-        bidx = torch.arange(n, device=dev)
 
         move = a < 4
         dirn = torch.where(move, a, facing)

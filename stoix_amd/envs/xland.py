@@ -174,7 +174,6 @@ class DoorKeyGrid(StatefulVecEnv):
         wall_col = torch.randint(3, 6, (n,), device=dev, generator=self.gen)
         door_row = torch.randint(1, N - 1, (n,), device=dev, generator=self.gen)
         cols = torch.arange(N, device=dev)
-        rows = torch.arange(N, device=dev)
         walls = torch.zeros(n, N, N, dtype=torch.bool, device=dev)
         walls[:, 0, :] = walls[:, -1, :] = True
         walls[:, :, 0] = walls[:, :, -1] = True
