@@ -237,6 +237,15 @@ def _kinetix(name: str):
     return table[name]
 
 
+def _jaxarc(name: str):
+    from stoix_amd.envs.arc import GridCopy
+
+    table = {"grid_copy": GridCopy, "Arc-GridCopy-v1": GridCopy}
+    if name not in table:
+        raise ValueError(f"unknown jaxarc-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
 def _gymnasium(name: str):
     # the gymnasium suite is factory-only (stateful CPU envs for Sebulba,
     # reference utils/env_factory.py:71-86); Anakin's make_single cannot
@@ -263,6 +272,7 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "navix": _xland,
     "mujoco_playground": _playground,
     "kinetix": _kinetix,
+    "jaxarc": _jaxarc,
     "craftax": _craftax,
     "popjym": _popjym,
     "debug": _debug,

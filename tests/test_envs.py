@@ -1147,3 +1147,46 @@ def test_procedural_reacher_rules():
     ts = env.step(torch.zeros(4, 2))
     assert torch.all(ts.reward < -2 * GOAL_R)
     assert torch.all(ts.discount == 1.0)
+
+
+def test_grid_copy_rules_and_no_repaint_farming():
+    """jaxarc-class GridCopy: first-time-correct cells pay +1 exactly ONCE
+    per episode (regression for a repaint-cycle exploit a PPO probe found:
+    correct -> overwrite -> repaint farmed +0.95/cycle, return 66 on a
+    10-cell board); wrong paints cost -0.05; completion terminates."""
+    import torch
+
+    from stoix_amd.envs.arc import G, GridCopy
+
+    env = GridCopy(2, seed=0)
+    env.reset()
+    s = env._state
+    # deterministic scene: one coloured target cell under the cursor
+    s["target"][:] = 0.0
+    s["canvas"][:] = 0.0
+    s["rewarded"][:] = 0.0
+    pos = (G // 2) * G + G // 2
+    s["cursor"][:] = float(pos)
+    s["target"][:, pos] = 2.0  # colour 2
+    s["target"][:, 0] = 1.0  # a second cell so completion needs both
+
+    paint2 = torch.full((2,), 5, dtype=torch.long)  # paint colour 2
+    paint1 = torch.full((2,), 4, dtype=torch.long)  # paint colour 1
+
+    ts = env.step(paint2)  # correct first time: +1
+    assert torch.all(ts.reward == 1.0)
+    assert torch.all(ts.discount == 1.0)  # cell 0 still missing
+    ts = env.step(paint1)  # overwrite with wrong colour: -0.05
+    assert torch.allclose(ts.reward, torch.full((2,), -0.05))
+    ts = env.step(paint2)  # repaint correct: NO second payment
+    assert torch.all(ts.reward == 0.0)
+    # finish the board: walk to (0,0) and paint colour 1
+    up = torch.zeros(2, dtype=torch.long)
+    left = torch.full((2,), 3, dtype=torch.long)
+    for _ in range(G // 2):
+        env.step(up)
+    for _ in range(G // 2):
+        env.step(left)
+    ts = env.step(paint1)
+    assert torch.all(ts.reward == 1.0)
+    assert torch.all(ts.discount == 0.0)  # complete -> terminated
