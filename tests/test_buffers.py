@@ -126,3 +126,54 @@ def test_prioritised_add_revalidates_block_boundary_windows():
     assert (leaves[:, 6:8] == 0).all()
     # unwritten region stays unsampleable
     assert (leaves[:, 8:] == 0).all()
+
+
+def test_sum_tree_random_ops_match_naive_reference():
+    """Property test: arbitrary interleavings of batched set() calls keep
+    the tree consistent with a naive array-of-priorities reference —
+    root == sum, every internal node == sum of children, and stratified
+    samples only land on positive-priority items."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    @settings(max_examples=30, deadline=None)
+    @given(
+        st.lists(
+            st.lists(
+                st.tuples(st.integers(0, 99), st.floats(0.0, 10.0)),
+                min_size=1, max_size=16,
+            ),
+            min_size=1, max_size=8,
+        ),
+        st.randoms(use_true_random=False),
+    )
+    def run(batches, rnd):
+        tree = SumTree(100)
+        ref = torch.zeros(100)
+        for batch in batches:
+            idx = torch.tensor([i for i, _ in batch], dtype=torch.long)
+            pr = torch.tensor([p for _, p in batch])
+            # duplicate indices: LAST write wins in the reference; the
+            # tree's contract is "an arbitrary scatter winner among the
+            # batch" — feed deduped batches to compare exactly
+            seen = {}
+            for i, p in batch:
+                seen[i] = p
+            idx = torch.tensor(list(seen.keys()), dtype=torch.long)
+            pr = torch.tensor(list(seen.values()))
+            tree.set(idx, pr)
+            ref[idx] = pr
+            # invariant: every internal node equals its children's sum
+            t = tree.tree
+            cap = tree.capacity
+            torch.testing.assert_close(t[cap : cap + 100], ref, rtol=1e-5, atol=1e-5)
+            for node in range(1, cap):
+                torch.testing.assert_close(
+                    t[node], t[2 * node] + t[2 * node + 1], rtol=1e-4, atol=1e-4
+                )
+            if float(ref.sum()) > 0:
+                g = torch.Generator().manual_seed(rnd.randrange(2**31))
+                s = tree.sample(64, g)
+                assert (ref[s] > 0).all(), "sampled a zero-priority item"
+
+    run()
