@@ -359,3 +359,27 @@ def test_tanh_normal_boundary_log_prob_bounded():
     d3 = AffineTanhTransformedDistribution(loc3, torch.full((1, 1), 0.1), -1.0, 1.0)
     d3.log_prob(torch.tensor([[1.0]])).backward()
     assert torch.isfinite(loc3.grad).all() and float(loc3.grad.abs()) > 0
+
+
+def test_tanh_normal_density_normalises_to_one():
+    """The tanh-normal density (with the CDF-mass boundary branches)
+    integrates to ~1 over the action interval — including regimes where
+    most probability mass sits in the boundary strips."""
+    import torch
+
+    from stoix_amd.networks.distributions import AffineTanhTransformedDistribution
+
+    for loc_v, scale_v in [(0.0, 0.6), (1.5, 0.3), (6.0, 0.5), (-40.0, 1e-3)]:
+        d = AffineTanhTransformedDistribution(
+            torch.tensor([[loc_v]]), torch.tensor([[scale_v]]), -1.0, 1.0
+        )
+        eps = 1e-3
+        # interior trapezoid on [-1+eps, 1-eps]
+        xs = torch.linspace(-1 + eps, 1 - eps, 20001).view(-1, 1)
+        dens = d.log_prob(xs).exp()
+        interior = torch.trapz(dens.view(-1), xs.view(-1))
+        # boundary strips carry mass = avg_density * eps
+        left = d.log_prob(torch.tensor([[-1.0]])).exp() * eps
+        right = d.log_prob(torch.tensor([[1.0]])).exp() * eps
+        total = float(interior + left + right)
+        assert abs(total - 1.0) < 0.02, (loc_v, scale_v, total)
