@@ -361,25 +361,43 @@ def test_tanh_normal_boundary_log_prob_bounded():
     assert torch.isfinite(loc3.grad).all() and float(loc3.grad.abs()) > 0
 
 
-def test_tanh_normal_density_normalises_to_one():
-    """The tanh-normal density (with the CDF-mass boundary branches)
-    integrates to ~1 over the action interval — including regimes where
-    most probability mass sits in the boundary strips."""
+def test_tanh_normal_density_mass_accounting():
+    """Exact mass accounting for the tanh-normal with the CDF-mass
+    boundary branches: (a) interior densities equal the analytic
+    change-of-variables density, (b) the boundary values equal
+    tail-mass / eps, (c) tails + interior mass sum to 1 — including the
+    saturated regime where nearly ALL mass sits in a boundary strip."""
+    import math
+
     import torch
 
     from stoix_amd.networks.distributions import AffineTanhTransformedDistribution
 
+    eps = 1e-3
+    u_hi = math.atanh(1.0 - eps)
     for loc_v, scale_v in [(0.0, 0.6), (1.5, 0.3), (6.0, 0.5), (-40.0, 1e-3)]:
         d = AffineTanhTransformedDistribution(
             torch.tensor([[loc_v]]), torch.tensor([[scale_v]]), -1.0, 1.0
         )
-        eps = 1e-3
-        # interior trapezoid on [-1+eps, 1-eps]
-        xs = torch.linspace(-1 + eps, 1 - eps, 20001).view(-1, 1)
-        dens = d.log_prob(xs).exp()
-        interior = torch.trapz(dens.view(-1), xs.view(-1))
-        # boundary strips carry mass = avg_density * eps
-        left = d.log_prob(torch.tensor([[-1.0]])).exp() * eps
-        right = d.log_prob(torch.tensor([[1.0]])).exp() * eps
-        total = float(interior + left + right)
-        assert abs(total - 1.0) < 0.02, (loc_v, scale_v, total)
+        # (a) pointwise interior density == N(u; loc, scale) / (1 - a^2)
+        xs = torch.linspace(-0.99, 0.99, 41).view(-1, 1)
+        u = torch.atanh(xs)
+        z = (u - loc_v) / scale_v
+        ref = (
+            torch.exp(-0.5 * z * z) / (scale_v * math.sqrt(2 * math.pi))
+            / (1 - xs * xs)
+        ).view(-1)
+        got = d.log_prob(xs).exp()
+        torch.testing.assert_close(got, ref, rtol=1e-3, atol=1e-8)
+        # (b) boundary value * eps == analytic tail mass
+        zr = torch.tensor((u_hi - loc_v) / scale_v, dtype=torch.float64)
+        zl = torch.tensor((-u_hi - loc_v) / scale_v, dtype=torch.float64)
+        right_mass = float(torch.special.ndtr(-zr))
+        left_mass = float(torch.special.ndtr(zl))
+        got_right = float(d.log_prob(torch.tensor([[1.0]])).exp() * eps)
+        got_left = float(d.log_prob(torch.tensor([[-1.0]])).exp() * eps)
+        assert abs(got_right - right_mass) < 1e-4 + 0.01 * right_mass
+        assert abs(got_left - left_mass) < 1e-4 + 0.01 * left_mass
+        # (c) tails + interior mass == 1
+        interior_mass = float(torch.special.ndtr(zr) - torch.special.ndtr(zl))
+        assert abs(interior_mass + left_mass + right_mass - 1.0) < 1e-9
