@@ -74,3 +74,20 @@ def test_bench_world1_single_process():
     assert out.returncode == 0, out.stderr
     rec = json.loads([l for l in out.stdout.splitlines() if l.startswith("{")][0])
     assert rec["n_gpus"] == 1 and rec["value"] > 0
+
+
+@pytest.mark.slow
+def test_bench_world2_sac_and_rainbow_builders():
+    """The non-flagship bench builders also honour the torchrun world=2
+    contract (device-pinned learners, barrier+MAX timing, one JSON line)."""
+    rec = _run_bench(
+        ["--config=sac", "--steps=2", "--warmup=1", "--num-envs=8"],
+        timeout=600,
+    )
+    assert rec["n_gpus"] == 2 and rec["value"] > 0
+    assert rec["config"]["parallelism"] == "dp2"
+    rec = _run_bench(
+        ["--config=rainbow", "--steps=2", "--warmup=1", "--num-envs=8"],
+        timeout=600,
+    )
+    assert rec["n_gpus"] == 2 and rec["value"] > 0
