@@ -224,3 +224,54 @@ def test_gae_matches_bruteforce_property():
                 assert abs(acc + float(v[t, b]) - float(tgt[t, b])) < 1e-4
 
     run()
+
+
+def test_vtrace_matches_bruteforce_property():
+    """Property test: the v-trace recurrence equals the paper's closed form
+    (Espeholt et al. 2018, eq. 1) evaluated brute-force on random instances:
+
+      vs_s = V_s + sum_{t>=s} (prod_{i=s}^{t-1} gamma_i c_i) * delta_t,
+      delta_t = clip(rho_t) (r_t + gamma_t V_{t+1} - V_t),
+      c_i = lambda * clip(rho_i, 1),
+
+    and pg_adv_t = clip(rho_t, pg_thresh) (r_t + gamma_t vs_{t+1} - V_t).
+    """
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from stoix_amd.ops import multistep as ms
+
+    @settings(max_examples=25, deadline=None)
+    @given(st.integers(0, 10_000))
+    def run(seed):
+        g = torch.Generator().manual_seed(seed)
+        T, B = int(torch.randint(2, 12, (1,), generator=g)), 3
+        r = torch.randn(T, B, generator=g)
+        d = (torch.rand(T, B, generator=g) > 0.2).float() * 0.95
+        v_tm1 = torch.randn(T, B, generator=g)
+        v_t = torch.randn(T, B, generator=g)
+        # off-policy: log-ratio noise gives rhos straddling the clip at 1
+        rho = torch.exp(0.7 * torch.randn(T, B, generator=g))
+        lam = float(torch.rand(1, generator=g))
+        errors, pg_adv, q_est = ms.vtrace_td_error_and_advantage(
+            v_tm1, v_t, r, d, rho, lambda_=lam
+        )
+        delta = rho.clamp(max=1.0) * (r + d * v_t - v_tm1)
+        c = lam * rho.clamp(max=1.0)
+        vs = torch.empty(T, B)
+        for b in range(B):
+            for s in range(T):
+                acc, coef = 0.0, 1.0
+                for t in range(s, T):
+                    acc += coef * float(delta[t, b])
+                    coef *= float(d[t, b]) * float(c[t, b])
+                vs[s, b] = float(v_tm1[s, b]) + acc
+        torch.testing.assert_close(errors, vs - v_tm1, rtol=1e-4, atol=1e-5)
+        vs_t = torch.cat([vs[1:], v_t[-1:]], dim=0)
+        expect_q = r + d * vs_t
+        torch.testing.assert_close(q_est, expect_q, rtol=1e-4, atol=1e-5)
+        torch.testing.assert_close(
+            pg_adv, rho.clamp(max=1.0) * (expect_q - v_tm1), rtol=1e-4, atol=1e-5
+        )
+
+    run()
