@@ -177,3 +177,50 @@ def test_sum_tree_random_ops_match_naive_reference():
                 assert (ref[s] > 0).all(), "sampled a zero-priority item"
 
     run()
+
+
+def test_trajectory_windows_valid_under_random_adds_property():
+    """Property test: after ANY sequence of variable-sized adds (wrapping the
+    circular time axis arbitrarily), every sampled window is (a) contiguous
+    in global time, (b) entirely inside the currently-valid region (the last
+    t_filled written steps — never straddling the write pointer), and (c)
+    period-aligned relative to the oldest valid step."""
+    from hypothesis import given, settings
+    from hypothesis import strategies as st
+
+    from stoix_amd.buffers.trajectory import TrajectoryBuffer
+
+    @settings(max_examples=20, deadline=None)
+    @given(st.integers(0, 10_000))
+    def run(seed):
+        g = torch.Generator().manual_seed(seed)
+        rows, t_max, seq = 4, 16, 5
+        period = int(torch.randint(1, 4, (1,), generator=g))
+        buf = TrajectoryBuffer(rows, t_max, seq, device="cpu", seed=seed, period=period)
+        glob = 0  # global step counter; cell value = row*10_000 + global_t
+        for _ in range(int(torch.randint(3, 10, (1,), generator=g))):
+            t_block = int(torch.randint(1, t_max, (1,), generator=g))
+            vals = (
+                torch.arange(rows).unsqueeze(1) * 10_000
+                + (glob + torch.arange(t_block)).unsqueeze(0)
+            ).float()
+            buf.add({"x": vals})
+            glob += t_block
+            if not buf.can_sample:
+                continue
+            out = buf.sample(64)
+            x = out["x"]  # [64, seq]
+            row_id = (x[:, 0] // 10_000).long()
+            gt = x - row_id.unsqueeze(1).float() * 10_000
+            # (a) contiguity
+            assert torch.all(gt[:, 1:] - gt[:, :-1] == 1.0), seed
+            # row consistency across the window
+            assert torch.all((x // 10_000) == row_id.unsqueeze(1)), seed
+            # (b) inside the valid region
+            oldest = glob - buf.t_filled
+            assert torch.all(gt[:, 0] >= oldest), (seed, oldest, gt[:, 0])
+            assert torch.all(gt[:, -1] <= glob - 1), seed
+            # (c) period alignment of starts
+            assert torch.all((gt[:, 0] - oldest) % period == 0), seed
+
+    run()
