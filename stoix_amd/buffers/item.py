@@ -9,7 +9,7 @@ ops (index copies / gathers), graph-capturable.
 """
 from __future__ import annotations
 
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 

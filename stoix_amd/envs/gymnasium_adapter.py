@@ -17,7 +17,6 @@ duck-typed (tested against a mock vector env in tests/test_sebulba.py) and
 from __future__ import annotations
 
 import threading
-from typing import Any, Dict, Optional, Tuple
 
 import numpy as np
 import torch

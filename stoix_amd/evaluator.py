@@ -16,7 +16,6 @@ from typing import Callable, Dict, Optional
 import torch
 
 from stoix_amd.envs.env import StatefulVecEnv
-from stoix_amd.types import EvalOutput
 
 
 @torch.no_grad()

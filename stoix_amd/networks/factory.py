@@ -18,7 +18,7 @@ from stoix_amd.networks import heads as H
 from stoix_amd.networks import torso as T
 from stoix_amd.networks.base import FeedForwardActor, FeedForwardCritic, RecurrentActor, RecurrentCritic, ScannedRNN
 from stoix_amd.networks.dueling import DistributionalDuelingQNetwork, DuelingQNetwork
-from stoix_amd.networks.inputs import ArrayInput, EmbeddingActionInput, EmbeddingActionOnehotInput
+from stoix_amd.networks.inputs import EmbeddingActionInput, EmbeddingActionOnehotInput
 from stoix_amd.networks.resnet import ResNetTorso, VisualResNetTorso
 
 TORSOS = {

@@ -18,7 +18,7 @@ learner's optimiser step; rank 0 logs/evaluates.
 from __future__ import annotations
 
 import time
-from typing import Callable, Dict, Optional
+from typing import Callable, Dict
 
 import torch
 

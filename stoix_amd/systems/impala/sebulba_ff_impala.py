@@ -19,7 +19,6 @@ import torch.nn as nn
 
 from stoix_amd import envs as environments
 from stoix_amd.config import compose
-from stoix_amd.envs.env import get_final_step_metrics
 from stoix_amd.networks.factory import build_actor, build_critic
 from stoix_amd.ops import multistep
 from stoix_amd.parallel.dist import FlatGradReducer, broadcast_module, get_dist_context

@@ -18,7 +18,7 @@ across GPUs with one fused flat RCCL all-reduce per minibatch
 from __future__ import annotations
 
 import sys
-from typing import Dict, Optional
+from typing import Dict
 
 import torch
 import torch.nn as nn

@@ -25,7 +25,7 @@ from __future__ import annotations
 
 import math
 import sys
-from typing import Dict, Tuple
+from typing import Dict
 
 import torch
 import torch.nn as nn

@@ -14,7 +14,7 @@ import os
 import threading
 import time
 from enum import Enum
-from typing import Any, Dict, Optional
+from typing import Any, Dict
 
 import numpy as np
 import torch
