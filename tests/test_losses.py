@@ -208,3 +208,40 @@ def test_quantile_regression_loss_matches_scalar_reference():
     for huber in (1.0, 0.7, 0.0):
         out = L.quantile_regression_loss(src, taus, tgt, huber_param=huber)
         assert abs(out.item() - scalar(huber)) < 1e-5, huber
+
+
+def test_munchausen_matches_scalar_reference():
+    """Golden test: the Munchausen-DQN loss equals an explicit scalar
+    computation (softmax policies via scipy logsumexp, clipped log-policy
+    bonus, soft value target) on random instances."""
+    from scipy.special import logsumexp
+
+    g = torch.Generator().manual_seed(3)
+    B, A = 5, 4
+    q_tm1 = torch.randn(B, A, generator=g)
+    q_target_tm1 = torch.randn(B, A, generator=g)
+    q_t = torch.randn(B, A, generator=g)
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.randn(B, generator=g)
+    d = torch.full((B,), 0.9)
+    tau, alpha, clip_min = 0.05, 0.9, -1.0
+
+    total = 0.0
+    for b in range(B):
+        lt = (q_target_tm1[b] / tau).numpy()
+        log_pi_tm1 = lt - logsumexp(lt)
+        bonus = alpha * min(max(tau * log_pi_tm1[int(a[b])], clip_min), 0.0)
+        ltt = (q_t[b] / tau).numpy()
+        log_pi_t = ltt - logsumexp(ltt)
+        import numpy as np
+
+        pi_t = np.exp(log_pi_t)
+        soft_v = float((pi_t * (q_t[b].numpy() - tau * log_pi_t)).sum())
+        target = float(r[b]) + bonus + 0.9 * soft_v
+        td = target - float(q_tm1[b, int(a[b])])
+        total += 0.5 * td * td
+    expect = total / B
+    out = L.munchausen_q_learning(
+        q_tm1, a, r, d, q_t, q_target_tm1, tau, alpha, clip_min
+    )
+    assert abs(out.item() - expect) < 1e-4
