@@ -245,3 +245,46 @@ def test_munchausen_matches_scalar_reference():
         q_tm1, a, r, d, q_t, q_target_tm1, tau, alpha, clip_min
     )
     assert abs(out.item() - expect) < 1e-4
+
+
+def test_categorical_double_q_learning_scalar_reference():
+    """Golden test: the C51 double-Q loss equals a scalar reconstruction —
+    pick argmax action from the selector, shift target atoms by r + d*z,
+    two-bin project by hand, cross-entropy against the taken action's
+    log-softmax."""
+    import numpy as np
+    from scipy.special import log_softmax, softmax
+
+    g = torch.Generator().manual_seed(9)
+    B, A, N = 4, 3, 7
+    atoms = torch.linspace(-2.0, 2.0, N)
+    logits_tm1 = torch.randn(B, A, N, generator=g)
+    logits_t = torch.randn(B, A, N, generator=g)
+    sel = torch.randn(B, A, generator=g)
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.randn(B, generator=g) * 0.5
+    d = torch.full((B,), 0.9)
+
+    vmin, vmax = -2.0, 2.0
+    dz = (vmax - vmin) / (N - 1)
+    total = 0.0
+    for b in range(B):
+        best = int(sel[b].argmax())
+        p_best = softmax(logits_t[b, best].numpy())
+        proj = np.zeros(N)
+        for n in range(N):
+            z = float(r[b]) + 0.9 * float(atoms[n])
+            z = min(max(z, vmin), vmax)
+            pos = (z - vmin) / dz
+            lo = int(pos // 1)
+            hi = min(lo + 1, N - 1)
+            frac = pos - lo
+            proj[lo] += p_best[n] * (1 - frac)
+            proj[hi] += p_best[n] * frac
+        logp = log_softmax(logits_tm1[b, int(a[b])].numpy())
+        total += -(proj * logp).sum()
+    expect = total / B
+    out = L.categorical_double_q_learning(
+        logits_tm1, atoms, a, r, d, logits_t, atoms, sel
+    )
+    assert abs(out.item() - expect) < 1e-4
