@@ -1190,3 +1190,36 @@ def test_grid_copy_rules_and_no_repaint_farming():
     ts = env.step(paint1)
     assert torch.all(ts.reward == 1.0)
     assert torch.all(ts.discount == 0.0)  # complete -> terminated
+
+
+def test_every_env_config_composes_and_steps():
+    """Every shipped env yaml composes with the default config and builds a
+    steppable vec-env on CPU (2 envs, 3 random steps) — except suites whose
+    external dependency is a documented offline gate (gymnasium)."""
+    from pathlib import Path
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import CONFIG_ROOT, compose
+
+    GATED_SUITES = {"gymnasium"}  # lazy-imports the gymnasium package
+    yamls = sorted((CONFIG_ROOT / "env").rglob("*.yaml"))
+    assert len(yamls) >= 30
+    checked = 0
+    for y in yamls:
+        rel = y.relative_to(CONFIG_ROOT / "env").with_suffix("")
+        cfg = compose(
+            "default/anakin/default_ff_ppo.yaml", [f"env={rel.as_posix()}"]
+        )
+        if cfg.env.env_name in GATED_SUITES:
+            with pytest.raises(Exception, match="gymnasium"):
+                environments.make_single(cfg, 2, "cpu", seed=0)
+            continue
+        env = environments.make_single(cfg, 2, "cpu", seed=0)
+        ts = env.reset()
+        act_space = env.action_space
+        g = torch.Generator().manual_seed(0)
+        for _ in range(3):
+            ts = env.step(act_space.sample(2, "cpu", generator=g))
+        assert ts.observation is not None
+        checked += 1
+    assert checked >= 30
