@@ -271,3 +271,50 @@ def test_pipeline_stress_no_loss_no_deadlock():
     for t in blocked:
         t.join(timeout=5)
         assert not t.is_alive(), "producer did not drain on shutdown"
+
+
+def test_native_pool_concurrent_actor_threads_isolated():
+    """Race gate for the native C++ pool (SURVEY §5.2 posture): 6 actor
+    threads each drive their OWN pool instance concurrently (the Sebulba
+    deployment shape — the engine releases the GIL and steps serially per
+    call, threads are the parallelism). Any shared mutable C++ state would
+    corrupt per-thread determinism: each thread replays the same seed twice
+    and must get bit-identical observation streams both times, with and
+    without the other threads hammering in parallel."""
+    import threading
+
+    from stoix_amd.envs.envpool_cpu import BreakoutCpu, envpool_ext
+
+    if envpool_ext() is None:
+        pytest.skip("native pool extension not built")
+
+    def stream(seed, steps=40):
+        env = BreakoutCpu(4, device="cpu", seed=seed)
+        ts = env.reset()
+        g = torch.Generator().manual_seed(seed)
+        acc = [ts.observation.sum().item()]
+        for _ in range(steps):
+            a = torch.randint(0, env.action_space.num_values, (4,), generator=g)
+            ts = env.step(a)
+            acc.append(ts.observation.sum().item())
+        return acc
+
+    solo = {s: stream(s) for s in range(6)}
+    results = {}
+    errs = []
+
+    def worker(seed):
+        try:
+            results[seed] = stream(seed)
+        except Exception as e:  # pragma: no cover
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker, args=(s,)) for s in range(6)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=120)
+    assert not errs
+    assert set(results) == set(range(6))
+    for s in range(6):
+        assert results[s] == solo[s], f"seed {s} diverged under concurrency"
