@@ -189,7 +189,14 @@ def _craftax(name: str):
 
 def _popjym(name: str):
     from stoix_amd.envs.pomdp import (
+        AutoEncodeEasy,
+        AutoEncodeMedium,
+        CountRecallEasy,
+        CountRecallMedium,
         NoisyStatelessCartPole,
+        RepeatFirstEasy,
+        RepeatFirstHard,
+        RepeatFirstMedium,
         StatelessCartPole,
         StatelessPendulum,
     )
@@ -197,13 +204,49 @@ def _popjym(name: str):
     table = {
         "stateless_cartpole": StatelessCartPole,
         "StatelessCartPole": StatelessCartPole,
+        # the reference's popjym difficulty-suffixed scenario names
+        # (configs/env/popjym/*.yaml: StatelessCartPoleEasy, AutoencodeEasy,
+        # CountRecallEasy/Medium, RepeatFirstEasy/Medium/Hard)
+        "stateless_cartpole_easy": StatelessCartPole,
+        "StatelessCartPoleEasy": StatelessCartPole,
         "noisy_stateless_cartpole": NoisyStatelessCartPole,
         "NoisyStatelessCartPole": NoisyStatelessCartPole,
         "stateless_pendulum": StatelessPendulum,
         "StatelessPendulum": StatelessPendulum,
+        "auto_encode_easy": AutoEncodeEasy,
+        "AutoencodeEasy": AutoEncodeEasy,
+        "auto_encode_medium": AutoEncodeMedium,
+        "AutoencodeMedium": AutoEncodeMedium,
+        "count_recall_easy": CountRecallEasy,
+        "CountRecallEasy": CountRecallEasy,
+        "count_recall_medium": CountRecallMedium,
+        "CountRecallMedium": CountRecallMedium,
+        "repeat_first_easy": RepeatFirstEasy,
+        "RepeatFirstEasy": RepeatFirstEasy,
+        "repeat_first_medium": RepeatFirstMedium,
+        "RepeatFirstMedium": RepeatFirstMedium,
+        "repeat_first_hard": RepeatFirstHard,
+        "RepeatFirstHard": RepeatFirstHard,
     }
     if name not in table:
         raise ValueError(f"unknown popjym-suite env '{name}' (have {list(table)})")
+    return table[name]
+
+
+def _popgym_arcade(name: str):
+    # popgym_arcade suite (reference configs/env/popgym_arcade/
+    # noisy_cartpole.yaml: NoisyCartPoleEasy with partial_obs=True) — the
+    # partially-observable noisy cartpole maps to the same capability class.
+    from stoix_amd.envs.pomdp import NoisyStatelessCartPole
+
+    table = {
+        "noisy_cartpole": NoisyStatelessCartPole,
+        "NoisyCartPoleEasy": NoisyStatelessCartPole,
+    }
+    if name not in table:
+        raise ValueError(
+            f"unknown popgym_arcade-suite env '{name}' (have {list(table)})"
+        )
     return table[name]
 
 
@@ -275,6 +318,7 @@ ENV_REGISTRY: Dict[str, Callable] = {
     "jaxarc": _jaxarc,
     "craftax": _craftax,
     "popjym": _popjym,
+    "popgym_arcade": _popgym_arcade,
     "debug": _debug,
 }
 

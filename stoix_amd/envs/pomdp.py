@@ -20,7 +20,7 @@ import torch
 
 from stoix_amd.envs.classic import CartPole, Pendulum
 from stoix_amd.envs.env import State, StatefulVecEnv, Tensor
-from stoix_amd.envs.spaces import BoxSpace
+from stoix_amd.envs.spaces import BoxSpace, DiscreteSpace
 
 
 class _MaskedObsEnv(StatefulVecEnv):
@@ -143,3 +143,209 @@ class StatelessPendulum(_MaskedObsEnv):
             reward,
             terminated,
         )
+
+
+class _MemoryGameEnv(StatefulVecEnv):
+    """Base for POPGym-class pure memory games (AutoEncode / CountRecall /
+    RepeatFirst — the reference's popjym scenario families beyond the
+    stateless-control tasks, configs/env/popjym/*.yaml). Original vectorised
+    designs of the same capability class: episodic symbol tasks where the
+    score is the fraction of correct recalls, rewards ±1/n_scored (POPGym's
+    normalised-return convention), and the observation carries the
+    reference wrapper's start-flag + previous-action augmentation
+    (make_env.py:363-364)."""
+
+    capture_safe = True
+
+    def _prev_a_onehot(self, state: State) -> Tensor:
+        n_act = self.action_space.num_values
+        return torch.nn.functional.one_hot(
+            state["_prev_a"].clamp(0, n_act - 1), n_act
+        ).float() * (state["t"] > 0).unsqueeze(-1).float()
+
+
+class RepeatFirst(_MemoryGameEnv):
+    """Observe a symbol at t=0; repeat it at every later step.
+
+    Obs: [symbol one-hot (zeros after t=0), start flag, time fraction,
+    prev-action one-hot]. Reward ±1/(L-1) per answer step.
+    """
+
+    L = 16
+    A = 4
+    solved_return_threshold = 0.8
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.max_episode_steps = self.L + 1
+        self.action_space = DiscreteSpace(self.A)
+        self.observation_space = BoxSpace((self.A + 2 + self.A,), 0.0, 1.0)
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "target": torch.randint(0, self.A, (n,), device=dev, generator=self.gen),
+            "t": torch.zeros(n, dtype=torch.long, device=dev),
+            "_prev_a": torch.zeros(n, dtype=torch.long, device=dev),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        show = (state["t"] == 0).unsqueeze(-1).float()
+        sym = torch.nn.functional.one_hot(state["target"], self.A).float() * show
+        tf = (state["t"].float() / self.L).unsqueeze(-1)
+        return torch.cat([sym, show, tf, self._prev_a_onehot(state)], dim=-1)
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long().clamp(0, self.A - 1)
+        scored = state["t"] >= 1
+        correct = (a == state["target"]) & scored
+        reward = (correct.float() - (scored & ~correct).float()) / float(self.L - 1)
+        t = state["t"] + 1
+        return (
+            {"target": state["target"], "t": t, "_prev_a": a},
+            reward,
+            t >= self.L,
+        )
+
+
+class RepeatFirstEasy(RepeatFirst):
+    L, A = 16, 4
+
+
+class RepeatFirstMedium(RepeatFirst):
+    L, A = 48, 8
+
+
+class RepeatFirstHard(RepeatFirst):
+    L, A = 104, 16
+
+
+class AutoEncode(_MemoryGameEnv):
+    """Watch a symbol sequence (t = 0..L-1), then reproduce it in order
+    during the play phase (t = L..2L-1).
+
+    Obs: [symbol one-hot (watch phase only), play-phase flag, start flag,
+    time fraction, prev-action one-hot]. Reward ±1/L per play answer; the
+    play-step obs carries no symbol, so the whole sequence must be held in
+    memory.
+    """
+
+    L = 6
+    A = 4
+    solved_return_threshold = 0.8
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.max_episode_steps = 2 * self.L + 1
+        self.action_space = DiscreteSpace(self.A)
+        self.observation_space = BoxSpace((self.A + 3 + self.A,), 0.0, 1.0)
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        return {
+            "seq": torch.randint(0, self.A, (n, self.L), device=dev, generator=self.gen),
+            "t": torch.zeros(n, dtype=torch.long, device=dev),
+            "_prev_a": torch.zeros(n, dtype=torch.long, device=dev),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        t = state["t"]
+        watch = (t < self.L).unsqueeze(-1).float()
+        idx = t.clamp(max=self.L - 1)
+        cur = state["seq"].gather(1, idx.unsqueeze(1)).squeeze(1)
+        sym = torch.nn.functional.one_hot(cur, self.A).float() * watch
+        play = 1.0 - watch
+        start = (t == 0).unsqueeze(-1).float()
+        tf = (t.float() / (2 * self.L)).unsqueeze(-1)
+        return torch.cat([sym, play, start, tf, self._prev_a_onehot(state)], dim=-1)
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long().clamp(0, self.A - 1)
+        t = state["t"]
+        scored = t >= self.L
+        ans_idx = (t - self.L).clamp(0, self.L - 1)
+        expect = state["seq"].gather(1, ans_idx.unsqueeze(1)).squeeze(1)
+        correct = (a == expect) & scored
+        reward = (correct.float() - (scored & ~correct).float()) / float(self.L)
+        t = t + 1
+        return (
+            {"seq": state["seq"], "t": t, "_prev_a": a},
+            reward,
+            t >= 2 * self.L,
+        )
+
+
+class AutoEncodeEasy(AutoEncode):
+    L, A = 6, 4
+
+
+class AutoEncodeMedium(AutoEncode):
+    L, A = 12, 6
+
+
+class CountRecall(_MemoryGameEnv):
+    """Each step shows a value symbol and a query symbol; answer how many
+    times the query value has occurred so far (including this step's
+    value). Running counts must be maintained in memory.
+
+    Obs: [value one-hot, query one-hot, start flag, time fraction,
+    prev-action one-hot]. Action space Discrete(T+1) (the count). Reward
+    ±1/T per answer.
+    """
+
+    T = 16
+    V = 4
+    solved_return_threshold = 0.6
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed)
+        self.max_episode_steps = self.T + 1
+        self.action_space = DiscreteSpace(self.T + 1)
+        self.observation_space = BoxSpace(
+            (2 * self.V + 2 + self.T + 1,), 0.0, 1.0
+        )
+
+    def _reset_fn(self, n: int) -> State:
+        dev = self.device
+        s0 = torch.randint(0, self.V, (n,), device=dev, generator=self.gen)
+        q0 = torch.randint(0, self.V, (n,), device=dev, generator=self.gen)
+        counts = torch.nn.functional.one_hot(s0, self.V).long()
+        return {
+            "counts": counts,
+            "s": s0,
+            "q": q0,
+            "t": torch.zeros(n, dtype=torch.long, device=dev),
+            "_prev_a": torch.zeros(n, dtype=torch.long, device=dev),
+        }
+
+    def _obs_fn(self, state: State) -> Tensor:
+        sv = torch.nn.functional.one_hot(state["s"], self.V).float()
+        qv = torch.nn.functional.one_hot(state["q"], self.V).float()
+        start = (state["t"] == 0).unsqueeze(-1).float()
+        tf = (state["t"].float() / self.T).unsqueeze(-1)
+        return torch.cat([sv, qv, start, tf, self._prev_a_onehot(state)], dim=-1)
+
+    def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
+        a = action.long().clamp(0, self.T)
+        truth = state["counts"].gather(1, state["q"].unsqueeze(1)).squeeze(1)
+        correct = a == truth.clamp(max=self.T)
+        reward = (correct.float() - (~correct).float()) / float(self.T)
+        dev = self.device
+        n = a.shape[0]
+        s = torch.randint(0, self.V, (n,), device=dev, generator=self.gen)
+        q = torch.randint(0, self.V, (n,), device=dev, generator=self.gen)
+        counts = state["counts"] + torch.nn.functional.one_hot(s, self.V).long()
+        t = state["t"] + 1
+        return (
+            {"counts": counts, "s": s, "q": q, "t": t, "_prev_a": a},
+            reward,
+            t >= self.T,
+        )
+
+
+class CountRecallEasy(CountRecall):
+    T, V = 16, 4
+
+
+class CountRecallMedium(CountRecall):
+    T, V = 32, 8
