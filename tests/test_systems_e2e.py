@@ -426,3 +426,62 @@ def test_ppo_learns_grid_copy():
     )
     r = run(cfg)
     assert r > 3.0, f"grid copy not learned: return={r}"
+
+
+def test_rec_ppo_learns_repeat_first():
+    """Recurrent PPO solves the popjym RepeatFirstEasy memory game to the
+    optimal 1.0 return (observe a symbol once, repeat it for 15 steps —
+    measured 1.0 at this budget in ~25 s)."""
+    from stoix_amd.systems.ppo.rec_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_rec_ppo.yaml",
+        [
+            "env=popjym/repeat_first_easy",
+            "arch.total_num_envs=128",
+            "arch.total_timesteps=null",
+            "arch.num_updates=60",
+            "arch.num_evaluation=1",
+            "arch.num_eval_episodes=32",
+            "arch.absolute_metric=false",
+            "arch.seed=3",
+            "system.rollout_length=17",
+            "system.num_minibatches=4",
+            "system.epochs=4",
+            "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r > 0.8, f"rec_ppo failed repeat_first_easy: return={r}"
+
+
+@pytest.mark.slow
+def test_rec_ppo_learns_auto_encode_above_chance():
+    """Recurrent PPO makes clear progress on popjym AutoEncodeEasy (watch a
+    6-symbol sequence, reproduce it blind): random play scores -0.5, and
+    this budget measured 0.24 at 200 updates / 0.68 at 600. Gate well below
+    the measured point to absorb seed noise while still rejecting a
+    memoryless collapse."""
+    from stoix_amd.systems.ppo.rec_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_rec_ppo.yaml",
+        [
+            "env=popjym/auto_encode_easy",
+            "arch.total_num_envs=256",
+            "arch.total_timesteps=null",
+            "arch.num_updates=200",
+            "arch.num_evaluation=1",
+            "arch.num_eval_episodes=32",
+            "arch.absolute_metric=false",
+            "arch.seed=3",
+            "system.rollout_length=13",
+            "system.num_minibatches=4",
+            "system.epochs=4",
+            "logger.loggers=[]",
+            "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r > 0.0, f"rec_ppo below chance-clearing bar on auto_encode: {r}"
