@@ -43,6 +43,8 @@ run stoix_amd.systems.disco_rl.ff_disco103    "" system.num_minibatches=2 system
 run stoix_amd.systems.ppo.ff_ppo              "xland" env=xland_minigrid/goal_grid system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "craftax" env=craftax/crafting system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.rec_ppo             "popjym" env=popjym/stateless_cartpole system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.rec_ppo             "popjym-mem" env=popjym/repeat_first_easy system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.rec_ppo             "arcade" env=popgym_arcade/noisy_cartpole system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "doorkey" env=navix/doorkey system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "swingup" env=mjc_playground/cartpole_swingup system=ppo/ff_ppo_continuous network=mlp_continuous system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "reacher" env=kinetix/reacher system=ppo/ff_ppo_continuous network=mlp_continuous system.num_minibatches=2 system.epochs=1
