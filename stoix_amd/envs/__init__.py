@@ -164,14 +164,27 @@ def _debug(name: str):
 
 
 def _xland(name: str):
-    from stoix_amd.envs.xland import DoorKeyGrid, XLandGrid
+    from stoix_amd.envs.xland import (
+        DoorKeyGrid,
+        DoorKeyGrid5,
+        DoorKeyGrid8,
+        EmptyGrid5,
+        EmptyGrid6,
+        XLandGrid,
+    )
 
     table = {
         "goal_grid": XLandGrid,
         "XLand-MiniGrid-R1": XLandGrid,
-        # navix/MiniGrid-class tasks share the gridworld machinery
+        # navix/MiniGrid-class tasks share the gridworld machinery; the
+        # reference's sized scenarios (xland_minigrid/empty_*.yaml,
+        # door_key_5x5.yaml; navix/empty_5x5.yaml, door_key_8x8.yaml)
         "doorkey": DoorKeyGrid,
         "Navix-DoorKey-8x8-v0": DoorKeyGrid,
+        "empty_5x5": EmptyGrid5,
+        "empty_6x6": EmptyGrid6,
+        "door_key_5x5": DoorKeyGrid5,
+        "door_key_8x8": DoorKeyGrid8,
     }
     if name not in table:
         raise ValueError(f"unknown xland-suite env '{name}' (have {list(table)})")

@@ -8,13 +8,13 @@ observation — structured pixel-ish obs + goal conditioning + sparse
 reward. Original design (not a port): every op is batched over B boards,
 so it runs on CPU and as device tensors on GPU.
 
-Grid 9x9 with border walls plus random interior walls; ``NUM_COLORS``
+Grid 9x9 with border walls plus random interior walls; ``C`` (class attr)
 coloured objects are placed at random free cells; each episode samples a
 goal colour. Reaching the goal object's cell gives +1 and terminates;
 reaching a WRONG object gives -0.1 and removes it (the agent can recover).
-Observation [9, 9, 3 + NUM_COLORS] channels: walls, agent, objects by
+Observation [N, N, 3 + C] channels: walls, agent, objects by
 colour (one plane per colour)... plus a goal plane broadcasting the goal
-colour one-hot over the last NUM_COLORS channels' first row convention is
+colour one-hot over the last C channels' first row convention is
 avoided — the goal is a SEPARATE constant plane set: channel
 ``3 + goal`` is incremented by a constant 0.5 everywhere, so a CNN torso
 can read the goal from any receptive field.
@@ -36,45 +36,51 @@ _DC = [0, 1, 0, -1]
 
 
 class XLandGrid(StatefulVecEnv):
+    # grid size / colour count / interior walls are class attrs so the
+    # reference's sized scenarios (empty_5x5, empty_6x6, door_key_5x5, ...)
+    # instantiate as subclasses
+    N = 9
+    C = 4
+    N_WALLS = 6
     max_episode_steps = 100
     capture_safe = True
     solved_return_threshold = 0.9
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
-        self.observation_space = BoxSpace((N, N, 3 + NUM_COLORS), 0.0, 1.5)
+        self.observation_space = BoxSpace((self.N, self.N, 3 + self.C), 0.0, 1.5)
         self.action_space = DiscreteSpace(4)
         self._dr = torch.tensor(_DR, device=self.device)
         self._dc = torch.tensor(_DC, device=self.device)
         self._one_f = torch.ones((), device=self.device)
 
     def _sample_free(self, occupied: Tensor, n: int) -> Tensor:
-        """Gumbel-max a free cell per board; occupied [n, N*N] bool."""
-        u = torch.rand(n, N * N, device=self.device, generator=self.gen)
+        """Gumbel-max a free cell per board; occupied [n, self.N*self.N] bool."""
+        u = torch.rand(n, self.N * self.N, device=self.device, generator=self.gen)
         g = -torch.log(-torch.log(u.clamp(min=1e-12)).clamp(min=1e-12))
         scores = torch.where(occupied, torch.full_like(g, -torch.inf), g)
         return scores.argmax(dim=-1)
 
     def _reset_fn(self, n: int) -> State:
         dev = self.device
-        walls = torch.zeros(n, N, N, dtype=torch.bool, device=dev)
+        walls = torch.zeros(n, self.N, self.N, dtype=torch.bool, device=dev)
         walls[:, 0, :] = walls[:, -1, :] = True
         walls[:, :, 0] = walls[:, :, -1] = True
-        occ = walls.reshape(n, N * N).clone()
+        occ = walls.reshape(n, self.N * self.N).clone()
         # random interior walls (may carve dead ends; goals stay reachable
         # often enough for learning — wrong-object penalties keep signal)
-        for _ in range(N_WALLS):
+        for _ in range(self.N_WALLS):
             cell = self._sample_free(occ, n)
             occ.scatter_(1, cell.unsqueeze(1), True)
-            walls = occ.reshape(n, N, N).clone()
-        obj_pos = torch.zeros(n, NUM_COLORS, dtype=torch.long, device=dev)
-        for c in range(NUM_COLORS):
+            walls = occ.reshape(n, self.N, self.N).clone()
+        obj_pos = torch.zeros(n, self.C, dtype=torch.long, device=dev)
+        for c in range(self.C):
             cell = self._sample_free(occ, n)
             occ.scatter_(1, cell.unsqueeze(1), True)
             obj_pos[:, c] = cell
         agent = self._sample_free(occ, n)
-        goal = torch.randint(0, NUM_COLORS, (n,), device=dev, generator=self.gen)
-        alive = torch.ones(n, NUM_COLORS, dtype=torch.bool, device=dev)
+        goal = torch.randint(0, self.C, (n,), device=dev, generator=self.gen)
+        alive = torch.ones(n, self.C, dtype=torch.bool, device=dev)
         return {
             "walls": walls.float(),
             "obj_pos": obj_pos.float(),
@@ -92,18 +98,18 @@ class XLandGrid(StatefulVecEnv):
         alive = state["obj_alive"]
         goal = state["goal"].long()
         bidx = torch.arange(n, device=dev)
-        obs = torch.zeros(n, N, N, 3 + NUM_COLORS, device=dev)
+        obs = torch.zeros(n, self.N, self.N, 3 + self.C, device=dev)
         obs[..., 0] = walls
-        a_r, a_c = agent // N, agent % N
+        a_r, a_c = agent // self.N, agent % self.N
         obs[bidx, a_r, a_c, 1] = self._one_f  # device scalar: capture-legal
         # channel 2: any-object plane; 3+c: per-colour planes
-        for c in range(NUM_COLORS):
-            r, cc = obj_pos[:, c] // N, obj_pos[:, c] % N
+        for c in range(self.C):
+            r, cc = obj_pos[:, c] // self.N, obj_pos[:, c] % self.N
             obs[bidx, r, cc, 2] = torch.maximum(obs[bidx, r, cc, 2], alive[:, c])
             obs[bidx, r, cc, 3 + c] = alive[:, c]
         # goal conditioning: +0.5 broadcast over the goal colour's plane
-        goal_onehot = torch.nn.functional.one_hot(goal, NUM_COLORS).float()
-        obs[..., 3:] = obs[..., 3:] + 0.5 * goal_onehot.view(n, 1, 1, NUM_COLORS)
+        goal_onehot = torch.nn.functional.one_hot(goal, self.C).float()
+        obs[..., 3:] = obs[..., 3:] + 0.5 * goal_onehot.view(n, 1, 1, self.C)
         return obs
 
     def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
@@ -111,21 +117,21 @@ class XLandGrid(StatefulVecEnv):
         dev = self.device
         a = action.long().clamp(0, 3)
         agent = state["agent"].long()
-        r, c = agent // N, agent % N
-        nr = (r + self._dr[a]).clamp(0, N - 1)
-        nc = (c + self._dc[a]).clamp(0, N - 1)
+        r, c = agent // self.N, agent % self.N
+        nr = (r + self._dr[a]).clamp(0, self.N - 1)
+        nc = (c + self._dc[a]).clamp(0, self.N - 1)
         walls = state["walls"] > 0.5
         bidx = torch.arange(n, device=dev)
         blocked = walls[bidx, nr, nc]
         nr = torch.where(blocked, r, nr)
         nc = torch.where(blocked, c, nc)
-        new_agent = nr * N + nc
+        new_agent = nr * self.N + nc
 
         obj_pos = state["obj_pos"].long()
         alive = state["obj_alive"] > 0.5
         goal = state["goal"].long()
         on_obj = (obj_pos == new_agent.unsqueeze(1)) & alive  # [n, C]
-        goal_onehot = torch.nn.functional.one_hot(goal, NUM_COLORS).bool()
+        goal_onehot = torch.nn.functional.one_hot(goal, self.C).bool()
         hit_goal = (on_obj & goal_onehot).any(dim=1)
         hit_wrong = (on_obj & ~goal_onehot).any(dim=1)
         reward = hit_goal.float() - 0.1 * hit_wrong.float()
@@ -156,13 +162,15 @@ class DoorKeyGrid(StatefulVecEnv):
     Actions: 0-3 move (pickup/open happen by walking into the cell).
     """
 
+    N = 9
+    WALL_COL_RANGE = (3, 6)  # sampled dividing-wall columns [lo, hi)
     max_episode_steps = 200
     capture_safe = True
     solved_return_threshold = 0.5
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
-        self.observation_space = BoxSpace((N, N, 6), 0.0, 1.5)
+        self.observation_space = BoxSpace((self.N, self.N, 6), 0.0, 1.5)
         self.action_space = DiscreteSpace(4)
         self._dr = torch.tensor(_DR, device=self.device)
         self._dc = torch.tensor(_DC, device=self.device)
@@ -171,25 +179,26 @@ class DoorKeyGrid(StatefulVecEnv):
     def _reset_fn(self, n: int) -> State:
         dev = self.device
         # dividing wall at a random column 3..5 with a door at a random row
-        wall_col = torch.randint(3, 6, (n,), device=dev, generator=self.gen)
-        door_row = torch.randint(1, N - 1, (n,), device=dev, generator=self.gen)
-        cols = torch.arange(N, device=dev)
-        walls = torch.zeros(n, N, N, dtype=torch.bool, device=dev)
+        lo, hi = self.WALL_COL_RANGE
+        wall_col = torch.randint(lo, hi, (n,), device=dev, generator=self.gen)
+        door_row = torch.randint(1, self.N - 1, (n,), device=dev, generator=self.gen)
+        cols = torch.arange(self.N, device=dev)
+        walls = torch.zeros(n, self.N, self.N, dtype=torch.bool, device=dev)
         walls[:, 0, :] = walls[:, -1, :] = True
         walls[:, :, 0] = walls[:, :, -1] = True
-        walls |= cols.view(1, 1, N) == wall_col.view(n, 1, 1)
+        walls |= cols.view(1, 1, self.N) == wall_col.view(n, 1, 1)
         # carve the door cell out of the wall mask (it is tracked separately)
         bidx = torch.arange(n, device=dev)
         walls[bidx, door_row, wall_col] = False
         # key on the LEFT side, goal on the RIGHT side, agent LEFT
         u = torch.rand(n, 4, device=dev, generator=self.gen)
-        key_r = 1 + (u[:, 0] * (N - 2)).long().clamp(max=N - 3)
+        key_r = 1 + (u[:, 0] * (self.N - 2)).long().clamp(max=self.N - 3)
         key_c = 1 + (u[:, 1] * (wall_col.float() - 1.0)).long().clamp(min=0)
         key_c = torch.minimum(key_c, wall_col - 1).clamp(min=1)
-        agent_r = 1 + (u[:, 2] * (N - 2)).long().clamp(max=N - 3)
+        agent_r = 1 + (u[:, 2] * (self.N - 2)).long().clamp(max=self.N - 3)
         agent_c = torch.ones(n, dtype=torch.long, device=dev)
         goal_r = door_row  # reachable by construction
-        goal_c = torch.full((n,), N - 2, dtype=torch.long, device=dev)
+        goal_c = torch.full((n,), self.N - 2, dtype=torch.long, device=dev)
         return {
             "walls": walls.float(),
             "door_r": door_row.float(), "door_c": wall_col.float(),
@@ -204,7 +213,7 @@ class DoorKeyGrid(StatefulVecEnv):
         n = state["agent_r"].shape[0]
         dev = self.device
         bidx = torch.arange(n, device=dev)
-        obs = torch.zeros(n, N, N, 6, device=dev)
+        obs = torch.zeros(n, self.N, self.N, 6, device=dev)
         obs[..., 0] = state["walls"]
         obs[bidx, state["agent_r"].long(), state["agent_c"].long(), 1] = self._one_f
         key_vis = (state["has_key"] < 0.5).float()
@@ -220,8 +229,8 @@ class DoorKeyGrid(StatefulVecEnv):
         dev = self.device
         a = action.long().clamp(0, 3)
         r, c = state["agent_r"].long(), state["agent_c"].long()
-        nr = (r + self._dr[a]).clamp(0, N - 1)
-        nc = (c + self._dc[a]).clamp(0, N - 1)
+        nr = (r + self._dr[a]).clamp(0, self.N - 1)
+        nc = (c + self._dc[a]).clamp(0, self.N - 1)
         bidx = torch.arange(n, device=dev)
         walls = state["walls"] > 0.5
         at_door = (nr == state["door_r"].long()) & (nc == state["door_c"].long())
@@ -252,3 +261,34 @@ class DoorKeyGrid(StatefulVecEnv):
             reward,
             at_goal,
         )
+
+
+class EmptyGrid5(XLandGrid):
+    """MiniGrid Empty-5x5-class (reference xland_minigrid/empty_5x5.yaml,
+    navix/empty_5x5.yaml): one object, no interior walls — reach it."""
+
+    N = 5
+    C = 1
+    N_WALLS = 0
+    max_episode_steps = 50
+
+
+class EmptyGrid6(EmptyGrid5):
+    """Empty-6x6 (reference xland_minigrid/empty_6x6.yaml)."""
+
+    N = 6
+
+
+class DoorKeyGrid5(DoorKeyGrid):
+    """DoorKey-5x5 (reference xland_minigrid/door_key_5x5.yaml)."""
+
+    N = 5
+    WALL_COL_RANGE = (2, 3)
+    max_episode_steps = 100
+
+
+class DoorKeyGrid8(DoorKeyGrid):
+    """DoorKey-8x8 (reference navix/door_key_8x8.yaml)."""
+
+    N = 8
+    WALL_COL_RANGE = (2, 6)
