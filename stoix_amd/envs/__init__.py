@@ -19,8 +19,11 @@ dispatch over suites (:420-433) and ``make(config) -> (train_env, eval_env)``
   * ``craftax`` — crafting/achievement-chain world (envs/crafting.py;
     capability-class equivalent of the JAX-only craftax suite).
 
-JAX-only external suites with no offline equivalent (navix, kinetix,
-popjym, playground, jaxarc) raise a clear error naming what is
+Also implemented in-repo (round 2): ``navix`` (DoorKey/Empty grids),
+``popjym`` + ``popgym_arcade`` (POMDPs and memory games), ``kinetix``
+(procedural reacher slice), ``mujoco_playground`` (swing-up + aliases),
+``jaxarc`` (GridCopy slice), ``gymnasium`` (adapter; needs the gymnasium
+package). Unknown scenario names raise a clear error listing what is
 available.
 """
 from __future__ import annotations
@@ -192,9 +195,19 @@ def _xland(name: str):
 
 
 def _craftax(name: str):
-    from stoix_amd.envs.crafting import Crafting
+    from stoix_amd.envs.crafting import Crafting, CraftingPixels
 
-    table = {"crafting": Crafting, "Craftax-Symbolic-v1": Crafting}
+    table = {
+        "crafting": Crafting,
+        "Craftax-Symbolic-v1": Crafting,
+        # the reference's four craftax scenario yamls (symbolic /
+        # classic_symbolic / pixels / classic_pixels)
+        "symbolic": Crafting,
+        "classic_symbolic": Crafting,
+        "crafting_pixels": CraftingPixels,
+        "pixels": CraftingPixels,
+        "classic_pixels": CraftingPixels,
+    }
     if name not in table:
         raise ValueError(f"unknown craftax-suite env '{name}' (have {list(table)})")
     return table[name]

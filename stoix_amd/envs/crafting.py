@@ -188,3 +188,22 @@ class Crafting(StatefulVecEnv):
             reward,
             all_done,
         )
+
+
+class CraftingPixels(Crafting):
+    """Pixel-style observation variant (reference craftax pixels.yaml /
+    classic_pixels.yaml scenarios feed image obs to CNN torsos): the same
+    world/dynamics with obs as [M, M, 12] image planes — 4 cell-type
+    one-hots, the agent plane, and the 7 inventory/achievement scalars
+    broadcast as constant planes (so any receptive field can read them)."""
+
+    def __init__(self, num_envs, device="cpu", seed=0, **kw):
+        super().__init__(num_envs, device, seed, **kw)
+        self.observation_space = BoxSpace((M, M, 12), 0.0, 10.0)
+
+    def _obs_fn(self, state: State) -> Tensor:
+        flat = super()._obs_fn(state)
+        n = flat.shape[0]
+        planes = flat[:, : M * M * 5].view(n, M, M, 5)
+        inv = flat[:, M * M * 5 :].view(n, 1, 1, 7).expand(n, M, M, 7)
+        return torch.cat([planes, inv], dim=-1)
