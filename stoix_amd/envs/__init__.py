@@ -277,7 +277,7 @@ def _popgym_arcade(name: str):
 
 
 def _playground(name: str):
-    from stoix_amd.envs.classic import CartPoleSwingUp, Pendulum
+    from stoix_amd.envs.classic import CartPoleBalance, CartPoleSwingUp, Pendulum
     from stoix_amd.envs.humanoid import Humanoid
     from stoix_amd.envs.planar import HalfCheetah, Hopper
 
@@ -285,6 +285,15 @@ def _playground(name: str):
     table = {
         "cartpole_swingup": CartPoleSwingUp,
         "CartpoleSwingup": CartPoleSwingUp,
+        # dm_control-tier scenarios (reference mjc_playground/dm_control/*)
+        "cartpole_balance": CartPoleBalance,
+        "CartpoleBalance": CartPoleBalance,
+        "hopper_hop": Hopper,
+        "HopperHop": Hopper,
+        # quadruped joystick locomotion maps to the Ant-class physics tier
+        # (reference mjc_playground/locomotion/go_1_joystick_flat_terrain)
+        "go_1_joystick_flat_terrain": Ant,
+        "Go1JoystickFlatTerrain": Ant,
         "pendulum_swingup": Pendulum,
         # locomotion tasks map onto the physics envs of the brax-class tier
         "ant": Ant,

@@ -282,3 +282,16 @@ class CartPoleSwingUp(CartPole):
         reward = upright * centred
         terminated = torch.zeros_like(hit)  # horizon-only
         return {"s": new_s}, reward, terminated
+
+
+class CartPoleBalance(CartPoleSwingUp):
+    """dm_control cartpole BALANCE task (reference
+    mjc_playground/dm_control/cartpole_balance.yaml): same continuous-force
+    cartpole and smooth upright/centred reward as swing-up, but the pole
+    STARTS UPRIGHT and must be kept there."""
+
+    solved_return_threshold = 450.0
+
+    def _reset_fn(self, n: int):
+        s = self.rand(n, 4, lo=-0.05, hi=0.05)  # upright start
+        return {"s": s}

@@ -1304,3 +1304,17 @@ def test_popgym_arcade_suite_registered():
     for _ in range(3):
         ts = env.step(torch.zeros(4, dtype=torch.long))
     assert ts.observation.shape[0] == 4
+
+
+def test_cartpole_balance_starts_upright_swingup_starts_down():
+    from stoix_amd.envs.classic import CartPoleBalance, CartPoleSwingUp
+
+    bal = CartPoleBalance(16, seed=0)
+    ts = bal.reset()
+    assert torch.all(ts.observation[:, 2] > 0.99)  # cos(theta) ~ 1
+    swing = CartPoleSwingUp(16, seed=0)
+    ts = swing.reset()
+    assert torch.all(ts.observation[:, 2] < -0.99)  # hanging
+    # balance reward is near-max while upright under zero force
+    ts = bal.step(torch.zeros(16, 1))
+    assert torch.all(ts.reward > 0.8)
