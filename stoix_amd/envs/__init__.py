@@ -138,6 +138,30 @@ def _envpool(name: str):
         if envpool_ext(required=True) is not None:
             return VizdoomBasicCpu(**kw)
 
+    def phoenix_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import PhoenixCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return PhoenixCpu(**kw)
+
+    def battlezone_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import BattlezoneCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return BattlezoneCpu(**kw)
+
+    def doubledunk_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import DoubledunkCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return DoubledunkCpu(**kw)
+
+    def namethisgame_maker(**kw):
+        from stoix_amd.envs.envpool_cpu import NameThisGameCpu, envpool_ext
+
+        if envpool_ext(required=True) is not None:
+            return NameThisGameCpu(**kw)
+
     table = {
         "breakout": breakout_maker,
         "vizdoom_basic": vizdoom_maker,
@@ -149,6 +173,14 @@ def _envpool(name: str):
         "SpaceInvaders-v5": spaceinv_maker,
         "qbert": qbert_maker,
         "Qbert-v5": qbert_maker,
+        "phoenix": phoenix_maker,
+        "Phoenix-v5": phoenix_maker,
+        "battlezone": battlezone_maker,
+        "BattleZone-v5": battlezone_maker,
+        "doubledunk": doubledunk_maker,
+        "DoubleDunk-v5": doubledunk_maker,
+        "namethisgame": namethisgame_maker,
+        "NameThisGame-v5": namethisgame_maker,
         # classic-control names envpool also serves
         "cartpole": CartPole,
         "CartPole-v1": CartPole,

@@ -21,7 +21,7 @@ def build(verbose: bool = False):
     BUILD_DIR.mkdir(exist_ok=True)
     return load(
         name="stoix_amd_envpool",
-        sources=[str(CSRC / "envpool_cpu.cpp")],
+        sources=[str(CSRC / "envpool_cpu.cpp"), str(CSRC / "envpool_games2.cpp")],
         extra_cflags=["-O3", "-std=c++17"],
         build_directory=str(BUILD_DIR),
         verbose=verbose,

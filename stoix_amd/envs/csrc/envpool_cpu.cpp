@@ -872,6 +872,10 @@ void vizdoom_cpu_step(torch::Tensor state, torch::Tensor action,
 }
 }  // namespace vizdoom
 
+// second-wave games (phoenix / battlezone / doubledunk / namethisgame)
+// live in envpool_games2.cpp and register through this hook
+void register_games2(pybind11::module_& m);
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   // gil_scoped_release: the whole fused step runs WITHOUT the GIL so
   // Sebulba's learner thread and sibling actor threads keep running
@@ -911,4 +915,5 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "batched VizDoom-basic-class fused step",
         py::call_guard<py::gil_scoped_release>());
   m.attr("VIZDOOM_STATE_DIM") = vizdoom::SDIM;
+  register_games2(m);
 }

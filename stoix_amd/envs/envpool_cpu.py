@@ -178,3 +178,55 @@ class BreakoutCpu(_PoolEnvBase):
     RESET_FN = "breakout_reset"
     STEP_FN = "breakout_step"
     N_ACTIONS = 4
+
+
+class PhoenixCpu(_PoolEnvBase):
+    """Native-engine Phoenix-class: a bird formation that periodically
+    swoops at the cannon; clearing it summons a multi-hit mothership.
+    +1/bird, +2/mothership hit, +10 destroy (win); a swooping bird
+    reaching the player ends the episode."""
+
+    max_episode_steps = 3000
+    STATE_DIM_ATTR = "PHOENIX_STATE_DIM"
+    RESET_FN = "phoenix_reset"
+    STEP_FN = "phoenix_step"
+    N_ACTIONS = 4  # noop / left / right / fire
+
+
+class BattlezoneCpu(_PoolEnvBase):
+    """Native-engine Battlezone-class first-person tank: rotate / drive /
+    fire on an open plane; enemies close in and fire on a timer when in
+    range. +10 per kill (fresh enemy spawns), -1 and termination when
+    shot. Obs: horizon render + bearing billboard + radar strip."""
+
+    max_episode_steps = 2000
+    STATE_DIM_ATTR = "BATTLEZONE_STATE_DIM"
+    RESET_FN = "battlezone_reset"
+    STEP_FN = "battlezone_step"
+    N_ACTIONS = 5  # noop / rot-left / rot-right / forward / fire
+
+
+class DoubledunkCpu(_PoolEnvBase):
+    """Native-engine DoubleDunk-class half-court basketball vs a scripted
+    defender: drive and shoot (+2 inside the arc), defender touch is a
+    steal, then the opponent drives for the hoop (-2 unless touched to
+    steal back). Fixed horizon; return = net points."""
+
+    max_episode_steps = 1000
+    STATE_DIM_ATTR = "DOUBLEDUNK_STATE_DIM"
+    RESET_FN = "doubledunk_reset"
+    STEP_FN = "doubledunk_step"
+    N_ACTIONS = 5  # up / down / left / right / shoot-or-steal
+
+
+class NameThisGameCpu(_PoolEnvBase):
+    """Native-engine NameThisGame-class undersea shooter: trim the
+    octopus tentacles growing toward the diver (+0.5/trim), shoot the
+    patrolling shark (+5); a tentacle reaching the sea floor ends the
+    episode."""
+
+    max_episode_steps = 3000
+    STATE_DIM_ATTR = "NAMETHISGAME_STATE_DIM"
+    RESET_FN = "namethisgame_reset"
+    STEP_FN = "namethisgame_step"
+    N_ACTIONS = 4  # noop / left / right / fire

@@ -1318,3 +1318,89 @@ def test_cartpole_balance_starts_upright_swingup_starts_down():
     # balance reward is near-max while upright under zero force
     ts = bal.step(torch.zeros(16, 1))
     assert torch.all(ts.reward > 0.8)
+
+
+def test_new_pool_games_rules():
+    """Rule tests for the second-wave native pool games (reference envpool
+    scenarios phoenix / battlezone / doubledunk / namethisgame)."""
+    import math
+
+    from stoix_amd.envs.envpool_cpu import (
+        BattlezoneCpu,
+        DoubledunkCpu,
+        NameThisGameCpu,
+        PhoenixCpu,
+        envpool_ext,
+    )
+
+    if envpool_ext() is None:
+        pytest.skip("native pool extension not built")
+
+    # Phoenix: track a formation bird's x and fire when aligned -> +1
+    env = PhoenixCpu(1, seed=5)
+    env.reset()
+    got = 0.0
+    for _ in range(200):
+        bird_x = float(env._s[0, 9 + 2])  # first alive bird's x
+        px = float(env._s[0, 0])
+        if abs(px - bird_x) < 1.5:
+            a = 3  # fire
+        else:
+            a = 2 if bird_x > px else 1
+        ts = env.step(torch.tensor([a]))
+        got += float(ts.reward)
+        if got >= 1.0:
+            break
+    assert got >= 1.0, "aimed shots never killed a bird"
+
+    # Battlezone: rotate onto the enemy bearing and fire -> +10
+    env = BattlezoneCpu(1, seed=1)
+    env.reset()
+    got = 0.0
+    for _ in range(300):
+        s = env._s[0]
+        rel = math.atan2(float(s[4] - s[0]), float(s[5] - s[1])) - float(s[2])
+        while rel > math.pi:
+            rel -= 2 * math.pi
+        while rel < -math.pi:
+            rel += 2 * math.pi
+        if abs(rel) < 0.06:
+            a = 4
+        else:
+            a = 2 if rel > 0 else 1
+        ts = env.step(torch.tensor([a]))
+        got += float(ts.reward)
+        if got >= 10.0:
+            break
+    assert got >= 10.0, "aimed tank shot never killed the enemy"
+
+    # DoubleDunk: a dunk from inside 6px of the hoop pays +2
+    env = DoubledunkCpu(1, seed=2)
+    env.reset()
+    env._s[0, 0] = 42.0  # teleport next to the hoop (rule check, not play)
+    env._s[0, 1] = 10.0
+    ts = env.step(torch.tensor([4]))
+    assert float(ts.reward) == 2.0
+    # positions reset after the score: player back near the baseline
+    assert float(env._s[0, 1]) == 70.0
+
+    # NameThisGame: firing up a tentacle column trims it (+0.5); pure noop
+    # eventually lets a tentacle reach the floor and terminate
+    env = NameThisGameCpu(1, seed=3)
+    env.reset()
+    env._s[0, 0] = 12.0  # under tentacle 0
+    len0 = float(env._s[0, 9])
+    got = 0.0
+    for t in range(40):
+        ts = env.step(torch.tensor([3 if t % 3 == 0 else 0]))
+        got += float(ts.reward)
+    assert got >= 0.5 and float(env._s[0, 9]) < len0 + 4.0
+    env = NameThisGameCpu(4, seed=4)
+    env.reset()
+    done = torch.zeros(4, dtype=torch.bool)
+    for _ in range(1500):
+        ts = env.step(torch.zeros(4, dtype=torch.long))
+        done |= ts.extras["episode_metrics"]["is_terminal_step"]
+        if bool(done.all()):
+            break
+    assert bool(done.all()), "tentacles never reached the floor under noop"
