@@ -49,6 +49,7 @@ run stoix_amd.systems.ppo.ff_ppo              "doorkey" env=navix/doorkey system
 run stoix_amd.systems.ppo.ff_ppo              "swingup" env=mjc_playground/cartpole_swingup system=ppo/ff_ppo_continuous network=mlp_continuous system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "reacher" env=kinetix/reacher system=ppo/ff_ppo_continuous network=mlp_continuous system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.ppo.ff_ppo              "vizdoom" env=envpool/vizdoom_basic network=cnn system.num_minibatches=2 system.epochs=1
+run stoix_amd.systems.ppo.ff_ppo              "battlezone" env=envpool/battlezone network=cnn system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.impala.sebulba_ff_impala "" arch.actor.actor_per_device=2 system.num_minibatches=2
 run stoix_amd.systems.ppo.sebulba_ff_ppo      "" arch.actor.actor_per_device=2 system.num_minibatches=2 system.epochs=1
 run stoix_amd.systems.impala.sebulba_ff_impala_shared_torso "" arch.actor.actor_per_device=2 system.num_minibatches=2
