@@ -528,3 +528,25 @@ def test_ppo_learns_battlezone():
     )
     r = run(cfg)
     assert r > 50.0, f"battlezone not learned: return={r}"
+
+
+@pytest.mark.slow
+def test_ppo_learns_doubledunk():
+    """PPO+CNN on the DoubleDunk-class pool game: random play concedes
+    heavily (-107 per episode); the policy must learn to protect the ball
+    and steal back. Measured -0.75 (near break-even) at this budget; gate
+    well above the random floor."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=envpool/doubledunk", "network=cnn", "arch.total_num_envs=16",
+         "arch.total_timesteps=null", "arch.num_updates=150",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=8",
+         "arch.absolute_metric=false", "system.rollout_length=32",
+         "system.num_minibatches=4", "system.epochs=2",
+         "system.ent_coef=0.02", "logger.loggers=[]",
+         "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > -30.0, f"doubledunk not learned: return={r}"
