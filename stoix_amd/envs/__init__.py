@@ -339,9 +339,26 @@ def _playground(name: str):
 
 
 def _kinetix(name: str):
-    from stoix_amd.envs.reacher import ProceduralReacher
+    from stoix_amd.envs.reacher import (
+        ProceduralReacher,
+        ProceduralReacher3,
+        ProceduralReacherSmall,
+    )
 
-    table = {"reacher": ProceduralReacher, "Kinetix-Reacher-v1": ProceduralReacher}
+    table = {
+        "reacher": ProceduralReacher,
+        "Kinetix-Reacher-v1": ProceduralReacher,
+        # the reference's env_size tiers (configs/env/kinetix/{small,medium,
+        # large,all}.yaml) scale scene complexity; here they scale the
+        # articulation (link count + morphology range) — documented slice
+        "small": ProceduralReacherSmall,
+        "s": ProceduralReacherSmall,
+        "medium": ProceduralReacher,
+        "m": ProceduralReacher,
+        "large": ProceduralReacher3,
+        "l": ProceduralReacher3,
+        "all": ProceduralReacher3,
+    }
     if name not in table:
         raise ValueError(f"unknown kinetix-suite env '{name}' (have {list(table)})")
     return table[name]

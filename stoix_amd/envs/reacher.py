@@ -30,23 +30,31 @@ GOAL_R = 0.10
 
 
 class ProceduralReacher(StatefulVecEnv):
+    # morphology tier knobs (the reference's kinetix env_size tiers
+    # small/medium/large scale scene complexity; here they scale the
+    # articulation: link count and per-episode length ranges)
+    NLINK = 2
+    LEN_LO, LEN_HI = 0.3, 0.7
     max_episode_steps = 150
     capture_safe = True
     solved_return_threshold = -10.0
 
     def __init__(self, num_envs, device="cpu", seed=0, **kw):
         super().__init__(num_envs, device, seed)
-        self.observation_space = BoxSpace((12,), -10.0, 10.0)
-        self.action_space = BoxSpace((2,), -1.0, 1.0)
+        L = self.NLINK
+        # obs: [cos q_i, sin q_i] per joint, dq (L), goal (2), tip (2), len (L)
+        self.observation_space = BoxSpace((3 * L + L + 4,), -10.0, 10.0)
+        self.action_space = BoxSpace((L,), -1.0, 1.0)
 
     def _reset_fn(self, n: int) -> State:
-        q = self.rand(n, 2, lo=-3.14159, hi=3.14159)
-        dq = torch.zeros(n, 2, device=self.device)
+        L = self.NLINK
+        q = self.rand(n, L, lo=-3.14159, hi=3.14159)
+        dq = torch.zeros(n, L, device=self.device)
         # procedural morphology: per-episode link lengths
-        lengths = self.rand(n, 2, lo=0.3, hi=0.7)
+        lengths = self.rand(n, L, lo=self.LEN_LO, hi=self.LEN_HI)
         # goal inside the annulus the arm can actually reach
         reach_max = lengths.sum(-1)
-        reach_min = (lengths[:, 0] - lengths[:, 1]).abs() + 0.05
+        reach_min = (2.0 * lengths.max(dim=-1).values - reach_max).clamp(min=0.0) + 0.05
         r = reach_min + self.rand(n) * (0.95 * reach_max - reach_min).clamp(min=0.01)
         ang = self.rand(n, lo=-3.14159, hi=3.14159)
         goal = torch.stack([r * torch.cos(ang), r * torch.sin(ang)], dim=-1)
@@ -54,32 +62,24 @@ class ProceduralReacher(StatefulVecEnv):
 
     @staticmethod
     def _tip(q: Tensor, lengths: Tensor) -> Tensor:
-        a1 = q[:, 0]
-        a2 = q[:, 0] + q[:, 1]
-        x = lengths[:, 0] * torch.cos(a1) + lengths[:, 1] * torch.cos(a2)
-        y = lengths[:, 0] * torch.sin(a1) + lengths[:, 1] * torch.sin(a2)
+        a = torch.cumsum(q, dim=-1)  # absolute link angles
+        x = (lengths * torch.cos(a)).sum(-1)
+        y = (lengths * torch.sin(a)).sum(-1)
         return torch.stack([x, y], dim=-1)
 
     def _obs_fn(self, state: State) -> Tensor:
         q, dq = state["q"], state["dq"]
         tip = self._tip(q, state["len"])
-        return torch.cat(
-            [
-                torch.cos(q[:, :1]), torch.sin(q[:, :1]),
-                torch.cos(q[:, 1:]), torch.sin(q[:, 1:]),
-                dq, state["goal"], tip, state["len"],
-            ],
-            dim=-1,
-        )
+        trig = torch.stack([torch.cos(q), torch.sin(q)], dim=-1).flatten(1)
+        return torch.cat([trig, dq, state["goal"], tip, state["len"]], dim=-1)
 
     def _step_fn(self, state: State, action: Tensor) -> Tuple[State, Tensor, Tensor]:
         q, dq = state["q"], state["dq"]
-        torque = action.reshape(-1, 2).clamp(-1.0, 1.0) * TORQUE
-        # inertia grows with the link lengths the joint must swing
+        torque = action.reshape(-1, self.NLINK).clamp(-1.0, 1.0) * TORQUE
+        # inertia per joint grows with the outboard link lengths it swings
         lengths = state["len"]
-        inertia1 = (lengths[:, 0] + lengths[:, 1]).pow(2).unsqueeze(-1)
-        inertia2 = lengths[:, 1:].pow(2)
-        inertia = torch.cat([inertia1, inertia2], dim=-1).clamp(min=0.05)
+        outboard = lengths.flip(-1).cumsum(-1).flip(-1)  # sum of links j..L-1
+        inertia = outboard.pow(2).clamp(min=0.05)
         ddq = torque / inertia - DAMP * dq
         dq = (dq + DT * ddq).clamp(-8.0, 8.0)
         q = q + DT * dq
@@ -92,3 +92,17 @@ class ProceduralReacher(StatefulVecEnv):
             reward,
             touched,
         )
+
+
+class ProceduralReacherSmall(ProceduralReacher):
+    """kinetix env_size small tier: 2 links, narrow morphology range."""
+
+    LEN_LO, LEN_HI = 0.4, 0.6
+
+
+class ProceduralReacher3(ProceduralReacher):
+    """kinetix env_size large tier: 3-link arm, full morphology range —
+    redundant kinematics the policy must resolve per episode."""
+
+    NLINK = 3
+    max_episode_steps = 200

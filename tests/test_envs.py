@@ -1404,3 +1404,23 @@ def test_new_pool_games_rules():
         if bool(done.all()):
             break
     assert bool(done.all()), "tentacles never reached the floor under noop"
+
+
+def test_reacher_nlink_kinematics():
+    """The N-link generalisation: 3-link tip position matches the manual
+    forward kinematics, and joint inertia uses the outboard link sums."""
+    from stoix_amd.envs.reacher import ProceduralReacher3
+
+    env = ProceduralReacher3(4, seed=0)
+    env.reset()
+    q = env._state["q"]
+    lens = env._state["len"]
+    a1 = q[:, 0]
+    a2 = q[:, 0] + q[:, 1]
+    a3 = a2 + q[:, 2]
+    x = lens[:, 0] * torch.cos(a1) + lens[:, 1] * torch.cos(a2) + lens[:, 2] * torch.cos(a3)
+    y = lens[:, 0] * torch.sin(a1) + lens[:, 1] * torch.sin(a2) + lens[:, 2] * torch.sin(a3)
+    tip = env._tip(q, lens)
+    torch.testing.assert_close(tip, torch.stack([x, y], dim=-1), rtol=1e-5, atol=1e-6)
+    # the goal is always reachable: |goal| <= 0.95 * total length
+    assert torch.all(env._state["goal"].norm(dim=-1) <= lens.sum(-1) * 0.951)
