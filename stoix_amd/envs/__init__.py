@@ -367,7 +367,16 @@ def _kinetix(name: str):
 def _jaxarc(name: str):
     from stoix_amd.envs.arc import GridCopy
 
-    table = {"grid_copy": GridCopy, "Arc-GridCopy-v1": GridCopy}
+    from stoix_amd.envs.arc import GridMirror
+
+    table = {
+        "grid_copy": GridCopy,
+        "Arc-GridCopy-v1": GridCopy,
+        # concept-class transformation slice (reference jaxarc concept
+        # grouping): paint the MIRROR of the shown sprite
+        "grid_mirror": GridMirror,
+        "Arc-GridMirror-v1": GridMirror,
+    }
     if name not in table:
         raise ValueError(f"unknown jaxarc-suite env '{name}' (have {list(table)})")
     return table[name]

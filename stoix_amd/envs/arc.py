@@ -56,6 +56,9 @@ class GridCopy(StatefulVecEnv):
             target.scatter_(1, cell.unsqueeze(1), color.unsqueeze(1))
         cursor = torch.full((n,), (G // 2) * G + G // 2, dtype=torch.long, device=dev)
         return {
+            # the grid the OBS displays; the copy task shows the scoring
+            # target itself, transformation subclasses show the input
+            "shown": target.float(),
             "target": target.float(),
             "canvas": torch.zeros(n, G * G, device=dev),
             "cursor": cursor.float(),
@@ -69,7 +72,7 @@ class GridCopy(StatefulVecEnv):
     def _obs_fn(self, state: State) -> Tensor:
         n = state["cursor"].shape[0]
         dev = self.device
-        target = state["target"].long().view(n, G, G)
+        target = state["shown"].long().view(n, G, G)
         canvas = state["canvas"].long().view(n, G, G)
         # colour one-hots without the empty class
         t_oh = torch.nn.functional.one_hot(target.clamp(0, K), K + 1)[..., 1:].float()
@@ -119,8 +122,22 @@ class GridCopy(StatefulVecEnv):
         # implication target>0 -> canvas==target)
         terminated = ((state["target"] > 0) <= (canvas == state["target"])).all(dim=-1)
         return (
-            {"target": state["target"], "canvas": canvas, "cursor": npos.float(),
-             "rewarded": rewarded},
+            {"shown": state["shown"], "target": state["target"], "canvas": canvas,
+             "cursor": npos.float(), "rewarded": rewarded},
             reward,
             terminated,
         )
+
+
+class GridMirror(GridCopy):
+    """ARC concept-class transformation task (reference jaxarc concept
+    grouping): the obs shows an INPUT sprite; the scored target is its
+    HORIZONTAL MIRROR. Unlike GridCopy (answer visible), the agent must
+    internalise the transformation rule — paint the mirror of what it
+    sees."""
+
+    def _reset_fn(self, n):
+        state = super()._reset_fn(n)
+        mirrored = state["shown"].view(n, G, G).flip(-1).reshape(n, G * G)
+        state["target"] = mirrored
+        return state

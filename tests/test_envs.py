@@ -1424,3 +1424,21 @@ def test_reacher_nlink_kinematics():
     torch.testing.assert_close(tip, torch.stack([x, y], dim=-1), rtol=1e-5, atol=1e-6)
     # the goal is always reachable: |goal| <= 0.95 * total length
     assert torch.all(env._state["goal"].norm(dim=-1) <= lens.sum(-1) * 0.951)
+
+
+def test_grid_mirror_scores_against_mirrored_input():
+    """GridMirror (ARC concept-class slice): the obs shows the input
+    sprite; scoring is against its horizontal mirror."""
+    from stoix_amd.envs.arc import G, GridMirror
+
+    env = GridMirror(4, seed=0)
+    env.reset()
+    shown = env._state["shown"].view(4, G, G)
+    target = env._state["target"].view(4, G, G)
+    assert torch.equal(shown.flip(-1), target)
+    # painting the mirrored cell with the right colour pays +1
+    idx = int((env._state["target"][0] > 0).nonzero()[0])
+    env._state["cursor"][:] = float(idx)
+    color = int(env._state["target"][0, idx])
+    ts = env.step(torch.full((4,), 3 + color, dtype=torch.long))
+    assert float(ts.reward[0]) == 1.0

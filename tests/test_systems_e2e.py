@@ -550,3 +550,25 @@ def test_ppo_learns_doubledunk():
     )
     r = run(cfg)
     assert r > -30.0, f"doubledunk not learned: return={r}"
+
+
+@pytest.mark.slow
+def test_ppo_learns_grid_mirror():
+    """PPO on the ARC concept-class mirror task (obs shows the input, the
+    scored target is its horizontal mirror — the transformation rule must
+    be internalised, unlike grid_copy where the answer is visible).
+    Measured 6.7/10 cells at this budget; random play hovers near 0."""
+    from stoix_amd.systems.ppo.ff_ppo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=jaxarc/grid_mirror", "arch.total_num_envs=256",
+         "arch.total_timesteps=null", "arch.num_updates=150",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=32",
+         "system.num_minibatches=4", "system.epochs=4",
+         "system.ent_coef=0.01", "logger.loggers=[]",
+         "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 3.0, f"grid mirror not learned: return={r}"
