@@ -318,3 +318,27 @@ def test_native_pool_concurrent_actor_threads_isolated():
     assert set(results) == set(range(6))
     for s in range(6):
         assert results[s] == solo[s], f"seed {s} diverged under concurrency"
+
+
+def test_sebulba_ppo_new_pool_game_end_to_end():
+    """The second-wave pool games flow through the Sebulba factory and
+    actor-thread machinery (fresh per-thread pool instances, GIL-released
+    stepping) exactly like Breakout — exercised on Phoenix."""
+    from stoix_amd.envs.envpool_cpu import envpool_ext
+    from stoix_amd.systems.ppo.sebulba_ff_ppo import run
+
+    if envpool_ext() is None:
+        pytest.skip("native pool extension not built")
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        [
+            "env=envpool/phoenix", "network=cnn",
+            "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
+            "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+            "logger.loggers=[]", "logger.checkpointing.save_model=false",
+        ],
+    )
+    r = run(cfg)
+    assert r == r
