@@ -118,6 +118,9 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             logger.log(eval_metrics, t_env, eval_idx, LogEvent.EVAL)
 
             if checkpointer is not None:
+                aux_fn = getattr(learner, "aux_checkpoint_state", None)
+                if aux_fn is not None:
+                    checkpointer.stage_aux(aux_fn())
                 checkpointer.save(t_env, learner.state_for_checkpoint(), mean_return)
             if mean_return >= best_return:
                 best_return = mean_return

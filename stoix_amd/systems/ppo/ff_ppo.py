@@ -320,6 +320,19 @@ class PPOLearner:
         if self.fused is not None:
             self.fused.refresh_masters()
 
+    def aux_checkpoint_state(self) -> dict:
+        """Optimizer moments for true resume (the reference's LearnerState
+        checkpoints include opt_states; template restore can't rebuild a
+        fresh optimizer's empty slots, so these ride as the aux payload)."""
+        return {
+            "actor_opt": self.actor_opt.state_dict(),
+            "critic_opt": self.critic_opt.state_dict(),
+        }
+
+    def load_aux_checkpoint_state(self, aux: dict) -> None:
+        self.actor_opt.load_state_dict(aux["actor_opt"])
+        self.critic_opt.load_state_dict(aux["critic_opt"])
+
 
 def learner_factory(config, env, device) -> PPOLearner:
     return PPOLearner(config, env, device)

@@ -47,12 +47,23 @@ class Checkpointer:
         self.save_best = save_best
         self.best_metric = best_metric
         self._best_value = float("-inf")
+        self._aux: Optional[dict] = None
         self.metadata = dict(metadata or {})
         self.metadata["checkpointer_version"] = CHECKPOINTER_VERSION
         with open(os.path.join(self.directory, "metadata.json"), "w") as f:
             json.dump(_jsonable(self.metadata), f, indent=2)
 
     # ------------------------------------------------------------------ save
+
+    def stage_aux(self, aux: Optional[dict]) -> None:
+        """Attach non-template state (optimizer state_dicts, counters) to
+        the NEXT save; written as aux.pt next to the safetensors payload.
+        The reference checkpoints the whole learner state incl. opt_state
+        (checkpointing.py save of the unreplicated LearnerState); template
+        restore cannot rebuild a fresh optimizer's empty state, so aux
+        rides alongside and restores through Optimizer.load_state_dict."""
+        self._aux = aux
+
 
     def save(self, timestep: int, state: Any, metric_value: Optional[float] = None) -> str:
         from safetensors.torch import save_file
@@ -71,6 +82,9 @@ class Checkpointer:
                 },
                 f,
             )
+        if self._aux is not None:
+            torch.save(self._aux, os.path.join(path, "aux.pt"))
+            self._aux = None
         if self.save_best and metric_value is not None and metric_value >= self._best_value:
             self._best_value = metric_value
             best = os.path.join(self.directory, "best")
@@ -125,6 +139,27 @@ class Checkpointer:
             path = os.path.join(self.directory, f"step_{timestep}")
         flat = load_file(os.path.join(path, "state.safetensors"))
         return _unflatten_into(template, flat)
+
+    def restore_aux(self, timestep: Optional[int] = None, best: bool = False) -> Optional[dict]:
+        """Load the aux.pt (optimizer state etc.) saved next to a
+        checkpoint, or None if that checkpoint has no aux payload."""
+        if best:
+            path = os.path.join(self.directory, "best")
+        else:
+            if timestep is None:
+                steps = sorted(
+                    int(d.split("_", 1)[1])
+                    for d in os.listdir(self.directory)
+                    if d.startswith("step_")
+                )
+                if not steps:
+                    raise FileNotFoundError(f"no checkpoints in {self.directory}")
+                timestep = steps[-1]
+            path = os.path.join(self.directory, f"step_{timestep}")
+        aux_path = os.path.join(path, "aux.pt")
+        if not os.path.exists(aux_path):
+            return None
+        return torch.load(aux_path, weights_only=False)
 
 
 def _unflatten_into(template: Any, flat: Dict[str, torch.Tensor], prefix: str = "") -> Any:

@@ -447,3 +447,48 @@ def test_distribution_numerics_vs_scipy_and_torch():
         ss.beta.logpdf(v.numpy(), alpha.numpy(), beta.numpy()), dtype=torch.float32
     ).sum(-1)
     torch.testing.assert_close(b.log_prob(v), ref, rtol=1e-4, atol=1e-5)
+
+
+def test_checkpoint_resume_restores_optimizer_moments(tmp_path):
+    """True-resume parity: the reference checkpoints the whole learner
+    state including opt_states. The aux payload must round-trip the Adam
+    moments so a restored learner continues with the same optimizer
+    trajectory (not freshly-zeroed moments)."""
+    import torch
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner
+    from stoix_amd.utils.checkpointing import Checkpointer
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8", "arch.total_timesteps=null",
+         "arch.num_updates=2", "arch.num_evaluation=1", "system.rollout_length=8",
+         "system.num_minibatches=2", "system.epochs=1", "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    torch.manual_seed(0)
+    env = environments.make_single(cfg, 8, torch.device("cpu"), seed=0)
+    learner = PPOLearner(cfg, env, torch.device("cpu"))
+    for _ in range(2):
+        learner.update_step()
+    ckpt = Checkpointer("resume", {}, directory=str(tmp_path))
+    ckpt.stage_aux(learner.aux_checkpoint_state())
+    ckpt.save(16, learner.state_for_checkpoint(), metric_value=1.0)
+
+    torch.manual_seed(99)
+    env2 = environments.make_single(cfg, 8, torch.device("cpu"), seed=0)
+    learner2 = PPOLearner(cfg, env2, torch.device("cpu"))
+    learner2.load_params(ckpt.restore_params(learner2.state_for_checkpoint()))
+    aux = ckpt.restore_aux()
+    assert aux is not None
+    learner2.load_aux_checkpoint_state(aux)
+    s1 = learner.actor_opt.state_dict()["state"]
+    s2 = learner2.actor_opt.state_dict()["state"]
+    assert set(s1) == set(s2) and len(s1) > 0
+    for k in s1:
+        torch.testing.assert_close(s1[k]["exp_avg"], s2[k]["exp_avg"])
+        torch.testing.assert_close(s1[k]["exp_avg_sq"], s2[k]["exp_avg_sq"])
