@@ -30,6 +30,26 @@ from stoix_amd.utils.logger import LogEvent, StoixLogger
 from stoix_amd.utils.total_timestep_checker import check_total_timesteps
 
 
+def default_aux_state(learner) -> dict:
+    """Optimizer state_dicts for every torch Optimizer the learner holds
+    (true-resume payload; learners can override with aux_checkpoint_state).
+    The reference checkpoints the full LearnerState incl. opt_states."""
+    return {
+        name: opt.state_dict()
+        for name, opt in vars(learner).items()
+        if isinstance(opt, torch.optim.Optimizer)
+    }
+
+
+def load_default_aux_state(learner, aux: dict) -> None:
+    """Counterpart of default_aux_state: restore optimizer moments by
+    attribute name, ignoring entries the learner does not have."""
+    for name, sd in (aux or {}).items():
+        opt = getattr(learner, name, None)
+        if isinstance(opt, torch.optim.Optimizer):
+            opt.load_state_dict(sd)
+
+
 def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = False) -> float:
     ctx = get_dist_context(force_cpu=force_cpu)
     device = ctx.device
@@ -119,8 +139,8 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
 
             if checkpointer is not None:
                 aux_fn = getattr(learner, "aux_checkpoint_state", None)
-                if aux_fn is not None:
-                    checkpointer.stage_aux(aux_fn())
+                aux = aux_fn() if aux_fn is not None else default_aux_state(learner)
+                checkpointer.stage_aux(aux or None)
                 checkpointer.save(t_env, learner.state_for_checkpoint(), mean_return)
             if mean_return >= best_return:
                 best_return = mean_return
