@@ -107,3 +107,39 @@ def test_logger_concurrent_threads_no_interleaving(tmp_path):
     assert len(parsed) == N_THREADS * N_LOGS
     seen = {int(p["episode_return"]) for p in parsed}
     assert len(seen) == N_THREADS * N_LOGS
+
+
+def test_evaluate_masks_post_done_rewards_exactly():
+    """evaluate() must accumulate each slot's reward ONLY until its first
+    done (autoreset keeps the env running underneath): cross-checked
+    against a hand-stepped env with identical seed and action stream."""
+    import torch
+
+    from stoix_amd.envs.classic import CartPole
+    from stoix_amd.evaluator import evaluate
+
+    B = 8
+
+    def act_fn(obs, greedy):
+        # deterministic position-dependent policy -> varied episode ends
+        return (obs[:, 0] > 0).long()
+
+    env = CartPole(B, seed=11)
+    out = evaluate(act_fn, env, greedy=True)
+
+    env2 = CartPole(B, seed=11)
+    ts = env2.reset()
+    ret = torch.zeros(B)
+    length = torch.zeros(B)
+    finished = torch.zeros(B, dtype=torch.bool)
+    for _ in range(env2.max_episode_steps + 1):
+        ts = env2.step(act_fn(ts.observation, True))
+        active = ~finished
+        ret += ts.reward * active
+        length += active.float()
+        finished |= ts.last()
+        if bool(finished.all()):
+            break
+    torch.testing.assert_close(out["episode_return"], ret)
+    torch.testing.assert_close(out["episode_length"], length)
+    assert bool(finished.all())
