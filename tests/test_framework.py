@@ -531,3 +531,53 @@ def test_default_aux_state_covers_all_learner_optimizers():
     s2 = learner2.q_opt.state_dict()["state"]
     for k in s1:
         torch.testing.assert_close(s1[k]["exp_avg"], s2[k]["exp_avg"])
+
+
+def test_noisy_linear_statistics_and_semantics():
+    """NoisyLinear (Fortunato et al. factorised noise): with mu=0, the
+    output's std over resamples matches sigma * E|f(eps_out)f(eps_in)| ...
+    checked empirically: Var[y_j] = sum_i sigma^2 E[(f_out f_in)^2] x_i^2 +
+    sigma_b^2 E[f_out^2], with E[f(eps)^2] = E[|eps|] = sqrt(2/pi).
+    Also: noise-off equals the plain linear; materialize caches the draw."""
+    import math
+
+    import torch
+
+    from stoix_amd.networks.layers import NoisyLinear
+
+    torch.manual_seed(0)
+    lin = NoisyLinear(32, 16, sigma_zero=0.5)
+    with torch.no_grad():
+        lin.weight_mu.zero_()
+        lin.bias_mu.zero_()
+    x = torch.randn(1, 32)
+    g = torch.Generator().manual_seed(1)
+    outs = []
+    for _ in range(4000):
+        lin.resample_noise(g)
+        outs.append(lin(x).detach())
+    y = torch.cat(outs)  # [4000, 16]
+    assert abs(float(y.mean())) < 0.02  # E[f(eps)] = 0
+    sig = 0.5 / math.sqrt(32)
+    e_abs = math.sqrt(2.0 / math.pi)  # E[f(eps)^2] = E|eps|
+    expect_var = (sig**2) * (e_abs**2) * float((x**2).sum()) + (sig**2) * e_abs
+    assert abs(float(y.var()) / expect_var - 1.0) < 0.15, (float(y.var()), expect_var)
+
+    # noise off == plain linear on the mus
+    with torch.no_grad():
+        lin.weight_mu.normal_()
+        lin.bias_mu.normal_()
+    lin.use_noise = False
+    torch.testing.assert_close(lin(x), torch.nn.functional.linear(x, lin.weight_mu, lin.bias_mu))
+    lin.use_noise = True
+
+    # materialize caches the current draw: same output until resample
+    lin.resample_noise(g)
+    lin.materialize()
+    a = lin(x).detach()
+    b = lin(x).detach()
+    torch.testing.assert_close(a, b)
+    lin._mat = None
+    lin.resample_noise(g)
+    c = lin(x).detach()
+    assert not torch.allclose(a, c)
