@@ -1442,3 +1442,48 @@ def test_grid_mirror_scores_against_mirrored_input():
     color = int(env._state["target"][0, idx])
     ts = env.step(torch.full((4,), 3 + color, dtype=torch.long))
     assert float(ts.reward[0]) == 1.0
+
+
+@pytest.mark.gpu
+def test_round2_envs_step_on_gpu():
+    """The round-2 env wave (gridworlds incl. sized variants, crafting,
+    memory games, N-link reacher, ARC slices, swing-up/balance) runs as
+    pure device tensor work on cuda (same generic torch path as the r1
+    wave; all declare capture_safe)."""
+    from stoix_amd.envs.arc import GridCopy, GridMirror
+    from stoix_amd.envs.classic import CartPoleBalance, CartPoleSwingUp
+    from stoix_amd.envs.crafting import Crafting, CraftingPixels
+    from stoix_amd.envs.pomdp import (
+        AutoEncodeEasy,
+        CountRecallEasy,
+        RepeatFirstEasy,
+        StatelessCartPole,
+    )
+    from stoix_amd.envs.reacher import ProceduralReacher, ProceduralReacher3
+    from stoix_amd.envs.xland import DoorKeyGrid, DoorKeyGrid5, EmptyGrid5, XLandGrid
+
+    dev = "cuda:0"
+    for cls, act in [
+        (XLandGrid, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (EmptyGrid5, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (DoorKeyGrid, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (DoorKeyGrid5, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (Crafting, lambda: torch.randint(0, 6, (8,), device=dev)),
+        (CraftingPixels, lambda: torch.randint(0, 6, (8,), device=dev)),
+        (RepeatFirstEasy, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (AutoEncodeEasy, lambda: torch.randint(0, 4, (8,), device=dev)),
+        (CountRecallEasy, lambda: torch.randint(0, 17, (8,), device=dev)),
+        (StatelessCartPole, lambda: torch.randint(0, 2, (8,), device=dev)),
+        (ProceduralReacher, lambda: torch.rand(8, 2, device=dev) * 2 - 1),
+        (ProceduralReacher3, lambda: torch.rand(8, 3, device=dev) * 2 - 1),
+        (GridCopy, lambda: torch.randint(0, 7, (8,), device=dev)),
+        (GridMirror, lambda: torch.randint(0, 7, (8,), device=dev)),
+        (CartPoleSwingUp, lambda: torch.rand(8, 1, device=dev) * 2 - 1),
+        (CartPoleBalance, lambda: torch.rand(8, 1, device=dev) * 2 - 1),
+    ]:
+        env = cls(num_envs=8, device=dev, seed=0)
+        ts = env.reset()
+        for _ in range(10):
+            ts = env.step(act())
+        assert ts.observation.is_cuda
+        assert torch.isfinite(ts.reward).all()
