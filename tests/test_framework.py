@@ -606,7 +606,7 @@ def test_mpo_temperature_dual_enforces_kl_budget():
         opt.zero_grad()
         dual.backward()
         opt.step()
-    eta = float(torch.nn.functional.softplus(log_eta))
+    eta = float(torch.nn.functional.softplus(log_eta.detach()))
     w = torch.softmax(q / eta, dim=0)
     # KL(w || uniform) per batch column, averaged
     kl = (w * (w.clamp_min(1e-12).log() + math.log(M))).sum(0).mean()
