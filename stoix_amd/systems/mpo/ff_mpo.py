@@ -77,13 +77,16 @@ class MPOLearner:
         # dual variables (softplus space)
         init = float(getattr(self.sys, "init_log_temperature", 10.0))
         init_a = float(getattr(self.sys, "init_log_alpha", 10.0))
+        # the reference's continuous configs pin the std dual much higher
+        # (init_log_alpha_stddev 500 vs alpha_mean 10) to freeze sigma early
+        init_as = float(getattr(self.sys, "init_log_alpha_stddev", init_a))
         self.log_temperature = nn.Parameter(torch.tensor(init, device=device))
         if self.discrete:
             self.log_alpha = nn.Parameter(torch.tensor(init_a, device=device))
             duals = [self.log_temperature, self.log_alpha]
         else:
             self.log_alpha_mean = nn.Parameter(torch.full((self.act_dim,), init_a, device=device))
-            self.log_alpha_std = nn.Parameter(torch.full((self.act_dim,), init_a, device=device))
+            self.log_alpha_std = nn.Parameter(torch.full((self.act_dim,), init_as, device=device))
             duals = [self.log_temperature, self.log_alpha_mean, self.log_alpha_std]
 
         self.actor_opt = torch.optim.Adam(self.actor.parameters(), lr=float(self.sys.actor_lr))

@@ -52,6 +52,9 @@ class VMPOLearner:
 
         init = float(getattr(self.sys, "init_log_temperature", 10.0))
         init_a = float(getattr(self.sys, "init_log_alpha", 10.0))
+        # the reference's continuous configs pin the std dual much higher
+        # (init_log_alpha_stddev 500 vs alpha_mean 10) to freeze sigma early
+        init_as = float(getattr(self.sys, "init_log_alpha_stddev", init_a))
         self.log_temperature = nn.Parameter(torch.tensor(init, device=device))
         if self.discrete:
             self.log_alpha = nn.Parameter(torch.tensor(init_a, device=device))
@@ -59,7 +62,7 @@ class VMPOLearner:
         else:
             act_dim = act_space.shape[0]
             self.log_alpha_mean = nn.Parameter(torch.full((act_dim,), init_a, device=device))
-            self.log_alpha_std = nn.Parameter(torch.full((act_dim,), init_a, device=device))
+            self.log_alpha_std = nn.Parameter(torch.full((act_dim,), init_as, device=device))
             duals = [self.log_temperature, self.log_alpha_mean, self.log_alpha_std]
         self.duals = duals
 
