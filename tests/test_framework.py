@@ -640,3 +640,32 @@ def test_mpo_decoupled_mstep_gradient_structure():
     lp_std = normal_logp(mu_t.unsqueeze(0), sigma_o.unsqueeze(0), a).sum()
     gm, gs = torch.autograd.grad(lp_std, [mu_o, sigma_o], allow_unused=True)
     assert gm is None and gs is not None and gs.abs().sum() > 0
+
+
+def test_spo_systematic_resample_properties():
+    """Systematic resampling (SPO's SMC backbone): (a) unbiased — the
+    resampled count of particle i is within 1 of P*w_i for EVERY draw
+    (the defining low-variance property of systematic resampling),
+    (b) expectation-preserving — weighted mean of values before equals
+    the unweighted mean after, up to that +-1/P quantisation."""
+    import torch
+
+    from stoix_amd.systems.spo.ff_spo import systematic_resample
+
+    g = torch.Generator().manual_seed(0)
+    B, P = 6, 64
+    logw = torch.randn(B, P, generator=g) * 2.0
+    w = torch.softmax(logw, dim=-1)
+    vals = torch.randn(B, P, generator=g)
+    for trial in range(20):
+        idx = systematic_resample(logw, g)
+        counts = torch.zeros(B, P)
+        counts.scatter_add_(1, idx, torch.ones(B, P))
+        # (a) deterministic-within-1 counts
+        assert torch.all((counts - P * w).abs() <= 1.0 + 1e-5), trial
+        # (b) post-resample mean ~= weighted mean: counts/P is within 1/P
+        # of w_i per particle, so |post - pre| <= (1/P) * sum_i |v_i|
+        post = (counts / P * vals).sum(-1)
+        pre = (w * vals).sum(-1)
+        bound = vals.abs().sum(-1) / P + 1e-6
+        assert torch.all((post - pre).abs() <= bound), trial
