@@ -288,3 +288,71 @@ def test_categorical_double_q_learning_scalar_reference():
         logits_tm1, atoms, a, r, d, logits_t, atoms, sel
     )
     assert abs(out.item() - expect) < 1e-4
+
+
+def test_categorical_td_learning_scalar_reference():
+    """Golden test: distributional TD (D4PG critic) equals the scalar
+    reconstruction — shift atoms by r + d*z, two-bin project, CE."""
+    import numpy as np
+    from scipy.special import log_softmax, softmax
+
+    g = torch.Generator().manual_seed(6)
+    B, N = 5, 9
+    atoms = torch.linspace(-3.0, 3.0, N)
+    logits_tm1 = torch.randn(B, N, generator=g)
+    logits_t = torch.randn(B, N, generator=g)
+    r = torch.randn(B, generator=g)
+    d = torch.full((B,), 0.95)
+    vmin, vmax = -3.0, 3.0
+    dz = (vmax - vmin) / (N - 1)
+    total = 0.0
+    for b in range(B):
+        p_t = softmax(logits_t[b].numpy())
+        proj = np.zeros(N)
+        for n in range(N):
+            z = min(max(float(r[b]) + 0.95 * float(atoms[n]), vmin), vmax)
+            pos = (z - vmin) / dz
+            lo = int(pos // 1)
+            hi = min(lo + 1, N - 1)
+            frac = pos - lo
+            proj[lo] += p_t[n] * (1 - frac)
+            proj[hi] += p_t[n] * frac
+        total += -(proj * log_softmax(logits_tm1[b].numpy())).sum()
+    out = L.categorical_td_learning(logits_tm1, atoms, r, d, logits_t, atoms)
+    assert abs(out.item() - total / B) < 1e-4
+
+
+def test_dpo_and_penalty_losses_scalar_reference():
+    """Golden tests at NON-trivial ratios: the DPO drift terms and the
+    penalty loss's (r-1)-log r KL estimator match scalar math."""
+    import math
+
+    lp = torch.tensor([0.4, -0.3, 0.2])
+    blp = torch.tensor([0.1, 0.1, 0.1])
+    adv = torch.tensor([1.5, -2.0, 0.5])
+    alpha, beta = 2.0, 0.6
+    total = 0.0
+    kl_total = 0.0
+    for i in range(3):
+        lr = float(lp[i] - blp[i])
+        r = math.exp(lr)
+        a = float(adv[i])
+        if a >= 0:
+            x = (r - 1.0) * a
+            drift = max(0.0, x - alpha * math.tanh(x / alpha))
+        else:
+            x = lr * a
+            drift = max(0.0, x - beta * math.tanh(x / beta))
+        total += -(r * a - drift)
+        kl_total += (r - 1.0) - lr
+    out = L.dpo_loss(lp, blp, adv, alpha=alpha, beta=beta)
+    assert abs(out.item() - total / 3) < 1e-5
+    coef = 3.0
+    loss, kl = L.ppo_penalty_loss(lp, blp, adv, coef)
+    expect = 0.0
+    for i in range(3):
+        lr = float(lp[i] - blp[i])
+        r = math.exp(lr)
+        expect += -(r * float(adv[i]) - coef * ((r - 1.0) - lr))
+    assert abs(loss.item() - expect / 3) < 1e-5
+    assert abs(kl.item() - kl_total / 3) < 1e-6
