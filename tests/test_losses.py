@@ -356,3 +356,36 @@ def test_dpo_and_penalty_losses_scalar_reference():
         expect += -(r * float(adv[i]) - coef * ((r - 1.0) - lr))
     assert abs(loss.item() - expect / 3) < 1e-5
     assert abs(kl.item() - kl_total / 3) < 1e-6
+
+
+def test_td_learning_and_quantile_q_learning_values():
+    """Golden values: 1-step TD (plain + Huber branch) and QR-DQN double-Q
+    selection — the loss must equal quantile_regression_loss applied to
+    the taken action's quantiles vs the selector-argmax target dist."""
+    v_tm1 = torch.tensor([1.0, -2.0])
+    r = torch.tensor([0.5, 1.0])
+    d = torch.tensor([0.9, 0.0])
+    v_t = torch.tensor([2.0, 7.0])
+    # plain: td = (0.5+1.8-1, 1+0-(-2)) = (1.3, 3.0); mse/2 mean
+    expect = 0.5 * (1.3**2 + 3.0**2) / 2
+    assert abs(L.td_learning(v_tm1, r, d, v_t).item() - expect) < 1e-6
+    # huber delta=1: |1.3|>1 -> 1*(1.3-0.5); |3|>1 -> (3-0.5)
+    expect_h = ((1.3 - 0.5) + (3.0 - 0.5)) / 2
+    assert abs(L.td_learning(v_tm1, r, d, v_t, 1.0).item() - expect_h) < 1e-6
+
+    g = torch.Generator().manual_seed(8)
+    B, N, A = 4, 6, 3
+    dist_tm1 = torch.randn(B, N, A, generator=g)
+    dist_t = torch.randn(B, N, A, generator=g)
+    sel = torch.randn(B, N, A, generator=g)
+    taus = (torch.arange(N, dtype=torch.float32) + 0.5) / N
+    a = torch.randint(0, A, (B,), generator=g)
+    r = torch.randn(B, generator=g)
+    d = torch.full((B,), 0.97)
+    out = L.quantile_q_learning(dist_tm1, taus, a, r, d, sel, dist_t)
+    best = sel.mean(dim=1).argmax(dim=-1)
+    bidx = torch.arange(B)
+    src = dist_tm1[bidx, :, a]
+    tgt = r.unsqueeze(-1) + d.unsqueeze(-1) * dist_t[bidx, :, best]
+    ref = L.quantile_regression_loss(src, taus, tgt, huber_param=1.0)
+    assert abs(out.item() - ref.item()) < 1e-6
