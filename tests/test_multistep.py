@@ -321,3 +321,13 @@ def test_retrace_matches_bruteforce_property():
                 assert abs(acc - float(out[s, b])) < 1e-4, (seed, s, b)
 
     run()
+
+
+def test_discounted_returns_scalar():
+    """Golden values for the plain discounted-return scan (REINFORCE)."""
+    r = _to_tb([1.0, 2.0, 3.0])
+    d = _to_tb([0.5, 0.5, 0.0])
+    v = _to_tb([0.0, 0.0, 0.0])
+    out = ms.batch_discounted_returns(r, d, v)
+    # G2 = 3; G1 = 2 + 0.5*3 = 3.5; G0 = 1 + 0.5*3.5 = 2.75
+    torch.testing.assert_close(out, _to_tb([2.75, 3.5, 3.0]))
