@@ -1024,12 +1024,19 @@ def test_graph_mode_step_equivalence_capture_safe_envs():
     rows only). Pinned for every capture_safe env family."""
     import torch
 
-    from stoix_amd.envs.crafting import Crafting
+    from stoix_amd.envs.arc import GridCopy, GridMirror
+    from stoix_amd.envs.crafting import Crafting, CraftingPixels
     from stoix_amd.envs.game2048 import Game2048
+    from stoix_amd.envs.pomdp import AutoEncodeEasy, CountRecallEasy, RepeatFirstEasy
     from stoix_amd.envs.snake import Snake
-    from stoix_amd.envs.xland import XLandGrid
+    from stoix_amd.envs.xland import DoorKeyGrid5, EmptyGrid5, XLandGrid
 
-    for cls, n_act in ((Snake, 4), (XLandGrid, 4), (Crafting, 6), (Game2048, 4)):
+    for cls, n_act in (
+        (Snake, 4), (XLandGrid, 4), (Crafting, 6), (Game2048, 4),
+        (CraftingPixels, 6), (EmptyGrid5, 4), (DoorKeyGrid5, 4),
+        (GridCopy, 7), (GridMirror, 7),
+        (RepeatFirstEasy, 4), (AutoEncodeEasy, 4), (CountRecallEasy, 17),
+    ):
         eager = cls(8, seed=3)
         graph = cls(8, seed=3)
         eager.reset()
