@@ -122,3 +122,46 @@ def test_sampled_mcts_prefers_rewarding_arm():
     # best arm (action +1) must dominate the visit distribution
     assert (out.action_weights.argmax(-1) == K - 1).float().mean() > 0.8
     assert (out.action.squeeze(-1) > 0.5).float().mean() > 0.8
+
+
+def test_mcts_backup_value_is_exact_on_deterministic_chain():
+    """Value-backup precision on a SELF-CONSISTENT deterministic model:
+    root arm 0 gives r=0.5, d=0.9 into a state worth exactly 2.0 (its own
+    expansion pays 2.0 then terminates, agreeing with the value net), so
+    EVERY simulation through arm 0 backs up exactly Q = 0.5 + 0.9*2.0 =
+    2.3 at any depth; arm 1 is worth exactly 0. The root search value
+    must converge to the visit-weighted mix and never exceed 2.3."""
+    A = 2
+
+    def fn(embedding, action):
+        B = action.shape[0]
+        depth = embedding["d"].squeeze(-1)
+        took0 = embedding["took0"].squeeze(-1)
+        at_root = depth == 0
+        new_took0 = torch.where(at_root, (action == 0).float(), took0)
+        reward = torch.where(
+            at_root,
+            torch.where(action == 0, torch.full((B,), 0.5), torch.zeros(B)),
+            torch.where(new_took0 > 0.5, torch.full((B,), 2.0), torch.zeros(B)),
+        )
+        discount = torch.where(at_root, torch.full((B,), 0.9), torch.zeros(B))
+        value = torch.where(
+            at_root & (action == 0), torch.full((B,), 2.0), torch.zeros(B)
+        )
+        new_emb = {"d": embedding["d"] + 1, "took0": new_took0.unsqueeze(-1)}
+        return new_emb, reward, discount, torch.zeros(B, A), value
+
+    out = mcts_search(
+        root_obs=torch.zeros(4, 2),
+        root_embedding={"d": torch.zeros(4, 1), "took0": torch.zeros(4, 1)},
+        root_prior_logits=torch.zeros(4, A),
+        root_value=torch.zeros(4),
+        recurrent_fn=fn,
+        num_simulations=60,
+        dirichlet_alpha=None,
+        temperature=0.0,
+    )
+    assert (out.action == 0).all()
+    assert (out.search_value <= 2.3 + 1e-5).all()
+    # visits concentrate on arm 0 -> the mixed value sits well above half
+    assert (out.search_value > 1.6).all(), out.search_value
