@@ -669,3 +669,24 @@ def test_spo_systematic_resample_properties():
         pre = (w * vals).sum(-1)
         bound = vals.abs().sum(-1) / P + 1e-6
         assert torch.all((post - pre).abs() <= bound), trial
+
+
+def test_scale_gradient_halves_backward_not_forward():
+    """MuZero's scale_gradient(x, 0.5) (reference jax_utils.py:12 /
+    ff_mz unroll): forward is identity, backward scales by 0.5 — and a
+    k-step unroll through it scales the state gradient by 0.5^k."""
+    import torch
+
+    from stoix_amd.systems.search.ff_mz import scale_gradient
+
+    x = torch.randn(5, requires_grad=True)
+    y = scale_gradient(x, 0.5)
+    torch.testing.assert_close(y, x)
+    (g,) = torch.autograd.grad(y.sum(), x)
+    torch.testing.assert_close(g, torch.full_like(x, 0.5))
+    # chained: 3 unroll steps -> 0.5^3
+    z = x
+    for _ in range(3):
+        z = scale_gradient(z * 1.0, 0.5)
+    (g,) = torch.autograd.grad(z.sum(), x)
+    torch.testing.assert_close(g, torch.full_like(x, 0.125))
