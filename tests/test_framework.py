@@ -690,3 +690,25 @@ def test_scale_gradient_halves_backward_not_forward():
         z = scale_gradient(z * 1.0, 0.5)
     (g,) = torch.autograd.grad(z.sum(), x)
     torch.testing.assert_close(g, torch.full_like(x, 0.125))
+
+
+def test_r2d2_value_transform_pair_is_inverse():
+    """R2D2's signed hyperbolic h and signed parabolic h^-1 (Pohlen et al.
+    transform pair) must be exact inverses over a wide range, odd, and
+    monotone — the properties the TD targets rely on."""
+    import torch
+
+    from stoix_amd.systems.q_learning.rec_r2d2 import (
+        signed_hyperbolic,
+        signed_parabolic,
+    )
+
+    x = torch.linspace(-500.0, 500.0, 2001, dtype=torch.float64)
+    h = signed_hyperbolic(x)
+    back = signed_parabolic(h)
+    torch.testing.assert_close(back, x, rtol=1e-6, atol=1e-6)
+    assert abs(float(signed_hyperbolic(torch.zeros(1, dtype=torch.float64)))) < 1e-12
+    torch.testing.assert_close(signed_hyperbolic(-x), -h)
+    assert torch.all(h[1:] > h[:-1])  # strictly increasing
+    # compresses: |h(x)| grows ~sqrt, so h(400) << 400
+    assert float(h[-1]) < 25.0
