@@ -712,3 +712,29 @@ def test_r2d2_value_transform_pair_is_inverse():
     assert torch.all(h[1:] > h[:-1])  # strictly increasing
     # compresses: |h(x)| grows ~sqrt, so h(400) << 400
     assert float(h[-1]) < 25.0
+
+
+def test_two_hot_encoding_properties():
+    """two_hot (MuZero/DisCo scalar targets): mass 1, expectation equals
+    the (clamped) input scalar, at most two adjacent non-zeros, exact
+    one-hot on atom values."""
+    import torch
+
+    from stoix_amd.networks.model_based import two_hot
+
+    atoms = torch.linspace(-5.0, 5.0, 21)
+    g = torch.Generator().manual_seed(0)
+    x = torch.rand(64, generator=g) * 14.0 - 7.0  # includes out-of-range
+    enc = two_hot(x, atoms)
+    torch.testing.assert_close(enc.sum(-1), torch.ones(64))
+    torch.testing.assert_close((enc * atoms).sum(-1), x.clamp(-5.0, 5.0), rtol=1e-5, atol=1e-6)
+    nz = (enc > 1e-8).sum(-1)
+    assert torch.all(nz <= 2)
+    # adjacency: the two mass-bearing bins are neighbours
+    for b in range(64):
+        on = (enc[b] > 1e-8).nonzero().flatten()
+        if on.numel() == 2:
+            assert int(on[1] - on[0]) == 1, (b, on)
+    # exact atoms -> one-hot
+    enc2 = two_hot(atoms.clone(), atoms)
+    torch.testing.assert_close(enc2, torch.eye(21), rtol=0, atol=1e-6)
