@@ -738,3 +738,38 @@ def test_two_hot_encoding_properties():
     # exact atoms -> one-hot
     enc2 = two_hot(atoms.clone(), atoms)
     torch.testing.assert_close(enc2, torch.eye(21), rtol=0, atol=1e-6)
+
+
+def test_discrete_valued_and_multidiscrete_distributions():
+    """DiscreteValuedDistribution (D4PG critic): mean = sum p*support and
+    mode = support[argmax]. MultiDiscreteDistribution: log_prob factorises
+    as the sum of per-dimension categorical log-probs."""
+    import torch
+
+    from stoix_amd.networks.distributions import (
+        Categorical,
+        DiscreteValuedDistribution,
+        MultiDiscreteDistribution,
+    )
+
+    g = torch.Generator().manual_seed(0)
+    logits = torch.randn(4, 11, generator=g)
+    support = torch.linspace(-2.0, 2.0, 11)
+    d = DiscreteValuedDistribution(logits, support)
+    p = torch.softmax(logits, -1)
+    torch.testing.assert_close(d.mean(), (p * support).sum(-1))
+    torch.testing.assert_close(d.mode(), support[logits.argmax(-1)])
+
+    nv = [3, 4, 2]
+    flat = torch.randn(5, sum(nv), generator=g)
+    md = MultiDiscreteDistribution(flat, nv)
+    a = torch.stack(
+        [torch.randint(0, n, (5,), generator=g) for n in nv], dim=-1
+    )
+    lp = md.log_prob(a)
+    expect = torch.zeros(5)
+    off = 0
+    for i, n in enumerate(nv):
+        expect = expect + Categorical(flat[:, off : off + n]).log_prob(a[:, i])
+        off += n
+    torch.testing.assert_close(lp, expect)
