@@ -773,3 +773,32 @@ def test_discrete_valued_and_multidiscrete_distributions():
         expect = expect + Categorical(flat[:, off : off + n]).log_prob(a[:, i])
         off += n
     torch.testing.assert_close(lp, expect)
+
+
+def test_dueling_decomposition_identity():
+    """Dueling Q nets: mean-centred advantages imply mean_a Q(s,a) = V(s)
+    — the identifiability constraint that motivates the architecture —
+    and the distributional variant satisfies it per atom (before the
+    softmax)."""
+    import torch
+
+    from stoix_amd.networks.dueling import (
+        DistributionalDuelingQNetwork,
+        DuelingQNetwork,
+    )
+
+    torch.manual_seed(0)
+    net = DuelingQNetwork(6, 4, layer_sizes=(32,))
+    x = torch.randn(10, 6)
+    q = net.q_values(x)
+    v = net.value(x).squeeze(-1)
+    torch.testing.assert_close(q.mean(-1), v, rtol=1e-5, atol=1e-5)
+
+    dnet = DistributionalDuelingQNetwork(6, 4, num_atoms=11, layer_sizes=(32,))
+    out = dnet(x)
+    logits = out.logits if hasattr(out, "logits") else out.q_logits
+    assert logits.shape == (10, 4, 11)
+    # per-atom identifiability: mean over actions of the dueled atom
+    # logits equals the value stream's atom logits
+    v_atoms = dnet.value(x)
+    torch.testing.assert_close(logits.mean(dim=1), v_atoms, rtol=1e-5, atol=1e-5)
