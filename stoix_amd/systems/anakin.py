@@ -100,6 +100,39 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             max_to_keep=config.logger.checkpointing.save_args.max_to_keep,
         )
 
+    # Restore-at-startup (reference ff_ppo.py:504-512: load_model rebuilds
+    # params from a saved checkpoint before training). checkpoint_uid is
+    # the DIRECTORY holding <system_name>/step_*/; optimizer moments are
+    # restored too when the checkpoint carries an aux payload.
+    load_cfg = getattr(config.logger.checkpointing, "load_model", False)
+    if load_cfg:
+        load_args = config.logger.checkpointing.load_args
+        load_dir = load_args.checkpoint_uid
+        if not load_dir:
+            raise ValueError(
+                "logger.checkpointing.load_model=true needs "
+                "logger.checkpointing.load_args.checkpoint_uid "
+                "(the directory that contains <system_name>/step_*/)"
+            )
+        loader = Checkpointer(
+            model_name=config.system.system_name, directory=str(load_dir)
+        )
+        ts_restore = load_args.timestep_to_restore
+        restored = loader.restore_params(
+            learner.state_for_checkpoint(),
+            timestep=None if ts_restore in (None, "null") else int(ts_restore),
+        )
+        learner.load_params(restored)
+        aux = loader.restore_aux(
+            timestep=None if ts_restore in (None, "null") else int(ts_restore)
+        )
+        if aux is not None:
+            aux_load = getattr(learner, "load_aux_checkpoint_state", None)
+            if aux_load is not None:
+                aux_load(aux)
+            else:
+                load_default_aux_state(learner, aux)
+
     num_updates_per_eval = int(config.arch.num_updates_per_eval)
     steps_per_update = (
         int(config.system.rollout_length) * num_envs * ctx.world_size

@@ -50,8 +50,12 @@ class Checkpointer:
         self._aux: Optional[dict] = None
         self.metadata = dict(metadata or {})
         self.metadata["checkpointer_version"] = CHECKPOINTER_VERSION
-        with open(os.path.join(self.directory, "metadata.json"), "w") as f:
-            json.dump(_jsonable(self.metadata), f, indent=2)
+        meta_path = os.path.join(self.directory, "metadata.json")
+        # a loader-only Checkpointer (no metadata given) must NOT clobber
+        # the saved run's metadata (and with it the version gate)
+        if metadata is not None or not os.path.exists(meta_path):
+            with open(meta_path, "w") as f:
+                json.dump(_jsonable(self.metadata), f, indent=2)
 
     # ------------------------------------------------------------------ save
 

@@ -802,3 +802,72 @@ def test_dueling_decomposition_identity():
     # logits equals the value stream's atom logits
     v_atoms = dnet.value(x)
     torch.testing.assert_close(logits.mean(dim=1), v_atoms, rtol=1e-5, atol=1e-5)
+
+
+def test_run_restores_from_checkpoint_at_startup(tmp_path):
+    """logger.checkpointing.load_model parity (reference ff_ppo.py:504-512):
+    a run pointed at a saved checkpoint starts from ITS params (and aux
+    optimizer moments), not from fresh init — and loading must not clobber
+    the saved metadata/version gate."""
+    import json
+
+    import torch
+
+    from stoix_amd import envs as environments
+    from stoix_amd.config import compose
+    from stoix_amd.systems.ppo.ff_ppo import PPOLearner, run
+    from stoix_amd.utils.checkpointing import Checkpointer
+    from stoix_amd.utils.total_timestep_checker import check_total_timesteps
+
+    cfg = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8", "arch.total_timesteps=null",
+         "arch.num_updates=2", "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+         "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+         "logger.loggers=[]"],
+    )
+    cfg.arch.n_devices = 1
+    check_total_timesteps(cfg)
+    torch.manual_seed(0)
+    env = environments.make_single(cfg, 8, torch.device("cpu"), seed=0)
+    learner = PPOLearner(cfg, env, torch.device("cpu"))
+    for _ in range(2):
+        learner.update_step()
+    ckpt = Checkpointer("ff_ppo", {"src": "test"}, directory=str(tmp_path))
+    ckpt.stage_aux(learner.aux_checkpoint_state())
+    ckpt.save(16, learner.state_for_checkpoint(), metric_value=1.0)
+    saved_w = {k: v.clone() for k, v in learner.actor.state_dict().items()}
+
+    cfg2 = compose(
+        "default/anakin/default_ff_ppo.yaml",
+        ["env=classic/cartpole", "arch.total_num_envs=8", "arch.total_timesteps=null",
+         "arch.num_updates=1", "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+         "arch.absolute_metric=false",
+         "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false",
+         "logger.checkpointing.load_model=true",
+         f"logger.checkpointing.load_args.checkpoint_uid={tmp_path}"],
+    )
+    # run() trains 1 more update from the restored weights; the restore
+    # itself is verified by rebuilding the learner the same way run does
+    # and catching the loaded weights before any update
+    import stoix_amd.systems.anakin as anakin_mod
+
+    captured = {}
+    orig = PPOLearner.update_step
+
+    def spy(self):
+        if "w" not in captured:
+            captured["w"] = {k: v.clone() for k, v in self.actor.state_dict().items()}
+        return orig(self)
+
+    PPOLearner.update_step = spy
+    try:
+        run(cfg2)
+    finally:
+        PPOLearner.update_step = orig
+    for k in saved_w:
+        torch.testing.assert_close(captured["w"][k], saved_w[k])
+    # metadata survived the loader
+    meta = json.loads((tmp_path / "ff_ppo" / "metadata.json").read_text())
+    assert meta.get("src") == "test"
