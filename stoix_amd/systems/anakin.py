@@ -98,6 +98,7 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             metadata=config.to_plain() if hasattr(config, "to_plain") else dict(config),
             directory=f"{logger.directory}/checkpoints",
             max_to_keep=config.logger.checkpointing.save_args.max_to_keep,
+            keep_period=config.logger.checkpointing.save_args.keep_period,
         )
 
     # Restore-at-startup (reference ff_ppo.py:504-512: load_model rebuilds
@@ -170,7 +171,11 @@ def run_anakin_experiment(config, learner_factory: Callable, force_cpu: bool = F
             final_return = mean_return
             logger.log(eval_metrics, t_env, eval_idx, LogEvent.EVAL)
 
-            if checkpointer is not None:
+            save_every = int(
+                getattr(config.logger.checkpointing.save_args, "save_interval_steps", 1)
+                or 1
+            )
+            if checkpointer is not None and eval_idx % save_every == 0:
                 aux_fn = getattr(learner, "aux_checkpoint_state", None)
                 aux = aux_fn() if aux_fn is not None else default_aux_state(learner)
                 checkpointer.stage_aux(aux or None)

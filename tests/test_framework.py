@@ -871,3 +871,16 @@ def test_run_restores_from_checkpoint_at_startup(tmp_path):
     # metadata survived the loader
     meta = json.loads((tmp_path / "ff_ppo" / "metadata.json").read_text())
     assert meta.get("src") == "test"
+
+
+def test_checkpointer_keep_period_and_save_interval(tmp_path):
+    """save_args parity: keep_period protects periodic checkpoints from
+    max_to_keep GC, and save_interval_steps thins the per-eval saves."""
+    ckpt = Checkpointer("m", {}, directory=str(tmp_path), max_to_keep=1, keep_period=20)
+    for t in (10, 20, 30, 40):
+        ckpt.save(t, {"w": torch.tensor([float(t)])})
+    kept = sorted(
+        int(d.split("_")[1]) for d in os.listdir(tmp_path / "m") if d.startswith("step_")
+    )
+    # 20 and 40 survive via keep_period (multiples of 20); 40 is also latest
+    assert kept == [20, 40], kept
