@@ -267,6 +267,20 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     param_server = ParameterServer(n_actors)
     timers = TimingTracker()
     logger = StoixLogger(config) if ctx.is_main else None
+    checkpointer = None
+    if ctx.is_main and logger is not None and config.logger.checkpointing.save_model:
+        from stoix_amd.utils.checkpointing import Checkpointer
+
+        # reference Sebulba parity: the async evaluator saves the evaluated
+        # snapshot per eval, best-by-return retained
+        checkpointer = Checkpointer(
+            model_name=config.system.system_name,
+            metadata=config.to_plain() if hasattr(config, "to_plain") else dict(config),
+            directory=f"{logger.directory}/checkpoints",
+            max_to_keep=config.logger.checkpointing.save_args.max_to_keep,
+            keep_period=config.logger.checkpointing.save_args.keep_period,
+        )
+
 
     # evaluation: fresh eval env on the evaluator device's host side
     eval_env = env_factory(int(config.arch.num_eval_episodes))
@@ -286,6 +300,10 @@ def run_experiment(config, force_cpu: bool = False) -> float:
         m = evaluate(act, eval_env, greedy=bool(config.arch.evaluation_greedy))
         if logger is not None:
             logger.log(m, t_env, 0, LogEvent.EVAL)
+        if checkpointer is not None:
+            checkpointer.save(
+                t_env, params, metric_value=float(m["episode_return"].mean())
+            )
         return m
 
     async_eval = AsyncEvaluator(evaluate_snapshot, lifetime)

