@@ -191,6 +191,26 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     obs_space, act_space = probe_env.observation_space, probe_env.action_space
     learner = SharedImpalaLearner(config, learner_device, obs_space, act_space)
 
+
+    # restore-at-startup (reference sebulba ff_ppo.py:783-789)
+    if getattr(config.logger.checkpointing, "load_model", False):
+        from stoix_amd.utils.checkpointing import Checkpointer
+
+        load_args = config.logger.checkpointing.load_args
+        if not load_args.checkpoint_uid:
+            raise ValueError(
+                "load_model=true needs logger.checkpointing.load_args.checkpoint_uid"
+            )
+        loader = Checkpointer(
+            model_name=config.system.system_name,
+            directory=str(load_args.checkpoint_uid),
+        )
+        ts = load_args.timestep_to_restore
+        restored = loader.restore_params(
+            {"net": learner.net.state_dict()}, timestep=None if ts in (None, "null") else int(ts)
+        )
+        for _name, _mod in {"net": learner.net}.items():
+            _mod.load_state_dict(restored[_name])
     lifetime = ThreadLifetime()
     pipeline = OnPolicyPipeline(n_actors)
     param_server = ParameterServer(n_actors)
@@ -265,6 +285,10 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     async_eval = AsyncEvaluator(evaluate_snapshot, lifetime)
 
     num_updates = int(config.arch.num_updates)
+    # publish the INITIAL learner params before any actor starts: rollout 0
+    # then acts with the learner's weights (matters after a load_model
+    # restore; otherwise actors would spend rollout 0 on their own init)
+    param_server.distribute_params(learner.cpu_params())
     threads = []
     for a_id in range(n_actors):
         dev = _device_of(actor_device_ids[a_id // per_dev], force_cpu)

@@ -263,6 +263,26 @@ def run_experiment(config, force_cpu: bool = False) -> float:
 
     learner = SebulbaPPOLearner(config, learner_device, obs_space, act_space)
     lifetime = ThreadLifetime()
+
+    # restore-at-startup (reference sebulba ff_ppo.py:783-789)
+    if getattr(config.logger.checkpointing, "load_model", False):
+        from stoix_amd.utils.checkpointing import Checkpointer
+
+        load_args = config.logger.checkpointing.load_args
+        if not load_args.checkpoint_uid:
+            raise ValueError(
+                "load_model=true needs logger.checkpointing.load_args.checkpoint_uid"
+            )
+        loader = Checkpointer(
+            model_name=config.system.system_name,
+            directory=str(load_args.checkpoint_uid),
+        )
+        ts = load_args.timestep_to_restore
+        restored = loader.restore_params(
+            {"actor": learner.actor.state_dict(), "critic": learner.critic.state_dict()}, timestep=None if ts in (None, "null") else int(ts)
+        )
+        for _name, _mod in {"actor": learner.actor, "critic": learner.critic}.items():
+            _mod.load_state_dict(restored[_name])
     pipeline = OnPolicyPipeline(n_actors)
     param_server = ParameterServer(n_actors)
     timers = TimingTracker()
@@ -313,6 +333,10 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     prewarm_convs(config, learner.actor, learner.critic, obs_space.shape, learner_device)
 
     num_updates = int(config.arch.num_updates)
+    # publish the INITIAL learner params before any actor starts: rollout 0
+    # then acts with the learner's weights (matters after a load_model
+    # restore; otherwise actors would spend rollout 0 on their own init)
+    param_server.distribute_params(learner.cpu_params())
     threads = []
     for a_id in range(n_actors):
         dev = _device_of(actor_device_ids[a_id // per_dev], force_cpu)

@@ -375,3 +375,57 @@ def test_sebulba_saves_checkpoints(tmp_path):
     flat = load_file(str(ckpt_dirs[0] / "state.safetensors"))
     assert any(k.startswith("actor.") for k in flat)
     assert all(torch.isfinite(v).all() for v in flat.values())
+
+
+def test_sebulba_restores_checkpoint_at_startup(tmp_path):
+    """Sebulba load_model parity (reference sebulba ff_ppo.py:783-789):
+    a second run pointed at the first run's checkpoint starts from its
+    saved actor params."""
+    import torch
+
+    from stoix_amd.systems.ppo.sebulba_ff_ppo import SebulbaPPOLearner, run
+    from safetensors.torch import load_file
+
+    base = [
+        "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
+        "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+        "arch.actor.actor_per_device=2",
+        "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+        "logger.loggers=[]",
+    ]
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        base + ["logger.checkpointing.save_model=true",
+                f"logger.base_exp_path={tmp_path}"],
+    )
+    run(cfg)
+    step_dir = sorted(tmp_path.rglob("step_*"))[-1]
+    saved = load_file(str(step_dir / "state.safetensors"))
+    ckpt_root = step_dir.parent.parent  # .../checkpoints
+
+    captured = {}
+    orig = SebulbaPPOLearner.learn
+
+    def spy(self, payloads):
+        # capture BEFORE the first update applies: these must be exactly
+        # the restored checkpoint weights
+        if "actor" not in captured:
+            captured["actor"] = {
+                k: v.detach().cpu().clone() for k, v in self.actor.state_dict().items()
+            }
+        return orig(self, payloads)
+
+    SebulbaPPOLearner.learn = spy
+    try:
+        cfg2 = compose(
+            "default/sebulba/default_ff_ppo.yaml",
+            base + ["logger.checkpointing.save_model=false",
+                    "logger.checkpointing.load_model=true",
+                    f"logger.checkpointing.load_args.checkpoint_uid={ckpt_root}",
+                    f"logger.base_exp_path={tmp_path}/second"],
+        )
+        run(cfg2)
+    finally:
+        SebulbaPPOLearner.learn = orig
+    for k, v in captured["actor"].items():
+        torch.testing.assert_close(v, saved[f"actor.{k}"])
