@@ -204,6 +204,24 @@ def run_experiment(config, force_cpu: bool = False) -> float:
             )
         return m
 
+
+    def absolute_snapshot_eval(params: Dict) -> Dict:
+        from stoix_amd.evaluator import evaluate
+
+        eval_actor.load_state_dict(params["actor"])
+
+        def act(obs, greedy):
+            dist = eval_actor(obs.to(eval_device))
+            return (dist.mode() if greedy else dist.sample(eval_gen)).cpu()
+
+        outs = [
+            evaluate(act, eval_env, greedy=bool(config.arch.evaluation_greedy))
+            for _ in range(10)
+        ]
+        return {
+            k: torch.cat([o[k].reshape(-1) for o in outs]) for k in outs[0]
+        }
+
     async_eval = AsyncEvaluator(evaluate_snapshot, lifetime)
 
     num_updates = int(config.arch.num_updates)
@@ -250,6 +268,17 @@ def run_experiment(config, force_cpu: bool = False) -> float:
     async_eval.join()
     if async_eval.last_metrics:
         final_return = float(async_eval.last_metrics["episode_return"].mean())
+    # absolute metric: 10x episodes with the BEST evaluated params
+    # (reference sebulba ff_ppo.py:994-1012)
+    if (
+        ctx.is_main
+        and bool(getattr(config.arch, "absolute_metric", True))
+        and async_eval.best_params is not None
+    ):
+        abs_m = absolute_snapshot_eval(async_eval.best_params)
+        if logger is not None:
+            logger.log(abs_m, t_env, int(config.arch.num_evaluation), LogEvent.ABSOLUTE)
+        final_return = float(abs_m["episode_return"].mean())
     if logger is not None:
         logger.close()
     return final_return

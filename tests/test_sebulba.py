@@ -68,7 +68,7 @@ def test_sebulba_ppo_end_to_end():
         "default/sebulba/default_ff_ppo.yaml",
         [
             "arch.total_num_envs=8", "arch.total_timesteps=null", "arch.num_updates=3",
-            "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "arch.num_evaluation=1", "arch.absolute_metric=false", "arch.num_eval_episodes=4",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
             "logger.loggers=[]", "logger.checkpointing.save_model=false",
@@ -85,7 +85,7 @@ def test_sebulba_impala_end_to_end():
         "default/sebulba/default_ff_impala.yaml",
         [
             "arch.total_num_envs=8", "arch.total_timesteps=null", "arch.num_updates=3",
-            "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "arch.num_evaluation=1", "arch.absolute_metric=false", "arch.num_eval_episodes=4",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2",
             "logger.loggers=[]", "logger.checkpointing.save_model=false",
@@ -102,7 +102,7 @@ def test_sebulba_impala_shared_torso_end_to_end():
         "default/sebulba/default_ff_impala_shared_torso.yaml",
         [
             "arch.total_num_envs=8", "arch.total_timesteps=null", "arch.num_updates=3",
-            "arch.num_evaluation=1", "arch.num_eval_episodes=4",
+            "arch.num_evaluation=1", "arch.absolute_metric=false", "arch.num_eval_episodes=4",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2",
             "logger.loggers=[]", "logger.checkpointing.save_model=false",
@@ -124,6 +124,7 @@ def test_sebulba_ppo_breakout_pixels():
             "env=envpool/breakout", "network=cnn",
             "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
             "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.absolute_metric=false",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
             "logger.loggers=[]", "logger.checkpointing.save_model=false",
@@ -335,6 +336,7 @@ def test_sebulba_ppo_new_pool_game_end_to_end():
             "env=envpool/phoenix", "network=cnn",
             "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
             "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.absolute_metric=false",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
             "logger.loggers=[]", "logger.checkpointing.save_model=false",
@@ -358,6 +360,7 @@ def test_sebulba_saves_checkpoints(tmp_path):
         [
             "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
             "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.absolute_metric=false",
             "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
             "logger.loggers=[]", "logger.checkpointing.save_model=true",
@@ -389,6 +392,7 @@ def test_sebulba_restores_checkpoint_at_startup(tmp_path):
     base = [
         "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
         "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.absolute_metric=false",
         "arch.actor.actor_per_device=2",
         "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
         "logger.loggers=[]",
