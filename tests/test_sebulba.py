@@ -433,3 +433,34 @@ def test_sebulba_restores_checkpoint_at_startup(tmp_path):
         SebulbaPPOLearner.learn = orig
     for k, v in captured["actor"].items():
         torch.testing.assert_close(v, saved[f"actor.{k}"])
+
+
+def test_sebulba_absolute_metric_logged(tmp_path):
+    """With arch.absolute_metric=true, the run ends with a 10x-episode
+    ABSOLUTE evaluation of the best params (reference sebulba
+    ff_ppo.py:994-1012), emitted to the sinks."""
+    import json
+
+    from stoix_amd.systems.ppo.sebulba_ff_ppo import run
+
+    cfg = compose(
+        "default/sebulba/default_ff_ppo.yaml",
+        [
+            "arch.total_num_envs=4", "arch.total_timesteps=null", "arch.num_updates=2",
+            "arch.num_evaluation=1", "arch.num_eval_episodes=2",
+            "arch.absolute_metric=true",
+            "arch.actor.actor_per_device=2",
+            "system.rollout_length=8", "system.num_minibatches=2", "system.epochs=1",
+            "logger.loggers=[json]", "logger.checkpointing.save_model=false",
+            f"logger.base_exp_path={tmp_path}",
+        ],
+    )
+    run(cfg)
+    lines = []
+    for f in tmp_path.rglob("*.json*"):
+        lines += [json.loads(l) for l in f.read_text().splitlines() if l.startswith("{")]
+    events = {l.get("event") for l in lines}
+    assert "absolute" in events, events
+    # 10x the eval episodes
+    abs_rows = [l for l in lines if l.get("event") == "absolute"]
+    assert abs_rows
