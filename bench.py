@@ -276,6 +276,9 @@ def bench_sebulba(args, ctx) -> None:
             f"arch.num_updates={updates}",
             "arch.num_evaluation=1",
             "arch.num_eval_episodes=4",
+            # no 10x absolute eval at shutdown: the bench measures the
+            # train phase; last_sps is recorded before the drain either way
+            "arch.absolute_metric=false",
             f"arch.actor.actor_per_device={actors}",
             f"system.rollout_length={rollout}",
             "system.num_minibatches=2",
