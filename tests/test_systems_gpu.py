@@ -84,7 +84,8 @@ def test_gpu_sebulba_ppo_breakout():
             "env=envpool/breakout", "network=cnn",
             "arch.total_num_envs=8", "arch.total_timesteps=null",
             "arch.num_updates=2", "arch.num_evaluation=1",
-            "arch.num_eval_episodes=2", "arch.actor.actor_per_device=2",
+            "arch.num_eval_episodes=2", "arch.absolute_metric=false",
+            "arch.actor.actor_per_device=2",
             "system.rollout_length=8", "system.num_minibatches=2",
             "system.epochs=1", "logger.loggers=[]",
             "logger.checkpointing.save_model=false",
