@@ -136,7 +136,7 @@ def test_sum_tree_random_ops_match_naive_reference():
     from hypothesis import given, settings
     from hypothesis import strategies as st
 
-    @settings(max_examples=30, deadline=None)
+    @settings(max_examples=30, deadline=None, derandomize=True)
     @given(
         st.lists(
             st.lists(
@@ -190,7 +190,7 @@ def test_trajectory_windows_valid_under_random_adds_property():
 
     from stoix_amd.buffers.trajectory import TrajectoryBuffer
 
-    @settings(max_examples=20, deadline=None)
+    @settings(max_examples=20, deadline=None, derandomize=True)
     @given(st.integers(0, 10_000))
     def run(seed):
         g = torch.Generator().manual_seed(seed)

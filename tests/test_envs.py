@@ -1521,7 +1521,7 @@ def test_game2048_slide_matches_scalar_reference_property():
                 i += 1
         return out + [0] * (4 - len(out)), rew
 
-    @settings(max_examples=40, deadline=None)
+    @settings(max_examples=40, deadline=None, derandomize=True)
     @given(st.integers(0, 100_000))
     def run(seed):
         g = torch.Generator().manual_seed(seed)

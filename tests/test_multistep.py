@@ -195,7 +195,7 @@ def test_gae_matches_bruteforce_property():
 
     from stoix_amd.ops import multistep as ms
 
-    @settings(max_examples=25, deadline=None)
+    @settings(max_examples=25, deadline=None, derandomize=True)
     @given(st.integers(0, 10_000))
     def run(seed):
         g = torch.Generator().manual_seed(seed)
@@ -241,7 +241,7 @@ def test_vtrace_matches_bruteforce_property():
 
     from stoix_amd.ops import multistep as ms
 
-    @settings(max_examples=25, deadline=None)
+    @settings(max_examples=25, deadline=None, derandomize=True)
     @given(st.integers(0, 10_000))
     def run(seed):
         g = torch.Generator().manual_seed(seed)
@@ -291,7 +291,7 @@ def test_retrace_matches_bruteforce_property():
 
     from stoix_amd.ops import multistep as ms
 
-    @settings(max_examples=25, deadline=None)
+    @settings(max_examples=25, deadline=None, derandomize=True)
     @given(st.integers(0, 10_000))
     def run(seed):
         g = torch.Generator().manual_seed(seed)
