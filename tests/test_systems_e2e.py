@@ -616,3 +616,64 @@ def test_ppo_learns_sized_gridworlds():
         )
         r = run(cfg)
         assert r > thr, f"{env_name} not learned: return={r}"
+
+
+@pytest.mark.slow
+def test_awr_learns_identity():
+    """EM-style family learning gate: AWR solves the identity debug game
+    to the optimal 10.0 at this budget (weighted regression on the
+    exp-advantage weights end-to-end)."""
+    from stoix_amd.systems.awr.ff_awr import run
+
+    cfg = compose(
+        "default/anakin/default_ff_awr.yaml",
+        ["env=debug/identity", "arch.total_num_envs=64",
+         "arch.total_timesteps=null", "arch.num_updates=60",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=8",
+         "system.batch_size=64", "system.buffer_size=4096",
+         "system.sample_sequence_length=4",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 8.0, f"AWR did not learn identity: {r}"
+
+
+@pytest.mark.slow
+def test_vmpo_learns_identity():
+    """V-MPO learning gate: solves the identity game to ~10 (measured
+    9.94; chance is 2.5) — validates the E-step weighting + KL-dual
+    M-step end-to-end."""
+    from stoix_amd.systems.mpo.ff_vmpo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_vmpo.yaml",
+        ["env=debug/identity", "arch.total_num_envs=64",
+         "arch.total_timesteps=null", "arch.num_updates=1600",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=8",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 8.0, f"V-MPO did not learn identity: {r}"
+
+
+@pytest.mark.slow
+def test_mpo_learns_identity_above_chance():
+    """MPO learning gate: clearly beats chance (2.5) on the identity game
+    (measured 6.6 at this budget; retrace critic + temperature/alpha
+    duals end-to-end)."""
+    from stoix_amd.systems.mpo.ff_mpo import run
+
+    cfg = compose(
+        "default/anakin/default_ff_mpo.yaml",
+        ["env=debug/identity", "arch.total_num_envs=64",
+         "arch.total_timesteps=null", "arch.num_updates=600",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=8",
+         "system.batch_size=128", "system.buffer_size=8192",
+         "system.sample_sequence_length=4", "system.epochs=8",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 4.0, f"MPO below the learning bar: {r}"
