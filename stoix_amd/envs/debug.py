@@ -35,9 +35,10 @@ class IdentityGame(StatefulVecEnv):
         return torch.nn.functional.one_hot(state["sym"], self.num_symbols).to(torch.float32)
 
     def _step_fn(self, state, action) -> Tuple[State, Tensor, Tensor]:
+        n = action.shape[0]
         reward = (action.long() == state["sym"]).to(torch.float32)
-        new_sym = self.randint(self.num_symbols, self.num_envs)
-        terminated = torch.zeros(self.num_envs, dtype=torch.bool, device=self.device)
+        new_sym = self.randint(self.num_symbols, n)
+        terminated = torch.zeros(n, dtype=torch.bool, device=self.device)
         return {"sym": new_sym}, reward, terminated
 
 
@@ -66,14 +67,15 @@ class SequenceGame(StatefulVecEnv):
         return torch.cat([one_hot * visible, t_frac], dim=-1)
 
     def _step_fn(self, state, action) -> Tuple[State, Tensor, Tensor]:
+        n = action.shape[0]
         t = state["t"] + 1
         is_final = t >= self.max_episode_steps
         reward = torch.where(
             is_final & (action.long() == state["sym"]),
-            torch.ones(self.num_envs, device=self.device),
-            torch.zeros(self.num_envs, device=self.device),
+            torch.ones(n, device=self.device),
+            torch.zeros(n, device=self.device),
         )
-        terminated = torch.zeros(self.num_envs, dtype=torch.bool, device=self.device)
+        terminated = torch.zeros(n, dtype=torch.bool, device=self.device)
         return {"sym": state["sym"], "t": t}, reward, terminated
 
 
@@ -104,10 +106,11 @@ class DelayedRewardGame(StatefulVecEnv):
         chose_good = torch.where(t == 0, action.long() == 1, state["chose_good"])
         t = t + 1
         is_final = t >= self.max_episode_steps
+        n = action.shape[0]
         reward = torch.where(
-            is_final, torch.where(chose_good, 1.0, -1.0), torch.zeros(self.num_envs, device=self.device)
+            is_final, torch.where(chose_good, 1.0, -1.0), torch.zeros(n, device=self.device)
         )
-        terminated = torch.zeros(self.num_envs, dtype=torch.bool, device=self.device)
+        terminated = torch.zeros(n, dtype=torch.bool, device=self.device)
         return {"t": t, "chose_good": chose_good}, reward, terminated
 
 

@@ -1536,3 +1536,23 @@ def test_game2048_slide_matches_scalar_reference_property():
             assert abs(float(r[e]) - expect_rew) < 1e-5, (seed, e)
 
     run()
+
+
+def test_env_step_fn_is_batch_size_agnostic():
+    """The functional ``_step_fn(state, action)`` must size every output by
+    the INCOMING batch, not the env's own num_envs: search systems (AZ)
+    drive the TRAIN env's step function with the EVAL env's smaller state
+    batch as the world model. Regression for a real crash found by an
+    AZ-on-identity probe (arena size 32 vs eval batch 16)."""
+    from stoix_amd.envs.debug import DEBUG_ENVIRONMENTS
+
+    for name, cls in DEBUG_ENVIRONMENTS.items():
+        env = cls(8, seed=0)
+        env.reset()
+        small = env._reset_fn(3)
+        a = torch.zeros(3, dtype=torch.long)
+        new_state, reward, terminated = env._step_fn(small, a)
+        assert reward.shape == (3,), (name, reward.shape)
+        assert terminated.shape == (3,), name
+        for k, v in new_state.items():
+            assert v.shape[0] == 3, (name, k, v.shape)

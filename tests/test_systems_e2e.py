@@ -677,3 +677,25 @@ def test_mpo_learns_identity_above_chance():
     )
     r = run(cfg)
     assert r > 4.0, f"MPO below the learning bar: {r}"
+
+
+@pytest.mark.slow
+def test_az_learns_identity():
+    """Search-family learning gate: AlphaZero (real-env model, 8 sims)
+    solves the identity game to the optimal 10.0 — also pins the search
+    EVALUATOR path with an eval batch different from the train batch
+    (regression: env._step_fn batch-safety)."""
+    from stoix_amd.systems.search.ff_az import run
+
+    cfg = compose(
+        "default/anakin/default_ff_az.yaml",
+        ["env=debug/identity", "arch.total_num_envs=32",
+         "arch.total_timesteps=null", "arch.num_updates=40",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=8",
+         "system.num_simulations=8", "system.num_minibatches=2",
+         "system.epochs=2", "logger.loggers=[]",
+         "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 8.0, f"AZ did not learn identity: {r}"
