@@ -207,6 +207,10 @@ class DoorKeyGrid(StatefulVecEnv):
             "has_key": torch.zeros(n, device=dev),
             "agent_r": agent_r.float(), "agent_c": agent_c.float(),
             "goal_r": goal_r.float(), "goal_c": goal_c.float(),
+            # episode time lives IN the state so _step_fn stays functional
+            # (search systems drive it with foreign batches); mirrors
+            # self._step_count under normal stepping
+            "t": torch.zeros(n, device=dev),
         }
 
     def _obs_fn(self, state: State) -> Tensor:
@@ -245,8 +249,9 @@ class DoorKeyGrid(StatefulVecEnv):
         on_key = (nr == state["key_r"].long()) & (nc == state["key_c"].long()) & ~has_key
         new_has_key = has_key | on_key
         at_goal = (nr == state["goal_r"].long()) & (nc == state["goal_c"].long())
-        # MiniGrid terminal shaping: 1 - 0.9 * t / T
-        frac = self._step_count.float() / float(self.max_episode_steps)
+        # MiniGrid terminal shaping: 1 - 0.9 * t / T (t carried in state)
+        t = state["t"] + 1.0
+        frac = t / float(self.max_episode_steps)
         reward = at_goal.float() * (1.0 - 0.9 * frac)
         return (
             {
@@ -257,6 +262,7 @@ class DoorKeyGrid(StatefulVecEnv):
                 "has_key": new_has_key.float(),
                 "agent_r": nr.float(), "agent_c": nc.float(),
                 "goal_r": state["goal_r"], "goal_c": state["goal_c"],
+                "t": t,
             },
             reward,
             at_goal,

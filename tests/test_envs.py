@@ -1544,14 +1544,28 @@ def test_env_step_fn_is_batch_size_agnostic():
     drive the TRAIN env's step function with the EVAL env's smaller state
     batch as the world model. Regression for a real crash found by an
     AZ-on-identity probe (arena size 32 vs eval batch 16)."""
+    from stoix_amd.envs.arc import GridCopy
+    from stoix_amd.envs.classic import CartPole, Pendulum
+    from stoix_amd.envs.crafting import Crafting
     from stoix_amd.envs.debug import DEBUG_ENVIRONMENTS
+    from stoix_amd.envs.game2048 import Game2048
+    from stoix_amd.envs.snake import Snake
+    from stoix_amd.envs.xland import DoorKeyGrid, XLandGrid
 
-    for name, cls in DEBUG_ENVIRONMENTS.items():
+    classes = list(DEBUG_ENVIRONMENTS.values()) + [
+        CartPole, Pendulum, Snake, Game2048, XLandGrid, DoorKeyGrid,
+        Crafting, GridCopy,
+    ]
+    for cls in classes:
         env = cls(8, seed=0)
         env.reset()
         small = env._reset_fn(3)
-        a = torch.zeros(3, dtype=torch.long)
+        if hasattr(env.action_space, "num_values"):
+            a = torch.zeros(3, dtype=torch.long)
+        else:
+            a = torch.zeros(3, *env.action_space.shape)
         new_state, reward, terminated = env._step_fn(small, a)
+        name = cls.__name__
         assert reward.shape == (3,), (name, reward.shape)
         assert terminated.shape == (3,), name
         for k, v in new_state.items():
