@@ -699,3 +699,30 @@ def test_az_learns_identity():
     )
     r = run(cfg)
     assert r > 8.0, f"AZ did not learn identity: {r}"
+
+
+def test_search_systems_eval_batch_mismatch():
+    """Search evaluators drive the TRAIN env's functional step with the
+    EVAL env's batch (32 train envs vs 16 eval episodes here) — the shape
+    that crashed AZ before envs' _step_fn became batch-agnostic. Quick
+    2-update smokes for SPO and Sampled-AZ."""
+    import importlib
+
+    for mod_path, default, extra in [
+        ("stoix_amd.systems.spo.ff_spo", "default/anakin/default_ff_spo.yaml",
+         ["system.num_particles=4", "system.search_depth=2",
+          "system.num_minibatches=2", "system.epochs=1"]),
+        ("stoix_amd.systems.search.ff_sampled_az",
+         "default/anakin/default_ff_sampled_az.yaml",
+         ["system.num_simulations=4", "system.num_sampled_actions=4",
+          "system.num_minibatches=2", "system.epochs=1"]),
+    ]:
+        run = importlib.import_module(mod_path).run
+        cfg = compose(default,
+            ["arch.total_num_envs=32", "arch.total_timesteps=null",
+             "arch.num_updates=2", "arch.num_evaluation=1",
+             "arch.num_eval_episodes=16", "arch.absolute_metric=false",
+             "system.rollout_length=8", "logger.loggers=[]",
+             "logger.checkpointing.save_model=false"] + extra)
+        r = run(cfg)
+        assert r == r, mod_path
