@@ -726,3 +726,25 @@ def test_search_systems_eval_batch_mismatch():
              "logger.checkpointing.save_model=false"] + extra)
         r = run(cfg)
         assert r == r, mod_path
+
+
+@pytest.mark.slow
+def test_mz_learns_identity_above_chance():
+    """MuZero learning gate: with the LEARNED world model (3-step unroll,
+    two-hot values) it clearly beats chance (2.5) on the identity game —
+    measured 5.0 at this budget."""
+    from stoix_amd.systems.search.ff_mz import run
+
+    cfg = compose(
+        "default/anakin/default_ff_mz.yaml",
+        ["env=debug/identity", "arch.total_num_envs=32",
+         "arch.total_timesteps=null", "arch.num_updates=500",
+         "arch.num_evaluation=1", "arch.num_eval_episodes=16",
+         "arch.absolute_metric=false", "system.rollout_length=8",
+         "system.num_simulations=8", "system.unroll_steps=3",
+         "system.n_step=3", "system.batch_size=128",
+         "system.buffer_size=8192", "system.epochs=2",
+         "logger.loggers=[]", "logger.checkpointing.save_model=false"],
+    )
+    r = run(cfg)
+    assert r > 3.5, f"MuZero below the learning bar: {r}"
