@@ -78,11 +78,18 @@ class DDPGFamilyLearner:
 
     @torch.no_grad()
     def act_fn(self, obs: Tensor, greedy: bool) -> Tensor:
+        """EVALUATION action: deterministic either way — the reference's
+        deterministic policies "sample" to their mean at eval
+        (exploration noise belongs to the ROLLOUT path only,
+        _explore_action); greedy is accepted for evaluator-API parity."""
+        del greedy
+        return self._actor_action(self.actor, obs)
+
+    @torch.no_grad()
+    def _explore_action(self, obs: Tensor) -> Tensor:
         a = self._actor_action(self.actor, obs)
-        if not greedy:
-            noise = torch.randn(a.shape, device=a.device, generator=self.gen) * self.exploration_sigma
-            a = (a + noise).clamp(self.act_min, self.act_max)
-        return a
+        noise = torch.randn(a.shape, device=a.device, generator=self.gen) * self.exploration_sigma
+        return (a + noise).clamp(self.act_min, self.act_max)
 
     # -------------------------------------------------------------- rollout
 
@@ -94,7 +101,7 @@ class DDPGFamilyLearner:
             if random_actions:
                 action = self.env.action_space.sample(self.B, self.device, self.gen)
             else:
-                action = self.act_fn(obs, greedy=False)
+                action = self._explore_action(obs)
             next_ts = self.env.step(action)
             self.buffer.add(
                 {
