@@ -226,3 +226,40 @@ def test_flat_grad_reducer_four_ranks():
     for r in range(1, 4):
         for a, b in zip(grads[0], grads[r]):
             torch.testing.assert_close(a, b)
+
+
+def _welford_worker(rank: int, port: int, outdir: str):
+    try:
+        _init(rank, port)
+        from stoix_amd.ops import running_statistics as rs
+
+        torch.manual_seed(500 + rank)  # DIFFERENT data per rank
+        chunks = [torch.randn(50, 6) * (rank + 1) + rank for _ in range(3)]
+        state = rs.init_state((6,))
+        for c in chunks:
+            state = rs.update(state, c, all_reduce=True)
+        torch.save(
+            ((state.mean.clone(), state.std.clone(), state.count.clone()),
+             torch.cat(chunks)),
+            os.path.join(outdir, f"r{rank}.pt"),
+        )
+    except Exception as e:  # pragma: no cover
+        torch.save(("ERROR", repr(e)), os.path.join(outdir, f"r{rank}.pt"))
+
+
+def test_welford_allreduce_matches_global_statistics():
+    """The distributed Welford path (all_reduce=True: partial sums merged
+    across ranks each update, reference running_statistics.py:297-310)
+    must equal the numpy statistics of the CONCATENATED data from every
+    rank — and both ranks must hold identical states."""
+    results = _run_workers(_welford_worker, 29874)
+    (m0, s0, c0), data0 = results[0]
+    (m1, s1, c1), data1 = results[1]
+    torch.testing.assert_close(m0, m1)
+    torch.testing.assert_close(s0, s1)
+    all_data = torch.cat([data0, data1])
+    assert float(c0) == all_data.shape[0]
+    torch.testing.assert_close(m0, all_data.mean(0), rtol=1e-4, atol=1e-5)
+    torch.testing.assert_close(
+        s0, all_data.std(0, unbiased=False), rtol=1e-3, atol=1e-4
+    )
