@@ -884,3 +884,25 @@ def test_checkpointer_keep_period_and_save_interval(tmp_path):
     )
     # 20 and 40 survive via keep_period (multiples of 20); 40 is also latest
     assert kept == [20, 40], kept
+
+
+def test_polyak_update_foreach_semantics():
+    """polyak_update(src, dst, tau): dst <- tau*src + (1-tau)*dst, exactly,
+    src untouched (the _foreach fused helper every off-policy system's
+    target nets rely on)."""
+    import torch
+
+    from stoix_amd.parallel.dist import polyak_update
+
+    torch.manual_seed(0)
+    src = torch.nn.Linear(6, 5)
+    dst = torch.nn.Linear(6, 5)
+    src_before = [p.clone() for p in src.parameters()]
+    dst_before = [p.clone() for p in dst.parameters()]
+    tau = 0.25
+    with torch.no_grad():
+        polyak_update(src.parameters(), dst.parameters(), tau)
+    for p, pb in zip(src.parameters(), src_before):
+        torch.testing.assert_close(p, pb)  # source unchanged
+    for p, sb, db in zip(dst.parameters(), src_before, dst_before):
+        torch.testing.assert_close(p, tau * sb + (1 - tau) * db)
